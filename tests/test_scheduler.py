@@ -1,0 +1,254 @@
+"""Scheduler unit tests (CPU-only), mirroring the reference's
+tests/v1/core/test_scheduler.py factory pattern (create_scheduler)."""
+
+import pytest
+
+from vllm_amd.config import (
+    CacheConfig,
+    DeviceConfig,
+    EngineConfig,
+    ModelConfig,
+    ParallelConfig,
+    SchedulerConfig,
+)
+from vllm_amd.core.scheduler import Scheduler
+from vllm_amd.core.sched_output import ModelRunnerOutput
+from vllm_amd.request import Request, RequestStatus
+from vllm_amd.sampling_params import SamplingParams
+
+
+def create_scheduler(
+    max_num_batched_tokens=256,
+    max_num_seqs=8,
+    num_gpu_blocks=128,
+    block_size=16,
+    enable_prefix_caching=True,
+    enable_chunked_prefill=True,
+    max_model_len=2048,
+):
+    cfg = EngineConfig(
+        model_config=ModelConfig(model="tiny-llama", dtype="fp32",
+                                 max_model_len=max_model_len),
+        cache_config=CacheConfig(
+            block_size=block_size,
+            num_gpu_blocks=num_gpu_blocks,
+            enable_prefix_caching=enable_prefix_caching,
+        ),
+        scheduler_config=SchedulerConfig(
+            max_num_batched_tokens=max_num_batched_tokens,
+            max_num_seqs=max_num_seqs,
+            enable_chunked_prefill=enable_chunked_prefill,
+        ),
+        parallel_config=ParallelConfig(),
+        device_config=DeviceConfig(device="cpu"),
+    )
+    return Scheduler(cfg, num_gpu_blocks=num_gpu_blocks)
+
+
+def make_request(req_id, num_tokens=32, max_tokens=16, prompt=None):
+    tokens = prompt if prompt is not None else list(range(num_tokens))
+    return Request(
+        request_id=req_id,
+        prompt_token_ids=tokens,
+        sampling_params=SamplingParams(max_tokens=max_tokens),
+        eos_token_id=0,
+    )
+
+
+def fake_runner_output(sched_out, requests, next_token=7):
+    """Emulates the model runner: sample a token for every request whose
+    scheduled extent reaches its total token count."""
+    req_ids, sampled = [], []
+    for req_id, n in sched_out.num_scheduled_tokens.items():
+        req = requests[req_id]
+        req_ids.append(req_id)
+        if req.num_computed_tokens + n >= req.num_tokens:
+            sampled.append([next_token])
+        else:
+            sampled.append([])
+    return ModelRunnerOutput(req_ids=req_ids, sampled_token_ids=sampled)
+
+
+class TestBasicScheduling:
+    def test_single_request_lifecycle(self):
+        sched = create_scheduler()
+        req = make_request("r1", num_tokens=32, max_tokens=4)
+        sched.add_request(req)
+
+        # Step 1: full prefill + first sampled token.
+        out = sched.schedule()
+        assert out.total_num_scheduled_tokens == 32
+        assert len(out.scheduled_new_reqs) == 1
+        ro = fake_runner_output(out, sched.requests)
+        eco = sched.update_from_output(out, ro)
+        assert len(eco) == 1 and eco[0].new_token_ids == [7]
+        assert req.num_computed_tokens == 32
+        assert req.num_output_tokens == 1
+
+        # Steps 2-4: decode one token each.
+        for i in range(3):
+            out = sched.schedule()
+            assert out.num_scheduled_tokens["r1"] == 1
+            assert out.scheduled_cached_reqs.num_reqs == 1
+            eco = sched.update_from_output(
+                out, fake_runner_output(out, sched.requests)
+            )
+        assert req.status == RequestStatus.FINISHED_LENGTH_CAPPED
+        assert eco[0].finish_reason == "length"
+        assert not sched.has_unfinished_requests()
+
+    def test_eos_stop(self):
+        sched = create_scheduler()
+        req = make_request("r1", num_tokens=16, max_tokens=100)
+        sched.add_request(req)
+        out = sched.schedule()
+        ro = fake_runner_output(out, sched.requests, next_token=0)  # eos
+        eco = sched.update_from_output(out, ro)
+        assert eco[0].finish_reason == "stop"
+
+    def test_ignore_eos(self):
+        sched = create_scheduler()
+        req = Request(
+            "r1", list(range(16)),
+            SamplingParams(max_tokens=4, ignore_eos=True), eos_token_id=0,
+        )
+        sched.add_request(req)
+        for _ in range(4):
+            out = sched.schedule()
+            eco = sched.update_from_output(
+                out, fake_runner_output(out, sched.requests, next_token=0)
+            )
+        assert req.status == RequestStatus.FINISHED_LENGTH_CAPPED
+
+    def test_stop_token_ids(self):
+        sched = create_scheduler()
+        req = Request(
+            "r1", list(range(16)),
+            SamplingParams(max_tokens=100, stop_token_ids=[42]),
+            eos_token_id=0,
+        )
+        sched.add_request(req)
+        out = sched.schedule()
+        eco = sched.update_from_output(
+            out, fake_runner_output(out, sched.requests, next_token=42)
+        )
+        assert eco[0].finish_reason == "stop"
+        assert eco[0].stop_reason == 42
+
+
+class TestChunkedPrefill:
+    def test_chunking(self):
+        sched = create_scheduler(max_num_batched_tokens=64)
+        req = make_request("r1", num_tokens=150, max_tokens=2)
+        sched.add_request(req)
+
+        out = sched.schedule()
+        assert out.num_scheduled_tokens["r1"] == 64
+        eco = sched.update_from_output(
+            out, fake_runner_output(out, sched.requests)
+        )
+        assert eco == []  # no token sampled mid-prefill
+        out = sched.schedule()
+        assert out.num_scheduled_tokens["r1"] == 64
+        sched.update_from_output(out, fake_runner_output(out, sched.requests))
+        out = sched.schedule()
+        assert out.num_scheduled_tokens["r1"] == 22  # 150 - 128
+        eco = sched.update_from_output(
+            out, fake_runner_output(out, sched.requests)
+        )
+        assert len(eco) == 1 and eco[0].new_token_ids == [7]
+
+    def test_budget_shared_across_requests(self):
+        sched = create_scheduler(max_num_batched_tokens=100)
+        sched.add_request(make_request("r1", prompt=list(range(60))))
+        sched.add_request(make_request("r2", prompt=list(range(100, 160))))
+        out = sched.schedule()
+        assert out.num_scheduled_tokens["r1"] == 60
+        assert out.num_scheduled_tokens["r2"] == 40  # chunked
+        assert out.total_num_scheduled_tokens == 100
+
+
+class TestPreemption:
+    def test_preempt_and_resume(self):
+        # Tiny pool: 8 blocks of 16 = 128 tokens of KV.
+        sched = create_scheduler(
+            num_gpu_blocks=8, max_num_batched_tokens=64,
+            enable_prefix_caching=False,
+        )
+        r1 = make_request("r1", num_tokens=60, max_tokens=100)
+        r2 = make_request("r2", num_tokens=60, max_tokens=100)
+        sched.add_request(r1)
+        sched.add_request(r2)
+
+        # Both prefill (60+4 rounds to 4 blocks each; pool fits both barely).
+        out = sched.schedule()
+        sched.update_from_output(out, fake_runner_output(out, sched.requests))
+        out = sched.schedule()
+        sched.update_from_output(out, fake_runner_output(out, sched.requests))
+
+        # Decode until one request needs a block that isn't there.
+        preempted = False
+        for _ in range(40):
+            out = sched.schedule()
+            if r2.status == RequestStatus.PREEMPTED or r1.status == RequestStatus.PREEMPTED:
+                preempted = True
+                break
+            sched.update_from_output(
+                out, fake_runner_output(out, sched.requests)
+            )
+        assert preempted
+        # The preempted request sits in waiting and can resume once the
+        # other finishes.
+        victim = r2 if r2.status == RequestStatus.PREEMPTED else r1
+        survivor = r1 if victim is r2 else r2
+        sched.finish_requests([survivor.request_id],
+                              RequestStatus.FINISHED_ABORTED)
+        out = sched.schedule()
+        assert victim.request_id in out.num_scheduled_tokens
+        assert victim.status == RequestStatus.RUNNING
+
+
+class TestPrefixCachingIntegration:
+    def test_shared_prefix_schedules_less(self):
+        sched = create_scheduler(max_num_batched_tokens=1024)
+        prompt = list(range(64))
+        r1 = make_request("r1", prompt=prompt, max_tokens=2)
+        sched.add_request(r1)
+        out = sched.schedule()
+        assert out.num_scheduled_tokens["r1"] == 64
+        sched.update_from_output(out, fake_runner_output(out, sched.requests))
+
+        r2 = make_request("r2", prompt=list(prompt), max_tokens=2)
+        sched.add_request(r2)
+        out = sched.schedule()
+        # 64 tokens, 4 full blocks cached but last token recomputed -> 16.
+        assert out.num_scheduled_tokens["r2"] == 16
+        new_req = [n for n in out.scheduled_new_reqs if n.req_id == "r2"][0]
+        assert new_req.num_computed_tokens == 48
+
+
+class TestMaxSeqs:
+    def test_seat_limit(self):
+        sched = create_scheduler(max_num_seqs=2)
+        for i in range(4):
+            sched.add_request(make_request(f"r{i}", num_tokens=16))
+        out = sched.schedule()
+        assert len(out.scheduled_new_reqs) == 2
+        assert len(sched.waiting) == 2
+
+
+class TestAbort:
+    def test_abort_running(self):
+        sched = create_scheduler()
+        req = make_request("r1", num_tokens=16, max_tokens=100)
+        sched.add_request(req)
+        out = sched.schedule()
+        sched.update_from_output(out, fake_runner_output(out, sched.requests))
+        sched.finish_requests(["r1"], RequestStatus.FINISHED_ABORTED)
+        assert not sched.has_unfinished_requests()
+        # Blocks all freed.
+        assert (sched.kv_cache_manager.block_pool.get_num_free_blocks()
+                == sched.kv_cache_manager.block_pool.num_gpu_blocks)
+        # Next schedule() reports it finished so the runner can clear state.
+        out = sched.schedule()
+        assert "r1" in out.finished_req_ids

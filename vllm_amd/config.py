@@ -1,0 +1,353 @@
+"""Engine configuration.
+
+One composite ``EngineConfig`` threads through every constructor, same
+discipline as the reference's VllmConfig (vllm/config/vllm.py:331).
+Model architectures are described by an explicit ``ModelSpec`` so the
+engine runs fully offline: built-in presets cover the benchmark configs
+(BASELINE.json), and any local HF ``config.json`` can also be loaded.
+"""
+
+from __future__ import annotations
+
+import json
+import os
+from dataclasses import dataclass, field
+from typing import Optional
+
+import torch
+
+
+@dataclass
+class ModelSpec:
+    """Explicit architecture description (replaces network HF config fetch)."""
+
+    name: str = "llama-3-8b"
+    architecture: str = "llama"  # llama | opt | mixtral | deepseek
+    vocab_size: int = 128256
+    hidden_size: int = 4096
+    intermediate_size: int = 14336
+    num_layers: int = 32
+    num_heads: int = 32
+    num_kv_heads: int = 8
+    head_dim: int = 128
+    rope_theta: float = 500000.0
+    rope_scaling: Optional[dict] = None
+    rms_norm_eps: float = 1e-5
+    max_position_embeddings: int = 8192
+    tie_word_embeddings: bool = False
+    # OPT-style specifics
+    activation: str = "silu"  # silu | relu | gelu
+    use_layernorm: bool = False  # OPT uses LayerNorm, llama uses RMSNorm
+    use_bias: bool = False
+    # MoE specifics (mixtral / deepseek)
+    num_experts: int = 0
+    num_experts_per_tok: int = 0
+    moe_intermediate_size: int = 0
+    num_shared_experts: int = 0
+    first_dense_layers: int = 0  # deepseek: leading dense layers
+    routed_scaling_factor: float = 1.0
+    norm_topk_prob: bool = True
+    n_group: int = 0  # deepseek group-limited routing
+    topk_group: int = 0
+    scoring_func: str = "softmax"  # softmax | sigmoid
+    # MLA specifics (deepseek)
+    q_lora_rank: int = 0
+    kv_lora_rank: int = 0
+    qk_nope_head_dim: int = 0
+    qk_rope_head_dim: int = 0
+    v_head_dim: int = 0
+    eos_token_id: int = 2
+    bos_token_id: int = 1
+
+    @property
+    def is_moe(self) -> bool:
+        return self.num_experts > 0
+
+    @property
+    def is_mla(self) -> bool:
+        return self.kv_lora_rank > 0
+
+    def num_kv_heads_per_rank(self, tp_size: int) -> int:
+        if self.num_kv_heads >= tp_size:
+            assert self.num_kv_heads % tp_size == 0
+            return self.num_kv_heads // tp_size
+        # KV heads replicated when tp > num_kv_heads.
+        assert tp_size % self.num_kv_heads == 0
+        return 1
+
+
+# Built-in presets matching BASELINE.json's named configs.
+MODEL_PRESETS: dict[str, ModelSpec] = {
+    "opt-125m": ModelSpec(
+        name="opt-125m",
+        architecture="opt",
+        vocab_size=50272,
+        hidden_size=768,
+        intermediate_size=3072,
+        num_layers=12,
+        num_heads=12,
+        num_kv_heads=12,
+        head_dim=64,
+        max_position_embeddings=2048,
+        activation="relu",
+        use_layernorm=True,
+        use_bias=True,
+        tie_word_embeddings=True,
+        rms_norm_eps=1e-5,
+        eos_token_id=2,
+    ),
+    "llama-3-8b": ModelSpec(
+        name="llama-3-8b",
+        architecture="llama",
+        vocab_size=128256,
+        hidden_size=4096,
+        intermediate_size=14336,
+        num_layers=32,
+        num_heads=32,
+        num_kv_heads=8,
+        head_dim=128,
+        rope_theta=500000.0,
+        max_position_embeddings=8192,
+        eos_token_id=128001,
+        bos_token_id=128000,
+    ),
+    "llama-3-70b": ModelSpec(
+        name="llama-3-70b",
+        architecture="llama",
+        vocab_size=128256,
+        hidden_size=8192,
+        intermediate_size=28672,
+        num_layers=80,
+        num_heads=64,
+        num_kv_heads=8,
+        head_dim=128,
+        rope_theta=500000.0,
+        max_position_embeddings=8192,
+        eos_token_id=128001,
+        bos_token_id=128000,
+    ),
+    "mixtral-8x7b": ModelSpec(
+        name="mixtral-8x7b",
+        architecture="mixtral",
+        vocab_size=32000,
+        hidden_size=4096,
+        intermediate_size=14336,
+        num_layers=32,
+        num_heads=32,
+        num_kv_heads=8,
+        head_dim=128,
+        rope_theta=1000000.0,
+        max_position_embeddings=32768,
+        num_experts=8,
+        num_experts_per_tok=2,
+        moe_intermediate_size=14336,
+        eos_token_id=2,
+    ),
+    # DeepSeek-V3-style MLA + MoE, scaled-down layer count fits 1 GPU bf16.
+    "deepseek-v3-lite": ModelSpec(
+        name="deepseek-v3-lite",
+        architecture="deepseek",
+        vocab_size=129280,
+        hidden_size=7168,
+        intermediate_size=18432,
+        num_layers=8,
+        num_heads=128,
+        num_kv_heads=128,
+        head_dim=192,
+        rope_theta=10000.0,
+        max_position_embeddings=16384,
+        num_experts=64,
+        num_experts_per_tok=8,
+        moe_intermediate_size=2048,
+        num_shared_experts=1,
+        first_dense_layers=1,
+        routed_scaling_factor=2.5,
+        scoring_func="sigmoid",
+        n_group=8,
+        topk_group=4,
+        q_lora_rank=1536,
+        kv_lora_rank=512,
+        qk_nope_head_dim=128,
+        qk_rope_head_dim=64,
+        v_head_dim=128,
+        eos_token_id=1,
+    ),
+    # Tiny models for tests.
+    "tiny-llama": ModelSpec(
+        name="tiny-llama",
+        architecture="llama",
+        vocab_size=1024,
+        hidden_size=128,
+        intermediate_size=256,
+        num_layers=2,
+        num_heads=4,
+        num_kv_heads=2,
+        head_dim=32,
+        rope_theta=10000.0,
+        max_position_embeddings=2048,
+        eos_token_id=2,
+    ),
+    "tiny-mixtral": ModelSpec(
+        name="tiny-mixtral",
+        architecture="mixtral",
+        vocab_size=1024,
+        hidden_size=128,
+        intermediate_size=256,
+        num_layers=2,
+        num_heads=4,
+        num_kv_heads=2,
+        head_dim=32,
+        rope_theta=10000.0,
+        max_position_embeddings=2048,
+        num_experts=4,
+        num_experts_per_tok=2,
+        moe_intermediate_size=256,
+        eos_token_id=2,
+    ),
+}
+
+
+def _spec_from_hf_config(path: str) -> ModelSpec:
+    """Build a ModelSpec from a local HF config.json directory/file."""
+    cfg_path = path if path.endswith(".json") else os.path.join(path, "config.json")
+    with open(cfg_path) as f:
+        hf = json.load(f)
+    arch = (hf.get("architectures") or ["LlamaForCausalLM"])[0].lower()
+    if "opt" in arch:
+        architecture = "opt"
+    elif "mixtral" in arch:
+        architecture = "mixtral"
+    elif "deepseek" in arch:
+        architecture = "deepseek"
+    else:
+        architecture = "llama"
+    hidden = hf["hidden_size"]
+    heads = hf["num_attention_heads"]
+    return ModelSpec(
+        name=os.path.basename(path.rstrip("/")),
+        architecture=architecture,
+        vocab_size=hf["vocab_size"],
+        hidden_size=hidden,
+        intermediate_size=hf.get("intermediate_size", 4 * hidden),
+        num_layers=hf.get("num_hidden_layers", hf.get("num_layers")),
+        num_heads=heads,
+        num_kv_heads=hf.get("num_key_value_heads", heads),
+        head_dim=hf.get("head_dim", hidden // heads),
+        rope_theta=hf.get("rope_theta", 10000.0),
+        rope_scaling=hf.get("rope_scaling"),
+        rms_norm_eps=hf.get("rms_norm_eps", 1e-5),
+        max_position_embeddings=hf.get("max_position_embeddings", 8192),
+        tie_word_embeddings=hf.get("tie_word_embeddings", False),
+        num_experts=hf.get("num_local_experts", hf.get("n_routed_experts", 0)) or 0,
+        num_experts_per_tok=hf.get("num_experts_per_tok", 0) or 0,
+        moe_intermediate_size=hf.get("moe_intermediate_size", 0) or 0,
+        eos_token_id=hf.get("eos_token_id", 2) or 2,
+        bos_token_id=hf.get("bos_token_id", 1) or 1,
+    )
+
+
+def get_model_spec(model: str) -> ModelSpec:
+    if model in MODEL_PRESETS:
+        return MODEL_PRESETS[model]
+    if os.path.exists(model):
+        return _spec_from_hf_config(model)
+    raise ValueError(
+        f"Unknown model {model!r}. Available presets: {sorted(MODEL_PRESETS)} "
+        "or a local directory containing config.json."
+    )
+
+
+_STR_TO_DTYPE = {
+    "bf16": torch.bfloat16,
+    "bfloat16": torch.bfloat16,
+    "fp16": torch.float16,
+    "half": torch.float16,
+    "float16": torch.float16,
+    "fp32": torch.float32,
+    "float32": torch.float32,
+}
+
+
+@dataclass
+class ModelConfig:
+    model: str = "llama-3-8b"
+    tokenizer: Optional[str] = None  # local path, or None -> mock tokenizer
+    dtype: str = "bf16"
+    max_model_len: int = 8192
+    load_format: str = "dummy"  # dummy | safetensors
+    model_path: Optional[str] = None  # weights dir for safetensors
+    seed: int = 0
+    enforce_eager: bool = False
+    spec: ModelSpec = None  # type: ignore[assignment]
+
+    def __post_init__(self) -> None:
+        if self.spec is None:
+            self.spec = get_model_spec(self.model)
+        self.max_model_len = min(
+            self.max_model_len, self.spec.max_position_embeddings
+        )
+
+    @property
+    def torch_dtype(self) -> torch.dtype:
+        return _STR_TO_DTYPE[self.dtype]
+
+
+@dataclass
+class CacheConfig:
+    block_size: int = 16
+    gpu_memory_utilization: float = 0.90
+    num_gpu_blocks: Optional[int] = None  # None -> profile at startup
+    enable_prefix_caching: bool = True
+    kv_cache_dtype: str = "auto"  # auto | bf16 | fp8
+
+
+@dataclass
+class SchedulerConfig:
+    max_num_batched_tokens: int = 8192
+    max_num_seqs: int = 256
+    enable_chunked_prefill: bool = True
+    long_prefill_token_threshold: int = 0  # 0 -> no cap beyond token budget
+    policy: str = "fcfs"  # fcfs | priority
+
+
+@dataclass
+class ParallelConfig:
+    tensor_parallel_size: int = 1
+    pipeline_parallel_size: int = 1
+    data_parallel_size: int = 1
+    enable_expert_parallel: bool = False
+    # Filled from env (RANK/LOCAL_RANK/WORLD_SIZE) when launched by torchrun.
+    rank: int = 0
+    local_rank: int = 0
+    world_size: int = 1
+    distributed_backend: str = "auto"  # auto -> nccl(RCCL) on GPU, gloo on CPU
+
+    @property
+    def needs_distributed(self) -> bool:
+        return self.world_size > 1 or self.tensor_parallel_size > 1
+
+
+@dataclass
+class DeviceConfig:
+    device: str = "auto"  # auto | cuda | cpu
+
+    def __post_init__(self) -> None:
+        if self.device == "auto":
+            self.device = "cuda" if torch.cuda.is_available() else "cpu"
+
+
+@dataclass
+class EngineConfig:
+    model_config: ModelConfig = field(default_factory=ModelConfig)
+    cache_config: CacheConfig = field(default_factory=CacheConfig)
+    scheduler_config: SchedulerConfig = field(default_factory=SchedulerConfig)
+    parallel_config: ParallelConfig = field(default_factory=ParallelConfig)
+    device_config: DeviceConfig = field(default_factory=DeviceConfig)
+
+    def __post_init__(self) -> None:
+        spec = self.model_config.spec
+        tp = self.parallel_config.tensor_parallel_size
+        if spec.num_heads % tp != 0:
+            raise ValueError(
+                f"num_heads={spec.num_heads} not divisible by tp={tp}"
+            )

@@ -1,0 +1,141 @@
+"""Per-request paged KV block accounting (role of vllm/v1/core/kv_cache_manager.py:118).
+
+Single KV-cache group (full attention) in v1; the structure leaves room
+for hybrid groups (sliding window / MLA) later.
+"""
+
+from __future__ import annotations
+
+from collections import defaultdict
+from typing import Optional
+
+from vllm_amd.core.block_pool import BlockPool
+from vllm_amd.core.kv_cache_utils import KVCacheBlock, hash_request_tokens
+from vllm_amd.request import Request
+
+
+class KVCacheManager:
+
+    def __init__(
+        self,
+        num_gpu_blocks: int,
+        block_size: int,
+        enable_caching: bool = True,
+    ) -> None:
+        self.block_size = block_size
+        self.enable_caching = enable_caching
+        self.block_pool = BlockPool(num_gpu_blocks, enable_caching)
+        self.req_to_blocks: dict[str, list[KVCacheBlock]] = defaultdict(list)
+        # How many blocks of each request are already content-cached.
+        self.num_cached_blocks: dict[str, int] = defaultdict(int)
+
+    @property
+    def usage(self) -> float:
+        return self.block_pool.get_usage()
+
+    def get_computed_blocks(
+        self, request: Request
+    ) -> tuple[list[KVCacheBlock], int]:
+        """Prefix-cache lookup: longest chain of cached full blocks matching
+        this request's prompt. Returns (blocks, num_computed_tokens)."""
+        if not self.enable_caching:
+            return [], 0
+        request.block_hashes = hash_request_tokens(
+            self.block_size,
+            request.all_token_ids,
+            prior_hashes=request.block_hashes,
+        )
+        computed: list[KVCacheBlock] = []
+        for h in request.block_hashes:
+            block = self.block_pool.get_cached_block(h)
+            if block is None:
+                break
+            computed.append(block)
+        num_computed = len(computed) * self.block_size
+        # Never report the full prompt as computed: at least the last token
+        # must be recomputed to produce logits.
+        if num_computed >= request.num_tokens:
+            computed.pop()
+            num_computed -= self.block_size
+        return computed, num_computed
+
+    def allocate_slots(
+        self,
+        request: Request,
+        num_new_tokens: int,
+        new_computed_blocks: Optional[list[KVCacheBlock]] = None,
+    ) -> Optional[list[KVCacheBlock]]:
+        """Allocate blocks so the request can hold
+        num_computed_tokens + len(new_computed_blocks)*bs + num_new_tokens.
+
+        Returns newly allocated blocks (excluding cache hits), or None if
+        the pool cannot satisfy the request (caller should preempt).
+        """
+        assert num_new_tokens > 0
+        new_computed_blocks = new_computed_blocks or []
+        req_blocks = self.req_to_blocks[request.request_id]
+
+        num_computed_tokens = request.num_computed_tokens + len(
+            new_computed_blocks
+        ) * self.block_size
+        total_tokens = num_computed_tokens + num_new_tokens
+        num_required_blocks = (
+            total_tokens + self.block_size - 1
+        ) // self.block_size
+        num_new_blocks = (
+            num_required_blocks - len(req_blocks) - len(new_computed_blocks)
+        )
+
+        if num_new_blocks > self.block_pool.get_num_free_blocks():
+            return None
+
+        # Commit cache hits (bump refs) only after we know allocation fits.
+        if new_computed_blocks:
+            self.block_pool.touch(new_computed_blocks)
+            req_blocks.extend(new_computed_blocks)
+            request.num_cached_tokens = (
+                len(new_computed_blocks) * self.block_size
+            )
+            # Hit blocks are already content-cached.
+            self.num_cached_blocks[request.request_id] = len(req_blocks)
+
+        if num_new_blocks <= 0:
+            new_blocks: list[KVCacheBlock] = []
+        else:
+            new_blocks = self.block_pool.get_new_blocks(num_new_blocks)
+            req_blocks.extend(new_blocks)
+
+        # Content-cache the blocks that become full after this step.
+        if self.enable_caching:
+            num_full_after = total_tokens // self.block_size
+            request.block_hashes = hash_request_tokens(
+                self.block_size,
+                request.all_token_ids,
+                prior_hashes=request.block_hashes,
+            )
+            # Only blocks whose tokens are all known can be hashed; with
+            # chunked prefill total_tokens <= num_tokens so this holds.
+            num_hashable = min(num_full_after, len(request.block_hashes))
+            cached = self.num_cached_blocks[request.request_id]
+            if num_hashable > cached:
+                self.block_pool.cache_full_blocks(
+                    req_blocks,
+                    request.block_hashes,
+                    cached,
+                    num_hashable,
+                )
+                self.num_cached_blocks[request.request_id] = num_hashable
+        return new_blocks
+
+    def free(self, request: Request) -> None:
+        blocks = self.req_to_blocks.pop(request.request_id, [])
+        self.num_cached_blocks.pop(request.request_id, None)
+        # Free in reverse so the tail blocks (least useful as prefix cache)
+        # are evicted first (LRU queue order).
+        self.block_pool.free_blocks(list(reversed(blocks)))
+
+    def get_block_ids(self, request_id: str) -> list[int]:
+        return [b.block_id for b in self.req_to_blocks[request_id]]
+
+    def reset_prefix_cache(self) -> bool:
+        return self.block_pool.reset_prefix_cache()

@@ -1,0 +1,356 @@
+"""Continuous-batching scheduler.
+
+Semantics follow the reference scheduler (vllm/v1/core/sched/scheduler.py:69):
+
+- No prefill/decode phases: each request's ``num_computed_tokens`` catches
+  up to ``num_tokens`` under a shared token budget (chunked prefill falls
+  out of the clamp).
+- RUNNING requests are scheduled first; on allocation failure the
+  lowest-priority (last) running request is preempted by evicting all its
+  blocks and requeuing it.
+- WAITING requests join while budget, seats, and KV blocks remain;
+  prefix-cache lookups skip already-cached prompt prefixes.
+- FCFS or priority policy.
+"""
+
+from __future__ import annotations
+
+import heapq
+from collections import deque
+from typing import Optional
+
+from vllm_amd.config import EngineConfig
+from vllm_amd.core.kv_cache_manager import KVCacheManager
+from vllm_amd.core.sched_output import (
+    CachedRequestData,
+    EngineCoreOutput,
+    ModelRunnerOutput,
+    NewRequestData,
+    SchedulerOutput,
+    SchedulerStats,
+)
+from vllm_amd.request import Request, RequestStatus
+
+
+class Scheduler:
+
+    def __init__(
+        self,
+        config: EngineConfig,
+        num_gpu_blocks: int,
+    ) -> None:
+        self.config = config
+        sched_cfg = config.scheduler_config
+        self.max_num_batched_tokens = sched_cfg.max_num_batched_tokens
+        self.max_num_seqs = sched_cfg.max_num_seqs
+        self.enable_chunked_prefill = sched_cfg.enable_chunked_prefill
+        self.long_prefill_token_threshold = sched_cfg.long_prefill_token_threshold
+        self.policy = sched_cfg.policy
+        self.max_model_len = config.model_config.max_model_len
+        self.block_size = config.cache_config.block_size
+
+        self.kv_cache_manager = KVCacheManager(
+            num_gpu_blocks=num_gpu_blocks,
+            block_size=config.cache_config.block_size,
+            enable_caching=config.cache_config.enable_prefix_caching,
+        )
+
+        self.requests: dict[str, Request] = {}
+        self.waiting: deque[Request] = deque()
+        self.running: list[Request] = []
+        # req_ids finished since the last schedule() call; the runner uses
+        # this to clear its persistent-batch rows.
+        self.finished_req_ids: set[str] = set()
+        # Stats
+        self.prefix_cache_queries = 0
+        self.prefix_cache_hits = 0
+
+    # ------------------------------------------------------------------
+    # Request admission / removal
+
+    def add_request(self, request: Request) -> None:
+        self.requests[request.request_id] = request
+        request.status = RequestStatus.WAITING
+        if self.policy == "priority":
+            # Maintain the waiting deque sorted by (priority, arrival).
+            items = list(self.waiting)
+            items.append(request)
+            items.sort()
+            self.waiting = deque(items)
+        else:
+            self.waiting.append(request)
+
+    def finish_requests(
+        self, request_ids: list[str], status: RequestStatus
+    ) -> None:
+        """Externally finish (abort) requests."""
+        for req_id in request_ids:
+            request = self.requests.get(req_id)
+            if request is None or request.is_finished():
+                continue
+            if request.status == RequestStatus.RUNNING:
+                self.running.remove(request)
+            else:
+                try:
+                    self.waiting.remove(request)
+                except ValueError:
+                    pass
+            request.status = status
+            self._free_request(request)
+
+    def has_unfinished_requests(self) -> bool:
+        return bool(self.waiting or self.running)
+
+    def get_num_unfinished_requests(self) -> int:
+        return len(self.waiting) + len(self.running)
+
+    # ------------------------------------------------------------------
+    # Scheduling
+
+    def schedule(self) -> SchedulerOutput:
+        token_budget = self.max_num_batched_tokens
+        scheduled_new_reqs: list[NewRequestData] = []
+        cached = CachedRequestData()
+        num_scheduled_tokens: dict[str, int] = {}
+        preempted_reqs: set[str] = set()
+
+        # ---- RUNNING loop (reference scheduler.py:484) ----
+        req_index = 0
+        while req_index < len(self.running) and token_budget > 0:
+            request = self.running[req_index]
+            num_new_tokens = request.num_tokens - request.num_computed_tokens
+            if self.long_prefill_token_threshold > 0:
+                num_new_tokens = min(
+                    num_new_tokens, self.long_prefill_token_threshold
+                )
+            num_new_tokens = min(num_new_tokens, token_budget)
+            # Don't run past the context window.
+            num_new_tokens = min(
+                num_new_tokens,
+                self.max_model_len - request.num_computed_tokens,
+            )
+            if num_new_tokens <= 0:
+                req_index += 1
+                continue
+
+            new_blocks = None
+            while True:
+                new_blocks = self.kv_cache_manager.allocate_slots(
+                    request, num_new_tokens
+                )
+                if new_blocks is not None:
+                    break
+                # Preempt the lowest-priority running request (the last one).
+                if self.policy == "priority":
+                    victim = max(
+                        self.running,
+                        key=lambda r: (r.priority, r.arrival_time),
+                    )
+                else:
+                    victim = self.running[-1]
+                self._preempt(victim)
+                preempted_reqs.add(victim.request_id)
+                if victim is request:
+                    break  # this very request was evicted
+            if request.request_id in preempted_reqs:
+                # Could not schedule it; it's back in waiting.
+                continue
+
+            cached.req_ids.append(request.request_id)
+            cached.resumed.append(False)
+            cached.new_block_ids.append([b.block_id for b in new_blocks])
+            cached.num_computed_tokens.append(request.num_computed_tokens)
+            # Tokens the runner hasn't seen yet: any output tokens generated
+            # since the prompt (runner keeps its own copy; for non-resumed
+            # requests only the newest token is unseen).
+            cached.new_token_ids.append(
+                request.all_token_ids[
+                    request.num_computed_tokens : request.num_computed_tokens
+                    + num_new_tokens
+                ]
+            )
+            num_scheduled_tokens[request.request_id] = num_new_tokens
+            token_budget -= num_new_tokens
+            req_index += 1
+
+        # ---- WAITING loop (reference scheduler.py:692) ----
+        while (
+            self.waiting
+            and token_budget > 0
+            and len(self.running) < self.max_num_seqs
+            and not preempted_reqs  # don't admit while evicting
+        ):
+            request = self.waiting[0]
+            resumed = request.status == RequestStatus.PREEMPTED
+
+            # Prefix-cache lookup for fresh requests.
+            new_computed_blocks = []
+            num_computed = request.num_computed_tokens
+            if num_computed == 0 and not resumed:
+                (
+                    new_computed_blocks,
+                    num_computed,
+                ) = self.kv_cache_manager.get_computed_blocks(request)
+                self.prefix_cache_queries += request.num_tokens
+                self.prefix_cache_hits += num_computed
+            elif resumed:
+                # Preemption dropped all blocks; try the prefix cache to
+                # recover whatever is still resident.
+                (
+                    new_computed_blocks,
+                    num_computed,
+                ) = self.kv_cache_manager.get_computed_blocks(request)
+
+            num_new_tokens = request.num_tokens - num_computed
+            if self.long_prefill_token_threshold > 0:
+                num_new_tokens = min(
+                    num_new_tokens, self.long_prefill_token_threshold
+                )
+            if num_new_tokens > token_budget:
+                if not self.enable_chunked_prefill:
+                    break
+                num_new_tokens = token_budget
+            if num_new_tokens <= 0:
+                break
+
+            new_blocks = self.kv_cache_manager.allocate_slots(
+                request, num_new_tokens, new_computed_blocks
+            )
+            if new_blocks is None:
+                break  # out of KV blocks; stop admitting
+
+            self.waiting.popleft()
+            self.running.append(request)
+            request.status = RequestStatus.RUNNING
+            request.num_computed_tokens = num_computed
+
+            all_block_ids = self.kv_cache_manager.get_block_ids(
+                request.request_id
+            )
+            if resumed:
+                cached.req_ids.append(request.request_id)
+                cached.resumed.append(True)
+                cached.new_block_ids.append(all_block_ids)
+                cached.num_computed_tokens.append(num_computed)
+                cached.new_token_ids.append(list(request.all_token_ids))
+            else:
+                scheduled_new_reqs.append(
+                    NewRequestData(
+                        req_id=request.request_id,
+                        prompt_token_ids=list(request.all_token_ids),
+                        block_ids=all_block_ids,
+                        num_computed_tokens=num_computed,
+                        sampling_params=request.sampling_params,
+                    )
+                )
+            num_scheduled_tokens[request.request_id] = num_new_tokens
+            token_budget -= num_new_tokens
+
+        total = sum(num_scheduled_tokens.values())
+        out = SchedulerOutput(
+            scheduled_new_reqs=scheduled_new_reqs,
+            scheduled_cached_reqs=cached,
+            num_scheduled_tokens=num_scheduled_tokens,
+            total_num_scheduled_tokens=total,
+            finished_req_ids=self.finished_req_ids,
+        )
+        self.finished_req_ids = set()
+        return out
+
+    def _preempt(self, request: Request) -> None:
+        self.running.remove(request)
+        self.kv_cache_manager.free(request)
+        request.status = RequestStatus.PREEMPTED
+        request.num_computed_tokens = 0
+        request.num_preemptions += 1
+        self.waiting.appendleft(request)
+
+    # ------------------------------------------------------------------
+    # Post-execution update
+
+    def update_from_output(
+        self,
+        scheduler_output: SchedulerOutput,
+        runner_output: ModelRunnerOutput,
+    ) -> list[EngineCoreOutput]:
+        outputs: list[EngineCoreOutput] = []
+        sampled_by_req = dict(
+            zip(runner_output.req_ids, runner_output.sampled_token_ids)
+        )
+        logprobs_by_req = runner_output.logprobs or {}
+
+        for req_id, num_sched in scheduler_output.num_scheduled_tokens.items():
+            request = self.requests.get(req_id)
+            if request is None or request.is_finished():
+                continue  # aborted mid-step
+            request.num_computed_tokens += num_sched
+
+            new_token_ids = sampled_by_req.get(req_id) or []
+            if not new_token_ids:
+                continue  # mid chunked-prefill, nothing sampled
+
+            stopped = False
+            kept_tokens: list[int] = []
+            for tok in new_token_ids:
+                kept_tokens.append(tok)
+                request.append_output_token_ids([tok])
+                stopped = self._check_stop(request, tok)
+                if stopped:
+                    break
+
+            out = EngineCoreOutput(
+                req_id=req_id,
+                new_token_ids=kept_tokens,
+                finish_reason=request.get_finished_reason(),
+                stop_reason=request.stop_reason,
+                num_cached_tokens=request.num_cached_tokens,
+                new_logprobs=logprobs_by_req.get(req_id),
+            )
+            outputs.append(out)
+            if stopped:
+                self.running.remove(request)
+                self._free_request(request)
+        return outputs
+
+    def _check_stop(self, request: Request, last_token: int) -> bool:
+        params = request.sampling_params
+        if (
+            request.num_output_tokens >= params.min_tokens
+            and not params.ignore_eos
+            and request.eos_token_id is not None
+            and last_token == request.eos_token_id
+        ):
+            request.status = RequestStatus.FINISHED_STOPPED
+            return True
+        if (
+            request.num_output_tokens >= params.min_tokens
+            and last_token in params.all_stop_token_ids
+        ):
+            request.status = RequestStatus.FINISHED_STOPPED
+            request.stop_reason = last_token
+            return True
+        if request.num_output_tokens >= request.max_tokens:
+            request.status = RequestStatus.FINISHED_LENGTH_CAPPED
+            return True
+        if request.num_tokens >= self.max_model_len:
+            request.status = RequestStatus.FINISHED_LENGTH_CAPPED
+            return True
+        return False
+
+    def _free_request(self, request: Request) -> None:
+        self.kv_cache_manager.free(request)
+        self.finished_req_ids.add(request.request_id)
+        # Keep the Request object in self.requests until the engine client
+        # drains its final output, then the engine calls release().
+
+    def release_request(self, req_id: str) -> None:
+        self.requests.pop(req_id, None)
+
+    def make_stats(self) -> SchedulerStats:
+        return SchedulerStats(
+            num_running_reqs=len(self.running),
+            num_waiting_reqs=len(self.waiting),
+            kv_cache_usage=self.kv_cache_manager.usage,
+            prefix_cache_queries=self.prefix_cache_queries,
+            prefix_cache_hits=self.prefix_cache_hits,
+        )

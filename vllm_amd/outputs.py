@@ -1,0 +1,38 @@
+"""User-facing request outputs (role of vllm/outputs.py in the reference)."""
+
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import Optional
+
+
+@dataclass
+class CompletionOutput:
+    index: int
+    text: str
+    token_ids: list[int]
+    cumulative_logprob: Optional[float] = None
+    logprobs: Optional[list[dict[int, float]]] = None
+    finish_reason: Optional[str] = None  # "stop" | "length" | "abort"
+    stop_reason: Optional[object] = None
+
+    @property
+    def finished(self) -> bool:
+        return self.finish_reason is not None
+
+
+@dataclass
+class RequestOutput:
+    request_id: str
+    prompt: Optional[str]
+    prompt_token_ids: list[int]
+    outputs: list[CompletionOutput]
+    finished: bool
+    metrics: Optional[dict] = None
+
+    def __repr__(self) -> str:
+        return (
+            f"RequestOutput(request_id={self.request_id!r}, "
+            f"finished={self.finished}, "
+            f"outputs={self.outputs!r})"
+        )
