@@ -1,0 +1,120 @@
+"""EngineArgs: dataclass <-> argparse bridge (role of vllm/engine/arg_utils.py:423)."""
+
+from __future__ import annotations
+
+import argparse
+from dataclasses import dataclass, fields
+from typing import Optional
+
+from vllm_amd.config import (
+    CacheConfig,
+    DeviceConfig,
+    EngineConfig,
+    ModelConfig,
+    ParallelConfig,
+    SchedulerConfig,
+)
+
+
+@dataclass
+class EngineArgs:
+    model: str = "llama-3-8b"
+    tokenizer: Optional[str] = None
+    dtype: str = "bf16"
+    max_model_len: int = 8192
+    load_format: str = "dummy"
+    model_path: Optional[str] = None
+    seed: int = 0
+    enforce_eager: bool = False
+
+    block_size: int = 16
+    gpu_memory_utilization: float = 0.90
+    num_gpu_blocks: Optional[int] = None
+    enable_prefix_caching: bool = True
+    kv_cache_dtype: str = "auto"
+
+    max_num_batched_tokens: int = 8192
+    max_num_seqs: int = 256
+    enable_chunked_prefill: bool = True
+    scheduling_policy: str = "fcfs"
+
+    tensor_parallel_size: int = 1
+    device: str = "auto"
+
+    @staticmethod
+    def add_cli_args(parser: argparse.ArgumentParser) -> argparse.ArgumentParser:
+        parser.add_argument("--model", type=str, default="llama-3-8b")
+        parser.add_argument("--tokenizer", type=str, default=None)
+        parser.add_argument("--dtype", type=str, default="bf16")
+        parser.add_argument("--max-model-len", type=int, default=8192)
+        parser.add_argument("--load-format", type=str, default="dummy",
+                            choices=["dummy", "safetensors"])
+        parser.add_argument("--model-path", type=str, default=None)
+        parser.add_argument("--seed", type=int, default=0)
+        parser.add_argument("--enforce-eager", action="store_true")
+        parser.add_argument("--block-size", type=int, default=16)
+        parser.add_argument("--gpu-memory-utilization", type=float,
+                            default=0.90)
+        parser.add_argument("--num-gpu-blocks", type=int, default=None)
+        parser.add_argument("--no-enable-prefix-caching",
+                            dest="enable_prefix_caching",
+                            action="store_false")
+        parser.add_argument("--kv-cache-dtype", type=str, default="auto")
+        parser.add_argument("--max-num-batched-tokens", type=int,
+                            default=8192)
+        parser.add_argument("--max-num-seqs", type=int, default=256)
+        parser.add_argument("--no-enable-chunked-prefill",
+                            dest="enable_chunked_prefill",
+                            action="store_false")
+        parser.add_argument("--scheduling-policy", type=str, default="fcfs",
+                            choices=["fcfs", "priority"])
+        parser.add_argument("--tensor-parallel-size", "-tp", type=int,
+                            default=1)
+        parser.add_argument("--device", type=str, default="auto")
+        return parser
+
+    @classmethod
+    def from_cli_args(cls, args: argparse.Namespace) -> "EngineArgs":
+        attrs = [f.name for f in fields(cls)]
+        kwargs = {a: getattr(args, a) for a in attrs if hasattr(args, a)}
+        if hasattr(args, "scheduling_policy"):
+            kwargs["scheduling_policy"] = args.scheduling_policy
+        return cls(**kwargs)
+
+    def create_engine_config(self) -> EngineConfig:
+        import os
+
+        world_size = int(os.environ.get("WORLD_SIZE", "1"))
+        pc = ParallelConfig(
+            tensor_parallel_size=self.tensor_parallel_size,
+            rank=int(os.environ.get("RANK", "0")),
+            local_rank=int(os.environ.get("LOCAL_RANK", "0")),
+            world_size=max(world_size, self.tensor_parallel_size),
+        )
+        return EngineConfig(
+            model_config=ModelConfig(
+                model=self.model,
+                tokenizer=self.tokenizer,
+                dtype=self.dtype,
+                max_model_len=self.max_model_len,
+                load_format=self.load_format,
+                model_path=self.model_path,
+                seed=self.seed,
+                enforce_eager=self.enforce_eager,
+            ),
+            cache_config=CacheConfig(
+                block_size=self.block_size,
+                gpu_memory_utilization=self.gpu_memory_utilization,
+                num_gpu_blocks=self.num_gpu_blocks,
+                enable_prefix_caching=self.enable_prefix_caching,
+                kv_cache_dtype=self.kv_cache_dtype,
+            ),
+            scheduler_config=SchedulerConfig(
+                max_num_batched_tokens=self.max_num_batched_tokens,
+                max_num_seqs=self.max_num_seqs,
+                enable_chunked_prefill=self.enable_chunked_prefill,
+                policy=self.scheduling_policy,
+            ),
+            parallel_config=pc,
+            device_config=DeviceConfig(device=self.device),
+        )

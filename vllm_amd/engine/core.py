@@ -1,0 +1,100 @@
+"""EngineCore: owns scheduler + worker(s); one step() = schedule ->
+execute -> update (role of vllm/v1/engine/core.py:103).
+
+Two launch modes:
+
+- Single process (tp=1): scheduler and worker in-proc.
+- SPMD (torchrun, world>1): every rank builds an EngineCore; rank 0 owns
+  the scheduler and broadcasts each SchedulerOutput over the CPU (gloo)
+  group; all ranks execute the model step collectively over RCCL/xGMI.
+  Non-zero ranks drive their loop via run_spmd_worker_loop().
+"""
+
+from __future__ import annotations
+
+import logging
+import time
+from typing import Optional
+
+import torch
+
+from vllm_amd.config import EngineConfig
+from vllm_amd.core.sched_output import EngineCoreOutput, SchedulerOutput
+from vllm_amd.core.scheduler import Scheduler
+from vllm_amd.request import Request, RequestStatus
+from vllm_amd.worker.worker import Worker
+
+logger = logging.getLogger(__name__)
+
+
+class EngineCore:
+
+    def __init__(self, config: EngineConfig):
+        self.config = config
+        self.worker = Worker(config)
+        self.worker.init_device()
+        self.worker.load_model()
+
+        from vllm_amd.parallel.state import get_world_group
+
+        self.world = get_world_group()
+        self.is_driver = self.world.rank_in_group == 0
+
+        # KV sizing must agree across ranks: take the min over ranks.
+        num_blocks = self.worker.determine_num_kv_blocks()
+        if self.world.world_size > 1:
+            t = torch.tensor([num_blocks], dtype=torch.int64)
+            if torch.cuda.is_available():
+                t = t.cuda()
+            torch.distributed.all_reduce(
+                t, op=torch.distributed.ReduceOp.MIN
+            )
+            num_blocks = int(t.item())
+        self.num_gpu_blocks = num_blocks
+        self.worker.initialize_kv_cache(num_blocks)
+
+        self.scheduler: Optional[Scheduler] = (
+            Scheduler(config, num_gpu_blocks=num_blocks)
+            if self.is_driver
+            else None
+        )
+
+    # ------------------------------------------------------------------
+    def add_request(self, request: Request) -> None:
+        assert self.is_driver
+        self.scheduler.add_request(request)
+
+    def abort_requests(self, request_ids: list[str]) -> None:
+        assert self.is_driver
+        self.scheduler.finish_requests(
+            request_ids, RequestStatus.FINISHED_ABORTED
+        )
+
+    def has_unfinished_requests(self) -> bool:
+        return self.scheduler is not None and \
+            self.scheduler.has_unfinished_requests()
+
+    # ------------------------------------------------------------------
+    def step(self) -> list[EngineCoreOutput]:
+        """One engine iteration on the driver rank."""
+        assert self.is_driver
+        scheduler_output = self.scheduler.schedule()
+        if self.world.world_size > 1:
+            self.world.broadcast_object(scheduler_output, src=0)
+        runner_output = self.worker.execute_model(scheduler_output)
+        return self.scheduler.update_from_output(
+            scheduler_output, runner_output
+        )
+
+    def run_spmd_worker_loop(self) -> None:
+        """Non-driver ranks: receive scheduler outputs forever."""
+        assert not self.is_driver
+        while True:
+            so = self.world.broadcast_object(None, src=0)
+            if so is None:  # shutdown sentinel
+                return
+            self.worker.execute_model(so)
+
+    def shutdown(self) -> None:
+        if self.is_driver and self.world.world_size > 1:
+            self.world.broadcast_object(None, src=0)

@@ -1,0 +1,225 @@
+"""Synchronous LLMEngine + output processing (detokenization, stop strings).
+
+Role of vllm/v1/engine/llm_engine.py and output_processor.py in one
+place: add_request -> step() -> RequestOutputs, with incremental
+detokenization and stop-string scanning on the client side of the engine.
+"""
+
+from __future__ import annotations
+
+import time
+from dataclasses import dataclass, field
+from typing import Optional, Union
+
+from vllm_amd.config import EngineConfig
+from vllm_amd.core.sched_output import EngineCoreOutput
+from vllm_amd.engine.core import EngineCore
+from vllm_amd.outputs import CompletionOutput, RequestOutput
+from vllm_amd.request import Request
+from vllm_amd.sampling_params import RequestOutputKind, SamplingParams
+from vllm_amd.tokenizer import IncrementalDetokenizer, TokenizerWrapper
+
+
+@dataclass
+class RequestState:
+    request_id: str
+    prompt: Optional[str]
+    prompt_token_ids: list[int]
+    params: SamplingParams
+    detokenizer: Optional[IncrementalDetokenizer]
+    output_token_ids: list[int] = field(default_factory=list)
+    logprobs: list = field(default_factory=list)
+    finished: bool = False
+    finish_reason: Optional[str] = None
+    stop_reason: Optional[object] = None
+    arrival_time: float = field(default_factory=time.time)
+    first_token_time: Optional[float] = None
+    finish_time: Optional[float] = None
+
+
+class OutputProcessor:
+    """EngineCoreOutput -> RequestOutput with stop-string handling."""
+
+    def __init__(self, tokenizer: TokenizerWrapper):
+        self.tokenizer = tokenizer
+        self.states: dict[str, RequestState] = {}
+
+    def add_request(
+        self,
+        request_id: str,
+        prompt: Optional[str],
+        prompt_token_ids: list[int],
+        params: SamplingParams,
+    ) -> None:
+        detok = (
+            IncrementalDetokenizer(
+                self.tokenizer, len(prompt_token_ids),
+                params.skip_special_tokens,
+            )
+            if params.detokenize
+            else None
+        )
+        self.states[request_id] = RequestState(
+            request_id=request_id,
+            prompt=prompt,
+            prompt_token_ids=prompt_token_ids,
+            params=params,
+            detokenizer=detok,
+        )
+
+    def process_outputs(
+        self, core_outputs: list[EngineCoreOutput]
+    ) -> tuple[list[RequestOutput], list[str]]:
+        """Returns (request_outputs, request_ids_to_abort)."""
+        outputs: list[RequestOutput] = []
+        to_abort: list[str] = []
+        now = time.time()
+        for co in core_outputs:
+            state = self.states.get(co.req_id)
+            if state is None or state.finished:
+                continue
+            if state.first_token_time is None:
+                state.first_token_time = now
+            state.output_token_ids.extend(co.new_token_ids)
+            if co.new_logprobs:
+                state.logprobs.extend(co.new_logprobs)
+            delta = ""
+            if state.detokenizer is not None:
+                delta = state.detokenizer.update(co.new_token_ids)
+            finish_reason = co.finish_reason
+            stop_reason = co.stop_reason
+
+            # Stop-string scan on the decoded text.
+            if state.params.stop and state.detokenizer is not None:
+                text = state.detokenizer.output_text
+                for s in state.params.stop:
+                    idx = text.find(s)
+                    if idx != -1:
+                        state.detokenizer.output_text = text[:idx]
+                        finish_reason = "stop"
+                        stop_reason = s
+                        if not co.finish_reason:
+                            to_abort.append(co.req_id)
+                        break
+
+            if finish_reason:
+                state.finished = True
+                state.finish_reason = finish_reason
+                state.stop_reason = stop_reason
+                state.finish_time = now
+            outputs.append(self._make_output(state, delta))
+        return outputs, to_abort
+
+    def _make_output(self, state: RequestState, delta: str) -> RequestOutput:
+        kind = state.params.output_kind
+        text = (
+            state.detokenizer.output_text
+            if state.detokenizer is not None
+            else ""
+        )
+        if kind == RequestOutputKind.DELTA:
+            text = delta
+        comp = CompletionOutput(
+            index=0,
+            text=text,
+            token_ids=list(state.output_token_ids),
+            logprobs=state.logprobs or None,
+            finish_reason=state.finish_reason,
+            stop_reason=state.stop_reason,
+        )
+        metrics = None
+        if state.finished:
+            metrics = {
+                "arrival_time": state.arrival_time,
+                "first_token_time": state.first_token_time,
+                "finish_time": state.finish_time,
+            }
+        return RequestOutput(
+            request_id=state.request_id,
+            prompt=state.prompt,
+            prompt_token_ids=state.prompt_token_ids,
+            outputs=[comp],
+            finished=state.finished,
+            metrics=metrics,
+        )
+
+    def release(self, request_id: str) -> None:
+        self.states.pop(request_id, None)
+
+
+class LLMEngine:
+
+    def __init__(self, config: EngineConfig):
+        self.config = config
+        self.engine_core = EngineCore(config)
+        self.tokenizer = TokenizerWrapper(config.model_config.tokenizer)
+        self.output_processor = OutputProcessor(self.tokenizer)
+        self._request_counter = 0
+
+    @property
+    def is_driver(self) -> bool:
+        return self.engine_core.is_driver
+
+    def add_request(
+        self,
+        request_id: Optional[str],
+        prompt: Union[str, list[int]],
+        params: Optional[SamplingParams] = None,
+    ) -> str:
+        if request_id is None:
+            request_id = f"req-{self._request_counter}"
+            self._request_counter += 1
+        params = params or SamplingParams()
+        if isinstance(prompt, str):
+            prompt_text = prompt
+            prompt_token_ids = self.tokenizer.encode(prompt)
+        else:
+            prompt_text = None
+            prompt_token_ids = list(prompt)
+        if not prompt_token_ids:
+            raise ValueError("empty prompt")
+        max_len = self.config.model_config.max_model_len
+        if len(prompt_token_ids) >= max_len:
+            raise ValueError(
+                f"prompt length {len(prompt_token_ids)} >= max_model_len "
+                f"{max_len}"
+            )
+        eos = self.tokenizer.eos_token_id
+        if eos is None:
+            eos = self.config.model_config.spec.eos_token_id
+        request = Request(
+            request_id=request_id,
+            prompt_token_ids=prompt_token_ids,
+            sampling_params=params,
+            eos_token_id=eos,
+            prompt=prompt_text,
+        )
+        self.engine_core.add_request(request)
+        self.output_processor.add_request(
+            request_id, prompt_text, prompt_token_ids, params
+        )
+        return request_id
+
+    def abort_request(self, request_ids: list[str]) -> None:
+        self.engine_core.abort_requests(request_ids)
+        for rid in request_ids:
+            self.output_processor.release(rid)
+
+    def step(self) -> list[RequestOutput]:
+        core_outputs = self.engine_core.step()
+        outputs, to_abort = self.output_processor.process_outputs(
+            core_outputs
+        )
+        if to_abort:
+            self.engine_core.abort_requests(to_abort)
+        for out in outputs:
+            if out.finished:
+                self.output_processor.release(out.request_id)
+                self.engine_core.scheduler.release_request(out.request_id)
+        return outputs
+
+    def has_unfinished_requests(self) -> bool:
+        return self.engine_core.has_unfinished_requests()
+
+    def shutdown(self) -> None:
+        self.engine_core.shutdown()

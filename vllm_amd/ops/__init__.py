@@ -1,0 +1,113 @@
+"""Custom op dispatch.
+
+On a GPU (MI355X) the hand-written HIP/CDNA4 kernels in vllm_amd/csrc are
+REQUIRED: if the compiled extension is missing we raise instead of
+silently falling back to eager PyTorch (per-op wrappers live in
+vllm_amd/ops/hip_ops.py). On CPU the plain-PyTorch reference
+implementations in _torch_ref.py run — they double as the numerics
+ground truth the GPU kernel tests compare against.
+"""
+
+from __future__ import annotations
+
+import torch
+
+from vllm_amd.ops import _torch_ref
+
+_HIP_LIB = None
+_HIP_CHECKED = False
+
+
+def _load_hip():
+    global _HIP_LIB, _HIP_CHECKED
+    if _HIP_CHECKED:
+        return _HIP_LIB
+    _HIP_CHECKED = True
+    try:
+        from vllm_amd.ops import hip_ops
+
+        _HIP_LIB = hip_ops
+    except ImportError as e:
+        _HIP_LIB = None
+        _HIP_IMPORT_ERROR[0] = e
+    return _HIP_LIB
+
+
+_HIP_IMPORT_ERROR = [None]
+
+
+def get_backend(device: torch.device):
+    if device.type == "cuda":
+        lib = _load_hip()
+        if lib is None:
+            raise RuntimeError(
+                "vllm_amd HIP extension not built — refusing to run the GPU "
+                "path on eager PyTorch. Run `python -m vllm_amd.build` or "
+                f"__graft_entry__.build(). Import error: {_HIP_IMPORT_ERROR[0]}"
+            )
+        return lib
+    return _torch_ref
+
+
+def rms_norm(x, weight, eps):
+    return get_backend(x.device).rms_norm(x, weight, eps)
+
+
+def fused_add_rms_norm(x, residual, weight, eps):
+    """Returns (normed, new_residual). May modify x/residual in place."""
+    return get_backend(x.device).fused_add_rms_norm(x, residual, weight, eps)
+
+
+def apply_rope(positions, q, k, cos_sin_cache, rotary_dim, is_neox=True):
+    """In-place rotary embedding on q [T, Hq*D] and k [T, Hkv*D]."""
+    return get_backend(q.device).apply_rope(
+        positions, q, k, cos_sin_cache, rotary_dim, is_neox
+    )
+
+
+def silu_and_mul(x):
+    return get_backend(x.device).silu_and_mul(x)
+
+
+def gelu_and_mul(x):
+    return get_backend(x.device).gelu_and_mul(x)
+
+
+def reshape_and_cache(key, value, kv_cache, slot_mapping):
+    return get_backend(key.device).reshape_and_cache(
+        key, value, kv_cache, slot_mapping
+    )
+
+
+def attention_unified(
+    q,
+    kv_cache,
+    block_table,
+    query_start_loc,
+    seq_lens,
+    scale,
+    num_decodes: int = 0,
+    sliding_window: int = 0,
+):
+    """Varlen attention over the paged KV cache (prefill + decode mixed).
+
+    q: [num_tokens, num_heads, head_dim]; kv_cache: [2, num_blocks,
+    block_size, num_kv_heads, head_dim]; returns [num_tokens, num_heads,
+    head_dim]. Requests are ordered decodes-first (query_len==1).
+    """
+    return get_backend(q.device).attention_unified(
+        q, kv_cache, block_table, query_start_loc, seq_lens, scale,
+        num_decodes, sliding_window,
+    )
+
+
+def topk_softmax(gating_logits, topk, renormalize=True):
+    return get_backend(gating_logits.device).topk_softmax(
+        gating_logits, topk, renormalize
+    )
+
+
+def fused_moe(hidden, w13, w2, topk_weights, topk_ids, activation="silu"):
+    return get_backend(hidden.device).fused_moe(
+        hidden, w13, w2, topk_weights, topk_ids, activation
+    )

@@ -1,0 +1,214 @@
+"""Distributed runtime state: TP groups over RCCL/xGMI.
+
+Role of the reference's vllm/distributed/parallel_state.py:380
+(GroupCoordinator / initialize_model_parallel), restructured for the
+MI355X topology: one process per GPU, torch.distributed with the
+"nccl" backend (RCCL on ROCm) over xGMI for the data plane, "gloo"
+for CPU-only control paths and tests.
+
+Single-process (tp=1, world=1) works with no init at all — every
+collective becomes a no-op — so CPU unit tests and the offline LLM
+path need no distributed bootstrap.
+"""
+
+from __future__ import annotations
+
+import os
+from dataclasses import dataclass
+from typing import Optional
+
+import torch
+import torch.distributed as dist
+
+
+@dataclass
+class GroupCoordinator:
+    """One communication group (TP today; PP/DP/EP reuse the same shape)."""
+
+    rank: int
+    world_size: int
+    rank_in_group: int
+    device_group: Optional[dist.ProcessGroup] = None
+    cpu_group: Optional[dist.ProcessGroup] = None
+
+    @property
+    def is_first_rank(self) -> bool:
+        return self.rank_in_group == 0
+
+    def all_reduce(self, t: torch.Tensor) -> torch.Tensor:
+        if self.world_size == 1:
+            return t
+        dist.all_reduce(t, group=self.device_group)
+        return t
+
+    def all_gather(self, t: torch.Tensor, dim: int = -1) -> torch.Tensor:
+        if self.world_size == 1:
+            return t
+        if dim < 0:
+            dim += t.dim()
+        # Gather along a new leading dim then reshape to concat on `dim`.
+        out = torch.empty(
+            (self.world_size,) + tuple(t.shape), dtype=t.dtype, device=t.device
+        )
+        dist.all_gather_into_tensor(out, t.contiguous(), group=self.device_group)
+        if dim == 0:
+            return out.reshape(-1, *t.shape[1:])
+        pieces = out.unbind(0)
+        return torch.cat(pieces, dim=dim)
+
+    def reduce_scatter(self, t: torch.Tensor, dim: int = 0) -> torch.Tensor:
+        if self.world_size == 1:
+            return t
+        assert dim == 0
+        out_shape = (t.shape[0] // self.world_size,) + tuple(t.shape[1:])
+        out = torch.empty(out_shape, dtype=t.dtype, device=t.device)
+        dist.reduce_scatter_tensor(out, t.contiguous(), group=self.device_group)
+        return out
+
+    def broadcast(self, t: torch.Tensor, src: int = 0) -> torch.Tensor:
+        if self.world_size == 1:
+            return t
+        dist.broadcast(t, src=src, group=self.device_group)
+        return t
+
+    def broadcast_object(self, obj=None, src: int = 0):
+        if self.world_size == 1:
+            return obj
+        holder = [obj]
+        dist.broadcast_object_list(holder, src=src,
+                                   group=self.cpu_group or self.device_group)
+        return holder[0]
+
+    def all_to_all(self, t: torch.Tensor) -> torch.Tensor:
+        """Equal-split all-to-all along dim 0 (EP dispatch/combine)."""
+        if self.world_size == 1:
+            return t
+        out = torch.empty_like(t)
+        dist.all_to_all_single(out, t.contiguous(), group=self.device_group)
+        return out
+
+    def barrier(self) -> None:
+        if self.world_size > 1:
+            dist.barrier(group=self.cpu_group or self.device_group)
+
+
+# Module-level groups (initialized once per process).
+_TP: Optional[GroupCoordinator] = None
+_EP: Optional[GroupCoordinator] = None
+_WORLD: Optional[GroupCoordinator] = None
+
+_SINGLE = GroupCoordinator(rank=0, world_size=1, rank_in_group=0)
+
+
+def get_tp_group() -> GroupCoordinator:
+    return _TP if _TP is not None else _SINGLE
+
+
+def get_ep_group() -> GroupCoordinator:
+    return _EP if _EP is not None else _SINGLE
+
+
+def get_world_group() -> GroupCoordinator:
+    return _WORLD if _WORLD is not None else _SINGLE
+
+
+def get_tp_rank() -> int:
+    return get_tp_group().rank_in_group
+
+
+def get_tp_world_size() -> int:
+    return get_tp_group().world_size
+
+
+def is_initialized() -> bool:
+    return _TP is not None
+
+
+def init_distributed(
+    tensor_parallel_size: int = 1,
+    backend: str = "auto",
+    rank: Optional[int] = None,
+    world_size: Optional[int] = None,
+    local_rank: Optional[int] = None,
+) -> None:
+    """Initialize torch.distributed and the model-parallel groups.
+
+    Reads RANK/WORLD_SIZE/LOCAL_RANK/MASTER_* from env when launched by
+    torchrun (the bench.py contract). world = TP in v1 (DP is one engine
+    per replica; PP later).
+    """
+    global _TP, _EP, _WORLD
+    rank = rank if rank is not None else int(os.environ.get("RANK", "0"))
+    world_size = (
+        world_size
+        if world_size is not None
+        else int(os.environ.get("WORLD_SIZE", "1"))
+    )
+    local_rank = (
+        local_rank
+        if local_rank is not None
+        else int(os.environ.get("LOCAL_RANK", str(rank)))
+    )
+
+    if world_size == 1 and tensor_parallel_size == 1:
+        return  # single-process fast path; no distributed state at all
+
+    if backend == "auto":
+        backend = "nccl" if torch.cuda.is_available() else "gloo"
+
+    if torch.cuda.is_available():
+        torch.cuda.set_device(local_rank)
+
+    if not dist.is_initialized():
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29500")
+        dist.init_process_group(
+            backend=backend, rank=rank, world_size=world_size
+        )
+
+    cpu_group = (
+        dist.new_group(backend="gloo") if backend != "gloo" else None
+    )
+
+    assert world_size % tensor_parallel_size == 0
+    # TP groups are contiguous ranks (adjacent ranks share xGMI hops).
+    tp_group = None
+    my_tp_ranks = None
+    for start in range(0, world_size, tensor_parallel_size):
+        ranks = list(range(start, start + tensor_parallel_size))
+        g = dist.new_group(ranks, backend=backend)
+        if rank in ranks:
+            tp_group = g
+            my_tp_ranks = ranks
+    _TP = GroupCoordinator(
+        rank=rank,
+        world_size=tensor_parallel_size,
+        rank_in_group=my_tp_ranks.index(rank),
+        device_group=tp_group,
+        cpu_group=cpu_group,
+    )
+    _WORLD = GroupCoordinator(
+        rank=rank,
+        world_size=world_size,
+        rank_in_group=rank,
+        device_group=dist.group.WORLD,
+        cpu_group=cpu_group,
+    )
+    # EP group == world group in v1 (experts sharded across all ranks).
+    _EP = _WORLD
+
+
+def destroy_distributed() -> None:
+    global _TP, _EP, _WORLD
+    _TP = _EP = _WORLD = None
+    if dist.is_initialized():
+        dist.destroy_process_group()
+
+
+# Convenience wrappers used by layers.
+def tensor_model_parallel_all_reduce(t: torch.Tensor) -> torch.Tensor:
+    return get_tp_group().all_reduce(t)
+
+
+def tensor_model_parallel_all_gather(t: torch.Tensor, dim: int = -1) -> torch.Tensor:
+    return get_tp_group().all_gather(t, dim=dim)

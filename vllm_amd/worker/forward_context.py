@@ -1,0 +1,59 @@
+"""Per-step ambient forward context (role of vllm/forward_context.py).
+
+The model runner sets one context per step; Attention layers read their
+metadata and KV cache tensor from it instead of threading them through
+every forward signature.
+"""
+
+from __future__ import annotations
+
+from contextlib import contextmanager
+from dataclasses import dataclass
+from typing import Optional
+
+import torch
+
+
+@dataclass
+class AttentionMetadata:
+    """One metadata build per step, shared by all layers
+    (shape of the reference's CommonAttentionMetadata, backend.py:412)."""
+
+    query_start_loc: torch.Tensor  # [num_reqs+1] int32
+    seq_lens: torch.Tensor  # [num_reqs] int32 (context incl. new tokens)
+    block_table: torch.Tensor  # [num_reqs, max_blocks] int32
+    slot_mapping: torch.Tensor  # [num_tokens] int64
+    num_reqs: int
+    num_actual_tokens: int
+    max_query_len: int
+    max_seq_len: int
+    # Requests are ordered decodes-first; first num_decodes have query_len 1.
+    num_decodes: int = 0
+
+
+@dataclass
+class ForwardContext:
+    attn_metadata: Optional[AttentionMetadata]
+    kv_caches: list[torch.Tensor]  # one per layer; [] during profiling
+
+
+_forward_context: Optional[ForwardContext] = None
+
+
+def get_forward_context() -> ForwardContext:
+    assert _forward_context is not None, (
+        "forward context not set — model forward must run under "
+        "set_forward_context()"
+    )
+    return _forward_context
+
+
+@contextmanager
+def set_forward_context(ctx: ForwardContext):
+    global _forward_context
+    prev = _forward_context
+    _forward_context = ctx
+    try:
+        yield
+    finally:
+        _forward_context = prev

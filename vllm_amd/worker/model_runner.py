@@ -1,0 +1,268 @@
+"""Model runner: per-step input prep + forward + sampling.
+
+Role of the reference's GPUModelRunner (vllm/v1/worker/gpu_model_runner.py):
+keeps persistent per-request state across steps, applies the scheduler's
+diffs, builds one flat token batch (decodes first), runs the model under a
+forward context, and samples for every request whose tokens are fully
+computed after this step.
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import Optional
+
+import numpy as np
+import torch
+
+from vllm_amd.config import EngineConfig
+from vllm_amd.core.sched_output import ModelRunnerOutput, SchedulerOutput
+from vllm_amd.layers.sampler import Sampler, SamplingMetadata
+from vllm_amd.models.registry import load_model
+from vllm_amd.sampling_params import SamplingParams
+from vllm_amd.worker.forward_context import (
+    AttentionMetadata,
+    ForwardContext,
+    set_forward_context,
+)
+
+
+@dataclass
+class CachedReqState:
+    req_id: str
+    token_ids: list[int]
+    prompt_len: int
+    num_computed_tokens: int
+    block_ids: list[int]
+    sampling_params: SamplingParams
+
+    @property
+    def output_token_ids(self) -> list[int]:
+        return self.token_ids[self.prompt_len:]
+
+
+class ModelRunner:
+
+    def __init__(self, config: EngineConfig, device: torch.device):
+        self.config = config
+        self.device = device
+        self.block_size = config.cache_config.block_size
+        self.spec = config.model_config.spec
+        self.dtype = config.model_config.torch_dtype
+        self.requests: dict[str, CachedReqState] = {}
+        self.kv_caches: list[torch.Tensor] = []
+        self.model: Optional[torch.nn.Module] = None
+        self.sampler = Sampler()
+        self.max_model_len = config.model_config.max_model_len
+
+    def load_model(self) -> None:
+        self.model = load_model(self.config.model_config, self.device)
+
+    # ------------------------------------------------------------------
+    def kv_cache_page_bytes(self) -> int:
+        """Bytes per KV block across all layers on THIS rank."""
+        from vllm_amd.parallel.state import get_tp_world_size
+
+        spec = self.spec
+        kv_heads = spec.num_kv_heads_per_rank(get_tp_world_size())
+        elt = torch.tensor([], dtype=self.dtype).element_size()
+        return (
+            2 * self.block_size * kv_heads * spec.head_dim * elt
+            * spec.num_layers
+        )
+
+    def allocate_kv_cache(self, num_blocks: int) -> None:
+        from vllm_amd.parallel.state import get_tp_world_size
+
+        spec = self.spec
+        kv_heads = spec.num_kv_heads_per_rank(get_tp_world_size())
+        self.kv_caches = [
+            torch.zeros(
+                2, num_blocks, self.block_size, kv_heads, spec.head_dim,
+                dtype=self.dtype, device=self.device,
+            )
+            for _ in range(spec.num_layers)
+        ]
+
+    # ------------------------------------------------------------------
+    def _update_states(self, so: SchedulerOutput) -> None:
+        for req_id in so.finished_req_ids:
+            self.requests.pop(req_id, None)
+        for nr in so.scheduled_new_reqs:
+            self.requests[nr.req_id] = CachedReqState(
+                req_id=nr.req_id,
+                token_ids=list(nr.prompt_token_ids),
+                prompt_len=len(nr.prompt_token_ids),
+                num_computed_tokens=nr.num_computed_tokens,
+                block_ids=list(nr.block_ids),
+                sampling_params=nr.sampling_params,
+            )
+        cr = so.scheduled_cached_reqs
+        for i, req_id in enumerate(cr.req_ids):
+            state = self.requests[req_id]
+            if cr.resumed[i]:
+                state.block_ids = list(cr.new_block_ids[i])
+                state.token_ids = list(cr.new_token_ids[i])
+                state.num_computed_tokens = cr.num_computed_tokens[i]
+            else:
+                state.block_ids.extend(cr.new_block_ids[i])
+                state.num_computed_tokens = cr.num_computed_tokens[i]
+
+    # ------------------------------------------------------------------
+    @torch.inference_mode()
+    def execute_model(self, so: SchedulerOutput) -> ModelRunnerOutput:
+        self._update_states(so)
+        if so.total_num_scheduled_tokens == 0:
+            return ModelRunnerOutput(req_ids=[], sampled_token_ids=[])
+
+        # Order requests decodes-first (query_len == 1).
+        items = sorted(
+            so.num_scheduled_tokens.items(), key=lambda kv: kv[1] != 1
+        )
+        req_ids = [rid for rid, _ in items]
+        num_decodes = sum(1 for _, n in items if n == 1)
+
+        total = so.total_num_scheduled_tokens
+        input_ids = np.empty(total, dtype=np.int64)
+        positions = np.empty(total, dtype=np.int64)
+        slot_mapping = np.empty(total, dtype=np.int64)
+        query_start_loc = np.zeros(len(items) + 1, dtype=np.int32)
+        seq_lens = np.empty(len(items), dtype=np.int32)
+        max_blocks = max(
+            (len(self.requests[rid].block_ids) for rid in req_ids), default=1
+        )
+        block_table = np.zeros((len(items), max_blocks), dtype=np.int32)
+
+        sampling_rows: list[int] = []  # row in `items` order
+        t = 0
+        for i, (rid, n) in enumerate(items):
+            state = self.requests[rid]
+            start = state.num_computed_tokens
+            toks = state.token_ids[start : start + n]
+            input_ids[t : t + n] = toks
+            positions[t : t + n] = np.arange(start, start + n)
+            # slot = block_ids[pos // bs] * bs + pos % bs
+            pos = np.arange(start, start + n)
+            bids = np.asarray(state.block_ids, dtype=np.int64)
+            slot_mapping[t : t + n] = (
+                bids[pos // self.block_size] * self.block_size
+                + pos % self.block_size
+            )
+            query_start_loc[i + 1] = query_start_loc[i] + n
+            seq_lens[i] = start + n
+            block_table[i, : len(state.block_ids)] = state.block_ids
+            if start + n >= len(state.token_ids):
+                sampling_rows.append(i)
+            t += n
+
+        dev = self.device
+        meta = AttentionMetadata(
+            query_start_loc=torch.from_numpy(query_start_loc).to(dev),
+            seq_lens=torch.from_numpy(seq_lens).to(dev),
+            block_table=torch.from_numpy(block_table).to(dev),
+            slot_mapping=torch.from_numpy(slot_mapping).to(dev),
+            num_reqs=len(items),
+            num_actual_tokens=total,
+            max_query_len=int(max(n for _, n in items)),
+            max_seq_len=int(seq_lens.max()),
+            num_decodes=num_decodes,
+        )
+        input_ids_t = torch.from_numpy(input_ids).to(dev)
+        positions_t = torch.from_numpy(positions).to(dev)
+
+        ctx = ForwardContext(attn_metadata=meta, kv_caches=self.kv_caches)
+        with set_forward_context(ctx):
+            hidden = self.model(input_ids_t, positions_t)
+
+        # Advance computed counts.
+        for rid, n in items:
+            self.requests[rid].num_computed_tokens += n
+
+        if not sampling_rows:
+            return ModelRunnerOutput(
+                req_ids=req_ids,
+                sampled_token_ids=[[] for _ in req_ids],
+            )
+
+        # Gather last-token hidden states of sampling requests.
+        last_idx = torch.tensor(
+            [int(query_start_loc[r + 1]) - 1 for r in sampling_rows],
+            device=dev,
+        )
+        logits = self.model.compute_logits(hidden[last_idx])
+
+        s_params = [
+            self.requests[req_ids[r]].sampling_params for r in sampling_rows
+        ]
+        s_prompts = [
+            self.requests[req_ids[r]].token_ids[
+                : self.requests[req_ids[r]].prompt_len
+            ]
+            for r in sampling_rows
+        ]
+        s_outputs = [
+            self.requests[req_ids[r]].output_token_ids for r in sampling_rows
+        ]
+        s_meta = SamplingMetadata.build(
+            s_params,
+            s_prompts,
+            s_outputs,
+            dev,
+            seeds_offset=[len(o) for o in s_outputs],
+        )
+        s_out = self.sampler(logits, s_meta)
+        sampled = s_out.sampled_token_ids.tolist()
+
+        sampled_per_req: list[list[int]] = [[] for _ in req_ids]
+        logprobs_per_req: dict[str, list[dict[int, float]]] = {}
+        for j, r in enumerate(sampling_rows):
+            tok = int(sampled[j])
+            sampled_per_req[r] = [tok]
+            # Runner keeps its own copy of generated tokens.
+            self.requests[req_ids[r]].token_ids.append(tok)
+            if s_out.logprobs is not None and s_out.logprobs[j] is not None:
+                logprobs_per_req[req_ids[r]] = [s_out.logprobs[j]]
+        return ModelRunnerOutput(
+            req_ids=req_ids,
+            sampled_token_ids=sampled_per_req,
+            logprobs=logprobs_per_req or None,
+        )
+
+    # ------------------------------------------------------------------
+    @torch.inference_mode()
+    def profile_run(self) -> None:
+        """Dummy forward at the worst-case token count for KV-memory sizing
+        (role of determine_available_memory, gpu_worker.py:461)."""
+        max_tokens = self.config.scheduler_config.max_num_batched_tokens
+        max_reqs = min(
+            self.config.scheduler_config.max_num_seqs, max_tokens
+        )
+        # One big prefill: worst-case activation memory.
+        tokens_per_req = max_tokens // max_reqs
+        counts = [tokens_per_req] * max_reqs
+        counts[0] += max_tokens - sum(counts)
+        total = sum(counts)
+        qsl = np.zeros(max_reqs + 1, dtype=np.int32)
+        qsl[1:] = np.cumsum(counts)
+        meta = AttentionMetadata(
+            query_start_loc=torch.from_numpy(qsl).to(self.device),
+            seq_lens=torch.tensor(counts, dtype=torch.int32,
+                                  device=self.device),
+            block_table=torch.zeros(max_reqs, 1, dtype=torch.int32,
+                                    device=self.device),
+            slot_mapping=torch.zeros(total, dtype=torch.int64,
+                                     device=self.device),
+            num_reqs=max_reqs,
+            num_actual_tokens=total,
+            max_query_len=max(counts),
+            max_seq_len=max(counts),
+        )
+        input_ids = torch.zeros(total, dtype=torch.int64, device=self.device)
+        positions = torch.cat(
+            [torch.arange(c, device=self.device) for c in counts]
+        )
+        ctx = ForwardContext(attn_metadata=meta, kv_caches=[])
+        with set_forward_context(ctx):
+            hidden = self.model(input_ids, positions)
+            # Include logits in the peak (all rows worst case).
+            self.model.compute_logits(hidden[: max_reqs])
