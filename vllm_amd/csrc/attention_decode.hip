@@ -1,0 +1,351 @@
+// Paged decode attention for MI355X (gfx950, CDNA4).
+//
+// Design (MI355X-first, not a port):
+//   Decode attention is HBM-bandwidth bound: per context token and KV head
+//   the kernel reads 2*D*2 bytes of K+V and does GQA_group*4*D flops —
+//   arithmetic intensity = group flops/byte (≈4-8), far below the
+//   157 TF / 6.3 TB/s = 25 flops/byte VALU roofline. So this kernel is a
+//   pure-VALU streaming kernel tuned for the load path, not MFMA:
+//   16B/lane short8 K/V loads (G13), 16-lane thread groups per token,
+//   flash-decoding partitions of PART tokens with a separate LSE-merge
+//   reduce kernel for long contexts.
+//
+//   The KV cache layout [2, blocks, kv_heads, block_size=64, D] keeps each
+//   (block, head) tile contiguous (16 KB at D=128), so a partition is a
+//   handful of fully-coalesced streams.
+//
+// Role of the reference's paged_attention (csrc/rocm/attention.cu:321,1420),
+// re-designed: the reference packs GQA into MFMA16 tiles; on MI355X the
+// memory-bound regime makes VALU dot products simpler and just as fast.
+
+#include <torch/all.h>
+#include <c10/hip/HIPStream.h>
+
+#include "common.h"
+
+namespace vllm_amd {
+
+// Tunables. PART must be a multiple of the token-iteration width
+// (BLOCK/16 tokens per iteration) and of the cache block size (64).
+constexpr int DEC_BLOCK = 256;          // threads per workgroup (4 waves)
+constexpr int DEC_PART = 512;           // context tokens per partition
+constexpr int MAX_GROUP = 8;            // max GQA ratio handled in registers
+
+// One workgroup: (decode seq, kv head, partition). Computes the partial
+// attention output for GROUP query heads over PART context tokens with an
+// unnormalized softmax (running max m, exp-sum l), writing either the
+// final bf16 output (single partition) or fp32 partials + (m, l) scratch.
+//
+// Thread layout: 16 lanes per token (lane d covers dims 8d..8d+7 as one
+// short8 = 16 B load), so a 256-thread block streams 16 tokens per
+// iteration; DEC_PART/16 = 32 iterations.
+template <typename Tag, int GROUP, bool FINAL>
+__global__ __launch_bounds__(DEC_BLOCK) void paged_decode_kernel(
+    short* __restrict__ out,            // FINAL: [Tdec, Hq, D] (16-bit)
+    float* __restrict__ tmp_out,        // else: [Tdec, Hq, parts, D]
+    float* __restrict__ tmp_lse,        // else: [Tdec, Hq, parts, 2]
+    const short* __restrict__ q,        // [Tdec, Hq, D]
+    const short* __restrict__ kv_cache, // [2, blocks, Hkv, 64, D]
+    const int* __restrict__ block_table,   // [num_reqs, max_blocks]
+    const int* __restrict__ seq_lens,      // [num_reqs]
+    const float scale, const int num_q_heads, const int num_kv_heads,
+    const int head_dim, const int max_blocks_per_req, const int num_parts,
+    const int64_t kv_plane_stride,      // elements between K and V planes
+    const int sliding_window) {
+  const int seq = blockIdx.x;       // decode row == request row
+  const int kvh = blockIdx.y;
+  const int part = blockIdx.z;
+  const int ctx = seq_lens[seq];
+
+  int t_begin = part * DEC_PART;
+  int t_end = min(ctx, t_begin + DEC_PART);
+  if (sliding_window > 0) {
+    // Query position is ctx-1; keys < ctx - sliding_window are masked.
+    t_begin = max(t_begin, ctx - sliding_window);
+  }
+  const int hq0 = kvh * GROUP;  // first query head of this group
+
+  if (t_begin >= t_end) {
+    if (!FINAL && threadIdx.x < GROUP * 2) {
+      const int g = threadIdx.x / 2;
+      float* lse = tmp_lse +
+          (((int64_t)seq * num_q_heads + hq0 + g) * num_parts + part) * 2;
+      lse[threadIdx.x % 2] = (threadIdx.x % 2 == 0) ? -3.0e38f : 0.f;
+    }
+    return;
+  }
+
+  const int lane16 = threadIdx.x & 15;        // dim-slice owner
+  const int tok_slot = threadIdx.x >> 4;      // 0..15: token within iter
+  const int dvec = head_dim / 8;              // short8 vectors per row (16)
+
+  // LDS: physical block ids for this partition + score/accum scratch.
+  __shared__ int blk_ids[DEC_PART / 64];
+  __shared__ float scores[GROUP][DEC_PART];
+  __shared__ float red_scratch[DEC_BLOCK / WAVE_SIZE];
+  // Cross-wave O reduction: [4 waves][GROUP][D] reused after score phase.
+  __shared__ float o_red[4][GROUP][128];
+
+  if (threadIdx.x < DEC_PART / 64) {
+    const int cache_blk = (part * DEC_PART) / 64 + threadIdx.x;
+    blk_ids[threadIdx.x] =
+        block_table[(int64_t)seq * max_blocks_per_req + cache_blk];
+  }
+  __syncthreads();
+
+  // Q fragment: each lane16 holds its 8-dim slice for all GROUP heads.
+  float qf[GROUP][8];
+  {
+    const short* qbase =
+        q + ((int64_t)seq * num_q_heads + hq0) * head_dim + lane16 * 8;
+#pragma unroll
+    for (int g = 0; g < GROUP; ++g) {
+      s16x8 v = *reinterpret_cast<const s16x8*>(qbase + g * head_dim);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) qf[g][j] = to_f32<Tag>(v[j]);
+    }
+  }
+
+  // ---- K pass: scores[g][t] = scale * q_g . k_t --------------------------
+  const int64_t head_tile_stride = (int64_t)64 * head_dim;  // one (blk,head)
+  for (int t = t_begin + tok_slot; t < t_end; t += 16) {
+    const int local = t - part * DEC_PART;
+    const int phys = blk_ids[local / 64];
+    const short* krow = kv_cache +
+        ((int64_t)phys * num_kv_heads + kvh) * head_tile_stride +
+        (int64_t)(t % 64) * head_dim + lane16 * 8;
+    s16x8 kv = *reinterpret_cast<const s16x8*>(krow);
+    float kf[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) kf[j] = to_f32<Tag>(kv[j]);
+#pragma unroll
+    for (int g = 0; g < GROUP; ++g) {
+      float s = 0.f;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) s += qf[g][j] * kf[j];
+      // Reduce over the 16 dim-slice lanes.
+#pragma unroll
+      for (int off = 8; off > 0; off >>= 1) s += __shfl_xor(s, off, 64);
+      if (lane16 == g % 16) scores[g][t - t_begin] = s * scale;
+    }
+  }
+  __syncthreads();
+
+  const int n_tok = t_end - t_begin;
+
+  // ---- softmax over the partition (block-wide, per head) -----------------
+  float m_g[GROUP], l_g[GROUP];
+#pragma unroll
+  for (int g = 0; g < GROUP; ++g) {
+    float m = -3.0e38f;
+    for (int i = threadIdx.x; i < n_tok; i += DEC_BLOCK)
+      m = fmaxf(m, scores[g][i]);
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1)
+      m = fmaxf(m, __shfl_xor(m, off, 64));
+    if ((threadIdx.x & 63) == 0) red_scratch[threadIdx.x >> 6] = m;
+    __syncthreads();
+    m = fmaxf(fmaxf(red_scratch[0], red_scratch[1]),
+              fmaxf(red_scratch[2], red_scratch[3]));
+    float l = 0.f;
+    for (int i = threadIdx.x; i < n_tok; i += DEC_BLOCK) {
+      const float p = __expf(scores[g][i] - m);
+      scores[g][i] = p;
+      l += p;
+    }
+    __syncthreads();  // red_scratch reuse
+    l = block_reduce_sum<DEC_BLOCK / WAVE_SIZE>(l, red_scratch);
+    m_g[g] = m;
+    l_g[g] = l;
+    __syncthreads();
+  }
+
+  // ---- V pass: acc[g][j] += p[t] * v_t[dims] -----------------------------
+  float acc[GROUP][8];
+#pragma unroll
+  for (int g = 0; g < GROUP; ++g)
+#pragma unroll
+    for (int j = 0; j < 8; ++j) acc[g][j] = 0.f;
+
+  for (int t = t_begin + tok_slot; t < t_end; t += 16) {
+    const int local = t - part * DEC_PART;
+    const int phys = blk_ids[local / 64];
+    const short* vrow = kv_cache + kv_plane_stride +
+        ((int64_t)phys * num_kv_heads + kvh) * head_tile_stride +
+        (int64_t)(t % 64) * head_dim + lane16 * 8;
+    s16x8 vv = *reinterpret_cast<const s16x8*>(vrow);
+    float vf[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) vf[j] = to_f32<Tag>(vv[j]);
+#pragma unroll
+    for (int g = 0; g < GROUP; ++g) {
+      const float p = scores[g][t - t_begin];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) acc[g][j] += p * vf[j];
+    }
+  }
+
+  // Reduce the 4 token-slots within each wave (lanes 16 apart share dims).
+#pragma unroll
+  for (int g = 0; g < GROUP; ++g)
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      acc[g][j] += __shfl_xor(acc[g][j], 16, 64);
+      acc[g][j] += __shfl_xor(acc[g][j], 32, 64);
+    }
+  // Cross-wave reduce via LDS (token-slot lane of each wave contributes).
+  const int wave = threadIdx.x / WAVE_SIZE;
+  __syncthreads();  // scores no longer needed; o_red aliases fresh use
+  if ((threadIdx.x & 63) < 16) {
+#pragma unroll
+    for (int g = 0; g < GROUP; ++g)
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        o_red[wave][g][lane16 * 8 + j] = acc[g][j];
+  }
+  __syncthreads();
+  if (wave == 0 && lane16 < 16) {
+#pragma unroll
+    for (int g = 0; g < GROUP; ++g) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int d = lane16 * 8 + j;
+        float o = o_red[0][g][d] + o_red[1][g][d] + o_red[2][g][d] +
+                  o_red[3][g][d];
+        if (FINAL) {
+          if (tok_slot == 0) {
+            const float inv_l = l_g[g] > 0.f ? 1.f / l_g[g] : 0.f;
+            out[((int64_t)seq * num_q_heads + hq0 + g) * head_dim + d] =
+                from_f32<Tag>(o * inv_l);
+          }
+        } else {
+          if (tok_slot == 0) {
+            tmp_out[((((int64_t)seq * num_q_heads + hq0 + g) * num_parts +
+                      part) * head_dim) + d] = o;
+          }
+        }
+      }
+    }
+    if (!FINAL && tok_slot == 0 && lane16 < GROUP) {
+      float* lse = tmp_lse +
+          (((int64_t)seq * num_q_heads + hq0 + lane16) * num_parts + part) *
+              2;
+      lse[0] = m_g[lane16];
+      lse[1] = l_g[lane16];
+    }
+  }
+}
+
+// Merge per-partition partials: out = sum_p o_p * exp(m_p - M) / sum_p
+// l_p * exp(m_p - M). Grid (Tdec, Hq), block = D threads (one per dim).
+template <typename Tag>
+__global__ void paged_decode_reduce_kernel(
+    short* __restrict__ out,           // [Tdec, Hq, D]
+    const float* __restrict__ tmp_out, // [Tdec, Hq, parts, D]
+    const float* __restrict__ tmp_lse, // [Tdec, Hq, parts, 2]
+    const int* __restrict__ seq_lens, const int num_parts,
+    const int head_dim, const int num_q_heads) {
+  const int seq = blockIdx.x;
+  const int h = blockIdx.y;
+  const int d = threadIdx.x;
+  const int used_parts = min(num_parts, (seq_lens[seq] + DEC_PART - 1) /
+                                            DEC_PART);
+  const float* lse =
+      tmp_lse + (((int64_t)seq * num_q_heads + h) * num_parts) * 2;
+  float M = -3.0e38f;
+  for (int p = 0; p < used_parts; ++p) M = fmaxf(M, lse[p * 2]);
+  float L = 0.f;
+  for (int p = 0; p < used_parts; ++p)
+    L += lse[p * 2 + 1] * __expf(lse[p * 2] - M);
+  const float inv_l = L > 0.f ? 1.f / L : 0.f;
+  const float* o_base =
+      tmp_out + (((int64_t)seq * num_q_heads + h) * num_parts) * head_dim;
+  float o = 0.f;
+  for (int p = 0; p < used_parts; ++p)
+    o += o_base[p * head_dim + d] * __expf(lse[p * 2] - M);
+  out[((int64_t)seq * num_q_heads + h) * head_dim + d] =
+      from_f32<Tag>(o * inv_l);
+}
+
+// ---------------------------------------------------------------------------
+// Host launcher. q: [Tdec, Hq, D] (contiguous slice of the step's q);
+// out: [Tdec, Hq, D]; kv_cache: [2, blocks, Hkv, 64, D];
+// block_table/seq_lens cover the decode rows (rows 0..Tdec-1).
+void paged_decode_attention(torch::Tensor out, torch::Tensor q,
+                            torch::Tensor kv_cache,
+                            torch::Tensor block_table,
+                            torch::Tensor seq_lens, double scale,
+                            int64_t max_seq_len, int64_t sliding_window,
+                            torch::Tensor tmp_out, torch::Tensor tmp_lse) {
+  const int num_seqs = q.size(0);
+  if (num_seqs == 0) return;
+  const int num_q_heads = q.size(1);
+  const int head_dim = q.size(2);
+  const int num_kv_heads = kv_cache.size(2);
+  const int group = num_q_heads / num_kv_heads;
+  TORCH_CHECK(head_dim == 128, "decode kernel supports head_dim=128");
+  TORCH_CHECK(kv_cache.size(3) == 64, "cache block_size must be 64");
+  TORCH_CHECK(group >= 1 && group <= MAX_GROUP && num_q_heads ==
+              group * num_kv_heads, "GQA group must be 1..8, got ", group);
+  const int num_parts =
+      (int)((max_seq_len + DEC_PART - 1) / DEC_PART);
+  const int max_blocks = block_table.size(1);
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+
+  dim3 grid(num_seqs, num_kv_heads, num_parts);
+  dim3 block(DEC_BLOCK);
+
+#define LAUNCH_DEC(TAG, G, FIN)                                              \
+  hipLaunchKernelGGL((paged_decode_kernel<TAG, G, FIN>), grid, block, 0,     \
+                     stream, (short*)out.data_ptr(),                         \
+                     FIN ? nullptr : tmp_out.data_ptr<float>(),              \
+                     FIN ? nullptr : tmp_lse.data_ptr<float>(),              \
+                     (const short*)q.data_ptr(),                             \
+                     (const short*)kv_cache.data_ptr(),                      \
+                     block_table.data_ptr<int>(), seq_lens.data_ptr<int>(),  \
+                     (float)scale, num_q_heads, num_kv_heads, head_dim,      \
+                     max_blocks, num_parts, kv_cache.stride(0),              \
+                     (int)sliding_window)
+
+#define DISPATCH_GROUP(TAG, FIN)                                             \
+  switch (group) {                                                           \
+    case 1: LAUNCH_DEC(TAG, 1, FIN); break;                                  \
+    case 2: LAUNCH_DEC(TAG, 2, FIN); break;                                  \
+    case 4: LAUNCH_DEC(TAG, 4, FIN); break;                                  \
+    case 5: LAUNCH_DEC(TAG, 5, FIN); break;                                  \
+    case 6: LAUNCH_DEC(TAG, 6, FIN); break;                                  \
+    case 8: LAUNCH_DEC(TAG, 8, FIN); break;                                  \
+    case 3: LAUNCH_DEC(TAG, 3, FIN); break;                                  \
+    case 7: LAUNCH_DEC(TAG, 7, FIN); break;                                  \
+    default: TORCH_CHECK(false, "unsupported GQA group ", group);            \
+  }
+
+  const bool is_bf16 = q.scalar_type() == torch::kBFloat16;
+  if (num_parts == 1) {
+    if (is_bf16) { DISPATCH_GROUP(BF16Tag, true); }
+    else         { DISPATCH_GROUP(FP16Tag, true); }
+  } else {
+    if (is_bf16) { DISPATCH_GROUP(BF16Tag, false); }
+    else         { DISPATCH_GROUP(FP16Tag, false); }
+    HIP_CHECK_KERNEL();
+    dim3 rgrid(num_seqs, num_q_heads);
+    if (is_bf16) {
+      hipLaunchKernelGGL((paged_decode_reduce_kernel<BF16Tag>), rgrid,
+                         dim3(head_dim), 0, stream, (short*)out.data_ptr(),
+                         tmp_out.data_ptr<float>(),
+                         tmp_lse.data_ptr<float>(), seq_lens.data_ptr<int>(),
+                         num_parts, head_dim, num_q_heads);
+    } else {
+      hipLaunchKernelGGL((paged_decode_reduce_kernel<FP16Tag>), rgrid,
+                         dim3(head_dim), 0, stream, (short*)out.data_ptr(),
+                         tmp_out.data_ptr<float>(),
+                         tmp_lse.data_ptr<float>(), seq_lens.data_ptr<int>(),
+                         num_parts, head_dim, num_q_heads);
+    }
+  }
+  HIP_CHECK_KERNEL();
+#undef DISPATCH_GROUP
+#undef LAUNCH_DEC
+}
+
+}  // namespace vllm_amd

@@ -1,0 +1,64 @@
+// torch.library bindings for the vllm_amd CDNA4 kernels.
+//
+// Loaded via torch.ops.load_library("vllm_amd/_C.so"); ops appear as
+// torch.ops.vllm_amd.<name>. No Python C API — pure libtorch, same
+// pattern as the reference's csrc/libtorch_stable/torch_bindings.cpp.
+
+#include <torch/all.h>
+#include <torch/library.h>
+
+namespace vllm_amd {
+
+torch::Tensor rms_norm(torch::Tensor x, torch::Tensor weight, double eps);
+void fused_add_rms_norm(torch::Tensor x, torch::Tensor residual,
+                        torch::Tensor weight, double eps);
+torch::Tensor silu_and_mul(torch::Tensor x);
+torch::Tensor gelu_and_mul(torch::Tensor x);
+void rotary_embedding(torch::Tensor positions, torch::Tensor q,
+                      torch::Tensor k, torch::Tensor cos_sin_cache,
+                      int64_t rot);
+void reshape_and_cache(torch::Tensor key, torch::Tensor value,
+                       torch::Tensor kv_cache, torch::Tensor slot_mapping);
+void paged_decode_attention(torch::Tensor out, torch::Tensor q,
+                            torch::Tensor kv_cache,
+                            torch::Tensor block_table,
+                            torch::Tensor seq_lens, double scale,
+                            int64_t max_seq_len, int64_t sliding_window,
+                            torch::Tensor tmp_out, torch::Tensor tmp_lse);
+void prefill_attention(torch::Tensor out, torch::Tensor q,
+                       torch::Tensor kv_cache, torch::Tensor block_table,
+                       torch::Tensor query_start_loc, torch::Tensor seq_lens,
+                       double scale, int64_t num_decodes,
+                       int64_t max_query_len, int64_t sliding_window);
+
+}  // namespace vllm_amd
+
+TORCH_LIBRARY(vllm_amd, m) {
+  m.def("rms_norm(Tensor x, Tensor weight, float eps) -> Tensor");
+  m.def("fused_add_rms_norm(Tensor(a!) x, Tensor(b!) residual, "
+        "Tensor weight, float eps) -> ()");
+  m.def("silu_and_mul(Tensor x) -> Tensor");
+  m.def("gelu_and_mul(Tensor x) -> Tensor");
+  m.def("rotary_embedding(Tensor positions, Tensor(a!) q, Tensor(b!) k, "
+        "Tensor cos_sin_cache, int rot) -> ()");
+  m.def("reshape_and_cache(Tensor key, Tensor value, Tensor(a!) kv_cache, "
+        "Tensor slot_mapping) -> ()");
+  m.def("paged_decode_attention(Tensor(a!) out, Tensor q, Tensor kv_cache, "
+        "Tensor block_table, Tensor seq_lens, float scale, int max_seq_len, "
+        "int sliding_window, Tensor(b!) tmp_out, Tensor(c!) tmp_lse) -> ()");
+  m.def("prefill_attention(Tensor(a!) out, Tensor q, Tensor kv_cache, "
+        "Tensor block_table, Tensor query_start_loc, Tensor seq_lens, "
+        "float scale, int num_decodes, int max_query_len, "
+        "int sliding_window) -> ()");
+}
+
+TORCH_LIBRARY_IMPL(vllm_amd, CUDA, m) {
+  m.impl("rms_norm", &vllm_amd::rms_norm);
+  m.impl("fused_add_rms_norm", &vllm_amd::fused_add_rms_norm);
+  m.impl("silu_and_mul", &vllm_amd::silu_and_mul);
+  m.impl("gelu_and_mul", &vllm_amd::gelu_and_mul);
+  m.impl("rotary_embedding", &vllm_amd::rotary_embedding);
+  m.impl("reshape_and_cache", &vllm_amd::reshape_and_cache);
+  m.impl("paged_decode_attention", &vllm_amd::paged_decode_attention);
+  m.impl("prefill_attention", &vllm_amd::prefill_attention);
+}

@@ -62,6 +62,8 @@ class Attention(nn.Module):
             self.scale,
             num_decodes=meta.num_decodes,
             sliding_window=self.sliding_window,
+            max_seq_len=meta.max_seq_len,
+            max_query_len=meta.max_query_len,
         )
         return out.reshape(T, -1)
 

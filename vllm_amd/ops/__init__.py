@@ -88,16 +88,20 @@ def attention_unified(
     scale,
     num_decodes: int = 0,
     sliding_window: int = 0,
+    max_seq_len: int = 0,
+    max_query_len: int = 0,
 ):
     """Varlen attention over the paged KV cache (prefill + decode mixed).
 
     q: [num_tokens, num_heads, head_dim]; kv_cache: [2, num_blocks,
-    block_size, num_kv_heads, head_dim]; returns [num_tokens, num_heads,
+    num_kv_heads, block_size, head_dim]; returns [num_tokens, num_heads,
     head_dim]. Requests are ordered decodes-first (query_len==1).
+    max_seq_len/max_query_len are CPU-known hints so the GPU path never
+    syncs to size its scratch.
     """
     return get_backend(q.device).attention_unified(
         q, kv_cache, block_table, query_start_loc, seq_lens, scale,
-        num_decodes, sliding_window,
+        num_decodes, sliding_window, max_seq_len, max_query_len,
     )
 
 

@@ -73,24 +73,26 @@ def gelu_and_mul(x: torch.Tensor) -> torch.Tensor:
 
 
 def reshape_and_cache(key, value, kv_cache, slot_mapping):
-    """key/value: [T, Hkv, D]; kv_cache: [2, num_blocks, block_size, Hkv, D];
-    slot_mapping: [T] int64 (block_id * block_size + offset)."""
-    block_size = kv_cache.shape[2]
+    """key/value: [T, Hkv, D]; kv_cache: [2, num_blocks, Hkv, block_size, D]
+    (head-major inside the block — each (block, head) tile is one contiguous
+    chunk for the HIP kernels); slot_mapping: [T] int64
+    (block_id * block_size + offset)."""
+    block_size = kv_cache.shape[3]
     block_ids = slot_mapping // block_size
     offsets = slot_mapping % block_size
-    kv_cache[0, block_ids, offsets] = key
-    kv_cache[1, block_ids, offsets] = value
+    kv_cache[0, block_ids, :, offsets] = key
+    kv_cache[1, block_ids, :, offsets] = value
 
 
 def _gather_kv(kv_cache, block_table_row, ctx_len):
     """Gather [ctx_len, Hkv, D] K and V for one request from paged cache."""
-    block_size = kv_cache.shape[2]
+    block_size = kv_cache.shape[3]
     num_blocks_needed = (ctx_len + block_size - 1) // block_size
     blocks = block_table_row[:num_blocks_needed].long()
-    k = kv_cache[0, blocks]  # [nb, bs, Hkv, D]
+    k = kv_cache[0, blocks]  # [nb, Hkv, bs, D]
     v = kv_cache[1, blocks]
-    k = k.reshape(-1, *k.shape[2:])[:ctx_len]
-    v = v.reshape(-1, *v.shape[2:])[:ctx_len]
+    k = k.transpose(1, 2).reshape(-1, k.shape[1], k.shape[3])[:ctx_len]
+    v = v.transpose(1, 2).reshape(-1, v.shape[1], v.shape[3])[:ctx_len]
     return k, v
 
 
@@ -103,9 +105,11 @@ def attention_unified(
     scale,
     num_decodes=0,
     sliding_window=0,
+    max_seq_len=0,
+    max_query_len=0,
 ):
     num_tokens, num_heads, head_dim = q.shape
-    num_kv_heads = kv_cache.shape[3]
+    num_kv_heads = kv_cache.shape[2]
     group = num_heads // num_kv_heads
     head_dim_v = kv_cache.shape[4]
     out = q.new_empty(num_tokens, num_heads, head_dim_v)

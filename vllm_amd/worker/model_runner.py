@@ -76,9 +76,12 @@ class ModelRunner:
 
         spec = self.spec
         kv_heads = spec.num_kv_heads_per_rank(get_tp_world_size())
+        # Head-major block layout: each (block, head) KV tile is one
+        # contiguous block_size*head_dim chunk (16 KB at 64x128 bf16) —
+        # the unit the HIP attention kernels read/stage.
         self.kv_caches = [
             torch.zeros(
-                2, num_blocks, self.block_size, kv_heads, spec.head_dim,
+                2, num_blocks, kv_heads, self.block_size, spec.head_dim,
                 dtype=self.dtype, device=self.device,
             )
             for _ in range(spec.num_layers)
