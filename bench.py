@@ -1,0 +1,178 @@
+"""Flagship serving benchmark (driver contract).
+
+Measures whole-node output tokens/sec on the BASELINE.json headline
+config: Llama-3-8B bf16, synthetic random prompts of the CI benchmark
+shape (random 128-token inputs), random-init (dummy) weights, continuous
+decode at a fixed running batch.
+
+    python bench.py --gpus N --steps K --warmup W
+
+For N>1 the driver launches this under torch.distributed.run with one
+rank per GPU; ranks form a TP=N group over RCCL/xGMI (the reference's
+headline TP configs — BASELINE.md). One "step" is one engine iteration:
+after warmup (which absorbs the prefill of the whole batch) every step
+is a full-batch decode producing `batch` output tokens.
+
+Timing: barrier + torch.cuda.synchronize on both sides of exactly
+--steps engine steps; elapsed is MAX over ranks; rank 0 prints one JSON
+line.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import numpy as np
+import torch
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=32)
+    p.add_argument("--warmup", type=int, default=8)
+    p.add_argument("--model", type=str, default="llama-3-8b")
+    p.add_argument("--batch", type=int, default=256,
+                   help="running batch (= concurrent sequences)")
+    p.add_argument("--input-len", type=int, default=128,
+                   help="synthetic prompt length (CI 'random 128' shape)")
+    p.add_argument("--dtype", type=str, default="bf16")
+    p.add_argument("--block-size", type=int, default=64)
+    p.add_argument("--num-gpu-blocks", type=int, default=None,
+                   help="skip the profiling pass with a fixed KV pool size")
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    on_gpu = torch.cuda.is_available()
+    if world > 1:
+        assert on_gpu, "multi-rank bench requires GPUs"
+
+    if not on_gpu and args.model == "llama-3-8b":
+        # CPU plumbing check only (no GPU in the build container).
+        args.model = "tiny-llama"
+        args.dtype = "fp32"
+        args.batch = min(args.batch, 8)
+        args.block_size = 16
+
+    from vllm_amd.config import (
+        CacheConfig, DeviceConfig, EngineConfig, ModelConfig,
+        ParallelConfig, SchedulerConfig,
+    )
+    from vllm_amd.engine.core import EngineCore
+    from vllm_amd.request import Request
+    from vllm_amd.sampling_params import SamplingParams
+
+    prefill_tokens = args.batch * args.input_len
+    config = EngineConfig(
+        model_config=ModelConfig(
+            model=args.model, dtype=args.dtype,
+            max_model_len=args.input_len + args.warmup + args.steps + 16,
+            load_format="dummy",
+        ),
+        cache_config=CacheConfig(
+            block_size=args.block_size,
+            num_gpu_blocks=args.num_gpu_blocks,
+            enable_prefix_caching=False,
+        ),
+        scheduler_config=SchedulerConfig(
+            max_num_batched_tokens=max(prefill_tokens, 8192),
+            max_num_seqs=args.batch,
+            enable_chunked_prefill=True,
+        ),
+        parallel_config=ParallelConfig(tensor_parallel_size=world),
+        device_config=DeviceConfig(device="cuda" if on_gpu else "cpu"),
+    )
+
+    engine = EngineCore(config)
+    is_driver = rank == 0
+
+    def sync():
+        if world > 1:
+            torch.distributed.barrier()
+        if on_gpu:
+            torch.cuda.synchronize()
+
+    total_steps = args.warmup + args.steps
+    if is_driver:
+        rng = np.random.default_rng(0)
+        params = SamplingParams(
+            temperature=0.0,
+            max_tokens=total_steps + 8,
+            ignore_eos=True,
+            detokenize=False,
+        )
+        for i in range(args.batch):
+            toks = rng.integers(
+                16, config.model_config.spec.vocab_size - 16,
+                size=args.input_len,
+            ).tolist()
+            engine.add_request(Request(
+                request_id=f"r{i}", prompt_token_ids=toks,
+                sampling_params=params, arrival_time=time.time(),
+            ))
+
+        for _ in range(args.warmup):
+            engine.step()
+        sync()
+        t0 = time.perf_counter()
+        for _ in range(args.steps):
+            engine.step()
+        sync()
+        t1 = time.perf_counter()
+        elapsed = t1 - t0
+    else:
+        for _ in range(args.warmup):
+            engine.step_worker()
+        sync()
+        t0 = time.perf_counter()
+        for _ in range(args.steps):
+            engine.step_worker()
+        sync()
+        t1 = time.perf_counter()
+        elapsed = t1 - t0
+
+    # MAX elapsed over ranks.
+    if world > 1:
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device="cuda" if on_gpu else "cpu")
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    if is_driver:
+        out_tokens = args.batch * args.steps
+        result = {
+            "metric": "output tokens/sec (whole node)",
+            "value": round(out_tokens / elapsed, 2),
+            "unit": "tokens/s",
+            "n_gpus": world if on_gpu else 0,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000, 3),
+            "higher_is_better": True,
+            "scaling": "strong",
+            "vs_baseline": None,
+            "dtype": args.dtype,
+            "data": "synthetic (random 128-token prompts, random-init "
+                    "dummy weights; no network)",
+            "config": {
+                "model": args.model,
+                "global_batch": args.batch,
+                "seq_len": args.input_len,
+                "parallelism": f"tp{world}",
+            },
+        }
+        print(json.dumps(result))
+    engine.shutdown()
+    if world > 1:
+        torch.distributed.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

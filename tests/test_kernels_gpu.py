@@ -1,0 +1,216 @@
+"""Numerics tests for the hand-written CDNA4 HIP kernels.
+
+Every kernel is compared against the plain-PyTorch fp32 reference in
+vllm_amd/ops/_torch_ref.py (the pattern of the reference repo's
+tests/kernels/). Runs only on an MI355X (-m gpu).
+"""
+
+import pytest
+import torch
+
+from vllm_amd.ops import _torch_ref as ref
+
+pytestmark = pytest.mark.gpu
+
+
+def _hip():
+    from vllm_amd.ops import hip_ops
+
+    return hip_ops
+
+
+def assert_close(hip_out, ref_out, atol=2e-2, rtol=2e-2, msg=""):
+    hip_f = hip_out.float()
+    ref_f = ref_out.float()
+    torch.testing.assert_close(hip_f, ref_f, atol=atol, rtol=rtol, msg=msg)
+
+
+@pytest.fixture(autouse=True)
+def _seed():
+    torch.manual_seed(0)
+
+
+@pytest.mark.parametrize("shape", [(1, 128), (17, 4096), (256, 8192)])
+def test_rms_norm(shape):
+    hip = _hip()
+    x = torch.randn(shape, dtype=torch.bfloat16, device="cuda")
+    w = torch.randn(shape[-1], dtype=torch.bfloat16, device="cuda")
+    out = hip.rms_norm(x, w, 1e-5)
+    expect = ref.rms_norm(x.float(), w.float(), 1e-5)
+    assert_close(out, expect)
+
+
+@pytest.mark.parametrize("shape", [(3, 512), (129, 4096)])
+def test_fused_add_rms_norm(shape):
+    hip = _hip()
+    x = torch.randn(shape, dtype=torch.bfloat16, device="cuda")
+    res = torch.randn(shape, dtype=torch.bfloat16, device="cuda")
+    w = torch.randn(shape[-1], dtype=torch.bfloat16, device="cuda")
+    ex_out, ex_res = ref.fused_add_rms_norm(
+        x.clone().float(), res.clone().float(), w.float(), 1e-5
+    )
+    out, new_res = hip.fused_add_rms_norm(x, res, w, 1e-5)
+    assert_close(new_res, ex_res)
+    assert_close(out, ex_out)
+
+
+@pytest.mark.parametrize("act", ["silu", "gelu"])
+@pytest.mark.parametrize("shape", [(5, 256), (300, 28672)])
+def test_act_and_mul(act, shape):
+    hip = _hip()
+    x = torch.randn(shape, dtype=torch.bfloat16, device="cuda")
+    fn = hip.silu_and_mul if act == "silu" else hip.gelu_and_mul
+    rfn = ref.silu_and_mul if act == "silu" else ref.gelu_and_mul
+    assert_close(fn(x), rfn(x.float()))
+
+
+@pytest.mark.parametrize("rotary_dim", [128, 64])
+def test_rope_neox(rotary_dim):
+    hip = _hip()
+    T, Hq, Hkv, D = 33, 8, 2, 128
+    max_pos = 2048
+    inv = 1.0 / (10000.0 ** (
+        torch.arange(0, rotary_dim, 2).float() / rotary_dim))
+    t = torch.arange(max_pos).float()
+    freqs = torch.outer(t, inv)
+    cache = torch.cat([freqs.cos(), freqs.sin()], dim=-1).cuda()
+    pos = torch.randint(0, max_pos, (T,), device="cuda")
+    q = torch.randn(T, Hq, D, dtype=torch.bfloat16, device="cuda")
+    k = torch.randn(T, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    eq, ek = ref.apply_rope(pos, q.clone().float(), k.clone().float(),
+                            cache, rotary_dim)
+    hq, hk = hip.apply_rope(pos, q, k, cache, rotary_dim)
+    assert_close(hq, eq)
+    assert_close(hk, ek)
+
+
+def test_reshape_and_cache():
+    hip = _hip()
+    T, Hkv, D, BS, NB = 200, 8, 128, 64, 32
+    k = torch.randn(T, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    v = torch.randn(T, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    cache = torch.zeros(2, NB, Hkv, BS, D, dtype=torch.bfloat16,
+                        device="cuda")
+    ref_cache = cache.clone()
+    slots = torch.randperm(NB * BS, device="cuda")[:T]
+    hip.reshape_and_cache(k, v, cache, slots)
+    ref.reshape_and_cache(k, v, ref_cache, slots)
+    torch.testing.assert_close(cache, ref_cache)
+
+
+def _make_paged(num_reqs, q_lens, ctx_lens, Hq, Hkv, D=128, BS=64):
+    """Build q + populated paged cache + metadata for attention tests."""
+    total_q = sum(q_lens)
+    max_blocks = max((c + BS - 1) // BS for c in ctx_lens)
+    num_blocks = num_reqs * max_blocks + 1
+    q = torch.randn(total_q, Hq, D, dtype=torch.bfloat16, device="cuda")
+    cache = torch.randn(2, num_blocks, Hkv, BS, D, dtype=torch.bfloat16,
+                        device="cuda")
+    block_table = torch.zeros(num_reqs, max_blocks, dtype=torch.int32,
+                              device="cuda")
+    nxt = 1
+    for i, c in enumerate(ctx_lens):
+        nb = (c + BS - 1) // BS
+        block_table[i, :nb] = torch.arange(nxt, nxt + nb)
+        nxt += nb
+    qsl = torch.zeros(num_reqs + 1, dtype=torch.int32, device="cuda")
+    qsl[1:] = torch.cumsum(
+        torch.tensor(q_lens, device="cuda", dtype=torch.int32), 0)
+    seq_lens = torch.tensor(ctx_lens, dtype=torch.int32, device="cuda")
+    return q, cache, block_table, qsl, seq_lens
+
+
+@pytest.mark.parametrize("group", [1, 4, 8])
+@pytest.mark.parametrize("ctx", [1, 63, 64, 200, 511, 513, 2000])
+def test_decode_attention(group, ctx):
+    hip = _hip()
+    Hkv = 2
+    Hq = Hkv * group
+    n = 4
+    q, cache, bt, qsl, sl = _make_paged(n, [1] * n, [ctx] * n, Hq, Hkv)
+    scale = 1.0 / 128**0.5
+    out = hip.attention_unified(q, cache, bt, qsl, sl, scale,
+                                num_decodes=n, max_seq_len=ctx,
+                                max_query_len=1)
+    expect = ref.attention_unified(q, cache, bt, qsl, sl, scale,
+                                   num_decodes=n)
+    assert_close(out, expect, msg=f"group={group} ctx={ctx}")
+
+
+@pytest.mark.parametrize("group", [4])
+@pytest.mark.parametrize(
+    "q_lens,ctx_lens",
+    [
+        ([16, 33], [16, 33]),            # pure prefill from scratch
+        ([64, 128], [64, 128]),
+        ([17, 70], [100, 300]),           # chunked-prefill continuation
+        ([1, 1, 5, 200], [40, 513, 60, 200]),  # mixed decode+prefill
+    ],
+)
+def test_prefill_attention(group, q_lens, ctx_lens):
+    hip = _hip()
+    Hkv = 2
+    Hq = Hkv * group
+    n = len(q_lens)
+    num_decodes = sum(1 for x in q_lens if x == 1)
+    # decodes-first ordering as the model runner produces
+    order = sorted(range(n), key=lambda i: q_lens[i] != 1)
+    q_lens = [q_lens[i] for i in order]
+    ctx_lens = [ctx_lens[i] for i in order]
+    q, cache, bt, qsl, sl = _make_paged(n, q_lens, ctx_lens, Hq, Hkv)
+    scale = 1.0 / 128**0.5
+    out = hip.attention_unified(
+        q, cache, bt, qsl, sl, scale, num_decodes=num_decodes,
+        max_seq_len=max(ctx_lens), max_query_len=max(q_lens))
+    expect = ref.attention_unified(q, cache, bt, qsl, sl, scale,
+                                   num_decodes=num_decodes)
+    assert_close(out, expect)
+
+
+def test_model_forward_matches_cpu_reference():
+    """Tiny Llama forward on GPU (bf16, HIP kernels) vs CPU fp32 torch
+    reference: logits must be directionally identical (cosine > 0.99)."""
+    from vllm_amd.config import ModelConfig
+    from vllm_amd.models.registry import load_model
+    from vllm_amd.worker.forward_context import (
+        AttentionMetadata, ForwardContext, set_forward_context)
+
+    def run(device, dtype):
+        torch.manual_seed(0)
+        cfg = ModelConfig(model="tiny-llama-128", dtype=dtype,
+                          load_format="dummy")
+        model = load_model(cfg, torch.device(device))
+        T = 9
+        ids = torch.arange(1, T + 1, device=device)
+        pos = torch.arange(T, device=device)
+        spec = cfg.spec
+        caches = [
+            torch.zeros(2, 4, spec.num_kv_heads, 64, spec.head_dim,
+                        dtype=cfg.torch_dtype, device=device)
+            for _ in range(spec.num_layers)
+        ]
+        meta = AttentionMetadata(
+            query_start_loc=torch.tensor([0, T], dtype=torch.int32,
+                                         device=device),
+            seq_lens=torch.tensor([T], dtype=torch.int32, device=device),
+            block_table=torch.tensor([[1]], dtype=torch.int32,
+                                     device=device),
+            slot_mapping=torch.arange(64, 64 + T, dtype=torch.int64,
+                                      device=device),
+            num_reqs=1,
+            num_actual_tokens=T,
+            max_query_len=T,
+            max_seq_len=T,
+            num_decodes=0,
+        )
+        ctx = ForwardContext(attn_metadata=meta, kv_caches=caches)
+        with set_forward_context(ctx), torch.inference_mode():
+            h = model(ids, pos)
+            logits = model.compute_logits(h)
+        return logits.float().cpu()
+
+    gpu = run("cuda", "bf16")
+    cpu = run("cpu", "fp32")
+    cos = torch.nn.functional.cosine_similarity(gpu, cpu, dim=-1)
+    assert cos.min() > 0.99, cos
+    assert (gpu.argmax(-1) == cpu.argmax(-1)).float().mean() > 0.8

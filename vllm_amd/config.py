@@ -187,6 +187,22 @@ MODEL_PRESETS: dict[str, ModelSpec] = {
         max_position_embeddings=2048,
         eos_token_id=2,
     ),
+    # Tiny model with the HIP kernels' native geometry (head_dim=128) —
+    # used by GPU e2e tests and smoke runs.
+    "tiny-llama-128": ModelSpec(
+        name="tiny-llama-128",
+        architecture="llama",
+        vocab_size=1024,
+        hidden_size=512,
+        intermediate_size=1024,
+        num_layers=2,
+        num_heads=4,
+        num_kv_heads=2,
+        head_dim=128,
+        rope_theta=10000.0,
+        max_position_embeddings=2048,
+        eos_token_id=2,
+    ),
     "tiny-mixtral": ModelSpec(
         name="tiny-mixtral",
         architecture="mixtral",

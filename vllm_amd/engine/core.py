@@ -86,14 +86,22 @@ class EngineCore:
             scheduler_output, runner_output
         )
 
+    def step_worker(self) -> bool:
+        """Non-driver ranks: receive ONE SchedulerOutput broadcast and
+        execute it collectively. Returns False on the shutdown sentinel.
+        Exactly one broadcast happens per driver step(), so counted
+        lockstep loops (bench.py) stay in sync by construction."""
+        assert not self.is_driver
+        so = self.world.broadcast_object(None, src=0)
+        if so is None:
+            return False
+        self.worker.execute_model(so)
+        return True
+
     def run_spmd_worker_loop(self) -> None:
         """Non-driver ranks: receive scheduler outputs forever."""
-        assert not self.is_driver
-        while True:
-            so = self.world.broadcast_object(None, src=0)
-            if so is None:  # shutdown sentinel
-                return
-            self.worker.execute_model(so)
+        while self.step_worker():
+            pass
 
     def shutdown(self) -> None:
         if self.is_driver and self.world.world_size > 1:
