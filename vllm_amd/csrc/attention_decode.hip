@@ -51,6 +51,7 @@ __global__ __launch_bounds__(DEC_BLOCK) void paged_decode_kernel(
     const float scale, const int num_q_heads, const int num_kv_heads,
     const int head_dim, const int max_blocks_per_req, const int num_parts,
     const int64_t kv_plane_stride,      // elements between K and V planes
+    const int64_t q_stride,             // elements between q token rows
     const int sliding_window) {
   const int seq = blockIdx.x;       // decode row == request row
   const int kvh = blockIdx.y;
@@ -97,7 +98,7 @@ __global__ __launch_bounds__(DEC_BLOCK) void paged_decode_kernel(
   float qf[GROUP][8];
   {
     const short* qbase =
-        q + ((int64_t)seq * num_q_heads + hq0) * head_dim + lane16 * 8;
+        q + (int64_t)seq * q_stride + hq0 * head_dim + lane16 * 8;
 #pragma unroll
     for (int g = 0; g < GROUP; ++g) {
       s16x8 v = *reinterpret_cast<const s16x8*>(qbase + g * head_dim);
@@ -284,6 +285,9 @@ void paged_decode_attention(torch::Tensor out, torch::Tensor q,
   const int num_kv_heads = kv_cache.size(2);
   const int group = num_q_heads / num_kv_heads;
   TORCH_CHECK(head_dim == 128, "decode kernel supports head_dim=128");
+  TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == head_dim &&
+              q.stride(0) % 8 == 0, "q must be head-contiguous [T,H,D]");
+  TORCH_CHECK(out.is_contiguous(), "out must be contiguous");
   TORCH_CHECK(kv_cache.size(3) == 64, "cache block_size must be 64");
   TORCH_CHECK(group >= 1 && group <= MAX_GROUP && num_q_heads ==
               group * num_kv_heads, "GQA group must be 1..8, got ", group);
@@ -305,7 +309,7 @@ void paged_decode_attention(torch::Tensor out, torch::Tensor q,
                      block_table.data_ptr<int>(), seq_lens.data_ptr<int>(),  \
                      (float)scale, num_q_heads, num_kv_heads, head_dim,      \
                      max_blocks, num_parts, kv_cache.stride(0),              \
-                     (int)sliding_window)
+                     q.stride(0), (int)sliding_window)
 
 #define DISPATCH_GROUP(TAG, FIN)                                             \
   switch (group) {                                                           \

@@ -48,7 +48,8 @@ __global__ __launch_bounds__(PF_BLOCK) void prefill_attention_kernel(
     const int* __restrict__ seq_lens,         // [num_reqs]
     const float scale, const int num_q_heads, const int num_kv_heads,
     const int max_blocks_per_req, const int num_decodes,
-    const int64_t kv_plane_stride, const int sliding_window) {
+    const int64_t kv_plane_stride, const int64_t q_stride,
+    const int sliding_window) {
   const int req = num_decodes + blockIdx.z;
   const int h = blockIdx.y;
   const int kvh = h / (num_q_heads / num_kv_heads);
@@ -76,7 +77,7 @@ __global__ __launch_bounds__(PF_BLOCK) void prefill_attention_kernel(
   s16x8 qfrag[4];
   {
     const short* qbase =
-        q + ((int64_t)(q_start + q_row_safe) * num_q_heads + h) * HEAD_DIM;
+        q + (int64_t)(q_start + q_row_safe) * q_stride + (int64_t)h * HEAD_DIM;
 #pragma unroll
     for (int kk = 0; kk < 4; ++kk)
       qfrag[kk] =
@@ -260,6 +261,9 @@ void prefill_attention(torch::Tensor out, torch::Tensor q,
   const int head_dim = q.size(2);
   const int num_kv_heads = kv_cache.size(2);
   TORCH_CHECK(head_dim == HEAD_DIM, "prefill kernel supports head_dim=128");
+  TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == head_dim &&
+              q.stride(0) % 8 == 0, "q must be head-contiguous [T,H,D]");
+  TORCH_CHECK(out.is_contiguous(), "out must be contiguous");
   TORCH_CHECK(kv_cache.size(3) == KVTILE, "cache block_size must be 64");
   TORCH_CHECK(num_q_heads % num_kv_heads == 0);
   const int max_tiles = (int)((max_query_len + QTILE - 1) / QTILE);
@@ -276,7 +280,7 @@ void prefill_attention(torch::Tensor out, torch::Tensor q,
                      seq_lens.data_ptr<int>(), (float)scale, num_q_heads,    \
                      num_kv_heads, (int)block_table.size(1),                 \
                      (int)num_decodes, kv_cache.stride(0),                   \
-                     (int)sliding_window)
+                     q.stride(0), (int)sliding_window)
 
   if (q.scalar_type() == torch::kBFloat16) { LAUNCH_PF(BF16Tag); }
   else                                     { LAUNCH_PF(FP16Tag); }

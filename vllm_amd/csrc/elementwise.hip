@@ -152,7 +152,8 @@ __global__ void rope_kernel(short* __restrict__ q, short* __restrict__ k,
                             const int64_t* __restrict__ positions,
                             const float* __restrict__ cos_sin,
                             const int rot, const int head_dim,
-                            const int num_q_heads, const int num_kv_heads) {
+                            const int num_q_heads, const int num_kv_heads,
+                            const int64_t q_stride, const int64_t k_stride) {
   const int64_t t = blockIdx.x;
   const int64_t pos = positions[t];
   const float* cs = cos_sin + pos * rot;
@@ -164,9 +165,9 @@ __global__ void rope_kernel(short* __restrict__ q, short* __restrict__ k,
     const int p = idx % half;
     short* base;
     if (h < num_q_heads) {
-      base = q + (t * num_q_heads + h) * head_dim;
+      base = q + t * q_stride + (int64_t)h * head_dim;
     } else {
-      base = k + (t * num_kv_heads + (h - num_q_heads)) * head_dim;
+      base = k + t * k_stride + (int64_t)(h - num_q_heads) * head_dim;
     }
     const float c = cs[p];
     const float s = cs[half + p];
@@ -188,17 +189,17 @@ __global__ void reshape_and_cache_kernel(
     const short* __restrict__ key, const short* __restrict__ value,
     short* __restrict__ kv_cache, const int64_t* __restrict__ slot_mapping,
     const int64_t kv_stride,  // elements between K and V planes
+    const int64_t k_row_stride, const int64_t v_row_stride,
     const int num_kv_heads, const int head_dim, const int block_size) {
   const int64_t t = blockIdx.x;
   const int64_t slot = slot_mapping[t];
   if (slot < 0) return;  // padding token
   const int64_t blk = slot / block_size;
   const int off = (int)(slot % block_size);
-  const int row_elems = num_kv_heads * head_dim;
-  const short* krow = key + t * row_elems;
-  const short* vrow = value + t * row_elems;
+  const short* krow = key + t * k_row_stride;
+  const short* vrow = value + t * v_row_stride;
   const int dvec = head_dim / 8;
-  const int nvec = row_elems / 8;
+  const int nvec = (num_kv_heads * head_dim) / 8;
   for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
     const int h = i / dvec;
     const int dv = i % dvec;
@@ -217,6 +218,19 @@ __global__ void reshape_and_cache_kernel(
 
 static inline void check_16b(const torch::Tensor& t, const char* name) {
   TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+  TORCH_CHECK(
+      t.scalar_type() == torch::kBFloat16 || t.scalar_type() == torch::kHalf,
+      name, " must be bf16 or fp16");
+}
+
+// [T, H, D] view whose heads are contiguous within a row but whose rows may
+// be strided (a head slice of the fused QKV projection output). Rows must
+// stay 16-byte aligned for the short8 loads.
+static inline void check_rows_16b(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.dim() == 3, name, " must be [T, H, D]");
+  TORCH_CHECK(t.stride(2) == 1 && t.stride(1) == t.size(2), name,
+              " heads must be contiguous within a row");
+  TORCH_CHECK(t.stride(0) % 8 == 0, name, " row stride must be 16B-aligned");
   TORCH_CHECK(
       t.scalar_type() == torch::kBFloat16 || t.scalar_type() == torch::kHalf,
       name, " must be bf16 or fp16");
@@ -308,9 +322,8 @@ torch::Tensor gelu_and_mul(torch::Tensor x) {
 void rotary_embedding(torch::Tensor positions, torch::Tensor q,
                       torch::Tensor k, torch::Tensor cos_sin_cache,
                       int64_t rot) {
-  check_16b(q, "q");
-  check_16b(k, "k");
-  TORCH_CHECK(q.dim() == 3 && k.dim() == 3, "q/k must be [T, H, D]");
+  check_rows_16b(q, "q");
+  check_rows_16b(k, "k");
   TORCH_CHECK(cos_sin_cache.scalar_type() == torch::kFloat32);
   const int T = q.size(0);
   const int num_q_heads = q.size(1);
@@ -322,7 +335,7 @@ void rotary_embedding(torch::Tensor positions, torch::Tensor q,
                      (short*)q.data_ptr(), (short*)k.data_ptr(),           \
                      positions.data_ptr<int64_t>(),                        \
                      cos_sin_cache.data_ptr<float>(), (int)rot, head_dim,  \
-                     num_q_heads, num_kv_heads)
+                     num_q_heads, num_kv_heads, q.stride(0), k.stride(0))
   DISPATCH_16B(q, LAUNCH_ROPE);
 #undef LAUNCH_ROPE
   HIP_CHECK_KERNEL();
@@ -330,7 +343,8 @@ void rotary_embedding(torch::Tensor positions, torch::Tensor q,
 
 void reshape_and_cache(torch::Tensor key, torch::Tensor value,
                        torch::Tensor kv_cache, torch::Tensor slot_mapping) {
-  check_16b(key, "key");
+  check_rows_16b(key, "key");
+  check_rows_16b(value, "value");
   check_16b(kv_cache, "kv_cache");
   const int T = key.size(0);
   if (T == 0) return;
@@ -348,6 +362,7 @@ void reshape_and_cache(torch::Tensor key, torch::Tensor value,
                      (const short*)value.data_ptr(),                   \
                      (short*)kv_cache.data_ptr(),                      \
                      slot_mapping.data_ptr<int64_t>(), kv_stride,      \
+                     key.stride(0), value.stride(0),                    \
                      num_kv_heads, head_dim, block_size)
   DISPATCH_16B(key, LAUNCH_RC);
 #undef LAUNCH_RC

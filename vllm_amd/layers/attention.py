@@ -37,13 +37,20 @@ class Attention(nn.Module):
     def forward(
         self, q: torch.Tensor, k: torch.Tensor, v: torch.Tensor
     ) -> torch.Tensor:
-        """q: [T, Hq*D], k/v: [T, Hkv*D] -> [T, Hq*D]."""
+        """q: [T, Hq, D] or [T, Hq*D]; k/v likewise -> [T, Hq*D].
+
+        Inputs may be strided head-slices of the fused QKV projection —
+        the HIP kernels take row strides, so no contiguity copies here.
+        """
         ctx = get_forward_context()
         meta = ctx.attn_metadata
         T = q.shape[0]
-        q = q.view(T, self.num_heads, self.head_dim)
-        k = k.view(T, self.num_kv_heads, self.head_dim)
-        v = v.view(T, self.num_kv_heads, self.head_dim)
+        if q.dim() == 2:
+            q = q.view(T, self.num_heads, self.head_dim)
+        if k.dim() == 2:
+            k = k.view(T, self.num_kv_heads, self.head_dim)
+        if v.dim() == 2:
+            v = v.view(T, self.num_kv_heads, self.head_dim)
 
         if not ctx.kv_caches:
             # Memory-profiling run: no cache allocated; compute attention

@@ -86,7 +86,7 @@ class LlamaAttention(nn.Module):
         q = q.view(T, self.num_heads, self.head_dim)
         k = k.view(T, self.num_kv_heads, self.head_dim)
         self.rotary_emb(positions, q, k)
-        out = self.attn(q.reshape(T, -1), k.reshape(T, -1), v)
+        out = self.attn(q, k, v)
         return self.o_proj(out)
 
 

@@ -71,7 +71,11 @@ def attention_unified(
     kernel. max_seq_len/max_query_len come from the scheduler (CPU) so
     no device sync is needed here."""
     num_tokens, num_heads, head_dim = q.shape
-    out = torch.empty_like(q)
+    # q may be a strided head-slice of the fused QKV output; the kernels
+    # take its row stride. The output is always freshly contiguous.
+    out = torch.empty(
+        (num_tokens, num_heads, head_dim), dtype=q.dtype, device=q.device
+    )
     if max_seq_len <= 0:
         max_seq_len = int(seq_lens.max().item())
     num_reqs = seq_lens.shape[0]
