@@ -77,15 +77,13 @@ __global__ __launch_bounds__(DEC_BLOCK) void paged_decode_kernel(
   }
 
   const int lane16 = threadIdx.x & 15;        // dim-slice owner
-  const int tok_slot = threadIdx.x >> 4;      // 0..15: token within iter
-  const int dvec = head_dim / 8;              // short8 vectors per row (16)
+  const int slot = (threadIdx.x >> 4) & 3;    // token slot within wave
+  const int wave = threadIdx.x / WAVE_SIZE;   // 0..3
 
-  // LDS: physical block ids for this partition + score/accum scratch.
+  // LDS: physical block ids + cross-wave merge scratch.
   __shared__ int blk_ids[DEC_PART / 64];
-  __shared__ float scores[GROUP][DEC_PART];
-  __shared__ float red_scratch[DEC_BLOCK / WAVE_SIZE];
-  // Cross-wave O reduction: [4 waves][GROUP][D] reused after score phase.
-  __shared__ float o_red[4][GROUP][128];
+  __shared__ float ml_red[4][GROUP][2];       // per-wave (m, l)
+  __shared__ float o_red[4][GROUP][128];      // per-wave accumulators
 
   if (threadIdx.x < DEC_PART / 64) {
     const int cache_blk = (part * DEC_PART) / 64 + threadIdx.x;
@@ -103,136 +101,124 @@ __global__ __launch_bounds__(DEC_BLOCK) void paged_decode_kernel(
     for (int g = 0; g < GROUP; ++g) {
       s16x8 v = *reinterpret_cast<const s16x8*>(qbase + g * head_dim);
 #pragma unroll
-      for (int j = 0; j < 8; ++j) qf[g][j] = to_f32<Tag>(v[j]);
+      for (int j = 0; j < 8; ++j) qf[g][j] = to_f32<Tag>(v[j]) * scale;
     }
   }
 
-  // ---- K pass: scores[g][t] = scale * q_g . k_t --------------------------
+  // ---- single pass: online softmax per (wave, slot) ----------------------
+  // Each slot streams tokens t = t_begin + wave*4 + slot + 16*i. K and V
+  // rows are read once, 16 B per lane (G13); no syncthreads in the loop, so
+  // the 4 waves keep the HBM pipeline full independently.
   const int64_t head_tile_stride = (int64_t)64 * head_dim;  // one (blk,head)
-  for (int t = t_begin + tok_slot; t < t_end; t += 16) {
-    const int local = t - part * DEC_PART;
-    const int phys = blk_ids[local / 64];
-    const short* krow = kv_cache +
-        ((int64_t)phys * num_kv_heads + kvh) * head_tile_stride +
-        (int64_t)(t % 64) * head_dim + lane16 * 8;
-    s16x8 kv = *reinterpret_cast<const s16x8*>(krow);
-    float kf[8];
-#pragma unroll
-    for (int j = 0; j < 8; ++j) kf[j] = to_f32<Tag>(kv[j]);
-#pragma unroll
-    for (int g = 0; g < GROUP; ++g) {
-      float s = 0.f;
-#pragma unroll
-      for (int j = 0; j < 8; ++j) s += qf[g][j] * kf[j];
-      // Reduce over the 16 dim-slice lanes.
-#pragma unroll
-      for (int off = 8; off > 0; off >>= 1) s += __shfl_xor(s, off, 64);
-      if (lane16 == g % 16) scores[g][t - t_begin] = s * scale;
-    }
-  }
-  __syncthreads();
-
-  const int n_tok = t_end - t_begin;
-
-  // ---- softmax over the partition (block-wide, per head) -----------------
-  float m_g[GROUP], l_g[GROUP];
+  float m_s[GROUP], l_s[GROUP], acc[GROUP][8];
 #pragma unroll
   for (int g = 0; g < GROUP; ++g) {
-    float m = -3.0e38f;
-    for (int i = threadIdx.x; i < n_tok; i += DEC_BLOCK)
-      m = fmaxf(m, scores[g][i]);
-#pragma unroll
-    for (int off = 32; off > 0; off >>= 1)
-      m = fmaxf(m, __shfl_xor(m, off, 64));
-    if ((threadIdx.x & 63) == 0) red_scratch[threadIdx.x >> 6] = m;
-    __syncthreads();
-    m = fmaxf(fmaxf(red_scratch[0], red_scratch[1]),
-              fmaxf(red_scratch[2], red_scratch[3]));
-    float l = 0.f;
-    for (int i = threadIdx.x; i < n_tok; i += DEC_BLOCK) {
-      const float p = __expf(scores[g][i] - m);
-      scores[g][i] = p;
-      l += p;
-    }
-    __syncthreads();  // red_scratch reuse
-    l = block_reduce_sum<DEC_BLOCK / WAVE_SIZE>(l, red_scratch);
-    m_g[g] = m;
-    l_g[g] = l;
-    __syncthreads();
-  }
-
-  // ---- V pass: acc[g][j] += p[t] * v_t[dims] -----------------------------
-  float acc[GROUP][8];
-#pragma unroll
-  for (int g = 0; g < GROUP; ++g)
+    m_s[g] = -3.0e38f;
+    l_s[g] = 0.f;
 #pragma unroll
     for (int j = 0; j < 8; ++j) acc[g][j] = 0.f;
+  }
 
-  for (int t = t_begin + tok_slot; t < t_end; t += 16) {
+  for (int t = t_begin + wave * 4 + slot; t < t_end; t += 16) {
     const int local = t - part * DEC_PART;
     const int phys = blk_ids[local / 64];
-    const short* vrow = kv_cache + kv_plane_stride +
+    const short* base = kv_cache +
         ((int64_t)phys * num_kv_heads + kvh) * head_tile_stride +
         (int64_t)(t % 64) * head_dim + lane16 * 8;
-    s16x8 vv = *reinterpret_cast<const s16x8*>(vrow);
-    float vf[8];
+    s16x8 kv = *reinterpret_cast<const s16x8*>(base);
+    s16x8 vv = *reinterpret_cast<const s16x8*>(base + kv_plane_stride);
+    float kf[8], vf[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) kf[j] = to_f32<Tag>(kv[j]);
 #pragma unroll
     for (int j = 0; j < 8; ++j) vf[j] = to_f32<Tag>(vv[j]);
 #pragma unroll
     for (int g = 0; g < GROUP; ++g) {
-      const float p = scores[g][t - t_begin];
+      float sc = 0.f;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) acc[g][j] += p * vf[j];
+      for (int j = 0; j < 8; ++j) sc += qf[g][j] * kf[j];
+      // Reduce over the 16 dim lanes of this slot (consecutive lanes).
+      sc += __shfl_xor(sc, 1, 64);
+      sc += __shfl_xor(sc, 2, 64);
+      sc += __shfl_xor(sc, 4, 64);
+      sc += __shfl_xor(sc, 8, 64);
+      const float m_new = fmaxf(m_s[g], sc);
+      const float p = __expf(sc - m_new);
+      if (m_new > m_s[g]) {
+        const float corr = __expf(m_s[g] - m_new);
+        l_s[g] = l_s[g] * corr + p;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) acc[g][j] = acc[g][j] * corr + p * vf[j];
+        m_s[g] = m_new;
+      } else {
+        l_s[g] += p;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) acc[g][j] += p * vf[j];
+      }
     }
   }
 
-  // Reduce the 4 token-slots within each wave (lanes 16 apart share dims).
+  // ---- merge the 4 slots within each wave (butterfly over lanes 16,32) ---
 #pragma unroll
-  for (int g = 0; g < GROUP; ++g)
+  for (int off = 16; off <= 32; off <<= 1) {
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      acc[g][j] += __shfl_xor(acc[g][j], 16, 64);
-      acc[g][j] += __shfl_xor(acc[g][j], 32, 64);
+    for (int g = 0; g < GROUP; ++g) {
+      const float m_o = __shfl_xor(m_s[g], off, 64);
+      const float l_o = __shfl_xor(l_s[g], off, 64);
+      const float M = fmaxf(m_s[g], m_o);
+      const float c1 = __expf(m_s[g] - M);
+      const float c2 = __expf(m_o - M);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const float a_o = __shfl_xor(acc[g][j], off, 64);
+        acc[g][j] = acc[g][j] * c1 + a_o * c2;
+      }
+      l_s[g] = l_s[g] * c1 + l_o * c2;
+      m_s[g] = M;
     }
-  // Cross-wave reduce via LDS (token-slot lane of each wave contributes).
-  const int wave = threadIdx.x / WAVE_SIZE;
-  __syncthreads();  // scores no longer needed; o_red aliases fresh use
-  if ((threadIdx.x & 63) < 16) {
-#pragma unroll
-    for (int g = 0; g < GROUP; ++g)
-#pragma unroll
-      for (int j = 0; j < 8; ++j)
-        o_red[wave][g][lane16 * 8 + j] = acc[g][j];
   }
-  __syncthreads();
-  if (wave == 0 && lane16 < 16) {
+
+  // ---- merge the 4 waves via LDS ------------------------------------------
+  if ((threadIdx.x & 63) < 16) {
 #pragma unroll
     for (int g = 0; g < GROUP; ++g) {
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const int d = lane16 * 8 + j;
-        float o = o_red[0][g][d] + o_red[1][g][d] + o_red[2][g][d] +
-                  o_red[3][g][d];
-        if (FINAL) {
-          if (tok_slot == 0) {
-            const float inv_l = l_g[g] > 0.f ? 1.f / l_g[g] : 0.f;
-            out[((int64_t)seq * num_q_heads + hq0 + g) * head_dim + d] =
-                from_f32<Tag>(o * inv_l);
-          }
-        } else {
-          if (tok_slot == 0) {
-            tmp_out[((((int64_t)seq * num_q_heads + hq0 + g) * num_parts +
-                      part) * head_dim) + d] = o;
-          }
-        }
+      for (int j = 0; j < 8; ++j) o_red[wave][g][lane16 * 8 + j] = acc[g][j];
+      if (lane16 == 0) {
+        ml_red[wave][g][0] = m_s[g];
+        ml_red[wave][g][1] = l_s[g];
       }
     }
-    if (!FINAL && tok_slot == 0 && lane16 < GROUP) {
-      float* lse = tmp_lse +
-          (((int64_t)seq * num_q_heads + hq0 + lane16) * num_parts + part) *
-              2;
-      lse[0] = m_g[lane16];
-      lse[1] = l_g[lane16];
+  }
+  __syncthreads();
+  // 128 threads: one per dim; every thread recomputes the scalar merge.
+  if (threadIdx.x < 128) {
+    const int d = threadIdx.x;
+#pragma unroll
+    for (int g = 0; g < GROUP; ++g) {
+      float M = fmaxf(fmaxf(ml_red[0][g][0], ml_red[1][g][0]),
+                      fmaxf(ml_red[2][g][0], ml_red[3][g][0]));
+      float L = 0.f, o = 0.f;
+#pragma unroll
+      for (int w = 0; w < 4; ++w) {
+        const float c = __expf(ml_red[w][g][0] - M);
+        L += ml_red[w][g][1] * c;
+        o += o_red[w][g][d] * c;
+      }
+      if (FINAL) {
+        const float inv_l = L > 0.f ? 1.f / L : 0.f;
+        out[((int64_t)seq * num_q_heads + hq0 + g) * head_dim + d] =
+            from_f32<Tag>(o * inv_l);
+      } else {
+        tmp_out[((((int64_t)seq * num_q_heads + hq0 + g) * num_parts + part) *
+                 head_dim) + d] = o;
+        if (d == 0) {
+          float* lse = tmp_lse +
+              (((int64_t)seq * num_q_heads + hq0 + g) * num_parts + part) * 2;
+          lse[0] = M;
+          lse[1] = L;
+        }
+      }
     }
   }
 }

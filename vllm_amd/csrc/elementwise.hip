@@ -94,6 +94,8 @@ __global__ void fused_add_rms_norm_kernel(short* __restrict__ x,
 
 // ---------------------------------------------------------------------------
 // SiLU-and-mul: out[t, i] = silu(x[t, i]) * x[t, d+i].
+// 2D grid (row, column chunk): one short8 vector per thread per launch so
+// small decode batches still put >>256 workgroups in flight (8 XCDs).
 template <typename Tag>
 __global__ void silu_and_mul_kernel(short* __restrict__ out,
                                     const short* __restrict__ in,
@@ -103,7 +105,9 @@ __global__ void silu_and_mul_kernel(short* __restrict__ out,
   const short* up = gate + d;
   short* o = out + row * d;
   const int nvec = d / 8;
-  for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
+  {
+    const int i = blockIdx.y * blockDim.x + threadIdx.x;
+    if (i >= nvec) return;
     s16x8 g = reinterpret_cast<const s16x8*>(gate)[i];
     s16x8 u = reinterpret_cast<const s16x8*>(up)[i];
     s16x8 r;
@@ -117,7 +121,7 @@ __global__ void silu_and_mul_kernel(short* __restrict__ out,
   }
 }
 
-// GELU(tanh)-and-mul.
+// GELU(tanh)-and-mul. Same 2D grid as silu_and_mul_kernel.
 template <typename Tag>
 __global__ void gelu_and_mul_kernel(short* __restrict__ out,
                                     const short* __restrict__ in,
@@ -127,7 +131,9 @@ __global__ void gelu_and_mul_kernel(short* __restrict__ out,
   const short* up = gate + d;
   short* o = out + row * d;
   const int nvec = d / 8;
-  for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
+  {
+    const int i = blockIdx.y * blockDim.x + threadIdx.x;
+    if (i >= nvec) return;
     s16x8 g = reinterpret_cast<const s16x8*>(gate)[i];
     s16x8 u = reinterpret_cast<const s16x8*>(up)[i];
     s16x8 r;
@@ -290,8 +296,9 @@ torch::Tensor silu_and_mul(torch::Tensor x) {
   auto out = torch::empty(sizes, x.options());
   const int64_t rows = x.numel() / (2 * d);
   auto stream = c10::hip::getCurrentHIPStream().stream();
+  const int chunks = (d / 8 + 255) / 256;
 #define LAUNCH_SILU(TAG)                                              \
-  hipLaunchKernelGGL((silu_and_mul_kernel<TAG>), dim3(rows),          \
+  hipLaunchKernelGGL((silu_and_mul_kernel<TAG>), dim3(rows, chunks),  \
                      dim3(256), 0, stream, (short*)out.data_ptr(),    \
                      (const short*)x.data_ptr(), d)
   DISPATCH_16B(x, LAUNCH_SILU);
@@ -309,8 +316,9 @@ torch::Tensor gelu_and_mul(torch::Tensor x) {
   auto out = torch::empty(sizes, x.options());
   const int64_t rows = x.numel() / (2 * d);
   auto stream = c10::hip::getCurrentHIPStream().stream();
+  const int chunks = (d / 8 + 255) / 256;
 #define LAUNCH_GELU(TAG)                                              \
-  hipLaunchKernelGGL((gelu_and_mul_kernel<TAG>), dim3(rows),          \
+  hipLaunchKernelGGL((gelu_and_mul_kernel<TAG>), dim3(rows, chunks),  \
                      dim3(256), 0, stream, (short*)out.data_ptr(),    \
                      (const short*)x.data_ptr(), d)
   DISPATCH_16B(x, LAUNCH_GELU);

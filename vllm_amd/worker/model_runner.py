@@ -41,6 +41,127 @@ class CachedReqState:
         return self.token_ids[self.prompt_len:]
 
 
+def _cdiv(a: int, b: int) -> int:
+    return (a + b - 1) // b
+
+
+class DecodeGraphRunner:
+    """hipGraph-captured pure-decode steps (role of the reference's
+    CUDAGraph dispatch, gpu_model_runner.py:4025 — redesigned: persistent
+    input buffers + lazy per-(batch, kv-partition) capture; replay removes
+    the ~8 launches/layer × 32 layers of per-step launch latency).
+
+    Padding rows are inert by construction: seq_len=0 (attention reads
+    nothing), slot_mapping=-1 (cache write skipped), block_table=0.
+    """
+
+    BATCH_BUCKETS = (1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128, 192, 256,
+                     384, 512, 768, 1024)
+    DEC_PART = 512  # must match attention_decode.hip
+
+    def __init__(self, runner: "ModelRunner"):
+        self.runner = runner
+        cfg = runner.config
+        self.max_seqs = min(
+            cfg.scheduler_config.max_num_seqs, self.BATCH_BUCKETS[-1]
+        )
+        self.buckets = [b for b in self.BATCH_BUCKETS if b <= self.max_seqs]
+        if self.buckets[-1] != self.max_seqs:
+            self.buckets.append(self.max_seqs)
+        self.max_parts = max(1, _cdiv(runner.max_model_len, self.DEC_PART))
+        self.max_blocks = _cdiv(runner.max_model_len, runner.block_size)
+        dev = runner.device
+        n = self.max_seqs
+        self.input_ids = torch.zeros(n, dtype=torch.int64, device=dev)
+        self.positions = torch.zeros(n, dtype=torch.int64, device=dev)
+        self.slot_mapping = torch.full((n,), -1, dtype=torch.int64,
+                                       device=dev)
+        self.seq_lens = torch.zeros(n, dtype=torch.int32, device=dev)
+        self.query_start_loc = torch.arange(n + 1, dtype=torch.int32,
+                                            device=dev)
+        self.block_table = torch.zeros(n, self.max_blocks, dtype=torch.int32,
+                                       device=dev)
+        # Pinned staging buffers for async H2D.
+        self.pin_i64 = torch.empty(3, n, dtype=torch.int64, pin_memory=True)
+        self.pin_seq = torch.empty(n, dtype=torch.int32, pin_memory=True)
+        self.pin_bt = torch.empty(n, self.max_blocks, dtype=torch.int32,
+                                  pin_memory=True)
+        self.graphs: dict = {}
+        self.pool = None
+
+    def bucket_for(self, n: int) -> Optional[int]:
+        for b in self.buckets:
+            if b >= n:
+                return b
+        return None
+
+    def parts_bucket(self, max_seq_len: int) -> int:
+        p = _cdiv(max(max_seq_len, 1), self.DEC_PART)
+        b = 1
+        while b < p:
+            b *= 2
+        return min(b, self.max_parts) if self.max_parts > 1 else 1
+
+    def _meta(self, nb: int, parts: int) -> AttentionMetadata:
+        return AttentionMetadata(
+            query_start_loc=self.query_start_loc[: nb + 1],
+            seq_lens=self.seq_lens[:nb],
+            block_table=self.block_table[:nb],
+            slot_mapping=self.slot_mapping[:nb],
+            num_reqs=nb,
+            num_actual_tokens=nb,
+            max_query_len=1,
+            max_seq_len=parts * self.DEC_PART,
+            num_decodes=nb,
+        )
+
+    def _capture(self, nb: int, parts: int):
+        runner = self.runner
+        ctx = ForwardContext(attn_metadata=self._meta(nb, parts),
+                             kv_caches=runner.kv_caches)
+        # Warmup on a side stream (allocator state, RCCL lazy init).
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s), set_forward_context(ctx):
+            hidden = runner.model(self.input_ids[:nb], self.positions[:nb])
+            runner.model.compute_logits(hidden)
+        torch.cuda.current_stream().wait_stream(s)
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph, pool=self.pool), set_forward_context(ctx):
+            hidden = runner.model(self.input_ids[:nb], self.positions[:nb])
+            logits = runner.model.compute_logits(hidden)
+        if self.pool is None:
+            self.pool = graph.pool()
+        self.graphs[(nb, parts)] = (graph, logits)
+        return self.graphs[(nb, parts)]
+
+    def run(self, n: int, nb: int, parts: int, input_ids, positions,
+            slot_mapping, seq_lens, block_table) -> torch.Tensor:
+        """Stage numpy inputs (n real rows, nb bucket) and replay. Returns
+        logits for the n real rows."""
+        pi = self.pin_i64
+        pi[0, :n] = torch.from_numpy(input_ids)
+        pi[1, :n] = torch.from_numpy(positions)
+        pi[2, :n] = torch.from_numpy(slot_mapping)
+        pi[2, n:nb] = -1
+        self.pin_seq[:n] = torch.from_numpy(seq_lens)
+        self.pin_seq[n:nb] = 0
+        w = block_table.shape[1]
+        self.pin_bt[:n, :w] = torch.from_numpy(block_table)
+        self.input_ids[:nb].copy_(pi[0, :nb], non_blocking=True)
+        self.positions[:nb].copy_(pi[1, :nb], non_blocking=True)
+        self.slot_mapping[:nb].copy_(pi[2, :nb], non_blocking=True)
+        self.seq_lens[:nb].copy_(self.pin_seq[:nb], non_blocking=True)
+        self.block_table[:n, :w].copy_(self.pin_bt[:n, :w],
+                                       non_blocking=True)
+        entry = self.graphs.get((nb, parts))
+        if entry is None:
+            entry = self._capture(nb, parts)
+        graph, logits = entry
+        graph.replay()
+        return logits[:n]
+
+
 class ModelRunner:
 
     def __init__(self, config: EngineConfig, device: torch.device):
@@ -52,6 +173,7 @@ class ModelRunner:
         self.requests: dict[str, CachedReqState] = {}
         self.kv_caches: list[torch.Tensor] = []
         self.model: Optional[torch.nn.Module] = None
+        self.graph_runner: Optional[DecodeGraphRunner] = None
         self.sampler = Sampler()
         self.max_model_len = config.model_config.max_model_len
 
@@ -86,6 +208,9 @@ class ModelRunner:
             )
             for _ in range(spec.num_layers)
         ]
+        if (self.device.type == "cuda"
+                and not self.config.model_config.enforce_eager):
+            self.graph_runner = DecodeGraphRunner(self)
 
     # ------------------------------------------------------------------
     def _update_states(self, so: SchedulerOutput) -> None:
@@ -159,23 +284,38 @@ class ModelRunner:
             t += n
 
         dev = self.device
-        meta = AttentionMetadata(
-            query_start_loc=torch.from_numpy(query_start_loc).to(dev),
-            seq_lens=torch.from_numpy(seq_lens).to(dev),
-            block_table=torch.from_numpy(block_table).to(dev),
-            slot_mapping=torch.from_numpy(slot_mapping).to(dev),
-            num_reqs=len(items),
-            num_actual_tokens=total,
-            max_query_len=int(max(n for _, n in items)),
-            max_seq_len=int(seq_lens.max()),
-            num_decodes=num_decodes,
-        )
-        input_ids_t = torch.from_numpy(input_ids).to(dev)
-        positions_t = torch.from_numpy(positions).to(dev)
+        max_seq_len = int(seq_lens.max())
 
-        ctx = ForwardContext(attn_metadata=meta, kv_caches=self.kv_caches)
-        with set_forward_context(ctx):
-            hidden = self.model(input_ids_t, positions_t)
+        # hipGraph fast path: pure-decode batch within a captured bucket.
+        nb = None
+        if (self.graph_runner is not None and num_decodes == len(items)
+                and total == len(items)):
+            nb = self.graph_runner.bucket_for(total)
+        if nb is not None:
+            parts = self.graph_runner.parts_bucket(max_seq_len)
+            logits_all = self.graph_runner.run(
+                total, nb, parts, input_ids, positions, slot_mapping,
+                seq_lens, block_table,
+            )
+        else:
+            meta = AttentionMetadata(
+                query_start_loc=torch.from_numpy(query_start_loc).to(dev),
+                seq_lens=torch.from_numpy(seq_lens).to(dev),
+                block_table=torch.from_numpy(block_table).to(dev),
+                slot_mapping=torch.from_numpy(slot_mapping).to(dev),
+                num_reqs=len(items),
+                num_actual_tokens=total,
+                max_query_len=int(max(n for _, n in items)),
+                max_seq_len=max_seq_len,
+                num_decodes=num_decodes,
+            )
+            input_ids_t = torch.from_numpy(input_ids).to(dev)
+            positions_t = torch.from_numpy(positions).to(dev)
+
+            ctx = ForwardContext(attn_metadata=meta, kv_caches=self.kv_caches)
+            with set_forward_context(ctx):
+                hidden = self.model(input_ids_t, positions_t)
+            logits_all = None
 
         # Advance computed counts.
         for rid, n in items:
@@ -187,12 +327,16 @@ class ModelRunner:
                 sampled_token_ids=[[] for _ in req_ids],
             )
 
-        # Gather last-token hidden states of sampling requests.
-        last_idx = torch.tensor(
-            [int(query_start_loc[r + 1]) - 1 for r in sampling_rows],
-            device=dev,
-        )
-        logits = self.model.compute_logits(hidden[last_idx])
+        if logits_all is not None:
+            # Decode: every row samples, rows already in order.
+            logits = logits_all
+        else:
+            # Gather last-token hidden states of sampling requests.
+            last_idx = torch.tensor(
+                [int(query_start_loc[r + 1]) - 1 for r in sampling_rows],
+                device=dev,
+            )
+            logits = self.model.compute_logits(hidden[last_idx])
 
         s_params = [
             self.requests[req_ids[r]].sampling_params for r in sampling_rows
