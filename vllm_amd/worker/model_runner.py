@@ -177,6 +177,21 @@ class ModelRunner:
         self.sampler = Sampler()
         self.max_model_len = config.model_config.max_model_len
 
+        # Persistent per-row batch state (vectorized decode input prep —
+        # role of the reference's persistent batch, gpu_input_batch.py).
+        # Rows are assigned on first schedule and recycled on finish.
+        n = config.scheduler_config.max_num_seqs
+        mb = _cdiv(self.max_model_len, self.block_size)
+        self._row_of: dict[str, int] = {}
+        self._free_rows = list(range(n - 1, -1, -1))
+        self.np_last_tok = np.zeros(n, dtype=np.int64)
+        self.np_computed = np.zeros(n, dtype=np.int64)
+        self.np_block_table = np.zeros((n, mb), dtype=np.int32)
+        self.np_nblocks = np.zeros(n, dtype=np.int32)
+        # Sampling-metadata cache for steady-state decode batches.
+        self._samp_cache_key = None
+        self._samp_cache_val = None
+
     def load_model(self) -> None:
         self.model = load_model(self.config.model_config, self.device)
 
@@ -215,7 +230,9 @@ class ModelRunner:
     # ------------------------------------------------------------------
     def _update_states(self, so: SchedulerOutput) -> None:
         for req_id in so.finished_req_ids:
-            self.requests.pop(req_id, None)
+            if self.requests.pop(req_id, None) is not None:
+                row = self._row_of.pop(req_id)
+                self._free_rows.append(row)
         for nr in so.scheduled_new_reqs:
             self.requests[nr.req_id] = CachedReqState(
                 req_id=nr.req_id,
@@ -225,16 +242,135 @@ class ModelRunner:
                 block_ids=list(nr.block_ids),
                 sampling_params=nr.sampling_params,
             )
+            row = self._free_rows.pop()
+            self._row_of[nr.req_id] = row
+            nb = len(nr.block_ids)
+            self.np_block_table[row, :nb] = nr.block_ids
+            self.np_nblocks[row] = nb
+            self.np_computed[row] = nr.num_computed_tokens
         cr = so.scheduled_cached_reqs
         for i, req_id in enumerate(cr.req_ids):
             state = self.requests[req_id]
+            row = self._row_of[req_id]
             if cr.resumed[i]:
                 state.block_ids = list(cr.new_block_ids[i])
                 state.token_ids = list(cr.new_token_ids[i])
                 state.num_computed_tokens = cr.num_computed_tokens[i]
+                nb = len(state.block_ids)
+                self.np_block_table[row, :nb] = state.block_ids
+                self.np_nblocks[row] = nb
             else:
                 state.block_ids.extend(cr.new_block_ids[i])
                 state.num_computed_tokens = cr.num_computed_tokens[i]
+                newb = cr.new_block_ids[i]
+                if newb:
+                    nb0 = self.np_nblocks[row]
+                    self.np_block_table[row, nb0:nb0 + len(newb)] = newb
+                    self.np_nblocks[row] = nb0 + len(newb)
+            self.np_computed[row] = cr.num_computed_tokens[i]
+
+    # ------------------------------------------------------------------
+    @torch.inference_mode()
+    def _execute_decode(self, so: SchedulerOutput,
+                        req_ids: list[str]) -> ModelRunnerOutput:
+        """Pure-decode step: vectorized input prep from the persistent row
+        arrays; hipGraph replay when available."""
+        n = len(req_ids)
+        rows = np.fromiter((self._row_of[r] for r in req_ids), dtype=np.int64,
+                           count=n)
+        positions = self.np_computed[rows]
+        input_ids = self.np_last_tok[rows]
+        blk = positions // self.block_size
+        slot_mapping = (
+            self.np_block_table[rows, blk].astype(np.int64) * self.block_size
+            + positions % self.block_size
+        )
+        seq_lens = (positions + 1).astype(np.int32)
+        max_seq_len = int(seq_lens.max())
+        w = _cdiv(max_seq_len, self.block_size)
+        block_table = self.np_block_table[rows][:, :w]
+
+        dev = self.device
+        nb = (self.graph_runner.bucket_for(n)
+              if self.graph_runner is not None else None)
+        if nb is not None:
+            parts = self.graph_runner.parts_bucket(max_seq_len)
+            logits = self.graph_runner.run(
+                n, nb, parts, input_ids, positions, slot_mapping, seq_lens,
+                block_table,
+            )
+        else:
+            meta = AttentionMetadata(
+                query_start_loc=torch.arange(n + 1, dtype=torch.int32,
+                                             device=dev),
+                seq_lens=torch.from_numpy(seq_lens).to(dev),
+                block_table=torch.from_numpy(
+                    np.ascontiguousarray(block_table)).to(dev),
+                slot_mapping=torch.from_numpy(slot_mapping).to(dev),
+                num_reqs=n,
+                num_actual_tokens=n,
+                max_query_len=1,
+                max_seq_len=max_seq_len,
+                num_decodes=n,
+            )
+            ctx = ForwardContext(attn_metadata=meta, kv_caches=self.kv_caches)
+            with set_forward_context(ctx):
+                hidden = self.model(
+                    torch.from_numpy(input_ids).to(dev),
+                    torch.from_numpy(positions).to(dev),
+                )
+            logits = self.model.compute_logits(hidden)
+
+        self.np_computed[rows] += 1
+        states = [self.requests[r] for r in req_ids]
+        for st in states:
+            st.num_computed_tokens += 1
+
+        s_meta = self._sampling_meta(req_ids, states, dev)
+        s_out = self.sampler(logits, s_meta)
+        sampled_np = s_out.sampled_token_ids.cpu().numpy()
+        self.np_last_tok[rows] = sampled_np
+        sampled = sampled_np.tolist()
+        sampled_per_req = [[int(t)] for t in sampled]
+        logprobs_per_req: dict[str, list[dict[int, float]]] = {}
+        for j, st in enumerate(states):
+            st.token_ids.append(int(sampled[j]))
+            if s_out.logprobs is not None and s_out.logprobs[j] is not None:
+                logprobs_per_req[req_ids[j]] = [s_out.logprobs[j]]
+        return ModelRunnerOutput(
+            req_ids=req_ids,
+            sampled_token_ids=sampled_per_req,
+            logprobs=logprobs_per_req or None,
+        )
+
+    def _sampling_meta(self, req_ids, states, dev) -> SamplingMetadata:
+        """SamplingMetadata with a steady-state cache: reused while the
+        (request set, params) is unchanged and nothing stateful (penalties,
+        seeded generators, logprobs, masks) is requested."""
+        params = [st.sampling_params for st in states]
+        key = tuple((r, id(p)) for r, p in zip(req_ids, params))
+        if key == self._samp_cache_key:
+            return self._samp_cache_val
+        meta = SamplingMetadata.build(
+            params,
+            [st.token_ids[: st.prompt_len] for st in states],
+            [st.output_token_ids for st in states],
+            dev,
+            seeds_offset=[len(st.output_token_ids) for st in states],
+        )
+        cacheable = (
+            meta.no_penalties and meta.max_num_logprobs == 0
+            and not meta.generators and meta.logit_bias is None
+            and meta.allowed_token_ids is None
+            and meta.min_tokens_mask is None
+        )
+        if cacheable:
+            self._samp_cache_key = key
+            self._samp_cache_val = meta
+        else:
+            self._samp_cache_key = None
+            self._samp_cache_val = None
+        return meta
 
     # ------------------------------------------------------------------
     @torch.inference_mode()
@@ -251,6 +387,8 @@ class ModelRunner:
         num_decodes = sum(1 for _, n in items if n == 1)
 
         total = so.total_num_scheduled_tokens
+        if num_decodes == len(items) and total == len(items):
+            return self._execute_decode(so, req_ids)
         input_ids = np.empty(total, dtype=np.int64)
         positions = np.empty(total, dtype=np.int64)
         slot_mapping = np.empty(total, dtype=np.int64)
@@ -317,9 +455,10 @@ class ModelRunner:
                 hidden = self.model(input_ids_t, positions_t)
             logits_all = None
 
-        # Advance computed counts.
+        # Advance computed counts (python state + persistent rows).
         for rid, n in items:
             self.requests[rid].num_computed_tokens += n
+            self.np_computed[self._row_of[rid]] += n
 
         if not sampling_rows:
             return ModelRunnerOutput(
@@ -367,6 +506,7 @@ class ModelRunner:
             sampled_per_req[r] = [tok]
             # Runner keeps its own copy of generated tokens.
             self.requests[req_ids[r]].token_ids.append(tok)
+            self.np_last_tok[self._row_of[req_ids[r]]] = tok
             if s_out.logprobs is not None and s_out.logprobs[j] is not None:
                 logprobs_per_req[req_ids[r]] = [s_out.logprobs[j]]
         return ModelRunnerOutput(
