@@ -214,3 +214,21 @@ def test_model_forward_matches_cpu_reference():
     cos = torch.nn.functional.cosine_similarity(gpu, cpu, dim=-1)
     assert cos.min() > 0.99, cos
     assert (gpu.argmax(-1) == cpu.argmax(-1)).float().mean() > 0.8
+
+
+@pytest.mark.parametrize("m,k,n,bias", [
+    (7, 512, 1024, False),
+    (256, 4096, 6144, False),
+    (256, 14336, 4096, False),
+    (1024, 4096, 4096, True),
+])
+def test_lt_linear(m, k, n, bias):
+    hip = _hip()
+    a = torch.randn(m, k, dtype=torch.bfloat16, device="cuda") * 0.1
+    w = torch.randn(n, k, dtype=torch.bfloat16, device="cuda") * 0.1
+    b = (torch.randn(n, dtype=torch.bfloat16, device="cuda")
+         if bias else None)
+    out = hip.linear(a, w, b)
+    expect = torch.nn.functional.linear(
+        a.float(), w.float(), b.float() if bias else None)
+    assert_close(out, expect, atol=5e-2, rtol=5e-2)

@@ -24,6 +24,7 @@ SOURCES = [
     CSRC / "elementwise.hip",
     CSRC / "attention_decode.hip",
     CSRC / "attention_prefill.hip",
+    CSRC / "gemm_hipblaslt.cpp",
     CSRC / "bindings.cpp",
 ]
 
@@ -56,6 +57,9 @@ def _torch_flags() -> tuple[list[str], list[str]]:
         "-lc10",
         "-lc10_hip",
         "-lamdhip64",
+        "-lhipblaslt",
+        "-L/opt/rocm/lib",
+        "-Wl,-rpath,/opt/rocm/lib",
         f"-Wl,-rpath,{torch_dir}/lib",
     ]
     return inc + cxx, link

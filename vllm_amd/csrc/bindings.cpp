@@ -30,6 +30,8 @@ void prefill_attention(torch::Tensor out, torch::Tensor q,
                        torch::Tensor query_start_loc, torch::Tensor seq_lens,
                        double scale, int64_t num_decodes,
                        int64_t max_query_len, int64_t sliding_window);
+torch::Tensor lt_linear(torch::Tensor a, torch::Tensor w,
+                        c10::optional<torch::Tensor> bias);
 
 }  // namespace vllm_amd
 
@@ -50,6 +52,7 @@ TORCH_LIBRARY(vllm_amd, m) {
         "Tensor block_table, Tensor query_start_loc, Tensor seq_lens, "
         "float scale, int num_decodes, int max_query_len, "
         "int sliding_window) -> ()");
+  m.def("lt_linear(Tensor a, Tensor w, Tensor? bias) -> Tensor");
 }
 
 TORCH_LIBRARY_IMPL(vllm_amd, CUDA, m) {
@@ -61,4 +64,5 @@ TORCH_LIBRARY_IMPL(vllm_amd, CUDA, m) {
   m.impl("reshape_and_cache", &vllm_amd::reshape_and_cache);
   m.impl("paged_decode_attention", &vllm_amd::paged_decode_attention);
   m.impl("prefill_attention", &vllm_amd::prefill_attention);
+  m.impl("lt_linear", &vllm_amd::lt_linear);
 }

@@ -1,0 +1,198 @@
+// Tuned hipBLASLt GEMM for the linear layers (bf16/fp16, TN, row-major).
+//
+// torch's F.linear on ROCm picks heuristics that leave ~1.5-2x on the
+// table for both skinny decode shapes (weight-streaming bound) and big
+// prefill shapes (hipBLASLt standalone reaches ~2.0 PF bf16 @8k^3 vs
+// ~1.3 PF through torch; see profiles/). This op asks hipBLASLt for a
+// list of candidate algorithms per (M, N, K, dtype) and picks the
+// fastest by measuring each once on first use; the choice is cached for
+// the process lifetime. Role of the reference's tuned-GEMM dispatch
+// (vllm/model_executor/layers/utils.py apply_w8a8_* and rocm skinny
+// gemms csrc/rocm/skinny_gemms.cu) — re-designed around hipBLASLt's own
+// search instead of hand-picked kernels.
+
+#include <torch/all.h>
+#include <c10/hip/HIPStream.h>
+
+#include <hipblaslt/hipblaslt.h>
+#include <hipblaslt/hipblaslt-ext.hpp>
+
+#include <mutex>
+#include <unordered_map>
+#include <vector>
+
+#include "common.h"
+
+namespace vllm_amd {
+
+namespace {
+
+#define HIP_CHECK(expr)                                              \
+  do {                                                               \
+    hipError_t _e = (expr);                                          \
+    TORCH_CHECK(_e == hipSuccess, "HIP error: ",                     \
+                hipGetErrorString(_e), " at ", __FILE__, ":",        \
+                __LINE__);                                           \
+  } while (0)
+
+#define HIPBLASLT_CHECK(expr)                                        \
+  do {                                                               \
+    hipblasStatus_t _st = (expr);                                    \
+    TORCH_CHECK(_st == HIPBLAS_STATUS_SUCCESS, "hipBLASLt error ",   \
+                (int)_st, " at ", __FILE__, ":", __LINE__);          \
+  } while (0)
+
+constexpr size_t kWorkspaceBytes = 128ull * 1024 * 1024;
+
+struct LtContext {
+  hipblasLtHandle_t handle{nullptr};
+  void* workspace{nullptr};
+  std::unordered_map<uint64_t, hipblasLtMatmulAlgo_t> algo_cache;
+  std::mutex mu;
+};
+
+LtContext& ctx() {
+  static LtContext c;
+  static std::once_flag once;
+  std::call_once(once, [] {
+    HIPBLASLT_CHECK(hipblasLtCreate(&c.handle));
+    HIP_CHECK(hipMalloc(&c.workspace, kWorkspaceBytes));
+  });
+  return c;
+}
+
+uint64_t shape_key(int64_t m, int64_t n, int64_t k, bool bf16, bool bias) {
+  // m<=2^24 buckets suffice; key layout: [m:24][n:20][k:18][dtype:1][bias:1]
+  return ((uint64_t)m << 40) ^ ((uint64_t)n << 20) ^ ((uint64_t)k << 2) ^
+         ((uint64_t)bf16 << 1) ^ (uint64_t)bias;
+}
+
+// C[M,N] row-major = A[M,K] row-major * W[N,K]^T row-major (+bias[N]).
+// In hipBLASLt's column-major view: C'[N,M] = op_T(W'[K,N]) * op_N(A'[K,M]).
+void run_matmul(torch::Tensor& out, const torch::Tensor& a,
+                const torch::Tensor& w, const c10::optional<torch::Tensor>& bias) {
+  const int64_t M = a.size(0);
+  const int64_t K = a.size(1);
+  const int64_t N = w.size(0);
+  TORCH_CHECK(w.size(1) == K && out.size(0) == M && out.size(1) == N);
+  TORCH_CHECK(a.is_contiguous() && w.is_contiguous() && out.is_contiguous());
+  const bool is_bf16 = a.scalar_type() == torch::kBFloat16;
+  hipDataType dt = is_bf16 ? HIP_R_16BF : HIP_R_16F;
+
+  auto& c = ctx();
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+
+  hipblasLtMatmulDesc_t op_desc;
+  HIPBLASLT_CHECK(hipblasLtMatmulDescCreate(&op_desc, HIPBLAS_COMPUTE_32F,
+                                            HIP_R_32F));
+  hipblasOperation_t opA = HIPBLAS_OP_T;
+  hipblasOperation_t opB = HIPBLAS_OP_N;
+  HIPBLASLT_CHECK(hipblasLtMatmulDescSetAttribute(
+      op_desc, HIPBLASLT_MATMUL_DESC_TRANSA, &opA, sizeof(opA)));
+  HIPBLASLT_CHECK(hipblasLtMatmulDescSetAttribute(
+      op_desc, HIPBLASLT_MATMUL_DESC_TRANSB, &opB, sizeof(opB)));
+  const bool has_bias = bias.has_value();
+  if (has_bias) {
+    hipblasLtEpilogue_t epi = HIPBLASLT_EPILOGUE_BIAS;
+    HIPBLASLT_CHECK(hipblasLtMatmulDescSetAttribute(
+        op_desc, HIPBLASLT_MATMUL_DESC_EPILOGUE, &epi, sizeof(epi)));
+    const void* bptr = bias->data_ptr();
+    HIPBLASLT_CHECK(hipblasLtMatmulDescSetAttribute(
+        op_desc, HIPBLASLT_MATMUL_DESC_BIAS_POINTER, &bptr, sizeof(bptr)));
+  }
+
+  // Column-major descriptors: A-slot = W (K x N viewed, opT), ld = K.
+  hipblasLtMatrixLayout_t lw, la, lc;
+  HIPBLASLT_CHECK(hipblasLtMatrixLayoutCreate(&lw, dt, K, N, K));
+  HIPBLASLT_CHECK(hipblasLtMatrixLayoutCreate(&la, dt, K, M, K));
+  HIPBLASLT_CHECK(hipblasLtMatrixLayoutCreate(&lc, dt, N, M, N));
+
+  const float alpha = 1.f, beta = 0.f;
+  const uint64_t key = shape_key(M, N, K, is_bf16, has_bias);
+
+  hipblasLtMatmulAlgo_t algo;
+  bool have_algo = false;
+  {
+    std::lock_guard<std::mutex> g(c.mu);
+    auto it = c.algo_cache.find(key);
+    if (it != c.algo_cache.end()) {
+      algo = it->second;
+      have_algo = true;
+    }
+  }
+
+  if (!have_algo) {
+    // First use of this shape: ask for candidates and race them once.
+    hipblasLtMatmulPreference_t pref;
+    HIPBLASLT_CHECK(hipblasLtMatmulPreferenceCreate(&pref));
+    HIPBLASLT_CHECK(hipblasLtMatmulPreferenceSetAttribute(
+        pref, HIPBLASLT_MATMUL_PREF_MAX_WORKSPACE_BYTES, &kWorkspaceBytes,
+        sizeof(kWorkspaceBytes)));
+    constexpr int kMaxAlgos = 48;
+    std::vector<hipblasLtMatmulHeuristicResult_t> results(kMaxAlgos);
+    int found = 0;
+    HIPBLASLT_CHECK(hipblasLtMatmulAlgoGetHeuristic(
+        c.handle, op_desc, lw, la, lc, lc, pref, kMaxAlgos, results.data(),
+        &found));
+    HIPBLASLT_CHECK(hipblasLtMatmulPreferenceDestroy(pref));
+    TORCH_CHECK(found > 0, "hipBLASLt: no algorithms for shape ", M, "x", N,
+                "x", K);
+
+    float best = 1e30f;
+    int best_i = 0;
+    hipEvent_t ev0, ev1;
+    HIP_CHECK(hipEventCreate(&ev0));
+    HIP_CHECK(hipEventCreate(&ev1));
+    const int reps = 3;
+    for (int i = 0; i < found; ++i) {
+      // Warm once, then time `reps` runs.
+      auto call = [&](void) {
+        return hipblasLtMatmul(
+            c.handle, op_desc, &alpha, w.data_ptr(), lw, a.data_ptr(), la,
+            &beta, out.data_ptr(), lc, out.data_ptr(), lc,
+            &results[i].algo, c.workspace, kWorkspaceBytes, stream);
+      };
+      if (call() != HIPBLAS_STATUS_SUCCESS) continue;
+      HIP_CHECK(hipEventRecord(ev0, stream));
+      for (int r = 0; r < reps; ++r) (void)call();
+      HIP_CHECK(hipEventRecord(ev1, stream));
+      HIP_CHECK(hipEventSynchronize(ev1));
+      float ms = 0.f;
+      HIP_CHECK(hipEventElapsedTime(&ms, ev0, ev1));
+      if (ms < best) {
+        best = ms;
+        best_i = i;
+      }
+    }
+    HIP_CHECK(hipEventDestroy(ev0));
+    HIP_CHECK(hipEventDestroy(ev1));
+    algo = results[best_i].algo;
+    std::lock_guard<std::mutex> g(c.mu);
+    c.algo_cache.emplace(key, algo);
+  }
+
+  HIPBLASLT_CHECK(hipblasLtMatmul(
+      c.handle, op_desc, &alpha, w.data_ptr(), lw, a.data_ptr(), la, &beta,
+      out.data_ptr(), lc, out.data_ptr(), lc, &algo, c.workspace,
+      kWorkspaceBytes, stream));
+
+  HIPBLASLT_CHECK(hipblasLtMatrixLayoutDestroy(lw));
+  HIPBLASLT_CHECK(hipblasLtMatrixLayoutDestroy(la));
+  HIPBLASLT_CHECK(hipblasLtMatrixLayoutDestroy(lc));
+  HIPBLASLT_CHECK(hipblasLtMatmulDescDestroy(op_desc));
+}
+
+}  // namespace
+
+torch::Tensor lt_linear(torch::Tensor a, torch::Tensor w,
+                        c10::optional<torch::Tensor> bias) {
+  auto sizes = a.sizes().vec();
+  const int64_t K = sizes.back();
+  auto a2 = a.reshape({-1, K}).contiguous();
+  auto out = torch::empty({a2.size(0), w.size(0)}, a.options());
+  run_matmul(out, a2, w, bias);
+  sizes.back() = w.size(0);
+  return out.reshape(sizes);
+}
+
+}  // namespace vllm_amd

@@ -67,7 +67,8 @@ class ParallelLMHead(VocabParallelEmbedding):
     """LM head sharing the vocab-parallel sharding; logits are gathered."""
 
     def compute_logits(self, hidden: torch.Tensor) -> torch.Tensor:
-        logits = F.linear(hidden, self.weight)
+        from vllm_amd import ops
+        logits = ops.linear(hidden, self.weight, None)
         if get_tp_world_size() > 1:
             logits = tensor_model_parallel_all_gather(logits, dim=-1)
             logits = logits[..., : self.num_embeddings]

@@ -5,8 +5,8 @@ Merged/QKV/RowParallel): weights are sharded at construction time; a
 RowParallelLinear forward ends in one RCCL all-reduce over xGMI — the
 dominant collective of TP decode (2 per decoder layer).
 
-GEMMs go through torch.nn.functional.linear (hipBLASLt/rocBLAS on ROCm);
-fused/quantized GEMM variants get hand-written MFMA kernels later.
+On GPU, GEMMs go through the tuned hipBLASLt op (ops.linear — per-shape
+algo search, csrc/gemm_hipblaslt.cpp); on CPU, torch F.linear.
 """
 
 from __future__ import annotations
@@ -15,7 +15,8 @@ from typing import Optional
 
 import torch
 import torch.nn as nn
-import torch.nn.functional as F
+
+from vllm_amd import ops
 
 from vllm_amd.parallel.state import (
     get_tp_rank,
@@ -40,7 +41,7 @@ class ReplicatedLinear(nn.Module):
         )
 
     def forward(self, x):
-        return F.linear(x, self.weight, self.bias)
+        return ops.linear(x, self.weight, self.bias)
 
 
 class ColumnParallelLinear(nn.Module):
@@ -77,7 +78,7 @@ class ColumnParallelLinear(nn.Module):
         )
 
     def forward(self, x):
-        y = F.linear(x, self.weight, self.bias)
+        y = ops.linear(x, self.weight, self.bias)
         if self.gather_output:
             y = tensor_model_parallel_all_gather(y, dim=-1)
         return y
@@ -195,7 +196,7 @@ class RowParallelLinear(nn.Module):
         )
 
     def forward(self, x):
-        y = F.linear(x, self.weight)
+        y = ops.linear(x, self.weight, None)
         if self.reduce_results:
             y = tensor_model_parallel_all_reduce(y)
         if self.bias is not None:

@@ -105,6 +105,11 @@ def attention_unified(
     )
 
 
+def linear(x, weight, bias=None):
+    """x @ weight.T + bias — tuned hipBLASLt on GPU, F.linear on CPU."""
+    return get_backend(x.device).linear(x, weight, bias)
+
+
 def topk_softmax(gating_logits, topk, renormalize=True):
     return get_backend(gating_logits.device).topk_softmax(
         gating_logits, topk, renormalize

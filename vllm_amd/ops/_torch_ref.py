@@ -138,6 +138,10 @@ def attention_unified(
     return out
 
 
+def linear(x, weight, bias=None):
+    return F.linear(x, weight, bias)
+
+
 def topk_softmax(gating_logits, topk, renormalize=True):
     """gating_logits: [T, E] -> (topk_weights [T,k] f32, topk_ids [T,k] i32)."""
     probs = gating_logits.float().softmax(dim=-1)

@@ -50,6 +50,11 @@ def gelu_and_mul(x):
     return _C.gelu_and_mul(x)
 
 
+def linear(x, weight, bias=None):
+    # Tuned hipBLASLt GEMM (per-shape algo search on first use).
+    return _C.lt_linear(x, weight, bias)
+
+
 def reshape_and_cache(key, value, kv_cache, slot_mapping):
     _C.reshape_and_cache(key, value, kv_cache, slot_mapping)
 
