@@ -1,0 +1,177 @@
+"""OpenAI API server tests on CPU (tiny model, mock tokenizer).
+
+Covers completions (non-stream + stream), chat completions, models,
+tokenize/detokenize, health and metrics — the surface of
+vllm/entrypoints/openai (reference tests/entrypoints/openai pattern).
+"""
+
+import json
+
+import pytest
+from fastapi.testclient import TestClient
+
+from vllm_amd.engine.arg_utils import EngineArgs
+from vllm_amd.entrypoints.openai.api_server import make_server
+
+
+@pytest.fixture(scope="module")
+def client():
+    args = EngineArgs(
+        model="tiny-llama",
+        dtype="fp32",
+        device="cpu",
+        block_size=16,
+        num_gpu_blocks=256,
+        max_model_len=512,
+        max_num_batched_tokens=512,
+        max_num_seqs=8,
+    )
+    app, state = make_server(args, served_model_name="tiny-llama")
+    with TestClient(app) as c:
+        yield c
+    state.engine.shutdown()
+
+
+def test_health_and_version(client):
+    assert client.get("/health").status_code == 200
+    assert "version" in client.get("/version").json()
+
+
+def test_models(client):
+    data = client.get("/v1/models").json()
+    assert data["object"] == "list"
+    assert data["data"][0]["id"] == "tiny-llama"
+
+
+def test_completions(client):
+    r = client.post("/v1/completions", json={
+        "model": "tiny-llama",
+        "prompt": "hello world this is a test",
+        "max_tokens": 8,
+        "temperature": 0.0,
+        "ignore_eos": True,
+    })
+    assert r.status_code == 200, r.text
+    data = r.json()
+    assert data["object"] == "text_completion"
+    assert data["usage"]["completion_tokens"] == 8
+    assert data["choices"][0]["finish_reason"] == "length"
+
+
+def test_completions_token_ids_prompt(client):
+    r = client.post("/v1/completions", json={
+        "model": "tiny-llama",
+        "prompt": [5, 6, 7, 8, 9],
+        "max_tokens": 4,
+        "temperature": 0.0,
+        "ignore_eos": True,
+    })
+    assert r.status_code == 200, r.text
+    assert r.json()["usage"]["prompt_tokens"] == 5
+
+
+def test_completions_stream(client):
+    with client.stream("POST", "/v1/completions", json={
+        "model": "tiny-llama",
+        "prompt": "stream me",
+        "max_tokens": 6,
+        "temperature": 0.0,
+        "ignore_eos": True,
+        "stream": True,
+    }) as r:
+        assert r.status_code == 200
+        chunks = []
+        for line in r.iter_lines():
+            if line.startswith("data: "):
+                payload = line[6:]
+                if payload == "[DONE]":
+                    break
+                chunks.append(json.loads(payload))
+        assert len(chunks) >= 1
+        assert chunks[-1]["choices"][0]["finish_reason"] == "length"
+
+
+def test_chat_completions(client):
+    r = client.post("/v1/chat/completions", json={
+        "model": "tiny-llama",
+        "messages": [
+            {"role": "system", "content": "You are a test."},
+            {"role": "user", "content": "Say something."},
+        ],
+        "max_tokens": 5,
+        "temperature": 0.0,
+        "ignore_eos": True,
+    })
+    assert r.status_code == 200, r.text
+    data = r.json()
+    assert data["object"] == "chat.completion"
+    assert data["choices"][0]["message"]["role"] == "assistant"
+    assert data["usage"]["completion_tokens"] == 5
+
+
+def test_chat_stream(client):
+    with client.stream("POST", "/v1/chat/completions", json={
+        "model": "tiny-llama",
+        "messages": [{"role": "user", "content": "hi"}],
+        "max_tokens": 4,
+        "temperature": 0.0,
+        "ignore_eos": True,
+        "stream": True,
+    }) as r:
+        got_role = False
+        got_done = False
+        for line in r.iter_lines():
+            if line.startswith("data: "):
+                payload = line[6:]
+                if payload == "[DONE]":
+                    got_done = True
+                    break
+                d = json.loads(payload)
+                if d["choices"][0]["delta"].get("role") == "assistant":
+                    got_role = True
+        assert got_role and got_done
+
+
+def test_tokenize_roundtrip(client):
+    r = client.post("/tokenize", json={"prompt": "round trip"})
+    toks = r.json()["tokens"]
+    assert r.json()["count"] == len(toks) > 0
+    r2 = client.post("/detokenize", json={"tokens": toks})
+    assert isinstance(r2.json()["prompt"], str)
+
+
+def test_metrics(client):
+    text = client.get("/metrics").text
+    assert "vllm_amd:num_requests_total" in text
+    assert "vllm_amd:kv_blocks_free" in text
+
+
+def test_stop_string(client):
+    # Force a stop string that can't appear -> finishes by length.
+    r = client.post("/v1/completions", json={
+        "model": "tiny-llama",
+        "prompt": "check stops",
+        "max_tokens": 4,
+        "temperature": 0.0,
+        "stop": ["never"],
+        "ignore_eos": True,
+    })
+    assert r.json()["choices"][0]["finish_reason"] == "length"
+
+
+def test_concurrent_requests(client):
+    import concurrent.futures as cf
+
+    def one(i):
+        r = client.post("/v1/completions", json={
+            "model": "tiny-llama",
+            "prompt": f"request number {i}",
+            "max_tokens": 6,
+            "temperature": 0.0,
+            "ignore_eos": True,
+        })
+        return r.json()["usage"]["completion_tokens"]
+
+    with cf.ThreadPoolExecutor(4) as ex:
+        results = list(ex.map(one, range(8)))
+    assert results == [6] * 8

@@ -1,0 +1,28 @@
+"""CLI: `python -m vllm_amd {serve,bench,chat}` (role of the reference's
+vllm/entrypoints/cli/main.py)."""
+
+from __future__ import annotations
+
+import sys
+
+
+def main() -> None:
+    if len(sys.argv) < 2 or sys.argv[1] in ("-h", "--help"):
+        print("usage: python -m vllm_amd {serve} [args]\n"
+              "  serve  — start the OpenAI-compatible API server")
+        return
+    cmd = sys.argv.pop(1)
+    if cmd == "serve":
+        # `python -m vllm_amd serve <model> [args]` or with --model.
+        if len(sys.argv) > 1 and not sys.argv[1].startswith("-"):
+            model = sys.argv.pop(1)
+            sys.argv.extend(["--model", model])
+        from vllm_amd.entrypoints.openai.api_server import main as serve_main
+
+        serve_main()
+    else:
+        raise SystemExit(f"unknown command {cmd!r}")
+
+
+if __name__ == "__main__":
+    main()

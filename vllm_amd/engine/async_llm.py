@@ -1,0 +1,141 @@
+"""AsyncLLM: asyncio front of the engine (role of vllm/v1/engine/
+async_llm.py:72).
+
+The engine loop runs on a dedicated thread (the GPU step is blocking);
+requests arrive through a thread-safe queue and outputs are delivered to
+per-request asyncio queues via call_soon_threadsafe. One engine step per
+loop iteration — continuous batching keeps all active requests moving.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import logging
+import queue
+import threading
+import time
+from typing import AsyncGenerator, Optional, Union
+
+from vllm_amd.config import EngineConfig
+from vllm_amd.engine.llm_engine import LLMEngine
+from vllm_amd.outputs import RequestOutput
+from vllm_amd.sampling_params import SamplingParams
+
+logger = logging.getLogger(__name__)
+
+
+class AsyncLLM:
+
+    def __init__(self, config: EngineConfig):
+        self.engine = LLMEngine(config)
+        self.config = config
+        # (request_id, prompt, params, loop, out_queue) | ("abort", rid)
+        self._inbox: queue.Queue = queue.Queue()
+        self._streams: dict[str, tuple[asyncio.AbstractEventLoop,
+                                       asyncio.Queue]] = {}
+        self._errors: dict[str, Exception] = {}
+        self._shutdown = False
+        self._counter = 0
+        self._counter_lock = threading.Lock()
+        self._thread = threading.Thread(
+            target=self._run_loop, daemon=True, name="engine-loop"
+        )
+        self._thread.start()
+
+    @property
+    def tokenizer(self):
+        return self.engine.tokenizer
+
+    # ------------------------------------------------------------------
+    def _run_loop(self) -> None:
+        while not self._shutdown:
+            # Drain the inbox (block briefly when idle).
+            block = not self.engine.has_unfinished_requests()
+            while True:
+                try:
+                    item = self._inbox.get(timeout=0.005 if block else 0)
+                except queue.Empty:
+                    break
+                block = False
+                kind = item[0]
+                if kind == "add":
+                    _, rid, prompt, params, loop, out_q = item
+                    try:
+                        self.engine.add_request(rid, prompt, params)
+                        self._streams[rid] = (loop, out_q)
+                    except Exception as e:  # noqa: BLE001
+                        loop.call_soon_threadsafe(out_q.put_nowait, e)
+                elif kind == "abort":
+                    _, rid = item
+                    if rid in self._streams:
+                        self.engine.abort_request([rid])
+                        self._streams.pop(rid, None)
+
+            if not self.engine.has_unfinished_requests():
+                continue
+            try:
+                outputs = self.engine.step()
+            except Exception as e:  # noqa: BLE001
+                logger.exception("engine step failed")
+                for rid, (loop, out_q) in self._streams.items():
+                    loop.call_soon_threadsafe(out_q.put_nowait, e)
+                self._streams.clear()
+                continue
+            for out in outputs:
+                entry = self._streams.get(out.request_id)
+                if entry is None:
+                    continue
+                loop, out_q = entry
+                loop.call_soon_threadsafe(out_q.put_nowait, out)
+                if out.finished:
+                    self._streams.pop(out.request_id, None)
+
+    # ------------------------------------------------------------------
+    def _next_id(self) -> str:
+        with self._counter_lock:
+            self._counter += 1
+            return f"req-{self._counter}"
+
+    async def generate(
+        self,
+        prompt: Union[str, list[int]],
+        sampling_params: Optional[SamplingParams] = None,
+        request_id: Optional[str] = None,
+    ) -> AsyncGenerator[RequestOutput, None]:
+        """Submit a request and stream RequestOutputs until finished."""
+        rid = request_id or self._next_id()
+        params = sampling_params or SamplingParams()
+        loop = asyncio.get_running_loop()
+        out_q: asyncio.Queue = asyncio.Queue()
+        self._inbox.put(("add", rid, prompt, params, loop, out_q))
+        try:
+            while True:
+                item = await out_q.get()
+                if isinstance(item, Exception):
+                    raise item
+                yield item
+                if item.finished:
+                    return
+        finally:
+            # Client disconnected / cancelled: abort in the engine.
+            self._inbox.put(("abort", rid))
+
+    async def abort(self, request_id: str) -> None:
+        self._inbox.put(("abort", request_id))
+
+    def stats(self) -> dict:
+        sched = self.engine.engine_core.scheduler
+        if sched is None:
+            return {}
+        return {
+            "num_running": len(sched.running),
+            "num_waiting": len(sched.waiting),
+            "kv_blocks_total": self.engine.engine_core.num_gpu_blocks,
+            "kv_blocks_free":
+                sched.kv_cache_manager.block_pool.get_num_free_blocks(),
+        }
+
+    def shutdown(self) -> None:
+        self._shutdown = True
+        self._thread.join(timeout=5)
+        self.engine.shutdown()

@@ -1,0 +1,253 @@
+"""OpenAI-compatible request/response schemas (role of the reference's
+vllm/entrypoints/openai/protocol.py, trimmed to the supported surface)."""
+
+from __future__ import annotations
+
+import time
+import uuid
+from typing import Any, Literal, Optional, Union
+
+from pydantic import BaseModel, Field
+
+from vllm_amd.sampling_params import RequestOutputKind, SamplingParams
+
+
+def random_id(prefix: str) -> str:
+    return f"{prefix}-{uuid.uuid4().hex[:24]}"
+
+
+class CompletionRequest(BaseModel):
+    model: str
+    prompt: Union[str, list[str], list[int], list[list[int]]]
+    best_of: Optional[int] = None
+    echo: bool = False
+    frequency_penalty: float = 0.0
+    logit_bias: Optional[dict[str, float]] = None
+    logprobs: Optional[int] = None
+    max_tokens: Optional[int] = 16
+    n: int = 1
+    presence_penalty: float = 0.0
+    seed: Optional[int] = None
+    stop: Optional[Union[str, list[str]]] = None
+    stream: bool = False
+    stream_options: Optional[dict[str, Any]] = None
+    suffix: Optional[str] = None
+    temperature: float = 1.0
+    top_p: float = 1.0
+    user: Optional[str] = None
+    # Extensions (same names as the reference).
+    top_k: int = 0
+    min_p: float = 0.0
+    repetition_penalty: float = 1.0
+    min_tokens: int = 0
+    stop_token_ids: Optional[list[int]] = None
+    ignore_eos: bool = False
+    skip_special_tokens: bool = True
+
+    def to_sampling_params(self, stream: bool) -> SamplingParams:
+        logit_bias = (
+            {int(k): v for k, v in self.logit_bias.items()}
+            if self.logit_bias else None
+        )
+        return SamplingParams(
+            n=1,
+            presence_penalty=self.presence_penalty,
+            frequency_penalty=self.frequency_penalty,
+            repetition_penalty=self.repetition_penalty,
+            temperature=self.temperature,
+            top_p=self.top_p,
+            top_k=self.top_k,
+            min_p=self.min_p,
+            seed=self.seed,
+            stop=self.stop,
+            stop_token_ids=self.stop_token_ids,
+            ignore_eos=self.ignore_eos,
+            max_tokens=self.max_tokens,
+            min_tokens=self.min_tokens,
+            logprobs=self.logprobs,
+            logit_bias=logit_bias,
+            skip_special_tokens=self.skip_special_tokens,
+            output_kind=(RequestOutputKind.DELTA if stream
+                         else RequestOutputKind.FINAL_ONLY),
+        )
+
+
+class ChatMessage(BaseModel):
+    role: str
+    content: Optional[Union[str, list[dict[str, Any]]]] = None
+    name: Optional[str] = None
+
+    def text(self) -> str:
+        if isinstance(self.content, str):
+            return self.content
+        if self.content is None:
+            return ""
+        parts = []
+        for seg in self.content:
+            if seg.get("type") == "text":
+                parts.append(seg.get("text", ""))
+        return "".join(parts)
+
+
+class ChatCompletionRequest(BaseModel):
+    model: str
+    messages: list[ChatMessage]
+    frequency_penalty: float = 0.0
+    logit_bias: Optional[dict[str, float]] = None
+    logprobs: bool = False
+    top_logprobs: Optional[int] = None
+    max_tokens: Optional[int] = None
+    max_completion_tokens: Optional[int] = None
+    n: int = 1
+    presence_penalty: float = 0.0
+    seed: Optional[int] = None
+    stop: Optional[Union[str, list[str]]] = None
+    stream: bool = False
+    stream_options: Optional[dict[str, Any]] = None
+    temperature: float = 1.0
+    top_p: float = 1.0
+    user: Optional[str] = None
+    top_k: int = 0
+    min_p: float = 0.0
+    repetition_penalty: float = 1.0
+    min_tokens: int = 0
+    stop_token_ids: Optional[list[int]] = None
+    ignore_eos: bool = False
+    skip_special_tokens: bool = True
+    add_generation_prompt: bool = True
+
+    def to_sampling_params(self, stream: bool,
+                           default_max_tokens: int) -> SamplingParams:
+        max_tokens = (self.max_completion_tokens or self.max_tokens
+                      or default_max_tokens)
+        logit_bias = (
+            {int(k): v for k, v in self.logit_bias.items()}
+            if self.logit_bias else None
+        )
+        n_logprobs = (self.top_logprobs or 1) if self.logprobs else None
+        return SamplingParams(
+            n=1,
+            presence_penalty=self.presence_penalty,
+            frequency_penalty=self.frequency_penalty,
+            repetition_penalty=self.repetition_penalty,
+            temperature=self.temperature,
+            top_p=self.top_p,
+            top_k=self.top_k,
+            min_p=self.min_p,
+            seed=self.seed,
+            stop=self.stop,
+            stop_token_ids=self.stop_token_ids,
+            ignore_eos=self.ignore_eos,
+            max_tokens=max_tokens,
+            min_tokens=self.min_tokens,
+            logprobs=n_logprobs,
+            logit_bias=logit_bias,
+            skip_special_tokens=self.skip_special_tokens,
+            output_kind=(RequestOutputKind.DELTA if stream
+                         else RequestOutputKind.FINAL_ONLY),
+        )
+
+
+class UsageInfo(BaseModel):
+    prompt_tokens: int = 0
+    completion_tokens: int = 0
+    total_tokens: int = 0
+
+
+class CompletionChoice(BaseModel):
+    index: int
+    text: str
+    logprobs: Optional[dict] = None
+    finish_reason: Optional[str] = None
+    stop_reason: Optional[Union[int, str]] = None
+
+
+class CompletionResponse(BaseModel):
+    id: str = Field(default_factory=lambda: random_id("cmpl"))
+    object: Literal["text_completion"] = "text_completion"
+    created: int = Field(default_factory=lambda: int(time.time()))
+    model: str
+    choices: list[CompletionChoice]
+    usage: UsageInfo = Field(default_factory=UsageInfo)
+
+
+class ChatCompletionMessage(BaseModel):
+    role: str = "assistant"
+    content: Optional[str] = None
+
+
+class ChatChoice(BaseModel):
+    index: int
+    message: ChatCompletionMessage
+    logprobs: Optional[dict] = None
+    finish_reason: Optional[str] = None
+
+
+class ChatCompletionResponse(BaseModel):
+    id: str = Field(default_factory=lambda: random_id("chatcmpl"))
+    object: Literal["chat.completion"] = "chat.completion"
+    created: int = Field(default_factory=lambda: int(time.time()))
+    model: str
+    choices: list[ChatChoice]
+    usage: UsageInfo = Field(default_factory=UsageInfo)
+
+
+class DeltaMessage(BaseModel):
+    role: Optional[str] = None
+    content: Optional[str] = None
+
+
+class ChatStreamChoice(BaseModel):
+    index: int
+    delta: DeltaMessage
+    finish_reason: Optional[str] = None
+
+
+class ChatCompletionStreamResponse(BaseModel):
+    id: str
+    object: Literal["chat.completion.chunk"] = "chat.completion.chunk"
+    created: int = Field(default_factory=lambda: int(time.time()))
+    model: str
+    choices: list[ChatStreamChoice]
+    usage: Optional[UsageInfo] = None
+
+
+class ModelCard(BaseModel):
+    id: str
+    object: Literal["model"] = "model"
+    created: int = Field(default_factory=lambda: int(time.time()))
+    owned_by: str = "vllm_amd"
+    max_model_len: Optional[int] = None
+
+
+class ModelList(BaseModel):
+    object: Literal["list"] = "list"
+    data: list[ModelCard] = Field(default_factory=list)
+
+
+class TokenizeRequest(BaseModel):
+    model: Optional[str] = None
+    prompt: str
+    add_special_tokens: bool = True
+
+
+class TokenizeResponse(BaseModel):
+    tokens: list[int]
+    count: int
+    max_model_len: int
+
+
+class DetokenizeRequest(BaseModel):
+    model: Optional[str] = None
+    tokens: list[int]
+
+
+class DetokenizeResponse(BaseModel):
+    prompt: str
+
+
+class ErrorResponse(BaseModel):
+    object: Literal["error"] = "error"
+    message: str
+    type: str = "invalid_request_error"
+    code: int = 400
