@@ -158,3 +158,19 @@ class TestTextPath:
         out = finals[rid]
         assert out.prompt == "hello world"
         assert isinstance(out.outputs[0].text, str)
+
+
+def test_mixtral_cpu_decode():
+    """Tiny Mixtral end-to-end on CPU (MoE routing + expert MLP)."""
+    from vllm_amd.entrypoints.llm import LLM
+    from vllm_amd.sampling_params import SamplingParams
+
+    llm = LLM(model="tiny-mixtral", dtype="fp32", device="cpu",
+              block_size=16, num_gpu_blocks=128, max_model_len=256,
+              max_num_batched_tokens=256, max_num_seqs=4)
+    outs = llm.generate(
+        [[1, 2, 3, 4, 5], [9, 8, 7]],
+        SamplingParams(temperature=0.0, max_tokens=6, ignore_eos=True),
+    )
+    llm.shutdown()
+    assert all(len(o.outputs[0].token_ids) == 6 for o in outs)

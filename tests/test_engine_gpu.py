@@ -97,3 +97,15 @@ def test_native_extension_is_loaded():
     from pathlib import Path
     so = Path(vllm_amd.__file__).parent / "_C.so"
     assert so.exists()
+
+
+def test_mixtral_gpu_decode():
+    from vllm_amd.sampling_params import SamplingParams
+
+    llm = _llm(model="tiny-mixtral-128")
+    outs = llm.generate(
+        [list(range(3, 40)), [5, 6, 7]],
+        SamplingParams(temperature=0.0, max_tokens=8, ignore_eos=True),
+    )
+    llm.shutdown()
+    assert all(len(o.outputs[0].token_ids) == 8 for o in outs)

@@ -232,3 +232,16 @@ def test_lt_linear(m, k, n, bias):
     expect = torch.nn.functional.linear(
         a.float(), w.float(), b.float() if bias else None)
     assert_close(out, expect, atol=5e-2, rtol=5e-2)
+
+
+def test_fused_moe_matches_reference():
+    T, H, I, E, K = 37, 256, 512, 4, 2
+    hidden = torch.randn(T, H, dtype=torch.bfloat16, device="cuda") * 0.1
+    w13 = torch.randn(E, 2 * I, H, dtype=torch.bfloat16, device="cuda") * 0.1
+    w2 = torch.randn(E, H, I, dtype=torch.bfloat16, device="cuda") * 0.1
+    logits = torch.randn(T, E, dtype=torch.float32, device="cuda")
+    hip = _hip()
+    tw, ti = hip.topk_softmax(logits, K)
+    out = hip.fused_moe(hidden, w13, w2, tw, ti)
+    expect = ref.fused_moe(hidden.float(), w13.float(), w2.float(), tw, ti)
+    assert_close(out, expect, atol=5e-2, rtol=5e-2)

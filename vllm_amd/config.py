@@ -203,6 +203,23 @@ MODEL_PRESETS: dict[str, ModelSpec] = {
         max_position_embeddings=2048,
         eos_token_id=2,
     ),
+    "tiny-mixtral-128": ModelSpec(
+        name="tiny-mixtral-128",
+        architecture="mixtral",
+        vocab_size=1024,
+        hidden_size=512,
+        intermediate_size=1024,
+        num_layers=2,
+        num_heads=4,
+        num_kv_heads=2,
+        head_dim=128,
+        rope_theta=10000.0,
+        max_position_embeddings=2048,
+        num_experts=4,
+        num_experts_per_tok=2,
+        moe_intermediate_size=512,
+        eos_token_id=2,
+    ),
     "tiny-mixtral": ModelSpec(
         name="tiny-mixtral",
         architecture="mixtral",

@@ -116,7 +116,39 @@ def attention_unified(
     return out
 
 
-# MoE ops: HIP grouped-GEMM kernels land with the Mixtral milestone; until
-# then the documented GPU implementation is the eager composition below
-# (routing + expert loop) — not a silent fallback of an existing kernel.
-from vllm_amd.ops._torch_ref import fused_moe, topk_softmax  # noqa: E402,F401
+# MoE routing (small [T, E] tensors — torch ops are fine here).
+from vllm_amd.ops._torch_ref import topk_softmax  # noqa: E402,F401
+
+
+def fused_moe(hidden, w13, w2, topk_weights, topk_ids, activation="silu"):
+    """MoE expert MLP on GPU: tokens sorted by expert, one tuned hipBLASLt
+    GEMM pair per non-empty expert, weighted scatter-add combine.
+
+    hidden: [T, H]; w13: [E, 2I, H]; w2: [E, H, I]. One D2H sync per call
+    for the segment sizes (a dedicated grouped-GEMM MFMA kernel replaces
+    this; role of the reference's fused_moe.py:299 Triton grouped GEMM).
+    """
+    T, H = hidden.shape
+    E = w13.shape[0]
+    k = topk_ids.shape[1]
+    flat = topk_ids.long().flatten()
+    order = torch.argsort(flat, stable=True)
+    token_of = order // k
+    counts = torch.bincount(flat, minlength=E).cpu().tolist()
+    x = hidden[token_of].contiguous()
+    out_sorted = torch.empty(T * k, H, dtype=hidden.dtype,
+                             device=hidden.device)
+    act = silu_and_mul if activation == "silu" else gelu_and_mul
+    offset = 0
+    for e in range(E):
+        c = counts[e]
+        if c == 0:
+            continue
+        seg = x[offset:offset + c]
+        h = act(linear(seg, w13[e]))
+        out_sorted[offset:offset + c] = linear(h, w2[e])
+        offset += c
+    w = topk_weights.flatten()[order].unsqueeze(1).float()
+    out = torch.zeros(T, H, dtype=torch.float32, device=hidden.device)
+    out.index_add_(0, token_of, out_sorted.float() * w)
+    return out.to(hidden.dtype)

@@ -223,8 +223,12 @@ class ModelRunner:
             )
             for _ in range(spec.num_layers)
         ]
+        # MoE models: the segmented-GEMM fused_moe syncs for expert
+        # counts, which is not graph-capturable (until the grouped-GEMM
+        # kernel lands).
         if (self.device.type == "cuda"
-                and not self.config.model_config.enforce_eager):
+                and not self.config.model_config.enforce_eager
+                and not self.spec.is_moe):
             self.graph_runner = DecodeGraphRunner(self)
 
     # ------------------------------------------------------------------
