@@ -46,11 +46,14 @@ class GroupCoordinator:
             return t
         if dim < 0:
             dim += t.dim()
-        # Gather along a new leading dim then reshape to concat on `dim`.
+        # Gather flat on dim 0 (collective layout), then view as
+        # [world, *shape] to concat on `dim`.
         out = torch.empty(
-            (self.world_size,) + tuple(t.shape), dtype=t.dtype, device=t.device
+            (self.world_size * t.shape[0],) + tuple(t.shape[1:]),
+            dtype=t.dtype, device=t.device,
         )
         dist.all_gather_into_tensor(out, t.contiguous(), group=self.device_group)
+        out = out.view((self.world_size,) + tuple(t.shape))
         if dim == 0:
             return out.reshape(-1, *t.shape[1:])
         pieces = out.unbind(0)
