@@ -1,0 +1,135 @@
+"""Safetensors checkpoint loading with TP-aware sharding.
+
+Role of the reference's model_loader (vllm/model_executor/model_loader/
+default_loader.py + per-model load_weights): iterate *.safetensors
+shards in a local directory and route each HF-named tensor to the right
+sharded parameter. Works for the Llama and Mixtral families (OPT uses
+the same q/k/v + fc naming mapped below).
+"""
+
+from __future__ import annotations
+
+import glob
+import logging
+import os
+
+import torch
+
+logger = logging.getLogger(__name__)
+
+
+def _iter_safetensors(path: str):
+    from safetensors import safe_open
+
+    files = sorted(glob.glob(os.path.join(path, "*.safetensors")))
+    if not files:
+        raise FileNotFoundError(f"no *.safetensors under {path}")
+    for f in files:
+        with safe_open(f, framework="pt", device="cpu") as sf:
+            for name in sf.keys():
+                yield name, sf.get_tensor(name)
+
+
+def load_safetensors_weights(model: torch.nn.Module, config) -> None:
+    """Load HF-layout weights into the TP-sharded model. The model is on
+    CPU at this point (moved to device afterwards by load_model)."""
+    path = config.model_path or config.model
+    spec = config.spec
+    dtype = config.torch_dtype
+
+    # Collect q/k/v and gate/up pieces so fused layers load atomically.
+    pending: dict[str, dict[str, torch.Tensor]] = {}
+    moe_pending: dict[str, dict[str, torch.Tensor]] = {}
+
+    root = model
+    layers = root.model.layers
+
+    def layer_of(name: str):
+        parts = name.split(".")
+        idx = int(parts[parts.index("layers") + 1])
+        return layers[idx]
+
+    n_loaded = 0
+    for name, w in _iter_safetensors(path):
+        w = w.to(dtype)
+        n_loaded += 1
+        if name.endswith("rotary_emb.inv_freq"):
+            continue
+        if name == "model.embed_tokens.weight":
+            root.model.embed_tokens.load_weight(w)
+            if spec.tie_word_embeddings:
+                pass  # lm_head shares the parameter
+            continue
+        if name == "lm_head.weight":
+            if not spec.tie_word_embeddings:
+                root.lm_head.load_weight(w)
+            continue
+        if name == "model.norm.weight":
+            root.model.norm.weight.data.copy_(w)
+            continue
+        if ".layers." not in name:
+            logger.warning("unmatched tensor %s", name)
+            continue
+
+        layer = layer_of(name)
+        key = name.rsplit(".", 2)[0]  # strip trailing proj.weight
+
+        if ".self_attn." in name:
+            attn = layer.self_attn
+            if any(p in name for p in ("q_proj", "k_proj", "v_proj")):
+                d = pending.setdefault(name.split(".self_attn.")[0], {})
+                which = name.split("self_attn.")[1].split(".")[0]
+                kind = "bias" if name.endswith("bias") else "weight"
+                d[f"{which}.{kind}"] = w
+                if all(f"{p}.weight" in d for p in
+                       ("q_proj", "k_proj", "v_proj")):
+                    attn.qkv_proj.load_qkv(
+                        d["q_proj.weight"], d["k_proj.weight"],
+                        d["v_proj.weight"],
+                        d.get("q_proj.bias"), d.get("k_proj.bias"),
+                        d.get("v_proj.bias"),
+                    )
+            elif "o_proj" in name or "out_proj" in name:
+                if name.endswith("bias"):
+                    attn.o_proj.load_bias(w)
+                else:
+                    attn.o_proj.load_weight(w)
+            continue
+
+        if ".mlp." in name:
+            mlp = layer.mlp
+            if "gate_proj" in name:
+                mlp.gate_up_proj.load_sub_weight(0, w)
+            elif "up_proj" in name:
+                mlp.gate_up_proj.load_sub_weight(1, w)
+            elif "down_proj" in name:
+                mlp.down_proj.load_weight(w)
+            continue
+
+        if ".block_sparse_moe." in name:
+            moe = layer.block_sparse_moe
+            if "gate.weight" in name:
+                moe.gate.weight.data.copy_(w)
+                continue
+            # experts.N.w{1,2,3}.weight
+            lkey = name.split(".block_sparse_moe.")[0]
+            d = moe_pending.setdefault(lkey, {})
+            seg = name.split(".experts.")[1]  # "N.wX.weight"
+            d[seg] = w
+            E = spec.num_experts
+            if len(d) == 3 * E:
+                w1 = torch.stack([d[f"{e}.w1.weight"] for e in range(E)])
+                w2 = torch.stack([d[f"{e}.w2.weight"] for e in range(E)])
+                w3 = torch.stack([d[f"{e}.w3.weight"] for e in range(E)])
+                moe.load_full_weights(w1, w3, w2)
+                moe_pending.pop(lkey)
+            continue
+
+        if "input_layernorm" in name:
+            layer.input_layernorm.weight.data.copy_(w)
+        elif "post_attention_layernorm" in name:
+            layer.post_attention_layernorm.weight.data.copy_(w)
+        else:
+            logger.warning("unmatched tensor %s (key %s)", name, key)
+
+    logger.info("loaded %d tensors from %s", n_loaded, path)
