@@ -174,3 +174,22 @@ def test_mixtral_cpu_decode():
     )
     llm.shutdown()
     assert all(len(o.outputs[0].token_ids) == 6 for o in outs)
+
+
+def test_deepseek_mla_cpu_decode():
+    """Tiny DeepSeek (MLA + sigmoid group-routed MoE + shared experts)
+    end-to-end on CPU, incl. chunked prefill over the compressed cache."""
+    from vllm_amd.entrypoints.llm import LLM
+    from vllm_amd.sampling_params import SamplingParams
+
+    llm = LLM(model="tiny-deepseek", dtype="fp32", device="cpu",
+              block_size=16, num_gpu_blocks=128, max_model_len=256,
+              max_num_batched_tokens=32, max_num_seqs=4)
+    prompts = [[(i * 11 + j) % 900 + 3 for j in range(50)] for i in range(2)]
+    p = SamplingParams(temperature=0.0, max_tokens=6, ignore_eos=True)
+    outs = llm.generate(prompts, p)
+    outs2 = llm.generate(prompts, p)
+    llm.shutdown()
+    assert all(len(o.outputs[0].token_ids) == 6 for o in outs)
+    for a, b in zip(outs, outs2):
+        assert a.outputs[0].token_ids == b.outputs[0].token_ids

@@ -109,3 +109,14 @@ def test_mixtral_gpu_decode():
     )
     llm.shutdown()
     assert all(len(o.outputs[0].token_ids) == 8 for o in outs)
+
+
+def test_deepseek_mla_gpu_decode():
+    from vllm_amd.sampling_params import SamplingParams
+
+    llm = _llm(model="tiny-deepseek", block_size=16)
+    prompts = [[(i * 11 + j) % 900 + 3 for j in range(50)] for i in range(2)]
+    p = SamplingParams(temperature=0.0, max_tokens=6, ignore_eos=True)
+    outs = llm.generate(prompts, p)
+    llm.shutdown()
+    assert all(len(o.outputs[0].token_ids) == 6 for o in outs)

@@ -220,6 +220,34 @@ MODEL_PRESETS: dict[str, ModelSpec] = {
         moe_intermediate_size=512,
         eos_token_id=2,
     ),
+    "tiny-deepseek": ModelSpec(
+        name="tiny-deepseek",
+        architecture="deepseek",
+        vocab_size=1024,
+        hidden_size=256,
+        intermediate_size=512,
+        num_layers=2,
+        num_heads=8,
+        num_kv_heads=8,
+        head_dim=64,
+        rope_theta=10000.0,
+        max_position_embeddings=2048,
+        num_experts=8,
+        num_experts_per_tok=2,
+        moe_intermediate_size=128,
+        num_shared_experts=1,
+        first_dense_layers=1,
+        routed_scaling_factor=1.5,
+        scoring_func="sigmoid",
+        n_group=4,
+        topk_group=2,
+        q_lora_rank=64,
+        kv_lora_rank=64,
+        qk_nope_head_dim=32,
+        qk_rope_head_dim=16,
+        v_head_dim=32,
+        eos_token_id=2,
+    ),
     "tiny-mixtral": ModelSpec(
         name="tiny-mixtral",
         architecture="mixtral",

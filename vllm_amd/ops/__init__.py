@@ -110,6 +110,28 @@ def linear(x, weight, bias=None):
     return get_backend(x.device).linear(x, weight, bias)
 
 
+def concat_and_cache_mla(c_kv, k_pe, kv_cache, slot_mapping):
+    return get_backend(c_kv.device).concat_and_cache_mla(
+        c_kv, k_pe, kv_cache, slot_mapping
+    )
+
+
+def mla_attention(q_nope, q_pe, kv_cache, block_table, query_start_loc,
+                  seq_lens, scale):
+    return get_backend(q_nope.device).mla_attention(
+        q_nope, q_pe, kv_cache, block_table, query_start_loc, seq_lens, scale
+    )
+
+
+def grouped_topk(gating, topk, renormalize=True, num_groups=0,
+                 topk_groups=0, scoring_func="softmax", e_score_bias=None,
+                 routed_scaling_factor=1.0):
+    return get_backend(gating.device).grouped_topk(
+        gating, topk, renormalize, num_groups, topk_groups, scoring_func,
+        e_score_bias, routed_scaling_factor
+    )
+
+
 def topk_softmax(gating_logits, topk, renormalize=True):
     return get_backend(gating_logits.device).topk_softmax(
         gating_logits, topk, renormalize

@@ -173,3 +173,92 @@ def fused_moe(hidden, w13, w2, topk_weights, topk_ids, activation="silu"):
         w = topk_weights[token_idx, k_idx].unsqueeze(-1).float()
         out.index_add_(0, token_idx, y.float() * w)
     return out.to(hidden.dtype)
+
+
+def concat_and_cache_mla(c_kv, k_pe, kv_cache, slot_mapping):
+    """MLA cache write: c_kv [T, lora], k_pe [T, rope] ->
+    kv_cache [num_blocks, block_size, lora+rope]."""
+    block_size = kv_cache.shape[1]
+    blk = slot_mapping // block_size
+    off = slot_mapping % block_size
+    lora = c_kv.shape[1]
+    kv_cache[blk, off, :lora] = c_kv
+    kv_cache[blk, off, lora:] = k_pe
+
+
+def mla_attention(
+    q_nope,       # [T, Hq, d_nope]  (already absorbed: d_nope == kv_lora)
+    q_pe,         # [T, Hq, d_rope]
+    kv_cache,     # [num_blocks, block_size, kv_lora + d_rope]
+    block_table,  # [num_reqs, max_blocks] int32
+    query_start_loc,  # [num_reqs+1]
+    seq_lens,     # [num_reqs]
+    scale,
+):
+    """Attention in the compressed (absorbed) MLA space: scores =
+    q_nope . c_kv + q_pe . k_pe; output is in kv_lora space [T, Hq, lora]
+    (caller applies W_UV). Causal over each request's own new tokens.
+
+    Role of the reference's MLA backends (vllm/v1/attention/backends/mla/)
+    in plain torch — the numerics ground truth for the HIP MLA kernel.
+    """
+    T, Hq, lora = q_nope.shape
+    d_rope = q_pe.shape[2]
+    block_size = kv_cache.shape[1]
+    out = q_nope.new_empty(T, Hq, lora)
+    qs = query_start_loc.tolist()
+    num_reqs = seq_lens.shape[0]
+    for i in range(num_reqs):
+        s, e = qs[i], qs[i + 1]
+        ql = e - s
+        ctx = int(seq_lens[i])
+        nb = (ctx + block_size - 1) // block_size
+        blocks = block_table[i, :nb].long()
+        kv = kv_cache[blocks].reshape(-1, lora + d_rope)[:ctx].float()
+        c_kv, k_pe = kv[:, :lora], kv[:, lora:]
+        qn = q_nope[s:e].float()  # [ql, Hq, lora]
+        qp = q_pe[s:e].float()
+        scores = (
+            torch.einsum("qhl,kl->hqk", qn, c_kv)
+            + torch.einsum("qhr,kr->hqk", qp, k_pe)
+        ) * scale
+        kpos = torch.arange(ctx, device=qn.device)
+        qpos = ctx - ql + torch.arange(ql, device=qn.device)
+        mask = kpos.unsqueeze(0) > qpos.unsqueeze(1)
+        scores.masked_fill_(mask.unsqueeze(0), float("-inf"))
+        p = scores.softmax(dim=-1)
+        o = torch.einsum("hqk,kl->qhl", p, c_kv)
+        out[s:e] = o.to(out.dtype)
+    return out
+
+
+def grouped_topk(
+    gating, topk, renormalize=True, num_groups=0, topk_groups=0,
+    scoring_func="softmax", e_score_bias=None, routed_scaling_factor=1.0,
+):
+    """DeepSeek-style routing: sigmoid/softmax scoring, optional
+    group-limited expert selection (n_group/topk_group), optional
+    per-expert score bias (role of fused_moe grouped_topk)."""
+    T, E = gating.shape
+    if scoring_func == "sigmoid":
+        scores = gating.float().sigmoid()
+    else:
+        scores = gating.float().softmax(dim=-1)
+    select = scores if e_score_bias is None else scores + e_score_bias
+    if num_groups > 0 and topk_groups > 0:
+        g = select.view(T, num_groups, E // num_groups)
+        # Group score: sum of top-2 experts inside each group.
+        group_score = g.topk(min(2, g.shape[-1]), dim=-1).values.sum(-1)
+        keep = group_score.topk(topk_groups, dim=-1).indices
+        mask = torch.zeros(T, num_groups, dtype=torch.bool,
+                           device=gating.device)
+        mask.scatter_(1, keep, True)
+        select = select.masked_fill(
+            ~mask.unsqueeze(-1).expand_as(g).reshape(T, E), float("-inf")
+        )
+    topk_ids = select.topk(topk, dim=-1).indices
+    topk_weights = scores.gather(1, topk_ids)
+    if renormalize:
+        topk_weights = topk_weights / topk_weights.sum(-1, keepdim=True)
+    topk_weights = topk_weights * routed_scaling_factor
+    return topk_weights, topk_ids.to(torch.int32)

@@ -116,8 +116,15 @@ def attention_unified(
     return out
 
 
-# MoE routing (small [T, E] tensors — torch ops are fine here).
-from vllm_amd.ops._torch_ref import topk_softmax  # noqa: E402,F401
+# MoE routing (small [T, E] tensors — torch ops are fine here) and the
+# MLA compressed-attention path (torch composition today; dedicated HIP
+# MLA decode kernel is the known next step — see SURVEY §9).
+from vllm_amd.ops._torch_ref import (  # noqa: E402,F401
+    concat_and_cache_mla,
+    grouped_topk,
+    mla_attention,
+    topk_softmax,
+)
 
 
 def fused_moe(hidden, w13, w2, topk_weights, topk_ids, activation="silu"):

@@ -201,8 +201,13 @@ class ModelRunner:
         from vllm_amd.parallel.state import get_tp_world_size
 
         spec = self.spec
-        kv_heads = spec.num_kv_heads_per_rank(get_tp_world_size())
         elt = torch.tensor([], dtype=self.dtype).element_size()
+        if spec.is_mla:
+            # Compressed MLA cache: kv_lora + rope values per token,
+            # replicated across TP ranks (vs per-head K+V for GQA).
+            per_tok = spec.kv_lora_rank + spec.qk_rope_head_dim
+            return self.block_size * per_tok * elt * spec.num_layers
+        kv_heads = spec.num_kv_heads_per_rank(get_tp_world_size())
         return (
             2 * self.block_size * kv_heads * spec.head_dim * elt
             * spec.num_layers
@@ -212,6 +217,14 @@ class ModelRunner:
         from vllm_amd.parallel.state import get_tp_world_size
 
         spec = self.spec
+        if spec.is_mla:
+            per_tok = spec.kv_lora_rank + spec.qk_rope_head_dim
+            self.kv_caches = [
+                torch.zeros(num_blocks, self.block_size, per_tok,
+                            dtype=self.dtype, device=self.device)
+                for _ in range(spec.num_layers)
+            ]
+            return
         kv_heads = spec.num_kv_heads_per_rank(get_tp_world_size())
         # Head-major block layout: each (block, head) KV tile is one
         # contiguous block_size*head_dim chunk (16 KB at 64x128 bf16) —
@@ -228,7 +241,7 @@ class ModelRunner:
         # kernel lands).
         if (self.device.type == "cuda"
                 and not self.config.model_config.enforce_eager
-                and not self.spec.is_moe):
+                and not self.spec.is_moe and not self.spec.is_mla):
             self.graph_runner = DecodeGraphRunner(self)
 
     # ------------------------------------------------------------------
