@@ -57,12 +57,14 @@ def make_request(req_id, num_tokens=32, max_tokens=16, prompt=None):
 
 def fake_runner_output(sched_out, requests, next_token=7):
     """Emulates the model runner: sample a token for every request whose
-    scheduled extent reaches its total token count."""
+    scheduled extent reaches its total token count. (The scheduler
+    advances num_computed_tokens at schedule time, so a request samples
+    when its post-schedule computed count covers all known tokens.)"""
     req_ids, sampled = [], []
     for req_id, n in sched_out.num_scheduled_tokens.items():
         req = requests[req_id]
         req_ids.append(req_id)
-        if req.num_computed_tokens + n >= req.num_tokens:
+        if req.num_computed_tokens >= req.num_tokens:
             sampled.append([next_token])
         else:
             sampled.append([])

@@ -371,6 +371,9 @@ class SchedulerConfig:
     max_num_batched_tokens: int = 8192
     max_num_seqs: int = 256
     enable_chunked_prefill: bool = True
+    # Pipeline CPU scheduling with GPU execution (one step in flight;
+    # role of the reference's AsyncScheduler + async model-runner output).
+    async_scheduling: bool = True
     long_prefill_token_threshold: int = 0  # 0 -> no cap beyond token budget
     policy: str = "fcfs"  # fcfs | priority
 

@@ -118,7 +118,9 @@ class Scheduler:
         req_index = 0
         while req_index < len(self.running) and token_budget > 0:
             request = self.running[req_index]
-            num_new_tokens = request.num_tokens - request.num_computed_tokens
+            num_new_tokens = (request.num_tokens
+                              + request.num_output_placeholders
+                              - request.num_computed_tokens)
             if self.long_prefill_token_threshold > 0:
                 num_new_tokens = min(
                     num_new_tokens, self.long_prefill_token_threshold
@@ -171,6 +173,12 @@ class Scheduler:
             )
             num_scheduled_tokens[request.request_id] = num_new_tokens
             token_budget -= num_new_tokens
+            # Advance optimistically at schedule time (async scheduling:
+            # the next schedule() may run before this step's results).
+            request.num_computed_tokens += num_new_tokens
+            if (request.num_computed_tokens
+                    == request.num_tokens + request.num_output_placeholders):
+                request.num_output_placeholders += 1
             req_index += 1
 
         # ---- WAITING loop (reference scheduler.py:692) ----
@@ -245,6 +253,10 @@ class Scheduler:
                 )
             num_scheduled_tokens[request.request_id] = num_new_tokens
             token_budget -= num_new_tokens
+            request.num_computed_tokens += num_new_tokens
+            if (request.num_computed_tokens
+                    == request.num_tokens + request.num_output_placeholders):
+                request.num_output_placeholders += 1
 
         total = sum(num_scheduled_tokens.values())
         out = SchedulerOutput(
@@ -262,6 +274,7 @@ class Scheduler:
         self.kv_cache_manager.free(request)
         request.status = RequestStatus.PREEMPTED
         request.num_computed_tokens = 0
+        request.num_output_placeholders = 0
         request.num_preemptions += 1
         self.waiting.appendleft(request)
 
@@ -283,7 +296,6 @@ class Scheduler:
             request = self.requests.get(req_id)
             if request is None or request.is_finished():
                 continue  # aborted mid-step
-            request.num_computed_tokens += num_sched
 
             new_token_ids = sampled_by_req.get(req_id) or []
             if not new_token_ids:
@@ -294,6 +306,8 @@ class Scheduler:
             for tok in new_token_ids:
                 kept_tokens.append(tok)
                 request.append_output_token_ids([tok])
+                request.num_output_placeholders = max(
+                    0, request.num_output_placeholders - 1)
                 stopped = self._check_stop(request, tok)
                 if stopped:
                     break

@@ -58,6 +58,9 @@ class EngineCore:
             if self.is_driver
             else None
         )
+        self.async_scheduling = config.scheduler_config.async_scheduling
+        # One in-flight (SchedulerOutput, AsyncModelOutput) when pipelining.
+        self._pending = None
 
     # ------------------------------------------------------------------
     def add_request(self, request: Request) -> None:
@@ -75,16 +78,39 @@ class EngineCore:
             self.scheduler.has_unfinished_requests()
 
     # ------------------------------------------------------------------
+    def _drain(self) -> list[EngineCoreOutput]:
+        if self._pending is None:
+            return []
+        so_prev, fut_prev = self._pending
+        self._pending = None
+        return self.scheduler.update_from_output(so_prev, fut_prev.result())
+
     def step(self) -> list[EngineCoreOutput]:
-        """One engine iteration on the driver rank."""
+        """One engine iteration on the driver rank. With async scheduling
+        the CPU schedules step N+1 while the GPU runs step N (placeholder
+        tokens; results consumed one step later)."""
         assert self.is_driver
         scheduler_output = self.scheduler.schedule()
         if self.world.world_size > 1:
             self.world.broadcast_object(scheduler_output, src=0)
-        runner_output = self.worker.execute_model(scheduler_output)
-        return self.scheduler.update_from_output(
-            scheduler_output, runner_output
+        pure_decode = (
+            scheduler_output.total_num_scheduled_tokens > 0
+            and scheduler_output.total_num_scheduled_tokens
+            == len(scheduler_output.num_scheduled_tokens)
         )
+        if not (self.async_scheduling and pure_decode):
+            # Mixed/prefill/empty steps run synchronously: the runner's
+            # slow path reads token values the pending step produces.
+            outputs = self._drain()
+            runner_output = self.worker.execute_model(scheduler_output)
+            outputs += self.scheduler.update_from_output(
+                scheduler_output, runner_output
+            )
+            return outputs
+        fut = self.worker.execute_model_async(scheduler_output)
+        outputs = self._drain()
+        self._pending = (scheduler_output, fut)
+        return outputs
 
     def step_worker(self) -> bool:
         """Non-driver ranks: receive ONE SchedulerOutput broadcast and
@@ -104,5 +130,10 @@ class EngineCore:
             pass
 
     def shutdown(self) -> None:
+        if self.is_driver and self._pending is not None:
+            try:
+                self._drain()
+            except Exception:  # noqa: BLE001
+                pass
         if self.is_driver and self.world.world_size > 1:
             self.world.broadcast_object(None, src=0)

@@ -62,6 +62,10 @@ class Request:
         self.stop_reason: Optional[object] = None
 
         self.output_token_ids: list[int] = []
+        # Async scheduling: tokens scheduled whose sampled values are not
+        # yet known (role of AsyncScheduler's num_output_placeholders,
+        # vllm/v1/core/sched/async_scheduler.py:12).
+        self.num_output_placeholders = 0
         # All token ids: prompt + generated. Kept as one list so attention
         # metadata / block hashing index into a single sequence.
         self._all_token_ids: list[int] = list(prompt_token_ids)

@@ -45,6 +45,24 @@ def _cdiv(a: int, b: int) -> int:
     return (a + b - 1) // b
 
 
+class AsyncModelOutput:
+    """Deferred ModelRunnerOutput: GPU work + D2H copy are in flight when
+    this is returned; result() synchronizes and runs the CPU bookkeeping
+    (role of the reference's AsyncGPUModelRunnerOutput,
+    gpu_model_runner.py:286)."""
+
+    def __init__(self, finish):
+        self._finish = finish
+        self._result = None
+        self._done = False
+
+    def result(self) -> "ModelRunnerOutput":
+        if not self._done:
+            self._result = self._finish()
+            self._done = True
+        return self._result
+
+
 class DecodeGraphRunner:
     """hipGraph-captured pure-decode steps (role of the reference's
     CUDAGraph dispatch, gpu_model_runner.py:4025 — redesigned: persistent
@@ -136,11 +154,15 @@ class DecodeGraphRunner:
         return self.graphs[(nb, parts)]
 
     def run(self, n: int, nb: int, parts: int, input_ids, positions,
-            slot_mapping, seq_lens, block_table) -> torch.Tensor:
-        """Stage numpy inputs (n real rows, nb bucket) and replay. Returns
-        logits for the n real rows."""
+            slot_mapping, seq_lens, block_table,
+            ids_dev: Optional[torch.Tensor] = None) -> torch.Tensor:
+        """Stage inputs (n real rows, nb bucket) and replay. Returns
+        logits for the n real rows. ids_dev, when given, is a device
+        tensor of decode input ids (previous step's sampled tokens — no
+        host round-trip)."""
         pi = self.pin_i64
-        pi[0, :n] = torch.from_numpy(input_ids)
+        if ids_dev is None:
+            pi[0, :n] = torch.from_numpy(input_ids)
         pi[1, :n] = torch.from_numpy(positions)
         pi[2, :n] = torch.from_numpy(slot_mapping)
         pi[2, n:nb] = -1
@@ -148,7 +170,10 @@ class DecodeGraphRunner:
         self.pin_seq[n:nb] = 0
         w = block_table.shape[1]
         self.pin_bt[:n, :w] = torch.from_numpy(block_table)
-        self.input_ids[:nb].copy_(pi[0, :nb], non_blocking=True)
+        if ids_dev is None:
+            self.input_ids[:nb].copy_(pi[0, :nb], non_blocking=True)
+        else:
+            self.input_ids[:n].copy_(ids_dev)
         self.positions[:nb].copy_(pi[1, :nb], non_blocking=True)
         self.slot_mapping[:nb].copy_(pi[2, :nb], non_blocking=True)
         self.seq_lens[:nb].copy_(self.pin_seq[:nb], non_blocking=True)
@@ -191,6 +216,19 @@ class ModelRunner:
         # Sampling-metadata cache for steady-state decode batches.
         self._samp_cache_key = None
         self._samp_cache_val = None
+        # Async sampled-token plumbing: last step's sampled tokens stay on
+        # the GPU and feed the next decode step's input_ids directly; the
+        # CPU copy lands in a double-buffered pinned staging area.
+        self._last_sampled: Optional[tuple[torch.Tensor, np.ndarray]] = None
+        self._row_pos = np.full(n, -1, dtype=np.int64)
+        if torch.cuda.is_available():
+            self._pin_sampled = [
+                torch.empty(n, dtype=torch.int64, pin_memory=True)
+                for _ in range(2)
+            ]
+        else:
+            self._pin_sampled = None
+        self._pin_idx = 0
 
     def load_model(self) -> None:
         self.model = load_model(self.config.model_config, self.device)
@@ -288,6 +326,22 @@ class ModelRunner:
 
     # ------------------------------------------------------------------
     @torch.inference_mode()
+    def execute_model_async(self, so: SchedulerOutput) -> "AsyncModelOutput":
+        """Launch one step without waiting for the sampled tokens (pure
+        decode); falls back to synchronous execution otherwise."""
+        self._update_states(so)
+        if so.total_num_scheduled_tokens == 0:
+            out = ModelRunnerOutput(req_ids=[], sampled_token_ids=[])
+            return AsyncModelOutput(lambda: out)
+        items = so.num_scheduled_tokens
+        if (so.total_num_scheduled_tokens == len(items)
+                and all(v == 1 for v in items.values())):
+            return self._execute_decode(so, list(items.keys()))
+        out = self._execute_inner(so)
+        return AsyncModelOutput(lambda: out)
+
+    # ------------------------------------------------------------------
+    @torch.inference_mode()
     def _execute_decode(self, so: SchedulerOutput,
                         req_ids: list[str]) -> ModelRunnerOutput:
         """Pure-decode step: vectorized input prep from the persistent row
@@ -297,6 +351,25 @@ class ModelRunner:
                            count=n)
         positions = self.np_computed[rows]
         input_ids = self.np_last_tok[rows]
+        # Decode inputs come straight from the previous step's sampled
+        # tokens ON DEVICE when available (async scheduling: the CPU copy
+        # may not have landed yet).
+        ids_dev: Optional[torch.Tensor] = None
+        if self._last_sampled is not None:
+            last_t, last_rows = self._last_sampled
+            self._row_pos[last_rows] = np.arange(len(last_rows))
+            cur = self._row_pos[rows]
+            self._row_pos[last_rows] = -1
+            hit = cur >= 0
+            if hit.all():
+                idx = torch.from_numpy(cur).to(last_t.device)
+                ids_dev = last_t.index_select(0, idx)
+            elif hit.any():
+                ids_dev = torch.from_numpy(input_ids).to(last_t.device)
+                sel = torch.from_numpy(cur[hit]).to(last_t.device)
+                dst = torch.from_numpy(np.nonzero(hit)[0]).to(last_t.device)
+                ids_dev = ids_dev.index_put(
+                    (dst,), last_t.index_select(0, sel))
         blk = positions // self.block_size
         slot_mapping = (
             self.np_block_table[rows, blk].astype(np.int64) * self.block_size
@@ -314,7 +387,7 @@ class ModelRunner:
             parts = self.graph_runner.parts_bucket(max_seq_len)
             logits = self.graph_runner.run(
                 n, nb, parts, input_ids, positions, slot_mapping, seq_lens,
-                block_table,
+                block_table, ids_dev=ids_dev,
             )
         else:
             meta = AttentionMetadata(
@@ -330,12 +403,11 @@ class ModelRunner:
                 max_seq_len=max_seq_len,
                 num_decodes=n,
             )
+            ids_t = (ids_dev if ids_dev is not None
+                     else torch.from_numpy(input_ids).to(dev))
             ctx = ForwardContext(attn_metadata=meta, kv_caches=self.kv_caches)
             with set_forward_context(ctx):
-                hidden = self.model(
-                    torch.from_numpy(input_ids).to(dev),
-                    torch.from_numpy(positions).to(dev),
-                )
+                hidden = self.model(ids_t, torch.from_numpy(positions).to(dev))
             logits = self.model.compute_logits(hidden)
 
         self.np_computed[rows] += 1
@@ -345,15 +417,39 @@ class ModelRunner:
 
         s_meta = self._sampling_meta(req_ids, states, dev)
         s_out = self.sampler(logits, s_meta)
-        sampled_np = s_out.sampled_token_ids.cpu().numpy()
+        sampled_t = s_out.sampled_token_ids
+        self._last_sampled = (sampled_t, rows)
+
+        if self.device.type == "cuda" and s_out.logprobs is None:
+            pin = self._pin_sampled[self._pin_idx]
+            self._pin_idx ^= 1
+            pin[:n].copy_(sampled_t, non_blocking=True)
+            ev = torch.cuda.Event()
+            ev.record()
+
+            def finish() -> ModelRunnerOutput:
+                ev.synchronize()
+                sampled_np = pin[:n].numpy().copy()
+                return self._finish_decode(req_ids, states, rows, sampled_np,
+                                           None)
+
+            return AsyncModelOutput(finish)
+
+        sampled_np = sampled_t.cpu().numpy()
+        out = self._finish_decode(req_ids, states, rows, sampled_np,
+                                  s_out.logprobs)
+        return AsyncModelOutput(lambda: out)
+
+    def _finish_decode(self, req_ids, states, rows, sampled_np,
+                       logprobs) -> ModelRunnerOutput:
         self.np_last_tok[rows] = sampled_np
         sampled = sampled_np.tolist()
         sampled_per_req = [[int(t)] for t in sampled]
         logprobs_per_req: dict[str, list[dict[int, float]]] = {}
         for j, st in enumerate(states):
             st.token_ids.append(int(sampled[j]))
-            if s_out.logprobs is not None and s_out.logprobs[j] is not None:
-                logprobs_per_req[req_ids[j]] = [s_out.logprobs[j]]
+            if logprobs is not None and logprobs[j] is not None:
+                logprobs_per_req[req_ids[j]] = [logprobs[j]]
         return ModelRunnerOutput(
             req_ids=req_ids,
             sampled_token_ids=sampled_per_req,
@@ -393,6 +489,10 @@ class ModelRunner:
     @torch.inference_mode()
     def execute_model(self, so: SchedulerOutput) -> ModelRunnerOutput:
         self._update_states(so)
+        return self._execute_inner(so)
+
+    @torch.inference_mode()
+    def _execute_inner(self, so: SchedulerOutput) -> ModelRunnerOutput:
         if so.total_num_scheduled_tokens == 0:
             return ModelRunnerOutput(req_ids=[], sampled_token_ids=[])
 
@@ -405,7 +505,7 @@ class ModelRunner:
 
         total = so.total_num_scheduled_tokens
         if num_decodes == len(items) and total == len(items):
-            return self._execute_decode(so, req_ids)
+            return self._execute_decode(so, req_ids).result()
         input_ids = np.empty(total, dtype=np.int64)
         positions = np.empty(total, dtype=np.int64)
         slot_mapping = np.empty(total, dtype=np.int64)
@@ -524,6 +624,10 @@ class ModelRunner:
             # Runner keeps its own copy of generated tokens.
             self.requests[req_ids[r]].token_ids.append(tok)
             self.np_last_tok[self._row_of[req_ids[r]]] = tok
+        # Mixed steps resolve on the CPU; invalidate the device-side
+        # sampled-token carry so the next decode reads np_last_tok.
+        self._last_sampled = None
+        for j, r in enumerate(sampling_rows):
             if s_out.logprobs is not None and s_out.logprobs[j] is not None:
                 logprobs_per_req[req_ids[r]] = [s_out.logprobs[j]]
         return ModelRunnerOutput(

@@ -84,3 +84,8 @@ class Worker:
 
     def execute_model(self, so: SchedulerOutput) -> ModelRunnerOutput:
         return self.runner.execute_model(so)
+
+    def execute_model_async(self, so: SchedulerOutput):
+        """Launch a step; returns an AsyncModelOutput (result() waits for
+        the sampled-token copy)."""
+        return self.runner.execute_model_async(so)
