@@ -1,0 +1,91 @@
+"""Speculative decoding (ngram prompt-lookup) tests: spec-on greedy
+output must equal spec-off output token-for-token, and drafts must be
+accepted on repetitive text."""
+
+import pytest
+
+from vllm_amd.entrypoints.llm import LLM
+from vllm_amd.sampling_params import SamplingParams
+from vllm_amd.spec_decode.ngram import NgramProposer
+
+
+def test_ngram_proposer_basic():
+    p = NgramProposer(min_n=2, max_n=3, k=3)
+    # "5 6 7 8" repeats: suffix [7, 8] seen before, followed by 9, 5, 6.
+    toks = [5, 6, 7, 8, 9, 5, 6, 7, 8]
+    assert p.propose(toks) == [9, 5, 6]
+    assert p.propose([1, 2, 3]) is None
+
+
+def _generate(spec_tokens: int):
+    llm = LLM(model="tiny-llama", dtype="fp32", device="cpu",
+              block_size=16, num_gpu_blocks=256, max_model_len=512,
+              max_num_batched_tokens=512, max_num_seqs=4,
+              num_speculative_tokens=spec_tokens)
+    # Strongly repetitive prompt so n-gram lookup fires.
+    prompt = [7, 8, 9, 10] * 12
+    outs = llm.generate(
+        [prompt, list(range(30, 60))],
+        SamplingParams(temperature=0.0, max_tokens=24, ignore_eos=True),
+    )
+    stats = None
+    sched = llm.engine.engine_core.scheduler
+    stats = (sched.spec_stats_drafted, sched.spec_stats_accepted)
+    llm.shutdown()
+    return [o.outputs[0].token_ids for o in outs], stats
+
+
+def test_spec_decode_matches_baseline():
+    base, _ = _generate(0)
+    spec, _ = _generate(4)
+    assert base == spec
+    assert all(len(t) == 24 for t in spec)
+
+
+def test_scheduler_spec_rollback():
+    """Scheduler-level: draft scheduling, acceptance accounting and KV
+    rollback of rejected positions (reference update_from_output:1679)."""
+    from tests.test_scheduler import create_scheduler, make_request
+    from vllm_amd.core.sched_output import ModelRunnerOutput
+
+    sched = create_scheduler()
+    sched.spec_proposer = object()  # enable the propose path marker
+    req = make_request("r1", num_tokens=16, max_tokens=32)
+    req.sampling_params.ignore_eos = True
+    sched.add_request(req)
+
+    # Prefill + first token.
+    out = sched.schedule()
+    sched.update_from_output(out, ModelRunnerOutput(
+        req_ids=["r1"], sampled_token_ids=[[7]]))
+    # Manually attach drafts (as the ngram proposer would).
+    req.spec_token_ids = [100, 101, 102]
+    out = sched.schedule()
+    assert out.num_scheduled_tokens["r1"] == 4  # 1 real + 3 drafts
+    assert out.scheduled_spec_decode_tokens["r1"] == [100, 101, 102]
+    computed_after_sched = req.num_computed_tokens
+    # Runner accepts draft 100 but rejects 101: returns 3 tokens
+    # (s0=100 accepted? semantics: runner returns [s0, s1] when first
+    # draft matched, second did not).
+    sched.update_from_output(out, ModelRunnerOutput(
+        req_ids=["r1"], sampled_token_ids=[[100, 55]]))
+    # 4 scheduled, 2 kept -> 2 rejected positions rolled back.
+    assert req.num_computed_tokens == computed_after_sched - 2
+    assert req.all_token_ids[-2:] == [100, 55]
+    assert sched.spec_stats_drafted == 3
+    assert sched.spec_stats_accepted == 1
+
+
+def test_spec_decode_accepts_on_repetition():
+    llm = LLM(model="tiny-llama", dtype="fp32", device="cpu",
+              block_size=16, num_gpu_blocks=256, max_model_len=512,
+              max_num_batched_tokens=512, max_num_seqs=4,
+              num_speculative_tokens=4)
+    # Single-token loop prompt: the model likely repeats; even if not,
+    # the engine must stay correct (count check only).
+    outs = llm.generate(
+        [[3, 4] * 20],
+        SamplingParams(temperature=0.0, max_tokens=16, ignore_eos=True),
+    )
+    llm.shutdown()
+    assert len(outs[0].outputs[0].token_ids) == 16

@@ -374,6 +374,15 @@ class SchedulerConfig:
     # Pipeline CPU scheduling with GPU execution (one step in flight;
     # role of the reference's AsyncScheduler + async model-runner output).
     async_scheduling: bool = True
+    # Speculative decoding: ngram prompt-lookup drafts (0 = off). Spec
+    # decode forces synchronous scheduling.
+    num_speculative_tokens: int = 0
+    ngram_prompt_lookup_min: int = 2
+    ngram_prompt_lookup_max: int = 4
+
+    def __post_init__(self) -> None:
+        if self.num_speculative_tokens > 0:
+            self.async_scheduling = False
     long_prefill_token_threshold: int = 0  # 0 -> no cap beyond token budget
     policy: str = "fcfs"  # fcfs | priority
 

@@ -47,6 +47,10 @@ class SchedulerOutput:
     num_scheduled_tokens: dict[str, int]
     total_num_scheduled_tokens: int
     finished_req_ids: set[str]
+    # req_id -> draft tokens scheduled for verification this step
+    # (spec decode; role of scheduled_spec_decode_tokens in the reference).
+    scheduled_spec_decode_tokens: dict[str, list[int]] = field(
+        default_factory=dict)
 
     @property
     def num_reqs(self) -> int:

@@ -49,6 +49,19 @@ class Scheduler:
         self.max_model_len = config.model_config.max_model_len
         self.block_size = config.cache_config.block_size
 
+        sc = config.scheduler_config
+        if sc.num_speculative_tokens > 0:
+            from vllm_amd.spec_decode.ngram import NgramProposer
+
+            self.spec_proposer = NgramProposer(
+                min_n=sc.ngram_prompt_lookup_min,
+                max_n=sc.ngram_prompt_lookup_max,
+                k=sc.num_speculative_tokens,
+            )
+        else:
+            self.spec_proposer = None
+        self.spec_stats_drafted = 0
+        self.spec_stats_accepted = 0
         self.kv_cache_manager = KVCacheManager(
             num_gpu_blocks=num_gpu_blocks,
             block_size=config.cache_config.block_size,
@@ -109,6 +122,7 @@ class Scheduler:
 
     def schedule(self) -> SchedulerOutput:
         token_budget = self.max_num_batched_tokens
+        scheduled_spec_tokens: dict[str, list[int]] = {}
         scheduled_new_reqs: list[NewRequestData] = []
         cached = CachedRequestData()
         num_scheduled_tokens: dict[str, int] = {}
@@ -118,7 +132,8 @@ class Scheduler:
         req_index = 0
         while req_index < len(self.running) and token_budget > 0:
             request = self.running[req_index]
-            num_new_tokens = (request.num_tokens
+            num_spec = len(request.spec_token_ids)
+            num_new_tokens = (request.num_tokens + num_spec
                               + request.num_output_placeholders
                               - request.num_computed_tokens)
             if self.long_prefill_token_threshold > 0:
@@ -171,6 +186,10 @@ class Scheduler:
                     + num_new_tokens
                 ]
             )
+            if num_spec > 0:
+                scheduled_spec_tokens[request.request_id] = \
+                    list(request.spec_token_ids)
+                request.spec_token_ids = []
             num_scheduled_tokens[request.request_id] = num_new_tokens
             token_budget -= num_new_tokens
             # Advance optimistically at schedule time (async scheduling:
@@ -265,6 +284,7 @@ class Scheduler:
             num_scheduled_tokens=num_scheduled_tokens,
             total_num_scheduled_tokens=total,
             finished_req_ids=self.finished_req_ids,
+            scheduled_spec_decode_tokens=scheduled_spec_tokens,
         )
         self.finished_req_ids = set()
         return out
@@ -275,6 +295,7 @@ class Scheduler:
         request.status = RequestStatus.PREEMPTED
         request.num_computed_tokens = 0
         request.num_output_placeholders = 0
+        request.spec_token_ids = []
         request.num_preemptions += 1
         self.waiting.appendleft(request)
 
@@ -298,6 +319,18 @@ class Scheduler:
                 continue  # aborted mid-step
 
             new_token_ids = sampled_by_req.get(req_id) or []
+            num_spec_sched = len(
+                scheduler_output.scheduled_spec_decode_tokens.get(req_id, ())
+            )
+            if num_spec_sched > 0:
+                # Roll back KV positions of rejected draft tokens: they
+                # were computed from draft values that turned out wrong
+                # (reference scheduler.py:1679 spec-token accounting).
+                num_rejected = num_spec_sched + 1 - len(new_token_ids)
+                if num_rejected > 0:
+                    request.num_computed_tokens -= num_rejected
+                self.spec_stats_drafted += num_spec_sched
+                self.spec_stats_accepted += len(new_token_ids) - 1
             if not new_token_ids:
                 continue  # mid chunked-prefill, nothing sampled
 
@@ -324,6 +357,10 @@ class Scheduler:
             if stopped:
                 self.running.remove(request)
                 self._free_request(request)
+            elif (self.spec_proposer is not None
+                    and request.sampling_params.temperature == 0.0):
+                drafts = self.spec_proposer.propose(request.all_token_ids)
+                request.spec_token_ids = drafts or []
         return outputs
 
     def _check_stop(self, request: Request, last_token: int) -> bool:

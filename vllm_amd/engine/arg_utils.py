@@ -38,6 +38,11 @@ class EngineArgs:
     enable_chunked_prefill: bool = True
     scheduling_policy: str = "fcfs"
 
+    num_speculative_tokens: int = 0
+    ngram_prompt_lookup_min: int = 2
+    ngram_prompt_lookup_max: int = 4
+    async_scheduling: bool = True
+
     tensor_parallel_size: int = 1
     device: str = "auto"
 
@@ -68,6 +73,11 @@ class EngineArgs:
                             action="store_false")
         parser.add_argument("--scheduling-policy", type=str, default="fcfs",
                             choices=["fcfs", "priority"])
+        parser.add_argument("--num-speculative-tokens", type=int, default=0)
+        parser.add_argument("--ngram-prompt-lookup-min", type=int, default=2)
+        parser.add_argument("--ngram-prompt-lookup-max", type=int, default=4)
+        parser.add_argument("--no-async-scheduling",
+                            dest="async_scheduling", action="store_false")
         parser.add_argument("--tensor-parallel-size", "-tp", type=int,
                             default=1)
         parser.add_argument("--device", type=str, default="auto")
@@ -114,6 +124,10 @@ class EngineArgs:
                 max_num_seqs=self.max_num_seqs,
                 enable_chunked_prefill=self.enable_chunked_prefill,
                 policy=self.scheduling_policy,
+                async_scheduling=self.async_scheduling,
+                num_speculative_tokens=self.num_speculative_tokens,
+                ngram_prompt_lookup_min=self.ngram_prompt_lookup_min,
+                ngram_prompt_lookup_max=self.ngram_prompt_lookup_max,
             ),
             parallel_config=pc,
             device_config=DeviceConfig(device=self.device),

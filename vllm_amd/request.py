@@ -66,6 +66,8 @@ class Request:
         # yet known (role of AsyncScheduler's num_output_placeholders,
         # vllm/v1/core/sched/async_scheduler.py:12).
         self.num_output_placeholders = 0
+        # Draft tokens proposed for the next step (spec decode).
+        self.spec_token_ids: list[int] = []
         # All token ids: prompt + generated. Kept as one list so attention
         # metadata / block hashing index into a single sequence.
         self._all_token_ids: list[int] = list(prompt_token_ids)

@@ -516,12 +516,17 @@ class ModelRunner:
         )
         block_table = np.zeros((len(items), max_blocks), dtype=np.int32)
 
+        spec_map = so.scheduled_spec_decode_tokens or {}
         sampling_rows: list[int] = []  # row in `items` order
+        sampling_npos: list[int] = []  # positions sampled per row (spec)
         t = 0
         for i, (rid, n) in enumerate(items):
             state = self.requests[rid]
             start = state.num_computed_tokens
-            toks = state.token_ids[start : start + n]
+            if rid in spec_map:
+                toks = (state.token_ids + spec_map[rid])[start : start + n]
+            else:
+                toks = state.token_ids[start : start + n]
             input_ids[t : t + n] = toks
             positions[t : t + n] = np.arange(start, start + n)
             # slot = block_ids[pos // bs] * bs + pos % bs
@@ -534,8 +539,12 @@ class ModelRunner:
             query_start_loc[i + 1] = query_start_loc[i] + n
             seq_lens[i] = start + n
             block_table[i, : len(state.block_ids)] = state.block_ids
-            if start + n >= len(state.token_ids):
+            if rid in spec_map:
                 sampling_rows.append(i)
+                sampling_npos.append(len(spec_map[rid]) + 1)
+            elif start + n >= len(state.token_ids):
+                sampling_rows.append(i)
+                sampling_npos.append(1)
             t += n
 
         dev = self.device
@@ -587,25 +596,25 @@ class ModelRunner:
             # Decode: every row samples, rows already in order.
             logits = logits_all
         else:
-            # Gather last-token hidden states of sampling requests.
-            last_idx = torch.tensor(
-                [int(query_start_loc[r + 1]) - 1 for r in sampling_rows],
-                device=dev,
-            )
+            # Gather sampling positions: the last `npos` tokens of each
+            # sampling request (npos > 1 verifies draft tokens in place).
+            idx = []
+            for r, npos in zip(sampling_rows, sampling_npos):
+                end = int(query_start_loc[r + 1])
+                idx.extend(range(end - npos, end))
+            last_idx = torch.tensor(idx, device=dev)
             logits = self.model.compute_logits(hidden[last_idx])
 
-        s_params = [
-            self.requests[req_ids[r]].sampling_params for r in sampling_rows
-        ]
-        s_prompts = [
-            self.requests[req_ids[r]].token_ids[
-                : self.requests[req_ids[r]].prompt_len
-            ]
-            for r in sampling_rows
-        ]
-        s_outputs = [
-            self.requests[req_ids[r]].output_token_ids for r in sampling_rows
-        ]
+        def rep(make):
+            vals = []
+            for r, npos in zip(sampling_rows, sampling_npos):
+                v = make(self.requests[req_ids[r]])
+                vals.extend([v] * npos)
+            return vals
+
+        s_params = rep(lambda st: st.sampling_params)
+        s_prompts = rep(lambda st: st.token_ids[: st.prompt_len])
+        s_outputs = rep(lambda st: st.output_token_ids)
         s_meta = SamplingMetadata.build(
             s_params,
             s_prompts,
@@ -618,18 +627,44 @@ class ModelRunner:
 
         sampled_per_req: list[list[int]] = [[] for _ in req_ids]
         logprobs_per_req: dict[str, list[dict[int, float]]] = {}
-        for j, r in enumerate(sampling_rows):
-            tok = int(sampled[j])
-            sampled_per_req[r] = [tok]
-            # Runner keeps its own copy of generated tokens.
-            self.requests[req_ids[r]].token_ids.append(tok)
-            self.np_last_tok[self._row_of[req_ids[r]]] = tok
+        flat = 0
+        for j, (r, npos) in enumerate(zip(sampling_rows, sampling_npos)):
+            rid = req_ids[r]
+            state = self.requests[rid]
+            row_sampled = sampled[flat: flat + npos]
+            flat += npos
+            if npos == 1:
+                accepted = [int(row_sampled[0])]
+            else:
+                # Greedy draft verification: accept draft j while the
+                # model's token at the previous position equals it.
+                drafts = spec_map[rid]
+                accepted = [int(row_sampled[0])]
+                for d_j in range(len(drafts)):
+                    if int(row_sampled[d_j]) != drafts[d_j]:
+                        break
+                    accepted.append(int(row_sampled[d_j + 1]))
+            sampled_per_req[r] = accepted
+            state.token_ids.extend(accepted)
+            self.np_last_tok[self._row_of[rid]] = accepted[-1]
+            # Runner-side rollback of rejected draft positions (the
+            # scheduler does the same with its own counters).
+            if npos > 1:
+                rejected = npos - len(accepted)
+                state.num_computed_tokens -= rejected
+                self.np_computed[self._row_of[rid]] -= rejected
         # Mixed steps resolve on the CPU; invalidate the device-side
         # sampled-token carry so the next decode reads np_last_tok.
         self._last_sampled = None
-        for j, r in enumerate(sampling_rows):
-            if s_out.logprobs is not None and s_out.logprobs[j] is not None:
-                logprobs_per_req[req_ids[r]] = [s_out.logprobs[j]]
+        if s_out.logprobs is not None:
+            flat = 0
+            for r, npos in zip(sampling_rows, sampling_npos):
+                lps = [lp for lp in s_out.logprobs[flat: flat + npos]
+                       if lp is not None]
+                flat += npos
+                if lps:
+                    logprobs_per_req[req_ids[r]] = lps[
+                        : len(sampled_per_req[r])]
         return ModelRunnerOutput(
             req_ids=req_ids,
             sampled_token_ids=sampled_per_req,
