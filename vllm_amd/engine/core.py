@@ -30,28 +30,47 @@ logger = logging.getLogger(__name__)
 class EngineCore:
 
     def __init__(self, config: EngineConfig):
+        import os
+
         self.config = config
-        self.worker = Worker(config)
-        self.worker.init_device()
-        self.worker.load_model()
+        tp = config.parallel_config.tensor_parallel_size
+        launched_world = int(os.environ.get("WORLD_SIZE", "1"))
+        self._multiproc = tp > 1 and launched_world == 1
 
         from vllm_amd.parallel.state import get_world_group
 
-        self.world = get_world_group()
-        self.is_driver = self.world.rank_in_group == 0
+        if self._multiproc:
+            # Engine owns the scheduler; one spawned worker process per
+            # GPU (RCCL group lives in the workers).
+            from vllm_amd.executor.multiproc import MultiprocExecutor
 
-        # KV sizing must agree across ranks: take the min over ranks.
-        num_blocks = self.worker.determine_num_kv_blocks()
-        if self.world.world_size > 1:
-            t = torch.tensor([num_blocks], dtype=torch.int64)
-            if torch.cuda.is_available():
-                t = t.cuda()
-            torch.distributed.all_reduce(
-                t, op=torch.distributed.ReduceOp.MIN
-            )
-            num_blocks = int(t.item())
-        self.num_gpu_blocks = num_blocks
-        self.worker.initialize_kv_cache(num_blocks)
+            self.worker = MultiprocExecutor(config)
+            self.world = get_world_group()  # engine proc: world of 1
+            self.is_driver = True
+            num_blocks = self.worker.determine_num_kv_blocks()
+            self.num_gpu_blocks = num_blocks
+            self.worker.initialize_kv_cache(num_blocks)
+        else:
+            # In-process worker; SPMD under torchrun (one rank per GPU).
+            self.worker = Worker(config)
+            self.worker.init_device()
+            self.worker.load_model()
+
+            self.world = get_world_group()
+            self.is_driver = self.world.rank_in_group == 0
+
+            # KV sizing must agree across ranks: min over ranks.
+            num_blocks = self.worker.determine_num_kv_blocks()
+            if self.world.world_size > 1:
+                t = torch.tensor([num_blocks], dtype=torch.int64)
+                if torch.cuda.is_available():
+                    t = t.cuda()
+                torch.distributed.all_reduce(
+                    t, op=torch.distributed.ReduceOp.MIN
+                )
+                num_blocks = int(t.item())
+            self.num_gpu_blocks = num_blocks
+            self.worker.initialize_kv_cache(num_blocks)
 
         self.scheduler: Optional[Scheduler] = (
             Scheduler(config, num_gpu_blocks=num_blocks)
@@ -135,5 +154,7 @@ class EngineCore:
                 self._drain()
             except Exception:  # noqa: BLE001
                 pass
+        if self._multiproc:
+            self.worker.shutdown()
         if self.is_driver and self.world.world_size > 1:
             self.world.broadcast_object(None, src=0)

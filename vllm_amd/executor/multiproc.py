@@ -1,0 +1,185 @@
+"""Multi-process executor: one worker process per GPU, spawned by the
+engine (role of the reference's MultiprocExecutor,
+vllm/v1/executor/multiproc_executor.py:108 — pipes instead of the shm
+MessageQueue; the data plane between workers is still RCCL over xGMI).
+
+The engine process owns the scheduler; worker rank 0 returns sampled
+tokens. Workers execute steps asynchronously (a sender thread ships each
+result when its GPU work completes), so the engine can schedule step N+1
+while the workers run step N — the same one-step pipeline as the
+in-process path.
+"""
+
+from __future__ import annotations
+
+import logging
+import os
+import threading
+from typing import Any, Optional
+
+import torch.multiprocessing as mp
+
+from vllm_amd.config import EngineConfig
+
+logger = logging.getLogger(__name__)
+
+
+def _worker_main(rank: int, config: EngineConfig, conn, master_port: int):
+    os.environ["RANK"] = str(rank)
+    os.environ["LOCAL_RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(config.parallel_config.tensor_parallel_size)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(master_port)
+    config.parallel_config.rank = rank
+    config.parallel_config.local_rank = rank
+    config.parallel_config.world_size = \
+        config.parallel_config.tensor_parallel_size
+
+    from vllm_amd.worker.worker import Worker
+
+    worker = Worker(config)
+    worker.init_device()
+    worker.load_model()
+
+    send_lock = threading.Lock()
+
+    def send(obj):
+        with send_lock:
+            conn.send(obj)
+
+    pending = []  # queue of AsyncModelOutput futures (rank 0 only)
+    pending_cv = threading.Condition()
+
+    def sender_loop():
+        while True:
+            with pending_cv:
+                while not pending:
+                    pending_cv.wait()
+                fut = pending.pop(0)
+            if fut is None:
+                return
+            try:
+                send(("ok", fut.result()))
+            except Exception as e:  # noqa: BLE001
+                send(("err", repr(e)))
+
+    sender = None
+    if rank == 0:
+        sender = threading.Thread(target=sender_loop, daemon=True)
+        sender.start()
+
+    try:
+        while True:
+            msg = conn.recv()
+            kind = msg[0]
+            if kind == "rpc":
+                _, method, args, kwargs = msg
+                try:
+                    result = getattr(worker, method)(*args, **kwargs)
+                    send(("ok", result))
+                except Exception as e:  # noqa: BLE001
+                    logger.exception("worker rpc %s failed", method)
+                    send(("err", repr(e)))
+            elif kind == "execute":
+                so = msg[1]
+                fut = worker.execute_model_async(so)
+                if rank == 0:
+                    with pending_cv:
+                        pending.append(fut)
+                        pending_cv.notify()
+                else:
+                    fut.result()
+            elif kind == "execute_sync":
+                so = msg[1]
+                out = worker.execute_model(so)
+                if rank == 0:
+                    send(("ok", out))
+            elif kind == "shutdown":
+                if sender is not None:
+                    with pending_cv:
+                        pending.append(None)
+                        pending_cv.notify()
+                    sender.join(timeout=5)
+                return
+    except (EOFError, KeyboardInterrupt):
+        pass
+
+
+class _Future:
+    def __init__(self, conn):
+        self._conn = conn
+        self._val = None
+        self._done = False
+
+    def result(self):
+        if not self._done:
+            status, val = self._conn.recv()
+            if status == "err":
+                raise RuntimeError(f"worker error: {val}")
+            self._val = val
+            self._done = True
+        return self._val
+
+
+class MultiprocExecutor:
+    """Engine-side handle to the worker processes."""
+
+    def __init__(self, config: EngineConfig):
+        self.config = config
+        tp = config.parallel_config.tensor_parallel_size
+        ctx = mp.get_context("spawn")
+        port = int(os.environ.get("VLLM_AMD_WORKER_PORT", "29533"))
+        self.conns = []
+        self.procs = []
+        for rank in range(tp):
+            parent, child = ctx.Pipe()
+            p = ctx.Process(
+                target=_worker_main, args=(rank, config, child, port),
+                daemon=True,
+            )
+            p.start()
+            self.conns.append(parent)
+            self.procs.append(p)
+
+    # ---- control-plane RPC -------------------------------------------
+    def collective_rpc(self, method: str, *args: Any, **kwargs: Any) -> list:
+        for conn in self.conns:
+            conn.send(("rpc", method, args, kwargs))
+        results = []
+        for conn in self.conns:
+            status, val = conn.recv()
+            if status == "err":
+                raise RuntimeError(f"worker rpc {method} failed: {val}")
+            results.append(val)
+        return results
+
+    def determine_num_kv_blocks(self) -> int:
+        return min(self.collective_rpc("determine_num_kv_blocks"))
+
+    def initialize_kv_cache(self, num_blocks: int) -> None:
+        self.collective_rpc("initialize_kv_cache", num_blocks)
+
+    # ---- data plane ---------------------------------------------------
+    def execute_model(self, so):
+        for conn in self.conns:
+            conn.send(("execute_sync", so))
+        status, val = self.conns[0].recv()
+        if status == "err":
+            raise RuntimeError(f"worker step failed: {val}")
+        return val
+
+    def execute_model_async(self, so) -> _Future:
+        for conn in self.conns:
+            conn.send(("execute", so))
+        return _Future(self.conns[0])
+
+    def shutdown(self) -> None:
+        for conn in self.conns:
+            try:
+                conn.send(("shutdown",))
+            except (BrokenPipeError, OSError):
+                pass
+        for p in self.procs:
+            p.join(timeout=10)
+            if p.is_alive():
+                p.terminate()
