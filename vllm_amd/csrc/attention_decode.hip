@@ -119,41 +119,67 @@ __global__ __launch_bounds__(DEC_BLOCK) void paged_decode_kernel(
     for (int j = 0; j < 8; ++j) acc[g][j] = 0.f;
   }
 
-  for (int t = t_begin + wave * 4 + slot; t < t_end; t += 16) {
-    const int local = t - part * DEC_PART;
-    const int phys = blk_ids[local / 64];
-    const short* base = kv_cache +
-        ((int64_t)phys * num_kv_heads + kvh) * head_tile_stride +
+  // Two tokens per (wave, slot) iteration: doubles the K/V loads in
+  // flight per lane and halves the online-softmax update chain.
+  const int t0_base = t_begin + wave * 4 + slot;
+  for (int t = t0_base; t < t_end; t += 32) {
+    const int t1 = t + 16;
+    const bool has1 = t1 < t_end;
+    const int local0 = t - part * DEC_PART;
+    const short* base0 = kv_cache +
+        ((int64_t)blk_ids[local0 / 64] * num_kv_heads + kvh) *
+            head_tile_stride +
         (int64_t)(t % 64) * head_dim + lane16 * 8;
-    s16x8 kv = *reinterpret_cast<const s16x8*>(base);
-    s16x8 vv = *reinterpret_cast<const s16x8*>(base + kv_plane_stride);
-    float kf[8], vf[8];
+    const int local1 = has1 ? t1 - part * DEC_PART : local0;
+    const short* base1 = has1 ? kv_cache +
+        ((int64_t)blk_ids[local1 / 64] * num_kv_heads + kvh) *
+            head_tile_stride +
+        (int64_t)(t1 % 64) * head_dim + lane16 * 8 : base0;
+    s16x8 k0 = *reinterpret_cast<const s16x8*>(base0);
+    s16x8 v0 = *reinterpret_cast<const s16x8*>(base0 + kv_plane_stride);
+    s16x8 k1 = *reinterpret_cast<const s16x8*>(base1);
+    s16x8 v1 = *reinterpret_cast<const s16x8*>(base1 + kv_plane_stride);
+    float kf0[8], vf0[8], kf1[8], vf1[8];
 #pragma unroll
-    for (int j = 0; j < 8; ++j) kf[j] = to_f32<Tag>(kv[j]);
-#pragma unroll
-    for (int j = 0; j < 8; ++j) vf[j] = to_f32<Tag>(vv[j]);
+    for (int j = 0; j < 8; ++j) {
+      kf0[j] = to_f32<Tag>(k0[j]);
+      vf0[j] = to_f32<Tag>(v0[j]);
+      kf1[j] = to_f32<Tag>(k1[j]);
+      vf1[j] = to_f32<Tag>(v1[j]);
+    }
 #pragma unroll
     for (int g = 0; g < GROUP; ++g) {
-      float sc = 0.f;
+      float sc0 = 0.f, sc1 = 0.f;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) sc += qf[g][j] * kf[j];
+      for (int j = 0; j < 8; ++j) {
+        sc0 += qf[g][j] * kf0[j];
+        sc1 += qf[g][j] * kf1[j];
+      }
       // Reduce over the 16 dim lanes of this slot (consecutive lanes).
-      sc += __shfl_xor(sc, 1, 64);
-      sc += __shfl_xor(sc, 2, 64);
-      sc += __shfl_xor(sc, 4, 64);
-      sc += __shfl_xor(sc, 8, 64);
-      const float m_new = fmaxf(m_s[g], sc);
-      const float p = __expf(sc - m_new);
+      sc0 += __shfl_xor(sc0, 1, 64);
+      sc1 += __shfl_xor(sc1, 1, 64);
+      sc0 += __shfl_xor(sc0, 2, 64);
+      sc1 += __shfl_xor(sc1, 2, 64);
+      sc0 += __shfl_xor(sc0, 4, 64);
+      sc1 += __shfl_xor(sc1, 4, 64);
+      sc0 += __shfl_xor(sc0, 8, 64);
+      sc1 += __shfl_xor(sc1, 8, 64);
+      if (!has1) sc1 = -3.0e38f;
+      const float m_new = fmaxf(m_s[g], fmaxf(sc0, sc1));
+      const float p0 = __expf(sc0 - m_new);
+      const float p1 = has1 ? __expf(sc1 - m_new) : 0.f;
       if (m_new > m_s[g]) {
         const float corr = __expf(m_s[g] - m_new);
-        l_s[g] = l_s[g] * corr + p;
+        l_s[g] = l_s[g] * corr + p0 + p1;
 #pragma unroll
-        for (int j = 0; j < 8; ++j) acc[g][j] = acc[g][j] * corr + p * vf[j];
+        for (int j = 0; j < 8; ++j)
+          acc[g][j] = acc[g][j] * corr + p0 * vf0[j] + p1 * vf1[j];
         m_s[g] = m_new;
       } else {
-        l_s[g] += p;
+        l_s[g] += p0 + p1;
 #pragma unroll
-        for (int j = 0; j < 8; ++j) acc[g][j] += p * vf[j];
+        for (int j = 0; j < 8; ++j)
+          acc[g][j] += p0 * vf0[j] + p1 * vf1[j];
       }
     }
   }
