@@ -20,6 +20,7 @@ class NewRequestData:
     block_ids: list[int]
     num_computed_tokens: int
     sampling_params: SamplingParams
+    grammar: object = None  # CompiledGrammar for structured output
 
 
 @dataclass

@@ -111,6 +111,9 @@ class Scheduler:
             request.status = status
             self._free_request(request)
 
+    def has_guided_requests(self) -> bool:
+        return any(r.grammar is not None for r in self.running)
+
     def has_unfinished_requests(self) -> bool:
         return bool(self.waiting or self.running)
 
@@ -268,6 +271,7 @@ class Scheduler:
                         block_ids=all_block_ids,
                         num_computed_tokens=num_computed,
                         sampling_params=request.sampling_params,
+                        grammar=request.grammar,
                     )
                 )
             num_scheduled_tokens[request.request_id] = num_new_tokens
@@ -358,6 +362,7 @@ class Scheduler:
                 self.running.remove(request)
                 self._free_request(request)
             elif (self.spec_proposer is not None
+                    and request.grammar is None
                     and request.sampling_params.temperature == 0.0):
                 drafts = self.spec_proposer.propose(request.all_token_ids)
                 request.spec_token_ids = drafts or []

@@ -117,7 +117,9 @@ class EngineCore:
             and scheduler_output.total_num_scheduled_tokens
             == len(scheduler_output.num_scheduled_tokens)
         )
-        if not (self.async_scheduling and pure_decode):
+        use_async = (self.async_scheduling and pure_decode
+                     and not self.scheduler.has_guided_requests())
+        if not use_async:
             # Mixed/prefill/empty steps run synchronously: the runner's
             # slow path reads token values the pending step produces.
             outputs = self._drain()

@@ -194,6 +194,12 @@ class LLMEngine:
             eos_token_id=eos,
             prompt=prompt_text,
         )
+        if params.guided_choice:
+            from vllm_amd.structured_output import compile_choice_grammar
+
+            request.grammar = compile_choice_grammar(
+                params.guided_choice, self.tokenizer, eos
+            )
         self.engine_core.add_request(request)
         self.output_processor.add_request(
             request_id, prompt_text, prompt_token_ids, params

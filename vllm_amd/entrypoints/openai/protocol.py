@@ -43,6 +43,7 @@ class CompletionRequest(BaseModel):
     stop_token_ids: Optional[list[int]] = None
     ignore_eos: bool = False
     skip_special_tokens: bool = True
+    guided_choice: Optional[list[str]] = None
 
     def to_sampling_params(self, stream: bool) -> SamplingParams:
         logit_bias = (
@@ -67,6 +68,7 @@ class CompletionRequest(BaseModel):
             logprobs=self.logprobs,
             logit_bias=logit_bias,
             skip_special_tokens=self.skip_special_tokens,
+            guided_choice=self.guided_choice,
             output_kind=(RequestOutputKind.DELTA if stream
                          else RequestOutputKind.FINAL_ONLY),
         )
@@ -114,6 +116,7 @@ class ChatCompletionRequest(BaseModel):
     stop_token_ids: Optional[list[int]] = None
     ignore_eos: bool = False
     skip_special_tokens: bool = True
+    guided_choice: Optional[list[str]] = None
     add_generation_prompt: bool = True
 
     def to_sampling_params(self, stream: bool,
@@ -143,6 +146,7 @@ class ChatCompletionRequest(BaseModel):
             logprobs=n_logprobs,
             logit_bias=logit_bias,
             skip_special_tokens=self.skip_special_tokens,
+            guided_choice=self.guided_choice,
             output_kind=(RequestOutputKind.DELTA if stream
                          else RequestOutputKind.FINAL_ONLY),
         )

@@ -42,6 +42,9 @@ class SamplingMetadata:
     logit_bias: Optional[list[Optional[dict[int, float]]]] = None
     allowed_token_ids: Optional[list[Optional[list[int]]]] = None
     min_tokens_mask: Optional[list[Optional[tuple[int, set[int]]]]] = None
+    # Structured output: per-row allowed-token sets for THIS step
+    # (attached by the runner; None rows are unconstrained).
+    grammar_masks: Optional[list[Optional[set]]] = None
 
     @classmethod
     def build(
@@ -147,6 +150,15 @@ class Sampler(torch.nn.Module):
                     mask = torch.ones(V, dtype=torch.bool,
                                       device=logits.device)
                     mask[torch.tensor(allowed, device=logits.device)] = False
+                    logits[i].masked_fill_(mask, float("-inf"))
+        if meta.grammar_masks is not None:
+            for i, allowed in enumerate(meta.grammar_masks):
+                if allowed is not None:
+                    mask = torch.ones(V, dtype=torch.bool,
+                                      device=logits.device)
+                    if allowed:
+                        mask[torch.tensor(sorted(allowed),
+                                          device=logits.device)] = False
                     logits[i].masked_fill_(mask, float("-inf"))
         if meta.min_tokens_mask is not None:
             for i, mt in enumerate(meta.min_tokens_mask):

@@ -68,6 +68,8 @@ class Request:
         self.num_output_placeholders = 0
         # Draft tokens proposed for the next step (spec decode).
         self.spec_token_ids: list[int] = []
+        # Structured output (compiled by the engine frontend).
+        self.grammar = None
         # All token ids: prompt + generated. Kept as one list so attention
         # metadata / block hashing index into a single sequence.
         self._all_token_ids: list[int] = list(prompt_token_ids)

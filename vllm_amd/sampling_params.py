@@ -55,6 +55,9 @@ class SamplingParams:
     logit_bias: Optional[dict[int, float]] = None
     allowed_token_ids: Optional[list[int]] = None
     bad_words: Optional[list[str]] = None
+    # Structured output: generation constrained to one of these strings
+    # (compiled to a token trie by the engine; see structured_output.py).
+    guided_choice: Optional[list[str]] = None
     output_kind: RequestOutputKind = RequestOutputKind.CUMULATIVE
 
     def __post_init__(self) -> None:

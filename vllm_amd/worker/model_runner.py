@@ -9,6 +9,7 @@ computed after this step.
 
 from __future__ import annotations
 
+import dataclasses
 from dataclasses import dataclass, field
 from typing import Optional
 
@@ -35,6 +36,8 @@ class CachedReqState:
     num_computed_tokens: int
     block_ids: list[int]
     sampling_params: SamplingParams
+    grammar: object = None
+    grammar_state: object = None
 
     @property
     def output_token_ids(self) -> list[int]:
@@ -296,6 +299,9 @@ class ModelRunner:
                 num_computed_tokens=nr.num_computed_tokens,
                 block_ids=list(nr.block_ids),
                 sampling_params=nr.sampling_params,
+                grammar=nr.grammar,
+                grammar_state=(nr.grammar.initial_state()
+                               if nr.grammar is not None else None),
             )
             row = self._free_rows.pop()
             self._row_of[nr.req_id] = row
@@ -416,6 +422,7 @@ class ModelRunner:
             st.num_computed_tokens += 1
 
         s_meta = self._sampling_meta(req_ids, states, dev)
+        s_meta = self._with_grammar_masks(s_meta, states)
         s_out = self.sampler(logits, s_meta)
         sampled_t = s_out.sampled_token_ids
         self._last_sampled = (sampled_t, rows)
@@ -440,6 +447,17 @@ class ModelRunner:
                                   s_out.logprobs)
         return AsyncModelOutput(lambda: out)
 
+    def _with_grammar_masks(self, meta, states):
+        if all(st.grammar is None for st in states):
+            return meta
+        meta = dataclasses.replace(meta, grammar_masks=[
+            (st.grammar.allowed_tokens(st.grammar_state)
+             if st.grammar is not None and st.grammar_state is not None
+             else None)
+            for st in states
+        ])
+        return meta
+
     def _finish_decode(self, req_ids, states, rows, sampled_np,
                        logprobs) -> ModelRunnerOutput:
         self.np_last_tok[rows] = sampled_np
@@ -447,6 +465,9 @@ class ModelRunner:
         sampled_per_req = [[int(t)] for t in sampled]
         logprobs_per_req: dict[str, list[dict[int, float]]] = {}
         for j, st in enumerate(states):
+            if st.grammar is not None and st.grammar_state is not None:
+                st.grammar_state = st.grammar.advance(
+                    st.grammar_state, int(sampled[j]))
             st.token_ids.append(int(sampled[j]))
             if logprobs is not None and logprobs[j] is not None:
                 logprobs_per_req[req_ids[j]] = [logprobs[j]]
@@ -476,6 +497,7 @@ class ModelRunner:
             and not meta.generators and meta.logit_bias is None
             and meta.allowed_token_ids is None
             and meta.min_tokens_mask is None
+            and all(st.grammar is None for st in states)
         )
         if cacheable:
             self._samp_cache_key = key
@@ -622,6 +644,8 @@ class ModelRunner:
             dev,
             seeds_offset=[len(o) for o in s_outputs],
         )
+        g_states = rep(lambda st: st)
+        s_meta = self._with_grammar_masks(s_meta, g_states)
         s_out = self.sampler(logits, s_meta)
         sampled = s_out.sampled_token_ids.tolist()
 
@@ -645,6 +669,12 @@ class ModelRunner:
                         break
                     accepted.append(int(row_sampled[d_j + 1]))
             sampled_per_req[r] = accepted
+            if state.grammar is not None:
+                for tok in accepted:
+                    if state.grammar_state is None:
+                        break
+                    state.grammar_state = state.grammar.advance(
+                        state.grammar_state, tok)
             state.token_ids.extend(accepted)
             self.np_last_tok[self._row_of[rid]] = accepted[-1]
             # Runner-side rollback of rejected draft positions (the
