@@ -120,3 +120,15 @@ def test_deepseek_mla_gpu_decode():
     outs = llm.generate(prompts, p)
     llm.shutdown()
     assert all(len(o.outputs[0].token_ids) == 6 for o in outs)
+
+
+def test_fp8_kv_cache_e2e():
+    from vllm_amd.sampling_params import SamplingParams
+
+    llm = _llm(kv_cache_dtype="fp8")
+    outs = llm.generate(
+        [list(range(1, 40)), [5, 6, 7]],
+        SamplingParams(temperature=0.0, max_tokens=12, ignore_eos=True),
+    )
+    llm.shutdown()
+    assert all(len(o.outputs[0].token_ids) == 12 for o in outs)

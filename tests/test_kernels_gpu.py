@@ -245,3 +245,53 @@ def test_fused_moe_matches_reference():
     out = hip.fused_moe(hidden, w13, w2, tw, ti)
     expect = ref.fused_moe(hidden.float(), w13.float(), w2.float(), tw, ti)
     assert_close(out, expect, atol=5e-2, rtol=5e-2)
+
+
+def test_fp8_kv_cache_attention():
+    """fp8 e4m3fn KV cache: write-path conversion + decode/prefill reads
+    must match the fp32 reference within fp8 quantization error."""
+    hip = _hip()
+    n, Hq, Hkv, D, BS = 4, 8, 2, 128, 64
+    ctx = 300
+    nb = (ctx + BS - 1) // BS
+    kfull = torch.randn(n * ctx, Hkv, D, dtype=torch.bfloat16,
+                        device="cuda")
+    vfull = torch.randn(n * ctx, Hkv, D, dtype=torch.bfloat16,
+                        device="cuda")
+    cache = torch.zeros(2, n * nb + 1, Hkv, BS, D,
+                        dtype=torch.float8_e4m3fn, device="cuda")
+    bt = torch.zeros(n, nb, dtype=torch.int32, device="cuda")
+    slots = []
+    for i in range(n):
+        bt[i] = torch.arange(1 + i * nb, 1 + (i + 1) * nb)
+        base = (1 + i * nb) * BS
+        slots.extend(range(base, base + ctx))
+    slots = torch.tensor(slots, dtype=torch.int64, device="cuda")
+    hip.reshape_and_cache(kfull, vfull, cache, slots)
+    # Reference cache written by torch conversions.
+    ref_cache = torch.zeros_like(cache)
+    ref.reshape_and_cache(kfull, vfull, ref_cache, slots)
+    assert (cache.view(torch.uint8) == ref_cache.view(torch.uint8)) \
+        .float().mean() > 0.999
+
+    q = torch.randn(n, Hq, D, dtype=torch.bfloat16, device="cuda")
+    qsl = torch.arange(n + 1, dtype=torch.int32, device="cuda")
+    sl = torch.full((n,), ctx, dtype=torch.int32, device="cuda")
+    scale = D**-0.5
+    out = hip.attention_unified(q, cache, bt, qsl, sl, scale,
+                                num_decodes=n, max_seq_len=ctx,
+                                max_query_len=1)
+    expect = ref.attention_unified(q, ref_cache, bt, qsl, sl, scale,
+                                   num_decodes=n)
+    assert_close(out, expect, atol=6e-2, rtol=6e-2)
+
+    # Prefill read path over the fp8 cache.
+    ql = 32
+    q2 = torch.randn(n * ql, Hq, D, dtype=torch.bfloat16, device="cuda")
+    qsl2 = torch.arange(0, n * ql + 1, ql, dtype=torch.int32, device="cuda")
+    out2 = hip.attention_unified(q2, cache, bt, qsl2, sl, scale,
+                                 num_decodes=0, max_seq_len=ctx,
+                                 max_query_len=ql)
+    expect2 = ref.attention_unified(q2, ref_cache, bt, qsl2, sl, scale,
+                                    num_decodes=0)
+    assert_close(out2, expect2, atol=6e-2, rtol=6e-2)

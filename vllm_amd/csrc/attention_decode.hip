@@ -39,13 +39,14 @@ constexpr int MAX_GROUP = 8;            // max GQA ratio handled in registers
 // Thread layout: 16 lanes per token (lane d covers dims 8d..8d+7 as one
 // short8 = 16 B load), so a 256-thread block streams 16 tokens per
 // iteration; DEC_PART/16 = 32 iterations.
-template <typename Tag, int GROUP, bool FINAL>
+template <typename Tag, typename CTag, int GROUP, bool FINAL>
 __global__ __launch_bounds__(DEC_BLOCK) void paged_decode_kernel(
     short* __restrict__ out,            // FINAL: [Tdec, Hq, D] (16-bit)
     float* __restrict__ tmp_out,        // else: [Tdec, Hq, parts, D]
     float* __restrict__ tmp_lse,        // else: [Tdec, Hq, parts, 2]
     const short* __restrict__ q,        // [Tdec, Hq, D]
-    const short* __restrict__ kv_cache, // [2, blocks, Hkv, 64, D]
+    const typename CacheTraits<CTag>::elem* __restrict__ kv_cache,
+                                        // [2, blocks, Hkv, 64, D]
     const int* __restrict__ block_table,   // [num_reqs, max_blocks]
     const int* __restrict__ seq_lens,      // [num_reqs]
     const float scale, const int num_q_heads, const int num_kv_heads,
@@ -125,27 +126,29 @@ __global__ __launch_bounds__(DEC_BLOCK) void paged_decode_kernel(
   for (int t = t0_base; t < t_end; t += 32) {
     const int t1 = t + 16;
     const bool has1 = t1 < t_end;
+    using CT = CacheTraits<CTag>;
+    using cvec = typename CT::vec8;
     const int local0 = t - part * DEC_PART;
-    const short* base0 = kv_cache +
+    const typename CT::elem* base0 = kv_cache +
         ((int64_t)blk_ids[local0 / 64] * num_kv_heads + kvh) *
             head_tile_stride +
         (int64_t)(t % 64) * head_dim + lane16 * 8;
     const int local1 = has1 ? t1 - part * DEC_PART : local0;
-    const short* base1 = has1 ? kv_cache +
+    const typename CT::elem* base1 = has1 ? kv_cache +
         ((int64_t)blk_ids[local1 / 64] * num_kv_heads + kvh) *
             head_tile_stride +
         (int64_t)(t1 % 64) * head_dim + lane16 * 8 : base0;
-    s16x8 k0 = *reinterpret_cast<const s16x8*>(base0);
-    s16x8 v0 = *reinterpret_cast<const s16x8*>(base0 + kv_plane_stride);
-    s16x8 k1 = *reinterpret_cast<const s16x8*>(base1);
-    s16x8 v1 = *reinterpret_cast<const s16x8*>(base1 + kv_plane_stride);
+    cvec k0 = *reinterpret_cast<const cvec*>(base0);
+    cvec v0 = *reinterpret_cast<const cvec*>(base0 + kv_plane_stride);
+    cvec k1 = *reinterpret_cast<const cvec*>(base1);
+    cvec v1 = *reinterpret_cast<const cvec*>(base1 + kv_plane_stride);
     float kf0[8], vf0[8], kf1[8], vf1[8];
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      kf0[j] = to_f32<Tag>(k0[j]);
-      vf0[j] = to_f32<Tag>(v0[j]);
-      kf1[j] = to_f32<Tag>(k1[j]);
-      vf1[j] = to_f32<Tag>(v1[j]);
+      kf0[j] = CT::get(k0, j);
+      vf0[j] = CT::get(v0, j);
+      kf1[j] = CT::get(k1, j);
+      vf1[j] = CT::get(v1, j);
     }
 #pragma unroll
     for (int g = 0; g < GROUP; ++g) {
@@ -311,38 +314,45 @@ void paged_decode_attention(torch::Tensor out, torch::Tensor q,
   dim3 grid(num_seqs, num_kv_heads, num_parts);
   dim3 block(DEC_BLOCK);
 
-#define LAUNCH_DEC(TAG, G, FIN)                                              \
-  hipLaunchKernelGGL((paged_decode_kernel<TAG, G, FIN>), grid, block, 0,     \
-                     stream, (short*)out.data_ptr(),                         \
+#define LAUNCH_DEC(TAG, CTAG, G, FIN)                                        \
+  hipLaunchKernelGGL((paged_decode_kernel<TAG, CTAG, G, FIN>), grid, block,  \
+                     0, stream, (short*)out.data_ptr(),                      \
                      FIN ? nullptr : tmp_out.data_ptr<float>(),              \
                      FIN ? nullptr : tmp_lse.data_ptr<float>(),              \
                      (const short*)q.data_ptr(),                             \
-                     (const short*)kv_cache.data_ptr(),                      \
+                     (const CacheTraits<CTAG>::elem*)kv_cache.data_ptr(),    \
                      block_table.data_ptr<int>(), seq_lens.data_ptr<int>(),  \
                      (float)scale, num_q_heads, num_kv_heads, head_dim,      \
                      max_blocks, num_parts, kv_cache.stride(0),              \
                      q.stride(0), (int)sliding_window)
 
-#define DISPATCH_GROUP(TAG, FIN)                                             \
+#define DISPATCH_GROUP(TAG, CTAG, FIN)                                       \
   switch (group) {                                                           \
-    case 1: LAUNCH_DEC(TAG, 1, FIN); break;                                  \
-    case 2: LAUNCH_DEC(TAG, 2, FIN); break;                                  \
-    case 4: LAUNCH_DEC(TAG, 4, FIN); break;                                  \
-    case 5: LAUNCH_DEC(TAG, 5, FIN); break;                                  \
-    case 6: LAUNCH_DEC(TAG, 6, FIN); break;                                  \
-    case 8: LAUNCH_DEC(TAG, 8, FIN); break;                                  \
-    case 3: LAUNCH_DEC(TAG, 3, FIN); break;                                  \
-    case 7: LAUNCH_DEC(TAG, 7, FIN); break;                                  \
+    case 1: LAUNCH_DEC(TAG, CTAG, 1, FIN); break;                            \
+    case 2: LAUNCH_DEC(TAG, CTAG, 2, FIN); break;                            \
+    case 4: LAUNCH_DEC(TAG, CTAG, 4, FIN); break;                            \
+    case 5: LAUNCH_DEC(TAG, CTAG, 5, FIN); break;                            \
+    case 6: LAUNCH_DEC(TAG, CTAG, 6, FIN); break;                            \
+    case 8: LAUNCH_DEC(TAG, CTAG, 8, FIN); break;                            \
+    case 3: LAUNCH_DEC(TAG, CTAG, 3, FIN); break;                            \
+    case 7: LAUNCH_DEC(TAG, CTAG, 7, FIN); break;                            \
     default: TORCH_CHECK(false, "unsupported GQA group ", group);            \
+  }
+
+#define DISPATCH_CACHE(TAG, FIN)                                             \
+  if (kv_cache.scalar_type() == torch::kFloat8_e4m3fn) {                     \
+    DISPATCH_GROUP(TAG, FP8CacheTag, FIN);                                   \
+  } else {                                                                   \
+    DISPATCH_GROUP(TAG, TAG, FIN);                                           \
   }
 
   const bool is_bf16 = q.scalar_type() == torch::kBFloat16;
   if (num_parts == 1) {
-    if (is_bf16) { DISPATCH_GROUP(BF16Tag, true); }
-    else         { DISPATCH_GROUP(FP16Tag, true); }
+    if (is_bf16) { DISPATCH_CACHE(BF16Tag, true); }
+    else         { DISPATCH_CACHE(FP16Tag, true); }
   } else {
-    if (is_bf16) { DISPATCH_GROUP(BF16Tag, false); }
-    else         { DISPATCH_GROUP(FP16Tag, false); }
+    if (is_bf16) { DISPATCH_CACHE(BF16Tag, false); }
+    else         { DISPATCH_CACHE(FP16Tag, false); }
     HIP_CHECK_KERNEL();
     dim3 rgrid(num_seqs, num_q_heads);
     if (is_bf16) {
@@ -360,6 +370,7 @@ void paged_decode_attention(torch::Tensor out, torch::Tensor q,
     }
   }
   HIP_CHECK_KERNEL();
+#undef DISPATCH_CACHE
 #undef DISPATCH_GROUP
 #undef LAUNCH_DEC
 }

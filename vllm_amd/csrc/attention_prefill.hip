@@ -38,11 +38,12 @@ DEVINL int swz(int row, int col_byte) {
   return (col_byte ^ ((row & 7) << 4));
 }
 
-template <typename Tag>
+template <typename Tag, typename CTag>
 __global__ __launch_bounds__(PF_BLOCK) void prefill_attention_kernel(
     short* __restrict__ out,             // [T, Hq, D]
     const short* __restrict__ q,         // [T, Hq, D]
-    const short* __restrict__ kv_cache,  // [2, blocks, Hkv, 64, D]
+    const typename CacheTraits<CTag>::elem* __restrict__ kv_cache,
+                                         // [2, blocks, Hkv, 64, D]
     const int* __restrict__ block_table, // [num_reqs, max_blocks]
     const int* __restrict__ query_start_loc,  // [num_reqs+1]
     const int* __restrict__ seq_lens,         // [num_reqs]
@@ -109,28 +110,36 @@ __global__ __launch_bounds__(PF_BLOCK) void prefill_attention_kernel(
 
   for (int kt = kt0; kt * KVTILE < kv_end; ++kt) {
     // --- stage K [64][128] and V^T [128][64] (all 4 waves cooperate) ----
+    using CT = CacheTraits<CTag>;
+    using cvec = typename CT::vec8;
     const int phys = block_table[(int64_t)req * max_blocks_per_req + kt];
-    const short* ksrc =
+    const typename CT::elem* ksrc =
         kv_cache + ((int64_t)phys * num_kv_heads + kvh) * head_tile;
-    const short* vsrc = ksrc + kv_plane_stride;
-    // K: 256 threads x 4 iters x 16B, swizzled ds_write_b128.
+    const typename CT::elem* vsrc = ksrc + kv_plane_stride;
+    // K: 256 threads x 4 iters x 8 elements, swizzled ds_write_b128.
+    // (fp8 cache: converted to the compute dtype while staging to LDS —
+    // the MFMA tiles always run bf16/fp16.)
 #pragma unroll
     for (int it = 0; it < 4; ++it) {
-      const int vec = it * PF_BLOCK + threadIdx.x;  // 16B chunk index
+      const int vec = it * PF_BLOCK + threadIdx.x;  // 8-elem chunk index
       const int tok = vec >> 4;                     // 16 chunks per row
       const int cb = (vec & 15) * 16;               // col byte
-      s16x8 kv8 = *reinterpret_cast<const s16x8*>(ksrc + vec * 8);
+      cvec kraw = *reinterpret_cast<const cvec*>(ksrc + vec * 8);
+      s16x8 kv8;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        kv8[j] = from_f32<Tag>(CT::get(kraw, j));
       *reinterpret_cast<s16x8*>(
           reinterpret_cast<char*>(k_lds) + tok * 256 + swz(tok, cb)) = kv8;
       // V: read the same shape, scatter-transpose into vt_lds.
-      s16x8 vv8 = *reinterpret_cast<const s16x8*>(vsrc + vec * 8);
+      cvec vraw = *reinterpret_cast<const cvec*>(vsrc + vec * 8);
       const int d0 = (vec & 15) * 8;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         const int d = d0 + j;
         *reinterpret_cast<short*>(
             reinterpret_cast<char*>(vt_lds) + d * 128 +
-            swz(d, tok * 2)) = vv8[j];
+            swz(d, tok * 2)) = from_f32<Tag>(CT::get(vraw, j));
       }
     }
     __syncthreads();
@@ -270,11 +279,12 @@ void prefill_attention(torch::Tensor out, torch::Tensor q,
   auto stream = c10::hip::getCurrentHIPStream().stream();
   dim3 grid(max_tiles, num_q_heads, num_prefills);
 
-#define LAUNCH_PF(TAG)                                                       \
-  hipLaunchKernelGGL((prefill_attention_kernel<TAG>), grid, dim3(PF_BLOCK),  \
+#define LAUNCH_PF(TAG, CTAG)                                                 \
+  hipLaunchKernelGGL((prefill_attention_kernel<TAG, CTAG>), grid,            \
+                     dim3(PF_BLOCK),                                         \
                      0, stream, (short*)out.data_ptr(),                      \
                      (const short*)q.data_ptr(),                             \
-                     (const short*)kv_cache.data_ptr(),                      \
+                     (const CacheTraits<CTAG>::elem*)kv_cache.data_ptr(),    \
                      block_table.data_ptr<int>(),                            \
                      query_start_loc.data_ptr<int>(),                        \
                      seq_lens.data_ptr<int>(), (float)scale, num_q_heads,    \
@@ -282,8 +292,14 @@ void prefill_attention(torch::Tensor out, torch::Tensor q,
                      (int)num_decodes, kv_cache.stride(0),                   \
                      q.stride(0), (int)sliding_window)
 
-  if (q.scalar_type() == torch::kBFloat16) { LAUNCH_PF(BF16Tag); }
-  else                                     { LAUNCH_PF(FP16Tag); }
+  const bool fp8c = kv_cache.scalar_type() == torch::kFloat8_e4m3fn;
+  if (q.scalar_type() == torch::kBFloat16) {
+    if (fp8c) { LAUNCH_PF(BF16Tag, FP8CacheTag); }
+    else      { LAUNCH_PF(BF16Tag, BF16Tag); }
+  } else {
+    if (fp8c) { LAUNCH_PF(FP16Tag, FP8CacheTag); }
+    else      { LAUNCH_PF(FP16Tag, FP16Tag); }
+  }
 #undef LAUNCH_PF
   HIP_CHECK_KERNEL();
 }

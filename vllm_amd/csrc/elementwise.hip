@@ -190,13 +190,16 @@ __global__ void rope_kernel(short* __restrict__ q, short* __restrict__ k,
 // (head-major inside the block so each (block, head) tile is contiguous);
 // slot_mapping: [T] int64 = block_id * block_size + offset.
 // One workgroup per token; vectorized 16B copies.
-template <typename Tag>
+template <typename Tag, typename CTag>
 __global__ void reshape_and_cache_kernel(
     const short* __restrict__ key, const short* __restrict__ value,
-    short* __restrict__ kv_cache, const int64_t* __restrict__ slot_mapping,
+    typename CacheTraits<CTag>::elem* __restrict__ kv_cache,
+    const int64_t* __restrict__ slot_mapping,
     const int64_t kv_stride,  // elements between K and V planes
     const int64_t k_row_stride, const int64_t v_row_stride,
     const int num_kv_heads, const int head_dim, const int block_size) {
+  using CT = CacheTraits<CTag>;
+  using cvec = typename CT::vec8;
   const int64_t t = blockIdx.x;
   const int64_t slot = slot_mapping[t];
   if (slot < 0) return;  // padding token
@@ -211,10 +214,16 @@ __global__ void reshape_and_cache_kernel(
     const int dv = i % dvec;
     const int64_t dst =
         (((blk * num_kv_heads + h) * block_size + off) * head_dim) / 8 + dv;
-    reinterpret_cast<s16x8*>(kv_cache)[dst] =
-        reinterpret_cast<const s16x8*>(krow)[i];
-    reinterpret_cast<s16x8*>(kv_cache + kv_stride)[dst] =
-        reinterpret_cast<const s16x8*>(vrow)[i];
+    s16x8 kv8 = reinterpret_cast<const s16x8*>(krow)[i];
+    s16x8 vv8 = reinterpret_cast<const s16x8*>(vrow)[i];
+    cvec kc, vc;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      kc[j] = CT::put(to_f32<Tag>(kv8[j]));
+      vc[j] = CT::put(to_f32<Tag>(vv8[j]));
+    }
+    reinterpret_cast<cvec*>(kv_cache)[dst] = kc;
+    reinterpret_cast<cvec*>(kv_cache + kv_stride)[dst] = vc;
   }
 }
 
@@ -353,7 +362,7 @@ void reshape_and_cache(torch::Tensor key, torch::Tensor value,
                        torch::Tensor kv_cache, torch::Tensor slot_mapping) {
   check_rows_16b(key, "key");
   check_rows_16b(value, "value");
-  check_16b(kv_cache, "kv_cache");
+  TORCH_CHECK(kv_cache.is_contiguous(), "kv_cache must be contiguous");
   const int T = key.size(0);
   if (T == 0) return;
   const int num_kv_heads = key.size(1);
@@ -363,17 +372,24 @@ void reshape_and_cache(torch::Tensor key, torch::Tensor value,
   TORCH_CHECK(kv_cache.size(2) == num_kv_heads);
   const int64_t kv_stride = kv_cache.stride(0);
   auto stream = c10::hip::getCurrentHIPStream().stream();
-#define LAUNCH_RC(TAG)                                                 \
-  hipLaunchKernelGGL((reshape_and_cache_kernel<TAG>), dim3(T),         \
+#define LAUNCH_RC2(TAG, CTAG)                                          \
+  hipLaunchKernelGGL((reshape_and_cache_kernel<TAG, CTAG>), dim3(T),   \
                      dim3(128), 0, stream,                             \
                      (const short*)key.data_ptr(),                     \
                      (const short*)value.data_ptr(),                   \
-                     (short*)kv_cache.data_ptr(),                      \
+                     (CacheTraits<CTAG>::elem*)kv_cache.data_ptr(),    \
                      slot_mapping.data_ptr<int64_t>(), kv_stride,      \
                      key.stride(0), value.stride(0),                    \
                      num_kv_heads, head_dim, block_size)
-  DISPATCH_16B(key, LAUNCH_RC);
-#undef LAUNCH_RC
+  const bool fp8c = kv_cache.scalar_type() == torch::kFloat8_e4m3fn;
+  if (key.scalar_type() == torch::kBFloat16) {
+    if (fp8c) { LAUNCH_RC2(BF16Tag, FP8CacheTag); }
+    else      { LAUNCH_RC2(BF16Tag, BF16Tag); }
+  } else {
+    if (fp8c) { LAUNCH_RC2(FP16Tag, FP8CacheTag); }
+    else      { LAUNCH_RC2(FP16Tag, FP16Tag); }
+  }
+#undef LAUNCH_RC2
   HIP_CHECK_KERNEL();
 }
 

@@ -80,8 +80,8 @@ def reshape_and_cache(key, value, kv_cache, slot_mapping):
     block_size = kv_cache.shape[3]
     block_ids = slot_mapping // block_size
     offsets = slot_mapping % block_size
-    kv_cache[0, block_ids, :, offsets] = key
-    kv_cache[1, block_ids, :, offsets] = value
+    kv_cache[0, block_ids, :, offsets] = key.to(kv_cache.dtype)
+    kv_cache[1, block_ids, :, offsets] = value.to(kv_cache.dtype)
 
 
 def _gather_kv(kv_cache, block_table_row, ctx_len):

@@ -199,6 +199,9 @@ class ModelRunner:
         self.spec = config.model_config.spec
         self.dtype = config.model_config.torch_dtype
         self.requests: dict[str, CachedReqState] = {}
+        kvd = config.cache_config.kv_cache_dtype
+        self.cache_dtype = (torch.float8_e4m3fn if kvd == "fp8"
+                            else self.dtype)
         self.kv_caches: list[torch.Tensor] = []
         self.model: Optional[torch.nn.Module] = None
         self.graph_runner: Optional[DecodeGraphRunner] = None
@@ -242,7 +245,7 @@ class ModelRunner:
         from vllm_amd.parallel.state import get_tp_world_size
 
         spec = self.spec
-        elt = torch.tensor([], dtype=self.dtype).element_size()
+        elt = torch.tensor([], dtype=self.cache_dtype).element_size()
         if spec.is_mla:
             # Compressed MLA cache: kv_lora + rope values per token,
             # replicated across TP ranks (vs per-head K+V for GQA).
@@ -259,6 +262,7 @@ class ModelRunner:
 
         spec = self.spec
         if spec.is_mla:
+            # MLA cache stays at model precision (fp8 MLA cache later).
             per_tok = spec.kv_lora_rank + spec.qk_rope_head_dim
             self.kv_caches = [
                 torch.zeros(num_blocks, self.block_size, per_tok,
@@ -273,7 +277,7 @@ class ModelRunner:
         self.kv_caches = [
             torch.zeros(
                 2, num_blocks, kv_heads, self.block_size, spec.head_dim,
-                dtype=self.dtype, device=self.device,
+                dtype=self.cache_dtype, device=self.device,
             )
             for _ in range(spec.num_layers)
         ]
