@@ -43,6 +43,9 @@ def parse_args():
     p.add_argument("--block-size", type=int, default=64)
     p.add_argument("--num-gpu-blocks", type=int, default=None,
                    help="skip the profiling pass with a fixed KV pool size")
+    p.add_argument("--kv-cache-dtype", type=str, default="auto",
+                   choices=["auto", "fp8"],
+                   help="fp8 halves KV bytes (attention-bound large-batch)")
     return p.parse_args()
 
 
@@ -81,6 +84,7 @@ def main():
             block_size=args.block_size,
             num_gpu_blocks=args.num_gpu_blocks,
             enable_prefix_caching=False,
+            kv_cache_dtype=args.kv_cache_dtype,
         ),
         scheduler_config=SchedulerConfig(
             max_num_batched_tokens=max(prefill_tokens, 8192),
