@@ -175,3 +175,19 @@ def test_concurrent_requests(client):
     with cf.ThreadPoolExecutor(4) as ex:
         results = list(ex.map(one, range(8)))
     assert results == [6] * 8
+
+
+def test_bench_serving_tool():
+    """The serving benchmark tool runs end-to-end on CPU."""
+    import subprocess
+    import sys
+
+    r = subprocess.run(
+        [sys.executable, "benchmarks/bench_serving.py", "--num-prompts", "4",
+         "--qps", "100", "--input-len", "16", "--output-len", "4"],
+        capture_output=True, text=True, timeout=240,
+    )
+    assert r.returncode == 0, r.stderr[-800:]
+    out = json.loads(r.stdout.strip().splitlines()[-1])
+    assert out["output_tokens_per_s"] > 0
+    assert out["ttft_ms"]["p50"] > 0
