@@ -339,6 +339,8 @@ class ModelConfig:
     model_path: Optional[str] = None  # weights dir for safetensors
     seed: int = 0
     enforce_eager: bool = False
+    # name -> local PEFT checkpoint dir; ids are 1-based in listed order.
+    lora_modules: Optional[dict] = None
     spec: ModelSpec = None  # type: ignore[assignment]
 
     def __post_init__(self) -> None:
@@ -351,6 +353,14 @@ class ModelConfig:
     @property
     def torch_dtype(self) -> torch.dtype:
         return _STR_TO_DTYPE[self.dtype]
+
+    def lora_id_of(self, name: Optional[str]) -> int:
+        if name is None:
+            return 0
+        names = list(self.lora_modules or {})
+        if name not in names:
+            raise ValueError(f"unknown LoRA adapter {name!r}")
+        return names.index(name) + 1
 
 
 @dataclass

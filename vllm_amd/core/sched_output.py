@@ -21,6 +21,7 @@ class NewRequestData:
     num_computed_tokens: int
     sampling_params: SamplingParams
     grammar: object = None  # CompiledGrammar for structured output
+    lora_id: int = 0
 
 
 @dataclass

@@ -272,6 +272,7 @@ class Scheduler:
                         num_computed_tokens=num_computed,
                         sampling_params=request.sampling_params,
                         grammar=request.grammar,
+                        lora_id=request.lora_id,
                     )
                 )
             num_scheduled_tokens[request.request_id] = num_new_tokens

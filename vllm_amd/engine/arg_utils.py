@@ -38,6 +38,7 @@ class EngineArgs:
     enable_chunked_prefill: bool = True
     scheduling_policy: str = "fcfs"
 
+    lora_modules: Optional[dict] = None  # name -> PEFT dir
     num_speculative_tokens: int = 0
     ngram_prompt_lookup_min: int = 2
     ngram_prompt_lookup_max: int = 4
@@ -73,6 +74,10 @@ class EngineArgs:
                             action="store_false")
         parser.add_argument("--scheduling-policy", type=str, default="fcfs",
                             choices=["fcfs", "priority"])
+        parser.add_argument(
+            "--lora-modules", type=str, nargs="*", default=None,
+            metavar="NAME=PATH",
+            help="LoRA adapters to serve (requests select by model name)")
         parser.add_argument("--num-speculative-tokens", type=int, default=0)
         parser.add_argument("--ngram-prompt-lookup-min", type=int, default=2)
         parser.add_argument("--ngram-prompt-lookup-max", type=int, default=4)
@@ -87,6 +92,10 @@ class EngineArgs:
     def from_cli_args(cls, args: argparse.Namespace) -> "EngineArgs":
         attrs = [f.name for f in fields(cls)]
         kwargs = {a: getattr(args, a) for a in attrs if hasattr(args, a)}
+        lm = kwargs.get("lora_modules")
+        if isinstance(lm, list):
+            kwargs["lora_modules"] = dict(
+                item.split("=", 1) for item in lm) if lm else None
         if hasattr(args, "scheduling_policy"):
             kwargs["scheduling_policy"] = args.scheduling_policy
         return cls(**kwargs)
@@ -104,6 +113,7 @@ class EngineArgs:
         return EngineConfig(
             model_config=ModelConfig(
                 model=self.model,
+                lora_modules=self.lora_modules,
                 tokenizer=self.tokenizer,
                 dtype=self.dtype,
                 max_model_len=self.max_model_len,

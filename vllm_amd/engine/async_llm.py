@@ -59,9 +59,10 @@ class AsyncLLM:
                 block = False
                 kind = item[0]
                 if kind == "add":
-                    _, rid, prompt, params, loop, out_q = item
+                    _, rid, prompt, params, lora, loop, out_q = item
                     try:
-                        self.engine.add_request(rid, prompt, params)
+                        self.engine.add_request(rid, prompt, params,
+                                                lora=lora)
                         self._streams[rid] = (loop, out_q)
                     except Exception as e:  # noqa: BLE001
                         loop.call_soon_threadsafe(out_q.put_nowait, e)
@@ -101,13 +102,14 @@ class AsyncLLM:
         prompt: Union[str, list[int]],
         sampling_params: Optional[SamplingParams] = None,
         request_id: Optional[str] = None,
+        lora: Optional[str] = None,
     ) -> AsyncGenerator[RequestOutput, None]:
         """Submit a request and stream RequestOutputs until finished."""
         rid = request_id or self._next_id()
         params = sampling_params or SamplingParams()
         loop = asyncio.get_running_loop()
         out_q: asyncio.Queue = asyncio.Queue()
-        self._inbox.put(("add", rid, prompt, params, loop, out_q))
+        self._inbox.put(("add", rid, prompt, params, lora, loop, out_q))
         try:
             while True:
                 item = await out_q.get()

@@ -165,6 +165,7 @@ class LLMEngine:
         request_id: Optional[str],
         prompt: Union[str, list[int]],
         params: Optional[SamplingParams] = None,
+        lora: Optional[str] = None,
     ) -> str:
         if request_id is None:
             request_id = f"req-{self._request_counter}"
@@ -193,6 +194,7 @@ class LLMEngine:
             sampling_params=params,
             eos_token_id=eos,
             prompt=prompt_text,
+            lora_id=self.config.model_config.lora_id_of(lora),
         )
         if params.guided_choice:
             from vllm_amd.structured_output import compile_choice_grammar

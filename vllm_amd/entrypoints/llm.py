@@ -22,6 +22,7 @@ class LLM:
         sampling_params: Optional[
             Union[SamplingParams, list[SamplingParams]]
         ] = None,
+        lora: Optional[str] = None,
     ) -> list[RequestOutput]:
         if isinstance(prompts, str) or (
             prompts and isinstance(prompts, list)
@@ -35,7 +36,8 @@ class LLM:
             sampling_params = [sampling_params] * n
         req_ids = []
         for prompt, params in zip(prompts, sampling_params):
-            req_ids.append(self.engine.add_request(None, prompt, params))
+            req_ids.append(
+                self.engine.add_request(None, prompt, params, lora=lora))
         order = {rid: i for i, rid in enumerate(req_ids)}
         finals: dict[str, RequestOutput] = {}
         while self.engine.has_unfinished_requests():

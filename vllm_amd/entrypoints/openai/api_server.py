@@ -58,6 +58,8 @@ class ServerState:
     def __init__(self, engine: AsyncLLM, model_name: str):
         self.engine = engine
         self.model_name = model_name
+        self.lora_names = list(
+            engine.config.model_config.lora_modules or {})
         self.max_model_len = engine.config.model_config.max_model_len
         # Prometheus counters.
         self.num_requests = 0
@@ -80,10 +82,11 @@ def build_app(state: ServerState) -> FastAPI:
 
     @app.get("/v1/models")
     async def list_models() -> ModelList:
-        return ModelList(data=[
-            ModelCard(id=state.model_name,
-                      max_model_len=state.max_model_len)
-        ])
+        cards = [ModelCard(id=state.model_name,
+                           max_model_len=state.max_model_len)]
+        cards += [ModelCard(id=n, max_model_len=state.max_model_len)
+                  for n in state.lora_names]
+        return ModelList(data=cards)
 
     @app.post("/tokenize")
     async def tokenize(req: TokenizeRequest) -> TokenizeResponse:
@@ -129,13 +132,15 @@ def build_app(state: ServerState) -> FastAPI:
             return _error("batched prompts: send one prompt per request")
         prompt = prompts[0]
         params = req.to_sampling_params(req.stream)
+        lora = req.model if req.model in state.lora_names else None
         state.num_requests += 1
         rid = random_id("cmpl")
 
         if req.stream:
             async def gen() -> AsyncGenerator[str, None]:
                 try:
-                    async for out in engine.generate(prompt, params, rid):
+                    async for out in engine.generate(prompt, params, rid,
+                                                     lora=lora):
                         comp = out.outputs[0]
                         state.num_generation_tokens += len(comp.token_ids)
                         chunk = {
@@ -159,7 +164,7 @@ def build_app(state: ServerState) -> FastAPI:
 
         final = None
         try:
-            async for out in engine.generate(prompt, params, rid):
+            async for out in engine.generate(prompt, params, rid, lora=lora):
                 final = out
         except ValueError as e:
             return _error(str(e))
@@ -195,6 +200,7 @@ def build_app(state: ServerState) -> FastAPI:
         )
         default_max = state.max_model_len
         params = req.to_sampling_params(req.stream, default_max)
+        lora = req.model if req.model in state.lora_names else None
         state.num_requests += 1
         rid = random_id("chatcmpl")
 
@@ -208,7 +214,8 @@ def build_app(state: ServerState) -> FastAPI:
                 )
                 yield f"data: {first.model_dump_json()}\n\n"
                 try:
-                    async for out in engine.generate(prompt, params, rid):
+                    async for out in engine.generate(prompt, params, rid,
+                                                     lora=lora):
                         comp = out.outputs[0]
                         state.num_generation_tokens += len(comp.token_ids)
                         chunk = ChatCompletionStreamResponse(
@@ -228,7 +235,7 @@ def build_app(state: ServerState) -> FastAPI:
 
         final = None
         try:
-            async for out in engine.generate(prompt, params, rid):
+            async for out in engine.generate(prompt, params, rid, lora=lora):
                 final = out
         except ValueError as e:
             return _error(str(e))

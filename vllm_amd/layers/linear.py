@@ -26,6 +26,28 @@ from vllm_amd.parallel.state import (
 )
 
 
+def _maybe_apply_lora(module, x, y):
+    """Add per-request LoRA deltas when the forward context carries
+    adapter ids and this layer has registered slices (lora.py)."""
+    slices = getattr(module, "lora_slices", None)
+    if slices is None:
+        return y
+    from vllm_amd.worker.forward_context import get_forward_context
+
+    try:
+        ctx = get_forward_context()
+    except Exception:  # outside a model step (e.g. unit use)
+        return y
+    if ctx is None or ctx.lora_manager is None or ctx.lora_ids is None:
+        return y
+    from vllm_amd.lora import apply_lora_slices
+
+    x2 = x.reshape(-1, x.shape[-1])
+    y2 = y.reshape(-1, y.shape[-1])
+    apply_lora_slices(x2, y2, ctx.lora_ids, ctx.lora_manager, slices)
+    return y
+
+
 class ReplicatedLinear(nn.Module):
     def __init__(self, input_size, output_size, bias=False, dtype=None):
         super().__init__()
@@ -79,6 +101,7 @@ class ColumnParallelLinear(nn.Module):
 
     def forward(self, x):
         y = ops.linear(x, self.weight, self.bias)
+        y = _maybe_apply_lora(self, x, y)
         if self.gather_output:
             y = tensor_model_parallel_all_gather(y, dim=-1)
         return y
@@ -197,6 +220,7 @@ class RowParallelLinear(nn.Module):
 
     def forward(self, x):
         y = ops.linear(x, self.weight, None)
+        y = _maybe_apply_lora(self, x, y)
         if self.reduce_results:
             y = tensor_model_parallel_all_reduce(y)
         if self.bias is not None:

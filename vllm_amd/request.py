@@ -49,6 +49,7 @@ class Request:
         arrival_time: Optional[float] = None,
         priority: int = 0,
         prompt: Optional[str] = None,
+        lora_id: int = 0,
     ) -> None:
         self.request_id = request_id
         self.prompt_token_ids = prompt_token_ids
@@ -57,6 +58,7 @@ class Request:
         self.eos_token_id = eos_token_id
         self.arrival_time = arrival_time if arrival_time is not None else time.time()
         self.priority = priority
+        self.lora_id = lora_id
 
         self.status = RequestStatus.WAITING
         self.stop_reason: Optional[object] = None

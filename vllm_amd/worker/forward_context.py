@@ -35,6 +35,9 @@ class AttentionMetadata:
 class ForwardContext:
     attn_metadata: Optional[AttentionMetadata]
     kv_caches: list[torch.Tensor]  # one per layer; [] during profiling
+    # Multi-LoRA: per-token adapter ids (0 = base) + the manager.
+    lora_ids: Optional[torch.Tensor] = None
+    lora_manager: object = None
 
 
 _forward_context: Optional[ForwardContext] = None

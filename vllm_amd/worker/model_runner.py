@@ -38,6 +38,7 @@ class CachedReqState:
     sampling_params: SamplingParams
     grammar: object = None
     grammar_state: object = None
+    lora_id: int = 0
 
     @property
     def output_token_ids(self) -> list[int]:
@@ -219,6 +220,8 @@ class ModelRunner:
         self.np_computed = np.zeros(n, dtype=np.int64)
         self.np_block_table = np.zeros((n, mb), dtype=np.int32)
         self.np_nblocks = np.zeros(n, dtype=np.int32)
+        self.np_lora = np.zeros(n, dtype=np.int64)
+        self.lora_manager = None
         # Sampling-metadata cache for steady-state decode batches.
         self._samp_cache_key = None
         self._samp_cache_val = None
@@ -238,6 +241,21 @@ class ModelRunner:
 
     def load_model(self) -> None:
         self.model = load_model(self.config.model_config, self.device)
+        mc = self.config.model_config
+        self.lora_manager = None
+        if mc.lora_modules:
+            from vllm_amd.parallel.state import get_tp_world_size
+
+            if get_tp_world_size() > 1:
+                raise NotImplementedError("LoRA with TP>1 lands later")
+            from vllm_amd.lora import (LoRAAdapter, LoRAManager,
+                                       attach_lora_metadata)
+
+            self.lora_manager = LoRAManager()
+            for name, path in mc.lora_modules.items():
+                self.lora_manager.register(LoRAAdapter.from_path(
+                    name, path, mc.torch_dtype, self.device))
+            attach_lora_metadata(self.model)
 
     # ------------------------------------------------------------------
     def kv_cache_page_bytes(self) -> int:
@@ -286,6 +304,7 @@ class ModelRunner:
         # kernel lands).
         if (self.device.type == "cuda"
                 and not self.config.model_config.enforce_eager
+                and not self.config.model_config.lora_modules
                 and not self.spec.is_moe and not self.spec.is_mla):
             self.graph_runner = DecodeGraphRunner(self)
 
@@ -306,6 +325,7 @@ class ModelRunner:
                 grammar=nr.grammar,
                 grammar_state=(nr.grammar.initial_state()
                                if nr.grammar is not None else None),
+                lora_id=nr.lora_id,
             )
             row = self._free_rows.pop()
             self._row_of[nr.req_id] = row
@@ -313,6 +333,7 @@ class ModelRunner:
             self.np_block_table[row, :nb] = nr.block_ids
             self.np_nblocks[row] = nb
             self.np_computed[row] = nr.num_computed_tokens
+            self.np_lora[row] = nr.lora_id
         cr = so.scheduled_cached_reqs
         for i, req_id in enumerate(cr.req_ids):
             state = self.requests[req_id]
@@ -415,7 +436,11 @@ class ModelRunner:
             )
             ids_t = (ids_dev if ids_dev is not None
                      else torch.from_numpy(input_ids).to(dev))
-            ctx = ForwardContext(attn_metadata=meta, kv_caches=self.kv_caches)
+            ctx = ForwardContext(
+                attn_metadata=meta, kv_caches=self.kv_caches,
+                lora_ids=self._lora_ids_tensor(self.np_lora[rows]),
+                lora_manager=self.lora_manager,
+            )
             with set_forward_context(ctx):
                 hidden = self.model(ids_t, torch.from_numpy(positions).to(dev))
             logits = self.model.compute_logits(hidden)
@@ -480,6 +505,11 @@ class ModelRunner:
             sampled_token_ids=sampled_per_req,
             logprobs=logprobs_per_req or None,
         )
+
+    def _lora_ids_tensor(self, arr):
+        if self.lora_manager is None or arr is None:
+            return None
+        return torch.from_numpy(np.ascontiguousarray(arr)).to(self.device)
 
     def _sampling_meta(self, req_ids, states, dev) -> SamplingMetadata:
         """SamplingMetadata with a steady-state cache: reused while the
@@ -602,7 +632,17 @@ class ModelRunner:
             input_ids_t = torch.from_numpy(input_ids).to(dev)
             positions_t = torch.from_numpy(positions).to(dev)
 
-            ctx = ForwardContext(attn_metadata=meta, kv_caches=self.kv_caches)
+            lora_np = None
+            if self.lora_manager is not None:
+                lora_np = np.concatenate([
+                    np.full(n_, self.np_lora[self._row_of[rid]])
+                    for rid, n_ in items
+                ])
+            ctx = ForwardContext(
+                attn_metadata=meta, kv_caches=self.kv_caches,
+                lora_ids=self._lora_ids_tensor(lora_np),
+                lora_manager=self.lora_manager,
+            )
             with set_forward_context(ctx):
                 hidden = self.model(input_ids_t, positions_t)
             logits_all = None
