@@ -21,13 +21,15 @@ def _make_adapter(tmp_path, spec, rank=4, alpha=8, seed=7):
         for mod, out_dim, in_dim in [
             ("self_attn.q_proj", qs, H),
             ("self_attn.v_proj", ks, H),
+            ("self_attn.o_proj", H, qs),
             ("mlp.down_proj", H, spec.intermediate_size),
         ]:
-            # Large enough to flip greedy argmax on a dummy-init model.
+            # Large enough to flip greedy argmax on a dummy-init model
+            # (whose weights are ~1e-3, so activations are tiny).
             t[f"{base}.{mod}.lora_A.weight"] = \
-                torch.randn(rank, in_dim, generator=gen) * 0.5
+                torch.randn(rank, in_dim, generator=gen) * 5.0
             t[f"{base}.{mod}.lora_B.weight"] = \
-                torch.randn(out_dim, rank, generator=gen) * 0.5
+                torch.randn(out_dim, rank, generator=gen) * 5.0
     d = tmp_path / "adapter"
     d.mkdir()
     save_file(t, str(d / "adapter_model.safetensors"))
@@ -98,3 +100,28 @@ def test_lora_mixed_batch(tmp_path):
     assert done[r_base] == done[r_b2]
     assert done[r_lora] == done[r_l2]
     assert done[r_base] != done[r_lora]
+
+
+def test_lora_prefix_cache_isolation(tmp_path):
+    """The prefix cache must NOT share blocks across adapters: the same
+    prompt under base vs adapter has different K/V contents."""
+    from vllm_amd.config import get_model_spec
+    from vllm_amd.entrypoints.llm import LLM
+    from vllm_amd.sampling_params import SamplingParams
+
+    spec = get_model_spec("tiny-llama")
+    path = _make_adapter(tmp_path, spec)
+    llm = LLM(model="tiny-llama", dtype="fp32", device="cpu", block_size=16,
+              num_gpu_blocks=128, max_model_len=256,
+              max_num_batched_tokens=256, max_num_seqs=4,
+              lora_modules={"a": path})
+    p = SamplingParams(temperature=0.0, max_tokens=6, ignore_eos=True)
+    prompt = list(range(10, 42))  # two full 16-token blocks
+    tuned_cold = llm.generate([prompt], p, lora="a")[0].outputs[0].token_ids
+    base = llm.generate([prompt], p)[0].outputs[0].token_ids
+    # tuned again, now with base's blocks in the cache: must still match
+    # the cold tuned run (no cross-adapter block reuse).
+    tuned_warm = llm.generate([prompt], p, lora="a")[0].outputs[0].token_ids
+    llm.shutdown()
+    assert tuned_cold == tuned_warm
+    assert base != tuned_cold

@@ -44,6 +44,7 @@ class KVCacheManager:
             self.block_size,
             request.all_token_ids,
             prior_hashes=request.block_hashes,
+            salt=getattr(request, "lora_id", 0) or None,
         )
         computed: list[KVCacheBlock] = []
         for h in request.block_hashes:
@@ -112,6 +113,7 @@ class KVCacheManager:
                 self.block_size,
                 request.all_token_ids,
                 prior_hashes=request.block_hashes,
+                salt=getattr(request, "lora_id", 0) or None,
             )
             # Only blocks whose tokens are all known can be hashed; with
             # chunked prefill total_tokens <= num_tokens so this holds.

@@ -33,14 +33,19 @@ def hash_block_tokens(
 def hash_request_tokens(
     block_size: int, token_ids: list[int], start_block: int = 0,
     prior_hashes: Optional[list[BlockHash]] = None,
+    salt: Optional[int] = None,
 ) -> list[BlockHash]:
     """Hash all *full* blocks of a token stream, reusing prior prefix hashes.
 
+    `salt` distinguishes KV contents beyond token ids (LoRA adapter id —
+    an adapter's k/v projections change the cached values; role of the
+    reference's block-hash extra_keys, kv_cache_utils.py:576).
     Returns the complete list of full-block hashes (prior + new).
     """
     hashes: list[BlockHash] = list(prior_hashes) if prior_hashes else []
     num_full_blocks = len(token_ids) // block_size
-    parent = hashes[-1].value if hashes else None
+    parent = hashes[-1].value if hashes else (
+        -salt if salt else None)
     for i in range(len(hashes), num_full_blocks):
         block_tokens = tuple(token_ids[i * block_size : (i + 1) * block_size])
         h = hash_block_tokens(parent, block_tokens)
