@@ -132,3 +132,43 @@ def test_fp8_kv_cache_e2e():
     )
     llm.shutdown()
     assert all(len(o.outputs[0].token_ids) == 12 for o in outs)
+
+
+def test_lora_gpu(tmp_path):
+    from tests.test_lora import _make_adapter
+    from vllm_amd.config import get_model_spec
+    from vllm_amd.sampling_params import SamplingParams
+
+    spec = get_model_spec("tiny-llama-128")
+    path = _make_adapter(tmp_path, spec)
+    llm = _llm(lora_modules={"a": path})
+    p = SamplingParams(temperature=0.0, max_tokens=6, ignore_eos=True)
+    prompt = list(range(10, 42))
+    base = llm.generate([prompt], p)[0].outputs[0].token_ids
+    tuned = llm.generate([prompt], p, lora="a")[0].outputs[0].token_ids
+    llm.shutdown()
+    assert base != tuned
+
+
+def test_guided_choice_gpu():
+    from vllm_amd.sampling_params import SamplingParams
+
+    llm = _llm()
+    tok = llm.engine.tokenizer
+    choices = ["yes", "no"]
+    outs = llm.generate(
+        ["constrained?"],
+        SamplingParams(temperature=0.0, max_tokens=8,
+                       guided_choice=choices),
+    )
+    llm.shutdown()
+    toks = outs[0].outputs[0].token_ids
+    body = toks[:-1] if toks and toks[-1] == tok.eos_token_id else toks
+    bos = getattr(tok.tokenizer, "bos_token_id", None)
+    seqs = []
+    for c in choices:
+        ids = tok.encode(c)
+        if bos is not None and ids and ids[0] == bos:
+            ids = ids[1:]
+        seqs.append(ids)
+    assert any(body == s for s in seqs)

@@ -191,3 +191,31 @@ def test_bench_serving_tool():
     out = json.loads(r.stdout.strip().splitlines()[-1])
     assert out["output_tokens_per_s"] > 0
     assert out["ttft_ms"]["p50"] > 0
+
+
+def test_anthropic_messages(client):
+    r = client.post("/v1/messages", json={
+        "model": "tiny-llama",
+        "max_tokens": 6,
+        "messages": [{"role": "user", "content": "hello"}],
+        "temperature": 0.0,
+    })
+    assert r.status_code == 200, r.text
+    d = r.json()
+    assert d["type"] == "message" and d["role"] == "assistant"
+    assert d["usage"]["output_tokens"] >= 1
+    assert d["content"][0]["type"] == "text"
+
+
+def test_anthropic_messages_stream(client):
+    with client.stream("POST", "/v1/messages", json={
+        "model": "tiny-llama",
+        "max_tokens": 4,
+        "messages": [{"role": "user", "content": "hi"}],
+        "temperature": 0.0,
+        "stream": True,
+    }) as r:
+        events = [ln.split(": ", 1)[1] for ln in r.iter_lines()
+                  if ln.startswith("event: ")]
+    assert events[0] == "message_start"
+    assert "message_stop" in events

@@ -8,10 +8,20 @@ import sys
 
 def main() -> None:
     if len(sys.argv) < 2 or sys.argv[1] in ("-h", "--help"):
-        print("usage: python -m vllm_amd {serve} [args]\n"
-              "  serve  — start the OpenAI-compatible API server")
+        print("usage: python -m vllm_amd {serve,bench} [args]\n"
+              "  serve          — start the OpenAI-compatible API server\n"
+              "  bench serving  — TTFT/ITL + throughput at fixed QPS")
         return
     cmd = sys.argv.pop(1)
+    if cmd == "bench":
+        # `python -m vllm_amd bench serving [args]`
+        sub = sys.argv.pop(1) if len(sys.argv) > 1 else "serving"
+        if sub != "serving":
+            raise SystemExit("bench subcommands: serving")
+        from benchmarks.bench_serving import main as bench_main
+
+        bench_main()
+        return
     if cmd == "serve":
         # `python -m vllm_amd serve <model> [args]` or with --model.
         if len(sys.argv) > 1 and not sys.argv[1].startswith("-"):

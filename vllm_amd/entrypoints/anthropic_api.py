@@ -1,0 +1,150 @@
+"""Anthropic Messages API (/v1/messages) compatibility router (role of
+the reference's vllm/entrypoints/anthropic/). Mounted by the OpenAI API
+server onto the same app/engine."""
+
+from __future__ import annotations
+
+import json
+import time
+from typing import Any, AsyncGenerator, Optional, Union
+
+from fastapi import APIRouter
+from fastapi.responses import JSONResponse, StreamingResponse
+from pydantic import BaseModel, Field
+
+from vllm_amd.sampling_params import RequestOutputKind, SamplingParams
+
+
+class AnthropicMessage(BaseModel):
+    role: str
+    content: Union[str, list[dict[str, Any]]]
+
+    def text(self) -> str:
+        if isinstance(self.content, str):
+            return self.content
+        return "".join(seg.get("text", "") for seg in self.content
+                       if seg.get("type") == "text")
+
+
+class MessagesRequest(BaseModel):
+    model: str
+    messages: list[AnthropicMessage]
+    max_tokens: int
+    system: Optional[Union[str, list[dict[str, Any]]]] = None
+    stop_sequences: Optional[list[str]] = None
+    stream: bool = False
+    temperature: float = 1.0
+    top_k: int = 0
+    top_p: float = 1.0
+    metadata: Optional[dict] = None
+
+
+def build_anthropic_router(state) -> APIRouter:
+    router = APIRouter()
+    engine = state.engine
+
+    def to_prompt(req: MessagesRequest) -> str:
+        parts = []
+        if req.system:
+            sys_text = req.system if isinstance(req.system, str) else \
+                "".join(s.get("text", "") for s in req.system)
+            parts.append(f"<|system|>\n{sys_text}")
+        for m in req.messages:
+            parts.append(f"<|{m.role}|>\n{m.text()}")
+        parts.append("<|assistant|>\n")
+        return "\n".join(parts)
+
+    @router.post("/v1/messages")
+    async def messages(req: MessagesRequest):
+        params = SamplingParams(
+            temperature=req.temperature,
+            top_p=req.top_p,
+            top_k=req.top_k,
+            max_tokens=req.max_tokens,
+            stop=req.stop_sequences,
+            output_kind=(RequestOutputKind.DELTA if req.stream
+                         else RequestOutputKind.FINAL_ONLY),
+        )
+        prompt = to_prompt(req)
+        lora = req.model if req.model in state.lora_names else None
+        rid = f"msg_{int(time.time() * 1e6):x}"
+        state.num_requests += 1
+
+        if req.stream:
+            async def gen() -> AsyncGenerator[str, None]:
+                start = {
+                    "type": "message_start",
+                    "message": {"id": rid, "type": "message",
+                                "role": "assistant", "content": [],
+                                "model": req.model,
+                                "usage": {"input_tokens": 0,
+                                          "output_tokens": 0}},
+                }
+                yield ("event: message_start\n"
+                       f"data: {json.dumps(start)}\n\n")
+                yield ("event: content_block_start\n"
+                       'data: {"type": "content_block_start", "index": 0, '
+                       '"content_block": {"type": "text", "text": ""}}\n\n')
+                n_out = 0
+                stop_reason = "end_turn"
+                async for out in engine.generate(prompt, params, rid,
+                                                 lora=lora):
+                    comp = out.outputs[0]
+                    n_out = len(comp.token_ids)
+                    if comp.text:
+                        delta = {"type": "content_block_delta", "index": 0,
+                                 "delta": {"type": "text_delta",
+                                           "text": comp.text}}
+                        yield ("event: content_block_delta\n"
+                               f"data: {json.dumps(delta)}\n\n")
+                    if comp.finish_reason == "length":
+                        stop_reason = "max_tokens"
+                    elif comp.finish_reason == "stop":
+                        stop_reason = ("stop_sequence"
+                                       if comp.stop_reason is not None
+                                       and not isinstance(comp.stop_reason,
+                                                          int)
+                                       else "end_turn")
+                yield ("event: content_block_stop\n"
+                       'data: {"type": "content_block_stop", "index": 0}'
+                       "\n\n")
+                md = {"type": "message_delta",
+                      "delta": {"stop_reason": stop_reason},
+                      "usage": {"output_tokens": n_out}}
+                yield f"event: message_delta\ndata: {json.dumps(md)}\n\n"
+                yield ('event: message_stop\n'
+                       'data: {"type": "message_stop"}\n\n')
+            return StreamingResponse(gen(), media_type="text/event-stream")
+
+        final = None
+        try:
+            async for out in engine.generate(prompt, params, rid, lora=lora):
+                final = out
+        except ValueError as e:
+            return JSONResponse(
+                {"type": "error",
+                 "error": {"type": "invalid_request_error",
+                           "message": str(e)}},
+                status_code=400,
+            )
+        comp = final.outputs[0]
+        state.num_prompt_tokens += len(final.prompt_token_ids)
+        state.num_generation_tokens += len(comp.token_ids)
+        stop_reason = ("max_tokens" if comp.finish_reason == "length"
+                       else "end_turn")
+        return {
+            "id": rid,
+            "type": "message",
+            "role": "assistant",
+            "model": req.model,
+            "content": [{"type": "text", "text": comp.text}],
+            "stop_reason": stop_reason,
+            "stop_sequence": (comp.stop_reason
+                              if isinstance(comp.stop_reason, str) else None),
+            "usage": {
+                "input_tokens": len(final.prompt_token_ids),
+                "output_tokens": len(comp.token_ids),
+            },
+        }
+
+    return router

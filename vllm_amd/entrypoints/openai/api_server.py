@@ -72,6 +72,10 @@ def build_app(state: ServerState) -> FastAPI:
     app = FastAPI(title="vllm_amd OpenAI-compatible server")
     engine = state.engine
 
+    from vllm_amd.entrypoints.anthropic_api import build_anthropic_router
+
+    app.include_router(build_anthropic_router(state))
+
     @app.get("/health")
     async def health() -> Response:
         return Response(status_code=200)
