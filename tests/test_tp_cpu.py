@@ -13,7 +13,8 @@ import os
 import pytest
 
 
-def _tp_worker(rank: int, world: int, port: int, model: str, q):
+def _tp_worker(rank: int, world: int, port: int, model: str, q,
+               ep: bool = False):
     os.environ["RANK"] = str(rank)
     os.environ["LOCAL_RANK"] = str(rank)
     os.environ["WORLD_SIZE"] = str(world)
@@ -39,6 +40,7 @@ def _tp_worker(rank: int, world: int, port: int, model: str, q):
             scheduler_config=SchedulerConfig(max_num_batched_tokens=256,
                                              max_num_seqs=4),
             parallel_config=ParallelConfig(tensor_parallel_size=world,
+                                           enable_expert_parallel=ep,
                                            distributed_backend="gloo"),
             device_config=DeviceConfig(device="cpu"),
         )
@@ -75,13 +77,16 @@ def _tp_worker(rank: int, world: int, port: int, model: str, q):
         q.put(("err", f"{e}\n{traceback.format_exc()}"))
 
 
-@pytest.mark.parametrize("model,port", [("tiny-llama", 29611),
-                                        ("tiny-mixtral", 29613)])
-def test_tp2_spmd_cpu(model, port):
+@pytest.mark.parametrize("model,port,ep", [
+    ("tiny-llama", 29611, False),
+    ("tiny-mixtral", 29613, False),
+    ("tiny-mixtral", 29615, True),   # expert parallelism
+])
+def test_tp2_spmd_cpu(model, port, ep):
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
     procs = [
-        ctx.Process(target=_tp_worker, args=(r, 2, port, model, q))
+        ctx.Process(target=_tp_worker, args=(r, 2, port, model, q, ep))
         for r in range(2)
     ]
     for p in procs:

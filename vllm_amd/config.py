@@ -341,6 +341,7 @@ class ModelConfig:
     enforce_eager: bool = False
     # name -> local PEFT checkpoint dir; ids are 1-based in listed order.
     lora_modules: Optional[dict] = None
+    enable_expert_parallel: bool = False
     spec: ModelSpec = None  # type: ignore[assignment]
 
     def __post_init__(self) -> None:
@@ -401,6 +402,9 @@ class SchedulerConfig:
 class ParallelConfig:
     tensor_parallel_size: int = 1
     pipeline_parallel_size: int = 1
+    # Shard MoE experts across the world (EP) instead of TP-sharding each
+    # expert's intermediate dim; tokens stay replicated (AgRs combine).
+    enable_expert_parallel: bool = False
     data_parallel_size: int = 1
     enable_expert_parallel: bool = False
     # Filled from env (RANK/LOCAL_RANK/WORLD_SIZE) when launched by torchrun.
@@ -438,3 +442,6 @@ class EngineConfig:
             raise ValueError(
                 f"num_heads={spec.num_heads} not divisible by tp={tp}"
             )
+        # Models read the EP flag from ModelConfig at construction.
+        self.model_config.enable_expert_parallel = \
+            self.parallel_config.enable_expert_parallel

@@ -45,6 +45,7 @@ class EngineArgs:
     async_scheduling: bool = True
 
     tensor_parallel_size: int = 1
+    enable_expert_parallel: bool = False
     device: str = "auto"
 
     @staticmethod
@@ -85,6 +86,7 @@ class EngineArgs:
                             dest="async_scheduling", action="store_false")
         parser.add_argument("--tensor-parallel-size", "-tp", type=int,
                             default=1)
+        parser.add_argument("--enable-expert-parallel", action="store_true")
         parser.add_argument("--device", type=str, default="auto")
         return parser
 
@@ -106,6 +108,7 @@ class EngineArgs:
         world_size = int(os.environ.get("WORLD_SIZE", "1"))
         pc = ParallelConfig(
             tensor_parallel_size=self.tensor_parallel_size,
+            enable_expert_parallel=self.enable_expert_parallel,
             rank=int(os.environ.get("RANK", "0")),
             local_rank=int(os.environ.get("LOCAL_RANK", "0")),
             world_size=max(world_size, self.tensor_parallel_size),

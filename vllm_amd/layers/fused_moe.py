@@ -1,10 +1,15 @@
 """Fused Mixture-of-Experts layer (role of the reference's
 vllm/model_executor/layers/fused_moe/layer.py FusedMoE).
 
-TP sharding: each rank holds the full expert set with the intermediate
-dim sharded 1/tp (gate and up halves sharded separately inside w13);
-forward ends in one RCCL all-reduce over xGMI. EP (experts sharded
-across ranks + all-to-all dispatch) layers on later (all2all.py role).
+Two sharding modes over the same forward:
+
+- TP (default): every rank holds all experts with the intermediate dim
+  sharded 1/tp (gate/up halves sharded separately inside w13).
+- EP (enable_expert_parallel): experts sharded across ranks at full
+  intermediate width; tokens are replicated in SPMD TP serving, so each
+  rank computes its local experts for every token and the combine is
+  the same RCCL all-reduce (the naive AgRs manager of the reference's
+  all2all.py — DeepEP-style dispatch/combine lands later).
 """
 
 from __future__ import annotations
@@ -32,28 +37,47 @@ class FusedMoE(nn.Module):
         renormalize: bool = True,
         activation: str = "silu",
         dtype: torch.dtype = None,
+        enable_expert_parallel: bool = False,
     ):
         super().__init__()
+        from vllm_amd.config import EngineConfig  # noqa: F401 (doc only)
+        from vllm_amd.parallel.state import get_ep_group
+
         tp = get_tp_world_size()
-        assert intermediate_size % tp == 0, (
-            f"moe intermediate {intermediate_size} not divisible by tp {tp}"
-        )
+        self.ep_size = 1
+        self.ep_rank = 0
+        if enable_expert_parallel and tp > 1:
+            ep = get_ep_group()
+            self.ep_size = ep.world_size
+            self.ep_rank = ep.rank_in_group
+        if self.ep_size > 1:
+            assert num_experts % self.ep_size == 0, (
+                f"{num_experts} experts not divisible by ep {self.ep_size}")
+            self.num_local_experts = num_experts // self.ep_size
+            self.expert_lo = self.ep_rank * self.num_local_experts
+            self.i_shard = intermediate_size  # full width per local expert
+        else:
+            assert intermediate_size % tp == 0, (
+                f"moe intermediate {intermediate_size} not divisible by "
+                f"tp {tp}")
+            self.num_local_experts = num_experts
+            self.expert_lo = 0
+            self.i_shard = intermediate_size // tp
         self.num_experts = num_experts
         self.top_k = top_k
         self.renormalize = renormalize
         self.activation = activation
-        self.i_shard = intermediate_size // tp
         self.gate = ReplicatedLinear(hidden_size, num_experts, bias=False,
                                      dtype=dtype)
-        # [E, 2*I/tp, H]: gate rows then up rows, both sharded.
+        # TP: [E, 2*I/tp, H]; EP: [E/ep, 2*I, H].
         self.w13 = nn.Parameter(
-            torch.empty(num_experts, 2 * self.i_shard, hidden_size,
-                        dtype=dtype),
+            torch.empty(self.num_local_experts, 2 * self.i_shard,
+                        hidden_size, dtype=dtype),
             requires_grad=False,
         )
-        # [E, H, I/tp]
         self.w2 = nn.Parameter(
-            torch.empty(num_experts, hidden_size, self.i_shard, dtype=dtype),
+            torch.empty(self.num_local_experts, hidden_size, self.i_shard,
+                        dtype=dtype),
             requires_grad=False,
         )
 
@@ -62,19 +86,36 @@ class FusedMoE(nn.Module):
         topk_weights, topk_ids = ops.topk_softmax(
             router_logits, self.top_k, renormalize=self.renormalize
         )
-        out = ops.fused_moe(
-            hidden, self.w13, self.w2, topk_weights, topk_ids,
-            activation=self.activation,
-        )
+        out = self.run_experts(hidden, topk_weights, topk_ids)
         if get_tp_world_size() > 1:
             out = tensor_model_parallel_all_reduce(out)
         return out
 
+    def run_experts(self, hidden, topk_weights, topk_ids) -> torch.Tensor:
+        """Local expert computation (routing masked to this rank's expert
+        range in EP mode; the caller's all-reduce completes the combine)."""
+        if self.ep_size > 1:
+            lo = self.expert_lo
+            hi = lo + self.num_local_experts
+            mask = (topk_ids >= lo) & (topk_ids < hi)
+            topk_ids = topk_ids.masked_fill(~mask, lo) - lo
+            topk_weights = topk_weights * mask
+        return ops.fused_moe(
+            hidden, self.w13, self.w2, topk_weights, topk_ids,
+            activation=self.activation,
+        )
+
     def load_full_weights(self, w1_full, w3_full, w2_full) -> None:
         """Shard full expert weights onto this rank: w1/w3 [E, I, H]
         (gate / up), w2 [E, H, I]."""
-        tp, r = get_tp_world_size(), get_tp_rank()
         i = self.i_shard
+        if self.ep_size > 1:
+            lo, hi = self.expert_lo, self.expert_lo + self.num_local_experts
+            self.w13.data[:, :i] = w1_full[lo:hi]
+            self.w13.data[:, i:] = w3_full[lo:hi]
+            self.w2.data.copy_(w2_full[lo:hi])
+            return
+        tp, r = get_tp_world_size(), get_tp_rank()
         self.w13.data[:, :i] = w1_full[:, r * i:(r + 1) * i]
         self.w13.data[:, i:] = w3_full[:, r * i:(r + 1) * i]
         self.w2.data.copy_(w2_full[:, :, r * i:(r + 1) * i])

@@ -182,7 +182,7 @@ class DeepseekDenseMLP(nn.Module):
 class DeepseekMoE(nn.Module):
     """Routed experts (sigmoid + group-limited top-k) + shared experts."""
 
-    def __init__(self, spec: ModelSpec, dtype):
+    def __init__(self, spec: ModelSpec, dtype, ep: bool = False):
         super().__init__()
         self.spec = spec
         self.moe = FusedMoE(
@@ -192,6 +192,7 @@ class DeepseekMoE(nn.Module):
             intermediate_size=spec.moe_intermediate_size,
             renormalize=spec.norm_topk_prob,
             dtype=dtype,
+            enable_expert_parallel=ep,
         )
         # DeepSeek-V3 aux-loss-free balancing bias (inference: applied to
         # selection only).
@@ -220,8 +221,7 @@ class DeepseekMoE(nn.Module):
             e_score_bias=self.e_score_correction_bias,
             routed_scaling_factor=spec.routed_scaling_factor,
         )
-        out = ops.fused_moe(hidden, self.moe.w13, self.moe.w2, topk_weights,
-                            topk_ids)
+        out = self.moe.run_experts(hidden, topk_weights, topk_ids)
         if self.shared_experts is not None:
             out = out + self.shared_experts(hidden)
         from vllm_amd.parallel.state import (
@@ -233,7 +233,8 @@ class DeepseekMoE(nn.Module):
 
 
 class DeepseekDecoderLayer(nn.Module):
-    def __init__(self, spec: ModelSpec, layer_idx: int, dtype, max_position):
+    def __init__(self, spec: ModelSpec, layer_idx: int, dtype, max_position,
+                 ep: bool = False):
         super().__init__()
         self.self_attn = DeepseekMLAAttention(spec, layer_idx, dtype,
                                               max_position)
@@ -241,7 +242,7 @@ class DeepseekDecoderLayer(nn.Module):
             self.mlp = DeepseekDenseMLP(spec.hidden_size,
                                         spec.intermediate_size, dtype)
         else:
-            self.mlp = DeepseekMoE(spec, dtype)
+            self.mlp = DeepseekMoE(spec, dtype, ep=ep)
         self.input_layernorm = RMSNorm(spec.hidden_size, spec.rms_norm_eps,
                                        dtype=dtype)
         self.post_attention_layernorm = RMSNorm(
@@ -269,7 +270,8 @@ class DeepseekModel(nn.Module):
             spec.vocab_size, spec.hidden_size, dtype=dtype
         )
         self.layers = nn.ModuleList([
-            DeepseekDecoderLayer(spec, i, dtype, config.max_model_len)
+            DeepseekDecoderLayer(spec, i, dtype, config.max_model_len,
+                                 ep=config.enable_expert_parallel)
             for i in range(spec.num_layers)
         ])
         self.norm = RMSNorm(spec.hidden_size, spec.rms_norm_eps, dtype=dtype)

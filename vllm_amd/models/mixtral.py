@@ -54,7 +54,8 @@ class MixtralAttention(nn.Module):
 
 
 class MixtralDecoderLayer(nn.Module):
-    def __init__(self, spec: ModelSpec, layer_idx: int, dtype, max_position):
+    def __init__(self, spec: ModelSpec, layer_idx: int, dtype, max_position,
+                 ep: bool = False):
         super().__init__()
         self.self_attn = MixtralAttention(spec, layer_idx, dtype,
                                           max_position)
@@ -65,6 +66,7 @@ class MixtralDecoderLayer(nn.Module):
             intermediate_size=spec.moe_intermediate_size,
             renormalize=spec.norm_topk_prob,
             dtype=dtype,
+            enable_expert_parallel=ep,
         )
         self.input_layernorm = RMSNorm(spec.hidden_size, spec.rms_norm_eps,
                                        dtype=dtype)
@@ -93,7 +95,8 @@ class MixtralModel(nn.Module):
             spec.vocab_size, spec.hidden_size, dtype=dtype
         )
         self.layers = nn.ModuleList([
-            MixtralDecoderLayer(spec, i, dtype, config.max_model_len)
+            MixtralDecoderLayer(spec, i, dtype, config.max_model_len,
+                                ep=config.enable_expert_parallel)
             for i in range(spec.num_layers)
         ])
         self.norm = RMSNorm(spec.hidden_size, spec.rms_norm_eps, dtype=dtype)
