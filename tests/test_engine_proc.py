@@ -1,0 +1,32 @@
+"""Engine-core process split: the API-side LLMEngine proxies a core
+running in its own process (pickle pipes)."""
+
+from vllm_amd.entrypoints.llm import LLM
+from vllm_amd.sampling_params import SamplingParams
+
+
+def test_multiprocess_engine_core():
+    llm = LLM(model="tiny-llama", dtype="fp32", device="cpu",
+              block_size=16, num_gpu_blocks=128, max_model_len=256,
+              max_num_batched_tokens=256, max_num_seqs=4,
+              multiprocess_engine=True)
+    p = SamplingParams(temperature=0.0, max_tokens=8, ignore_eos=True)
+    prompts = [list(range(5, 30)), [7, 8, 9]]
+    outs = llm.generate(prompts, p)
+    outs2 = llm.generate(prompts, p)
+    llm.shutdown()
+    for a, b in zip(outs, outs2):
+        assert len(a.outputs[0].token_ids) == 8
+        assert a.outputs[0].token_ids == b.outputs[0].token_ids
+
+
+def test_multiprocess_engine_guided_and_stops():
+    llm = LLM(model="tiny-llama", dtype="fp32", device="cpu",
+              block_size=16, num_gpu_blocks=128, max_model_len=256,
+              max_num_batched_tokens=256, max_num_seqs=4,
+              multiprocess_engine=True)
+    g = SamplingParams(temperature=0.0, max_tokens=8,
+                       guided_choice=["yes", "no"])
+    out = llm.generate([[1, 2, 3]], g)[0]
+    llm.shutdown()
+    assert out.outputs[0].finish_reason == "stop"

@@ -405,8 +405,10 @@ class ParallelConfig:
     # Shard MoE experts across the world (EP) instead of TP-sharding each
     # expert's intermediate dim; tokens stay replicated (AgRs combine).
     enable_expert_parallel: bool = False
+    # Run the engine core (scheduler + executor) in its own process; the
+    # API process only tokenizes/detokenizes (reference EngineCoreProc).
+    multiprocess_engine: bool = False
     data_parallel_size: int = 1
-    enable_expert_parallel: bool = False
     # Filled from env (RANK/LOCAL_RANK/WORLD_SIZE) when launched by torchrun.
     rank: int = 0
     local_rank: int = 0

@@ -46,6 +46,7 @@ class EngineArgs:
 
     tensor_parallel_size: int = 1
     enable_expert_parallel: bool = False
+    multiprocess_engine: bool = False
     device: str = "auto"
 
     @staticmethod
@@ -87,6 +88,8 @@ class EngineArgs:
         parser.add_argument("--tensor-parallel-size", "-tp", type=int,
                             default=1)
         parser.add_argument("--enable-expert-parallel", action="store_true")
+        parser.add_argument("--multiprocess-engine", action="store_true",
+                            help="run the engine core in its own process")
         parser.add_argument("--device", type=str, default="auto")
         return parser
 
@@ -109,6 +112,7 @@ class EngineArgs:
         pc = ParallelConfig(
             tensor_parallel_size=self.tensor_parallel_size,
             enable_expert_parallel=self.enable_expert_parallel,
+            multiprocess_engine=self.multiprocess_engine,
             rank=int(os.environ.get("RANK", "0")),
             local_rank=int(os.environ.get("LOCAL_RANK", "0")),
             world_size=max(world_size, self.tensor_parallel_size),

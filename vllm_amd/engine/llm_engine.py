@@ -151,7 +151,12 @@ class LLMEngine:
 
     def __init__(self, config: EngineConfig):
         self.config = config
-        self.engine_core = EngineCore(config)
+        if config.parallel_config.multiprocess_engine:
+            from vllm_amd.engine.core_client import EngineCoreClient
+
+            self.engine_core = EngineCoreClient(config)
+        else:
+            self.engine_core = EngineCore(config)
         self.tokenizer = TokenizerWrapper(config.model_config.tokenizer)
         self.output_processor = OutputProcessor(self.tokenizer)
         self._request_counter = 0
@@ -223,7 +228,9 @@ class LLMEngine:
         for out in outputs:
             if out.finished:
                 self.output_processor.release(out.request_id)
-                self.engine_core.scheduler.release_request(out.request_id)
+                if self.engine_core.scheduler is not None:
+                    self.engine_core.scheduler.release_request(
+                        out.request_id)
         return outputs
 
     def has_unfinished_requests(self) -> bool:
