@@ -193,3 +193,21 @@ def test_deepseek_mla_cpu_decode():
     assert all(len(o.outputs[0].token_ids) == 6 for o in outs)
     for a, b in zip(outs, outs2):
         assert a.outputs[0].token_ids == b.outputs[0].token_ids
+
+
+def test_qwen3_cpu_decode():
+    """Tiny Qwen3 (qkv bias + per-head q/k RMSNorm) end-to-end on CPU."""
+    from vllm_amd.entrypoints.llm import LLM
+    from vllm_amd.sampling_params import SamplingParams
+
+    llm = LLM(model="tiny-qwen3", dtype="fp32", device="cpu",
+              block_size=16, num_gpu_blocks=128, max_model_len=256,
+              max_num_batched_tokens=256, max_num_seqs=4)
+    prompts = [[1, 2, 3, 4, 5], [9, 8, 7]]
+    p = SamplingParams(temperature=0.0, max_tokens=6, ignore_eos=True)
+    outs = llm.generate(prompts, p)
+    outs2 = llm.generate(prompts, p)
+    llm.shutdown()
+    assert all(len(o.outputs[0].token_ids) == 6 for o in outs)
+    for a, b in zip(outs, outs2):
+        assert a.outputs[0].token_ids == b.outputs[0].token_ids

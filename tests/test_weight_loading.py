@@ -27,6 +27,16 @@ def _export_hf_llama(model, spec, path):
         t[f"{p}.self_attn.v_proj.weight"] = qkv[qs + ks:].clone()
         t[f"{p}.self_attn.o_proj.weight"] = \
             layer.self_attn.o_proj.weight.data.clone()
+        if layer.self_attn.qkv_proj.bias is not None:
+            qkvb = layer.self_attn.qkv_proj.bias.data
+            t[f"{p}.self_attn.q_proj.bias"] = qkvb[:qs].clone()
+            t[f"{p}.self_attn.k_proj.bias"] = qkvb[qs:qs + ks].clone()
+            t[f"{p}.self_attn.v_proj.bias"] = qkvb[qs + ks:].clone()
+        if getattr(layer.self_attn, "q_norm", None) is not None:
+            t[f"{p}.self_attn.q_norm.weight"] = \
+                layer.self_attn.q_norm.weight.data.clone()
+            t[f"{p}.self_attn.k_norm.weight"] = \
+                layer.self_attn.k_norm.weight.data.clone()
         gu = layer.mlp.gate_up_proj.weight.data
         ii = spec.intermediate_size
         t[f"{p}.mlp.gate_proj.weight"] = gu[:ii].clone()
@@ -49,6 +59,27 @@ def test_llama_safetensors_roundtrip(tmp_path):
     _export_hf_llama(model_a, cfg_a.spec, tmp_path)
 
     cfg_b = ModelConfig(model="tiny-llama", dtype="fp32",
+                        load_format="safetensors",
+                        model_path=str(tmp_path))
+    model_b = load_model(cfg_b, torch.device("cpu"))
+
+    for (na, pa), (nb, pb) in zip(model_a.named_parameters(),
+                                  model_b.named_parameters()):
+        assert na == nb
+        assert torch.equal(pa, pb), na
+
+
+def test_qwen3_safetensors_roundtrip(tmp_path):
+    """Qwen3 adds qkv bias + q/k norms to the llama layout."""
+    from vllm_amd.config import ModelConfig
+    from vllm_amd.models.registry import load_model
+
+    cfg_a = ModelConfig(model="tiny-qwen3", dtype="fp32",
+                        load_format="dummy")
+    model_a = load_model(cfg_a, torch.device("cpu"))
+    _export_hf_llama(model_a, cfg_a.spec, tmp_path)
+
+    cfg_b = ModelConfig(model="tiny-qwen3", dtype="fp32",
                         load_format="safetensors",
                         model_path=str(tmp_path))
     model_b = load_model(cfg_b, torch.device("cpu"))

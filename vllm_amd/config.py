@@ -39,6 +39,9 @@ class ModelSpec:
     activation: str = "silu"  # silu | relu | gelu
     use_layernorm: bool = False  # OPT uses LayerNorm, llama uses RMSNorm
     use_bias: bool = False
+    # Qwen-family specifics
+    qkv_bias: bool = False  # Qwen2: bias on q/k/v projections only
+    qk_norm: bool = False  # Qwen3: per-head RMSNorm on q and k pre-RoPE
     # MoE specifics (mixtral / deepseek)
     num_experts: int = 0
     num_experts_per_tok: int = 0
@@ -172,7 +175,55 @@ MODEL_PRESETS: dict[str, ModelSpec] = {
         v_head_dim=128,
         eos_token_id=1,
     ),
+    "qwen3-8b": ModelSpec(
+        name="qwen3-8b",
+        architecture="qwen3",
+        vocab_size=151936,
+        hidden_size=4096,
+        intermediate_size=12288,
+        num_layers=36,
+        num_heads=32,
+        num_kv_heads=8,
+        head_dim=128,
+        rope_theta=1000000.0,
+        max_position_embeddings=32768,
+        rms_norm_eps=1e-6,
+        qk_norm=True,
+        eos_token_id=151645,
+    ),
+    "qwen2.5-7b": ModelSpec(
+        name="qwen2.5-7b",
+        architecture="qwen2",
+        vocab_size=152064,
+        hidden_size=3584,
+        intermediate_size=18944,
+        num_layers=28,
+        num_heads=28,
+        num_kv_heads=4,
+        head_dim=128,
+        rope_theta=1000000.0,
+        max_position_embeddings=32768,
+        rms_norm_eps=1e-6,
+        qkv_bias=True,
+        eos_token_id=151645,
+    ),
     # Tiny models for tests.
+    "tiny-qwen3": ModelSpec(
+        name="tiny-qwen3",
+        architecture="qwen3",
+        vocab_size=1024,
+        hidden_size=128,
+        intermediate_size=256,
+        num_layers=2,
+        num_heads=4,
+        num_kv_heads=2,
+        head_dim=32,
+        rope_theta=10000.0,
+        max_position_embeddings=2048,
+        qk_norm=True,
+        qkv_bias=True,  # exercises the Qwen2 bias path too
+        eos_token_id=2,
+    ),
     "tiny-llama": ModelSpec(
         name="tiny-llama",
         architecture="llama",
@@ -280,6 +331,10 @@ def _spec_from_hf_config(path: str) -> ModelSpec:
         architecture = "mixtral"
     elif "deepseek" in arch:
         architecture = "deepseek"
+    elif "qwen3" in arch:
+        architecture = "qwen3"
+    elif "qwen2" in arch:
+        architecture = "qwen2"
     else:
         architecture = "llama"
     hidden = hf["hidden_size"]
@@ -304,6 +359,8 @@ def _spec_from_hf_config(path: str) -> ModelSpec:
         moe_intermediate_size=hf.get("moe_intermediate_size", 0) or 0,
         eos_token_id=hf.get("eos_token_id", 2) or 2,
         bos_token_id=hf.get("bos_token_id", 1) or 1,
+        qkv_bias=architecture == "qwen2",
+        qk_norm=architecture == "qwen3",
     )
 
 

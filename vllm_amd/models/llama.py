@@ -54,7 +54,7 @@ class LlamaAttention(nn.Module):
             spec.head_dim,
             spec.num_heads,
             spec.num_kv_heads,
-            bias=spec.use_bias,
+            bias=spec.use_bias or spec.qkv_bias,
             dtype=dtype,
         )
         self.num_heads = self.qkv_proj.num_heads
@@ -71,6 +71,11 @@ class LlamaAttention(nn.Module):
             theta=spec.rope_theta,
             rope_scaling=spec.rope_scaling,
         )
+        # Qwen3-style per-head q/k RMSNorm (reference qwen3.py:60-61).
+        self.q_norm = (RMSNorm(spec.head_dim, spec.rms_norm_eps, dtype=dtype)
+                       if spec.qk_norm else None)
+        self.k_norm = (RMSNorm(spec.head_dim, spec.rms_norm_eps, dtype=dtype)
+                       if spec.qk_norm else None)
         self.attn = Attention(
             self.num_heads,
             spec.head_dim,
@@ -85,6 +90,9 @@ class LlamaAttention(nn.Module):
         q, k, v = self.qkv_proj.split_qkv(qkv)
         q = q.view(T, self.num_heads, self.head_dim)
         k = k.view(T, self.num_kv_heads, self.head_dim)
+        if self.q_norm is not None:
+            q = self.q_norm(q.reshape(-1, self.head_dim)).view(q.shape)
+            k = self.k_norm(k.reshape(-1, self.head_dim)).view(k.shape)
         self.rotary_emb(positions, q, k)
         out = self.attn(q, k, v)
         return self.o_proj(out)

@@ -10,7 +10,9 @@ from vllm_amd.config import ModelConfig
 
 
 def get_model_class(architecture: str):
-    if architecture == "llama":
+    if architecture in ("llama", "qwen2", "qwen3"):
+        # Qwen2/Qwen3 are llama-structured (reference qwen2.py/qwen3.py);
+        # they differ only by spec knobs (qkv_bias, qk_norm).
         from vllm_amd.models.llama import LlamaForCausalLM
 
         return LlamaForCausalLM

@@ -89,6 +89,10 @@ def load_safetensors_weights(model: torch.nn.Module, config) -> None:
                         d.get("q_proj.bias"), d.get("k_proj.bias"),
                         d.get("v_proj.bias"),
                     )
+            elif "q_norm" in name:
+                attn.q_norm.weight.data.copy_(w)
+            elif "k_norm" in name:
+                attn.k_norm.weight.data.copy_(w)
             elif "o_proj" in name or "out_proj" in name:
                 if name.endswith("bias"):
                     attn.o_proj.load_bias(w)
