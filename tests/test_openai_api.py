@@ -219,3 +219,35 @@ def test_anthropic_messages_stream(client):
                   if ln.startswith("event: ")]
     assert events[0] == "message_start"
     assert "message_stop" in events
+
+
+def test_chat_response_format_json_schema(client):
+    r = client.post("/v1/chat/completions", json={
+        "model": "tiny-llama",
+        "messages": [{"role": "user", "content": "Emit JSON."}],
+        "max_tokens": 60,
+        "temperature": 0.0,
+        "logit_bias": {"257": 50.0, "125": 20.0, "34": 10.0},
+        "response_format": {"type": "json_schema", "json_schema": {
+            "name": "out", "schema": {"type": "object", "properties": {
+                "ok": {"type": "boolean"},
+                "kind": {"enum": ["a", "b"]}}}}},
+    })
+    assert r.status_code == 200, r.text
+    import json as _json
+    doc = _json.loads(r.json()["choices"][0]["message"]["content"])
+    assert isinstance(doc["ok"], bool) and doc["kind"] in ("a", "b")
+
+
+def test_completions_guided_regex(client):
+    import re as _re
+    pattern = r"[ab]{3}-[0-9]{2}"
+    r = client.post("/v1/completions", json={
+        "model": "tiny-llama",
+        "prompt": "match: ",
+        "max_tokens": 20,
+        "temperature": 0.0,
+        "guided_regex": pattern,
+    })
+    assert r.status_code == 200, r.text
+    assert _re.fullmatch(pattern, r.json()["choices"][0]["text"])

@@ -207,6 +207,19 @@ class LLMEngine:
             request.grammar = compile_choice_grammar(
                 params.guided_choice, self.tokenizer, eos
             )
+        elif params.guided_regex or params.guided_json is not None \
+                or params.guided_json_object:
+            from vllm_amd.guided_json import any_json_regex, schema_to_regex
+            from vllm_amd.guided_regex import RegexGrammar
+
+            if params.guided_regex:
+                pattern = params.guided_regex
+            elif params.guided_json is not None \
+                    and params.guided_json is not True:
+                pattern = schema_to_regex(params.guided_json)
+            else:
+                pattern = any_json_regex()
+            request.grammar = RegexGrammar(pattern, self.tokenizer, eos)
         self.engine_core.add_request(request)
         self.output_processor.add_request(
             request_id, prompt_text, prompt_token_ids, params

@@ -44,6 +44,8 @@ class CompletionRequest(BaseModel):
     ignore_eos: bool = False
     skip_special_tokens: bool = True
     guided_choice: Optional[list[str]] = None
+    guided_regex: Optional[str] = None
+    guided_json: Optional[Union[dict, str]] = None
 
     def to_sampling_params(self, stream: bool) -> SamplingParams:
         logit_bias = (
@@ -69,6 +71,8 @@ class CompletionRequest(BaseModel):
             logit_bias=logit_bias,
             skip_special_tokens=self.skip_special_tokens,
             guided_choice=self.guided_choice,
+            guided_regex=self.guided_regex,
+            guided_json=self.guided_json,
             output_kind=(RequestOutputKind.DELTA if stream
                          else RequestOutputKind.FINAL_ONLY),
         )
@@ -117,6 +121,9 @@ class ChatCompletionRequest(BaseModel):
     ignore_eos: bool = False
     skip_special_tokens: bool = True
     guided_choice: Optional[list[str]] = None
+    guided_regex: Optional[str] = None
+    guided_json: Optional[Union[dict, str]] = None
+    response_format: Optional[dict[str, Any]] = None
     add_generation_prompt: bool = True
 
     def to_sampling_params(self, stream: bool,
@@ -128,6 +135,17 @@ class ChatCompletionRequest(BaseModel):
             if self.logit_bias else None
         )
         n_logprobs = (self.top_logprobs or 1) if self.logprobs else None
+        # response_format: json_object -> any-JSON grammar; json_schema ->
+        # schema grammar (OpenAI structured outputs shape).
+        guided_json = self.guided_json
+        json_object = False
+        if self.response_format:
+            kind = self.response_format.get("type")
+            if kind == "json_object":
+                json_object = True
+            elif kind == "json_schema":
+                js = self.response_format.get("json_schema") or {}
+                guided_json = js.get("schema") or js
         return SamplingParams(
             n=1,
             presence_penalty=self.presence_penalty,
@@ -147,6 +165,9 @@ class ChatCompletionRequest(BaseModel):
             logit_bias=logit_bias,
             skip_special_tokens=self.skip_special_tokens,
             guided_choice=self.guided_choice,
+            guided_regex=self.guided_regex,
+            guided_json=guided_json,
+            guided_json_object=json_object,
             output_kind=(RequestOutputKind.DELTA if stream
                          else RequestOutputKind.FINAL_ONLY),
         )

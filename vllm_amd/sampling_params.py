@@ -58,6 +58,12 @@ class SamplingParams:
     # Structured output: generation constrained to one of these strings
     # (compiled to a token trie by the engine; see structured_output.py).
     guided_choice: Optional[list[str]] = None
+    # Structured output: regex / JSON-schema constraints (compiled to a
+    # DFA with per-state token masks; see guided_regex.py / guided_json.py).
+    guided_regex: Optional[str] = None
+    guided_json: Optional[object] = None  # dict schema, JSON string, or True
+    # response_format={"type": "json_object"}: any JSON object.
+    guided_json_object: bool = False
     output_kind: RequestOutputKind = RequestOutputKind.CUMULATIVE
 
     def __post_init__(self) -> None:
