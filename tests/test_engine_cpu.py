@@ -211,3 +211,55 @@ def test_qwen3_cpu_decode():
     assert all(len(o.outputs[0].token_ids) == 6 for o in outs)
     for a, b in zip(outs, outs2):
         assert a.outputs[0].token_ids == b.outputs[0].token_ids
+
+
+def test_sliding_window_cpu():
+    """Sliding-window attention (Mistral/Gemma3 pattern): matches the
+    full-attention model while context < window, diverges beyond it."""
+    from vllm_amd.entrypoints.llm import LLM
+    from vllm_amd.sampling_params import SamplingParams
+
+    kw = dict(dtype="fp32", device="cpu", block_size=16,
+              num_gpu_blocks=128, max_model_len=256,
+              max_num_batched_tokens=256, max_num_seqs=4)
+    p = SamplingParams(temperature=0.0, max_tokens=4, ignore_eos=True)
+    short = [[5, 6, 7]]  # ctx stays under window=8
+    long = [[(i * 7) % 900 + 3 for i in range(40)]]
+
+    win = LLM(model="tiny-mistral", **kw)
+    w_short = win.generate(short, p)[0].outputs[0].token_ids
+    w_long = win.generate(long, p)[0].outputs[0].token_ids
+    w_long2 = win.generate(long, p)[0].outputs[0].token_ids
+    win.shutdown()
+
+    full = LLM(model="tiny-llama", **kw)  # same shape/seed, no window
+    f_short = full.generate(short, p)[0].outputs[0].token_ids
+    f_long = full.generate(long, p)[0].outputs[0].token_ids
+    full.shutdown()
+
+    assert w_short == f_short
+    assert w_long == w_long2  # deterministic
+    assert len(w_long) == 4 and len(f_long) == 4
+
+
+def test_sliding_window_op_bites():
+    """The window mask changes attention output once ctx > window."""
+    import torch
+    from vllm_amd.ops import _torch_ref as ref
+
+    torch.manual_seed(0)
+    ctx, bs, H, KV, D = 24, 16, 4, 2, 32
+    nblocks = (ctx + bs - 1) // bs
+    kv = torch.randn(nblocks, 2, KV, bs, D)
+    bt = torch.arange(nblocks, dtype=torch.int32).unsqueeze(0)
+    q = torch.randn(1, H, D)
+    qs = torch.tensor([0, 1], dtype=torch.int32)
+    sl = torch.tensor([ctx], dtype=torch.int32)
+    full = ref.attention_unified(q, kv, bt, qs, sl, 0.18, num_decodes=1)
+    win = ref.attention_unified(q, kv, bt, qs, sl, 0.18, num_decodes=1,
+                                sliding_window=8)
+    assert not torch.allclose(full, win)
+    # Window >= ctx is a no-op.
+    wide = ref.attention_unified(q, kv, bt, qs, sl, 0.18, num_decodes=1,
+                                 sliding_window=64)
+    assert torch.allclose(full, wide)

@@ -295,3 +295,37 @@ def test_fp8_kv_cache_attention():
     expect2 = ref.attention_unified(q2, ref_cache, bt, qsl2, sl, scale,
                                     num_decodes=0)
     assert_close(out2, expect2, atol=6e-2, rtol=6e-2)
+
+
+@pytest.mark.parametrize("window", [8, 64, 200])
+def test_decode_attention_sliding_window(window):
+    hip = _hip()
+    Hkv, group = 2, 4
+    Hq = Hkv * group
+    n, ctx = 4, 150
+    q, cache, bt, qsl, sl = _make_paged(n, [1] * n, [ctx] * n, Hq, Hkv)
+    scale = 1.0 / 128**0.5
+    out = hip.attention_unified(q, cache, bt, qsl, sl, scale,
+                                num_decodes=n, max_seq_len=ctx,
+                                max_query_len=1, sliding_window=window)
+    expect = ref.attention_unified(q, cache, bt, qsl, sl, scale,
+                                   num_decodes=n, sliding_window=window)
+    assert_close(out, expect, msg=f"window={window}")
+
+
+@pytest.mark.parametrize("window", [16, 128])
+def test_prefill_attention_sliding_window(window):
+    hip = _hip()
+    Hkv, group = 2, 4
+    Hq = Hkv * group
+    q_lens, ctx_lens = [17, 70], [100, 300]
+    n = len(q_lens)
+    q, cache, bt, qsl, sl = _make_paged(n, q_lens, ctx_lens, Hq, Hkv)
+    scale = 1.0 / 128**0.5
+    out = hip.attention_unified(
+        q, cache, bt, qsl, sl, scale, num_decodes=0,
+        max_seq_len=max(ctx_lens), max_query_len=max(q_lens),
+        sliding_window=window)
+    expect = ref.attention_unified(q, cache, bt, qsl, sl, scale,
+                                   num_decodes=0, sliding_window=window)
+    assert_close(out, expect, msg=f"window={window}")

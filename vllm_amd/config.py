@@ -42,6 +42,11 @@ class ModelSpec:
     # Qwen-family specifics
     qkv_bias: bool = False  # Qwen2: bias on q/k/v projections only
     qk_norm: bool = False  # Qwen3: per-head RMSNorm on q and k pre-RoPE
+    # Sliding-window attention (Mistral/Gemma). 0 = full attention.
+    sliding_window: int = 0
+    # Gemma3-style layer pattern: every Nth layer is global, the rest
+    # use the sliding window. 0 = all layers windowed (Mistral).
+    global_attn_every_n_layers: int = 0
     # MoE specifics (mixtral / deepseek)
     num_experts: int = 0
     num_experts_per_tok: int = 0
@@ -175,6 +180,21 @@ MODEL_PRESETS: dict[str, ModelSpec] = {
         v_head_dim=128,
         eos_token_id=1,
     ),
+    "mistral-7b": ModelSpec(
+        name="mistral-7b",
+        architecture="llama",
+        vocab_size=32000,
+        hidden_size=4096,
+        intermediate_size=14336,
+        num_layers=32,
+        num_heads=32,
+        num_kv_heads=8,
+        head_dim=128,
+        rope_theta=10000.0,
+        max_position_embeddings=32768,
+        sliding_window=4096,
+        eos_token_id=2,
+    ),
     "qwen3-8b": ModelSpec(
         name="qwen3-8b",
         architecture="qwen3",
@@ -208,6 +228,25 @@ MODEL_PRESETS: dict[str, ModelSpec] = {
         eos_token_id=151645,
     ),
     # Tiny models for tests.
+    # Same shape as tiny-llama but with a sliding window (+1 global layer
+    # in the Gemma3 pattern) — window/full equivalence tests rely on the
+    # shared shape/seed.
+    "tiny-mistral": ModelSpec(
+        name="tiny-mistral",
+        architecture="llama",
+        vocab_size=1024,
+        hidden_size=128,
+        intermediate_size=256,
+        num_layers=2,
+        num_heads=4,
+        num_kv_heads=2,
+        head_dim=32,
+        rope_theta=10000.0,
+        max_position_embeddings=2048,
+        sliding_window=8,
+        global_attn_every_n_layers=2,
+        eos_token_id=2,
+    ),
     "tiny-qwen3": ModelSpec(
         name="tiny-qwen3",
         architecture="qwen3",
@@ -361,6 +400,7 @@ def _spec_from_hf_config(path: str) -> ModelSpec:
         bos_token_id=hf.get("bos_token_id", 1) or 1,
         qkv_bias=architecture == "qwen2",
         qk_norm=architecture == "qwen3",
+        sliding_window=hf.get("sliding_window") or 0,
     )
 
 

@@ -76,12 +76,19 @@ class LlamaAttention(nn.Module):
                        if spec.qk_norm else None)
         self.k_norm = (RMSNorm(spec.head_dim, spec.rms_norm_eps, dtype=dtype)
                        if spec.qk_norm else None)
+        # Per-layer window: all layers windowed (Mistral) unless the spec
+        # marks every Nth layer global (Gemma3 pattern).
+        window = spec.sliding_window
+        if window and spec.global_attn_every_n_layers:
+            if (layer_idx + 1) % spec.global_attn_every_n_layers == 0:
+                window = 0
         self.attn = Attention(
             self.num_heads,
             spec.head_dim,
             scale=spec.head_dim**-0.5,
             num_kv_heads=self.num_kv_heads,
             layer_idx=layer_idx,
+            sliding_window=window,
         )
 
     def forward(self, positions, hidden):
