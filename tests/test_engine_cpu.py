@@ -263,3 +263,22 @@ def test_sliding_window_op_bites():
     wide = ref.attention_unified(q, kv, bt, qs, sl, 0.18, num_decodes=1,
                                  sliding_window=64)
     assert torch.allclose(full, wide)
+
+
+def test_gemma3_cpu_decode():
+    """Tiny Gemma3 (sandwich norms, GeGLU, scaled embeddings, qk-norm,
+    5:1 local/global sliding-window pattern) end-to-end on CPU."""
+    from vllm_amd.entrypoints.llm import LLM
+    from vllm_amd.sampling_params import SamplingParams
+
+    llm = LLM(model="tiny-gemma3", dtype="fp32", device="cpu",
+              block_size=16, num_gpu_blocks=128, max_model_len=256,
+              max_num_batched_tokens=64, max_num_seqs=4)
+    prompts = [[(i * 13 + j) % 900 + 3 for j in range(30)] for i in range(2)]
+    p = SamplingParams(temperature=0.0, max_tokens=6, ignore_eos=True)
+    outs = llm.generate(prompts, p)
+    outs2 = llm.generate(prompts, p)
+    llm.shutdown()
+    assert all(len(o.outputs[0].token_ids) == 6 for o in outs)
+    for a, b in zip(outs, outs2):
+        assert a.outputs[0].token_ids == b.outputs[0].token_ids

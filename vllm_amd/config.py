@@ -47,6 +47,11 @@ class ModelSpec:
     # Gemma3-style layer pattern: every Nth layer is global, the rest
     # use the sliding window. 0 = all layers windowed (Mistral).
     global_attn_every_n_layers: int = 0
+    # Gemma-family specifics
+    scale_embeddings: bool = False  # multiply embeddings by sqrt(hidden)
+    rmsnorm_unit_offset: bool = False  # norm gain is (1 + w)
+    rope_local_theta: float = 0.0  # rope theta for windowed layers (0=same)
+    query_pre_attn_scalar: int = 0  # attn scale = this**-0.5 (0=head_dim)
     # MoE specifics (mixtral / deepseek)
     num_experts: int = 0
     num_experts_per_tok: int = 0
@@ -195,6 +200,30 @@ MODEL_PRESETS: dict[str, ModelSpec] = {
         sliding_window=4096,
         eos_token_id=2,
     ),
+    "gemma3-12b": ModelSpec(
+        name="gemma3-12b",
+        architecture="gemma3",
+        vocab_size=262208,
+        hidden_size=3840,
+        intermediate_size=15360,
+        num_layers=48,
+        num_heads=16,
+        num_kv_heads=8,
+        head_dim=256,
+        rope_theta=1000000.0,
+        rope_local_theta=10000.0,
+        max_position_embeddings=32768,
+        rms_norm_eps=1e-6,
+        sliding_window=1024,
+        global_attn_every_n_layers=6,
+        scale_embeddings=True,
+        rmsnorm_unit_offset=True,
+        qk_norm=True,
+        query_pre_attn_scalar=256,
+        tie_word_embeddings=True,
+        activation="gelu",
+        eos_token_id=1,
+    ),
     "qwen3-8b": ModelSpec(
         name="qwen3-8b",
         architecture="qwen3",
@@ -245,6 +274,29 @@ MODEL_PRESETS: dict[str, ModelSpec] = {
         max_position_embeddings=2048,
         sliding_window=8,
         global_attn_every_n_layers=2,
+        eos_token_id=2,
+    ),
+    "tiny-gemma3": ModelSpec(
+        name="tiny-gemma3",
+        architecture="gemma3",
+        vocab_size=1024,
+        hidden_size=128,
+        intermediate_size=256,
+        num_layers=3,
+        num_heads=4,
+        num_kv_heads=2,
+        head_dim=32,
+        rope_theta=1000000.0,
+        rope_local_theta=10000.0,
+        max_position_embeddings=2048,
+        sliding_window=8,
+        global_attn_every_n_layers=3,
+        scale_embeddings=True,
+        rmsnorm_unit_offset=True,
+        qk_norm=True,
+        query_pre_attn_scalar=32,
+        tie_word_embeddings=True,
+        activation="gelu",
         eos_token_id=2,
     ),
     "tiny-qwen3": ModelSpec(
@@ -370,12 +422,18 @@ def _spec_from_hf_config(path: str) -> ModelSpec:
         architecture = "mixtral"
     elif "deepseek" in arch:
         architecture = "deepseek"
+    elif "gemma3" in arch:
+        architecture = "gemma3"
     elif "qwen3" in arch:
         architecture = "qwen3"
     elif "qwen2" in arch:
         architecture = "qwen2"
     else:
         architecture = "llama"
+    # Gemma3 nests the text config under text_config in the multimodal
+    # checkpoint layout.
+    if architecture == "gemma3" and "text_config" in hf:
+        hf = {**hf["text_config"], "architectures": hf["architectures"]}
     hidden = hf["hidden_size"]
     heads = hf["num_attention_heads"]
     return ModelSpec(
@@ -399,8 +457,14 @@ def _spec_from_hf_config(path: str) -> ModelSpec:
         eos_token_id=hf.get("eos_token_id", 2) or 2,
         bos_token_id=hf.get("bos_token_id", 1) or 1,
         qkv_bias=architecture == "qwen2",
-        qk_norm=architecture == "qwen3",
+        qk_norm=architecture in ("qwen3", "gemma3"),
         sliding_window=hf.get("sliding_window") or 0,
+        global_attn_every_n_layers=hf.get("sliding_window_pattern", 0) or 0,
+        rope_local_theta=hf.get("rope_local_base_freq", 0.0) or 0.0,
+        query_pre_attn_scalar=hf.get("query_pre_attn_scalar", 0) or 0,
+        scale_embeddings=architecture == "gemma3",
+        rmsnorm_unit_offset=architecture == "gemma3",
+        activation="gelu" if architecture == "gemma3" else "silu",
     )
 
 

@@ -24,6 +24,10 @@ def get_model_class(architecture: str):
         from vllm_amd.models.mixtral import MixtralForCausalLM
 
         return MixtralForCausalLM
+    if architecture == "gemma3":
+        from vllm_amd.models.gemma import GemmaForCausalLM
+
+        return GemmaForCausalLM
     if architecture == "deepseek":
         from vllm_amd.models.deepseek import DeepseekForCausalLM
 

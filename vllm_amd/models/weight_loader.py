@@ -37,6 +37,12 @@ def load_safetensors_weights(model: torch.nn.Module, config) -> None:
     spec = config.spec
     dtype = config.torch_dtype
 
+    def norm_w(w):
+        # Gemma stores RMSNorm gains zero-centered: effective gain is
+        # (1 + w). Fold the offset into the stored weight so the fused
+        # RMSNorm kernel stays a plain x̂*w.
+        return w + 1 if spec.rmsnorm_unit_offset else w
+
     # Collect q/k/v and gate/up pieces so fused layers load atomically.
     pending: dict[str, dict[str, torch.Tensor]] = {}
     moe_pending: dict[str, dict[str, torch.Tensor]] = {}
@@ -65,7 +71,7 @@ def load_safetensors_weights(model: torch.nn.Module, config) -> None:
                 root.lm_head.load_weight(w)
             continue
         if name == "model.norm.weight":
-            root.model.norm.weight.data.copy_(w)
+            root.model.norm.weight.data.copy_(norm_w(w))
             continue
         if ".layers." not in name:
             logger.warning("unmatched tensor %s", name)
@@ -90,9 +96,9 @@ def load_safetensors_weights(model: torch.nn.Module, config) -> None:
                         d.get("v_proj.bias"),
                     )
             elif "q_norm" in name:
-                attn.q_norm.weight.data.copy_(w)
+                attn.q_norm.weight.data.copy_(norm_w(w))
             elif "k_norm" in name:
-                attn.k_norm.weight.data.copy_(w)
+                attn.k_norm.weight.data.copy_(norm_w(w))
             elif "o_proj" in name or "out_proj" in name:
                 if name.endswith("bias"):
                     attn.o_proj.load_bias(w)
@@ -129,10 +135,14 @@ def load_safetensors_weights(model: torch.nn.Module, config) -> None:
                 moe_pending.pop(lkey)
             continue
 
-        if "input_layernorm" in name:
-            layer.input_layernorm.weight.data.copy_(w)
+        if "pre_feedforward_layernorm" in name:
+            layer.pre_feedforward_layernorm.weight.data.copy_(norm_w(w))
+        elif "post_feedforward_layernorm" in name:
+            layer.post_feedforward_layernorm.weight.data.copy_(norm_w(w))
+        elif "input_layernorm" in name:
+            layer.input_layernorm.weight.data.copy_(norm_w(w))
         elif "post_attention_layernorm" in name:
-            layer.post_attention_layernorm.weight.data.copy_(w)
+            layer.post_attention_layernorm.weight.data.copy_(norm_w(w))
         else:
             logger.warning("unmatched tensor %s (key %s)", name, key)
 
