@@ -38,26 +38,53 @@ def _error(msg: str, code: int = 400) -> JSONResponse:
     )
 
 
-def apply_chat_template(tokenizer, messages, add_generation_prompt=True):
+def apply_chat_template(tokenizer, messages, add_generation_prompt=True,
+                        tools=None):
     """HF chat template when a real tokenizer is loaded; otherwise a
-    simple role-tagged fallback (mock tokenizer / no template)."""
+    simple role-tagged fallback (mock tokenizer / no template). `tools`
+    go through the template's `tools=` kwarg when it takes one, else a
+    hermes-style system preamble."""
     hf = getattr(tokenizer, "tokenizer", None)
     if hf is not None and getattr(hf, "chat_template", None):
+        dicts = []
+        for m in messages:
+            d = {"role": m.role, "content": m.text()}
+            if getattr(m, "tool_calls", None):
+                d["tool_calls"] = m.tool_calls
+            if getattr(m, "tool_call_id", None):
+                d["tool_call_id"] = m.tool_call_id
+            dicts.append(d)
+        if tools:
+            try:
+                return hf.apply_chat_template(
+                    dicts, tokenize=False, tools=tools,
+                    add_generation_prompt=add_generation_prompt,
+                )
+            except Exception:  # template has no tools support
+                from vllm_amd.entrypoints.tool_parser import (
+                    render_tools_block)
+                dicts = ([{"role": "system", "content":
+                           render_tools_block(tools)}] + dicts)
         return hf.apply_chat_template(
-            [{"role": m.role, "content": m.text()} for m in messages],
-            tokenize=False,
-            add_generation_prompt=add_generation_prompt,
-        )
-    parts = [f"<|{m.role}|>\n{m.text()}" for m in messages]
+            dicts, tokenize=False,
+            add_generation_prompt=add_generation_prompt)
+    parts = []
+    if tools:
+        from vllm_amd.entrypoints.tool_parser import render_tools_block
+        parts.append(f"<|system|>\n{render_tools_block(tools)}")
+    parts += [f"<|{m.role}|>\n{m.text()}" for m in messages]
     if add_generation_prompt:
         parts.append("<|assistant|>\n")
     return "\n".join(parts)
 
 
 class ServerState:
-    def __init__(self, engine: AsyncLLM, model_name: str):
+    def __init__(self, engine: AsyncLLM, model_name: str,
+                 reasoning_parser: Optional[str] = None):
         self.engine = engine
         self.model_name = model_name
+        # "deepseek_r1" enables <think> splitting into reasoning_content.
+        self.reasoning_parser = reasoning_parser
         self.lora_names = list(
             engine.config.model_config.lora_modules or {})
         self.max_model_len = engine.config.model_config.max_model_len
@@ -199,8 +226,13 @@ def build_app(state: ServerState) -> FastAPI:
     async def chat_completions(req: ChatCompletionRequest, raw: Request):
         if req.n != 1:
             return _error("only n=1 is supported")
+        from vllm_amd.entrypoints import tool_parser as tp
+
+        tools_on = bool(req.tools) and req.tool_choice != "none"
+        named = req.named_tool() if tools_on else None
         prompt = apply_chat_template(
-            engine.tokenizer, req.messages, req.add_generation_prompt
+            engine.tokenizer, req.messages, req.add_generation_prompt,
+            tools=req.tools if tools_on and not named else None,
         )
         default_max = state.max_model_len
         params = req.to_sampling_params(req.stream, default_max)
@@ -217,20 +249,84 @@ def build_app(state: ServerState) -> FastAPI:
                                                     content=""))],
                 )
                 yield f"data: {first.model_dump_json()}\n\n"
+                rparse = (tp.StreamingReasoningParser()
+                          if state.reasoning_parser else None)
+                tparse = (tp.StreamingToolParser()
+                          if tools_on and not named else None)
+                named_id = tp._call_id() if named else None
+                named_first = True
+
+                def deltas(text, finish=None):
+                    """Split one raw text delta through the active
+                    parsers into zero or more DeltaMessages."""
+                    out = []
+                    reasoning = ""
+                    if rparse is not None:
+                        reasoning, text = rparse.feed(text)
+                        if finish is not None:
+                            r2, t2 = rparse.flush()
+                            reasoning += r2
+                            text += t2
+                    if named:
+                        calls = []
+                        if text:
+                            fn = {"arguments": text}
+                            nonlocal named_first
+                            if named_first:
+                                fn["name"] = named
+                                named_first = False
+                                calls = [{"index": 0, "id": named_id,
+                                          "type": "function",
+                                          "function": fn}]
+                            else:
+                                calls = [{"index": 0, "function": fn}]
+                        out.append(DeltaMessage(
+                            reasoning_content=reasoning or None,
+                            tool_calls=calls or None))
+                        return out, bool(calls)
+                    saw = False
+                    if tparse is not None:
+                        text, calls = tparse.feed(text)
+                        if finish is not None:
+                            t2, c2 = tparse.flush()
+                            text += t2
+                            calls += c2
+                        saw = tparse.saw_tool_call
+                        if calls:
+                            out.append(DeltaMessage(tool_calls=calls))
+                    if text or reasoning:
+                        out.append(DeltaMessage(
+                            content=text or None,
+                            reasoning_content=reasoning or None))
+                    return out, saw
                 try:
                     async for out in engine.generate(prompt, params, rid,
                                                      lora=lora):
                         comp = out.outputs[0]
                         state.num_generation_tokens += len(comp.token_ids)
-                        chunk = ChatCompletionStreamResponse(
-                            id=rid, model=req.model,
-                            choices=[ChatStreamChoice(
-                                index=0,
-                                delta=DeltaMessage(content=comp.text),
-                                finish_reason=comp.finish_reason,
-                            )],
-                        )
-                        yield f"data: {chunk.model_dump_json()}\n\n"
+                        msgs, saw = deltas(comp.text, comp.finish_reason)
+                        finish = comp.finish_reason
+                        if finish and (saw or named):
+                            finish = "tool_calls" if finish == "stop" \
+                                else finish
+                        for i, d in enumerate(msgs):
+                            last = comp.finish_reason and i == len(msgs) - 1
+                            chunk = ChatCompletionStreamResponse(
+                                id=rid, model=req.model,
+                                choices=[ChatStreamChoice(
+                                    index=0, delta=d,
+                                    finish_reason=finish if last else None,
+                                )],
+                            )
+                            yield f"data: {chunk.model_dump_json()}\n\n"
+                        if comp.finish_reason and not msgs:
+                            chunk = ChatCompletionStreamResponse(
+                                id=rid, model=req.model,
+                                choices=[ChatStreamChoice(
+                                    index=0, delta=DeltaMessage(),
+                                    finish_reason=finish)],
+                            )
+                            yield f"data: {chunk.model_dump_json()}\n\n"
                     yield "data: [DONE]\n\n"
                 except Exception as e:  # noqa: BLE001
                     err = {"error": {"message": str(e)}}
@@ -251,13 +347,34 @@ def build_app(state: ServerState) -> FastAPI:
             completion_tokens=len(comp.token_ids),
             total_tokens=len(final.prompt_token_ids) + len(comp.token_ids),
         )
+        text = comp.text
+        reasoning = None
+        if state.reasoning_parser:
+            reasoning, text = tp.split_reasoning(text)
+        tool_calls = None
+        finish = comp.finish_reason or "stop"
+        if named:
+            call = tp.ParsedToolCall(id=tp._call_id(), name=named,
+                                     arguments=text.strip())
+            tool_calls = [call.as_openai(0)]
+            text = None
+            finish = "tool_calls"
+        elif tools_on:
+            text, calls = tp.parse_hermes_tool_calls(text)
+            if calls:
+                tool_calls = [c.as_openai(i) for i, c in enumerate(calls)]
+                if finish == "stop":
+                    finish = "tool_calls"
         return ChatCompletionResponse(
             id=rid,
             model=req.model,
             choices=[ChatChoice(
                 index=0,
-                message=ChatCompletionMessage(content=comp.text),
-                finish_reason=comp.finish_reason or "stop",
+                message=ChatCompletionMessage(
+                    content=text if text else None,
+                    reasoning_content=reasoning,
+                    tool_calls=tool_calls),
+                finish_reason=finish,
             )],
             usage=usage,
         )
@@ -266,9 +383,11 @@ def build_app(state: ServerState) -> FastAPI:
 
 
 def make_server(engine_args: EngineArgs,
-                served_model_name: Optional[str] = None):
+                served_model_name: Optional[str] = None,
+                reasoning_parser: Optional[str] = None):
     engine = AsyncLLM(engine_args.create_engine_config())
-    state = ServerState(engine, served_model_name or engine_args.model)
+    state = ServerState(engine, served_model_name or engine_args.model,
+                        reasoning_parser=reasoning_parser)
     return build_app(state), state
 
 
@@ -280,10 +399,13 @@ def main() -> None:
     parser.add_argument("--host", type=str, default="0.0.0.0")
     parser.add_argument("--port", type=int, default=8000)
     parser.add_argument("--served-model-name", type=str, default=None)
+    parser.add_argument("--reasoning-parser", type=str, default=None,
+                        choices=["deepseek_r1"])
     EngineArgs.add_cli_args(parser)
     args = parser.parse_args()
     engine_args = EngineArgs.from_cli_args(args)
-    app, _ = make_server(engine_args, args.served_model_name)
+    app, _ = make_server(engine_args, args.served_model_name,
+                         reasoning_parser=args.reasoning_parser)
     uvicorn.run(app, host=args.host, port=args.port, log_level="info")
 
 

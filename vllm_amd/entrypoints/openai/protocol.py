@@ -82,6 +82,8 @@ class ChatMessage(BaseModel):
     role: str
     content: Optional[Union[str, list[dict[str, Any]]]] = None
     name: Optional[str] = None
+    tool_calls: Optional[list[dict[str, Any]]] = None
+    tool_call_id: Optional[str] = None
 
     def text(self) -> str:
         if isinstance(self.content, str):
@@ -125,6 +127,16 @@ class ChatCompletionRequest(BaseModel):
     guided_json: Optional[Union[dict, str]] = None
     response_format: Optional[dict[str, Any]] = None
     add_generation_prompt: bool = True
+    # Tool calling (OpenAI function-calling surface).
+    tools: Optional[list[dict[str, Any]]] = None
+    tool_choice: Optional[Union[str, dict[str, Any]]] = None
+
+    def named_tool(self) -> Optional[str]:
+        """Function name when tool_choice pins a single function."""
+        if isinstance(self.tool_choice, dict):
+            fn = self.tool_choice.get("function") or {}
+            return fn.get("name")
+        return None
 
     def to_sampling_params(self, stream: bool,
                            default_max_tokens: int) -> SamplingParams:
@@ -139,6 +151,15 @@ class ChatCompletionRequest(BaseModel):
         # schema grammar (OpenAI structured outputs shape).
         guided_json = self.guided_json
         json_object = False
+        # tool_choice naming one function: guide decoding with that
+        # function's parameter schema; the server wraps the raw JSON
+        # output as the tool call's arguments.
+        named = self.named_tool()
+        if named and self.tools:
+            from vllm_amd.entrypoints.tool_parser import named_tool_schema
+            schema = named_tool_schema(self.tools, named)
+            if schema is not None:
+                guided_json = schema
         if self.response_format:
             kind = self.response_format.get("type")
             if kind == "json_object":
@@ -199,6 +220,8 @@ class CompletionResponse(BaseModel):
 class ChatCompletionMessage(BaseModel):
     role: str = "assistant"
     content: Optional[str] = None
+    reasoning_content: Optional[str] = None
+    tool_calls: Optional[list[dict[str, Any]]] = None
 
 
 class ChatChoice(BaseModel):
@@ -220,6 +243,8 @@ class ChatCompletionResponse(BaseModel):
 class DeltaMessage(BaseModel):
     role: Optional[str] = None
     content: Optional[str] = None
+    reasoning_content: Optional[str] = None
+    tool_calls: Optional[list[dict[str, Any]]] = None
 
 
 class ChatStreamChoice(BaseModel):
