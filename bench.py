@@ -46,6 +46,9 @@ def parse_args():
     p.add_argument("--kv-cache-dtype", type=str, default="auto",
                    choices=["auto", "fp8"],
                    help="fp8 halves KV bytes (attention-bound large-batch)")
+    p.add_argument("--quantization", type=str, default=None,
+                   choices=["fp8"],
+                   help="fp8 W8A8 dense linears (~2x MFMA rate on gfx950)")
     return p.parse_args()
 
 
@@ -79,6 +82,7 @@ def main():
             model=args.model, dtype=args.dtype,
             max_model_len=args.input_len + args.warmup + args.steps + 16,
             load_format="dummy",
+            quantization=args.quantization,
         ),
         cache_config=CacheConfig(
             block_size=args.block_size,
@@ -166,7 +170,8 @@ def main():
             "higher_is_better": True,
             "scaling": "strong",
             "vs_baseline": None,
-            "dtype": args.dtype,
+            "dtype": (args.dtype if not args.quantization
+                      else f"{args.dtype}+w8a8-fp8"),
             "data": "synthetic (random 128-token prompts, random-init "
                     "dummy weights; no network)",
             "config": {

@@ -1,0 +1,145 @@
+"""fp8 W8A8 quantization tests.
+
+CPU part: the _torch_ref fp8 simulation's numerics (quant roundtrip
+error bounds, linear_fp8 vs fp32 linear) and an end-to-end engine run
+with quantization="fp8". GPU part (-m gpu): the HIP dynamic-quant kernel,
+the hipBLASLt fp8 GEMM and the fused rescale against the CPU reference
+(reference repo pattern: tests/kernels/quantization/test_fp8_quant.py and
+tests/quantization/test_fp8.py).
+"""
+
+import pytest
+import torch
+
+from vllm_amd.ops import _torch_ref as ref
+
+FP8_MAX = 448.0
+
+
+def test_quant_dynamic_roundtrip_error():
+    torch.manual_seed(0)
+    x = torch.randn(64, 512) * 3.0
+    q, s = ref.quant_fp8_dynamic(x)
+    deq = q.float() * s.unsqueeze(1)
+    # e4m3 has a 3-bit mantissa: relative error <= 2^-4 per element
+    # relative to the row max's magnitude bucket.
+    err = (deq - x).abs()
+    bound = (x.abs() / 16.0).clamp_min(s.unsqueeze(1) * 0.5)
+    assert (err <= bound + 1e-6).all()
+    # scales are absmax / 448
+    assert torch.allclose(s, x.abs().amax(dim=1) / FP8_MAX)
+
+
+def test_quant_weight_per_channel():
+    torch.manual_seed(1)
+    w = torch.randn(96, 256)
+    w[3] *= 100.0  # one hot channel must not poison the others
+    q, s = ref.quantize_weight_fp8(w)
+    deq = q.float() * s.unsqueeze(1)
+    rel = (deq - w).abs().amax(dim=1) / w.abs().amax(dim=1)
+    assert (rel < 0.07).all()
+
+
+def test_linear_fp8_close_to_fp32():
+    torch.manual_seed(2)
+    x = torch.randn(33, 256)
+    w = torch.randn(128, 256) * 0.05
+    b = torch.randn(128) * 0.1
+    w8, ws = ref.quantize_weight_fp8(w)
+    y = ref.linear_fp8(x, w8, ws, b)
+    y_ref = torch.nn.functional.linear(x, w, b)
+    # W8A8: both operands carry <=2^-4 relative rounding error.
+    denom = y_ref.abs().mean()
+    assert ((y - y_ref).abs().mean() / denom) < 0.05
+
+
+def test_zero_row_quant():
+    x = torch.zeros(4, 64)
+    q, s = ref.quant_fp8_dynamic(x)
+    assert (q.float() == 0).all() and (s > 0).all()
+
+
+def test_engine_fp8_cpu_e2e():
+    """tiny-llama with quantization=fp8 steps deterministically on CPU
+    (fp8 simulated in fp32 at the same rounding points as the GPU)."""
+    from vllm_amd.entrypoints.llm import LLM
+    from vllm_amd.sampling_params import SamplingParams
+
+    llm = LLM(model="tiny-llama", dtype="fp32", device="cpu",
+              quantization="fp8", block_size=16, num_gpu_blocks=128,
+              max_model_len=256, max_num_batched_tokens=64, max_num_seqs=4)
+    prompts = [[(i * 7 + j) % 900 + 3 for j in range(24)] for i in range(2)]
+    p = SamplingParams(temperature=0.0, max_tokens=6, ignore_eos=True)
+    outs = llm.generate(prompts, p)
+    outs2 = llm.generate(prompts, p)
+    llm.shutdown()
+    assert all(len(o.outputs[0].token_ids) == 6 for o in outs)
+    for a, b in zip(outs, outs2):
+        assert a.outputs[0].token_ids == b.outputs[0].token_ids
+    # the dense linears really are quantized
+    from vllm_amd.layers.linear import ColumnParallelLinear
+    # (engine is shut down; construct a fresh layer to sanity-check API)
+    lin = ColumnParallelLinear(64, 64, dtype=torch.float32)
+    lin.weight.data.normal_()
+    lin.quantize_fp8()
+    assert lin.weight_fp8.dtype == torch.float8_e4m3fn
+    assert lin.weight.numel() == 0
+
+
+# ------------------------------------------------------------------ GPU
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("shape", [(1, 4096), (17, 4096), (256, 8192)])
+def test_hip_dynamic_quant(shape):
+    from vllm_amd.ops import hip_ops as hip
+
+    torch.manual_seed(0)
+    x = (torch.randn(shape, dtype=torch.bfloat16, device="cuda") * 2.5)
+    q, s = hip.quant_fp8_dynamic(x)
+    q_ref, s_ref = ref.quant_fp8_dynamic(x.float().cpu())
+    torch.testing.assert_close(s.cpu(), s_ref, atol=1e-6, rtol=1e-5)
+    deq = q.float().cpu() * s.cpu().unsqueeze(1)
+    deq_ref = q_ref.float() * s_ref.unsqueeze(1)
+    # identical RNE rounding -> at most one quantum apart
+    quantum = s_ref.unsqueeze(1) * 0.0625
+    assert ((deq - deq_ref).abs() <= quantum + 1e-6).all()
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("mnk", [(7, 512, 1024), (128, 4096, 4096)])
+def test_hip_linear_fp8(mnk):
+    from vllm_amd.ops import hip_ops as hip
+
+    M, N, K = mnk
+    torch.manual_seed(1)
+    x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda")
+    w = torch.randn(N, K, dtype=torch.bfloat16, device="cuda") * 0.05
+    b = torch.randn(N, dtype=torch.bfloat16, device="cuda") * 0.1
+    w8, ws = ref.quantize_weight_fp8(w)
+    y = hip.linear_fp8(x, w8, ws, b)
+    y_ref = ref.linear_fp8(x.float().cpu(), w8.cpu(), ws.cpu(),
+                           b.float().cpu())
+    denom = y_ref.abs().mean()
+    assert ((y.float().cpu() - y_ref).abs().mean() / denom) < 0.02
+
+
+@pytest.mark.gpu
+def test_engine_fp8_gpu_e2e():
+    """llama tiny preset with quantization=fp8 decodes on the GPU through
+    the fp8 GEMM path (weights really converted, bf16 copies freed)."""
+    from vllm_amd.entrypoints.llm import LLM
+    from vllm_amd.sampling_params import SamplingParams
+
+    llm = LLM(model="tiny-llama", dtype="bf16", device="cuda",
+              quantization="fp8", block_size=16, num_gpu_blocks=512,
+              max_model_len=512, max_num_batched_tokens=512,
+              max_num_seqs=8)
+    prompts = [[(i * 11 + j) % 900 + 3 for j in range(48)]
+               for i in range(4)]
+    p = SamplingParams(temperature=0.0, max_tokens=8, ignore_eos=True)
+    outs = llm.generate(prompts, p)
+    outs2 = llm.generate(prompts, p)
+    llm.shutdown()
+    assert all(len(o.outputs[0].token_ids) == 8 for o in outs)
+    for a, b in zip(outs, outs2):
+        assert a.outputs[0].token_ids == b.outputs[0].token_ids

@@ -503,6 +503,10 @@ class ModelConfig:
     # name -> local PEFT checkpoint dir; ids are 1-based in listed order.
     lora_modules: Optional[dict] = None
     enable_expert_parallel: bool = False
+    # "fp8": W8A8 e4m3 — per-channel weight scales (quantized after
+    # load), per-token dynamic activation scales, fp8 MFMA GEMMs. Dense
+    # linears only; lm_head and MoE expert weights stay in model dtype.
+    quantization: Optional[str] = None
     spec: ModelSpec = None  # type: ignore[assignment]
 
     def __post_init__(self) -> None:
@@ -511,6 +515,14 @@ class ModelConfig:
         self.max_model_len = min(
             self.max_model_len, self.spec.max_position_embeddings
         )
+        if self.quantization is not None:
+            if self.quantization != "fp8":
+                raise ValueError(
+                    f"unsupported quantization {self.quantization!r}")
+            if self.dtype not in ("bf16", "fp32"):
+                raise ValueError(
+                    "quantization=fp8 requires dtype bf16 (GPU) or fp32 "
+                    "(CPU simulation)")
 
     @property
     def torch_dtype(self) -> torch.dtype:

@@ -32,6 +32,12 @@ void prefill_attention(torch::Tensor out, torch::Tensor q,
                        int64_t max_query_len, int64_t sliding_window);
 torch::Tensor lt_linear(torch::Tensor a, torch::Tensor w,
                         c10::optional<torch::Tensor> bias);
+torch::Tensor lt_linear_fp8(torch::Tensor a, torch::Tensor w);
+void dynamic_quant_fp8(torch::Tensor out, torch::Tensor scales,
+                       torch::Tensor x);
+void scale_rows_cols(torch::Tensor y, torch::Tensor row_scales,
+                     torch::Tensor col_scales,
+                     c10::optional<torch::Tensor> bias);
 
 }  // namespace vllm_amd
 
@@ -53,6 +59,11 @@ TORCH_LIBRARY(vllm_amd, m) {
         "float scale, int num_decodes, int max_query_len, "
         "int sliding_window) -> ()");
   m.def("lt_linear(Tensor a, Tensor w, Tensor? bias) -> Tensor");
+  m.def("lt_linear_fp8(Tensor a, Tensor w) -> Tensor");
+  m.def("dynamic_quant_fp8(Tensor(a!) out, Tensor(b!) scales, Tensor x)"
+        " -> ()");
+  m.def("scale_rows_cols(Tensor(a!) y, Tensor row_scales, Tensor col_scales,"
+        " Tensor? bias) -> ()");
 }
 
 TORCH_LIBRARY_IMPL(vllm_amd, CUDA, m) {
@@ -65,4 +76,7 @@ TORCH_LIBRARY_IMPL(vllm_amd, CUDA, m) {
   m.impl("paged_decode_attention", &vllm_amd::paged_decode_attention);
   m.impl("prefill_attention", &vllm_amd::prefill_attention);
   m.impl("lt_linear", &vllm_amd::lt_linear);
+  m.impl("lt_linear_fp8", &vllm_amd::lt_linear_fp8);
+  m.impl("dynamic_quant_fp8", &vllm_amd::dynamic_quant_fp8);
+  m.impl("scale_rows_cols", &vllm_amd::scale_rows_cols);
 }

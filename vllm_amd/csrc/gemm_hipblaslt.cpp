@@ -67,6 +67,62 @@ uint64_t shape_key(int64_t m, int64_t n, int64_t k, bool bf16, bool bias) {
          ((uint64_t)bf16 << 1) ^ (uint64_t)bias;
 }
 
+// Cache lookup, or race hipBLASLt's candidate algorithms for this
+// (desc, layouts) once — `call` runs the matmul with a given algo — and
+// cache the winner.
+template <typename Call>
+hipblasLtMatmulAlgo_t pick_algo(uint64_t key, hipblasLtMatmulDesc_t op_desc,
+                                hipblasLtMatrixLayout_t lw,
+                                hipblasLtMatrixLayout_t la,
+                                hipblasLtMatrixLayout_t lc,
+                                hipStream_t stream, Call call) {
+  auto& c = ctx();
+  {
+    std::lock_guard<std::mutex> g(c.mu);
+    auto it = c.algo_cache.find(key);
+    if (it != c.algo_cache.end()) return it->second;
+  }
+  hipblasLtMatmulPreference_t pref;
+  HIPBLASLT_CHECK(hipblasLtMatmulPreferenceCreate(&pref));
+  HIPBLASLT_CHECK(hipblasLtMatmulPreferenceSetAttribute(
+      pref, HIPBLASLT_MATMUL_PREF_MAX_WORKSPACE_BYTES, &kWorkspaceBytes,
+      sizeof(kWorkspaceBytes)));
+  constexpr int kMaxAlgos = 48;
+  std::vector<hipblasLtMatmulHeuristicResult_t> results(kMaxAlgos);
+  int found = 0;
+  HIPBLASLT_CHECK(hipblasLtMatmulAlgoGetHeuristic(
+      c.handle, op_desc, lw, la, lc, lc, pref, kMaxAlgos, results.data(),
+      &found));
+  HIPBLASLT_CHECK(hipblasLtMatmulPreferenceDestroy(pref));
+  TORCH_CHECK(found > 0, "hipBLASLt: no algorithms for this GEMM");
+
+  float best = 1e30f;
+  int best_i = 0;
+  hipEvent_t ev0, ev1;
+  HIP_CHECK(hipEventCreate(&ev0));
+  HIP_CHECK(hipEventCreate(&ev1));
+  const int reps = 3;
+  for (int i = 0; i < found; ++i) {
+    // Warm once, then time `reps` runs.
+    if (call(results[i].algo) != HIPBLAS_STATUS_SUCCESS) continue;
+    HIP_CHECK(hipEventRecord(ev0, stream));
+    for (int r = 0; r < reps; ++r) (void)call(results[i].algo);
+    HIP_CHECK(hipEventRecord(ev1, stream));
+    HIP_CHECK(hipEventSynchronize(ev1));
+    float ms = 0.f;
+    HIP_CHECK(hipEventElapsedTime(&ms, ev0, ev1));
+    if (ms < best) {
+      best = ms;
+      best_i = i;
+    }
+  }
+  HIP_CHECK(hipEventDestroy(ev0));
+  HIP_CHECK(hipEventDestroy(ev1));
+  std::lock_guard<std::mutex> g(c.mu);
+  c.algo_cache.emplace(key, results[best_i].algo);
+  return results[best_i].algo;
+}
+
 // C[M,N] row-major = A[M,K] row-major * W[N,K]^T row-major (+bias[N]).
 // In hipBLASLt's column-major view: C'[N,M] = op_T(W'[K,N]) * op_N(A'[K,M]).
 void run_matmul(torch::Tensor& out, const torch::Tensor& a,
@@ -110,66 +166,72 @@ void run_matmul(torch::Tensor& out, const torch::Tensor& a,
   const float alpha = 1.f, beta = 0.f;
   const uint64_t key = shape_key(M, N, K, is_bf16, has_bias);
 
-  hipblasLtMatmulAlgo_t algo;
-  bool have_algo = false;
-  {
-    std::lock_guard<std::mutex> g(c.mu);
-    auto it = c.algo_cache.find(key);
-    if (it != c.algo_cache.end()) {
-      algo = it->second;
-      have_algo = true;
-    }
-  }
+  auto call = [&](const hipblasLtMatmulAlgo_t& cand) {
+    return hipblasLtMatmul(
+        c.handle, op_desc, &alpha, w.data_ptr(), lw, a.data_ptr(), la,
+        &beta, out.data_ptr(), lc, out.data_ptr(), lc, &cand, c.workspace,
+        kWorkspaceBytes, stream);
+  };
+  hipblasLtMatmulAlgo_t algo =
+      pick_algo(key, op_desc, lw, la, lc, stream, call);
 
-  if (!have_algo) {
-    // First use of this shape: ask for candidates and race them once.
-    hipblasLtMatmulPreference_t pref;
-    HIPBLASLT_CHECK(hipblasLtMatmulPreferenceCreate(&pref));
-    HIPBLASLT_CHECK(hipblasLtMatmulPreferenceSetAttribute(
-        pref, HIPBLASLT_MATMUL_PREF_MAX_WORKSPACE_BYTES, &kWorkspaceBytes,
-        sizeof(kWorkspaceBytes)));
-    constexpr int kMaxAlgos = 48;
-    std::vector<hipblasLtMatmulHeuristicResult_t> results(kMaxAlgos);
-    int found = 0;
-    HIPBLASLT_CHECK(hipblasLtMatmulAlgoGetHeuristic(
-        c.handle, op_desc, lw, la, lc, lc, pref, kMaxAlgos, results.data(),
-        &found));
-    HIPBLASLT_CHECK(hipblasLtMatmulPreferenceDestroy(pref));
-    TORCH_CHECK(found > 0, "hipBLASLt: no algorithms for shape ", M, "x", N,
-                "x", K);
+  HIPBLASLT_CHECK(hipblasLtMatmul(
+      c.handle, op_desc, &alpha, w.data_ptr(), lw, a.data_ptr(), la, &beta,
+      out.data_ptr(), lc, out.data_ptr(), lc, &algo, c.workspace,
+      kWorkspaceBytes, stream));
 
-    float best = 1e30f;
-    int best_i = 0;
-    hipEvent_t ev0, ev1;
-    HIP_CHECK(hipEventCreate(&ev0));
-    HIP_CHECK(hipEventCreate(&ev1));
-    const int reps = 3;
-    for (int i = 0; i < found; ++i) {
-      // Warm once, then time `reps` runs.
-      auto call = [&](void) {
-        return hipblasLtMatmul(
-            c.handle, op_desc, &alpha, w.data_ptr(), lw, a.data_ptr(), la,
-            &beta, out.data_ptr(), lc, out.data_ptr(), lc,
-            &results[i].algo, c.workspace, kWorkspaceBytes, stream);
-      };
-      if (call() != HIPBLAS_STATUS_SUCCESS) continue;
-      HIP_CHECK(hipEventRecord(ev0, stream));
-      for (int r = 0; r < reps; ++r) (void)call();
-      HIP_CHECK(hipEventRecord(ev1, stream));
-      HIP_CHECK(hipEventSynchronize(ev1));
-      float ms = 0.f;
-      HIP_CHECK(hipEventElapsedTime(&ms, ev0, ev1));
-      if (ms < best) {
-        best = ms;
-        best_i = i;
-      }
-    }
-    HIP_CHECK(hipEventDestroy(ev0));
-    HIP_CHECK(hipEventDestroy(ev1));
-    algo = results[best_i].algo;
-    std::lock_guard<std::mutex> g(c.mu);
-    c.algo_cache.emplace(key, algo);
-  }
+  HIPBLASLT_CHECK(hipblasLtMatrixLayoutDestroy(lw));
+  HIPBLASLT_CHECK(hipblasLtMatrixLayoutDestroy(la));
+  HIPBLASLT_CHECK(hipblasLtMatrixLayoutDestroy(lc));
+  HIPBLASLT_CHECK(hipblasLtMatmulDescDestroy(op_desc));
+}
+
+// fp8 e4m3 x fp8 e4m3 -> bf16 raw GEMM (no scaling — the caller rescales
+// rows/cols in one fused pass; see quant_fp8.hip). Same tuned-algo cache
+// as the bf16 path, keyed with dtype bit = 0 and a distinct fp8 marker.
+// gfx950 fp8 MFMA peak is ~2x bf16, so big prefill GEMMs roughly double.
+void run_matmul_fp8(torch::Tensor& out, const torch::Tensor& a,
+                    const torch::Tensor& w) {
+  const int64_t M = a.size(0);
+  const int64_t K = a.size(1);
+  const int64_t N = w.size(0);
+  TORCH_CHECK(w.size(1) == K && out.size(0) == M && out.size(1) == N);
+  TORCH_CHECK(a.is_contiguous() && w.is_contiguous() && out.is_contiguous());
+  TORCH_CHECK(a.scalar_type() == torch::kFloat8_e4m3fn &&
+              w.scalar_type() == torch::kFloat8_e4m3fn);
+  TORCH_CHECK(out.scalar_type() == torch::kBFloat16);
+
+  auto& c = ctx();
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+
+  hipblasLtMatmulDesc_t op_desc;
+  HIPBLASLT_CHECK(hipblasLtMatmulDescCreate(&op_desc, HIPBLAS_COMPUTE_32F,
+                                            HIP_R_32F));
+  hipblasOperation_t opA = HIPBLAS_OP_T;
+  hipblasOperation_t opB = HIPBLAS_OP_N;
+  HIPBLASLT_CHECK(hipblasLtMatmulDescSetAttribute(
+      op_desc, HIPBLASLT_MATMUL_DESC_TRANSA, &opA, sizeof(opA)));
+  HIPBLASLT_CHECK(hipblasLtMatmulDescSetAttribute(
+      op_desc, HIPBLASLT_MATMUL_DESC_TRANSB, &opB, sizeof(opB)));
+
+  hipblasLtMatrixLayout_t lw, la, lc;
+  HIPBLASLT_CHECK(hipblasLtMatrixLayoutCreate(&lw, HIP_R_8F_E4M3, K, N, K));
+  HIPBLASLT_CHECK(hipblasLtMatrixLayoutCreate(&la, HIP_R_8F_E4M3, K, M, K));
+  HIPBLASLT_CHECK(hipblasLtMatrixLayoutCreate(&lc, HIP_R_16BF, N, M, N));
+
+  const float alpha = 1.f, beta = 0.f;
+  // fp8 keys get a high tag bit so they can't collide with bf16/fp16
+  // keys of the same shape.
+  const uint64_t key = shape_key(M, N, K, false, false) ^ (1ull << 63);
+
+  auto call = [&](const hipblasLtMatmulAlgo_t& cand) {
+    return hipblasLtMatmul(
+        c.handle, op_desc, &alpha, w.data_ptr(), lw, a.data_ptr(), la,
+        &beta, out.data_ptr(), lc, out.data_ptr(), lc, &cand, c.workspace,
+        kWorkspaceBytes, stream);
+  };
+  hipblasLtMatmulAlgo_t algo =
+      pick_algo(key, op_desc, lw, la, lc, stream, call);
 
   HIPBLASLT_CHECK(hipblasLtMatmul(
       c.handle, op_desc, &alpha, w.data_ptr(), lw, a.data_ptr(), la, &beta,
@@ -183,6 +245,13 @@ void run_matmul(torch::Tensor& out, const torch::Tensor& a,
 }
 
 }  // namespace
+
+torch::Tensor lt_linear_fp8(torch::Tensor a, torch::Tensor w) {
+  auto out = torch::empty({a.size(0), w.size(0)},
+                          a.options().dtype(torch::kBFloat16));
+  run_matmul_fp8(out, a, w);
+  return out;
+}
 
 torch::Tensor lt_linear(torch::Tensor a, torch::Tensor w,
                         c10::optional<torch::Tensor> bias) {

@@ -26,6 +26,7 @@ class EngineArgs:
     model_path: Optional[str] = None
     seed: int = 0
     enforce_eager: bool = False
+    quantization: Optional[str] = None
 
     block_size: int = 16
     gpu_memory_utilization: float = 0.90
@@ -60,6 +61,8 @@ class EngineArgs:
         parser.add_argument("--model-path", type=str, default=None)
         parser.add_argument("--seed", type=int, default=0)
         parser.add_argument("--enforce-eager", action="store_true")
+        parser.add_argument("--quantization", "-q", type=str,
+                            default=None, choices=["fp8"])
         parser.add_argument("--block-size", type=int, default=16)
         parser.add_argument("--gpu-memory-utilization", type=float,
                             default=0.90)
@@ -128,6 +131,7 @@ class EngineArgs:
                 model_path=self.model_path,
                 seed=self.seed,
                 enforce_eager=self.enforce_eager,
+                quantization=self.quantization,
             ),
             cache_config=CacheConfig(
                 block_size=self.block_size,

@@ -26,6 +26,24 @@ from vllm_amd.parallel.state import (
 )
 
 
+def _run_gemm(module, x, bias):
+    """Dispatch to the fp8 W8A8 path when this layer has been quantized
+    (registry.quantize_model_fp8), else the bf16/fp16 tuned GEMM."""
+    w8 = getattr(module, "weight_fp8", None)
+    if w8 is not None:
+        return ops.linear_fp8(x, w8, module.weight_scale, bias)
+    return ops.linear(x, module.weight, bias)
+
+
+def _quantize_fp8(module):
+    """Convert this layer's weight to e4m3 + per-channel scales and free
+    the wide copy (half the weight HBM traffic and footprint)."""
+    w8, scale = ops.quantize_weight_fp8(module.weight.data)
+    module.register_buffer("weight_fp8", w8)
+    module.register_buffer("weight_scale", scale)
+    module.weight.data = module.weight.data.new_empty(0)
+
+
 def _maybe_apply_lora(module, x, y):
     """Add per-request LoRA deltas when the forward context carries
     adapter ids and this layer has registered slices (lora.py)."""
@@ -63,7 +81,9 @@ class ReplicatedLinear(nn.Module):
         )
 
     def forward(self, x):
-        return ops.linear(x, self.weight, self.bias)
+        return _run_gemm(self, x, self.bias)
+
+    quantize_fp8 = _quantize_fp8
 
 
 class ColumnParallelLinear(nn.Module):
@@ -100,11 +120,13 @@ class ColumnParallelLinear(nn.Module):
         )
 
     def forward(self, x):
-        y = ops.linear(x, self.weight, self.bias)
+        y = _run_gemm(self, x, self.bias)
         y = _maybe_apply_lora(self, x, y)
         if self.gather_output:
             y = tensor_model_parallel_all_gather(y, dim=-1)
         return y
+
+    quantize_fp8 = _quantize_fp8
 
     def load_weight(self, full_weight: torch.Tensor) -> None:
         tp, rank = get_tp_world_size(), get_tp_rank()
@@ -219,7 +241,7 @@ class RowParallelLinear(nn.Module):
         )
 
     def forward(self, x):
-        y = ops.linear(x, self.weight, None)
+        y = _run_gemm(self, x, None)
         y = _maybe_apply_lora(self, x, y)
         if self.reduce_results:
             y = tensor_model_parallel_all_reduce(y)
@@ -234,3 +256,5 @@ class RowParallelLinear(nn.Module):
 
     def load_bias(self, full_bias: torch.Tensor) -> None:
         self.bias.data.copy_(full_bias)
+
+    quantize_fp8 = _quantize_fp8

@@ -59,4 +59,25 @@ def load_model(config: ModelConfig, device: torch.device) -> torch.nn.Module:
         load_safetensors_weights(model, config)
     else:
         raise ValueError(f"Unknown load_format {config.load_format}")
-    return model.to(device).eval()
+    model = model.to(device).eval()
+    if config.quantization == "fp8":
+        quantize_model_fp8(model)
+    return model
+
+
+def quantize_model_fp8(model: torch.nn.Module) -> None:
+    """Quantize every dense linear to W8A8 fp8 after load (per-channel
+    weight scales; activations get per-token scales at run time). The
+    lm_head stays in model dtype (logit sensitivity — same choice as the
+    reference's fp8 config), as do MoE expert weights (grouped-GEMM path
+    is bf16 this round) and the embedding."""
+    from vllm_amd.layers.linear import (
+        ColumnParallelLinear,
+        ReplicatedLinear,
+        RowParallelLinear,
+    )
+
+    for module in model.modules():
+        if isinstance(module, (ColumnParallelLinear, RowParallelLinear,
+                               ReplicatedLinear)):
+            module.quantize_fp8()

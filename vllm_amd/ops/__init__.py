@@ -110,6 +110,22 @@ def linear(x, weight, bias=None):
     return get_backend(x.device).linear(x, weight, bias)
 
 
+def linear_fp8(x, w_fp8, w_scale, bias=None):
+    """W8A8 fp8 linear: dynamic per-token activation scales, per-channel
+    weight scales. fp8 MFMA GEMM on GPU; bit-matching fp32 simulation on
+    CPU (_torch_ref.linear_fp8)."""
+    return get_backend(x.device).linear_fp8(x, w_fp8, w_scale, bias)
+
+
+def quant_fp8_dynamic(x):
+    """Per-token dynamic e4m3 quantization: [M,K] -> (fp8 [M,K], [M] f32)."""
+    return get_backend(x.device).quant_fp8_dynamic(x)
+
+
+# Weight quantization happens once at load — plain torch either device.
+from vllm_amd.ops._torch_ref import quantize_weight_fp8  # noqa: E402,F401
+
+
 def concat_and_cache_mla(c_kv, k_pe, kv_cache, slot_mapping):
     return get_backend(c_kv.device).concat_and_cache_mla(
         c_kv, k_pe, kv_cache, slot_mapping

@@ -142,6 +142,47 @@ def linear(x, weight, bias=None):
     return F.linear(x, weight, bias)
 
 
+FP8_MAX = 448.0  # OCP e4m3fn max finite
+
+
+def quant_fp8_dynamic(x):
+    """Per-token (row) dynamic fp8 quantization.
+
+    x: [M, K] -> (x_fp8 [M, K] float8_e4m3fn, scales [M] f32). Ground
+    truth for the HIP kernel in csrc/quant_fp8.hip."""
+    xf = x.float()
+    amax = xf.abs().amax(dim=-1).clamp_min(1e-12)
+    scales = amax / FP8_MAX
+    q = (xf / scales.unsqueeze(1)).to(torch.float8_e4m3fn)
+    return q, scales
+
+
+def quantize_weight_fp8(w):
+    """Per-output-channel fp8 weight quantization (done once at load).
+
+    w: [N, K] -> (w_fp8 [N, K] float8_e4m3fn, scales [N] f32)."""
+    wf = w.float()
+    amax = wf.abs().amax(dim=-1).clamp_min(1e-12)
+    scales = amax / FP8_MAX
+    q = (wf / scales.unsqueeze(1)).to(torch.float8_e4m3fn)
+    return q, scales
+
+
+def linear_fp8(x, w_fp8, w_scale, bias=None):
+    """W8A8 fp8 linear, CPU reference: quantize activations per-token
+    exactly as the GPU path does, then compute in fp32 on dequantized
+    values (same rounding points as fp8-GEMM + rs*cs rescale)."""
+    shape = list(x.shape)
+    x2 = x.reshape(-1, shape[-1])
+    x_fp8, x_scale = quant_fp8_dynamic(x2)
+    y = x_fp8.float() @ w_fp8.float().t()
+    y = y * x_scale.unsqueeze(1) * w_scale.unsqueeze(0)
+    if bias is not None:
+        y = y + bias.float()
+    shape[-1] = w_fp8.shape[0]
+    return y.to(x.dtype).reshape(shape)
+
+
 def topk_softmax(gating_logits, topk, renormalize=True):
     """gating_logits: [T, E] -> (topk_weights [T,k] f32, topk_ids [T,k] i32)."""
     probs = gating_logits.float().softmax(dim=-1)

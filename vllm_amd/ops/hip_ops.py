@@ -55,6 +55,28 @@ def linear(x, weight, bias=None):
     return _C.lt_linear(x, weight, bias)
 
 
+def quant_fp8_dynamic(x):
+    # Per-token dynamic quant (csrc/quant_fp8.hip): one workgroup/row.
+    M, K = x.shape
+    out = torch.empty(M, K, dtype=torch.float8_e4m3fn, device=x.device)
+    scales = torch.empty(M, dtype=torch.float32, device=x.device)
+    _C.dynamic_quant_fp8(out.view(torch.uint8), scales, x)
+    return out, scales
+
+
+def linear_fp8(x, w_fp8, w_scale, bias=None):
+    """W8A8 fp8 linear: per-token activation quant -> fp8 MFMA GEMM
+    (tuned hipBLASLt, raw bf16 out) -> fused rs[m]*cs[n] rescale + bias.
+    ~2x bf16 GEMM rate on gfx950 and half the weight HBM traffic."""
+    shape = list(x.shape)
+    x2 = x.reshape(-1, shape[-1]).contiguous()
+    x_fp8, x_scale = quant_fp8_dynamic(x2)
+    y = _C.lt_linear_fp8(x_fp8, w_fp8)
+    _C.scale_rows_cols(y, x_scale, w_scale, bias)
+    shape[-1] = w_fp8.shape[0]
+    return y.reshape(shape)
+
+
 def reshape_and_cache(key, value, kv_cache, slot_mapping):
     _C.reshape_and_cache(key, value, kv_cache, slot_mapping)
 
