@@ -49,6 +49,13 @@ def parse_args():
     p.add_argument("--quantization", type=str, default=None,
                    choices=["fp8"],
                    help="fp8 W8A8 dense linears (~2x MFMA rate on gfx950)")
+    p.add_argument("--parallelism", type=str, default="dp",
+                   choices=["dp", "tp"],
+                   help="dp: one independent engine replica per rank, each "
+                        "with its own --batch (weak scaling; xGMI-free, the "
+                        "throughput-optimal layout for an 8B model). tp: one "
+                        "engine, weights sharded over all ranks (strong "
+                        "scaling; the 70B layout)")
     return p.parse_args()
 
 
@@ -56,10 +63,9 @@ def main():
     args = parse_args()
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    dp_mode = args.parallelism == "dp" and world > 1
     on_gpu = torch.cuda.is_available()
-    if world > 1:
-        assert on_gpu, "multi-rank bench requires GPUs"
-
     if not on_gpu and args.model == "llama-3-8b":
         # CPU plumbing check only (no GPU in the build container).
         args.model = "tiny-llama"
@@ -95,12 +101,15 @@ def main():
             max_num_seqs=args.batch,
             enable_chunked_prefill=True,
         ),
-        parallel_config=ParallelConfig(tensor_parallel_size=world),
+        parallel_config=ParallelConfig(
+            tensor_parallel_size=1 if dp_mode else world,
+            rank=rank, local_rank=local_rank, world_size=world),
         device_config=DeviceConfig(device="cuda" if on_gpu else "cpu"),
     )
 
     engine = EngineCore(config)
-    is_driver = rank == 0
+    # dp: every rank drives its own engine replica over its own batch.
+    is_driver = engine.is_driver
 
     def sync():
         if world > 1:
@@ -139,6 +148,7 @@ def main():
         t1 = time.perf_counter()
         elapsed = t1 - t0
     else:
+        out_tokens = 0
         for _ in range(args.warmup):
             engine.step_worker()
         sync()
@@ -149,14 +159,18 @@ def main():
         t1 = time.perf_counter()
         elapsed = t1 - t0
 
-    # MAX elapsed over ranks.
     if world > 1:
-        t = torch.tensor([elapsed], dtype=torch.float64,
-                         device="cuda" if on_gpu else "cpu")
+        # MAX elapsed over ranks; SUM of sampled tokens (dp: every rank
+        # generated its own batch; tp: only rank 0 counted).
+        dev = "cuda" if on_gpu else "cpu"
+        t = torch.tensor([elapsed], dtype=torch.float64, device=dev)
         torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
         elapsed = float(t.item())
+        n = torch.tensor([out_tokens], dtype=torch.float64, device=dev)
+        torch.distributed.all_reduce(n, op=torch.distributed.ReduceOp.SUM)
+        out_tokens = int(n.item())
 
-    if is_driver:
+    if rank == 0:
         # Count ACTUAL sampled tokens in the timed region (robust to
         # prefill chunks spilling past warmup).
         result = {
@@ -168,7 +182,7 @@ def main():
             "warmup": args.warmup,
             "ms_per_step": round(elapsed / args.steps * 1000, 3),
             "higher_is_better": True,
-            "scaling": "strong",
+            "scaling": "weak" if dp_mode else "strong",
             "vs_baseline": None,
             "dtype": (args.dtype if not args.quantization
                       else f"{args.dtype}+w8a8-fp8"),
@@ -176,14 +190,19 @@ def main():
                     "dummy weights; no network)",
             "config": {
                 "model": args.model,
-                "global_batch": args.batch,
+                "global_batch": args.batch * (world if dp_mode else 1),
                 "seq_len": args.input_len,
-                "parallelism": f"tp{world}",
+                "parallelism": f"{args.parallelism}{world}"
+                if world > 1 else "tp1",
             },
         }
         print(json.dumps(result))
-    engine.shutdown()
+    if is_driver:
+        engine.shutdown()  # drains the async-pending step (broadcasts)
+    else:
+        engine.run_spmd_worker_loop()  # consume drain + stop sentinel
     if world > 1:
+        torch.distributed.barrier()
         torch.distributed.destroy_process_group()
 
 

@@ -588,6 +588,13 @@ class ParallelConfig:
     world_size: int = 1
     distributed_backend: str = "auto"  # auto -> nccl(RCCL) on GPU, gloo on CPU
 
+    def __post_init__(self) -> None:
+        if (self.enable_expert_parallel
+                and self.tensor_parallel_size < self.world_size):
+            raise ValueError(
+                "expert parallelism shards experts over the whole world; "
+                "it cannot be combined with DP replicas (tp < world)")
+
     @property
     def needs_distributed(self) -> bool:
         return self.world_size > 1 or self.tensor_parallel_size > 1

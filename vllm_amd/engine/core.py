@@ -37,7 +37,11 @@ class EngineCore:
         launched_world = int(os.environ.get("WORLD_SIZE", "1"))
         self._multiproc = tp > 1 and launched_world == 1
 
-        from vllm_amd.parallel.state import get_world_group
+        # The engine's lockstep domain is the TP group: with tp == world this
+        # is the whole job (classic SPMD TP); with tp < world each TP group
+        # runs an independent engine replica (SPMD data parallelism — the
+        # bench's dp mode; role of the reference's DP engine replicas).
+        from vllm_amd.parallel.state import get_tp_group
 
         if self._multiproc:
             # Engine owns the scheduler; one spawned worker process per
@@ -45,7 +49,7 @@ class EngineCore:
             from vllm_amd.executor.multiproc import MultiprocExecutor
 
             self.worker = MultiprocExecutor(config)
-            self.world = get_world_group()  # engine proc: world of 1
+            self.world = get_tp_group()  # engine proc: group of 1
             self.is_driver = True
             num_blocks = self.worker.determine_num_kv_blocks()
             self.num_gpu_blocks = num_blocks
@@ -56,7 +60,7 @@ class EngineCore:
             self.worker.init_device()
             self.worker.load_model()
 
-            self.world = get_world_group()
+            self.world = get_tp_group()
             self.is_driver = self.world.rank_in_group == 0
 
             # KV sizing must agree across ranks: min over ranks.
@@ -66,7 +70,8 @@ class EngineCore:
                 if torch.cuda.is_available():
                     t = t.cuda()
                 torch.distributed.all_reduce(
-                    t, op=torch.distributed.ReduceOp.MIN
+                    t, op=torch.distributed.ReduceOp.MIN,
+                    group=self.world.device_group,
                 )
                 num_blocks = int(t.item())
             self.num_gpu_blocks = num_blocks
