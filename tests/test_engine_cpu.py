@@ -282,3 +282,62 @@ def test_gemma3_cpu_decode():
     assert all(len(o.outputs[0].token_ids) == 6 for o in outs)
     for a, b in zip(outs, outs2):
         assert a.outputs[0].token_ids == b.outputs[0].token_ids
+
+
+def test_sleep_wake_cycle():
+    """sleep frees KV + offloads (level 1) or discards (level 2) weights;
+    wake_up restores; generation after wake matches generation before."""
+    from vllm_amd.entrypoints.llm import LLM
+    from vllm_amd.sampling_params import SamplingParams
+
+    llm = LLM(model="tiny-llama", dtype="fp32", device="cpu",
+              block_size=16, num_gpu_blocks=64, max_model_len=256,
+              max_num_batched_tokens=64, max_num_seqs=4)
+    prompts = [[(i * 5 + j) % 900 + 3 for j in range(16)] for i in range(2)]
+    p = SamplingParams(temperature=0.0, max_tokens=5, ignore_eos=True)
+    before = llm.generate(prompts, p)
+    core = llm.engine.engine_core
+    assert not core.is_sleeping()
+    for level in (1, 2):
+        core.sleep(level)
+        assert core.is_sleeping()
+        assert core.worker.runner.kv_caches == []
+        if level == 2:
+            assert core.worker.runner.model is None
+        core.wake_up()
+        assert not core.is_sleeping()
+        after = llm.generate(prompts, p)
+        for a, b in zip(before, after):
+            assert a.outputs[0].token_ids == b.outputs[0].token_ids, level
+    llm.shutdown()
+
+
+def test_sleep_rejected_while_busy():
+    from vllm_amd.config import (
+        CacheConfig, DeviceConfig, EngineConfig, ModelConfig,
+        SchedulerConfig,
+    )
+    from vllm_amd.engine.core import EngineCore
+    from vllm_amd.request import Request
+    from vllm_amd.sampling_params import SamplingParams
+
+    config = EngineConfig(
+        model_config=ModelConfig(model="tiny-llama", dtype="fp32",
+                                 max_model_len=128),
+        cache_config=CacheConfig(block_size=16, num_gpu_blocks=64),
+        scheduler_config=SchedulerConfig(max_num_batched_tokens=128,
+                                         max_num_seqs=2),
+        device_config=DeviceConfig(device="cpu"),
+    )
+    core = EngineCore(config)
+    core.add_request(Request(
+        request_id="r0", prompt_token_ids=list(range(3, 19)),
+        sampling_params=SamplingParams(max_tokens=4, ignore_eos=True)))
+    import pytest as _pytest
+    with _pytest.raises(RuntimeError):
+        core.sleep()
+    while core.has_unfinished_requests():
+        core.step()
+    core.sleep()
+    core.wake_up()
+    core.shutdown()

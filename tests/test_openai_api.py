@@ -251,3 +251,18 @@ def test_completions_guided_regex(client):
     })
     assert r.status_code == 200, r.text
     assert _re.fullmatch(pattern, r.json()["choices"][0]["text"])
+
+
+def test_sleep_wake_endpoints(client):
+    assert client.get("/is_sleeping").json() == {"is_sleeping": False}
+    r = client.post("/sleep?level=1")
+    assert r.status_code == 200
+    assert client.get("/is_sleeping").json() == {"is_sleeping": True}
+    assert client.post("/wake_up").status_code == 200
+    assert client.get("/is_sleeping").json() == {"is_sleeping": False}
+    r = client.post("/v1/completions", json={
+        "model": "tiny-llama", "prompt": "after wake", "max_tokens": 4,
+        "temperature": 0.0, "ignore_eos": True,
+    })
+    assert r.status_code == 200, r.text
+    assert r.json()["usage"]["completion_tokens"] == 4

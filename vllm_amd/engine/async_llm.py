@@ -125,6 +125,21 @@ class AsyncLLM:
     async def abort(self, request_id: str) -> None:
         self._inbox.put(("abort", request_id))
 
+    def sleep(self, level: int = 1) -> None:
+        """Release GPU memory (weights to host at level 1, discarded at
+        level 2; KV pool freed). Only valid with no unfinished requests —
+        the engine loop is parked in its idle poll then, so the call is
+        safe from the API thread."""
+        if self.engine.has_unfinished_requests():
+            raise RuntimeError("cannot sleep with unfinished requests")
+        self.engine.sleep(level)
+
+    def wake_up(self) -> None:
+        self.engine.wake_up()
+
+    def is_sleeping(self) -> bool:
+        return self.engine.is_sleeping()
+
     def stats(self) -> dict:
         sched = self.engine.engine_core.scheduler
         if sched is None:

@@ -102,6 +102,28 @@ class EngineCore:
             self.scheduler.has_unfinished_requests()
 
     # ------------------------------------------------------------------
+    # Sleep mode (role of the reference's engine sleep/wake_up): release
+    # GPU memory between serving bursts. Level 1 offloads weights to host
+    # RAM and frees the KV pool; level 2 discards weights too.
+    def sleep(self, level: int = 1) -> None:
+        if self.has_unfinished_requests():
+            raise RuntimeError("cannot sleep with unfinished requests")
+        if self._pending is not None:
+            self._drain()
+        self.worker.sleep(level)
+        if self.scheduler is not None:
+            # Cached prefix blocks point into the freed pool.
+            self.scheduler.kv_cache_manager.reset_prefix_cache()
+        self._sleeping = True
+
+    def wake_up(self) -> None:
+        self.worker.wake_up()
+        self._sleeping = False
+
+    def is_sleeping(self) -> bool:
+        return getattr(self, "_sleeping", False)
+
+    # ------------------------------------------------------------------
     def _drain(self) -> list[EngineCoreOutput]:
         if self._pending is None:
             return []

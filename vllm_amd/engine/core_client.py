@@ -49,6 +49,15 @@ def _engine_proc_main(config: EngineConfig, conn) -> None:
                 core.add_request(msg[1])
             elif kind == "abort":
                 core.abort_requests(msg[1])
+            elif kind == "sleep":
+                try:
+                    core.sleep(msg[1])
+                    conn.send(("slept", None))
+                except Exception as e:  # noqa: BLE001
+                    conn.send(("slept", repr(e)))
+            elif kind == "wake_up":
+                core.wake_up()
+                conn.send(("woke", None))
             elif kind == "shutdown":
                 running = False
         if not running:
@@ -88,6 +97,8 @@ class EngineCoreClient:
         self.scheduler = None  # lives in the engine process
         self._unfinished: set[str] = set()
         self._outq: "queue.Queue" = queue.Queue()
+        self._ctrlq: "queue.Queue" = queue.Queue()
+        self._sleeping = False
         self._recv_thread = threading.Thread(
             target=self._recv_loop, daemon=True, name="engine-core-recv"
         )
@@ -104,6 +115,8 @@ class EngineCoreClient:
                 return
             if kind == "outputs":
                 self._outq.put(payload)
+            elif kind in ("slept", "woke"):
+                self._ctrlq.put((kind, payload))
             elif kind == "error":
                 self._outq.put(RuntimeError(payload))
             elif kind == "bye":
@@ -121,6 +134,25 @@ class EngineCoreClient:
 
     def has_unfinished_requests(self) -> bool:
         return bool(self._unfinished)
+
+    def sleep(self, level: int = 1) -> None:
+        if self._unfinished:
+            raise RuntimeError("cannot sleep with unfinished requests")
+        self._conn.send(("sleep", level))
+        kind, err = self._ctrlq.get(timeout=120)
+        assert kind == "slept"
+        if err:
+            raise RuntimeError(err)
+        self._sleeping = True
+
+    def wake_up(self) -> None:
+        self._conn.send(("wake_up",))
+        kind, _ = self._ctrlq.get(timeout=300)
+        assert kind == "woke"
+        self._sleeping = False
+
+    def is_sleeping(self) -> bool:
+        return self._sleeping
 
     def step(self, timeout: float = 0.05) -> list[EngineCoreOutput]:
         """Dequeue one batch of outputs (the engine process steps on its

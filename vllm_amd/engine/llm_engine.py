@@ -249,5 +249,14 @@ class LLMEngine:
     def has_unfinished_requests(self) -> bool:
         return self.engine_core.has_unfinished_requests()
 
+    def sleep(self, level: int = 1) -> None:
+        self.engine_core.sleep(level)
+
+    def wake_up(self) -> None:
+        self.engine_core.wake_up()
+
+    def is_sleeping(self) -> bool:
+        return self.engine_core.is_sleeping()
+
     def shutdown(self) -> None:
         self.engine_core.shutdown()

@@ -129,6 +129,24 @@ def build_app(state: ServerState) -> FastAPI:
     async def detokenize(req: DetokenizeRequest) -> DetokenizeResponse:
         return DetokenizeResponse(prompt=engine.tokenizer.decode(req.tokens))
 
+    @app.post("/sleep")
+    async def sleep(raw: Request):
+        level = int(raw.query_params.get("level", "1"))
+        try:
+            engine.sleep(level)
+        except RuntimeError as e:
+            return _error(str(e), 400)
+        return Response(status_code=200)
+
+    @app.post("/wake_up")
+    async def wake_up():
+        engine.wake_up()
+        return Response(status_code=200)
+
+    @app.get("/is_sleeping")
+    async def is_sleeping():
+        return {"is_sleeping": engine.is_sleeping()}
+
     @app.get("/metrics")
     async def metrics() -> Response:
         s = engine.stats()

@@ -159,6 +159,12 @@ class MultiprocExecutor:
     def initialize_kv_cache(self, num_blocks: int) -> None:
         self.collective_rpc("initialize_kv_cache", num_blocks)
 
+    def sleep(self, level: int = 1) -> None:
+        self.collective_rpc("sleep", level)
+
+    def wake_up(self) -> None:
+        self.collective_rpc("wake_up")
+
     # ---- data plane ---------------------------------------------------
     def execute_model(self, so):
         for conn in self.conns:

@@ -10,6 +10,7 @@ computed after this step.
 from __future__ import annotations
 
 import dataclasses
+import gc
 from dataclasses import dataclass, field
 from typing import Optional
 
@@ -278,6 +279,7 @@ class ModelRunner:
     def allocate_kv_cache(self, num_blocks: int) -> None:
         from vllm_amd.parallel.state import get_tp_world_size
 
+        self.num_gpu_blocks = num_blocks
         spec = self.spec
         if spec.is_mla:
             # MLA cache stays at model precision (fp8 MLA cache later).
@@ -307,6 +309,32 @@ class ModelRunner:
                 and not self.config.model_config.lora_modules
                 and not self.spec.is_moe and not self.spec.is_mla):
             self.graph_runner = DecodeGraphRunner(self)
+
+    # ------------------------------------------------------------------
+    def sleep(self, level: int = 1) -> None:
+        """Release GPU memory between serving bursts (role of the
+        reference's sleep mode, vllm/device_allocator + worker sleep):
+        level 1 frees the KV pool and the hipGraphs holding pointers into
+        it and offloads weights to host RAM; level 2 also discards the
+        weights (wake_up reloads them from the configured source)."""
+        self.graph_runner = None
+        self.kv_caches = []
+        self._last_sampled = None
+        self._samp_cache_key = self._samp_cache_val = None
+        if level >= 2:
+            self.model = None
+        elif self.model is not None:
+            self.model = self.model.to("cpu")
+        gc.collect()
+        if self.device.type == "cuda":
+            torch.cuda.empty_cache()
+
+    def wake_up(self) -> None:
+        if self.model is None:
+            self.load_model()
+        else:
+            self.model = self.model.to(self.device)
+        self.allocate_kv_cache(self.num_gpu_blocks)
 
     # ------------------------------------------------------------------
     def _update_states(self, so: SchedulerOutput) -> None:

@@ -82,6 +82,12 @@ class Worker:
     def initialize_kv_cache(self, num_blocks: int) -> None:
         self.runner.allocate_kv_cache(num_blocks)
 
+    def sleep(self, level: int = 1) -> None:
+        self.runner.sleep(level)
+
+    def wake_up(self) -> None:
+        self.runner.wake_up()
+
     def execute_model(self, so: SchedulerOutput) -> ModelRunnerOutput:
         return self.runner.execute_model(so)
 
