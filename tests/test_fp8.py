@@ -98,11 +98,20 @@ def test_hip_dynamic_quant(shape):
     q, s = hip.quant_fp8_dynamic(x)
     q_ref, s_ref = ref.quant_fp8_dynamic(x.float().cpu())
     torch.testing.assert_close(s.cpu(), s_ref, atol=1e-6, rtol=1e-5)
-    deq = q.float().cpu() * s.cpu().unsqueeze(1)
-    deq_ref = q_ref.float() * s_ref.unsqueeze(1)
-    # identical RNE rounding -> at most one quantum apart
-    quantum = s_ref.unsqueeze(1) * 0.0625
-    assert ((deq - deq_ref).abs() <= quantum + 1e-6).all()
+
+    # The kernel scales by a reciprocal while the torch ref divides, so a
+    # value can cross a rounding boundary: allow at most ONE e4m3 code of
+    # disagreement. e4m3 codes are sign-magnitude and monotone in the
+    # low 7 bits, so compare in signed-magnitude integer space.
+    def codes(t):
+        b = t.view(torch.uint8).int()
+        mag = b & 0x7F
+        return torch.where(b >= 128, -mag, mag)
+
+    d = (codes(q.cpu()) - codes(q_ref)).abs()
+    assert (d <= 1).all(), f"max code diff {d.max().item()}"
+    # and >99% must agree exactly
+    assert (d == 0).float().mean() > 0.99
 
 
 @pytest.mark.gpu
