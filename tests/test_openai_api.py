@@ -266,3 +266,26 @@ def test_sleep_wake_endpoints(client):
     })
     assert r.status_code == 200, r.text
     assert r.json()["usage"]["completion_tokens"] == 4
+
+
+def test_metrics_histograms(client):
+    # non-stream + stream requests populate the latency histograms
+    client.post("/v1/completions", json={
+        "model": "tiny-llama", "prompt": "hist test", "max_tokens": 4,
+        "temperature": 0.0, "ignore_eos": True})
+    with client.stream("POST", "/v1/completions", json={
+            "model": "tiny-llama", "prompt": "hist stream", "max_tokens": 4,
+            "temperature": 0.0, "ignore_eos": True, "stream": True}) as r:
+        for _ in r.iter_lines():
+            pass
+    text = client.get("/metrics").text
+    assert "vllm_amd:e2e_request_latency_seconds_count" in text
+    assert "vllm_amd:time_to_first_token_seconds_bucket" in text
+    assert "vllm_amd:time_per_output_token_seconds_sum" in text
+    assert "vllm_amd:request_generation_tokens_count" in text
+    assert "vllm_amd:prefix_cache_queries_total" in text
+    assert "vllm_amd:num_preemptions_total" in text
+    # counts are cumulative and > 0
+    for line in text.splitlines():
+        if line.startswith("vllm_amd:e2e_request_latency_seconds_count"):
+            assert float(line.split()[-1]) >= 2

@@ -77,6 +77,7 @@ class Scheduler:
         # Stats
         self.prefix_cache_queries = 0
         self.prefix_cache_hits = 0
+        self.num_preemptions_total = 0
 
     # ------------------------------------------------------------------
     # Request admission / removal
@@ -302,6 +303,7 @@ class Scheduler:
         request.num_output_placeholders = 0
         request.spec_token_ids = []
         request.num_preemptions += 1
+        self.num_preemptions_total += 1
         self.waiting.appendleft(request)
 
     # ------------------------------------------------------------------

@@ -150,6 +150,9 @@ class AsyncLLM:
             "kv_blocks_total": self.engine.engine_core.num_gpu_blocks,
             "kv_blocks_free":
                 sched.kv_cache_manager.block_pool.get_num_free_blocks(),
+            "prefix_cache_queries": sched.prefix_cache_queries,
+            "prefix_cache_hits": sched.prefix_cache_hits,
+            "num_preemptions": sched.num_preemptions_total,
         }
 
     def shutdown(self) -> None:

@@ -88,7 +88,10 @@ class ServerState:
         self.lora_names = list(
             engine.config.model_config.lora_modules or {})
         self.max_model_len = engine.config.model_config.max_model_len
-        # Prometheus counters.
+        # Prometheus counters + histograms.
+        from vllm_amd.metrics import ServerMetrics
+
+        self.metrics = ServerMetrics()
         self.num_requests = 0
         self.num_prompt_tokens = 0
         self.num_generation_tokens = 0
@@ -163,7 +166,15 @@ def build_app(state: ServerState) -> FastAPI:
             f"vllm_amd:num_requests_waiting {s.get('num_waiting', 0)}",
             "# TYPE vllm_amd:kv_blocks_free gauge",
             f"vllm_amd:kv_blocks_free {s.get('kv_blocks_free', 0)}",
+            "# TYPE vllm_amd:prefix_cache_queries_total counter",
+            "vllm_amd:prefix_cache_queries_total "
+            f"{s.get('prefix_cache_queries', 0)}",
+            "# TYPE vllm_amd:prefix_cache_hits_total counter",
+            f"vllm_amd:prefix_cache_hits_total {s.get('prefix_cache_hits', 0)}",
+            "# TYPE vllm_amd:num_preemptions_total counter",
+            f"vllm_amd:num_preemptions_total {s.get('num_preemptions', 0)}",
         ]
+        lines += state.metrics.render()
         return Response("\n".join(lines) + "\n",
                         media_type="text/plain; version=0.0.4")
 
@@ -187,11 +198,18 @@ def build_app(state: ServerState) -> FastAPI:
 
         if req.stream:
             async def gen() -> AsyncGenerator[str, None]:
+                from vllm_amd.metrics import RequestTimer
+                timer = RequestTimer(state.metrics)
+                n_gen = 0
+                n_prompt = 0
                 try:
                     async for out in engine.generate(prompt, params, rid,
                                                      lora=lora):
                         comp = out.outputs[0]
                         state.num_generation_tokens += len(comp.token_ids)
+                        timer.on_tokens(len(comp.token_ids))
+                        n_gen += len(comp.token_ids)
+                        n_prompt = len(out.prompt_token_ids)
                         chunk = {
                             "id": rid,
                             "object": "text_completion",
@@ -205,12 +223,15 @@ def build_app(state: ServerState) -> FastAPI:
                             }],
                         }
                         yield f"data: {json.dumps(chunk)}\n\n"
+                    timer.on_finish(n_prompt, n_gen)
                     yield "data: [DONE]\n\n"
                 except Exception as e:  # noqa: BLE001
                     err = {"error": {"message": str(e)}}
                     yield f"data: {json.dumps(err)}\n\n"
             return StreamingResponse(gen(), media_type="text/event-stream")
 
+        from vllm_amd.metrics import RequestTimer
+        timer = RequestTimer(state.metrics)
         final = None
         try:
             async for out in engine.generate(prompt, params, rid, lora=lora):
@@ -218,6 +239,7 @@ def build_app(state: ServerState) -> FastAPI:
         except ValueError as e:
             return _error(str(e))
         comp = final.outputs[0]
+        timer.on_finish(len(final.prompt_token_ids), len(comp.token_ids))
         state.num_prompt_tokens += len(final.prompt_token_ids)
         state.num_generation_tokens += len(comp.token_ids)
         usage = UsageInfo(
@@ -267,6 +289,10 @@ def build_app(state: ServerState) -> FastAPI:
                                                     content=""))],
                 )
                 yield f"data: {first.model_dump_json()}\n\n"
+                from vllm_amd.metrics import RequestTimer
+                timer = RequestTimer(state.metrics)
+                n_gen = 0
+                n_prompt = 0
                 rparse = (tp.StreamingReasoningParser()
                           if state.reasoning_parser else None)
                 tparse = (tp.StreamingToolParser()
@@ -322,6 +348,9 @@ def build_app(state: ServerState) -> FastAPI:
                                                      lora=lora):
                         comp = out.outputs[0]
                         state.num_generation_tokens += len(comp.token_ids)
+                        timer.on_tokens(len(comp.token_ids))
+                        n_gen += len(comp.token_ids)
+                        n_prompt = len(out.prompt_token_ids)
                         msgs, saw = deltas(comp.text, comp.finish_reason)
                         finish = comp.finish_reason
                         if finish and (saw or named):
@@ -345,12 +374,15 @@ def build_app(state: ServerState) -> FastAPI:
                                     finish_reason=finish)],
                             )
                             yield f"data: {chunk.model_dump_json()}\n\n"
+                    timer.on_finish(n_prompt, n_gen)
                     yield "data: [DONE]\n\n"
                 except Exception as e:  # noqa: BLE001
                     err = {"error": {"message": str(e)}}
                     yield f"data: {json.dumps(err)}\n\n"
             return StreamingResponse(gen(), media_type="text/event-stream")
 
+        from vllm_amd.metrics import RequestTimer
+        timer = RequestTimer(state.metrics)
         final = None
         try:
             async for out in engine.generate(prompt, params, rid, lora=lora):
@@ -358,6 +390,7 @@ def build_app(state: ServerState) -> FastAPI:
         except ValueError as e:
             return _error(str(e))
         comp = final.outputs[0]
+        timer.on_finish(len(final.prompt_token_ids), len(comp.token_ids))
         state.num_prompt_tokens += len(final.prompt_token_ids)
         state.num_generation_tokens += len(comp.token_ids)
         usage = UsageInfo(
