@@ -289,3 +289,39 @@ def test_metrics_histograms(client):
     for line in text.splitlines():
         if line.startswith("vllm_amd:e2e_request_latency_seconds_count"):
             assert float(line.split()[-1]) >= 2
+
+
+def test_run_batch_jsonl(tmp_path):
+    """Offline OpenAI batch runner: JSONL in -> JSONL out through the
+    in-process ASGI app (reference: entrypoints/openai/run_batch.py)."""
+    import subprocess
+    import sys
+
+    inp = tmp_path / "in.jsonl"
+    outp = tmp_path / "out.jsonl"
+    lines = [
+        {"custom_id": "a", "method": "POST", "url": "/v1/completions",
+         "body": {"model": "tiny-llama", "prompt": "one", "max_tokens": 4,
+                  "temperature": 0.0, "ignore_eos": True}},
+        {"custom_id": "b", "method": "POST", "url": "/v1/chat/completions",
+         "body": {"model": "tiny-llama",
+                  "messages": [{"role": "user", "content": "two"}],
+                  "max_tokens": 4, "temperature": 0.0, "ignore_eos": True}},
+        {"custom_id": "c", "method": "GET", "url": "/nope", "body": {}},
+    ]
+    inp.write_text("\n".join(json.dumps(x) for x in lines))
+    r = subprocess.run(
+        [sys.executable, "-m", "vllm_amd", "run-batch", "-i", str(inp),
+         "-o", str(outp), "--model", "tiny-llama", "--dtype", "fp32",
+         "--device", "cpu", "--block-size", "16", "--num-gpu-blocks", "64",
+         "--max-model-len", "256"],
+        capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr
+    results = {json.loads(ln)["custom_id"]: json.loads(ln)
+               for ln in outp.read_text().splitlines()}
+    assert results["a"]["error"] is None
+    assert results["a"]["response"]["body"]["usage"][
+        "completion_tokens"] == 4
+    assert results["b"]["response"]["body"]["choices"][0][
+        "message"] is not None
+    assert results["c"]["error"] is not None

@@ -8,9 +8,10 @@ import sys
 
 def main() -> None:
     if len(sys.argv) < 2 or sys.argv[1] in ("-h", "--help"):
-        print("usage: python -m vllm_amd {serve,bench} [args]\n"
+        print("usage: python -m vllm_amd {serve,bench,run-batch} [args]\n"
               "  serve          — start the OpenAI-compatible API server\n"
-              "  bench serving  — TTFT/ITL + throughput at fixed QPS")
+              "  bench serving  — TTFT/ITL + throughput at fixed QPS\n"
+              "  run-batch      — offline OpenAI batch-format JSONL runner")
         return
     cmd = sys.argv.pop(1)
     if cmd == "bench":
@@ -21,6 +22,11 @@ def main() -> None:
         from benchmarks.bench_serving import main as bench_main
 
         bench_main()
+        return
+    if cmd == "run-batch":
+        from vllm_amd.entrypoints.run_batch import main as rb_main
+
+        rb_main()
         return
     if cmd == "serve":
         # `python -m vllm_amd serve <model> [args]` or with --model.
