@@ -558,11 +558,15 @@ class SchedulerConfig:
     # Pipeline CPU scheduling with GPU execution (one step in flight;
     # role of the reference's AsyncScheduler + async model-runner output).
     async_scheduling: bool = True
-    # Speculative decoding: ngram prompt-lookup drafts (0 = off). Spec
-    # decode forces synchronous scheduling.
+    # Speculative decoding (0 = off). Spec decode forces synchronous
+    # scheduling. Methods: "ngram" (prompt-lookup, CPU) or "medusa"
+    # (model-based multi-head drafts from the runner).
     num_speculative_tokens: int = 0
+    spec_decode_method: str = "ngram"
     ngram_prompt_lookup_min: int = 2
     ngram_prompt_lookup_max: int = 4
+    # Optional dir with medusa-head safetensors; None -> random heads.
+    medusa_path: Optional[str] = None
 
     def __post_init__(self) -> None:
         if self.num_speculative_tokens > 0:

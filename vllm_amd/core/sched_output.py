@@ -70,6 +70,9 @@ class ModelRunnerOutput:
     # Optional per-request logprobs of sampled tokens:
     # req_id -> list of {token_id: logprob} dicts, one per sampled token.
     logprobs: Optional[dict[str, list[dict[int, float]]]] = None
+    # Model-based draft proposals (medusa heads): req_id -> draft tokens
+    # for the NEXT step, conditioned on the last accepted position.
+    draft_token_ids: Optional[dict[str, list[int]]] = None
 
 
 EMPTY_MODEL_RUNNER_OUTPUT = ModelRunnerOutput(req_ids=[], sampled_token_ids=[])

@@ -51,13 +51,18 @@ class Scheduler:
 
         sc = config.scheduler_config
         if sc.num_speculative_tokens > 0:
-            from vllm_amd.spec_decode.ngram import NgramProposer
+            if sc.spec_decode_method == "medusa":
+                from vllm_amd.spec_decode.medusa import MedusaProposer
 
-            self.spec_proposer = NgramProposer(
-                min_n=sc.ngram_prompt_lookup_min,
-                max_n=sc.ngram_prompt_lookup_max,
-                k=sc.num_speculative_tokens,
-            )
+                self.spec_proposer = MedusaProposer()
+            else:
+                from vllm_amd.spec_decode.ngram import NgramProposer
+
+                self.spec_proposer = NgramProposer(
+                    min_n=sc.ngram_prompt_lookup_min,
+                    max_n=sc.ngram_prompt_lookup_max,
+                    k=sc.num_speculative_tokens,
+                )
         else:
             self.spec_proposer = None
         self.spec_stats_drafted = 0
@@ -367,7 +372,12 @@ class Scheduler:
             elif (self.spec_proposer is not None
                     and request.grammar is None
                     and request.sampling_params.temperature == 0.0):
-                drafts = self.spec_proposer.propose(request.all_token_ids)
+                if getattr(self.spec_proposer, "model_based", False):
+                    drafts = (runner_output.draft_token_ids or {}).get(
+                        req_id)
+                else:
+                    drafts = self.spec_proposer.propose(
+                        request.all_token_ids)
                 request.spec_token_ids = drafts or []
         return outputs
 

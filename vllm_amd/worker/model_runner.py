@@ -155,7 +155,8 @@ class DecodeGraphRunner:
             logits = runner.model.compute_logits(hidden)
         if self.pool is None:
             self.pool = graph.pool()
-        self.graphs[(nb, parts)] = (graph, logits)
+        # hidden is kept for model-based draft proposers (medusa heads).
+        self.graphs[(nb, parts)] = (graph, logits, hidden)
         return self.graphs[(nb, parts)]
 
     def run(self, n: int, nb: int, parts: int, input_ids, positions,
@@ -187,8 +188,9 @@ class DecodeGraphRunner:
         entry = self.graphs.get((nb, parts))
         if entry is None:
             entry = self._capture(nb, parts)
-        graph, logits = entry
+        graph, logits, hidden = entry
         graph.replay()
+        self.last_hidden = hidden
         return logits[:n]
 
 
@@ -206,6 +208,7 @@ class ModelRunner:
                             else self.dtype)
         self.kv_caches: list[torch.Tensor] = []
         self.model: Optional[torch.nn.Module] = None
+        self.medusa = None  # MedusaHeads when spec_decode_method=medusa
         self.graph_runner: Optional[DecodeGraphRunner] = None
         self.sampler = Sampler()
         self.max_model_len = config.model_config.max_model_len
@@ -242,6 +245,19 @@ class ModelRunner:
 
     def load_model(self) -> None:
         self.model = load_model(self.config.model_config, self.device)
+        sc = self.config.scheduler_config
+        if (sc.num_speculative_tokens > 0
+                and sc.spec_decode_method == "medusa"):
+            from vllm_amd.spec_decode.medusa import MedusaHeads
+
+            self.medusa = MedusaHeads(
+                self.spec.hidden_size, self.spec.vocab_size,
+                sc.num_speculative_tokens, self.dtype)
+            if sc.medusa_path:
+                self.medusa.load_safetensors(sc.medusa_path, self.dtype)
+            else:
+                self.medusa.init_dummy(self.config.model_config.seed)
+            self.medusa = self.medusa.to(self.device).eval()
         mc = self.config.model_config
         self.lora_manager = None
         if mc.lora_modules:
@@ -442,12 +458,14 @@ class ModelRunner:
         dev = self.device
         nb = (self.graph_runner.bucket_for(n)
               if self.graph_runner is not None else None)
+        hidden = None
         if nb is not None:
             parts = self.graph_runner.parts_bucket(max_seq_len)
             logits = self.graph_runner.run(
                 n, nb, parts, input_ids, positions, slot_mapping, seq_lens,
                 block_table, ids_dev=ids_dev,
             )
+            hidden = self.graph_runner.last_hidden
         else:
             meta = AttentionMetadata(
                 query_start_loc=torch.arange(n + 1, dtype=torch.int32,
@@ -484,6 +502,14 @@ class ModelRunner:
         sampled_t = s_out.sampled_token_ids
         self._last_sampled = (sampled_t, rows)
 
+        draft_map = None
+        if self.medusa is not None:
+            # Heads condition on this step's hidden; drafts verify next
+            # step. Medusa forces sync scheduling, so the sync here is on
+            # the critical path anyway.
+            draft_map = dict(zip(
+                req_ids, self.medusa.propose(hidden[:n]).cpu().tolist()))
+
         if self.device.type == "cuda" and s_out.logprobs is None:
             pin = self._pin_sampled[self._pin_idx]
             self._pin_idx ^= 1
@@ -495,13 +521,13 @@ class ModelRunner:
                 ev.synchronize()
                 sampled_np = pin[:n].numpy().copy()
                 return self._finish_decode(req_ids, states, rows, sampled_np,
-                                           None)
+                                           None, draft_map)
 
             return AsyncModelOutput(finish)
 
         sampled_np = sampled_t.cpu().numpy()
         out = self._finish_decode(req_ids, states, rows, sampled_np,
-                                  s_out.logprobs)
+                                  s_out.logprobs, draft_map)
         return AsyncModelOutput(lambda: out)
 
     def _with_grammar_masks(self, meta, states):
@@ -516,7 +542,7 @@ class ModelRunner:
         return meta
 
     def _finish_decode(self, req_ids, states, rows, sampled_np,
-                       logprobs) -> ModelRunnerOutput:
+                       logprobs, draft_map=None) -> ModelRunnerOutput:
         self.np_last_tok[rows] = sampled_np
         sampled = sampled_np.tolist()
         sampled_per_req = [[int(t)] for t in sampled]
@@ -532,6 +558,7 @@ class ModelRunner:
             req_ids=req_ids,
             sampled_token_ids=sampled_per_req,
             logprobs=logprobs_per_req or None,
+            draft_token_ids=draft_map,
         )
 
     def _lora_ids_tensor(self, arr):
@@ -723,6 +750,8 @@ class ModelRunner:
 
         sampled_per_req: list[list[int]] = [[] for _ in req_ids]
         logprobs_per_req: dict[str, list[dict[int, float]]] = {}
+        med_rows: list[int] = []
+        med_pos: list[int] = []
         flat = 0
         for j, (r, npos) in enumerate(zip(sampling_rows, sampling_npos)):
             rid = req_ids[r]
@@ -741,6 +770,12 @@ class ModelRunner:
                         break
                     accepted.append(int(row_sampled[d_j + 1]))
             sampled_per_req[r] = accepted
+            if self.medusa is not None:
+                # hidden index of the last ACCEPTED position: heads
+                # there propose the next round's drafts.
+                end = int(query_start_loc[r + 1])
+                med_rows.append(r)
+                med_pos.append(end - npos + len(accepted) - 1)
             if state.grammar is not None:
                 for tok in accepted:
                     if state.grammar_state is None:
@@ -755,6 +790,11 @@ class ModelRunner:
                 rejected = npos - len(accepted)
                 state.num_computed_tokens -= rejected
                 self.np_computed[self._row_of[rid]] -= rejected
+        draft_map = None
+        if self.medusa is not None and med_rows:
+            h = hidden[torch.tensor(med_pos, device=dev)]
+            drafts = self.medusa.propose(h).cpu().tolist()
+            draft_map = {req_ids[r]: d for r, d in zip(med_rows, drafts)}
         # Mixed steps resolve on the CPU; invalidate the device-side
         # sampled-token carry so the next decode reads np_last_tok.
         self._last_sampled = None
@@ -771,6 +811,7 @@ class ModelRunner:
             req_ids=req_ids,
             sampled_token_ids=sampled_per_req,
             logprobs=logprobs_per_req or None,
+            draft_token_ids=draft_map,
         )
 
     # ------------------------------------------------------------------
