@@ -139,8 +139,8 @@ def test_engine_fp8_gpu_e2e():
     from vllm_amd.entrypoints.llm import LLM
     from vllm_amd.sampling_params import SamplingParams
 
-    llm = LLM(model="tiny-llama", dtype="bf16", device="cuda",
-              quantization="fp8", block_size=16, num_gpu_blocks=512,
+    llm = LLM(model="tiny-llama-128", dtype="bf16", device="cuda",
+              quantization="fp8", block_size=64, num_gpu_blocks=256,
               max_model_len=512, max_num_batched_tokens=512,
               max_num_seqs=8)
     prompts = [[(i * 11 + j) % 900 + 3 for j in range(48)]

@@ -276,6 +276,8 @@ MODEL_PRESETS: dict[str, ModelSpec] = {
         global_attn_every_n_layers=2,
         eos_token_id=2,
     ),
+    # head_dim=128: the HIP attention kernels' native geometry, so the
+    # same preset runs the GPU path (test_engine_gpu).
     "tiny-gemma3": ModelSpec(
         name="tiny-gemma3",
         architecture="gemma3",
@@ -283,9 +285,9 @@ MODEL_PRESETS: dict[str, ModelSpec] = {
         hidden_size=128,
         intermediate_size=256,
         num_layers=3,
-        num_heads=4,
-        num_kv_heads=2,
-        head_dim=32,
+        num_heads=2,
+        num_kv_heads=1,
+        head_dim=128,
         rope_theta=1000000.0,
         rope_local_theta=10000.0,
         max_position_embeddings=2048,
@@ -294,7 +296,7 @@ MODEL_PRESETS: dict[str, ModelSpec] = {
         scale_embeddings=True,
         rmsnorm_unit_offset=True,
         qk_norm=True,
-        query_pre_attn_scalar=32,
+        query_pre_attn_scalar=128,
         tie_word_embeddings=True,
         activation="gelu",
         eos_token_id=2,
@@ -548,6 +550,9 @@ class CacheConfig:
     num_gpu_blocks: Optional[int] = None  # None -> profile at startup
     enable_prefix_caching: bool = True
     kv_cache_dtype: str = "auto"  # auto | bf16 | fp8
+    # Host-RAM tier for evicted prefix-cache blocks (0 = off): evicted
+    # GPU blocks are saved D2H and restored on later prefix hits.
+    cpu_offload_gb: float = 0.0
 
 
 @dataclass

@@ -19,10 +19,14 @@ class BlockPool:
     free block (touch) instead of recomputing its KV.
     """
 
-    def __init__(self, num_gpu_blocks: int, enable_caching: bool = True) -> None:
+    def __init__(self, num_gpu_blocks: int, enable_caching: bool = True,
+                 on_evict=None) -> None:
         assert num_gpu_blocks > 0
         self.num_gpu_blocks = num_gpu_blocks
         self.enable_caching = enable_caching
+        # Called as on_evict(block_hash, block_id) when a content-cached
+        # free block is about to be reused (KV offload hook).
+        self.on_evict = on_evict
         self.blocks: list[KVCacheBlock] = [
             KVCacheBlock(block_id=i) for i in range(num_gpu_blocks)
         ]
@@ -64,6 +68,8 @@ class BlockPool:
                 cached = self.cached_block_hash_to_block.get(block.block_hash.value)
                 if cached is block:
                     del self.cached_block_hash_to_block[block.block_hash.value]
+                    if self.on_evict is not None:
+                        self.on_evict(block.block_hash, block.block_id)
                 block.reset_hash()
             block.ref_cnt = 1
             out.append(block)

@@ -53,6 +53,11 @@ class SchedulerOutput:
     # (spec decode; role of scheduled_spec_decode_tokens in the reference).
     scheduled_spec_decode_tokens: dict[str, list[int]] = field(
         default_factory=dict)
+    # Ordered KV offload copies for the runner, executed BEFORE the
+    # forward: ("out", gpu_block_id, host_slot) = D2H save of an evicted
+    # prefix block; ("in", gpu_block_id, host_slot) = H2D restore of a
+    # host-tier prefix hit.
+    kv_swap_ops: list[tuple[str, int, int]] = field(default_factory=list)
 
     @property
     def num_reqs(self) -> int:

@@ -38,6 +38,7 @@ class Scheduler:
         self,
         config: EngineConfig,
         num_gpu_blocks: int,
+        num_host_blocks: int = 0,
     ) -> None:
         self.config = config
         sched_cfg = config.scheduler_config
@@ -71,6 +72,7 @@ class Scheduler:
             num_gpu_blocks=num_gpu_blocks,
             block_size=config.cache_config.block_size,
             enable_caching=config.cache_config.enable_prefix_caching,
+            num_host_blocks=num_host_blocks,
         )
 
         self.requests: dict[str, Request] = {}
@@ -296,6 +298,7 @@ class Scheduler:
             total_num_scheduled_tokens=total,
             finished_req_ids=self.finished_req_ids,
             scheduled_spec_decode_tokens=scheduled_spec_tokens,
+            kv_swap_ops=self.kv_cache_manager.take_swap_ops(),
         )
         self.finished_req_ids = set()
         return out

@@ -33,6 +33,7 @@ class EngineArgs:
     num_gpu_blocks: Optional[int] = None
     enable_prefix_caching: bool = True
     kv_cache_dtype: str = "auto"
+    cpu_offload_gb: float = 0.0
 
     max_num_batched_tokens: int = 8192
     max_num_seqs: int = 256
@@ -73,6 +74,7 @@ class EngineArgs:
                             dest="enable_prefix_caching",
                             action="store_false")
         parser.add_argument("--kv-cache-dtype", type=str, default="auto")
+        parser.add_argument("--cpu-offload-gb", type=float, default=0.0)
         parser.add_argument("--max-num-batched-tokens", type=int,
                             default=8192)
         parser.add_argument("--max-num-seqs", type=int, default=256)
@@ -144,6 +146,7 @@ class EngineArgs:
                 num_gpu_blocks=self.num_gpu_blocks,
                 enable_prefix_caching=self.enable_prefix_caching,
                 kv_cache_dtype=self.kv_cache_dtype,
+                cpu_offload_gb=self.cpu_offload_gb,
             ),
             scheduler_config=SchedulerConfig(
                 max_num_batched_tokens=self.max_num_batched_tokens,

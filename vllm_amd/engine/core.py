@@ -77,8 +77,15 @@ class EngineCore:
             self.num_gpu_blocks = num_blocks
             self.worker.initialize_kv_cache(num_blocks)
 
+        offload_gb = config.cache_config.cpu_offload_gb
+        num_host_blocks = 0
+        if offload_gb > 0:
+            page = self.worker.kv_cache_page_bytes()
+            num_host_blocks = int(offload_gb * (1 << 30) // page)
+            self.worker.allocate_host_kv_pool(num_host_blocks)
         self.scheduler: Optional[Scheduler] = (
-            Scheduler(config, num_gpu_blocks=num_blocks)
+            Scheduler(config, num_gpu_blocks=num_blocks,
+                      num_host_blocks=num_host_blocks)
             if self.is_driver
             else None
         )

@@ -159,6 +159,12 @@ class MultiprocExecutor:
     def initialize_kv_cache(self, num_blocks: int) -> None:
         self.collective_rpc("initialize_kv_cache", num_blocks)
 
+    def kv_cache_page_bytes(self) -> int:
+        return self.collective_rpc("kv_cache_page_bytes")[0]
+
+    def allocate_host_kv_pool(self, num_host_blocks: int) -> None:
+        self.collective_rpc("allocate_host_kv_pool", num_host_blocks)
+
     def sleep(self, level: int = 1) -> None:
         self.collective_rpc("sleep", level)
 

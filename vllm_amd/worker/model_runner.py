@@ -207,6 +207,8 @@ class ModelRunner:
         self.cache_dtype = (torch.float8_e4m3fn if kvd == "fp8"
                             else self.dtype)
         self.kv_caches: list[torch.Tensor] = []
+        self.host_kv_caches: list[torch.Tensor] = []
+        self.num_host_blocks = 0
         self.model: Optional[torch.nn.Module] = None
         self.medusa = None  # MedusaHeads when spec_decode_method=medusa
         self.graph_runner: Optional[DecodeGraphRunner] = None
@@ -326,6 +328,35 @@ class ModelRunner:
                 and not self.spec.is_moe and not self.spec.is_mla):
             self.graph_runner = DecodeGraphRunner(self)
 
+    def allocate_host_kv_pool(self, num_host_blocks: int) -> None:
+        """Pinned host-RAM pool for offloaded prefix blocks: one tensor
+        per layer, slot-major, same per-block layout as the GPU cache so
+        swaps are single contiguous copies per layer."""
+        self.num_host_blocks = num_host_blocks
+        pin = self.device.type == "cuda"
+        self.host_kv_caches = []
+        for cache in self.kv_caches:
+            if self.spec.is_mla:
+                shape = (num_host_blocks,) + tuple(cache.shape[1:])
+            else:
+                # GPU cache [2, N, H, B, D] -> host [M, 2, H, B, D]
+                shape = (num_host_blocks, 2) + tuple(cache.shape[2:])
+            self.host_kv_caches.append(torch.zeros(
+                shape, dtype=cache.dtype, pin_memory=pin))
+
+    def _run_kv_swaps(self, ops) -> None:
+        """Execute the scheduler's ordered offload copies before the
+        forward (swap-outs precede swap-ins in the list by construction;
+        everything is stream-ordered with the step's kernels)."""
+        mla = self.spec.is_mla
+        for kind, bid, slot in ops:
+            for cache, host in zip(self.kv_caches, self.host_kv_caches):
+                src_gpu = cache[bid] if mla else cache[:, bid]
+                if kind == "out":
+                    host[slot].copy_(src_gpu, non_blocking=True)
+                else:
+                    src_gpu.copy_(host[slot], non_blocking=True)
+
     # ------------------------------------------------------------------
     def sleep(self, level: int = 1) -> None:
         """Release GPU memory between serving bursts (role of the
@@ -335,6 +366,7 @@ class ModelRunner:
         weights (wake_up reloads them from the configured source)."""
         self.graph_runner = None
         self.kv_caches = []
+        self.host_kv_caches = []
         self._last_sampled = None
         self._samp_cache_key = self._samp_cache_val = None
         if level >= 2:
@@ -351,9 +383,13 @@ class ModelRunner:
         else:
             self.model = self.model.to(self.device)
         self.allocate_kv_cache(self.num_gpu_blocks)
+        if self.num_host_blocks:
+            self.allocate_host_kv_pool(self.num_host_blocks)
 
     # ------------------------------------------------------------------
     def _update_states(self, so: SchedulerOutput) -> None:
+        if so.kv_swap_ops:
+            self._run_kv_swaps(so.kv_swap_ops)
         for req_id in so.finished_req_ids:
             if self.requests.pop(req_id, None) is not None:
                 row = self._row_of.pop(req_id)

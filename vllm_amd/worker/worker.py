@@ -82,6 +82,12 @@ class Worker:
     def initialize_kv_cache(self, num_blocks: int) -> None:
         self.runner.allocate_kv_cache(num_blocks)
 
+    def kv_cache_page_bytes(self) -> int:
+        return self.runner.kv_cache_page_bytes()
+
+    def allocate_host_kv_pool(self, num_host_blocks: int) -> None:
+        self.runner.allocate_host_kv_pool(num_host_blocks)
+
     def sleep(self, level: int = 1) -> None:
         self.runner.sleep(level)
 
