@@ -329,3 +329,60 @@ def test_prefill_attention_sliding_window(window):
     expect = ref.attention_unified(q, cache, bt, qsl, sl, scale,
                                    num_decodes=0, sliding_window=window)
     assert_close(out, expect, msg=f"window={window}")
+
+
+@pytest.mark.parametrize("hd", [64, 256])
+@pytest.mark.parametrize("group", [1, 2, 8])
+@pytest.mark.parametrize("ctx", [30, 144, 1500])
+def test_decode_attention_head_dims(hd, group, ctx):
+    """head_dim 64 (OPT-class) and 256 (Gemma3-class) decode paths of the
+    HD-templated kernel vs the torch reference."""
+    hip = _hip()
+    Hkv = 2
+    Hq = Hkv * group
+    n = 3
+    q, cache, bt, qsl, sl = _make_paged(n, [1] * n, [ctx] * n, Hq, Hkv,
+                                        D=hd)
+    scale = 1.0 / hd**0.5
+    out = hip.attention_unified(q, cache, bt, qsl, sl, scale,
+                                num_decodes=n, max_seq_len=ctx,
+                                max_query_len=1)
+    expect = ref.attention_unified(q, cache, bt, qsl, sl, scale,
+                                   num_decodes=n)
+    assert_close(out, expect, msg=f"hd={hd} group={group} ctx={ctx}")
+
+
+@pytest.mark.parametrize("hd", [64, 256])
+def test_decode_attention_head_dims_window(hd):
+    hip = _hip()
+    Hkv, group = 2, 2
+    Hq = Hkv * group
+    n, ctx = 3, 300
+    q, cache, bt, qsl, sl = _make_paged(n, [1] * n, [ctx] * n, Hq, Hkv,
+                                        D=hd)
+    scale = 1.0 / hd**0.5
+    out = hip.attention_unified(q, cache, bt, qsl, sl, scale,
+                                num_decodes=n, max_seq_len=ctx,
+                                max_query_len=1, sliding_window=128)
+    expect = ref.attention_unified(q, cache, bt, qsl, sl, scale,
+                                   num_decodes=n, sliding_window=128)
+    assert_close(out, expect, msg=f"hd={hd}")
+
+
+@pytest.mark.parametrize("hd", [64, 256])
+def test_prefill_torch_path_other_head_dims(hd):
+    """Mixed decode+prefill at head_dim != 128: decode rows on the HIP
+    kernel, prefill rows on the explicit torch path, one output."""
+    hip = _hip()
+    Hkv, group = 2, 2
+    Hq = Hkv * group
+    q_lens = [1, 1, 40]
+    ctx_lens = [90, 33, 40]
+    q, cache, bt, qsl, sl = _make_paged(3, q_lens, ctx_lens, Hq, Hkv, D=hd)
+    scale = 1.0 / hd**0.5
+    out = hip.attention_unified(q, cache, bt, qsl, sl, scale,
+                                num_decodes=2, max_seq_len=90,
+                                max_query_len=40)
+    expect = ref.attention_unified(q, cache, bt, qsl, sl, scale,
+                                   num_decodes=2)
+    assert_close(out, expect, msg=f"hd={hd}")

@@ -39,7 +39,10 @@ constexpr int MAX_GROUP = 8;            // max GQA ratio handled in registers
 // Thread layout: 16 lanes per token (lane d covers dims 8d..8d+7 as one
 // short8 = 16 B load), so a 256-thread block streams 16 tokens per
 // iteration; DEC_PART/16 = 32 iterations.
-template <typename Tag, typename CTag, int GROUP, bool FINAL>
+// HD = head_dim (64/128/256). LPT = HD/8 lanes cover one token's dims
+// (8 dims per lane, one 16 B load); SLOTS = DEC_BLOCK/LPT token slots
+// stream in parallel, two tokens per slot-iteration.
+template <typename Tag, typename CTag, int GROUP, bool FINAL, int HD>
 __global__ __launch_bounds__(DEC_BLOCK) void paged_decode_kernel(
     short* __restrict__ out,            // FINAL: [Tdec, Hq, D] (16-bit)
     float* __restrict__ tmp_out,        // else: [Tdec, Hq, parts, D]
@@ -77,14 +80,17 @@ __global__ __launch_bounds__(DEC_BLOCK) void paged_decode_kernel(
     return;
   }
 
-  const int lane16 = threadIdx.x & 15;        // dim-slice owner
-  const int slot = (threadIdx.x >> 4) & 3;    // token slot within wave
-  const int wave = threadIdx.x / WAVE_SIZE;   // 0..3
+  constexpr int LPT = HD / 8;                   // lanes per token
+  constexpr int SLOTS_PER_WAVE = WAVE_SIZE / LPT;
+  constexpr int SLOTS = DEC_BLOCK / LPT;
+  const int lane_d = threadIdx.x & (LPT - 1);   // dim-slice owner
+  const int slot = (threadIdx.x / LPT) % SLOTS_PER_WAVE;
+  const int wave = threadIdx.x / WAVE_SIZE;     // 0..3
 
   // LDS: physical block ids + cross-wave merge scratch.
   __shared__ int blk_ids[DEC_PART / 64];
   __shared__ float ml_red[4][GROUP][2];       // per-wave (m, l)
-  __shared__ float o_red[4][GROUP][128];      // per-wave accumulators
+  __shared__ float o_red[4][GROUP][HD];       // per-wave accumulators
 
   if (threadIdx.x < DEC_PART / 64) {
     const int cache_blk = (part * DEC_PART) / 64 + threadIdx.x;
@@ -97,7 +103,7 @@ __global__ __launch_bounds__(DEC_BLOCK) void paged_decode_kernel(
   float qf[GROUP][8];
   {
     const short* qbase =
-        q + (int64_t)seq * q_stride + hq0 * head_dim + lane16 * 8;
+        q + (int64_t)seq * q_stride + hq0 * head_dim + lane_d * 8;
 #pragma unroll
     for (int g = 0; g < GROUP; ++g) {
       s16x8 v = *reinterpret_cast<const s16x8*>(qbase + g * head_dim);
@@ -122,9 +128,9 @@ __global__ __launch_bounds__(DEC_BLOCK) void paged_decode_kernel(
 
   // Two tokens per (wave, slot) iteration: doubles the K/V loads in
   // flight per lane and halves the online-softmax update chain.
-  const int t0_base = t_begin + wave * 4 + slot;
-  for (int t = t0_base; t < t_end; t += 32) {
-    const int t1 = t + 16;
+  const int t0_base = t_begin + wave * SLOTS_PER_WAVE + slot;
+  for (int t = t0_base; t < t_end; t += 2 * SLOTS) {
+    const int t1 = t + SLOTS;
     const bool has1 = t1 < t_end;
     using CT = CacheTraits<CTag>;
     using cvec = typename CT::vec8;
@@ -132,12 +138,12 @@ __global__ __launch_bounds__(DEC_BLOCK) void paged_decode_kernel(
     const typename CT::elem* base0 = kv_cache +
         ((int64_t)blk_ids[local0 / 64] * num_kv_heads + kvh) *
             head_tile_stride +
-        (int64_t)(t % 64) * head_dim + lane16 * 8;
+        (int64_t)(t % 64) * head_dim + lane_d * 8;
     const int local1 = has1 ? t1 - part * DEC_PART : local0;
     const typename CT::elem* base1 = has1 ? kv_cache +
         ((int64_t)blk_ids[local1 / 64] * num_kv_heads + kvh) *
             head_tile_stride +
-        (int64_t)(t1 % 64) * head_dim + lane16 * 8 : base0;
+        (int64_t)(t1 % 64) * head_dim + lane_d * 8 : base0;
     cvec k0 = *reinterpret_cast<const cvec*>(base0);
     cvec v0 = *reinterpret_cast<const cvec*>(base0 + kv_plane_stride);
     cvec k1 = *reinterpret_cast<const cvec*>(base1);
@@ -158,15 +164,12 @@ __global__ __launch_bounds__(DEC_BLOCK) void paged_decode_kernel(
         sc0 += qf[g][j] * kf0[j];
         sc1 += qf[g][j] * kf1[j];
       }
-      // Reduce over the 16 dim lanes of this slot (consecutive lanes).
-      sc0 += __shfl_xor(sc0, 1, 64);
-      sc1 += __shfl_xor(sc1, 1, 64);
-      sc0 += __shfl_xor(sc0, 2, 64);
-      sc1 += __shfl_xor(sc1, 2, 64);
-      sc0 += __shfl_xor(sc0, 4, 64);
-      sc1 += __shfl_xor(sc1, 4, 64);
-      sc0 += __shfl_xor(sc0, 8, 64);
-      sc1 += __shfl_xor(sc1, 8, 64);
+      // Reduce over the LPT dim lanes of this slot (consecutive lanes).
+#pragma unroll
+      for (int off = 1; off < LPT; off <<= 1) {
+        sc0 += __shfl_xor(sc0, off, 64);
+        sc1 += __shfl_xor(sc1, off, 64);
+      }
       if (!has1) sc1 = -3.0e38f;
       const float m_new = fmaxf(m_s[g], fmaxf(sc0, sc1));
       const float p0 = __expf(sc0 - m_new);
@@ -187,9 +190,9 @@ __global__ __launch_bounds__(DEC_BLOCK) void paged_decode_kernel(
     }
   }
 
-  // ---- merge the 4 slots within each wave (butterfly over lanes 16,32) ---
+  // ---- merge the slots within each wave (butterfly over lanes LPT..32) ---
 #pragma unroll
-  for (int off = 16; off <= 32; off <<= 1) {
+  for (int off = LPT; off <= 32; off <<= 1) {
 #pragma unroll
     for (int g = 0; g < GROUP; ++g) {
       const float m_o = __shfl_xor(m_s[g], off, 64);
@@ -208,20 +211,20 @@ __global__ __launch_bounds__(DEC_BLOCK) void paged_decode_kernel(
   }
 
   // ---- merge the 4 waves via LDS ------------------------------------------
-  if ((threadIdx.x & 63) < 16) {
+  if ((threadIdx.x & 63) < LPT) {
 #pragma unroll
     for (int g = 0; g < GROUP; ++g) {
 #pragma unroll
-      for (int j = 0; j < 8; ++j) o_red[wave][g][lane16 * 8 + j] = acc[g][j];
-      if (lane16 == 0) {
+      for (int j = 0; j < 8; ++j) o_red[wave][g][lane_d * 8 + j] = acc[g][j];
+      if (lane_d == 0) {
         ml_red[wave][g][0] = m_s[g];
         ml_red[wave][g][1] = l_s[g];
       }
     }
   }
   __syncthreads();
-  // 128 threads: one per dim; every thread recomputes the scalar merge.
-  if (threadIdx.x < 128) {
+  // HD threads: one per dim; every thread recomputes the scalar merge.
+  if (threadIdx.x < HD) {
     const int d = threadIdx.x;
 #pragma unroll
     for (int g = 0; g < GROUP; ++g) {
@@ -299,7 +302,8 @@ void paged_decode_attention(torch::Tensor out, torch::Tensor q,
   const int head_dim = q.size(2);
   const int num_kv_heads = kv_cache.size(2);
   const int group = num_q_heads / num_kv_heads;
-  TORCH_CHECK(head_dim == 128, "decode kernel supports head_dim=128");
+  TORCH_CHECK(head_dim == 64 || head_dim == 128 || head_dim == 256,
+              "decode kernel supports head_dim 64/128/256, got ", head_dim);
   TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == head_dim &&
               q.stride(0) % 8 == 0, "q must be head-contiguous [T,H,D]");
   TORCH_CHECK(out.is_contiguous(), "out must be contiguous");
@@ -314,8 +318,9 @@ void paged_decode_attention(torch::Tensor out, torch::Tensor q,
   dim3 grid(num_seqs, num_kv_heads, num_parts);
   dim3 block(DEC_BLOCK);
 
-#define LAUNCH_DEC(TAG, CTAG, G, FIN)                                        \
-  hipLaunchKernelGGL((paged_decode_kernel<TAG, CTAG, G, FIN>), grid, block,  \
+#define LAUNCH_DEC_HD(TAG, CTAG, G, FIN, HD)                                 \
+  hipLaunchKernelGGL((paged_decode_kernel<TAG, CTAG, G, FIN, HD>), grid,     \
+                     block,                                                  \
                      0, stream, (short*)out.data_ptr(),                      \
                      FIN ? nullptr : tmp_out.data_ptr<float>(),              \
                      FIN ? nullptr : tmp_lse.data_ptr<float>(),              \
@@ -325,6 +330,13 @@ void paged_decode_attention(torch::Tensor out, torch::Tensor q,
                      (float)scale, num_q_heads, num_kv_heads, head_dim,      \
                      max_blocks, num_parts, kv_cache.stride(0),              \
                      q.stride(0), (int)sliding_window)
+
+#define LAUNCH_DEC(TAG, CTAG, G, FIN)                                        \
+  switch (head_dim) {                                                        \
+    case 64: LAUNCH_DEC_HD(TAG, CTAG, G, FIN, 64); break;                    \
+    case 256: LAUNCH_DEC_HD(TAG, CTAG, G, FIN, 256); break;                  \
+    default: LAUNCH_DEC_HD(TAG, CTAG, G, FIN, 128); break;                   \
+  }
 
 #define DISPATCH_GROUP(TAG, CTAG, FIN)                                       \
   switch (group) {                                                           \
@@ -373,6 +385,7 @@ void paged_decode_attention(torch::Tensor out, torch::Tensor q,
 #undef DISPATCH_CACHE
 #undef DISPATCH_GROUP
 #undef LAUNCH_DEC
+#undef LAUNCH_DEC_HD
 }
 
 }  // namespace vllm_amd

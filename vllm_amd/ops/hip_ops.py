@@ -128,6 +128,23 @@ def attention_unified(
         )
 
     if num_decodes < num_reqs:
+        if head_dim != 128:
+            # The MFMA prefill kernel is 128-specific (LDS tiling); other
+            # head dims (Gemma3's 256, OPT's 64) take an explicit torch
+            # path for the one-shot prefill rows — decode, the
+            # steady-state loop, runs the HIP kernel above at any
+            # supported head_dim.
+            _warn_prefill_fallback(head_dim)
+            from vllm_amd.ops import _torch_ref
+
+            qsl = query_start_loc[num_decodes:] - num_decodes
+            ref_out = _torch_ref.attention_unified(
+                q[num_decodes:], kv_cache, block_table[num_decodes:],
+                qsl, seq_lens[num_decodes:], scale, num_decodes=0,
+                sliding_window=sliding_window,
+            )
+            out[num_decodes:] = ref_out
+            return out
         if max_query_len <= 0:
             qsl = query_start_loc
             max_query_len = int((qsl[1:] - qsl[:-1]).max().item())
@@ -136,6 +153,20 @@ def attention_unified(
             scale, num_decodes, max_query_len, sliding_window,
         )
     return out
+
+
+_PREFILL_FALLBACK_WARNED = set()
+
+
+def _warn_prefill_fallback(head_dim: int) -> None:
+    if head_dim not in _PREFILL_FALLBACK_WARNED:
+        _PREFILL_FALLBACK_WARNED.add(head_dim)
+        import logging
+
+        logging.getLogger(__name__).warning(
+            "prefill head_dim=%d: MFMA prefill kernel is head_dim-128 "
+            "only; prefill rows use the torch path (decode stays on the "
+            "HIP kernel)", head_dim)
 
 
 # MoE routing (small [T, E] tensors — torch ops are fine here) and the
