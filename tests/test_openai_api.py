@@ -325,3 +325,37 @@ def test_run_batch_jsonl(tmp_path):
     assert results["b"]["response"]["body"]["choices"][0][
         "message"] is not None
     assert results["c"]["error"] is not None
+
+
+def test_embeddings_endpoint(client):
+    r = client.post("/v1/embeddings", json={
+        "model": "tiny-llama",
+        "input": ["hello world", "second input text"],
+    })
+    assert r.status_code == 200, r.text
+    data = r.json()
+    assert data["object"] == "list"
+    assert len(data["data"]) == 2
+    emb = data["data"][0]["embedding"]
+    assert isinstance(emb, list) and len(emb) > 0
+    assert data["usage"]["prompt_tokens"] > 0
+    # base64 format round-trips
+    r2 = client.post("/v1/embeddings", json={
+        "model": "tiny-llama", "input": "hello world",
+        "encoding_format": "base64"})
+    import base64
+    import struct
+    raw = base64.b64decode(r2.json()["data"][0]["embedding"])
+    vec = struct.unpack(f"<{len(raw)//4}f", raw)
+    assert len(vec) == len(emb)
+    # determinism: same input -> same embedding
+    assert list(vec) == pytest.approx(emb, abs=1e-6)
+
+
+def test_embeddings_mean_pooling(client):
+    r = client.post("/v1/embeddings", json={
+        "model": "tiny-llama", "input": "mean pool me",
+        "pooling": "mean"})
+    assert r.status_code == 200, r.text
+    vec = r.json()["data"][0]["embedding"]
+    assert len(vec) > 0

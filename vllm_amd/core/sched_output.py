@@ -78,6 +78,9 @@ class ModelRunnerOutput:
     # Model-based draft proposals (medusa heads): req_id -> draft tokens
     # for the NEXT step, conditioned on the last accepted position.
     draft_token_ids: Optional[dict[str, list[int]]] = None
+    # Pooling requests that completed prefill this step: req_id -> pooled
+    # hidden vector (embedding models).
+    pooled: Optional[dict[str, list[float]]] = None
 
 
 EMPTY_MODEL_RUNNER_OUTPUT = ModelRunnerOutput(req_ids=[], sampled_token_ids=[])
@@ -91,6 +94,7 @@ class EngineCoreOutput:
     stop_reason: Optional[object] = None
     num_cached_tokens: int = 0
     new_logprobs: Optional[list[dict[int, float]]] = None
+    pooled: Optional[list[float]] = None
 
     @property
     def finished(self) -> bool:

@@ -122,6 +122,9 @@ class Scheduler:
     def has_guided_requests(self) -> bool:
         return any(r.grammar is not None for r in self.running)
 
+    def has_pooling_requests(self) -> bool:
+        return any(r.sampling_params.pooling for r in self.running)
+
     def has_unfinished_requests(self) -> bool:
         return bool(self.waiting or self.running)
 
@@ -224,7 +227,10 @@ class Scheduler:
             # Prefix-cache lookup for fresh requests.
             new_computed_blocks = []
             num_computed = request.num_computed_tokens
-            if num_computed == 0 and not resumed:
+            # mean-pooling needs hidden states of EVERY prompt position,
+            # so cached prefixes cannot be skipped.
+            skip_cache = request.sampling_params.pooling == "mean"
+            if num_computed == 0 and not resumed and not skip_cache:
                 (
                     new_computed_blocks,
                     num_computed,
@@ -332,6 +338,21 @@ class Scheduler:
             request = self.requests.get(req_id)
             if request is None or request.is_finished():
                 continue  # aborted mid-step
+
+            pooled_vec = (runner_output.pooled or {}).get(req_id)
+            if pooled_vec is not None:
+                # Pooling request: prefill done, no tokens — finish now.
+                request.status = RequestStatus.FINISHED_STOPPED
+                outputs.append(EngineCoreOutput(
+                    req_id=req_id,
+                    new_token_ids=[],
+                    finish_reason=request.get_finished_reason(),
+                    num_cached_tokens=request.num_cached_tokens,
+                    pooled=pooled_vec,
+                ))
+                self.running.remove(request)
+                self._free_request(request)
+                continue
 
             new_token_ids = sampled_by_req.get(req_id) or []
             num_spec_sched = len(

@@ -152,7 +152,8 @@ class EngineCore:
             == len(scheduler_output.num_scheduled_tokens)
         )
         use_async = (self.async_scheduling and pure_decode
-                     and not self.scheduler.has_guided_requests())
+                     and not self.scheduler.has_guided_requests()
+                     and not self.scheduler.has_pooling_requests())
         if not use_async:
             # Mixed/prefill/empty steps run synchronously: the runner's
             # slow path reads token values the pending step produces.

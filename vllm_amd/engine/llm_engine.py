@@ -35,6 +35,7 @@ class RequestState:
     arrival_time: float = field(default_factory=time.time)
     first_token_time: Optional[float] = None
     finish_time: Optional[float] = None
+    pooled: Optional[list[float]] = None
 
 
 class OutputProcessor:
@@ -102,6 +103,8 @@ class OutputProcessor:
                             to_abort.append(co.req_id)
                         break
 
+            if co.pooled is not None:
+                state.pooled = co.pooled
             if finish_reason:
                 state.finished = True
                 state.finish_reason = finish_reason
@@ -141,6 +144,7 @@ class OutputProcessor:
             outputs=[comp],
             finished=state.finished,
             metrics=metrics,
+            pooled=getattr(state, "pooled", None),
         )
 
     def release(self, request_id: str) -> None:

@@ -16,6 +16,15 @@ class LLM:
         engine_args = EngineArgs(model=model, **kwargs)
         self.engine = LLMEngine(engine_args.create_engine_config())
 
+    def embed(self, prompts, pooling: str = "last"):
+        """Embedding API: returns one pooled hidden-state vector per
+        prompt (role of the reference's LLM.embed / encode)."""
+        from vllm_amd.sampling_params import SamplingParams
+
+        outs = self.generate(
+            prompts, SamplingParams(pooling=pooling, max_tokens=1))
+        return [o.pooled for o in outs]
+
     def generate(
         self,
         prompts: Union[str, list[int], Sequence[Union[str, list[int]]]],

@@ -25,8 +25,9 @@ from vllm_amd.entrypoints.openai.protocol import (
     ChatChoice, ChatCompletionMessage, ChatCompletionRequest,
     ChatCompletionResponse, ChatCompletionStreamResponse, ChatStreamChoice,
     CompletionChoice, CompletionRequest, CompletionResponse, DeltaMessage,
-    DetokenizeRequest, DetokenizeResponse, ErrorResponse, ModelCard,
-    ModelList, TokenizeRequest, TokenizeResponse, UsageInfo, random_id,
+    DetokenizeRequest, DetokenizeResponse, EmbeddingData, EmbeddingRequest,
+    EmbeddingResponse, ErrorResponse, ModelCard, ModelList, TokenizeRequest,
+    TokenizeResponse, UsageInfo, random_id,
 )
 
 VERSION = "0.1.0"
@@ -260,6 +261,50 @@ def build_app(state: ServerState) -> FastAPI:
             )],
             usage=usage,
         )
+
+    # ------------------------------------------------------------------
+    @app.post("/v1/embeddings")
+    async def embeddings(req: EmbeddingRequest):
+        import asyncio
+        import base64
+        import struct
+
+        from vllm_amd.sampling_params import SamplingParams
+
+        inputs = req.input
+        if isinstance(inputs, str):
+            inputs = [inputs]
+        elif inputs and isinstance(inputs[0], int):
+            inputs = [inputs]
+        params = SamplingParams(pooling=req.pooling, max_tokens=1)
+        state.num_requests += len(inputs)
+
+        async def one(prompt):
+            final = None
+            async for out in engine.generate(prompt, params,
+                                             random_id("embd")):
+                final = out
+            return final
+
+        try:
+            finals = await asyncio.gather(*(one(p) for p in inputs))
+        except ValueError as e:
+            return _error(str(e))
+        data = []
+        n_prompt = 0
+        for i, final in enumerate(finals):
+            n_prompt += len(final.prompt_token_ids)
+            vec = final.pooled or []
+            if req.encoding_format == "base64":
+                emb = base64.b64encode(
+                    struct.pack(f"<{len(vec)}f", *vec)).decode()
+            else:
+                emb = vec
+            data.append(EmbeddingData(index=i, embedding=emb))
+        state.num_prompt_tokens += n_prompt
+        return EmbeddingResponse(
+            data=data, model=req.model,
+            usage=UsageInfo(prompt_tokens=n_prompt, total_tokens=n_prompt))
 
     # ------------------------------------------------------------------
     @app.post("/v1/chat/completions")

@@ -262,6 +262,28 @@ class ChatCompletionStreamResponse(BaseModel):
     usage: Optional[UsageInfo] = None
 
 
+class EmbeddingRequest(BaseModel):
+    model: str
+    input: Union[str, list[str], list[int], list[list[int]]]
+    encoding_format: Literal["float", "base64"] = "float"
+    user: Optional[str] = None
+    # Extension: pooling strategy (reference pools per model config).
+    pooling: Literal["last", "mean"] = "last"
+
+
+class EmbeddingData(BaseModel):
+    object: Literal["embedding"] = "embedding"
+    index: int
+    embedding: Union[list[float], str]
+
+
+class EmbeddingResponse(BaseModel):
+    object: Literal["list"] = "list"
+    data: list[EmbeddingData]
+    model: str
+    usage: UsageInfo = Field(default_factory=UsageInfo)
+
+
 class ModelCard(BaseModel):
     id: str
     object: Literal["model"] = "model"

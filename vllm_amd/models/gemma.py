@@ -82,8 +82,12 @@ class GemmaAttention(nn.Module):
         q, k, v = self.qkv_proj.split_qkv(qkv)
         q = q.view(T, self.num_heads, self.head_dim)
         k = k.view(T, self.num_kv_heads, self.head_dim)
-        q = self.q_norm(q.reshape(-1, self.head_dim)).view(q.shape)
-        k = self.k_norm(k.reshape(-1, self.head_dim)).view(k.shape)
+        # contiguous(): reshape of a strided head slice can stay a view
+        # (num_kv_heads=1) and the HIP rms_norm requires contiguous rows.
+        q = self.q_norm(
+            q.reshape(-1, self.head_dim).contiguous()).view(q.shape)
+        k = self.k_norm(
+            k.reshape(-1, self.head_dim).contiguous()).view(k.shape)
         self.rotary_emb(positions, q, k)
         return self.o_proj(self.attn(q, k, v))
 

@@ -98,8 +98,12 @@ class LlamaAttention(nn.Module):
         q = q.view(T, self.num_heads, self.head_dim)
         k = k.view(T, self.num_kv_heads, self.head_dim)
         if self.q_norm is not None:
-            q = self.q_norm(q.reshape(-1, self.head_dim)).view(q.shape)
-            k = self.k_norm(k.reshape(-1, self.head_dim)).view(k.shape)
+            # reshape() of a fused-QKV head slice may stay a strided VIEW
+            # (e.g. num_kv_heads=1); the HIP rms_norm wants contiguous.
+            q = self.q_norm(
+                q.reshape(-1, self.head_dim).contiguous()).view(q.shape)
+            k = self.k_norm(
+                k.reshape(-1, self.head_dim).contiguous()).view(k.shape)
         self.rotary_emb(positions, q, k)
         out = self.attn(q, k, v)
         return self.o_proj(out)

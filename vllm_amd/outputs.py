@@ -29,6 +29,8 @@ class RequestOutput:
     outputs: list[CompletionOutput]
     finished: bool
     metrics: Optional[dict] = None
+    # Embedding/pooling requests: the pooled hidden-state vector.
+    pooled: Optional[list[float]] = None
 
     def __repr__(self) -> str:
         return (

@@ -46,6 +46,10 @@ class SamplingParams:
     stop_token_ids: Optional[list[int]] = None
     ignore_eos: bool = False
     max_tokens: Optional[int] = 16
+    # Pooling request ("last" | "mean"): no tokens are sampled — the
+    # request finishes after prefill with the pooled hidden state
+    # (embedding models; role of the reference's PoolingParams).
+    pooling: Optional[str] = None
     min_tokens: int = 0
     logprobs: Optional[int] = None
     prompt_logprobs: Optional[int] = None
@@ -96,6 +100,8 @@ class SamplingParams:
             raise ValueError("min_p must be in [0, 1]")
         if self.max_tokens is not None and self.max_tokens < 1:
             raise ValueError("max_tokens must be >= 1")
+        if self.pooling not in (None, "last", "mean"):
+            raise ValueError("pooling must be 'last' or 'mean'")
         if self.min_tokens < 0:
             raise ValueError("min_tokens must be >= 0")
         if self.temperature < _SAMPLING_EPS:

@@ -40,6 +40,7 @@ class CachedReqState:
     grammar: object = None
     grammar_state: object = None
     lora_id: int = 0
+    pool_acc: object = None  # running hidden sum for mean pooling
 
     @property
     def output_token_ids(self) -> list[int]:
@@ -651,7 +652,9 @@ class ModelRunner:
         num_decodes = sum(1 for _, n in items if n == 1)
 
         total = so.total_num_scheduled_tokens
-        if num_decodes == len(items) and total == len(items):
+        if (num_decodes == len(items) and total == len(items)
+                and not any(self.requests[r].sampling_params.pooling
+                            for r in req_ids)):
             return self._execute_decode(so, req_ids).result()
         input_ids = np.empty(total, dtype=np.int64)
         positions = np.empty(total, dtype=np.int64)
@@ -666,6 +669,7 @@ class ModelRunner:
         spec_map = so.scheduled_spec_decode_tokens or {}
         sampling_rows: list[int] = []  # row in `items` order
         sampling_npos: list[int] = []  # positions sampled per row (spec)
+        pooling_rows: list[tuple] = []  # (rid, row, n, start)
         t = 0
         for i, (rid, n) in enumerate(items):
             state = self.requests[rid]
@@ -689,6 +693,8 @@ class ModelRunner:
             if rid in spec_map:
                 sampling_rows.append(i)
                 sampling_npos.append(len(spec_map[rid]) + 1)
+            elif state.sampling_params.pooling:
+                pooling_rows.append((rid, i, n, start))
             elif start + n >= len(state.token_ids):
                 sampling_rows.append(i)
                 sampling_npos.append(1)
@@ -743,10 +749,29 @@ class ModelRunner:
             self.requests[rid].num_computed_tokens += n
             self.np_computed[self._row_of[rid]] += n
 
+        pooled_map: dict[str, list[float]] = {}
+        for rid, i, n, start in pooling_rows:
+            # hidden is defined: pooling rows force the eager branch.
+            state = self.requests[rid]
+            sp = state.sampling_params
+            seg0 = int(query_start_loc[i])
+            seg1 = int(query_start_loc[i + 1])
+            if sp.pooling == "mean":
+                chunk_sum = hidden[seg0:seg1].float().sum(dim=0)
+                state.pool_acc = (chunk_sum if state.pool_acc is None
+                                  else state.pool_acc + chunk_sum)
+            if start + n >= len(state.token_ids):  # final prompt chunk
+                if sp.pooling == "mean":
+                    vec = state.pool_acc / len(state.token_ids)
+                else:  # last-token pooling
+                    vec = hidden[seg1 - 1].float()
+                pooled_map[rid] = vec.cpu().tolist()
+
         if not sampling_rows:
             return ModelRunnerOutput(
                 req_ids=req_ids,
                 sampled_token_ids=[[] for _ in req_ids],
+                pooled=pooled_map or None,
             )
 
         if logits_all is not None:
@@ -848,6 +873,7 @@ class ModelRunner:
             sampled_token_ids=sampled_per_req,
             logprobs=logprobs_per_req or None,
             draft_token_ids=draft_map,
+            pooled=pooled_map or None,
         )
 
     # ------------------------------------------------------------------
