@@ -606,7 +606,8 @@ class ParallelConfig:
 
     @property
     def needs_distributed(self) -> bool:
-        return self.world_size > 1 or self.tensor_parallel_size > 1
+        return (self.world_size > 1 or self.tensor_parallel_size > 1
+                or self.pipeline_parallel_size > 1)
 
 
 @dataclass

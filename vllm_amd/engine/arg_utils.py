@@ -49,6 +49,7 @@ class EngineArgs:
     async_scheduling: bool = True
 
     tensor_parallel_size: int = 1
+    pipeline_parallel_size: int = 1
     enable_expert_parallel: bool = False
     multiprocess_engine: bool = False
     device: str = "auto"
@@ -97,6 +98,8 @@ class EngineArgs:
                             dest="async_scheduling", action="store_false")
         parser.add_argument("--tensor-parallel-size", "-tp", type=int,
                             default=1)
+        parser.add_argument("--pipeline-parallel-size", "-pp", type=int,
+                            default=1)
         parser.add_argument("--enable-expert-parallel", action="store_true")
         parser.add_argument("--multiprocess-engine", action="store_true",
                             help="run the engine core in its own process")
@@ -121,11 +124,13 @@ class EngineArgs:
         world_size = int(os.environ.get("WORLD_SIZE", "1"))
         pc = ParallelConfig(
             tensor_parallel_size=self.tensor_parallel_size,
+            pipeline_parallel_size=self.pipeline_parallel_size,
             enable_expert_parallel=self.enable_expert_parallel,
             multiprocess_engine=self.multiprocess_engine,
             rank=int(os.environ.get("RANK", "0")),
             local_rank=int(os.environ.get("LOCAL_RANK", "0")),
-            world_size=max(world_size, self.tensor_parallel_size),
+            world_size=max(world_size, self.tensor_parallel_size
+                           * self.pipeline_parallel_size),
         )
         return EngineConfig(
             model_config=ModelConfig(

@@ -33,6 +33,12 @@ class EngineCore:
         import os
 
         self.config = config
+        pp = config.parallel_config.pipeline_parallel_size
+        if pp > 1:
+            # v1 pipeline: synchronous lockstep; no spec decode/pooling.
+            config.scheduler_config.async_scheduling = False
+            if config.scheduler_config.num_speculative_tokens > 0:
+                raise ValueError("speculative decoding requires pp=1")
         tp = config.parallel_config.tensor_parallel_size
         launched_world = int(os.environ.get("WORLD_SIZE", "1"))
         self._multiproc = tp > 1 and launched_world == 1
@@ -41,7 +47,7 @@ class EngineCore:
         # is the whole job (classic SPMD TP); with tp < world each TP group
         # runs an independent engine replica (SPMD data parallelism — the
         # bench's dp mode; role of the reference's DP engine replicas).
-        from vllm_amd.parallel.state import get_tp_group
+        from vllm_amd.parallel.state import get_replica_group
 
         if self._multiproc:
             # Engine owns the scheduler; one spawned worker process per
@@ -49,7 +55,7 @@ class EngineCore:
             from vllm_amd.executor.multiproc import MultiprocExecutor
 
             self.worker = MultiprocExecutor(config)
-            self.world = get_tp_group()  # engine proc: group of 1
+            self.world = get_replica_group()  # engine proc: group of 1
             self.is_driver = True
             num_blocks = self.worker.determine_num_kv_blocks()
             self.num_gpu_blocks = num_blocks
@@ -60,7 +66,7 @@ class EngineCore:
             self.worker.init_device()
             self.worker.load_model()
 
-            self.world = get_tp_group()
+            self.world = get_replica_group()
             self.is_driver = self.world.rank_in_group == 0
 
             # KV sizing must agree across ranks: min over ranks.

@@ -22,6 +22,11 @@ from vllm_amd.layers.linear import (
     RowParallelLinear,
 )
 from vllm_amd.layers.rotary import RotaryEmbedding
+from vllm_amd.parallel.state import (
+    is_first_pp_rank,
+    is_last_pp_rank,
+    pp_layer_range,
+)
 
 
 class LlamaMLP(nn.Module):
@@ -47,7 +52,7 @@ class LlamaMLP(nn.Module):
 
 class LlamaAttention(nn.Module):
     def __init__(self, spec: ModelSpec, layer_idx: int, dtype: torch.dtype,
-                 max_position: int):
+                 max_position: int, cache_idx: int = -1):
         super().__init__()
         self.qkv_proj = QKVParallelLinear(
             spec.hidden_size,
@@ -87,7 +92,7 @@ class LlamaAttention(nn.Module):
             spec.head_dim,
             scale=spec.head_dim**-0.5,
             num_kv_heads=self.num_kv_heads,
-            layer_idx=layer_idx,
+            layer_idx=cache_idx if cache_idx >= 0 else layer_idx,
             sliding_window=window,
         )
 
@@ -110,9 +115,11 @@ class LlamaAttention(nn.Module):
 
 
 class LlamaDecoderLayer(nn.Module):
-    def __init__(self, spec: ModelSpec, layer_idx: int, dtype, max_position):
+    def __init__(self, spec: ModelSpec, layer_idx: int, dtype, max_position,
+                 cache_idx: int = -1):
         super().__init__()
-        self.self_attn = LlamaAttention(spec, layer_idx, dtype, max_position)
+        self.self_attn = LlamaAttention(spec, layer_idx, dtype, max_position,
+                                        cache_idx)
         self.mlp = LlamaMLP(spec, dtype)
         self.input_layernorm = RMSNorm(spec.hidden_size, spec.rms_norm_eps,
                                        dtype=dtype)
@@ -133,26 +140,46 @@ class LlamaDecoderLayer(nn.Module):
 
 
 class LlamaModel(nn.Module):
+    """With pipeline parallelism, this stage materializes only its layer
+    slice [lo, hi) — other slots hold nn.Identity so parameter NAMES keep
+    their global layer indices (checkpoint loading and the name-seeded
+    dummy init stay partition-invariant). The embedding lives on the
+    first stage, final norm on the last; stage boundaries carry one
+    [T, hidden] activation (hidden + residual pre-combined)."""
+
     def __init__(self, config: ModelConfig):
         super().__init__()
         spec = config.spec
         dtype = config.torch_dtype
-        self.embed_tokens = VocabParallelEmbedding(
-            spec.vocab_size, spec.hidden_size, dtype=dtype
+        self.lo, self.hi = pp_layer_range(spec.num_layers)
+        self.embed_tokens = (
+            VocabParallelEmbedding(spec.vocab_size, spec.hidden_size,
+                                   dtype=dtype)
+            if is_first_pp_rank() else None
         )
         self.layers = nn.ModuleList(
             [
-                LlamaDecoderLayer(spec, i, dtype, config.max_model_len)
+                LlamaDecoderLayer(spec, i, dtype, config.max_model_len,
+                                  cache_idx=i - self.lo)
+                if self.lo <= i < self.hi else nn.Identity()
                 for i in range(spec.num_layers)
             ]
         )
-        self.norm = RMSNorm(spec.hidden_size, spec.rms_norm_eps, dtype=dtype)
+        self.norm = (RMSNorm(spec.hidden_size, spec.rms_norm_eps,
+                             dtype=dtype)
+                     if is_last_pp_rank() else None)
 
-    def forward(self, input_ids, positions):
-        hidden = self.embed_tokens(input_ids)
-        residual = None
-        for layer in self.layers:
+    def forward(self, input_ids, positions, hidden_in=None):
+        if self.embed_tokens is not None:
+            hidden = self.embed_tokens(input_ids)
+            residual = None
+        else:
+            hidden = hidden_in
+            residual = None
+        for layer in self.layers[self.lo:self.hi]:
             hidden, residual = layer(positions, hidden, residual)
+        if self.norm is None:
+            return hidden + residual  # stage-boundary activation
         hidden, _ = self.norm(hidden, residual)
         return hidden
 
@@ -163,14 +190,20 @@ class LlamaForCausalLM(nn.Module):
         spec = config.spec
         self.config = config
         self.model = LlamaModel(config)
-        self.lm_head = ParallelLMHead(
-            spec.vocab_size, spec.hidden_size, dtype=config.torch_dtype
+        self.lm_head = (
+            ParallelLMHead(spec.vocab_size, spec.hidden_size,
+                           dtype=config.torch_dtype)
+            if is_last_pp_rank() else None
         )
-        if spec.tie_word_embeddings:
+        if spec.tie_word_embeddings and self.lm_head is not None:
+            if self.model.embed_tokens is None:
+                raise ValueError(
+                    "tie_word_embeddings requires the embedding and the "
+                    "lm_head on the same PP stage (pp=1)")
             self.lm_head.weight = self.model.embed_tokens.weight
 
-    def forward(self, input_ids, positions):
-        return self.model(input_ids, positions)
+    def forward(self, input_ids, positions, hidden_in=None):
+        return self.model(input_ids, positions, hidden_in)
 
     def compute_logits(self, hidden):
         return self.lm_head.compute_logits(hidden)

@@ -37,10 +37,17 @@ def get_model_class(architecture: str):
 
 def initialize_dummy_weights(model: torch.nn.Module, seed: int = 0) -> None:
     """Random-init all weights (reference dummy_loader.py:22 pattern) —
-    the offline benchmark path; no checkpoints are available."""
-    gen = torch.Generator()
-    gen.manual_seed(seed)
-    for param in model.parameters():
+    the offline benchmark path; no checkpoints are available.
+
+    Seeded PER PARAMETER NAME (stable crc32), not by iteration order, so
+    a pipeline-parallel stage holding layers [lo, hi) materializes
+    exactly the weights a single-process model would — pp1 and pp2 runs
+    are bit-comparable."""
+    import zlib
+
+    for name, param in model.named_parameters():
+        gen = torch.Generator()
+        gen.manual_seed(seed ^ zlib.crc32(name.encode()))
         # Uniform in [-1e-3, 1e-3]: keeps bf16 activations finite through
         # 80-layer stacks while exercising real arithmetic.
         cpu_val = torch.rand(param.shape, generator=gen,

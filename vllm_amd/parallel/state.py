@@ -101,6 +101,65 @@ _EP: Optional[GroupCoordinator] = None
 _WORLD: Optional[GroupCoordinator] = None
 
 _SINGLE = GroupCoordinator(rank=0, world_size=1, rank_in_group=0)
+_PP: Optional[GroupCoordinator] = None
+_REPLICA: Optional[GroupCoordinator] = None
+
+
+def get_replica_group() -> GroupCoordinator:
+    """Engine lockstep domain: the pp*tp model replica (== the TP group
+    when pp == 1)."""
+    if _REPLICA is not None:
+        return _REPLICA
+    return get_tp_group()
+
+
+def get_pp_group() -> GroupCoordinator:
+    return _PP if _PP is not None else _SINGLE
+
+
+def get_pp_rank() -> int:
+    return get_pp_group().rank_in_group
+
+
+def get_pp_world_size() -> int:
+    return get_pp_group().world_size
+
+
+def is_first_pp_rank() -> bool:
+    return get_pp_rank() == 0
+
+
+def is_last_pp_rank() -> bool:
+    return get_pp_rank() == get_pp_world_size() - 1
+
+
+def pp_send(t: torch.Tensor, dst_in_group: int) -> None:
+    g = get_pp_group()
+    dist.send(t.contiguous(), dst=g.group_ranks[dst_in_group],
+              group=g.device_group)
+
+
+def pp_recv(t: torch.Tensor, src_in_group: int) -> None:
+    g = get_pp_group()
+    dist.recv(t, src=g.group_ranks[src_in_group], group=g.device_group)
+
+
+def pp_broadcast_object(obj=None, src_in_group: int = 0):
+    """Object broadcast within the PP group (src is a PP-group rank)."""
+    g = get_pp_group()
+    if g.world_size == 1:
+        return obj
+    return g.broadcast_object(obj, src=g.group_ranks[src_in_group])
+
+
+def pp_layer_range(num_layers: int) -> tuple[int, int]:
+    """[lo, hi) decoder layers owned by this PP stage (remainder layers
+    go to the EARLY stages, which also hold the embedding)."""
+    pp, r = get_pp_world_size(), get_pp_rank()
+    per, extra = divmod(num_layers, pp)
+    lo = r * per + min(r, extra)
+    hi = lo + per + (1 if r < extra else 0)
+    return lo, hi
 
 
 def get_tp_group() -> GroupCoordinator:
@@ -129,6 +188,7 @@ def is_initialized() -> bool:
 
 def init_distributed(
     tensor_parallel_size: int = 1,
+    pipeline_parallel_size: int = 1,
     backend: str = "auto",
     rank: Optional[int] = None,
     world_size: Optional[int] = None,
@@ -153,7 +213,8 @@ def init_distributed(
         else int(os.environ.get("LOCAL_RANK", str(rank)))
     )
 
-    if world_size == 1 and tensor_parallel_size == 1:
+    if (world_size == 1 and tensor_parallel_size == 1
+            and pipeline_parallel_size == 1):
         return  # single-process fast path; no distributed state at all
 
     if backend == "auto":
@@ -173,8 +234,12 @@ def init_distributed(
         dist.new_group(backend="gloo") if backend != "gloo" else None
     )
 
-    assert world_size % tensor_parallel_size == 0
-    # TP groups are contiguous ranks (adjacent ranks share xGMI hops).
+    model_world = tensor_parallel_size * pipeline_parallel_size
+    assert world_size % model_world == 0
+    # TP groups are contiguous ranks (adjacent ranks share xGMI hops);
+    # PP groups stride across TP groups within one model replica:
+    # rank = replica*pp*tp + pp_rank*tp + tp_rank.
+    global _PP
     tp_group = None
     my_tp_ranks = None
     for start in range(0, world_size, tensor_parallel_size):
@@ -190,6 +255,47 @@ def init_distributed(
         device_group=tp_group,
         cpu_group=cpu_group,
     )
+    if pipeline_parallel_size > 1:
+        pp_group = None
+        my_pp_ranks = None
+        for rep in range(0, world_size, model_world):
+            for t in range(tensor_parallel_size):
+                ranks = [rep + p * tensor_parallel_size + t
+                         for p in range(pipeline_parallel_size)]
+                g = dist.new_group(ranks, backend=backend)
+                if rank in ranks:
+                    pp_group = g
+                    my_pp_ranks = ranks
+        # cpu_group stays None: object collectives must run on THIS
+        # pp group (the global gloo group would collide across groups).
+        _PP = GroupCoordinator(
+            rank=rank,
+            world_size=pipeline_parallel_size,
+            rank_in_group=my_pp_ranks.index(rank),
+            device_group=pp_group,
+        )
+        _PP.group_ranks = my_pp_ranks
+        # The engine lockstep (scheduler-broadcast) domain with PP is the
+        # whole model replica: the contiguous pp*tp block.
+        global _REPLICA
+        rep_group = None
+        my_rep_ranks = None
+        for rep in range(0, world_size, model_world):
+            ranks = list(range(rep, rep + model_world))
+            g = dist.new_group(ranks, backend=backend)
+            cg = (dist.new_group(ranks, backend="gloo")
+                  if backend != "gloo" else None)
+            if rank in ranks:
+                rep_group = g
+                my_rep_ranks = ranks
+                rep_cpu = cg
+        _REPLICA = GroupCoordinator(
+            rank=rank,
+            world_size=model_world,
+            rank_in_group=my_rep_ranks.index(rank),
+            device_group=rep_group,
+            cpu_group=rep_cpu,
+        )
     _WORLD = GroupCoordinator(
         rank=rank,
         world_size=world_size,
@@ -202,8 +308,8 @@ def init_distributed(
 
 
 def destroy_distributed() -> None:
-    global _TP, _EP, _WORLD
-    _TP = _EP = _WORLD = None
+    global _TP, _EP, _WORLD, _PP, _REPLICA
+    _TP = _EP = _WORLD = _PP = _REPLICA = None
     if dist.is_initialized():
         dist.destroy_process_group()
 

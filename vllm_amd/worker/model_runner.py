@@ -207,6 +207,9 @@ class ModelRunner:
         kvd = config.cache_config.kv_cache_dtype
         self.cache_dtype = (torch.float8_e4m3fn if kvd == "fp8"
                             else self.dtype)
+        # Pipeline parallelism: this rank caches only its layer slice.
+        from vllm_amd.parallel.state import get_pp_world_size
+        self.pp_size = config.parallel_config.pipeline_parallel_size
         self.kv_caches: list[torch.Tensor] = []
         self.host_kv_caches: list[torch.Tensor] = []
         self.num_host_blocks = 0
@@ -278,6 +281,12 @@ class ModelRunner:
             attach_lora_metadata(self.model)
 
     # ------------------------------------------------------------------
+    def _num_local_layers(self) -> int:
+        from vllm_amd.parallel.state import pp_layer_range
+
+        lo, hi = pp_layer_range(self.spec.num_layers)
+        return hi - lo
+
     def kv_cache_page_bytes(self) -> int:
         """Bytes per KV block across all layers on THIS rank."""
         from vllm_amd.parallel.state import get_tp_world_size
@@ -288,11 +297,12 @@ class ModelRunner:
             # Compressed MLA cache: kv_lora + rope values per token,
             # replicated across TP ranks (vs per-head K+V for GQA).
             per_tok = spec.kv_lora_rank + spec.qk_rope_head_dim
-            return self.block_size * per_tok * elt * spec.num_layers
+            return (self.block_size * per_tok * elt
+                    * self._num_local_layers())
         kv_heads = spec.num_kv_heads_per_rank(get_tp_world_size())
         return (
             2 * self.block_size * kv_heads * spec.head_dim * elt
-            * spec.num_layers
+            * self._num_local_layers()
         )
 
     def allocate_kv_cache(self, num_blocks: int) -> None:
@@ -306,7 +316,7 @@ class ModelRunner:
             self.kv_caches = [
                 torch.zeros(num_blocks, self.block_size, per_tok,
                             dtype=self.dtype, device=self.device)
-                for _ in range(spec.num_layers)
+                for _ in range(self._num_local_layers())
             ]
             return
         kv_heads = spec.num_kv_heads_per_rank(get_tp_world_size())
@@ -318,7 +328,7 @@ class ModelRunner:
                 2, num_blocks, kv_heads, self.block_size, spec.head_dim,
                 dtype=self.cache_dtype, device=self.device,
             )
-            for _ in range(spec.num_layers)
+            for _ in range(self._num_local_layers())
         ]
         # MoE models: the segmented-GEMM fused_moe syncs for expert
         # counts, which is not graph-capturable (until the grouped-GEMM
@@ -326,6 +336,7 @@ class ModelRunner:
         if (self.device.type == "cuda"
                 and not self.config.model_config.enforce_eager
                 and not self.config.model_config.lora_modules
+                and self.pp_size == 1
                 and not self.spec.is_moe and not self.spec.is_mla):
             self.graph_runner = DecodeGraphRunner(self)
 
@@ -653,6 +664,7 @@ class ModelRunner:
 
         total = so.total_num_scheduled_tokens
         if (num_decodes == len(items) and total == len(items)
+                and self.pp_size == 1
                 and not any(self.requests[r].sampling_params.pooling
                             for r in req_ids)):
             return self._execute_decode(so, req_ids).result()
@@ -740,8 +752,27 @@ class ModelRunner:
                 lora_ids=self._lora_ids_tensor(lora_np),
                 lora_manager=self.lora_manager,
             )
-            with set_forward_context(ctx):
-                hidden = self.model(input_ids_t, positions_t)
+            if self.pp_size > 1:
+                # Stage boundary: recv [T, hidden] from the previous
+                # stage, send to the next (non-overlapped v1 pipeline —
+                # capacity sharding; microbatched overlap is future work).
+                from vllm_amd.parallel.state import (
+                    get_pp_rank, is_first_pp_rank, is_last_pp_rank,
+                    pp_recv, pp_send)
+
+                hidden_in = None
+                if not is_first_pp_rank():
+                    hidden_in = torch.empty(
+                        total, self.spec.hidden_size, dtype=self.dtype,
+                        device=dev)
+                    pp_recv(hidden_in, get_pp_rank() - 1)
+                with set_forward_context(ctx):
+                    hidden = self.model(input_ids_t, positions_t, hidden_in)
+                if not is_last_pp_rank():
+                    pp_send(hidden, get_pp_rank() + 1)
+            else:
+                with set_forward_context(ctx):
+                    hidden = self.model(input_ids_t, positions_t)
             logits_all = None
 
         # Advance computed counts (python state + persistent rows).
@@ -773,6 +804,30 @@ class ModelRunner:
                 sampled_token_ids=[[] for _ in req_ids],
                 pooled=pooled_map or None,
             )
+
+        if self.pp_size > 1:
+            from vllm_amd.parallel.state import (
+                is_last_pp_rank, pp_broadcast_object)
+
+            if not is_last_pp_rank():
+                # Sampling happens on the last stage; every stage applies
+                # the sampled tokens to stay bookkeeping-consistent (the
+                # first stage embeds them next step).
+                sampled_per_req, logprobs_pp = pp_broadcast_object(
+                    None, src_in_group=self.pp_size - 1)
+                for r, npos in zip(sampling_rows, sampling_npos):
+                    rid = req_ids[r]
+                    state = self.requests[rid]
+                    accepted = sampled_per_req[r]
+                    if accepted:
+                        state.token_ids.extend(accepted)
+                        self.np_last_tok[self._row_of[rid]] = accepted[-1]
+                self._last_sampled = None
+                return ModelRunnerOutput(
+                    req_ids=req_ids,
+                    sampled_token_ids=sampled_per_req,
+                    logprobs=logprobs_pp or None,
+                )
 
         if logits_all is not None:
             # Decode: every row samples, rows already in order.
@@ -868,6 +923,11 @@ class ModelRunner:
                 if lps:
                     logprobs_per_req[req_ids[r]] = lps[
                         : len(sampled_per_req[r])]
+        if self.pp_size > 1:
+            from vllm_amd.parallel.state import pp_broadcast_object
+
+            pp_broadcast_object((sampled_per_req, logprobs_per_req),
+                                src_in_group=self.pp_size - 1)
         return ModelRunnerOutput(
             req_ids=req_ids,
             sampled_token_ids=sampled_per_req,
@@ -911,6 +971,17 @@ class ModelRunner:
         )
         ctx = ForwardContext(attn_metadata=meta, kv_caches=[])
         with set_forward_context(ctx):
-            hidden = self.model(input_ids, positions)
-            # Include logits in the peak (all rows worst case).
-            self.model.compute_logits(hidden[: max_reqs])
+            if self.pp_size > 1:
+                from vllm_amd.parallel.state import (
+                    is_first_pp_rank, is_last_pp_rank)
+
+                hidden_in = (None if is_first_pp_rank() else torch.zeros(
+                    total, self.spec.hidden_size, dtype=self.dtype,
+                    device=self.device))
+                hidden = self.model(input_ids, positions, hidden_in)
+                if is_last_pp_rank():
+                    self.model.compute_logits(hidden[: max_reqs])
+            else:
+                hidden = self.model(input_ids, positions)
+                # Include logits in the peak (all rows worst case).
+                self.model.compute_logits(hidden[: max_reqs])

@@ -38,6 +38,7 @@ class Worker:
         if pc.needs_distributed:
             init_distributed(
                 tensor_parallel_size=pc.tensor_parallel_size,
+                pipeline_parallel_size=pc.pipeline_parallel_size,
                 backend=pc.distributed_backend,
             )
         torch.manual_seed(self.config.model_config.seed)
