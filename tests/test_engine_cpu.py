@@ -341,3 +341,30 @@ def test_sleep_rejected_while_busy():
     core.sleep()
     core.wake_up()
     core.shutdown()
+
+
+def test_parallel_sampling_n():
+    """params.n > 1: n independent sampled branches per prompt (seeded
+    branches differ; greedy branches are identical)."""
+    from vllm_amd.entrypoints.llm import LLM
+    from vllm_amd.sampling_params import SamplingParams
+
+    llm = LLM(model="tiny-llama", dtype="fp32", device="cpu",
+              block_size=16, num_gpu_blocks=128, max_model_len=256,
+              max_num_batched_tokens=128, max_num_seqs=8)
+    prompt = [(3 * j) % 900 + 5 for j in range(16)]
+    [out] = llm.generate(
+        [prompt], SamplingParams(temperature=1.0, seed=7, n=3,
+                                 max_tokens=8, ignore_eos=True))
+    assert len(out.outputs) == 3
+    assert [c.index for c in out.outputs] == [0, 1, 2]
+    assert all(len(c.token_ids) == 8 for c in out.outputs)
+    # different seeds -> at least one branch differs
+    token_sets = {tuple(c.token_ids) for c in out.outputs}
+    assert len(token_sets) > 1
+    # greedy: all branches identical
+    [g] = llm.generate(
+        [prompt], SamplingParams(temperature=0.0, n=2, max_tokens=8,
+                                 ignore_eos=True))
+    assert g.outputs[0].token_ids == g.outputs[1].token_ids
+    llm.shutdown()

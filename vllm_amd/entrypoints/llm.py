@@ -43,17 +43,46 @@ class LLM:
             sampling_params = [SamplingParams()] * n
         elif isinstance(sampling_params, SamplingParams):
             sampling_params = [sampling_params] * n
-        req_ids = []
-        for prompt, params in zip(prompts, sampling_params):
-            req_ids.append(
-                self.engine.add_request(None, prompt, params, lora=lora))
-        order = {rid: i for i, rid in enumerate(req_ids)}
+        # Parallel sampling (params.n > 1): fan out n engine requests per
+        # prompt — the prefix cache dedups the shared prompt KV; seeded
+        # runs offset the seed per branch so branches differ.
+        import dataclasses as _dc
+
+        req_ids = []          # one per engine request
+        req_of = {}           # engine req id -> (prompt_idx, branch_idx)
+        for p_idx, (prompt, params) in enumerate(
+                zip(prompts, sampling_params)):
+            for b in range(max(params.n, 1)):
+                branch = params
+                if params.n > 1:
+                    branch = _dc.replace(
+                        params, n=1,
+                        seed=(params.seed + b
+                              if params.seed is not None else None))
+                rid = self.engine.add_request(None, prompt, branch,
+                                              lora=lora)
+                req_ids.append(rid)
+                req_of[rid] = (p_idx, b)
         finals: dict[str, RequestOutput] = {}
         while self.engine.has_unfinished_requests():
             for out in self.engine.step():
                 if out.finished:
                     finals[out.request_id] = out
-        return [finals[rid] for rid in sorted(finals, key=order.get)]
+        results: list[RequestOutput] = []
+        for p_idx in range(len(prompts)):
+            branches = sorted(
+                (rid for rid in req_ids if req_of[rid][0] == p_idx),
+                key=lambda rid: req_of[rid][1])
+            base = finals[branches[0]]
+            if len(branches) > 1:
+                outs = []
+                for b, rid in enumerate(branches):
+                    comp = finals[rid].outputs[0]
+                    comp.index = b
+                    outs.append(comp)
+                base = _dc.replace(base, outputs=outs)
+            results.append(base)
+        return results
 
     def shutdown(self) -> None:
         self.engine.shutdown()

@@ -179,11 +179,55 @@ def build_app(state: ServerState) -> FastAPI:
         return Response("\n".join(lines) + "\n",
                         media_type="text/plain; version=0.0.4")
 
+    def _branch_params(params, n):
+        """n>1 parallel sampling: one engine request per branch, seed
+        offset per branch so seeded branches differ."""
+        import dataclasses as _dc
+
+        if n <= 1:
+            return [params]
+        return [
+            _dc.replace(params, seed=(params.seed + b
+                                      if params.seed is not None else None))
+            for b in range(n)
+        ]
+
+    async def _merge_streams(gens):
+        """Merge n async generators into (index, item) events."""
+        import asyncio
+
+        queue: asyncio.Queue = asyncio.Queue()
+        DONE = object()
+
+        async def pump(i, g):
+            try:
+                async for item in g:
+                    await queue.put((i, item, None))
+            except Exception as e:  # noqa: BLE001
+                await queue.put((i, None, e))
+            await queue.put((i, DONE, None))
+
+        tasks = [asyncio.ensure_future(pump(i, g))
+                 for i, g in enumerate(gens)]
+        done = 0
+        try:
+            while done < len(gens):
+                i, item, err = await queue.get()
+                if err is not None:
+                    raise err
+                if item is DONE:
+                    done += 1
+                    continue
+                yield i, item
+        finally:
+            for t in tasks:
+                t.cancel()
+
     # ------------------------------------------------------------------
     @app.post("/v1/completions")
     async def completions(req: CompletionRequest, raw: Request):
-        if req.n != 1 or (req.best_of or 1) != 1:
-            return _error("only n=1 is supported")
+        if (req.best_of or 1) != 1:
+            return _error("best_of is not supported")
         prompts = req.prompt
         if isinstance(prompts, str):
             prompts = [prompts]
@@ -193,6 +237,7 @@ def build_app(state: ServerState) -> FastAPI:
             return _error("batched prompts: send one prompt per request")
         prompt = prompts[0]
         params = req.to_sampling_params(req.stream)
+        branches = _branch_params(params, req.n)
         lora = req.model if req.model in state.lora_names else None
         state.num_requests += 1
         rid = random_id("cmpl")
@@ -203,9 +248,10 @@ def build_app(state: ServerState) -> FastAPI:
                 timer = RequestTimer(state.metrics)
                 n_gen = 0
                 n_prompt = 0
+                gens = [engine.generate(prompt, bp, f"{rid}-{b}", lora=lora)
+                        for b, bp in enumerate(branches)]
                 try:
-                    async for out in engine.generate(prompt, params, rid,
-                                                     lora=lora):
+                    async for b, out in _merge_streams(gens):
                         comp = out.outputs[0]
                         state.num_generation_tokens += len(comp.token_ids)
                         timer.on_tokens(len(comp.token_ids))
@@ -217,7 +263,7 @@ def build_app(state: ServerState) -> FastAPI:
                             "created": int(time.time()),
                             "model": req.model,
                             "choices": [{
-                                "index": 0,
+                                "index": b,
                                 "text": comp.text,
                                 "logprobs": None,
                                 "finish_reason": comp.finish_reason,
@@ -231,36 +277,46 @@ def build_app(state: ServerState) -> FastAPI:
                     yield f"data: {json.dumps(err)}\n\n"
             return StreamingResponse(gen(), media_type="text/event-stream")
 
+        import asyncio
+
         from vllm_amd.metrics import RequestTimer
         timer = RequestTimer(state.metrics)
-        final = None
-        try:
-            async for out in engine.generate(prompt, params, rid, lora=lora):
+
+        async def run_branch(b, bp):
+            final = None
+            async for out in engine.generate(prompt, bp, f"{rid}-{b}",
+                                             lora=lora):
                 final = out
+            return final
+
+        try:
+            finals = await asyncio.gather(
+                *(run_branch(b, bp) for b, bp in enumerate(branches)))
         except ValueError as e:
             return _error(str(e))
-        comp = final.outputs[0]
-        timer.on_finish(len(final.prompt_token_ids), len(comp.token_ids))
-        state.num_prompt_tokens += len(final.prompt_token_ids)
-        state.num_generation_tokens += len(comp.token_ids)
+        n_prompt = len(finals[0].prompt_token_ids)
+        n_gen = sum(len(f.outputs[0].token_ids) for f in finals)
+        timer.on_finish(n_prompt, n_gen)
+        state.num_prompt_tokens += n_prompt
+        state.num_generation_tokens += n_gen
         usage = UsageInfo(
-            prompt_tokens=len(final.prompt_token_ids),
-            completion_tokens=len(comp.token_ids),
-            total_tokens=len(final.prompt_token_ids) + len(comp.token_ids),
+            prompt_tokens=n_prompt,
+            completion_tokens=n_gen,
+            total_tokens=n_prompt + n_gen,
         )
-        return CompletionResponse(
-            id=rid,
-            model=req.model,
-            choices=[CompletionChoice(
-                index=0,
+        choices = []
+        for b, final in enumerate(finals):
+            comp = final.outputs[0]
+            choices.append(CompletionChoice(
+                index=b,
                 text=(final.prompt or "") + comp.text if req.echo
                 else comp.text,
                 finish_reason=comp.finish_reason,
                 stop_reason=comp.stop_reason
                 if isinstance(comp.stop_reason, (int, str)) else None,
-            )],
-            usage=usage,
-        )
+            ))
+        return CompletionResponse(
+            id=rid, model=req.model, choices=choices, usage=usage)
 
     # ------------------------------------------------------------------
     @app.post("/v1/embeddings")
@@ -309,8 +365,8 @@ def build_app(state: ServerState) -> FastAPI:
     # ------------------------------------------------------------------
     @app.post("/v1/chat/completions")
     async def chat_completions(req: ChatCompletionRequest, raw: Request):
-        if req.n != 1:
-            return _error("only n=1 is supported")
+        if req.n != 1 and req.stream:
+            return _error("n>1 is not supported with streaming chat")
         from vllm_amd.entrypoints import tool_parser as tp
 
         tools_on = bool(req.tools) and req.tool_choice != "none"
@@ -426,54 +482,66 @@ def build_app(state: ServerState) -> FastAPI:
                     yield f"data: {json.dumps(err)}\n\n"
             return StreamingResponse(gen(), media_type="text/event-stream")
 
+        import asyncio
+
         from vllm_amd.metrics import RequestTimer
         timer = RequestTimer(state.metrics)
-        final = None
-        try:
-            async for out in engine.generate(prompt, params, rid, lora=lora):
+        branches = _branch_params(params, req.n)
+
+        async def run_branch(b, bp):
+            final = None
+            async for out in engine.generate(prompt, bp, f"{rid}-{b}",
+                                             lora=lora):
                 final = out
+            return final
+
+        try:
+            finals = await asyncio.gather(
+                *(run_branch(b, bp) for b, bp in enumerate(branches)))
         except ValueError as e:
             return _error(str(e))
-        comp = final.outputs[0]
-        timer.on_finish(len(final.prompt_token_ids), len(comp.token_ids))
-        state.num_prompt_tokens += len(final.prompt_token_ids)
-        state.num_generation_tokens += len(comp.token_ids)
+        n_prompt = len(finals[0].prompt_token_ids)
+        n_gen = sum(len(f.outputs[0].token_ids) for f in finals)
+        timer.on_finish(n_prompt, n_gen)
+        state.num_prompt_tokens += n_prompt
+        state.num_generation_tokens += n_gen
         usage = UsageInfo(
-            prompt_tokens=len(final.prompt_token_ids),
-            completion_tokens=len(comp.token_ids),
-            total_tokens=len(final.prompt_token_ids) + len(comp.token_ids),
+            prompt_tokens=n_prompt,
+            completion_tokens=n_gen,
+            total_tokens=n_prompt + n_gen,
         )
-        text = comp.text
-        reasoning = None
-        if state.reasoning_parser:
-            reasoning, text = tp.split_reasoning(text)
-        tool_calls = None
-        finish = comp.finish_reason or "stop"
-        if named:
-            call = tp.ParsedToolCall(id=tp._call_id(), name=named,
-                                     arguments=text.strip())
-            tool_calls = [call.as_openai(0)]
-            text = None
-            finish = "tool_calls"
-        elif tools_on:
-            text, calls = tp.parse_hermes_tool_calls(text)
-            if calls:
-                tool_calls = [c.as_openai(i) for i, c in enumerate(calls)]
-                if finish == "stop":
-                    finish = "tool_calls"
-        return ChatCompletionResponse(
-            id=rid,
-            model=req.model,
-            choices=[ChatChoice(
-                index=0,
+        choices = []
+        for b, final in enumerate(finals):
+            comp = final.outputs[0]
+            text = comp.text
+            reasoning = None
+            if state.reasoning_parser:
+                reasoning, text = tp.split_reasoning(text)
+            tool_calls = None
+            finish = comp.finish_reason or "stop"
+            if named:
+                call = tp.ParsedToolCall(id=tp._call_id(), name=named,
+                                         arguments=text.strip())
+                tool_calls = [call.as_openai(0)]
+                text = None
+                finish = "tool_calls"
+            elif tools_on:
+                text, calls = tp.parse_hermes_tool_calls(text)
+                if calls:
+                    tool_calls = [c.as_openai(i)
+                                  for i, c in enumerate(calls)]
+                    if finish == "stop":
+                        finish = "tool_calls"
+            choices.append(ChatChoice(
+                index=b,
                 message=ChatCompletionMessage(
                     content=text if text else None,
                     reasoning_content=reasoning,
                     tool_calls=tool_calls),
                 finish_reason=finish,
-            )],
-            usage=usage,
-        )
+            ))
+        return ChatCompletionResponse(
+            id=rid, model=req.model, choices=choices, usage=usage)
 
     return app
 
