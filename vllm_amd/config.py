@@ -278,6 +278,23 @@ MODEL_PRESETS: dict[str, ModelSpec] = {
     ),
     # head_dim=128: the HIP attention kernels' native geometry, so the
     # same preset runs the GPU path (test_engine_gpu).
+    # Uniform sliding window (every layer windowed, Mistral-style):
+    # exercises the KV block-reclaim path.
+    "tiny-swa": ModelSpec(
+        name="tiny-swa",
+        architecture="llama",
+        vocab_size=1024,
+        hidden_size=128,
+        intermediate_size=256,
+        num_layers=2,
+        num_heads=4,
+        num_kv_heads=2,
+        head_dim=32,
+        rope_theta=10000.0,
+        max_position_embeddings=2048,
+        sliding_window=16,
+        eos_token_id=2,
+    ),
     "tiny-gemma3": ModelSpec(
         name="tiny-gemma3",
         architecture="gemma3",

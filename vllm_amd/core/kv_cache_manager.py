@@ -23,6 +23,7 @@ class KVCacheManager:
         block_size: int,
         enable_caching: bool = True,
         num_host_blocks: int = 0,
+        sliding_window: int = 0,
     ) -> None:
         self.block_size = block_size
         self.enable_caching = enable_caching
@@ -40,6 +41,16 @@ class KVCacheManager:
         self.req_to_blocks: dict[str, list[KVCacheBlock]] = defaultdict(list)
         # How many blocks of each request are already content-cached.
         self.num_cached_blocks: dict[str, int] = defaultdict(int)
+        # Uniform sliding-window models (EVERY attention layer windowed,
+        # e.g. Mistral): blocks that slid fully out of the window are
+        # returned to the pool mid-request. The runner's block table
+        # keeps the stale ids (never dereferenced — the kernels compute
+        # t_begin from the window), so only the MEMORY is reclaimed.
+        # Mixed-pattern models (Gemma3 global layers) need per-group
+        # tables — tracked for round 2.
+        self.sliding_window = sliding_window
+        # req_id -> number of leading blocks already returned to the pool.
+        self.num_reclaimed: dict[str, int] = defaultdict(int)
 
     def _on_evict(self, block_hash, block_id: int) -> None:
         slot = self.host_pool.put(block_hash)
@@ -132,6 +143,19 @@ class KVCacheManager:
             new_blocks = self.block_pool.get_new_blocks(num_new_blocks)
             req_blocks.extend(new_blocks)
 
+        # Sliding window: return leading blocks whose every token lies
+        # outside the window of ALL future queries. One block of margin
+        # absorbs spec-decode rollbacks (<= a few tokens).
+        if self.sliding_window > 0:
+            reclaim_below = (
+                total_tokens - self.sliding_window - self.block_size
+            ) // self.block_size
+            done = self.num_reclaimed[request.request_id]
+            if reclaim_below > done:
+                stale = req_blocks[done:reclaim_below]
+                self.block_pool.free_blocks(list(reversed(stale)))
+                self.num_reclaimed[request.request_id] = reclaim_below
+
         # Content-cache the blocks that become full after this step.
         if self.enable_caching:
             num_full_after = total_tokens // self.block_size
@@ -145,6 +169,9 @@ class KVCacheManager:
             # chunked prefill total_tokens <= num_tokens so this holds.
             num_hashable = min(num_full_after, len(request.block_hashes))
             cached = self.num_cached_blocks[request.request_id]
+            # Never (re)hash a reclaimed block: the pool may have handed
+            # it to another owner; tagging it would poison the cache.
+            cached = max(cached, self.num_reclaimed[request.request_id])
             if num_hashable > cached:
                 self.block_pool.cache_full_blocks(
                     req_blocks,
@@ -177,9 +204,11 @@ class KVCacheManager:
     def free(self, request: Request) -> None:
         blocks = self.req_to_blocks.pop(request.request_id, [])
         self.num_cached_blocks.pop(request.request_id, None)
+        reclaimed = self.num_reclaimed.pop(request.request_id, 0)
         # Free in reverse so the tail blocks (least useful as prefix cache)
-        # are evicted first (LRU queue order).
-        self.block_pool.free_blocks(list(reversed(blocks)))
+        # are evicted first (LRU queue order). Skip blocks the sliding
+        # window already returned.
+        self.block_pool.free_blocks(list(reversed(blocks[reclaimed:])))
 
     def get_block_ids(self, request_id: str) -> list[int]:
         return [b.block_id for b in self.req_to_blocks[request_id]]

@@ -68,11 +68,16 @@ class Scheduler:
             self.spec_proposer = None
         self.spec_stats_drafted = 0
         self.spec_stats_accepted = 0
+        spec = config.model_config.spec
+        uniform_window = (spec.sliding_window
+                          if spec.sliding_window > 0
+                          and spec.global_attn_every_n_layers == 0 else 0)
         self.kv_cache_manager = KVCacheManager(
             num_gpu_blocks=num_gpu_blocks,
             block_size=config.cache_config.block_size,
             enable_caching=config.cache_config.enable_prefix_caching,
             num_host_blocks=num_host_blocks,
+            sliding_window=uniform_window,
         )
 
         self.requests: dict[str, Request] = {}
