@@ -368,3 +368,16 @@ def test_parallel_sampling_n():
                                  ignore_eos=True))
     assert g.outputs[0].token_ids == g.outputs[1].token_ids
     llm.shutdown()
+
+
+def test_pooling_one_token_prompt():
+    """1-token-prompt pooling request: must not take any pure-decode fast
+    path (graph replay has no hidden) — regression guard."""
+    from vllm_amd.entrypoints.llm import LLM
+
+    llm = LLM(model="tiny-llama", dtype="fp32", device="cpu",
+              block_size=16, num_gpu_blocks=64, max_model_len=128,
+              max_num_batched_tokens=64, max_num_seqs=4)
+    [vec] = llm.embed([[42]])
+    assert vec is not None and len(vec) == 128
+    llm.shutdown()

@@ -715,10 +715,12 @@ class ModelRunner:
         dev = self.device
         max_seq_len = int(seq_lens.max())
 
-        # hipGraph fast path: pure-decode batch within a captured bucket.
+        # hipGraph fast path: pure-decode batch within a captured bucket
+        # (pooling rows need `hidden`, which graph replay doesn't expose
+        # on this path — they stay eager).
         nb = None
         if (self.graph_runner is not None and num_decodes == len(items)
-                and total == len(items)):
+                and total == len(items) and not pooling_rows):
             nb = self.graph_runner.bucket_for(total)
         if nb is not None:
             parts = self.graph_runner.parts_bucket(max_seq_len)
