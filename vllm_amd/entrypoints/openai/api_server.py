@@ -565,11 +565,24 @@ def main() -> None:
     parser.add_argument("--served-model-name", type=str, default=None)
     parser.add_argument("--reasoning-parser", type=str, default=None,
                         choices=["deepseek_r1"])
+    parser.add_argument("--grpc-port", type=int, default=None,
+                        help="also serve the gRPC Inference service "
+                             "(entrypoints/grpc/inference.proto)")
     EngineArgs.add_cli_args(parser)
     args = parser.parse_args()
     engine_args = EngineArgs.from_cli_args(args)
-    app, _ = make_server(engine_args, args.served_model_name,
-                         reasoning_parser=args.reasoning_parser)
+    app, state = make_server(engine_args, args.served_model_name,
+                             reasoning_parser=args.reasoning_parser)
+    if args.grpc_port is not None:
+        from vllm_amd.entrypoints.grpc.server import make_grpc_server
+
+        @app.on_event("startup")
+        async def _start_grpc() -> None:
+            server, port = make_grpc_server(
+                state.engine, state.lora_names,
+                f"{args.host}:{args.grpc_port}")
+            await server.start()
+            app.state.grpc_server = server
     uvicorn.run(app, host=args.host, port=args.port, log_level="info")
 
 
