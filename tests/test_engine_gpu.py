@@ -227,3 +227,25 @@ def test_sleep_wake_gpu():
     after = [o.outputs[0].token_ids for o in llm.generate(prompts, p)]
     llm.shutdown()
     assert before == after
+
+
+def test_sliding_window_reclaim_gpu():
+    """Uniform-window model on a pool smaller than the total sequence:
+    out-of-window blocks are reclaimed mid-request; stale block-table
+    ids are never dereferenced by the HIP kernels."""
+    from vllm_amd.sampling_params import SamplingParams
+
+    prompt = [(5 * j) % 900 + 3 for j in range(80)]
+    p = SamplingParams(temperature=0.0, max_tokens=200, ignore_eos=True)
+    # reference with a roomy pool
+    llm = _llm(model="tiny-swa-128", num_gpu_blocks=64, block_size=64,
+               max_model_len=512)
+    [ref] = llm.generate([prompt], p)
+    llm.shutdown()
+    # 3 blocks = 192 slots < 280 total tokens; window 64 fits
+    llm = _llm(model="tiny-swa-128", num_gpu_blocks=3, block_size=64,
+               max_model_len=512)
+    [out] = llm.generate([prompt], p)
+    llm.shutdown()
+    assert len(out.outputs[0].token_ids) == 200
+    assert out.outputs[0].token_ids == ref.outputs[0].token_ids

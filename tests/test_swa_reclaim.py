@@ -73,3 +73,23 @@ def test_e2e_window_bound_memory():
     small.shutdown()
     assert len(out.outputs[0].token_ids) == 120
     assert out.outputs[0].token_ids == ref.outputs[0].token_ids
+
+
+def test_e2e_gpu_geometry_on_cpu():
+    """Same shapes as the GPU reclaim test (block 64, window 64, 3-block
+    pool) on the CPU reference path — de-risks the @gpu variant."""
+    from vllm_amd.entrypoints.llm import LLM
+    from vllm_amd.sampling_params import SamplingParams
+
+    prompt = [(5 * j) % 900 + 3 for j in range(80)]
+    p = SamplingParams(temperature=0.0, max_tokens=200, ignore_eos=True)
+    kw = dict(model="tiny-swa-128", dtype="fp32", device="cpu",
+              max_model_len=512, max_num_batched_tokens=512,
+              max_num_seqs=2, block_size=64)
+    big = LLM(num_gpu_blocks=64, **kw)
+    [ref] = big.generate([prompt], p)
+    big.shutdown()
+    small = LLM(num_gpu_blocks=3, **kw)
+    [out] = small.generate([prompt], p)
+    small.shutdown()
+    assert out.outputs[0].token_ids == ref.outputs[0].token_ids

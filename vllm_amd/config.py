@@ -295,6 +295,22 @@ MODEL_PRESETS: dict[str, ModelSpec] = {
         sliding_window=16,
         eos_token_id=2,
     ),
+    # GPU-geometry variant of tiny-swa (head_dim 128, window = 1 block).
+    "tiny-swa-128": ModelSpec(
+        name="tiny-swa-128",
+        architecture="llama",
+        vocab_size=1024,
+        hidden_size=256,
+        intermediate_size=512,
+        num_layers=2,
+        num_heads=2,
+        num_kv_heads=1,
+        head_dim=128,
+        rope_theta=10000.0,
+        max_position_embeddings=2048,
+        sliding_window=64,
+        eos_token_id=2,
+    ),
     "tiny-gemma3": ModelSpec(
         name="tiny-gemma3",
         architecture="gemma3",
