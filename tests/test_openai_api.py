@@ -390,3 +390,29 @@ def test_parallel_sampling_api(client):
             if line.startswith("data: ") and line != "data: [DONE]":
                 idxs.add(json.loads(line[6:])["choices"][0]["index"])
     assert idxs == {0, 1}
+
+
+def test_score_and_rerank(client):
+    r = client.post("/v1/score", json={
+        "model": "tiny-llama", "text_1": "query text",
+        "text_2": ["doc one", "doc two", "query text"]})
+    assert r.status_code == 200, r.text
+    data = r.json()["data"]
+    assert len(data) == 3
+    # identical texts -> maximal self-similarity
+    assert data[2]["score"] == pytest.approx(1.0, abs=1e-5)
+    assert all(-1.0 <= d["score"] <= 1.0 + 1e-6 for d in data)
+
+    r = client.post("/v1/rerank", json={
+        "model": "tiny-llama", "query": "the query",
+        "documents": ["aaa", "the query", "bbb"], "top_n": 2})
+    assert r.status_code == 200, r.text
+    results = r.json()["results"]
+    assert len(results) == 2
+    # exact-match document ranks first
+    assert results[0]["index"] == 1
+    assert results[0]["relevance_score"] >= results[1]["relevance_score"]
+
+    r = client.post("/pooling", json={
+        "model": "tiny-llama", "input": "pool me"})
+    assert r.status_code == 200 and len(r.json()["data"]) == 1

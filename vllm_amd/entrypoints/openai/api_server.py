@@ -26,8 +26,9 @@ from vllm_amd.entrypoints.openai.protocol import (
     ChatCompletionResponse, ChatCompletionStreamResponse, ChatStreamChoice,
     CompletionChoice, CompletionRequest, CompletionResponse, DeltaMessage,
     DetokenizeRequest, DetokenizeResponse, EmbeddingData, EmbeddingRequest,
-    EmbeddingResponse, ErrorResponse, ModelCard, ModelList, TokenizeRequest,
-    TokenizeResponse, UsageInfo, random_id,
+    EmbeddingResponse, ErrorResponse, ModelCard, ModelList, RerankRequest,
+    RerankResponse, RerankResult, ScoreData, ScoreRequest, ScoreResponse,
+    TokenizeRequest, TokenizeResponse, UsageInfo, random_id,
 )
 
 VERSION = "0.1.0"
@@ -361,6 +362,81 @@ def build_app(state: ServerState) -> FastAPI:
         return EmbeddingResponse(
             data=data, model=req.model,
             usage=UsageInfo(prompt_tokens=n_prompt, total_tokens=n_prompt))
+
+    async def _embed_many(prompts, pooling):
+        import asyncio
+
+        from vllm_amd.sampling_params import SamplingParams
+
+        params = SamplingParams(pooling=pooling, max_tokens=1)
+
+        async def one(prompt):
+            final = None
+            async for out in engine.generate(prompt, params,
+                                             random_id("pool")):
+                final = out
+            return final
+
+        return await asyncio.gather(*(one(p) for p in prompts))
+
+    def _cosine(a, b):
+        import math
+
+        dot = sum(x * y for x, y in zip(a, b))
+        na = math.sqrt(sum(x * x for x in a)) or 1e-12
+        nb = math.sqrt(sum(x * x for x in b)) or 1e-12
+        return dot / (na * nb)
+
+    @app.post("/v1/score")
+    async def score(req: ScoreRequest):
+        """Similarity scores between text_1 and text_2 (embedding-model
+        scoring: cosine of pooled hidden states; the reference's
+        cross-encoder path needs a classifier-head model class)."""
+        t1 = [req.text_1] if isinstance(req.text_1, str) else req.text_1
+        t2 = [req.text_2] if isinstance(req.text_2, str) else req.text_2
+        if len(t1) == 1 and len(t2) > 1:
+            t1 = t1 * len(t2)
+        if len(t1) != len(t2):
+            return _error("text_1/text_2 length mismatch")
+        finals = await _embed_many(t1 + t2, req.pooling)
+        n = len(t1)
+        n_prompt = sum(len(f.prompt_token_ids) for f in finals)
+        data = [ScoreData(index=i,
+                          score=_cosine(finals[i].pooled,
+                                        finals[n + i].pooled))
+                for i in range(n)]
+        state.num_requests += 1
+        return ScoreResponse(
+            data=data, model=req.model,
+            usage=UsageInfo(prompt_tokens=n_prompt,
+                            total_tokens=n_prompt))
+
+    @app.post("/rerank")
+    @app.post("/v1/rerank")
+    async def rerank(req: RerankRequest):
+        finals = await _embed_many([req.query] + req.documents,
+                                   req.pooling)
+        qv = finals[0].pooled
+        scored = sorted(
+            ((i, _cosine(qv, f.pooled))
+             for i, f in enumerate(finals[1:])),
+            key=lambda x: -x[1])
+        if req.top_n:
+            scored = scored[:req.top_n]
+        n_prompt = sum(len(f.prompt_token_ids) for f in finals)
+        state.num_requests += 1
+        return RerankResponse(
+            model=req.model,
+            results=[RerankResult(index=i,
+                                  document={"text": req.documents[i]},
+                                  relevance_score=s)
+                     for i, s in scored],
+            usage=UsageInfo(prompt_tokens=n_prompt,
+                            total_tokens=n_prompt))
+
+    @app.post("/pooling")
+    async def pooling(req: EmbeddingRequest):
+        return await embeddings(req)
 
     # ------------------------------------------------------------------
     @app.post("/v1/chat/completions")

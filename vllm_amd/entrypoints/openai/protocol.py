@@ -284,6 +284,47 @@ class EmbeddingResponse(BaseModel):
     usage: UsageInfo = Field(default_factory=UsageInfo)
 
 
+class ScoreRequest(BaseModel):
+    model: str
+    text_1: Union[str, list[str]]
+    text_2: Union[str, list[str]]
+    pooling: Literal["last", "mean"] = "mean"
+
+
+class ScoreData(BaseModel):
+    object: Literal["score"] = "score"
+    index: int
+    score: float
+
+
+class ScoreResponse(BaseModel):
+    object: Literal["list"] = "list"
+    data: list[ScoreData]
+    model: str
+    usage: UsageInfo = Field(default_factory=UsageInfo)
+
+
+class RerankRequest(BaseModel):
+    model: str
+    query: str
+    documents: list[str]
+    top_n: Optional[int] = None
+    pooling: Literal["last", "mean"] = "mean"
+
+
+class RerankResult(BaseModel):
+    index: int
+    document: dict
+    relevance_score: float
+
+
+class RerankResponse(BaseModel):
+    id: str = Field(default_factory=lambda: random_id("rerank"))
+    model: str
+    results: list[RerankResult]
+    usage: UsageInfo = Field(default_factory=UsageInfo)
+
+
 class ModelCard(BaseModel):
     id: str
     object: Literal["model"] = "model"
