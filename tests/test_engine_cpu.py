@@ -381,3 +381,33 @@ def test_pooling_one_token_prompt():
     [vec] = llm.embed([[42]])
     assert vec is not None and len(vec) == 128
     llm.shutdown()
+
+
+def test_beam_search():
+    """Beam search must find a continuation whose cumulative logprob is
+    >= the greedy continuation's (greedy is one member of the beam)."""
+    import math
+
+    from vllm_amd.entrypoints.llm import LLM
+    from vllm_amd.sampling_params import SamplingParams
+
+    llm = LLM(model="tiny-llama", dtype="fp32", device="cpu",
+              block_size=16, num_gpu_blocks=256, max_model_len=256,
+              max_num_batched_tokens=256, max_num_seqs=8)
+    prompt = [(3 * j) % 900 + 5 for j in range(12)]
+    [beams] = llm.beam_search([prompt], beam_width=3, max_tokens=5)
+    assert 1 <= len(beams) <= 3
+    for toks, score in beams:
+        assert len(toks) <= 5
+        assert math.isfinite(score)
+    # greedy baseline: compute its cumulative logprob via logprobs=1
+    [g] = llm.generate([prompt], SamplingParams(
+        temperature=0.0, max_tokens=5, logprobs=1, ignore_eos=True,
+        detokenize=False))
+    greedy_cum = sum(list(d.values())[0] if d else 0.0
+                     for d in (g.outputs[0].logprobs or []))
+    best = beams[0][1]
+    assert best >= greedy_cum - 1e-4
+    # beams are distinct
+    assert len({tuple(t) for t, _ in beams}) == len(beams)
+    llm.shutdown()

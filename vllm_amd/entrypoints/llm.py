@@ -25,6 +25,68 @@ class LLM:
             prompts, SamplingParams(pooling=pooling, max_tokens=1))
         return [o.pooled for o in outs]
 
+    def beam_search(
+        self,
+        prompts,
+        beam_width: int = 4,
+        max_tokens: int = 16,
+        length_penalty: float = 1.0,
+    ) -> list[list[tuple[list[int], float]]]:
+        """Beam search emulated over the engine (role of the reference's
+        LLM.beam_search, vllm/entrypoints/llm.py beam_search): each round
+        extends every live beam by one token using top-`beam_width`
+        logprobs, then keeps the best `beam_width` beams by cumulative
+        logprob. The prefix cache makes each one-token extension a
+        cache-hit prefill, so a round costs ~one decode step per beam.
+
+        Returns, per prompt, beams as (token_ids, cumulative_logprob)
+        sorted best-first (token_ids exclude the prompt)."""
+        from vllm_amd.sampling_params import SamplingParams
+
+        if isinstance(prompts, str) or (
+            prompts and isinstance(prompts, list)
+            and isinstance(prompts[0], int)
+        ):
+            prompts = [prompts]
+        results = []
+        for prompt in prompts:
+            if isinstance(prompt, str):
+                prompt = self.engine.tokenizer.encode(prompt)
+            # beams: (tokens_so_far_incl_prompt, cum_logprob, finished)
+            beams = [(list(prompt), 0.0, False)]
+            for _ in range(max_tokens):
+                live = [b for b in beams if not b[2]]
+                if not live:
+                    break
+                params = SamplingParams(
+                    temperature=0.0, max_tokens=1, logprobs=beam_width,
+                    ignore_eos=False, detokenize=False)
+                outs = self.generate([b[0] for b in live], params)
+                candidates = [b for b in beams if b[2]]
+                for (toks, cum, _), out in zip(live, outs):
+                    comp = out.outputs[0]
+                    eos_id = self.engine.config.model_config.spec.                         eos_token_id
+                    lps = (comp.logprobs or [{}])[0]
+                    if not lps:
+                        tok = comp.token_ids[0]
+                        candidates.append((toks + [tok], cum, True))
+                        continue
+                    for tok, lp in lps.items():
+                        done = tok == eos_id
+                        candidates.append(
+                            (toks + [int(tok)], cum + lp, done))
+                # keep the best beam_width by length-penalized logprob
+                def key(b):
+                    gen_len = max(len(b[0]) - len(prompt), 1)
+                    return b[1] / (gen_len ** length_penalty)
+
+                beams = sorted(candidates, key=key, reverse=True)
+                beams = beams[:beam_width]
+            results.append(sorted(
+                [(b[0][len(prompt):], b[1]) for b in beams],
+                key=lambda x: -x[1]))
+        return results
+
     def generate(
         self,
         prompts: Union[str, list[int], Sequence[Union[str, list[int]]]],
