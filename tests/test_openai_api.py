@@ -416,3 +416,22 @@ def test_score_and_rerank(client):
     r = client.post("/pooling", json={
         "model": "tiny-llama", "input": "pool me"})
     assert r.status_code == 200 and len(r.json()["data"]) == 1
+
+
+def test_stream_include_usage(client):
+    for url, body in [
+        ("/v1/completions", {"model": "tiny-llama", "prompt": "u",
+                             "max_tokens": 3, "temperature": 0.0,
+                             "ignore_eos": True}),
+        ("/v1/chat/completions", {"model": "tiny-llama",
+                                  "messages": [{"role": "user",
+                                                "content": "u"}],
+                                  "max_tokens": 3, "temperature": 0.0,
+                                  "ignore_eos": True}),
+    ]:
+        body.update(stream=True, stream_options={"include_usage": True})
+        with client.stream("POST", url, json=body) as r:
+            chunks = [json.loads(ln[6:]) for ln in r.iter_lines()
+                      if ln.startswith("data: ") and ln != "data: [DONE]"]
+        assert chunks[-1].get("usage", {}).get("completion_tokens") == 3, url
+        assert chunks[-1]["choices"] == []

@@ -251,12 +251,16 @@ def build_app(state: ServerState) -> FastAPI:
                 n_prompt = 0
                 gens = [engine.generate(prompt, bp, f"{rid}-{b}", lora=lora)
                         for b, bp in enumerate(branches)]
+                seen: dict[int, int] = {}
                 try:
                     async for b, out in _merge_streams(gens):
                         comp = out.outputs[0]
-                        state.num_generation_tokens += len(comp.token_ids)
-                        timer.on_tokens(len(comp.token_ids))
-                        n_gen += len(comp.token_ids)
+                        # comp.token_ids is cumulative; count the delta.
+                        new = len(comp.token_ids) - seen.get(b, 0)
+                        seen[b] = len(comp.token_ids)
+                        state.num_generation_tokens += new
+                        timer.on_tokens(new)
+                        n_gen += new
                         n_prompt = len(out.prompt_token_ids)
                         chunk = {
                             "id": rid,
@@ -272,6 +276,18 @@ def build_app(state: ServerState) -> FastAPI:
                         }
                         yield f"data: {json.dumps(chunk)}\n\n"
                     timer.on_finish(n_prompt, n_gen)
+                    if (req.stream_options or {}).get("include_usage"):
+                        usage_chunk = {
+                            "id": rid, "object": "text_completion",
+                            "created": int(time.time()),
+                            "model": req.model, "choices": [],
+                            "usage": {
+                                "prompt_tokens": n_prompt,
+                                "completion_tokens": n_gen,
+                                "total_tokens": n_prompt + n_gen,
+                            },
+                        }
+                        yield f"data: {json.dumps(usage_chunk)}\n\n"
                     yield "data: [DONE]\n\n"
                 except Exception as e:  # noqa: BLE001
                     err = {"error": {"message": str(e)}}
@@ -520,13 +536,16 @@ def build_app(state: ServerState) -> FastAPI:
                             content=text or None,
                             reasoning_content=reasoning or None))
                     return out, saw
+                seen_toks = 0
                 try:
                     async for out in engine.generate(prompt, params, rid,
                                                      lora=lora):
                         comp = out.outputs[0]
-                        state.num_generation_tokens += len(comp.token_ids)
-                        timer.on_tokens(len(comp.token_ids))
-                        n_gen += len(comp.token_ids)
+                        new = len(comp.token_ids) - seen_toks
+                        seen_toks = len(comp.token_ids)
+                        state.num_generation_tokens += new
+                        timer.on_tokens(new)
+                        n_gen += new
                         n_prompt = len(out.prompt_token_ids)
                         msgs, saw = deltas(comp.text, comp.finish_reason)
                         finish = comp.finish_reason
@@ -552,6 +571,14 @@ def build_app(state: ServerState) -> FastAPI:
                             )
                             yield f"data: {chunk.model_dump_json()}\n\n"
                     timer.on_finish(n_prompt, n_gen)
+                    if (req.stream_options or {}).get("include_usage"):
+                        final_chunk = ChatCompletionStreamResponse(
+                            id=rid, model=req.model, choices=[],
+                            usage=UsageInfo(
+                                prompt_tokens=n_prompt,
+                                completion_tokens=n_gen,
+                                total_tokens=n_prompt + n_gen))
+                        yield f"data: {final_chunk.model_dump_json()}\n\n"
                     yield "data: [DONE]\n\n"
                 except Exception as e:  # noqa: BLE001
                     err = {"error": {"message": str(e)}}
