@@ -411,3 +411,38 @@ def test_beam_search():
     # beams are distinct
     assert len({tuple(t) for t, _ in beams}) == len(beams)
     llm.shutdown()
+
+
+def test_prompt_logprobs():
+    """prompt_logprobs=k: one {token: logprob} dict per prompt token from
+    index 1; invariant under chunked prefill; contains the actual token
+    plus the top-k."""
+    from vllm_amd.entrypoints.llm import LLM
+    from vllm_amd.sampling_params import SamplingParams
+
+    prompt = [(7 * j) % 900 + 3 for j in range(40)]
+    p = SamplingParams(temperature=0.0, max_tokens=2, ignore_eos=True,
+                       prompt_logprobs=3, detokenize=False)
+
+    def run(chunk):
+        llm = LLM(model="tiny-llama", dtype="fp32", device="cpu",
+                  block_size=16, num_gpu_blocks=128, max_model_len=256,
+                  max_num_batched_tokens=chunk, max_num_seqs=4)
+        [out] = llm.generate([prompt], p)
+        llm.shutdown()
+        return out
+
+    out = run(256)
+    assert out.prompt_logprobs is not None
+    assert len(out.prompt_logprobs) == len(prompt) - 1
+    for j, d in enumerate(out.prompt_logprobs):
+        assert prompt[j + 1] in d          # actual next token present
+        assert len(d) >= 3                 # top-k included
+        assert all(v <= 0.0 for v in d.values())
+    # chunked prefill (16-token chunks) must give identical results
+    out2 = run(16)
+    assert len(out2.prompt_logprobs) == len(prompt) - 1
+    for d1, d2 in zip(out.prompt_logprobs, out2.prompt_logprobs):
+        assert set(d1) == set(d2)
+        for t in d1:
+            assert abs(d1[t] - d2[t]) < 1e-5

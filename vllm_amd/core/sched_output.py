@@ -81,6 +81,9 @@ class ModelRunnerOutput:
     # Pooling requests that completed prefill this step: req_id -> pooled
     # hidden vector (embedding models).
     pooled: Optional[dict[str, list[float]]] = None
+    # Prompt logprobs computed this step (chunked-prefill order):
+    # req_id -> one {token_id: logprob} dict per prompt position covered.
+    prompt_logprobs: Optional[dict[str, list[dict[int, float]]]] = None
 
 
 EMPTY_MODEL_RUNNER_OUTPUT = ModelRunnerOutput(req_ids=[], sampled_token_ids=[])
@@ -94,6 +97,7 @@ class EngineCoreOutput:
     stop_reason: Optional[object] = None
     num_cached_tokens: int = 0
     new_logprobs: Optional[list[dict[int, float]]] = None
+    new_prompt_logprobs: Optional[list[dict[int, float]]] = None
     pooled: Optional[list[float]] = None
 
     @property

@@ -232,9 +232,11 @@ class Scheduler:
             # Prefix-cache lookup for fresh requests.
             new_computed_blocks = []
             num_computed = request.num_computed_tokens
-            # mean-pooling needs hidden states of EVERY prompt position,
-            # so cached prefixes cannot be skipped.
-            skip_cache = request.sampling_params.pooling == "mean"
+            # mean-pooling and prompt_logprobs need hidden states /
+            # logits of EVERY prompt position — cached prefixes cannot
+            # be skipped.
+            skip_cache = (request.sampling_params.pooling == "mean"
+                          or bool(request.sampling_params.prompt_logprobs))
             if num_computed == 0 and not resumed and not skip_cache:
                 (
                     new_computed_blocks,
@@ -359,6 +361,11 @@ class Scheduler:
                 self._free_request(request)
                 continue
 
+            new_plp = (runner_output.prompt_logprobs or {}).get(req_id)
+            if new_plp and not request.is_finished():
+                request.prompt_logprob_chunks = (
+                    getattr(request, "prompt_logprob_chunks", []) + new_plp)
+
             new_token_ids = sampled_by_req.get(req_id) or []
             num_spec_sched = len(
                 scheduler_output.scheduled_spec_decode_tokens.get(req_id, ())
@@ -393,7 +400,10 @@ class Scheduler:
                 stop_reason=request.stop_reason,
                 num_cached_tokens=request.num_cached_tokens,
                 new_logprobs=logprobs_by_req.get(req_id),
+                new_prompt_logprobs=getattr(
+                    request, "prompt_logprob_chunks", None),
             )
+            request.prompt_logprob_chunks = None
             outputs.append(out)
             if stopped:
                 self.running.remove(request)

@@ -36,6 +36,7 @@ class RequestState:
     first_token_time: Optional[float] = None
     finish_time: Optional[float] = None
     pooled: Optional[list[float]] = None
+    prompt_logprobs: Optional[list] = None
 
 
 class OutputProcessor:
@@ -105,6 +106,9 @@ class OutputProcessor:
 
             if co.pooled is not None:
                 state.pooled = co.pooled
+            if co.new_prompt_logprobs:
+                state.prompt_logprobs = (
+                    (state.prompt_logprobs or []) + co.new_prompt_logprobs)
             if finish_reason:
                 state.finished = True
                 state.finish_reason = finish_reason
@@ -145,6 +149,7 @@ class OutputProcessor:
             finished=state.finished,
             metrics=metrics,
             pooled=getattr(state, "pooled", None),
+            prompt_logprobs=getattr(state, "prompt_logprobs", None),
         )
 
     def release(self, request_id: str) -> None:

@@ -31,6 +31,9 @@ class RequestOutput:
     metrics: Optional[dict] = None
     # Embedding/pooling requests: the pooled hidden-state vector.
     pooled: Optional[list[float]] = None
+    # SamplingParams.prompt_logprobs: one {token: logprob} dict per
+    # prompt token starting at index 1 (token 0 is unconditioned).
+    prompt_logprobs: Optional[list[dict[int, float]]] = None
 
     def __repr__(self) -> str:
         return (

@@ -682,6 +682,7 @@ class ModelRunner:
         sampling_rows: list[int] = []  # row in `items` order
         sampling_npos: list[int] = []  # positions sampled per row (spec)
         pooling_rows: list[tuple] = []  # (rid, row, n, start)
+        plp_rows: list[tuple] = []      # prompt-logprob rows
         t = 0
         for i, (rid, n) in enumerate(items):
             state = self.requests[rid]
@@ -710,6 +711,9 @@ class ModelRunner:
             elif start + n >= len(state.token_ids):
                 sampling_rows.append(i)
                 sampling_npos.append(1)
+            if (state.sampling_params.prompt_logprobs
+                    and start < state.prompt_len):
+                plp_rows.append((rid, i, n, start))
             t += n
 
         dev = self.device
@@ -782,6 +786,40 @@ class ModelRunner:
             self.requests[rid].num_computed_tokens += n
             self.np_computed[self._row_of[rid]] += n
 
+        plp_map: dict[str, list[dict[int, float]]] = {}
+        for rid, i, n, start in plp_rows:
+            # Logits at position p predict prompt token p+1; emit one
+            # {token: logprob} dict per covered prompt position (top-k
+            # plus the actual next prompt token, reference convention).
+            state = self.requests[rid]
+            k = state.sampling_params.prompt_logprobs
+            s0 = int(query_start_loc[i])
+            # positions start..start+n-1 predict tokens start+1..start+n,
+            # clipped to prompt tokens only (token 0 has no logprob —
+            # consumers index entries from prompt token 1).
+            lo_pos = start
+            hi_pos = min(start + n, state.prompt_len - 1)
+            if hi_pos <= lo_pos:
+                continue
+            seg = hidden[s0 + (lo_pos - start):s0 + (hi_pos - start)]
+            logits = self.model.compute_logits(seg).float()
+            lsm = torch.log_softmax(logits, dim=-1)
+            topv, topi = lsm.topk(k, dim=-1)
+            targets = torch.tensor(
+                state.token_ids[lo_pos + 1:hi_pos + 1],
+                device=lsm.device)
+            tgt_lp = lsm.gather(1, targets.unsqueeze(1)).squeeze(1)
+            entries = plp_map.setdefault(rid, [])
+            topv_l = topv.tolist()
+            topi_l = topi.tolist()
+            tgt_l = tgt_lp.tolist()
+            tids = targets.tolist()
+            for j in range(len(tids)):
+                d = {int(t): float(v)
+                     for t, v in zip(topi_l[j], topv_l[j])}
+                d[int(tids[j])] = float(tgt_l[j])
+                entries.append(d)
+
         pooled_map: dict[str, list[float]] = {}
         for rid, i, n, start in pooling_rows:
             # hidden is defined: pooling rows force the eager branch.
@@ -805,6 +843,7 @@ class ModelRunner:
                 req_ids=req_ids,
                 sampled_token_ids=[[] for _ in req_ids],
                 pooled=pooled_map or None,
+                prompt_logprobs=plp_map or None,
             )
 
         if self.pp_size > 1:
@@ -936,6 +975,7 @@ class ModelRunner:
             logprobs=logprobs_per_req or None,
             draft_token_ids=draft_map,
             pooled=pooled_map or None,
+            prompt_logprobs=plp_map or None,
         )
 
     # ------------------------------------------------------------------
