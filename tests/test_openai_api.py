@@ -435,3 +435,14 @@ def test_stream_include_usage(client):
                       if ln.startswith("data: ") and ln != "data: [DONE]"]
         assert chunks[-1].get("usage", {}).get("completion_tokens") == 3, url
         assert chunks[-1]["choices"] == []
+
+
+def test_completions_prompt_logprobs(client):
+    r = client.post("/v1/completions", json={
+        "model": "tiny-llama", "prompt": [5, 9, 13, 17, 21],
+        "max_tokens": 1, "temperature": 0.0, "ignore_eos": True,
+        "prompt_logprobs": 2})
+    assert r.status_code == 200, r.text
+    plp = r.json()["choices"][0]["prompt_logprobs"]
+    assert len(plp) == 4  # prompt_len - 1
+    assert all(len(d) >= 2 for d in plp)

@@ -328,6 +328,7 @@ def build_app(state: ServerState) -> FastAPI:
                 index=b,
                 text=(final.prompt or "") + comp.text if req.echo
                 else comp.text,
+                prompt_logprobs=final.prompt_logprobs,
                 finish_reason=comp.finish_reason,
                 stop_reason=comp.stop_reason
                 if isinstance(comp.stop_reason, (int, str)) else None,

@@ -46,6 +46,8 @@ class CompletionRequest(BaseModel):
     guided_choice: Optional[list[str]] = None
     guided_regex: Optional[str] = None
     guided_json: Optional[Union[dict, str]] = None
+    # Extension (same name as the reference): per-prompt-token logprobs.
+    prompt_logprobs: Optional[int] = None
 
     def to_sampling_params(self, stream: bool) -> SamplingParams:
         logit_bias = (
@@ -68,6 +70,7 @@ class CompletionRequest(BaseModel):
             max_tokens=self.max_tokens,
             min_tokens=self.min_tokens,
             logprobs=self.logprobs,
+            prompt_logprobs=self.prompt_logprobs,
             logit_bias=logit_bias,
             skip_special_tokens=self.skip_special_tokens,
             guided_choice=self.guided_choice,
@@ -204,6 +207,7 @@ class CompletionChoice(BaseModel):
     index: int
     text: str
     logprobs: Optional[dict] = None
+    prompt_logprobs: Optional[list[dict[int, float]]] = None
     finish_reason: Optional[str] = None
     stop_reason: Optional[Union[int, str]] = None
 
