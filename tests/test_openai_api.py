@@ -446,3 +446,32 @@ def test_completions_prompt_logprobs(client):
     plp = r.json()["choices"][0]["prompt_logprobs"]
     assert len(plp) == 4  # prompt_len - 1
     assert all(len(d) >= 2 for d in plp)
+
+
+def test_responses_api(client):
+    r = client.post("/v1/responses", json={
+        "model": "tiny-llama", "input": "say hi",
+        "instructions": "be brief", "max_output_tokens": 5,
+        "temperature": 0.0})
+    assert r.status_code == 200, r.text
+    data = r.json()
+    assert data["object"] == "response"
+    assert data["status"] == "completed"
+    (item,) = data["output"]
+    assert item["type"] == "message" and item["role"] == "assistant"
+    assert item["content"][0]["type"] == "output_text"
+    assert data["usage"]["output_tokens"] == 5
+
+
+def test_responses_api_stream(client):
+    with client.stream("POST", "/v1/responses", json={
+            "model": "tiny-llama",
+            "input": [{"role": "user", "content": [
+                {"type": "input_text", "text": "stream"}]}],
+            "max_output_tokens": 4, "temperature": 0.0,
+            "stream": True}) as r:
+        events = [ln.split(": ", 1)[1] for ln in r.iter_lines()
+                  if ln.startswith("event: ")]
+    assert events[0] == "response.created"
+    assert "response.output_item.added" in events
+    assert "response.completed" in events

@@ -22,11 +22,12 @@ from fastapi.responses import JSONResponse, Response, StreamingResponse
 from vllm_amd.engine.arg_utils import EngineArgs
 from vllm_amd.engine.async_llm import AsyncLLM
 from vllm_amd.entrypoints.openai.protocol import (
-    ChatChoice, ChatCompletionMessage, ChatCompletionRequest,
+    ChatChoice, ChatCompletionMessage, ChatCompletionRequest, ChatMessage,
     ChatCompletionResponse, ChatCompletionStreamResponse, ChatStreamChoice,
     CompletionChoice, CompletionRequest, CompletionResponse, DeltaMessage,
     DetokenizeRequest, DetokenizeResponse, EmbeddingData, EmbeddingRequest,
-    EmbeddingResponse, ErrorResponse, ModelCard, ModelList, RerankRequest,
+    EmbeddingResponse, ErrorResponse, ModelCard, ModelList,
+    ResponsesRequest, RerankRequest,
     RerankResponse, RerankResult, ScoreData, ScoreRequest, ScoreResponse,
     TokenizeRequest, TokenizeResponse, UsageInfo, random_id,
 )
@@ -335,6 +336,127 @@ def build_app(state: ServerState) -> FastAPI:
             ))
         return CompletionResponse(
             id=rid, model=req.model, choices=choices, usage=usage)
+
+    # ------------------------------------------------------------------
+    @app.post("/v1/responses")
+    async def responses(req: "ResponsesRequest", raw: Request):
+        """OpenAI Responses API over the chat path: input items become
+        chat messages (instructions -> system), output is one assistant
+        message item. Streaming emits the typed response.* SSE events."""
+        from vllm_amd.entrypoints.openai.protocol import random_id
+        from vllm_amd.sampling_params import (
+            RequestOutputKind, SamplingParams)
+
+        msgs: list[ChatMessage] = []
+        if req.instructions:
+            msgs.append(ChatMessage(role="system",
+                                    content=req.instructions))
+        if isinstance(req.input, str):
+            msgs.append(ChatMessage(role="user", content=req.input))
+        else:
+            for item in req.input:
+                content = item.get("content")
+                if isinstance(content, list):
+                    content = "".join(
+                        seg.get("text", "") for seg in content
+                        if seg.get("type") in ("input_text", "text",
+                                               "output_text"))
+                msgs.append(ChatMessage(role=item.get("role", "user"),
+                                        content=content))
+        prompt = apply_chat_template(engine.tokenizer, msgs, True)
+        params = SamplingParams(
+            temperature=req.temperature,
+            top_p=req.top_p,
+            max_tokens=req.max_output_tokens or state.max_model_len,
+            output_kind=(RequestOutputKind.DELTA if req.stream
+                         else RequestOutputKind.FINAL_ONLY),
+        )
+        rid = random_id("resp")
+        msg_id = random_id("msg")
+        state.num_requests += 1
+        created = int(time.time())
+
+        def envelope(status, output, usage=None):
+            return {
+                "id": rid, "object": "response", "created_at": created,
+                "status": status, "model": req.model,
+                "output": output, "metadata": req.metadata or {},
+                **({"usage": usage} if usage else {}),
+            }
+
+        def msg_item(status, text):
+            return {"type": "message", "id": msg_id, "role": "assistant",
+                    "status": status,
+                    "content": [{"type": "output_text", "text": text,
+                                 "annotations": []}]}
+
+        if req.stream:
+            async def gen() -> AsyncGenerator[str, None]:
+                def ev(etype, data):
+                    return (f"event: {etype}\n"
+                            f"data: {json.dumps(data)}\n\n")
+
+                yield ev("response.created",
+                         {"type": "response.created",
+                          "response": envelope("in_progress", [])})
+                yield ev("response.output_item.added",
+                         {"type": "response.output_item.added",
+                          "output_index": 0,
+                          "item": msg_item("in_progress", "")})
+                text = ""
+                n_gen = 0
+                n_prompt = 0
+                try:
+                    async for out in engine.generate(prompt, params, rid):
+                        comp = out.outputs[0]
+                        n_gen = len(comp.token_ids)
+                        n_prompt = len(out.prompt_token_ids)
+                        if comp.text:
+                            text += comp.text
+                            yield ev("response.output_text.delta",
+                                     {"type": "response.output_text.delta",
+                                      "item_id": msg_id,
+                                      "output_index": 0,
+                                      "content_index": 0,
+                                      "delta": comp.text})
+                    yield ev("response.output_text.done",
+                             {"type": "response.output_text.done",
+                              "item_id": msg_id, "output_index": 0,
+                              "content_index": 0, "text": text})
+                    yield ev("response.output_item.done",
+                             {"type": "response.output_item.done",
+                              "output_index": 0,
+                              "item": msg_item("completed", text)})
+                    usage = {"input_tokens": n_prompt,
+                             "output_tokens": n_gen,
+                             "total_tokens": n_prompt + n_gen}
+                    yield ev("response.completed",
+                             {"type": "response.completed",
+                              "response": envelope(
+                                  "completed",
+                                  [msg_item("completed", text)], usage)})
+                except Exception as e:  # noqa: BLE001
+                    yield ev("error", {"type": "error",
+                                       "message": str(e)})
+            return StreamingResponse(gen(), media_type="text/event-stream")
+
+        final = None
+        try:
+            async for out in engine.generate(prompt, params, rid):
+                final = out
+        except ValueError as e:
+            return _error(str(e))
+        comp = final.outputs[0]
+        state.num_prompt_tokens += len(final.prompt_token_ids)
+        state.num_generation_tokens += len(comp.token_ids)
+        usage = {"input_tokens": len(final.prompt_token_ids),
+                 "output_tokens": len(comp.token_ids),
+                 "total_tokens": len(final.prompt_token_ids)
+                 + len(comp.token_ids)}
+        body = envelope("completed", [msg_item("completed", comp.text)],
+                        usage)
+        body["output_text"] = comp.text
+        return body
 
     # ------------------------------------------------------------------
     @app.post("/v1/embeddings")

@@ -288,6 +288,19 @@ class EmbeddingResponse(BaseModel):
     usage: UsageInfo = Field(default_factory=UsageInfo)
 
 
+class ResponsesRequest(BaseModel):
+    """OpenAI Responses API (role of the reference's /v1/responses)."""
+
+    model: str
+    input: Union[str, list[dict[str, Any]]]
+    instructions: Optional[str] = None
+    max_output_tokens: Optional[int] = None
+    temperature: float = 1.0
+    top_p: float = 1.0
+    stream: bool = False
+    metadata: Optional[dict[str, Any]] = None
+
+
 class ScoreRequest(BaseModel):
     model: str
     text_1: Union[str, list[str]]
