@@ -475,3 +475,25 @@ def test_responses_api_stream(client):
     assert events[0] == "response.created"
     assert "response.output_item.added" in events
     assert "response.completed" in events
+
+
+def test_openai_shaped_logprobs(client):
+    r = client.post("/v1/completions", json={
+        "model": "tiny-llama", "prompt": [5, 9, 13], "max_tokens": 3,
+        "temperature": 0.0, "ignore_eos": True, "logprobs": 2})
+    assert r.status_code == 200, r.text
+    lp = r.json()["choices"][0]["logprobs"]
+    assert len(lp["tokens"]) == 3
+    assert len(lp["token_logprobs"]) == 3
+    assert all(v is None or v <= 0 for v in lp["token_logprobs"])
+    assert all(len(t) >= 1 for t in lp["top_logprobs"])
+
+    r = client.post("/v1/chat/completions", json={
+        "model": "tiny-llama",
+        "messages": [{"role": "user", "content": "hi"}],
+        "max_tokens": 3, "temperature": 0.0, "ignore_eos": True,
+        "logprobs": True, "top_logprobs": 2})
+    assert r.status_code == 200, r.text
+    content = r.json()["choices"][0]["logprobs"]["content"]
+    assert len(content) == 3
+    assert {"token", "logprob", "bytes", "top_logprobs"} <= set(content[0])

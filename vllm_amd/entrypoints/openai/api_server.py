@@ -181,6 +181,40 @@ def build_app(state: ServerState) -> FastAPI:
         return Response("\n".join(lines) + "\n",
                         media_type="text/plain; version=0.0.4")
 
+    def _completion_logprobs(comp):
+        """OpenAI completions logprobs object from the engine's raw
+        per-token {token_id: logprob} dicts."""
+        if not comp.logprobs:
+            return None
+        tok = engine.tokenizer
+        tokens, token_logprobs, top = [], [], []
+        for tid, d in zip(comp.token_ids, comp.logprobs):
+            tokens.append(tok.decode([tid]))
+            token_logprobs.append(d.get(tid))
+            top.append({tok.decode([t]): lp for t, lp in d.items()})
+        return {"tokens": tokens, "token_logprobs": token_logprobs,
+                "top_logprobs": top, "text_offset": []}
+
+    def _chat_logprobs(comp):
+        """OpenAI chat logprobs object ({"content": [...]})."""
+        if not comp.logprobs:
+            return None
+        tok = engine.tokenizer
+        content = []
+        for tid, d in zip(comp.token_ids, comp.logprobs):
+            tstr = tok.decode([tid])
+            content.append({
+                "token": tstr,
+                "logprob": d.get(tid),
+                "bytes": list(tstr.encode()),
+                "top_logprobs": [
+                    {"token": tok.decode([t]), "logprob": lp,
+                     "bytes": list(tok.decode([t]).encode())}
+                    for t, lp in d.items()
+                ],
+            })
+        return {"content": content}
+
     def _branch_params(params, n):
         """n>1 parallel sampling: one engine request per branch, seed
         offset per branch so seeded branches differ."""
@@ -329,6 +363,7 @@ def build_app(state: ServerState) -> FastAPI:
                 index=b,
                 text=(final.prompt or "") + comp.text if req.echo
                 else comp.text,
+                logprobs=_completion_logprobs(comp),
                 prompt_logprobs=final.prompt_logprobs,
                 finish_reason=comp.finish_reason,
                 stop_reason=comp.stop_reason
@@ -764,6 +799,7 @@ def build_app(state: ServerState) -> FastAPI:
                     content=text if text else None,
                     reasoning_content=reasoning,
                     tool_calls=tool_calls),
+                logprobs=_chat_logprobs(comp),
                 finish_reason=finish,
             ))
         return ChatCompletionResponse(
