@@ -88,3 +88,60 @@ def test_qwen3_safetensors_roundtrip(tmp_path):
                                   model_b.named_parameters()):
         assert na == nb
         assert torch.equal(pa, pb), na
+
+
+def _export_hf_phi3(model, spec, path):
+    """Phi-3 checkpoint layout: qkv_proj and gate_up_proj pre-fused."""
+    from safetensors.torch import save_file
+
+    t = {}
+    t["model.embed_tokens.weight"] = \
+        model.model.embed_tokens.weight.data[: spec.vocab_size].clone()
+    t["lm_head.weight"] = \
+        model.lm_head.weight.data[: spec.vocab_size].clone()
+    t["model.norm.weight"] = model.model.norm.weight.data.clone()
+    for i, layer in enumerate(model.model.layers):
+        p = f"model.layers.{i}"
+        t[f"{p}.self_attn.qkv_proj.weight"] = \
+            layer.self_attn.qkv_proj.weight.data.clone()
+        t[f"{p}.self_attn.o_proj.weight"] = \
+            layer.self_attn.o_proj.weight.data.clone()
+        t[f"{p}.mlp.gate_up_proj.weight"] = \
+            layer.mlp.gate_up_proj.weight.data.clone()
+        t[f"{p}.mlp.down_proj.weight"] = \
+            layer.mlp.down_proj.weight.data.clone()
+        t[f"{p}.input_layernorm.weight"] = \
+            layer.input_layernorm.weight.data.clone()
+        t[f"{p}.post_attention_layernorm.weight"] = \
+            layer.post_attention_layernorm.weight.data.clone()
+    save_file(t, str(path / "model.safetensors"))
+
+
+def test_phi3_fused_checkpoint_roundtrip(tmp_path):
+    """Phi-3's fused qkv_proj/gate_up_proj checkpoint names load into the
+    llama trunk bit-exactly."""
+    from vllm_amd.config import MODEL_PRESETS, ModelConfig, ModelSpec
+    import dataclasses
+
+    tiny = dataclasses.replace(
+        MODEL_PRESETS["tiny-llama"], name="tiny-phi3",
+        architecture="phi3")
+    MODEL_PRESETS["tiny-phi3"] = tiny
+    try:
+        from vllm_amd.models.registry import load_model
+
+        cfg_a = ModelConfig(model="tiny-phi3", dtype="fp32",
+                            load_format="dummy")
+        model_a = load_model(cfg_a, torch.device("cpu"))
+        _export_hf_phi3(model_a, cfg_a.spec, tmp_path)
+
+        cfg_b = ModelConfig(model="tiny-phi3", dtype="fp32",
+                            load_format="safetensors",
+                            model_path=str(tmp_path))
+        model_b = load_model(cfg_b, torch.device("cpu"))
+        for (na, pa), (nb, pb) in zip(model_a.named_parameters(),
+                                      model_b.named_parameters()):
+            assert na == nb
+            assert torch.equal(pa, pb), na
+    finally:
+        MODEL_PRESETS.pop("tiny-phi3", None)

@@ -224,6 +224,21 @@ MODEL_PRESETS: dict[str, ModelSpec] = {
         activation="gelu",
         eos_token_id=1,
     ),
+    "phi-3-mini": ModelSpec(
+        name="phi-3-mini",
+        architecture="phi3",
+        vocab_size=32064,
+        hidden_size=3072,
+        intermediate_size=8192,
+        num_layers=32,
+        num_heads=32,
+        num_kv_heads=32,
+        head_dim=96,
+        rope_theta=10000.0,
+        max_position_embeddings=4096,
+        rms_norm_eps=1e-5,
+        eos_token_id=32000,
+    ),
     "qwen3-8b": ModelSpec(
         name="qwen3-8b",
         architecture="qwen3",
@@ -459,6 +474,11 @@ def _spec_from_hf_config(path: str) -> ModelSpec:
         architecture = "deepseek"
     elif "gemma3" in arch:
         architecture = "gemma3"
+    elif "phi3" in arch:
+        # Phi-3 is llama-structured (fused qkv/gate_up checkpoints);
+        # longrope scaling beyond the original 4k context is a tracked
+        # round-2 item — contexts <= original_max_position need none.
+        architecture = "phi3"
     elif "qwen3" in arch:
         architecture = "qwen3"
     elif "qwen2" in arch:

@@ -10,9 +10,10 @@ from vllm_amd.config import ModelConfig
 
 
 def get_model_class(architecture: str):
-    if architecture in ("llama", "qwen2", "qwen3"):
-        # Qwen2/Qwen3 are llama-structured (reference qwen2.py/qwen3.py);
-        # they differ only by spec knobs (qkv_bias, qk_norm).
+    if architecture in ("llama", "qwen2", "qwen3", "phi3"):
+        # Qwen2/Qwen3/Phi-3 are llama-structured (reference qwen2.py /
+        # qwen3.py / phi3.py); they differ only by spec knobs
+        # (qkv_bias, qk_norm) and checkpoint fusion (phi3 loader).
         from vllm_amd.models.llama import LlamaForCausalLM
 
         return LlamaForCausalLM

@@ -82,7 +82,14 @@ def load_safetensors_weights(model: torch.nn.Module, config) -> None:
 
         if ".self_attn." in name:
             attn = layer.self_attn
-            if any(p in name for p in ("q_proj", "k_proj", "v_proj")):
+            if "qkv_proj" in name:
+                # Phi-3 layout: q/k/v pre-fused in the checkpoint.
+                spec_ = spec
+                q_sz = spec_.num_heads * spec_.head_dim
+                kv_sz = spec_.num_kv_heads * spec_.head_dim
+                qw, kw, vw = w.split([q_sz, kv_sz, kv_sz], dim=0)
+                attn.qkv_proj.load_qkv(qw, kw, vw)
+            elif any(p in name for p in ("q_proj", "k_proj", "v_proj")):
                 d = pending.setdefault(name.split(".self_attn.")[0], {})
                 which = name.split("self_attn.")[1].split(".")[0]
                 kind = "bias" if name.endswith("bias") else "weight"
@@ -108,7 +115,12 @@ def load_safetensors_weights(model: torch.nn.Module, config) -> None:
 
         if ".mlp." in name:
             mlp = layer.mlp
-            if "gate_proj" in name:
+            if "gate_up_proj" in name:
+                # Phi-3 layout: gate/up pre-fused in the checkpoint.
+                gw, uw = w.chunk(2, dim=0)
+                mlp.gate_up_proj.load_sub_weight(0, gw)
+                mlp.gate_up_proj.load_sub_weight(1, uw)
+            elif "gate_proj" in name:
                 mlp.gate_up_proj.load_sub_weight(0, w)
             elif "up_proj" in name:
                 mlp.gate_up_proj.load_sub_weight(1, w)
