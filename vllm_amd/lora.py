@@ -1,8 +1,10 @@
 """Multi-LoRA serving (role of the reference's vllm/lora/): per-request
 low-rank adapters applied inside the parallel linear layers.
 
-Round-1 scope: TP=1, torch-composed segment matmuls (rows grouped by
-adapter; rank r is small so the A/B GEMMs are cheap). HF PEFT checkpoint
+Torch-composed segment matmuls (rows grouped by adapter; rank r is
+small so the A/B GEMMs are cheap), TP-aware: column-parallel layers take
+this rank's rows of B, row-parallel layers this rank's columns of A
+(partial deltas sum in the layer's all-reduce). HF PEFT checkpoint
 layout (adapter_config.json + adapter_model.safetensors) or in-memory
 tensors. Adapters are served under their registered name: an OpenAI
 request whose `model` equals an adapter name runs with that adapter.
@@ -88,11 +90,16 @@ def apply_lora_slices(
     out: torch.Tensor,
     lora_ids: torch.Tensor,  # [T] int32/int64; 0 = none
     manager: LoRAManager,
-    slices: list[tuple[str, int, int]],  # (module_key, out_off, out_len)
+    # (module_key, out_off_local, out_len, b_off_full, a_off, a_len):
+    # column-parallel layers shard B rows (b_off_full selects this
+    # rank's slice of the FULL adapter B); row-parallel layers shard A
+    # columns (a_off/a_len select this rank's input slice — the partial
+    # deltas sum in the layer's existing all-reduce).
+    slices: list[tuple],
 ) -> None:
     """out[rows, off:off+len] += scaling * (x A^T) B^T per adapter, per
     fused-output slice. Rows are grouped per adapter id (few adapters
-    per batch)."""
+    per batch). TP-aware: adapters store FULL A/B; shards are views."""
     unique = torch.unique(lora_ids)
     for lid_t in unique:
         lid = int(lid_t)
@@ -101,11 +108,14 @@ def apply_lora_slices(
         adapter = manager.get(lid)
         rows = (lora_ids == lid_t).nonzero(as_tuple=True)[0]
         xi = x[rows]
-        for key, off, length in slices:
+        for key, off, length, b_off, a_off, a_len in slices:
             ab = adapter.weights.get(key)
             if ab is None:
                 continue
             A, B = ab
+            if a_len is not None:
+                A = A[:, a_off:a_off + a_len]
+            B = B[b_off:b_off + length]
             delta = (xi @ A.t()) @ B.t() * adapter.scaling
             out[rows, off:off + length] += delta.to(out.dtype)
 
@@ -118,22 +128,36 @@ def attach_lora_metadata(model: torch.nn.Module) -> None:
         QKVParallelLinear, RowParallelLinear,
     )
 
+    from vllm_amd.parallel.state import get_tp_rank
+
+    rank = get_tp_rank()
     for name, mod in model.named_modules():
         if isinstance(mod, QKVParallelLinear):
             base = name.rsplit(".", 1)[0]
-            qs = mod.num_heads * mod.head_dim
+            qs = mod.num_heads * mod.head_dim      # per-rank sizes
             ks = mod.num_kv_heads * mod.head_dim
+            kv_rank = rank // mod.num_kv_head_replicas
+            kv_chunks = max(mod.total_num_kv_heads
+                            // max(mod.num_kv_heads, 1), 1)
             mod.lora_slices = [
-                (f"{base}.q_proj", 0, qs),
-                (f"{base}.k_proj", qs, ks),
-                (f"{base}.v_proj", qs + ks, ks),
+                (f"{base}.q_proj", 0, qs, rank * qs, None, None),
+                (f"{base}.k_proj", qs, ks,
+                 (kv_rank % kv_chunks) * ks, None, None),
+                (f"{base}.v_proj", qs + ks, ks,
+                 (kv_rank % kv_chunks) * ks, None, None),
             ]
         elif isinstance(mod, MergedColumnParallelLinear):
             base = name.rsplit(".", 1)[0]
-            half = mod.weight.shape[0] // 2
+            half = mod.weight.shape[0] // 2        # per-rank half
             mod.lora_slices = [
-                (f"{base}.gate_proj", 0, half),
-                (f"{base}.up_proj", half, half),
+                (f"{base}.gate_proj", 0, half, rank * half, None, None),
+                (f"{base}.up_proj", half, half, rank * half, None, None),
             ]
-        elif isinstance(mod, (ColumnParallelLinear, RowParallelLinear)):
-            mod.lora_slices = [(name, 0, mod.weight.shape[0])]
+        elif isinstance(mod, RowParallelLinear):
+            in_len = mod.input_size_per_partition
+            mod.lora_slices = [
+                (name, 0, mod.weight.shape[0], 0, rank * in_len, in_len)]
+        elif isinstance(mod, ColumnParallelLinear):
+            out_len = mod.weight.shape[0]
+            mod.lora_slices = [
+                (name, 0, out_len, rank * out_len, None, None)]

@@ -267,10 +267,6 @@ class ModelRunner:
         mc = self.config.model_config
         self.lora_manager = None
         if mc.lora_modules:
-            from vllm_amd.parallel.state import get_tp_world_size
-
-            if get_tp_world_size() > 1:
-                raise NotImplementedError("LoRA with TP>1 lands later")
             from vllm_amd.lora import (LoRAAdapter, LoRAManager,
                                        attach_lora_metadata)
 
