@@ -1,0 +1,67 @@
+"""Failure-detection sentinels (role of the reference's EngineDeadError /
+worker monitors): a dead worker or engine-core process surfaces as a
+clean EngineDeadError instead of a pipe hang, and /health turns 503."""
+
+import os
+import signal
+import time
+
+import pytest
+
+from vllm_amd.config import (
+    CacheConfig, DeviceConfig, EngineConfig, ModelConfig, ParallelConfig,
+    SchedulerConfig,
+)
+from vllm_amd.executor.multiproc import EngineDeadError
+
+
+def _config(**kw):
+    return EngineConfig(
+        model_config=ModelConfig(model="tiny-llama", dtype="fp32",
+                                 max_model_len=128),
+        cache_config=CacheConfig(block_size=16, num_gpu_blocks=64),
+        scheduler_config=SchedulerConfig(max_num_batched_tokens=128,
+                                         max_num_seqs=2),
+        parallel_config=ParallelConfig(**kw),
+        device_config=DeviceConfig(device="cpu"),
+    )
+
+
+def test_dead_worker_raises_engine_dead():
+    from vllm_amd.engine.core import EngineCore
+    from vllm_amd.request import Request
+    from vllm_amd.sampling_params import SamplingParams
+
+    core = EngineCore(_config(tensor_parallel_size=2,
+                              multiprocess_engine=False,
+                              distributed_backend="gloo"))
+    assert core._multiproc
+    core.check_health()  # alive
+    # generation works before the fault
+    core.add_request(Request(
+        request_id="r0", prompt_token_ids=list(range(3, 15)),
+        sampling_params=SamplingParams(max_tokens=2, ignore_eos=True)))
+    while core.has_unfinished_requests():
+        core.step()
+    # kill one worker by exact PID
+    victim = core.worker.procs[1]
+    os.kill(victim.pid, signal.SIGKILL)
+    victim.join(timeout=10)
+    with pytest.raises(EngineDeadError):
+        core.worker.check_health()
+    with pytest.raises((EngineDeadError, RuntimeError)):
+        core.worker.collective_rpc("kv_cache_page_bytes")
+    core.worker.shutdown()
+
+
+def test_dead_engine_proc_detected():
+    from vllm_amd.engine.core_client import EngineCoreClient
+
+    client = EngineCoreClient(_config(tensor_parallel_size=1,
+                                      multiprocess_engine=True))
+    client.check_health()
+    os.kill(client._proc.pid, signal.SIGKILL)
+    client._proc.join(timeout=10)
+    with pytest.raises(EngineDeadError):
+        client.check_health()
+    client.shutdown()

@@ -35,6 +35,7 @@ class AsyncLLM:
                                        asyncio.Queue]] = {}
         self._errors: dict[str, Exception] = {}
         self._shutdown = False
+        self._engine_error: Optional[BaseException] = None
         self._counter = 0
         self._counter_lock = threading.Lock()
         self._thread = threading.Thread(
@@ -78,9 +79,17 @@ class AsyncLLM:
                 outputs = self.engine.step()
             except Exception as e:  # noqa: BLE001
                 logger.exception("engine step failed")
+                from vllm_amd.executor.multiproc import EngineDeadError
+
+                if isinstance(e, EngineDeadError):
+                    # Unrecoverable: mark errored (health reports it) and
+                    # fail all in-flight streams.
+                    self._engine_error = e
                 for rid, (loop, out_q) in self._streams.items():
                     loop.call_soon_threadsafe(out_q.put_nowait, e)
                 self._streams.clear()
+                if self._engine_error is not None:
+                    return
                 continue
             for out in outputs:
                 entry = self._streams.get(out.request_id)
@@ -124,6 +133,13 @@ class AsyncLLM:
 
     async def abort(self, request_id: str) -> None:
         self._inbox.put(("abort", request_id))
+
+    def check_health(self) -> None:
+        """Raise when the engine cannot serve (dead worker/engine proc or
+        a fatal engine-loop error)."""
+        if self._engine_error is not None:
+            raise self._engine_error
+        self.engine.check_health()
 
     def sleep(self, level: int = 1) -> None:
         """Release GPU memory (weights to host at level 1, discarded at

@@ -136,6 +136,12 @@ class EngineCore:
     def is_sleeping(self) -> bool:
         return getattr(self, "_sleeping", False)
 
+    def check_health(self) -> None:
+        """Raise EngineDeadError if a worker process died (in-proc
+        workers cannot die independently — no-op)."""
+        if self._multiproc:
+            self.worker.check_health()
+
     # ------------------------------------------------------------------
     def _drain(self) -> list[EngineCoreOutput]:
         if self._pending is None:

@@ -154,6 +154,14 @@ class EngineCoreClient:
     def is_sleeping(self) -> bool:
         return self._sleeping
 
+    def check_health(self) -> None:
+        from vllm_amd.executor.multiproc import EngineDeadError
+
+        if not self._proc.is_alive():
+            raise EngineDeadError(
+                f"engine core process died (exit code "
+                f"{self._proc.exitcode})")
+
     def step(self, timeout: float = 0.05) -> list[EngineCoreOutput]:
         """Dequeue one batch of outputs (the engine process steps on its
         own cadence; this just drains)."""

@@ -267,5 +267,10 @@ class LLMEngine:
     def is_sleeping(self) -> bool:
         return self.engine_core.is_sleeping()
 
+    def check_health(self) -> None:
+        fn = getattr(self.engine_core, "check_health", None)
+        if fn is not None:
+            fn()
+
     def shutdown(self) -> None:
         self.engine_core.shutdown()

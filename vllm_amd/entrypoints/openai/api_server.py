@@ -111,6 +111,10 @@ def build_app(state: ServerState) -> FastAPI:
 
     @app.get("/health")
     async def health() -> Response:
+        try:
+            engine.check_health()
+        except Exception as e:  # noqa: BLE001
+            return JSONResponse({"error": str(e)}, status_code=503)
         return Response(status_code=200)
 
     @app.get("/version")

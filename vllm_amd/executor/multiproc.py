@@ -106,19 +106,29 @@ def _worker_main(rank: int, config: EngineConfig, conn, master_port: int):
 
 
 class _Future:
-    def __init__(self, conn):
+    def __init__(self, conn, executor=None):
         self._conn = conn
+        self._executor = executor
         self._val = None
         self._done = False
 
     def result(self):
         if not self._done:
-            status, val = self._conn.recv()
+            if self._executor is not None:
+                status, val = self._executor._recv(self._conn)
+            else:
+                status, val = self._conn.recv()
             if status == "err":
                 raise RuntimeError(f"worker error: {val}")
             self._val = val
             self._done = True
         return self._val
+
+
+class EngineDeadError(RuntimeError):
+    """A worker (or the engine core) process died; the engine cannot make
+    progress (role of the reference's EngineDeadError / fault sentinels).
+    """
 
 
 class MultiprocExecutor:
@@ -141,13 +151,39 @@ class MultiprocExecutor:
             self.conns.append(parent)
             self.procs.append(p)
 
+    def check_health(self) -> None:
+        """Raise EngineDeadError if any worker process has died."""
+        dead = [i for i, p in enumerate(self.procs) if not p.is_alive()]
+        if dead:
+            codes = [self.procs[i].exitcode for i in dead]
+            raise EngineDeadError(
+                f"worker process(es) {dead} died (exit codes {codes})")
+
+    def _recv(self, conn):
+        # Poll with a health check so a dead worker surfaces as a clean
+        # EngineDeadError instead of a hang on the pipe.
+        while not conn.poll(1.0):
+            self.check_health()
+        return conn.recv()
+
+    def _send_all(self, msg) -> None:
+        for conn in self.conns:
+            try:
+                conn.send(msg)
+            except (BrokenPipeError, OSError) as e:
+                self.check_health()
+                raise EngineDeadError(f"worker pipe closed: {e}") from e
+
     # ---- control-plane RPC -------------------------------------------
     def collective_rpc(self, method: str, *args: Any, **kwargs: Any) -> list:
-        for conn in self.conns:
-            conn.send(("rpc", method, args, kwargs))
+        self._send_all(("rpc", method, args, kwargs))
         results = []
         for conn in self.conns:
-            status, val = conn.recv()
+            try:
+                status, val = self._recv(conn)
+            except (EOFError, OSError) as e:
+                self.check_health()
+                raise EngineDeadError(f"worker pipe closed: {e}") from e
             if status == "err":
                 raise RuntimeError(f"worker rpc {method} failed: {val}")
             results.append(val)
@@ -173,17 +209,19 @@ class MultiprocExecutor:
 
     # ---- data plane ---------------------------------------------------
     def execute_model(self, so):
-        for conn in self.conns:
-            conn.send(("execute_sync", so))
-        status, val = self.conns[0].recv()
+        self._send_all(("execute_sync", so))
+        try:
+            status, val = self._recv(self.conns[0])
+        except (EOFError, OSError) as e:
+            self.check_health()
+            raise EngineDeadError(f"worker pipe closed: {e}") from e
         if status == "err":
             raise RuntimeError(f"worker step failed: {val}")
         return val
 
     def execute_model_async(self, so) -> _Future:
-        for conn in self.conns:
-            conn.send(("execute", so))
-        return _Future(self.conns[0])
+        self._send_all(("execute", so))
+        return _Future(self.conns[0], self)
 
     def shutdown(self) -> None:
         for conn in self.conns:
