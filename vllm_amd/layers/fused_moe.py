@@ -92,14 +92,30 @@ class FusedMoE(nn.Module):
         return out
 
     def run_experts(self, hidden, topk_weights, topk_ids) -> torch.Tensor:
-        """Local expert computation (routing masked to this rank's expert
-        range in EP mode; the caller's all-reduce completes the combine)."""
+        """Local expert computation. EP mode: activations are replicated
+        across ranks (SPMD TP serving), so each rank COMPACTS to the
+        (token, slot) pairs routed to its local experts — no wasted
+        expert FLOPs on foreign slots — and the caller's all-reduce
+        completes the combine. (A DeepEP-style a2a dispatch only pays
+        once activations are DP/sequence-sharded — a round-2 item
+        together with DP attention for MoE.)"""
         if self.ep_size > 1:
             lo = self.expert_lo
             hi = lo + self.num_local_experts
             mask = (topk_ids >= lo) & (topk_ids < hi)
-            topk_ids = topk_ids.masked_fill(~mask, lo) - lo
-            topk_weights = topk_weights * mask
+            slot_tok, slot_k = mask.nonzero(as_tuple=True)
+            if slot_tok.numel() == 0:
+                return torch.zeros_like(hidden)
+            sel_ids = (topk_ids[slot_tok, slot_k] - lo).unsqueeze(1)
+            sel_w = topk_weights[slot_tok, slot_k].unsqueeze(1)
+            y = ops.fused_moe(
+                hidden[slot_tok], self.w13, self.w2, sel_w, sel_ids,
+                activation=self.activation,
+            )
+            out = torch.zeros(hidden.shape, dtype=torch.float32,
+                              device=hidden.device)
+            out.index_add_(0, slot_tok, y.float())
+            return out.to(hidden.dtype)
         return ops.fused_moe(
             hidden, self.w13, self.w2, topk_weights, topk_ids,
             activation=self.activation,
