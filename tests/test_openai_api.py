@@ -497,3 +497,29 @@ def test_openai_shaped_logprobs(client):
     content = r.json()["choices"][0]["logprobs"]["content"]
     assert len(content) == 3
     assert {"token", "logprob", "bytes", "top_logprobs"} <= set(content[0])
+
+
+def test_api_key_auth_and_request_id():
+    from vllm_amd.engine.arg_utils import EngineArgs
+    from vllm_amd.entrypoints.openai.api_server import make_server
+
+    args = EngineArgs(model="tiny-llama", dtype="fp32", device="cpu",
+                      block_size=16, num_gpu_blocks=64, max_model_len=128,
+                      max_num_batched_tokens=64, max_num_seqs=2)
+    app, state = make_server(args, api_key="sk-test")
+    with TestClient(app) as c:
+        # /v1 without key -> 401; /health stays open
+        r = c.post("/v1/completions", json={
+            "model": "tiny-llama", "prompt": "x", "max_tokens": 2})
+        assert r.status_code == 401
+        assert c.get("/health").status_code == 200
+        r = c.post("/v1/completions", json={
+            "model": "tiny-llama", "prompt": "x", "max_tokens": 2,
+            "temperature": 0.0, "ignore_eos": True},
+            headers={"Authorization": "Bearer sk-test"})
+        assert r.status_code == 200
+        assert r.headers.get("X-Request-Id")
+        # client-supplied id is echoed
+        r = c.get("/health", headers={"X-Request-Id": "rid-42"})
+        assert r.headers["X-Request-Id"] == "rid-42"
+    state.engine.shutdown()

@@ -83,9 +83,12 @@ def apply_chat_template(tokenizer, messages, add_generation_prompt=True,
 
 class ServerState:
     def __init__(self, engine: AsyncLLM, model_name: str,
-                 reasoning_parser: Optional[str] = None):
+                 reasoning_parser: Optional[str] = None,
+                 api_key: Optional[str] = None):
         self.engine = engine
         self.model_name = model_name
+        # Bearer token required on /v1/* when set (reference --api-key).
+        self.api_key = api_key
         # "deepseek_r1" enables <think> splitting into reasoning_content.
         self.reasoning_parser = reasoning_parser
         self.lora_names = list(
@@ -104,6 +107,21 @@ class ServerState:
 def build_app(state: ServerState) -> FastAPI:
     app = FastAPI(title="vllm_amd OpenAI-compatible server")
     engine = state.engine
+
+    @app.middleware("http")
+    async def request_id_and_auth(request: Request, call_next):
+        if (state.api_key is not None
+                and request.url.path.startswith("/v1")):
+            auth = request.headers.get("Authorization", "")
+            if auth != f"Bearer {state.api_key}":
+                return JSONResponse(
+                    {"error": {"message": "invalid or missing API key",
+                               "type": "authentication_error"}},
+                    status_code=401)
+        response = await call_next(request)
+        rid = request.headers.get("X-Request-Id") or random_id("req")
+        response.headers["X-Request-Id"] = rid
+        return response
 
     from vllm_amd.entrypoints.anthropic_api import build_anthropic_router
 
@@ -814,10 +832,12 @@ def build_app(state: ServerState) -> FastAPI:
 
 def make_server(engine_args: EngineArgs,
                 served_model_name: Optional[str] = None,
-                reasoning_parser: Optional[str] = None):
+                reasoning_parser: Optional[str] = None,
+                api_key: Optional[str] = None):
     engine = AsyncLLM(engine_args.create_engine_config())
     state = ServerState(engine, served_model_name or engine_args.model,
-                        reasoning_parser=reasoning_parser)
+                        reasoning_parser=reasoning_parser,
+                        api_key=api_key)
     return build_app(state), state
 
 
@@ -831,6 +851,8 @@ def main() -> None:
     parser.add_argument("--served-model-name", type=str, default=None)
     parser.add_argument("--reasoning-parser", type=str, default=None,
                         choices=["deepseek_r1"])
+    parser.add_argument("--api-key", type=str, default=None,
+                        help="require this bearer token on /v1 routes")
     parser.add_argument("--grpc-port", type=int, default=None,
                         help="also serve the gRPC Inference service "
                              "(entrypoints/grpc/inference.proto)")
@@ -838,7 +860,8 @@ def main() -> None:
     args = parser.parse_args()
     engine_args = EngineArgs.from_cli_args(args)
     app, state = make_server(engine_args, args.served_model_name,
-                             reasoning_parser=args.reasoning_parser)
+                             reasoning_parser=args.reasoning_parser,
+                             api_key=args.api_key)
     if args.grpc_port is not None:
         from vllm_amd.entrypoints.grpc.server import make_grpc_server
 
