@@ -235,3 +235,68 @@ def test_chat_tool_role_message_accepted(client):
         "ignore_eos": True,
     })
     assert r.status_code == 200, r.text
+
+
+def test_parse_mistral_format():
+    from vllm_amd.entrypoints.tool_parser import parse_mistral_tool_calls
+
+    text = ('Let me check. [TOOL_CALLS][{"name": "get_weather", '
+            '"arguments": {"city": "Paris"}}, {"name": "b", '
+            '"arguments": {}}]')
+    content, calls = parse_mistral_tool_calls(text)
+    assert content == "Let me check."
+    assert [c.name for c in calls] == ["get_weather", "b"]
+    assert json.loads(calls[0].arguments) == {"city": "Paris"}
+    # no marker -> plain content
+    c2, k2 = parse_mistral_tool_calls("no tools here")
+    assert c2 == "no tools here" and k2 == []
+
+
+def test_parse_llama3_json_format():
+    from vllm_amd.entrypoints.tool_parser import (
+        parse_llama3_json_tool_calls)
+
+    content, calls = parse_llama3_json_tool_calls(
+        '{"name": "f", "parameters": {"x": 1}}')
+    assert content == ""
+    assert calls[0].name == "f"
+    assert json.loads(calls[0].arguments) == {"x": 1}
+    c2, k2 = parse_llama3_json_tool_calls("plain answer")
+    assert c2 == "plain answer" and k2 == []
+
+
+@pytest.mark.parametrize("chunk", [1, 7, 1000])
+def test_streaming_mistral_parser(chunk):
+    from vllm_amd.entrypoints.tool_parser import (
+        make_streaming_tool_parser)
+
+    text = ('before [TOOL_CALLS][{"name": "a", "arguments": {"k": 2}}]')
+    p = make_streaming_tool_parser("mistral")
+    content, calls = "", []
+    for i in range(0, len(text), chunk):
+        c, k = p.feed(text[i:i + chunk])
+        content += c
+        calls += k
+    c, k = p.flush()
+    content += c
+    calls += k
+    assert content.strip() == "before"
+    assert len(calls) == 1 and calls[0]["function"]["name"] == "a"
+    assert p.saw_tool_call
+
+
+def test_streaming_llama3_json_parser():
+    from vllm_amd.entrypoints.tool_parser import (
+        make_streaming_tool_parser)
+
+    p = make_streaming_tool_parser("llama3_json")
+    c1, k1 = p.feed('{"name": "f", "param')
+    c2, k2 = p.feed('eters": {}}')
+    c3, k3 = p.flush()
+    assert c1 == c2 == "" and not k1 and not k2
+    assert len(k3) == 1 and k3[0]["function"]["name"] == "f"
+    # non-JSON output streams through
+    p2 = make_streaming_tool_parser("llama3_json")
+    c, _ = p2.feed("hello world")
+    c2, k = p2.flush()
+    assert c + c2 == "hello world" and not k

@@ -84,11 +84,14 @@ def apply_chat_template(tokenizer, messages, add_generation_prompt=True,
 class ServerState:
     def __init__(self, engine: AsyncLLM, model_name: str,
                  reasoning_parser: Optional[str] = None,
-                 api_key: Optional[str] = None):
+                 api_key: Optional[str] = None,
+                 tool_call_parser: str = "hermes"):
         self.engine = engine
         self.model_name = model_name
         # Bearer token required on /v1/* when set (reference --api-key).
         self.api_key = api_key
+        # hermes | mistral | llama3_json (tool_parser.py formats).
+        self.tool_call_parser = tool_call_parser
         # "deepseek_r1" enables <think> splitting into reasoning_content.
         self.reasoning_parser = reasoning_parser
         self.lora_names = list(
@@ -668,7 +671,8 @@ def build_app(state: ServerState) -> FastAPI:
                 n_prompt = 0
                 rparse = (tp.StreamingReasoningParser()
                           if state.reasoning_parser else None)
-                tparse = (tp.StreamingToolParser()
+                tparse = (tp.make_streaming_tool_parser(
+                              state.tool_call_parser)
                           if tools_on and not named else None)
                 named_id = tp._call_id() if named else None
                 named_first = True
@@ -809,7 +813,8 @@ def build_app(state: ServerState) -> FastAPI:
                 text = None
                 finish = "tool_calls"
             elif tools_on:
-                text, calls = tp.parse_hermes_tool_calls(text)
+                text, calls = tp.parse_tool_calls(
+                    state.tool_call_parser, text)
                 if calls:
                     tool_calls = [c.as_openai(i)
                                   for i, c in enumerate(calls)]
@@ -833,11 +838,13 @@ def build_app(state: ServerState) -> FastAPI:
 def make_server(engine_args: EngineArgs,
                 served_model_name: Optional[str] = None,
                 reasoning_parser: Optional[str] = None,
-                api_key: Optional[str] = None):
+                api_key: Optional[str] = None,
+                tool_call_parser: str = "hermes"):
     engine = AsyncLLM(engine_args.create_engine_config())
     state = ServerState(engine, served_model_name or engine_args.model,
                         reasoning_parser=reasoning_parser,
-                        api_key=api_key)
+                        api_key=api_key,
+                        tool_call_parser=tool_call_parser)
     return build_app(state), state
 
 
@@ -851,6 +858,8 @@ def main() -> None:
     parser.add_argument("--served-model-name", type=str, default=None)
     parser.add_argument("--reasoning-parser", type=str, default=None,
                         choices=["deepseek_r1"])
+    parser.add_argument("--tool-call-parser", type=str, default="hermes",
+                        choices=["hermes", "mistral", "llama3_json"])
     parser.add_argument("--api-key", type=str, default=None,
                         help="require this bearer token on /v1 routes")
     parser.add_argument("--grpc-port", type=int, default=None,
@@ -861,7 +870,8 @@ def main() -> None:
     engine_args = EngineArgs.from_cli_args(args)
     app, state = make_server(engine_args, args.served_model_name,
                              reasoning_parser=args.reasoning_parser,
-                             api_key=args.api_key)
+                             api_key=args.api_key,
+                             tool_call_parser=args.tool_call_parser)
     if args.grpc_port is not None:
         from vllm_amd.entrypoints.grpc.server import make_grpc_server
 

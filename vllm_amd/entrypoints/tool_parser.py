@@ -262,3 +262,117 @@ def named_tool_schema(tools: list[dict[str, Any]],
         if fn.get("name") == name:
             return fn.get("parameters") or {"type": "object"}
     return None
+
+
+# --------------------------------------------------------------------------
+# Mistral format: optional content, then "[TOOL_CALLS]" followed by a JSON
+# array of {"name": ..., "arguments": {...}} (reference
+# tool_parsers/mistral_tool_parser.py).
+MISTRAL_MARKER = "[TOOL_CALLS]"
+
+
+def parse_mistral_tool_calls(text: str) -> tuple[str, list[ParsedToolCall]]:
+    idx = text.find(MISTRAL_MARKER)
+    if idx < 0:
+        return text.strip(), []
+    content = text[:idx].strip()
+    body = text[idx + len(MISTRAL_MARKER):].strip()
+    try:
+        arr = json.loads(body)
+    except (json.JSONDecodeError, ValueError):
+        return text.strip(), []
+    calls = []
+    for obj in arr if isinstance(arr, list) else [arr]:
+        if isinstance(obj, dict) and "name" in obj:
+            args = obj.get("arguments", obj.get("parameters", {}))
+            if not isinstance(args, str):
+                args = json.dumps(args)
+            calls.append(ParsedToolCall(id=_call_id(),
+                                        name=str(obj["name"]),
+                                        arguments=args))
+    return content, calls
+
+
+# Llama-3 JSON format: the WHOLE assistant message is one JSON object
+# {"name": ..., "parameters": {...}} (reference llama_tool_parser.py).
+def parse_llama3_json_tool_calls(
+        text: str) -> tuple[str, list[ParsedToolCall]]:
+    stripped = text.strip()
+    if not stripped.startswith("{"):
+        return stripped, []
+    call = _parse_call_body(stripped)
+    if call is None:
+        return stripped, []
+    return "", [call]
+
+
+@dataclass
+class StreamingBufferedToolParser:
+    """Streaming wrapper for formats that cannot be parsed incrementally
+    (mistral marker / whole-message JSON): content before the trigger
+    streams through; once triggered, the rest buffers and parses at
+    flush. `parse` is the complete-output parser; `trigger` returns the
+    index where buffering must start, or -1."""
+
+    parse: Any = None
+    trigger: Any = None
+    _buf: str = ""
+    _buffering: bool = False
+    _saw_call: bool = False
+
+    @property
+    def saw_tool_call(self) -> bool:
+        return self._saw_call
+
+    def feed(self, delta: str) -> tuple[str, list[dict[str, Any]]]:
+        self._buf += delta
+        if self._buffering:
+            return "", []
+        idx = self.trigger(self._buf)
+        if idx >= 0:
+            content = self._buf[:idx]
+            self._buf = self._buf[idx:]
+            self._buffering = True
+            return content, []
+        # Withhold a potential trigger prefix at the tail.
+        hold = self._hold_len(self._buf)
+        emit = len(self._buf) - hold
+        out = self._buf[:emit]
+        self._buf = self._buf[emit:]
+        return out, []
+
+    def _hold_len(self, text: str) -> int:
+        return _longest_tag_prefix(text, MISTRAL_MARKER)
+
+    def flush(self) -> tuple[str, list[dict[str, Any]]]:
+        content, calls = self.parse(self._buf)
+        self._buf = ""
+        out = [c.as_openai(i) for i, c in enumerate(calls)]
+        if out:
+            self._saw_call = True
+        return content, out
+
+
+def make_streaming_tool_parser(fmt: str):
+    if fmt == "mistral":
+        return StreamingBufferedToolParser(
+            parse=parse_mistral_tool_calls,
+            trigger=lambda s: s.find(MISTRAL_MARKER))
+    if fmt == "llama3_json":
+        def trig(s):
+            st = s.lstrip()
+            return (len(s) - len(st)) if st.startswith("{") else -1
+
+        p = StreamingBufferedToolParser(
+            parse=parse_llama3_json_tool_calls, trigger=trig)
+        p._hold_len = lambda text: 0
+        return p
+    return StreamingToolParser()
+
+
+def parse_tool_calls(fmt: str, text: str):
+    if fmt == "mistral":
+        return parse_mistral_tool_calls(text)
+    if fmt == "llama3_json":
+        return parse_llama3_json_tool_calls(text)
+    return parse_hermes_tool_calls(text)
