@@ -446,3 +446,28 @@ def test_prompt_logprobs():
         assert set(d1) == set(d2)
         for t in d1:
             assert abs(d1[t] - d2[t]) < 1e-5
+
+
+def test_request_trace_file(tmp_path):
+    """--trace-file: one JSON line per finished request with timing and
+    token counts (ObservabilityConfig)."""
+    import json as _json
+
+    from vllm_amd.entrypoints.llm import LLM
+    from vllm_amd.sampling_params import SamplingParams
+
+    trace = tmp_path / "trace.jsonl"
+    llm = LLM(model="tiny-llama", dtype="fp32", device="cpu",
+              block_size=16, num_gpu_blocks=64, max_model_len=128,
+              max_num_batched_tokens=64, max_num_seqs=4,
+              trace_file=str(trace))
+    llm.generate([[3, 4, 5, 6, 7], [8, 9, 10]],
+                 SamplingParams(temperature=0.0, max_tokens=4,
+                                ignore_eos=True))
+    llm.shutdown()
+    lines = [_json.loads(x) for x in trace.read_text().splitlines()]
+    assert len(lines) == 2
+    for ln in lines:
+        assert ln["output_tokens"] == 4
+        assert ln["finish_reason"] == "length"
+        assert ln["e2e_s"] > 0 and ln["ttft_s"] > 0

@@ -673,12 +673,23 @@ class DeviceConfig:
 
 
 @dataclass
+class ObservabilityConfig:
+    """Request-level tracing (role of the reference's
+    ObservabilityConfig): one JSON line per finished request — arrival /
+    first-token / finish timestamps, token counts, finish reason."""
+
+    trace_file: Optional[str] = None
+
+
+@dataclass
 class EngineConfig:
     model_config: ModelConfig = field(default_factory=ModelConfig)
     cache_config: CacheConfig = field(default_factory=CacheConfig)
     scheduler_config: SchedulerConfig = field(default_factory=SchedulerConfig)
     parallel_config: ParallelConfig = field(default_factory=ParallelConfig)
     device_config: DeviceConfig = field(default_factory=DeviceConfig)
+    observability_config: "ObservabilityConfig" = field(
+        default_factory=lambda: ObservabilityConfig())
 
     def __post_init__(self) -> None:
         spec = self.model_config.spec

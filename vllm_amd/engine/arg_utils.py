@@ -11,6 +11,7 @@ from vllm_amd.config import (
     DeviceConfig,
     EngineConfig,
     ModelConfig,
+    ObservabilityConfig,
     ParallelConfig,
     SchedulerConfig,
 )
@@ -53,6 +54,7 @@ class EngineArgs:
     enable_expert_parallel: bool = False
     multiprocess_engine: bool = False
     device: str = "auto"
+    trace_file: Optional[str] = None
 
     @staticmethod
     def add_cli_args(parser: argparse.ArgumentParser) -> argparse.ArgumentParser:
@@ -104,6 +106,8 @@ class EngineArgs:
         parser.add_argument("--multiprocess-engine", action="store_true",
                             help="run the engine core in its own process")
         parser.add_argument("--device", type=str, default="auto")
+        parser.add_argument("--trace-file", type=str, default=None,
+                            help="JSONL request-trace output path")
         return parser
 
     @classmethod
@@ -167,4 +171,6 @@ class EngineArgs:
             ),
             parallel_config=pc,
             device_config=DeviceConfig(device=self.device),
+            observability_config=ObservabilityConfig(
+                trace_file=self.trace_file),
         )

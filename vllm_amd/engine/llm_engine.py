@@ -169,6 +169,10 @@ class LLMEngine:
         self.tokenizer = TokenizerWrapper(config.model_config.tokenizer)
         self.output_processor = OutputProcessor(self.tokenizer)
         self._request_counter = 0
+        self._trace_fh = None
+        trace_file = config.observability_config.trace_file
+        if trace_file:
+            self._trace_fh = open(trace_file, "a", buffering=1)
 
     @property
     def is_driver(self) -> bool:
@@ -249,11 +253,35 @@ class LLMEngine:
             self.engine_core.abort_requests(to_abort)
         for out in outputs:
             if out.finished:
+                if self._trace_fh is not None:
+                    self._write_trace(out)
                 self.output_processor.release(out.request_id)
                 if self.engine_core.scheduler is not None:
                     self.engine_core.scheduler.release_request(
                         out.request_id)
         return outputs
+
+    def _write_trace(self, out: RequestOutput) -> None:
+        import json as _json
+
+        m = out.metrics or {}
+        comp = out.outputs[0]
+        arrival = m.get("arrival_time")
+        first = m.get("first_token_time")
+        finish = m.get("finish_time")
+        self._trace_fh.write(_json.dumps({
+            "request_id": out.request_id,
+            "arrival_time": arrival,
+            "first_token_time": first,
+            "finish_time": finish,
+            "ttft_s": (first - arrival
+                       if first is not None and arrival else None),
+            "e2e_s": (finish - arrival
+                      if finish is not None and arrival else None),
+            "prompt_tokens": len(out.prompt_token_ids),
+            "output_tokens": len(comp.token_ids),
+            "finish_reason": comp.finish_reason,
+        }) + "\n")
 
     def has_unfinished_requests(self) -> bool:
         return self.engine_core.has_unfinished_requests()
@@ -273,4 +301,7 @@ class LLMEngine:
             fn()
 
     def shutdown(self) -> None:
+        if self._trace_fh is not None:
+            self._trace_fh.close()
+            self._trace_fh = None
         self.engine_core.shutdown()
