@@ -220,13 +220,18 @@ class LLMEngine:
             request.grammar = compile_choice_grammar(
                 params.guided_choice, self.tokenizer, eos
             )
-        elif params.guided_regex or params.guided_json is not None \
-                or params.guided_json_object:
+        elif (params.guided_regex or params.guided_grammar
+                or params.guided_json is not None
+                or params.guided_json_object):
             from vllm_amd.guided_json import any_json_regex, schema_to_regex
             from vllm_amd.guided_regex import RegexGrammar
 
             if params.guided_regex:
                 pattern = params.guided_regex
+            elif params.guided_grammar:
+                from vllm_amd.guided_grammar import grammar_to_regex
+
+                pattern = grammar_to_regex(params.guided_grammar)
             elif params.guided_json is not None \
                     and params.guided_json is not True:
                 pattern = schema_to_regex(params.guided_json)

@@ -46,6 +46,7 @@ class CompletionRequest(BaseModel):
     guided_choice: Optional[list[str]] = None
     guided_regex: Optional[str] = None
     guided_json: Optional[Union[dict, str]] = None
+    guided_grammar: Optional[str] = None
     # Extension (same name as the reference): per-prompt-token logprobs.
     prompt_logprobs: Optional[int] = None
 
@@ -76,6 +77,7 @@ class CompletionRequest(BaseModel):
             guided_choice=self.guided_choice,
             guided_regex=self.guided_regex,
             guided_json=self.guided_json,
+            guided_grammar=self.guided_grammar,
             output_kind=(RequestOutputKind.DELTA if stream
                          else RequestOutputKind.FINAL_ONLY),
         )
@@ -128,6 +130,7 @@ class ChatCompletionRequest(BaseModel):
     guided_choice: Optional[list[str]] = None
     guided_regex: Optional[str] = None
     guided_json: Optional[Union[dict, str]] = None
+    guided_grammar: Optional[str] = None
     response_format: Optional[dict[str, Any]] = None
     add_generation_prompt: bool = True
     # Tool calling (OpenAI function-calling surface).
@@ -191,6 +194,7 @@ class ChatCompletionRequest(BaseModel):
             guided_choice=self.guided_choice,
             guided_regex=self.guided_regex,
             guided_json=guided_json,
+            guided_grammar=self.guided_grammar,
             guided_json_object=json_object,
             output_kind=(RequestOutputKind.DELTA if stream
                          else RequestOutputKind.FINAL_ONLY),

@@ -66,6 +66,8 @@ class SamplingParams:
     # DFA with per-state token masks; see guided_regex.py / guided_json.py).
     guided_regex: Optional[str] = None
     guided_json: Optional[object] = None  # dict schema, JSON string, or True
+    # GBNF-style EBNF grammar (non-recursive subset; guided_grammar.py).
+    guided_grammar: Optional[str] = None
     # response_format={"type": "json_object"}: any JSON object.
     guided_json_object: bool = False
     output_kind: RequestOutputKind = RequestOutputKind.CUMULATIVE
