@@ -217,3 +217,27 @@ def test_tp2_pp2_grid():
         assert status == "ok", payload
     toks = next(p for s, p in outs if p is not None)
     assert all(len(t) == 6 for t in toks)
+
+
+def test_pp_rejects_pooling_requests():
+    """pooling/prompt_logprobs need the last stage's hidden/logits on the
+    driver — LLMEngine.add_request rejects them under pp>1."""
+    import pytest
+
+    from vllm_amd.config import EngineConfig, ModelConfig, ParallelConfig
+    from vllm_amd.engine.llm_engine import LLMEngine
+    from vllm_amd.sampling_params import SamplingParams
+    from vllm_amd.tokenizer import TokenizerWrapper
+
+    eng = LLMEngine.__new__(LLMEngine)  # skip engine-core construction
+    eng.config = EngineConfig(
+        model_config=ModelConfig(model="tiny-llama", dtype="fp32"),
+        parallel_config=ParallelConfig(pipeline_parallel_size=2,
+                                       tensor_parallel_size=1,
+                                       world_size=2))
+    eng.tokenizer = TokenizerWrapper(None)
+    eng._request_counter = 0
+    for bad in (SamplingParams(pooling="last", max_tokens=1),
+                SamplingParams(prompt_logprobs=2, max_tokens=1)):
+        with pytest.raises(ValueError, match="pp=1"):
+            eng.add_request(None, [3, 4, 5], bad)

@@ -214,6 +214,12 @@ class LLMEngine:
             prompt=prompt_text,
             lora_id=self.config.model_config.lora_id_of(lora),
         )
+        pp = self.config.parallel_config.pipeline_parallel_size
+        if pp > 1 and (params.pooling or params.prompt_logprobs):
+            # Hidden states / logits live on the LAST pp stage while the
+            # driver is stage 0; routing them back is a round-2 item.
+            raise ValueError(
+                "pooling and prompt_logprobs require pp=1")
         if params.guided_choice:
             from vllm_amd.structured_output import compile_choice_grammar
 
