@@ -715,67 +715,54 @@ class ModelRunner:
         dev = self.device
         max_seq_len = int(seq_lens.max())
 
-        # hipGraph fast path: pure-decode batch within a captured bucket
-        # (pooling rows need `hidden`, which graph replay doesn't expose
-        # on this path — they stay eager).
-        nb = None
-        if (self.graph_runner is not None and num_decodes == len(items)
-                and total == len(items) and not pooling_rows):
-            nb = self.graph_runner.bucket_for(total)
-        if nb is not None:
-            parts = self.graph_runner.parts_bucket(max_seq_len)
-            logits_all = self.graph_runner.run(
-                total, nb, parts, input_ids, positions, slot_mapping,
-                seq_lens, block_table,
-            )
+        # (Pure-decode batches already returned via _execute_decode —
+        # everything here runs eagerly: mixed/prefill/spec/pooling/PP.)
+        meta = AttentionMetadata(
+            query_start_loc=torch.from_numpy(query_start_loc).to(dev),
+            seq_lens=torch.from_numpy(seq_lens).to(dev),
+            block_table=torch.from_numpy(block_table).to(dev),
+            slot_mapping=torch.from_numpy(slot_mapping).to(dev),
+            num_reqs=len(items),
+            num_actual_tokens=total,
+            max_query_len=int(max(n for _, n in items)),
+            max_seq_len=max_seq_len,
+            num_decodes=num_decodes,
+        )
+        input_ids_t = torch.from_numpy(input_ids).to(dev)
+        positions_t = torch.from_numpy(positions).to(dev)
+
+        lora_np = None
+        if self.lora_manager is not None:
+            lora_np = np.concatenate([
+                np.full(n_, self.np_lora[self._row_of[rid]])
+                for rid, n_ in items
+            ])
+        ctx = ForwardContext(
+            attn_metadata=meta, kv_caches=self.kv_caches,
+            lora_ids=self._lora_ids_tensor(lora_np),
+            lora_manager=self.lora_manager,
+        )
+        if self.pp_size > 1:
+            # Stage boundary: recv [T, hidden] from the previous
+            # stage, send to the next (non-overlapped v1 pipeline —
+            # capacity sharding; microbatched overlap is future work).
+            from vllm_amd.parallel.state import (
+                get_pp_rank, is_first_pp_rank, is_last_pp_rank,
+                pp_recv, pp_send)
+
+            hidden_in = None
+            if not is_first_pp_rank():
+                hidden_in = torch.empty(
+                    total, self.spec.hidden_size, dtype=self.dtype,
+                    device=dev)
+                pp_recv(hidden_in, get_pp_rank() - 1)
+            with set_forward_context(ctx):
+                hidden = self.model(input_ids_t, positions_t, hidden_in)
+            if not is_last_pp_rank():
+                pp_send(hidden, get_pp_rank() + 1)
         else:
-            meta = AttentionMetadata(
-                query_start_loc=torch.from_numpy(query_start_loc).to(dev),
-                seq_lens=torch.from_numpy(seq_lens).to(dev),
-                block_table=torch.from_numpy(block_table).to(dev),
-                slot_mapping=torch.from_numpy(slot_mapping).to(dev),
-                num_reqs=len(items),
-                num_actual_tokens=total,
-                max_query_len=int(max(n for _, n in items)),
-                max_seq_len=max_seq_len,
-                num_decodes=num_decodes,
-            )
-            input_ids_t = torch.from_numpy(input_ids).to(dev)
-            positions_t = torch.from_numpy(positions).to(dev)
-
-            lora_np = None
-            if self.lora_manager is not None:
-                lora_np = np.concatenate([
-                    np.full(n_, self.np_lora[self._row_of[rid]])
-                    for rid, n_ in items
-                ])
-            ctx = ForwardContext(
-                attn_metadata=meta, kv_caches=self.kv_caches,
-                lora_ids=self._lora_ids_tensor(lora_np),
-                lora_manager=self.lora_manager,
-            )
-            if self.pp_size > 1:
-                # Stage boundary: recv [T, hidden] from the previous
-                # stage, send to the next (non-overlapped v1 pipeline —
-                # capacity sharding; microbatched overlap is future work).
-                from vllm_amd.parallel.state import (
-                    get_pp_rank, is_first_pp_rank, is_last_pp_rank,
-                    pp_recv, pp_send)
-
-                hidden_in = None
-                if not is_first_pp_rank():
-                    hidden_in = torch.empty(
-                        total, self.spec.hidden_size, dtype=self.dtype,
-                        device=dev)
-                    pp_recv(hidden_in, get_pp_rank() - 1)
-                with set_forward_context(ctx):
-                    hidden = self.model(input_ids_t, positions_t, hidden_in)
-                if not is_last_pp_rank():
-                    pp_send(hidden, get_pp_rank() + 1)
-            else:
-                with set_forward_context(ctx):
-                    hidden = self.model(input_ids_t, positions_t)
-            logits_all = None
+            with set_forward_context(ctx):
+                hidden = self.model(input_ids_t, positions_t)
 
         # Advance computed counts (python state + persistent rows).
         for rid, n in items:
@@ -866,18 +853,14 @@ class ModelRunner:
                     logprobs=logprobs_pp or None,
                 )
 
-        if logits_all is not None:
-            # Decode: every row samples, rows already in order.
-            logits = logits_all
-        else:
-            # Gather sampling positions: the last `npos` tokens of each
-            # sampling request (npos > 1 verifies draft tokens in place).
-            idx = []
-            for r, npos in zip(sampling_rows, sampling_npos):
-                end = int(query_start_loc[r + 1])
-                idx.extend(range(end - npos, end))
-            last_idx = torch.tensor(idx, device=dev)
-            logits = self.model.compute_logits(hidden[last_idx])
+        # Gather sampling positions: the last `npos` tokens of each
+        # sampling request (npos > 1 verifies draft tokens in place).
+        idx = []
+        for r, npos in zip(sampling_rows, sampling_npos):
+            end = int(query_start_loc[r + 1])
+            idx.extend(range(end - npos, end))
+        last_idx = torch.tensor(idx, device=dev)
+        logits = self.model.compute_logits(hidden[last_idx])
 
         def rep(make):
             vals = []
