@@ -471,3 +471,25 @@ def test_request_trace_file(tmp_path):
         assert ln["output_tokens"] == 4
         assert ln["finish_reason"] == "length"
         assert ln["e2e_s"] > 0 and ln["ttft_s"] > 0
+
+
+def test_offline_chat_api():
+    from vllm_amd.entrypoints.llm import LLM
+    from vllm_amd.sampling_params import SamplingParams
+
+    llm = LLM(model="tiny-llama", dtype="fp32", device="cpu",
+              block_size=16, num_gpu_blocks=64, max_model_len=256,
+              max_num_batched_tokens=128, max_num_seqs=4)
+    outs = llm.chat(
+        [{"role": "system", "content": "be brief"},
+         {"role": "user", "content": "hello"}],
+        SamplingParams(temperature=0.0, max_tokens=4, ignore_eos=True))
+    assert len(outs) == 1
+    assert len(outs[0].outputs[0].token_ids) == 4
+    # batched conversations
+    outs = llm.chat(
+        [[{"role": "user", "content": "a"}],
+         [{"role": "user", "content": "b"}]],
+        SamplingParams(temperature=0.0, max_tokens=3, ignore_eos=True))
+    assert len(outs) == 2
+    llm.shutdown()

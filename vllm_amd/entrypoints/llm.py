@@ -25,6 +25,32 @@ class LLM:
             prompts, SamplingParams(pooling=pooling, max_tokens=1))
         return [o.pooled for o in outs]
 
+    def chat(
+        self,
+        messages,
+        sampling_params: Optional[SamplingParams] = None,
+        add_generation_prompt: bool = True,
+    ) -> list[RequestOutput]:
+        """Offline chat API (role of the reference's LLM.chat): messages
+        go through the same chat templating as the server, then
+        generate(). Accepts one conversation (list of dicts) or a list
+        of conversations."""
+        from vllm_amd.entrypoints.openai.api_server import (
+            apply_chat_template)
+        from vllm_amd.entrypoints.openai.protocol import ChatMessage
+
+        if messages and isinstance(messages[0], dict):
+            messages = [messages]
+        prompts = [
+            apply_chat_template(
+                self.engine.tokenizer,
+                [ChatMessage(**m) for m in conv],
+                add_generation_prompt,
+            )
+            for conv in messages
+        ]
+        return self.generate(prompts, sampling_params)
+
     def beam_search(
         self,
         prompts,
