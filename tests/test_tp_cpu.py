@@ -111,3 +111,86 @@ def test_tp2_spmd_cpu(model, port, ep):
         # Same prompt in the same world must reproduce exactly (prefix
         # cache hit on the second run).
         assert run0[k] == run1[k], k
+
+
+def _fp8_tp_worker(rank: int, world: int, port: int, q):
+    os.environ["RANK"] = str(rank)
+    os.environ["LOCAL_RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    try:
+        import torch
+
+        torch.set_num_threads(1)
+        from vllm_amd.config import (
+            CacheConfig, DeviceConfig, EngineConfig, ModelConfig,
+            ParallelConfig, SchedulerConfig,
+        )
+        from vllm_amd.engine.core import EngineCore
+        from vllm_amd.request import Request
+        from vllm_amd.sampling_params import SamplingParams
+
+        config = EngineConfig(
+            model_config=ModelConfig(model="tiny-llama", dtype="fp32",
+                                     max_model_len=256,
+                                     quantization="fp8"),
+            cache_config=CacheConfig(block_size=16, num_gpu_blocks=64),
+            scheduler_config=SchedulerConfig(max_num_batched_tokens=256,
+                                             max_num_seqs=4),
+            parallel_config=ParallelConfig(tensor_parallel_size=world,
+                                           distributed_backend="gloo"),
+            device_config=DeviceConfig(device="cpu"),
+        )
+        engine = EngineCore(config)
+        if rank == 0:
+            params = SamplingParams(temperature=0.0, max_tokens=6,
+                                    ignore_eos=True)
+            toks = {}
+            for run in range(2):
+                engine.add_request(Request(
+                    request_id=f"run{run}",
+                    prompt_token_ids=[j * 3 + 5 for j in range(14)],
+                    sampling_params=params))
+                while engine.has_unfinished_requests():
+                    for out in engine.step():
+                        toks.setdefault(out.req_id, []).extend(
+                            out.new_token_ids)
+            engine.shutdown()
+            q.put(("ok", toks))
+        else:
+            engine.run_spmd_worker_loop()
+            q.put(("ok", None))
+        import torch.distributed as dist
+
+        if dist.is_initialized():
+            dist.destroy_process_group()
+    except Exception as e:  # noqa: BLE001
+        import traceback
+
+        q.put(("err", f"{e}\n{traceback.format_exc()}"))
+
+
+def test_fp8_quantization_under_tp2():
+    """fp8 W8A8 with TP-sharded weights: per-rank per-channel scales,
+    deterministic decode across runs (cross-feature hardening)."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_fp8_tp_worker, args=(r, 2, 29661, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    outs = []
+    try:
+        for _ in range(2):
+            outs.append(q.get(timeout=180))
+    finally:
+        for p in procs:
+            p.join(timeout=30)
+            if p.is_alive():
+                p.terminate()
+    for status, payload in outs:
+        assert status == "ok", payload
+    toks = next(p for s, p in outs if p is not None)
+    assert len(toks["run0"]) == 6
+    assert toks["run0"] == toks["run1"]

@@ -39,6 +39,7 @@ class EngineArgs:
     max_num_batched_tokens: int = 8192
     max_num_seqs: int = 256
     enable_chunked_prefill: bool = True
+    long_prefill_token_threshold: int = 0
     scheduling_policy: str = "fcfs"
 
     lora_modules: Optional[dict] = None  # name -> PEFT dir
@@ -84,6 +85,10 @@ class EngineArgs:
         parser.add_argument("--no-enable-chunked-prefill",
                             dest="enable_chunked_prefill",
                             action="store_false")
+        parser.add_argument("--long-prefill-token-threshold", type=int,
+                            default=0,
+                            help="cap prefill chunk size (0 = only the "
+                                 "token-budget cap)")
         parser.add_argument("--scheduling-policy", type=str, default="fcfs",
                             choices=["fcfs", "priority"])
         parser.add_argument(
@@ -161,6 +166,8 @@ class EngineArgs:
                 max_num_batched_tokens=self.max_num_batched_tokens,
                 max_num_seqs=self.max_num_seqs,
                 enable_chunked_prefill=self.enable_chunked_prefill,
+                long_prefill_token_threshold=(
+                    self.long_prefill_token_threshold),
                 policy=self.scheduling_policy,
                 async_scheduling=self.async_scheduling,
                 num_speculative_tokens=self.num_speculative_tokens,

@@ -201,6 +201,12 @@ def build_app(state: ServerState) -> FastAPI:
             f"vllm_amd:prefix_cache_hits_total {s.get('prefix_cache_hits', 0)}",
             "# TYPE vllm_amd:num_preemptions_total counter",
             f"vllm_amd:num_preemptions_total {s.get('num_preemptions', 0)}",
+            "# TYPE vllm_amd:spec_decode_num_draft_tokens_total counter",
+            "vllm_amd:spec_decode_num_draft_tokens_total "
+            f"{s.get('spec_tokens_drafted', 0)}",
+            "# TYPE vllm_amd:spec_decode_num_accepted_tokens_total counter",
+            "vllm_amd:spec_decode_num_accepted_tokens_total "
+            f"{s.get('spec_tokens_accepted', 0)}",
         ]
         lines += state.metrics.render()
         return Response("\n".join(lines) + "\n",
