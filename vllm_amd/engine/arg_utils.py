@@ -64,7 +64,7 @@ class EngineArgs:
         parser.add_argument("--dtype", type=str, default="bf16")
         parser.add_argument("--max-model-len", type=int, default=8192)
         parser.add_argument("--load-format", type=str, default="dummy",
-                            choices=["dummy", "safetensors"])
+                            choices=["dummy", "safetensors", "sharded"])
         parser.add_argument("--model-path", type=str, default=None)
         parser.add_argument("--seed", type=int, default=0)
         parser.add_argument("--enforce-eager", action="store_true")

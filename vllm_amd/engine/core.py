@@ -136,6 +136,19 @@ class EngineCore:
     def is_sleeping(self) -> bool:
         return getattr(self, "_sleeping", False)
 
+    def update_weights(self, model_path: str) -> None:
+        """In-place weight refresh (RL-style update). Engine must be
+        idle — the next step serves the new weights."""
+        if self.has_unfinished_requests():
+            raise RuntimeError(
+                "cannot update weights with unfinished requests")
+        if self._pending is not None:
+            self._drain()
+        self.worker.update_weights(model_path)
+
+    def save_sharded_state(self, out_dir: str):
+        return self.worker.save_sharded_state(out_dir)
+
     def check_health(self) -> None:
         """Raise EngineDeadError if a worker process died (in-proc
         workers cannot die independently — no-op)."""

@@ -201,6 +201,12 @@ class MultiprocExecutor:
     def allocate_host_kv_pool(self, num_host_blocks: int) -> None:
         self.collective_rpc("allocate_host_kv_pool", num_host_blocks)
 
+    def update_weights(self, model_path: str) -> None:
+        self.collective_rpc("update_weights", model_path)
+
+    def save_sharded_state(self, out_dir: str) -> list:
+        return self.collective_rpc("save_sharded_state", out_dir)
+
     def sleep(self, level: int = 1) -> None:
         self.collective_rpc("sleep", level)
 

@@ -65,6 +65,10 @@ def load_model(config: ModelConfig, device: torch.device) -> torch.nn.Module:
         from vllm_amd.models.weight_loader import load_safetensors_weights
 
         load_safetensors_weights(model, config)
+    elif config.load_format == "sharded":
+        from vllm_amd.models.weight_loader import load_sharded_state
+
+        load_sharded_state(model, config.model_path or config.model)
     else:
         raise ValueError(f"Unknown load_format {config.load_format}")
     model = model.to(device).eval()

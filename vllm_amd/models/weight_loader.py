@@ -30,6 +30,28 @@ def _iter_safetensors(path: str):
                 yield name, sf.get_tensor(name)
 
 
+def sharded_state_path(out_dir: str) -> str:
+    import os
+
+    from vllm_amd.parallel.state import get_pp_rank, get_tp_rank
+
+    return os.path.join(
+        out_dir, f"rank{get_pp_rank()}_{get_tp_rank()}.safetensors")
+
+
+def load_sharded_state(model: torch.nn.Module, out_dir: str) -> None:
+    """Load THIS rank's pre-sharded parameters (fast TP/PP restart; role
+    of the reference's sharded_state_loader — no re-sharding pass)."""
+    from safetensors import safe_open
+
+    params = dict(model.named_parameters())
+    with safe_open(sharded_state_path(out_dir), framework="pt",
+                   device="cpu") as f:
+        for name in f.keys():
+            params[name].data.copy_(
+                f.get_tensor(name).to(params[name].dtype))
+
+
 def load_safetensors_weights(model: torch.nn.Module, config) -> None:
     """Load HF-layout weights into the TP-sharded model. The model is on
     CPU at this point (moved to device afterwards by load_model)."""

@@ -385,6 +385,39 @@ class ModelRunner:
         if self.device.type == "cuda":
             torch.cuda.empty_cache()
 
+    def update_weights(self, model_path: str) -> None:
+        """In-place weight refresh from a safetensors dir (role of the
+        reference's RL weight-update path / set_weight_version): the
+        engine must be idle; KV cache and hipGraphs stay valid because
+        parameter STORAGE is reused (copy_ into existing tensors)."""
+        from vllm_amd.models.weight_loader import load_safetensors_weights
+
+        mc = self.config.model_config
+        if mc.quantization == "fp8":
+            raise ValueError(
+                "update_weights into fp8-quantized layers needs "
+                "requantization — reload instead (sleep(2)/wake_up)")
+        cfg = dataclasses.replace(mc, model_path=model_path,
+                                  load_format="safetensors")
+        load_safetensors_weights(self.model, cfg)
+
+    def save_sharded_state(self, out_dir: str) -> str:
+        """Write THIS rank's (TP-sharded, PP-sliced) parameters to
+        `out_dir/rank{pp}_{tp}.safetensors` for fast restarts without
+        re-sharding (role of the reference's sharded_state_loader)."""
+        import os as _os
+
+        from safetensors.torch import save_file
+
+        from vllm_amd.models.weight_loader import sharded_state_path
+
+        _os.makedirs(out_dir, exist_ok=True)
+        path = sharded_state_path(out_dir)
+        tensors = {n: p.data.detach().cpu().contiguous()
+                   for n, p in self.model.named_parameters()}
+        save_file(tensors, path)
+        return path
+
     def wake_up(self) -> None:
         if self.model is None:
             self.load_model()

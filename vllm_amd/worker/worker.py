@@ -89,6 +89,12 @@ class Worker:
     def allocate_host_kv_pool(self, num_host_blocks: int) -> None:
         self.runner.allocate_host_kv_pool(num_host_blocks)
 
+    def update_weights(self, model_path: str) -> None:
+        self.runner.update_weights(model_path)
+
+    def save_sharded_state(self, out_dir: str) -> str:
+        return self.runner.save_sharded_state(out_dir)
+
     def sleep(self, level: int = 1) -> None:
         self.runner.sleep(level)
 
