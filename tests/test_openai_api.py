@@ -523,3 +523,30 @@ def test_api_key_auth_and_request_id():
         r = c.get("/health", headers={"X-Request-Id": "rid-42"})
         assert r.headers["X-Request-Id"] == "rid-42"
     state.engine.shutdown()
+
+
+def test_profile_endpoints(tmp_path):
+    from vllm_amd.config import ObservabilityConfig
+    from vllm_amd.engine.arg_utils import EngineArgs
+    from vllm_amd.entrypoints.openai.api_server import make_server
+
+    args = EngineArgs(model="tiny-llama", dtype="fp32", device="cpu",
+                      block_size=16, num_gpu_blocks=64, max_model_len=128,
+                      max_num_batched_tokens=64, max_num_seqs=2)
+    app, state = make_server(args)
+    state.engine.config.observability_config = ObservabilityConfig(
+        profile_dir=str(tmp_path))
+    with TestClient(app) as c:
+        assert c.post("/stop_profile").status_code == 400  # not running
+        assert c.post("/start_profile").status_code == 200
+        assert c.post("/start_profile").status_code == 400  # double start
+        c.post("/v1/completions", json={
+            "model": "tiny-llama", "prompt": "p", "max_tokens": 3,
+            "temperature": 0.0, "ignore_eos": True})
+        r = c.post("/stop_profile")
+        assert r.status_code == 200
+        trace = r.json()["trace"]
+        assert trace.startswith(str(tmp_path))
+        import os
+        assert os.path.getsize(trace) > 0
+    state.engine.shutdown()

@@ -679,6 +679,9 @@ class ObservabilityConfig:
     first-token / finish timestamps, token counts, finish reason."""
 
     trace_file: Optional[str] = None
+    # torch.profiler output dir for /start_profile `/stop_profile`
+    # (kineto -> chrome trace; roctracer GPU events on ROCm).
+    profile_dir: str = "profile_out"
 
 
 @dataclass

@@ -134,6 +134,12 @@ class AsyncLLM:
     async def abort(self, request_id: str) -> None:
         self._inbox.put(("abort", request_id))
 
+    def start_profile(self) -> None:
+        self.engine.engine_core.start_profile()
+
+    def stop_profile(self):
+        return self.engine.engine_core.stop_profile()
+
     def check_health(self) -> None:
         """Raise when the engine cannot serve (dead worker/engine proc or
         a fatal engine-loop error)."""

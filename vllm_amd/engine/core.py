@@ -149,6 +149,13 @@ class EngineCore:
     def save_sharded_state(self, out_dir: str):
         return self.worker.save_sharded_state(out_dir)
 
+    def start_profile(self) -> None:
+        self.worker.start_profile(
+            self.config.observability_config.profile_dir)
+
+    def stop_profile(self):
+        return self.worker.stop_profile()
+
     def check_health(self) -> None:
         """Raise EngineDeadError if a worker process died (in-proc
         workers cannot die independently — no-op)."""

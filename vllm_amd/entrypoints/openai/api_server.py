@@ -174,6 +174,22 @@ def build_app(state: ServerState) -> FastAPI:
         engine.wake_up()
         return Response(status_code=200)
 
+    @app.post("/start_profile")
+    async def start_profile():
+        try:
+            engine.start_profile()
+        except Exception as e:  # noqa: BLE001
+            return _error(str(e), 400)
+        return Response(status_code=200)
+
+    @app.post("/stop_profile")
+    async def stop_profile():
+        try:
+            path = engine.stop_profile()
+        except Exception as e:  # noqa: BLE001
+            return _error(str(e), 400)
+        return {"trace": path}
+
     @app.get("/is_sleeping")
     async def is_sleeping():
         return {"is_sleeping": engine.is_sleeping()}

@@ -204,6 +204,12 @@ class MultiprocExecutor:
     def update_weights(self, model_path: str) -> None:
         self.collective_rpc("update_weights", model_path)
 
+    def start_profile(self, out_dir: str) -> None:
+        self.collective_rpc("start_profile", out_dir)
+
+    def stop_profile(self) -> list:
+        return self.collective_rpc("stop_profile")
+
     def save_sharded_state(self, out_dir: str) -> list:
         return self.collective_rpc("save_sharded_state", out_dir)
 

@@ -385,6 +385,38 @@ class ModelRunner:
         if self.device.type == "cuda":
             torch.cuda.empty_cache()
 
+    def start_profile(self, out_dir: str) -> None:
+        """torch.profiler capture (role of the reference's Worker.profile
+        / layerwise profiling; on ROCm kineto records roctracer GPU
+        events). One capture at a time; stop_profile exports a chrome
+        trace under out_dir."""
+        if getattr(self, "_profiler", None) is not None:
+            raise RuntimeError("profiler already running")
+        acts = [torch.profiler.ProfilerActivity.CPU]
+        if self.device.type == "cuda":
+            acts.append(torch.profiler.ProfilerActivity.CUDA)
+        self._profiler = torch.profiler.profile(activities=acts)
+        self._profiler_dir = out_dir
+        self._profiler.__enter__()
+
+    def stop_profile(self) -> str:
+        import os as _os
+        import time as _time
+
+        prof = getattr(self, "_profiler", None)
+        if prof is None:
+            raise RuntimeError("profiler not running")
+        prof.__exit__(None, None, None)
+        self._profiler = None
+        _os.makedirs(self._profiler_dir, exist_ok=True)
+        from vllm_amd.parallel.state import get_tp_rank
+
+        path = _os.path.join(
+            self._profiler_dir,
+            f"trace_rank{get_tp_rank()}_{int(_time.time())}.json")
+        prof.export_chrome_trace(path)
+        return path
+
     def update_weights(self, model_path: str) -> None:
         """In-place weight refresh from a safetensors dir (role of the
         reference's RL weight-update path / set_weight_version): the

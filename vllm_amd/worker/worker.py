@@ -92,6 +92,12 @@ class Worker:
     def update_weights(self, model_path: str) -> None:
         self.runner.update_weights(model_path)
 
+    def start_profile(self, out_dir: str) -> None:
+        self.runner.start_profile(out_dir)
+
+    def stop_profile(self) -> str:
+        return self.runner.stop_profile()
+
     def save_sharded_state(self, out_dir: str) -> str:
         return self.runner.save_sharded_state(out_dir)
 
