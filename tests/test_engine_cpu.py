@@ -493,3 +493,35 @@ def test_offline_chat_api():
         SamplingParams(temperature=0.0, max_tokens=3, ignore_eos=True))
     assert len(outs) == 2
     llm.shutdown()
+
+
+def test_bad_words():
+    """bad_words: the greedy-chosen token gets banned; outputs avoid the
+    banned single- and multi-token sequences."""
+    from vllm_amd.entrypoints.llm import LLM
+    from vllm_amd.sampling_params import SamplingParams
+
+    llm = LLM(model="tiny-llama", dtype="fp32", device="cpu",
+              block_size=16, num_gpu_blocks=64, max_model_len=128,
+              max_num_batched_tokens=64, max_num_seqs=2)
+    prompt = [7, 8, 9, 10, 11]
+    base = llm.generate([prompt], SamplingParams(
+        temperature=0.0, max_tokens=4, ignore_eos=True))[0]
+    first = base.outputs[0].token_ids[0]
+    # Ban exactly the token greedy would pick first (inject pre-tokenized
+    # sequence the way the engine's tokenizer pass would).
+    p = SamplingParams(temperature=0.0, max_tokens=4, ignore_eos=True,
+                       bad_words=["x"])
+    p._bad_words_token_ids = [[first]]
+    out = llm.generate([prompt], p)[0]
+    toks = out.outputs[0].token_ids
+    assert first not in toks
+    # multi-token sequence: ban (t0, t1) pair from the base continuation
+    t0, t1 = base.outputs[0].token_ids[0], base.outputs[0].token_ids[1]
+    p2 = SamplingParams(temperature=0.0, max_tokens=4, ignore_eos=True,
+                        bad_words=["y"])
+    p2._bad_words_token_ids = [[t0, t1]]
+    out2 = llm.generate([prompt], p2)[0]
+    seq = out2.outputs[0].token_ids
+    assert seq[0] == t0 and seq[1] != t1  # pair completion blocked
+    llm.shutdown()

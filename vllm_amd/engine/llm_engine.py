@@ -214,6 +214,16 @@ class LLMEngine:
             prompt=prompt_text,
             lora_id=self.config.model_config.lora_id_of(lora),
         )
+        if params.bad_words and params._bad_words_token_ids is None:
+            # Tokenize both bare and space-prefixed spellings (the
+            # reference's bad_words semantics).
+            seqs = []
+            for w in params.bad_words:
+                for variant in (w, " " + w):
+                    ids = self.tokenizer.encode(variant)
+                    if ids:
+                        seqs.append(ids)
+            params._bad_words_token_ids = seqs
         pp = self.config.parallel_config.pipeline_parallel_size
         if pp > 1 and (params.pooling or params.prompt_logprobs):
             # Hidden states / logits live on the LAST pp stage while the

@@ -41,6 +41,8 @@ class SamplingMetadata:
     output_token_ids: Optional[list[list[int]]] = None
     logit_bias: Optional[list[Optional[dict[int, float]]]] = None
     allowed_token_ids: Optional[list[Optional[list[int]]]] = None
+    # Per-row tokens banned THIS step (bad_words sequence completion).
+    bad_token_ids: Optional[list[Optional[list[int]]]] = None
     min_tokens_mask: Optional[list[Optional[tuple[int, set[int]]]]] = None
     # Structured output: per-row allowed-token sets for THIS step
     # (attached by the runner; None rows are unconstrained).
@@ -92,6 +94,19 @@ class SamplingMetadata:
                 min_tok.append((p.min_tokens, p.all_stop_token_ids))
             else:
                 min_tok.append(None)
+        bad_ids: list = []
+        for p, prompt, out in zip(params, prompt_token_ids,
+                                  output_token_ids):
+            seqs = p._bad_words_token_ids
+            if not seqs:
+                bad_ids.append(None)
+                continue
+            tail = (list(prompt) + list(out))
+            banned = []
+            for seq in seqs:
+                if len(seq) == 1 or tail[-(len(seq) - 1):] == seq[:-1]:
+                    banned.append(seq[-1])
+            bad_ids.append(banned or None)
         num_logprobs = [p.logprobs or 0 for p in params]
         return cls(
             temperature=temp,
@@ -118,6 +133,8 @@ class SamplingMetadata:
                                else allowed),
             min_tokens_mask=(None if all(m is None for m in min_tok)
                              else min_tok),
+            bad_token_ids=(None if all(b is None for b in bad_ids)
+                           else bad_ids),
         )
 
 
@@ -144,6 +161,12 @@ class Sampler(torch.nn.Module):
                                         dtype=torch.float32,
                                         device=logits.device)
                     logits[i].index_add_(0, ids, vals)
+        if meta.bad_token_ids is not None:
+            for i, banned in enumerate(meta.bad_token_ids):
+                if banned:
+                    logits[i, torch.tensor(banned,
+                                           device=logits.device)] = \
+                        float("-inf")
         if meta.allowed_token_ids is not None:
             for i, allowed in enumerate(meta.allowed_token_ids):
                 if allowed is not None:

@@ -68,6 +68,11 @@ class SamplingParams:
     guided_json: Optional[object] = None  # dict schema, JSON string, or True
     # GBNF-style EBNF grammar (non-recursive subset; guided_grammar.py).
     guided_grammar: Optional[str] = None
+    # Words that must never appear in the output (tokenized at request
+    # admission into _bad_words_token_ids; a token is banned when it
+    # would complete one of the sequences).
+    bad_words: Optional[list[str]] = None
+    _bad_words_token_ids: Optional[list[list[int]]] = None
     # response_format={"type": "json_object"}: any JSON object.
     guided_json_object: bool = False
     output_kind: RequestOutputKind = RequestOutputKind.CUMULATIVE
