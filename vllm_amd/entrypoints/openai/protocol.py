@@ -47,6 +47,7 @@ class CompletionRequest(BaseModel):
     guided_regex: Optional[str] = None
     guided_json: Optional[Union[dict, str]] = None
     guided_grammar: Optional[str] = None
+    bad_words: Optional[list[str]] = None
     # Extension (same name as the reference): per-prompt-token logprobs.
     prompt_logprobs: Optional[int] = None
 
@@ -78,6 +79,7 @@ class CompletionRequest(BaseModel):
             guided_regex=self.guided_regex,
             guided_json=self.guided_json,
             guided_grammar=self.guided_grammar,
+            bad_words=self.bad_words,
             output_kind=(RequestOutputKind.DELTA if stream
                          else RequestOutputKind.FINAL_ONLY),
         )
@@ -131,6 +133,7 @@ class ChatCompletionRequest(BaseModel):
     guided_regex: Optional[str] = None
     guided_json: Optional[Union[dict, str]] = None
     guided_grammar: Optional[str] = None
+    bad_words: Optional[list[str]] = None
     response_format: Optional[dict[str, Any]] = None
     add_generation_prompt: bool = True
     # Tool calling (OpenAI function-calling surface).
@@ -195,6 +198,7 @@ class ChatCompletionRequest(BaseModel):
             guided_regex=self.guided_regex,
             guided_json=guided_json,
             guided_grammar=self.guided_grammar,
+            bad_words=self.bad_words,
             guided_json_object=json_object,
             output_kind=(RequestOutputKind.DELTA if stream
                          else RequestOutputKind.FINAL_ONLY),
