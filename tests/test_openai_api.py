@@ -550,3 +550,52 @@ def test_profile_endpoints(tmp_path):
         import os
         assert os.path.getsize(trace) > 0
     state.engine.shutdown()
+
+
+def test_anthropic_messages_tools(client):
+    """Anthropic tool specs render into the prompt and hermes-format
+    model output maps to tool_use content blocks (no tool call emitted
+    by the tiny model -> plain text content, request still valid)."""
+    r = client.post("/v1/messages", json={
+        "model": "tiny-llama", "max_tokens": 6,
+        "messages": [{"role": "user", "content": "weather?"}],
+        "tools": [{"name": "get_weather",
+                   "description": "weather lookup",
+                   "input_schema": {"type": "object", "properties": {
+                       "city": {"type": "string"}}}}],
+        "temperature": 0.0,
+    })
+    assert r.status_code == 200, r.text
+    data = r.json()
+    assert data["type"] == "message"
+    assert data["content"][0]["type"] in ("text", "tool_use")
+    # tool_result round-trip turn is accepted
+    r = client.post("/v1/messages", json={
+        "model": "tiny-llama", "max_tokens": 4,
+        "messages": [
+            {"role": "user", "content": "weather?"},
+            {"role": "assistant", "content": "checking"},
+            {"role": "user", "content": [
+                {"type": "tool_result", "tool_use_id": "toolu_1",
+                 "content": "sunny"}]},
+        ],
+        "temperature": 0.0,
+    })
+    assert r.status_code == 200, r.text
+
+
+def test_anthropic_tool_use_mapping_unit():
+    """Unit-map a hermes call into a tool_use block via the handler's
+    parser (deterministic path check)."""
+    from vllm_amd.entrypoints.tool_parser import parse_hermes_tool_calls
+
+    text = ('<tool_call>{"name": "get_weather", "arguments": '
+            '{"city": "Oslo"}}</tool_call>')
+    content, calls = parse_hermes_tool_calls(text)
+    block = {"type": "tool_use",
+             "id": calls[0].id.replace("call_", "toolu_", 1),
+             "name": calls[0].name,
+             "input": json.loads(calls[0].arguments)}
+    assert block["name"] == "get_weather"
+    assert block["input"] == {"city": "Oslo"}
+    assert block["id"].startswith("toolu_")

@@ -37,6 +37,9 @@ class MessagesRequest(BaseModel):
     top_k: int = 0
     top_p: float = 1.0
     metadata: Optional[dict] = None
+    # Anthropic tool use: [{name, description, input_schema}].
+    tools: Optional[list[dict[str, Any]]] = None
+    tool_choice: Optional[dict[str, Any]] = None
 
 
 def build_anthropic_router(state) -> APIRouter:
@@ -49,7 +52,24 @@ def build_anthropic_router(state) -> APIRouter:
             sys_text = req.system if isinstance(req.system, str) else \
                 "".join(s.get("text", "") for s in req.system)
             parts.append(f"<|system|>\n{sys_text}")
+        if req.tools:
+            # Map Anthropic tool specs onto the hermes preamble the
+            # open models are trained for (input_schema -> parameters).
+            from vllm_amd.entrypoints.tool_parser import render_tools_block
+
+            hermes = [{"name": t.get("name"),
+                       "description": t.get("description", ""),
+                       "parameters": t.get("input_schema", {})}
+                      for t in req.tools]
+            parts.append(f"<|system|>\n{render_tools_block(hermes)}")
         for m in req.messages:
+            if isinstance(m.content, list):
+                # tool_result blocks round-trip as tool turns.
+                for seg in m.content:
+                    if seg.get("type") == "tool_result":
+                        parts.append(
+                            "<|tool|>\n"
+                            + json.dumps(seg.get("content", "")))
             parts.append(f"<|{m.role}|>\n{m.text()}")
         parts.append("<|assistant|>\n")
         return "\n".join(parts)
@@ -132,12 +152,33 @@ def build_anthropic_router(state) -> APIRouter:
         state.num_generation_tokens += len(comp.token_ids)
         stop_reason = ("max_tokens" if comp.finish_reason == "length"
                        else "end_turn")
+        content: list[dict[str, Any]] = []
+        text = comp.text
+        if req.tools:
+            # Hermes-format calls map to Anthropic tool_use blocks.
+            from vllm_amd.entrypoints.tool_parser import (
+                parse_hermes_tool_calls)
+
+            text, calls = parse_hermes_tool_calls(text)
+            for c in calls:
+                content.append({
+                    "type": "tool_use",
+                    "id": c.id.replace("call_", "toolu_", 1),
+                    "name": c.name,
+                    "input": json.loads(c.arguments or "{}"),
+                })
+            if calls:
+                stop_reason = "tool_use"
+        if text:
+            content.insert(0, {"type": "text", "text": text})
+        if not content:
+            content = [{"type": "text", "text": ""}]
         return {
             "id": rid,
             "type": "message",
             "role": "assistant",
             "model": req.model,
-            "content": [{"type": "text", "text": comp.text}],
+            "content": content,
             "stop_reason": stop_reason,
             "stop_sequence": (comp.stop_reason
                               if isinstance(comp.stop_reason, str) else None),
