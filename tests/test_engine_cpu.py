@@ -525,3 +525,25 @@ def test_bad_words():
     seq = out2.outputs[0].token_ids
     assert seq[0] == t0 and seq[1] != t1  # pair completion blocked
     llm.shutdown()
+
+
+def test_qwen3_moe_cpu_decode():
+    """Qwen3-MoE (mixtral trunk + per-head qk-norm) decodes
+    deterministically on CPU."""
+    from vllm_amd.entrypoints.llm import LLM
+    from vllm_amd.sampling_params import SamplingParams
+
+    llm = LLM(model="tiny-qwen3-moe", dtype="fp32", device="cpu",
+              block_size=16, num_gpu_blocks=128, max_model_len=256,
+              max_num_batched_tokens=64, max_num_seqs=4)
+    prompts = [[(i * 9 + j) % 900 + 3 for j in range(24)] for i in range(2)]
+    p = SamplingParams(temperature=0.0, max_tokens=6, ignore_eos=True)
+    outs = llm.generate(prompts, p)
+    outs2 = llm.generate(prompts, p)
+    # qk-norm modules really exist on the MoE trunk
+    model = llm.engine.engine_core.worker.runner.model
+    assert model.model.layers[0].self_attn.q_norm is not None
+    llm.shutdown()
+    for a, b in zip(outs, outs2):
+        assert len(a.outputs[0].token_ids) == 6
+        assert a.outputs[0].token_ids == b.outputs[0].token_ids

@@ -224,6 +224,44 @@ MODEL_PRESETS: dict[str, ModelSpec] = {
         activation="gelu",
         eos_token_id=1,
     ),
+    "qwen3-30b-a3b": ModelSpec(
+        name="qwen3-30b-a3b",
+        architecture="qwen3_moe",
+        vocab_size=151936,
+        hidden_size=2048,
+        intermediate_size=6144,
+        num_layers=48,
+        num_heads=32,
+        num_kv_heads=4,
+        head_dim=128,
+        rope_theta=1000000.0,
+        max_position_embeddings=32768,
+        rms_norm_eps=1e-6,
+        qk_norm=True,
+        num_experts=128,
+        num_experts_per_tok=8,
+        moe_intermediate_size=768,
+        norm_topk_prob=True,
+        eos_token_id=151645,
+    ),
+    "tiny-qwen3-moe": ModelSpec(
+        name="tiny-qwen3-moe",
+        architecture="qwen3_moe",
+        vocab_size=1024,
+        hidden_size=128,
+        intermediate_size=256,
+        num_layers=2,
+        num_heads=4,
+        num_kv_heads=2,
+        head_dim=32,
+        rope_theta=1000000.0,
+        max_position_embeddings=2048,
+        qk_norm=True,
+        num_experts=8,
+        num_experts_per_tok=2,
+        moe_intermediate_size=64,
+        eos_token_id=2,
+    ),
     "phi-3-mini": ModelSpec(
         name="phi-3-mini",
         architecture="phi3",
@@ -474,6 +512,8 @@ def _spec_from_hf_config(path: str) -> ModelSpec:
         architecture = "deepseek"
     elif "gemma3" in arch:
         architecture = "gemma3"
+    elif "qwen3moe" in arch or "qwen3_moe" in arch:
+        architecture = "qwen3_moe"
     elif "phi3" in arch:
         # Phi-3 is llama-structured (fused qkv/gate_up checkpoints);
         # longrope scaling beyond the original 4k context is a tracked
@@ -506,13 +546,16 @@ def _spec_from_hf_config(path: str) -> ModelSpec:
         rms_norm_eps=hf.get("rms_norm_eps", 1e-5),
         max_position_embeddings=hf.get("max_position_embeddings", 8192),
         tie_word_embeddings=hf.get("tie_word_embeddings", False),
-        num_experts=hf.get("num_local_experts", hf.get("n_routed_experts", 0)) or 0,
+        num_experts=(hf.get("num_local_experts")
+                     or hf.get("n_routed_experts")
+                     or hf.get("num_experts") or 0),
         num_experts_per_tok=hf.get("num_experts_per_tok", 0) or 0,
         moe_intermediate_size=hf.get("moe_intermediate_size", 0) or 0,
+        norm_topk_prob=hf.get("norm_topk_prob", True),
         eos_token_id=hf.get("eos_token_id", 2) or 2,
         bos_token_id=hf.get("bos_token_id", 1) or 1,
         qkv_bias=architecture == "qwen2",
-        qk_norm=architecture in ("qwen3", "gemma3"),
+        qk_norm=architecture in ("qwen3", "qwen3_moe", "gemma3"),
         sliding_window=hf.get("sliding_window") or 0,
         global_attn_every_n_layers=hf.get("sliding_window_pattern", 0) or 0,
         rope_local_theta=hf.get("rope_local_base_freq", 0.0) or 0.0,

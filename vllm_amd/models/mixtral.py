@@ -37,6 +37,12 @@ class MixtralAttention(nn.Module):
             spec.head_dim, spec.head_dim, max_position, theta=spec.rope_theta,
             rope_scaling=spec.rope_scaling,
         )
+        # Qwen3-MoE: per-head RMSNorm on q/k pre-RoPE (same knob as the
+        # dense Qwen3 trunk).
+        self.q_norm = (RMSNorm(spec.head_dim, spec.rms_norm_eps,
+                               dtype=dtype) if spec.qk_norm else None)
+        self.k_norm = (RMSNorm(spec.head_dim, spec.rms_norm_eps,
+                               dtype=dtype) if spec.qk_norm else None)
         self.attn = Attention(
             self.num_heads, spec.head_dim, scale=spec.head_dim**-0.5,
             num_kv_heads=self.num_kv_heads, layer_idx=layer_idx,
@@ -48,6 +54,11 @@ class MixtralAttention(nn.Module):
         q, k, v = self.qkv_proj.split_qkv(qkv)
         q = q.view(T, self.num_heads, self.head_dim)
         k = k.view(T, self.num_kv_heads, self.head_dim)
+        if self.q_norm is not None:
+            q = self.q_norm(
+                q.reshape(-1, self.head_dim).contiguous()).view(q.shape)
+            k = self.k_norm(
+                k.reshape(-1, self.head_dim).contiguous()).view(k.shape)
         self.rotary_emb(positions, q, k)
         out = self.attn(q, k, v)
         return self.o_proj(out)

@@ -21,7 +21,8 @@ def get_model_class(architecture: str):
         from vllm_amd.models.opt import OPTForCausalLM
 
         return OPTForCausalLM
-    if architecture == "mixtral":
+    if architecture in ("mixtral", "qwen3_moe"):
+        # Qwen3-MoE is mixtral-structured + per-head qk-norm (spec knob).
         from vllm_amd.models.mixtral import MixtralForCausalLM
 
         return MixtralForCausalLM

@@ -150,6 +150,16 @@ def load_safetensors_weights(model: torch.nn.Module, config) -> None:
                 mlp.down_proj.load_weight(w)
             continue
 
+        if ".mlp.experts." in name or name.endswith(".mlp.gate.weight"):
+            # Qwen3-MoE checkpoint layout -> mixtral-trunk names:
+            # mlp.experts.N.{gate,up,down}_proj == block_sparse_moe
+            # .experts.N.{w1,w3,w2}; mlp.gate == router.
+            name = (name
+                    .replace(".mlp.experts.", ".block_sparse_moe.experts.")
+                    .replace(".mlp.gate.", ".block_sparse_moe.gate.")
+                    .replace(".gate_proj.", ".w1.")
+                    .replace(".up_proj.", ".w3.")
+                    .replace(".down_proj.", ".w2."))
         if ".block_sparse_moe." in name:
             moe = layer.block_sparse_moe
             if "gate.weight" in name:
