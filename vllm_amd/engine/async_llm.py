@@ -49,6 +49,10 @@ class AsyncLLM:
 
     # ------------------------------------------------------------------
     def _run_loop(self) -> None:
+        import time as _time
+
+        last_log = _time.monotonic()
+        tokens_since = 0
         while not self._shutdown:
             # Drain the inbox (block briefly when idle).
             block = not self.engine.has_unfinished_requests()
@@ -92,6 +96,9 @@ class AsyncLLM:
                     return
                 continue
             for out in outputs:
+                if out.outputs:
+                    tokens_since += len(out.outputs[0].token_ids) if \
+                        out.finished else 0
                 entry = self._streams.get(out.request_id)
                 if entry is None:
                     continue
@@ -99,6 +106,24 @@ class AsyncLLM:
                 loop.call_soon_threadsafe(out_q.put_nowait, out)
                 if out.finished:
                     self._streams.pop(out.request_id, None)
+
+            # Periodic human stats line (role of the reference's
+            # LoggingStatLogger).
+            now = _time.monotonic()
+            if now - last_log >= 10.0:
+                s = self.stats()
+                queries = s.get("prefix_cache_queries", 0)
+                hit = (100.0 * s.get("prefix_cache_hits", 0) / queries
+                       if queries else 0.0)
+                logger.info(
+                    "running=%d waiting=%d finished_toks/s=%.1f "
+                    "kv_free=%d/%d prefix_hit=%.1f%%",
+                    s.get("num_running", 0), s.get("num_waiting", 0),
+                    tokens_since / (now - last_log),
+                    s.get("kv_blocks_free", 0),
+                    s.get("kv_blocks_total", 0), hit)
+                last_log = now
+                tokens_since = 0
 
     # ------------------------------------------------------------------
     def _next_id(self) -> str:
