@@ -30,3 +30,39 @@ def test_multiprocess_engine_guided_and_stops():
     out = llm.generate([[1, 2, 3]], g)[0]
     llm.shutdown()
     assert out.outputs[0].finish_reason == "stop"
+
+
+def test_multiprocess_engine_control_plane(tmp_path):
+    """sleep/wake, sharded save and in-place weight update all work
+    through the engine-process control pipe."""
+    llm = LLM(model="tiny-llama", dtype="fp32", device="cpu",
+              block_size=16, num_gpu_blocks=64, max_model_len=128,
+              max_num_batched_tokens=64, max_num_seqs=2,
+              multiprocess_engine=True)
+    p = SamplingParams(temperature=0.0, max_tokens=4, ignore_eos=True)
+    before = llm.generate([[3, 4, 5, 6]], p)[0].outputs[0].token_ids
+
+    core = llm.engine.engine_core
+    core.sleep(1)
+    assert core.is_sleeping()
+    core.wake_up()
+    after = llm.generate([[3, 4, 5, 6]], p)[0].outputs[0].token_ids
+    assert after == before
+
+    path = core.save_sharded_state(str(tmp_path))
+    import os
+    assert os.path.exists(str(tmp_path) + "/rank0_0.safetensors")
+
+    # update_weights round-trips (same weights -> same tokens)
+    from tests.test_weight_loading import _export_hf_llama
+    from vllm_amd.config import ModelConfig
+    from vllm_amd.models.registry import load_model
+    import torch
+
+    cfg = ModelConfig(model="tiny-llama", dtype="fp32")
+    model = load_model(cfg, torch.device("cpu"))
+    _export_hf_llama(model, cfg.spec, tmp_path)
+    core.update_weights(str(tmp_path))
+    again = llm.generate([[3, 4, 5, 6]], p)[0].outputs[0].token_ids
+    assert again == before
+    llm.shutdown()

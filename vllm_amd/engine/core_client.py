@@ -58,6 +58,15 @@ def _engine_proc_main(config: EngineConfig, conn) -> None:
             elif kind == "wake_up":
                 core.wake_up()
                 conn.send(("woke", None))
+            elif kind == "call":
+                # Generic idle-time control RPC (update_weights,
+                # save_sharded_state, start/stop_profile, ...).
+                method, args = msg[1], msg[2]
+                try:
+                    result = getattr(core, method)(*args)
+                    conn.send(("called", (None, result)))
+                except Exception as e:  # noqa: BLE001
+                    conn.send(("called", (repr(e), None)))
             elif kind == "shutdown":
                 running = False
         if not running:
@@ -115,7 +124,7 @@ class EngineCoreClient:
                 return
             if kind == "outputs":
                 self._outq.put(payload)
-            elif kind in ("slept", "woke"):
+            elif kind in ("slept", "woke", "called"):
                 self._ctrlq.put((kind, payload))
             elif kind == "error":
                 self._outq.put(RuntimeError(payload))
@@ -153,6 +162,29 @@ class EngineCoreClient:
 
     def is_sleeping(self) -> bool:
         return self._sleeping
+
+    def _call(self, method: str, *args):
+        self._conn.send(("call", method, args))
+        kind, (err, result) = self._ctrlq.get(timeout=300)
+        assert kind == "called"
+        if err:
+            raise RuntimeError(err)
+        return result
+
+    def update_weights(self, model_path: str) -> None:
+        if self._unfinished:
+            raise RuntimeError(
+                "cannot update weights with unfinished requests")
+        self._call("update_weights", model_path)
+
+    def save_sharded_state(self, out_dir: str):
+        return self._call("save_sharded_state", out_dir)
+
+    def start_profile(self) -> None:
+        self._call("start_profile")
+
+    def stop_profile(self):
+        return self._call("stop_profile")
 
     def check_health(self) -> None:
         from vllm_amd.executor.multiproc import EngineDeadError
