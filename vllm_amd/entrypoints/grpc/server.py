@@ -10,8 +10,6 @@ the HTTP server uses.
 
 from __future__ import annotations
 
-from typing import Optional
-
 import grpc
 from google.protobuf import descriptor_pb2, descriptor_pool, message_factory
 
