@@ -138,3 +138,29 @@ def test_medusa_matches_baseline():
     assert base == med
     assert all(len(t) == 24 for t in med)
     assert stats[0] > 0  # drafts were actually proposed and verified
+
+
+def test_dynamic_speculation_length():
+    """Poor draft acceptance shrinks the speculation length; outputs stay
+    exact (drafts are only ever rejected work, never wrong tokens)."""
+    llm = LLM(model="tiny-llama", dtype="fp32", device="cpu",
+              block_size=16, num_gpu_blocks=256, max_model_len=512,
+              max_num_batched_tokens=512, max_num_seqs=4,
+              num_speculative_tokens=4, spec_decode_method="medusa")
+    # Random-head medusa drafts are ~never accepted -> k must shrink.
+    prompts = [[(i * 17 + j) % 900 + 3 for j in range(24)]
+               for i in range(3)]
+    p = SamplingParams(temperature=0.0, max_tokens=40, ignore_eos=True)
+    outs = llm.generate(prompts, p)
+    sched = llm.engine.engine_core.scheduler
+    assert sched.spec_k < 4
+    assert all(len(o.outputs[0].token_ids) == 40 for o in outs)
+    # equality with non-speculative run
+    llm2 = LLM(model="tiny-llama", dtype="fp32", device="cpu",
+               block_size=16, num_gpu_blocks=256, max_model_len=512,
+               max_num_batched_tokens=512, max_num_seqs=4)
+    base = llm2.generate(prompts, p)
+    llm2.shutdown()
+    llm.shutdown()
+    for a, b in zip(outs, base):
+        assert a.outputs[0].token_ids == b.outputs[0].token_ids

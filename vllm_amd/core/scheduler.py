@@ -68,6 +68,14 @@ class Scheduler:
             self.spec_proposer = None
         self.spec_stats_drafted = 0
         self.spec_stats_accepted = 0
+        # Dynamic speculation length (role of the reference's dynamic
+        # per-batch speculation): start at the configured k, shrink when
+        # acceptance is poor (wasted verify compute), grow back toward
+        # the configured cap when acceptance is high.
+        self.spec_k = sc.num_speculative_tokens
+        self._spec_k_max = sc.num_speculative_tokens
+        self._spec_window_drafted = 0
+        self._spec_window_accepted = 0
         spec = config.model_config.spec
         uniform_window = (spec.sliding_window
                           if spec.sliding_window > 0
@@ -379,6 +387,17 @@ class Scheduler:
                     request.num_computed_tokens -= num_rejected
                 self.spec_stats_drafted += num_spec_sched
                 self.spec_stats_accepted += len(new_token_ids) - 1
+                self._spec_window_drafted += num_spec_sched
+                self._spec_window_accepted += len(new_token_ids) - 1
+                if self._spec_window_drafted >= 64:
+                    rate = (self._spec_window_accepted
+                            / self._spec_window_drafted)
+                    if rate < 0.25 and self.spec_k > 1:
+                        self.spec_k -= 1
+                    elif rate > 0.7 and self.spec_k < self._spec_k_max:
+                        self.spec_k += 1
+                    self._spec_window_drafted = 0
+                    self._spec_window_accepted = 0
             if not new_token_ids:
                 continue  # mid chunked-prefill, nothing sampled
 
@@ -417,7 +436,7 @@ class Scheduler:
                 else:
                     drafts = self.spec_proposer.propose(
                         request.all_token_ids)
-                request.spec_token_ids = drafts or []
+                request.spec_token_ids = (drafts or [])[: self.spec_k]
         return outputs
 
     def _check_stop(self, request: Request, last_token: int) -> bool:
