@@ -18,16 +18,27 @@ import torch
 logger = logging.getLogger(__name__)
 
 
-def _iter_safetensors(path: str):
+def _iter_safetensors(path: str, dtype=None):
     from safetensors import safe_open
 
     files = sorted(glob.glob(os.path.join(path, "*.safetensors")))
     if not files:
         raise FileNotFoundError(f"no *.safetensors under {path}")
+    # AWQ-packed checkpoints (qweight/qzeros/scales triples) are
+    # dequantized per file at load (quant_loaders.py).
     for f in files:
         with safe_open(f, framework="pt", device="cpu") as sf:
-            for name in sf.keys():
-                yield name, sf.get_tensor(name)
+            names = list(sf.keys())
+            if any(n.endswith(".qweight") for n in names):
+                from vllm_amd.quant_loaders import dequantize_awq_stream
+
+                tensors = {n: sf.get_tensor(n) for n in names}
+                for name, t in dequantize_awq_stream(
+                        tensors, dtype or torch.float32).items():
+                    yield name, t
+            else:
+                for name in names:
+                    yield name, sf.get_tensor(name)
 
 
 def sharded_state_path(out_dir: str) -> str:
@@ -78,7 +89,7 @@ def load_safetensors_weights(model: torch.nn.Module, config) -> None:
         return layers[idx]
 
     n_loaded = 0
-    for name, w in _iter_safetensors(path):
+    for name, w in _iter_safetensors(path, dtype):
         w = w.to(dtype)
         n_loaded += 1
         if name.endswith("rotary_emb.inv_freq"):

@@ -1,0 +1,94 @@
+"""Quantized-checkpoint loaders.
+
+Role of the reference's layers/quantization/awq.py checkpoint handling:
+AWQ (4-bit, group-scaled, zero-pointed) checkpoints are DEQUANTIZED at
+load into the model dtype, so any AWQ-packed model runs today; combining
+with `--quantization fp8` re-quantizes the dequantized weights to W8A8
+for the fp8 MFMA path. A native int4 dequant-GEMM kernel (keeping the 4×
+weight-memory saving at run time) is a tracked round-2 item.
+
+AWQ packing: qweight int32 [K, N/8] — eight 4-bit values per int32
+along the OUTPUT dim in the interleaved order [0, 2, 4, 6, 1, 3, 5, 7];
+qzeros int32 [K/g, N/8] same packing; scales fp16 [K/g, N]:
+    w[k, n] = (q[k, n] - zero[k//g, n]) * scale[k//g, n]
+and the HF Linear weight is the transpose [N, K].
+"""
+
+from __future__ import annotations
+
+import torch
+
+# Nibble i of the int32 holds output column base*8 + _AWQ_ORDER[i].
+_AWQ_ORDER = (0, 2, 4, 6, 1, 3, 5, 7)
+
+
+def _unpack_int4(packed: torch.Tensor) -> torch.Tensor:
+    """int32 [R, C] -> int32 [R, C*8] in logical column order."""
+    R, C = packed.shape
+    out = torch.empty(R, C * 8, dtype=torch.int32)
+    p = packed.to(torch.int64)  # avoid sign trouble on >>28
+    for i, col in enumerate(_AWQ_ORDER):
+        out[:, col::8] = ((p >> (4 * i)) & 0xF).to(torch.int32)
+    return out
+
+
+def dequant_awq(qweight: torch.Tensor, qzeros: torch.Tensor,
+                scales: torch.Tensor, dtype: torch.dtype) -> torch.Tensor:
+    """Returns the dense Linear weight [out_features, in_features]."""
+    K = qweight.shape[0]
+    q = _unpack_int4(qweight).float()               # [K, N]
+    z = _unpack_int4(qzeros).float()                # [K/g, N]
+    s = scales.float()                              # [K/g, N]
+    group = K // z.shape[0]
+    z = z.repeat_interleave(group, dim=0)           # [K, N]
+    s = s.repeat_interleave(group, dim=0)
+    w = (q - z) * s                                 # [K, N]
+    return w.t().contiguous().to(dtype)             # [N, K]
+
+
+def pack_awq(weight: torch.Tensor, group_size: int = 128,
+             zero: int = 8) -> tuple[torch.Tensor, torch.Tensor,
+                                     torch.Tensor]:
+    """Quantize a dense [N, K] weight into AWQ tensors (tests + tooling;
+    symmetric zero-point `zero`). Returns (qweight, qzeros, scales)."""
+    N, K = weight.shape
+    assert K % group_size == 0
+    w = weight.float().t().contiguous()             # [K, N]
+    wg = w.view(K // group_size, group_size, N)
+    amax = wg.abs().amax(dim=1).clamp_min(1e-8)     # [K/g, N]
+    scales = amax / max(zero, 15 - zero)
+    q = torch.round(
+        w / scales.repeat_interleave(group_size, dim=0) + zero
+    ).clamp(0, 15).to(torch.int32)                  # [K, N]
+    zeros = torch.full((K // group_size, N), zero, dtype=torch.int32)
+
+    def pack(t):
+        R, C = t.shape
+        packed = torch.zeros(R, C // 8, dtype=torch.int64)
+        for i, col in enumerate(_AWQ_ORDER):
+            packed |= t[:, col::8].to(torch.int64) << (4 * i)
+        # reinterpret the low 32 bits as int32
+        return packed.to(torch.uint32).view(torch.int32) \
+            if hasattr(torch, "uint32") else packed.to(torch.int32)
+
+    return pack(q), pack(zeros), scales.to(torch.float16)
+
+
+def dequantize_awq_stream(tensors: "dict[str, torch.Tensor]",
+                          dtype: torch.dtype):
+    """Transform a checkpoint tensor map: every {prefix}.qweight /
+    .qzeros / .scales triple becomes one {prefix}.weight dense tensor;
+    other entries pass through."""
+    prefixes = {n[: -len(".qweight")] for n in tensors if
+                n.endswith(".qweight")}
+    out = {}
+    for name, t in tensors.items():
+        stem = name.rsplit(".", 1)[0]
+        if stem in prefixes:
+            if name.endswith(".qweight"):
+                out[f"{stem}.weight"] = dequant_awq(
+                    t, tensors[f"{stem}.qzeros"],
+                    tensors[f"{stem}.scales"], dtype)
+            continue  # qzeros/scales consumed above
+        out[name] = t
+    return out
