@@ -547,3 +547,38 @@ def test_qwen3_moe_cpu_decode():
     for a, b in zip(outs, outs2):
         assert len(a.outputs[0].token_ids) == 6
         assert a.outputs[0].token_ids == b.outputs[0].token_ids
+
+
+def test_yarn_rope_scaling():
+    """YaRN: high-frequency dims keep the base rotation, low-frequency
+    dims are interpolated by `factor`, with the mscale folded into the
+    cache; positions < original ctx stay close to unscaled."""
+    import math
+
+    import torch
+
+    from vllm_amd.layers.rotary import RotaryEmbedding, _compute_inv_freq
+
+    dim, theta = 64, 10000.0
+    scaling = {"rope_type": "yarn", "factor": 4.0,
+               "original_max_position_embeddings": 1024,
+               "beta_fast": 32, "beta_slow": 1}
+    r = RotaryEmbedding(dim, dim, 4096, theta=theta, rope_scaling=scaling)
+    base = RotaryEmbedding(dim, dim, 4096, theta=theta)
+    # mscale folded in: cos row 0 is the constant mscale
+    mscale = 0.1 * math.log(4.0) + 1.0
+    assert torch.allclose(r.cos_sin_cache[0, :dim // 2],
+                          torch.full((dim // 2,), mscale), atol=1e-5)
+    # lowest-frequency dim rotates ~factor x slower than base
+    inv = _compute_inv_freq(dim, theta)
+    pos = 1000
+    ang_scaled = torch.acos(
+        (r.cos_sin_cache[pos, dim // 2 - 1] / mscale).clamp(-1, 1))
+    ang_base = (inv[-1] * pos) % (2 * math.pi)
+    ang_interp = (inv[-1] / 4.0 * pos) % (2 * math.pi)
+    # the scaled angle should match the interpolated frequency branch
+    expect = min(ang_interp, 2 * math.pi - ang_interp)
+    assert abs(float(ang_scaled) - expect) < 1e-2
+    # highest-frequency dim unchanged (up to mscale)
+    assert torch.allclose(r.cos_sin_cache[pos, 0] / mscale,
+                          base.cos_sin_cache[pos, 0], atol=1e-4)
