@@ -194,3 +194,43 @@ def test_awq_checkpoint_dequant_roundtrip(tmp_path):
     want = model_a.model.layers[0].self_attn.o_proj.weight.data
     rel = (got - want).abs().max() / want.abs().max()
     assert rel < 0.15
+
+
+def test_gptq_dequant_roundtrip():
+    """GPTQ packing (input-dim nibbles, minus-one zeros) dequantizes to
+    the original within the 4-bit grid; format auto-detected from shape
+    orientation."""
+    from vllm_amd.quant_loaders import (
+        dequant_gptq, dequantize_awq_stream)
+
+    torch.manual_seed(1)
+    N, K, g = 48, 128, 64
+    w = torch.randn(N, K) * 0.1
+    # pack GPTQ-style by hand
+    wt = w.t().contiguous()                      # [K, N]
+    wg = wt.view(K // g, g, N)
+    amax = wg.abs().amax(dim=1).clamp_min(1e-8)
+    scales = amax / 7.0
+    q = torch.round(
+        wt / scales.repeat_interleave(g, dim=0) + 8).clamp(0, 15).int()
+    qweight = torch.zeros(K // 8, N, dtype=torch.int64)
+    for i in range(8):
+        qweight |= q[i::8, :].to(torch.int64) << (4 * i)
+    qweight = qweight.to(torch.int32)
+    zeros_m1 = torch.full((K // g, N), 7, dtype=torch.int64)  # 8 - 1
+    qzeros = torch.zeros(K // g, N // 8, dtype=torch.int64)
+    for i in range(8):
+        qzeros |= zeros_m1[:, i::8] << (4 * i)
+    qzeros = qzeros.to(torch.int32)
+
+    deq = dequant_gptq(qweight, qzeros, scales.half(), torch.float32)
+    rel = (deq - w).abs().max() / w.abs().max()
+    assert rel < 0.15
+
+    # auto-detection via the stream transformer
+    out = dequantize_awq_stream({
+        "m.qweight": qweight, "m.qzeros": qzeros,
+        "m.scales": scales.half(), "other": torch.ones(3)},
+        torch.float32)
+    assert torch.equal(out["m.weight"], deq)
+    assert "other" in out and "m.qzeros" not in out

@@ -74,6 +74,49 @@ def pack_awq(weight: torch.Tensor, group_size: int = 128,
     return pack(q), pack(zeros), scales.to(torch.float16)
 
 
+# GPTQ packing (non-act-order): qweight int32 [K/8, N] — eight 4-bit
+# values per int32 along the INPUT dim in plain sequential order;
+# qzeros packed like AWQ's along N but stored MINUS ONE; scales fp16
+# [K/g, N]. Detected by qweight orientation (rows*8 == scales groups*g).
+def _unpack_int4_rows(packed: torch.Tensor) -> torch.Tensor:
+    """int32 [R, C] -> int32 [R*8, C] sequential along rows."""
+    R, C = packed.shape
+    out = torch.empty(R * 8, C, dtype=torch.int32)
+    p = packed.to(torch.int64)
+    for i in range(8):
+        out[i::8, :] = ((p >> (4 * i)) & 0xF).to(torch.int32)
+    return out
+
+
+def _unpack_int4_cols_seq(packed: torch.Tensor) -> torch.Tensor:
+    """int32 [R, C] -> int32 [R, C*8] sequential along columns (GPTQ
+    qzeros layout)."""
+    R, C = packed.shape
+    out = torch.empty(R, C * 8, dtype=torch.int32)
+    p = packed.to(torch.int64)
+    for i in range(8):
+        out[:, i::8] = ((p >> (4 * i)) & 0xF).to(torch.int32)
+    return out
+
+
+def dequant_gptq(qweight: torch.Tensor, qzeros: torch.Tensor,
+                 scales: torch.Tensor, dtype: torch.dtype,
+                 g_idx: torch.Tensor = None) -> torch.Tensor:
+    """Dense Linear weight [out, in] from GPTQ tensors (act-order g_idx
+    supported when provided)."""
+    q = _unpack_int4_rows(qweight).float()          # [K, N]
+    z = _unpack_int4_cols_seq(qzeros).float() + 1   # stored minus one
+    s = scales.float()                              # [K/g, N]
+    K = q.shape[0]
+    group = K // s.shape[0]
+    if g_idx is not None and g_idx.numel() == K:
+        gi = g_idx.long()
+    else:
+        gi = torch.arange(K) // group
+    w = (q - z[gi]) * s[gi]                         # [K, N]
+    return w.t().contiguous().to(dtype)
+
+
 def dequantize_awq_stream(tensors: "dict[str, torch.Tensor]",
                           dtype: torch.dtype):
     """Transform a checkpoint tensor map: every {prefix}.qweight /
@@ -86,9 +129,25 @@ def dequantize_awq_stream(tensors: "dict[str, torch.Tensor]",
         stem = name.rsplit(".", 1)[0]
         if stem in prefixes:
             if name.endswith(".qweight"):
-                out[f"{stem}.weight"] = dequant_awq(
-                    t, tensors[f"{stem}.qzeros"],
-                    tensors[f"{stem}.scales"], dtype)
-            continue  # qzeros/scales consumed above
+                scales = tensors[f"{stem}.scales"]
+                qzeros = tensors[f"{stem}.qzeros"]
+                # Orientation tells the formats apart: AWQ packs the
+                # OUTPUT dim into int32 columns (cols*8 == N == scales
+                # cols); GPTQ packs the INPUT dim into rows (cols == N).
+                if t.shape[1] * 8 == scales.shape[1]:
+                    out[f"{stem}.weight"] = dequant_awq(
+                        t, qzeros, scales, dtype)
+                elif t.shape[1] == scales.shape[1]:
+                    out[f"{stem}.weight"] = dequant_gptq(
+                        t, qzeros, scales, dtype,
+                        tensors.get(f"{stem}.g_idx"))
+                else:
+                    raise ValueError(
+                        f"unrecognized int4 packing for {stem}: "
+                        f"qweight {tuple(t.shape)} vs scales "
+                        f"{tuple(scales.shape)}")
+            continue  # qzeros/scales/g_idx consumed above
+        if name.endswith(".g_idx"):
+            continue
         out[name] = t
     return out
