@@ -582,3 +582,25 @@ def test_yarn_rope_scaling():
     # highest-frequency dim unchanged (up to mscale)
     assert torch.allclose(r.cos_sin_cache[pos, 0] / mscale,
                           base.cos_sin_cache[pos, 0], atol=1e-4)
+
+
+def test_custom_logits_processors():
+    """Offline logits_processors: callables reshape the distribution per
+    step (here: force a fixed token)."""
+    from vllm_amd.entrypoints.llm import LLM
+    from vllm_amd.sampling_params import SamplingParams
+
+    def force_42(output_ids, logits):
+        logits = logits.clone()
+        logits[:] = float("-inf")
+        logits[42] = 0.0
+        return logits
+
+    llm = LLM(model="tiny-llama", dtype="fp32", device="cpu",
+              block_size=16, num_gpu_blocks=64, max_model_len=128,
+              max_num_batched_tokens=64, max_num_seqs=2)
+    [out] = llm.generate([[3, 4, 5]], SamplingParams(
+        temperature=0.0, max_tokens=5, ignore_eos=True,
+        logits_processors=[force_42]))
+    llm.shutdown()
+    assert out.outputs[0].token_ids == [42] * 5

@@ -43,6 +43,8 @@ class SamplingMetadata:
     allowed_token_ids: Optional[list[Optional[list[int]]]] = None
     # Per-row tokens banned THIS step (bad_words sequence completion).
     bad_token_ids: Optional[list[Optional[list[int]]]] = None
+    # Per-row custom processors fn(output_ids, logits_row) -> logits_row.
+    logits_processors: Optional[list] = None
     min_tokens_mask: Optional[list[Optional[tuple[int, set[int]]]]] = None
     # Structured output: per-row allowed-token sets for THIS step
     # (attached by the runner; None rows are unconstrained).
@@ -135,6 +137,9 @@ class SamplingMetadata:
                              else min_tok),
             bad_token_ids=(None if all(b is None for b in bad_ids)
                            else bad_ids),
+            logits_processors=(
+                None if all(not p.logits_processors for p in params)
+                else [p.logits_processors for p in params]),
         )
 
 
@@ -161,6 +166,13 @@ class Sampler(torch.nn.Module):
                                         dtype=torch.float32,
                                         device=logits.device)
                     logits[i].index_add_(0, ids, vals)
+        if meta.logits_processors is not None:
+            for i, procs in enumerate(meta.logits_processors):
+                if procs:
+                    row = logits[i]
+                    for fn in procs:
+                        row = fn(meta.output_token_ids[i], row)
+                    logits[i] = row
         if meta.bad_token_ids is not None:
             for i, banned in enumerate(meta.bad_token_ids):
                 if banned:

@@ -73,6 +73,10 @@ class SamplingParams:
     # would complete one of the sequences).
     bad_words: Optional[list[str]] = None
     _bad_words_token_ids: Optional[list[list[int]]] = None
+    # Offline-API pluggable processors: fn(output_token_ids, logits) ->
+    # logits, applied in request order before temperature (role of the
+    # reference's per-request logits_processors).
+    logits_processors: Optional[list] = None
     # response_format={"type": "json_object"}: any JSON object.
     guided_json_object: bool = False
     output_kind: RequestOutputKind = RequestOutputKind.CUMULATIVE
