@@ -599,3 +599,22 @@ def test_anthropic_tool_use_mapping_unit():
     assert block["name"] == "get_weather"
     assert block["input"] == {"city": "Oslo"}
     assert block["id"].startswith("toolu_")
+
+
+def test_chat_stream_n2(client):
+    with client.stream("POST", "/v1/chat/completions", json={
+            "model": "tiny-llama",
+            "messages": [{"role": "user", "content": "hi"}],
+            "max_tokens": 4, "n": 2, "temperature": 0.0,
+            "ignore_eos": True, "stream": True}) as r:
+        finishes = {}
+        idxs = set()
+        for ln in r.iter_lines():
+            if ln.startswith("data: ") and ln != "data: [DONE]":
+                ch = json.loads(ln[6:])["choices"]
+                if ch:
+                    idxs.add(ch[0]["index"])
+                    if ch[0].get("finish_reason"):
+                        finishes[ch[0]["index"]] = ch[0]["finish_reason"]
+    assert idxs == {0, 1}
+    assert finishes == {0: "length", 1: "length"}

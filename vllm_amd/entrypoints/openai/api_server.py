@@ -662,8 +662,6 @@ def build_app(state: ServerState) -> FastAPI:
     # ------------------------------------------------------------------
     @app.post("/v1/chat/completions")
     async def chat_completions(req: ChatCompletionRequest, raw: Request):
-        if req.n != 1 and req.stream:
-            return _error("n>1 is not supported with streaming chat")
         from vllm_amd.entrypoints import tool_parser as tp
 
         tools_on = bool(req.tools) and req.tool_choice != "none"
@@ -679,46 +677,37 @@ def build_app(state: ServerState) -> FastAPI:
         rid = random_id("chatcmpl")
 
         if req.stream:
-            async def gen() -> AsyncGenerator[str, None]:
-                first = ChatCompletionStreamResponse(
-                    id=rid, model=req.model,
-                    choices=[ChatStreamChoice(
-                        index=0, delta=DeltaMessage(role="assistant",
-                                                    content=""))],
-                )
-                yield f"data: {first.model_dump_json()}\n\n"
-                from vllm_amd.metrics import RequestTimer
-                timer = RequestTimer(state.metrics)
-                n_gen = 0
-                n_prompt = 0
-                rparse = (tp.StreamingReasoningParser()
-                          if state.reasoning_parser else None)
-                tparse = (tp.make_streaming_tool_parser(
-                              state.tool_call_parser)
-                          if tools_on and not named else None)
-                named_id = tp._call_id() if named else None
-                named_first = True
+            class _Branch:
+                """Per-branch streaming parse state (reasoning + tool
+                parsers, named-call bookkeeping, delta accounting)."""
 
-                def deltas(text, finish=None):
-                    """Split one raw text delta through the active
-                    parsers into zero or more DeltaMessages."""
+                def __init__(self):
+                    self.rparse = (tp.StreamingReasoningParser()
+                                   if state.reasoning_parser else None)
+                    self.tparse = (tp.make_streaming_tool_parser(
+                                       state.tool_call_parser)
+                                   if tools_on and not named else None)
+                    self.named_id = tp._call_id() if named else None
+                    self.named_first = True
+                    self.seen_toks = 0
+
+                def deltas(self, text, finish=None):
                     out = []
                     reasoning = ""
-                    if rparse is not None:
-                        reasoning, text = rparse.feed(text)
+                    if self.rparse is not None:
+                        reasoning, text = self.rparse.feed(text)
                         if finish is not None:
-                            r2, t2 = rparse.flush()
+                            r2, t2 = self.rparse.flush()
                             reasoning += r2
                             text += t2
                     if named:
                         calls = []
                         if text:
                             fn = {"arguments": text}
-                            nonlocal named_first
-                            if named_first:
+                            if self.named_first:
                                 fn["name"] = named
-                                named_first = False
-                                calls = [{"index": 0, "id": named_id,
+                                self.named_first = False
+                                calls = [{"index": 0, "id": self.named_id,
                                           "type": "function",
                                           "function": fn}]
                             else:
@@ -728,13 +717,13 @@ def build_app(state: ServerState) -> FastAPI:
                             tool_calls=calls or None))
                         return out, bool(calls)
                     saw = False
-                    if tparse is not None:
-                        text, calls = tparse.feed(text)
+                    if self.tparse is not None:
+                        text, calls = self.tparse.feed(text)
                         if finish is not None:
-                            t2, c2 = tparse.flush()
+                            t2, c2 = self.tparse.flush()
                             text += t2
                             calls += c2
-                        saw = tparse.saw_tool_call
+                        saw = self.tparse.saw_tool_call
                         if calls:
                             out.append(DeltaMessage(tool_calls=calls))
                     if text or reasoning:
@@ -742,18 +731,37 @@ def build_app(state: ServerState) -> FastAPI:
                             content=text or None,
                             reasoning_content=reasoning or None))
                     return out, saw
-                seen_toks = 0
+
+            branches = _branch_params(params, req.n)
+
+            async def gen() -> AsyncGenerator[str, None]:
+                from vllm_amd.metrics import RequestTimer
+                timer = RequestTimer(state.metrics)
+                n_gen = 0
+                n_prompt = 0
+                states = [_Branch() for _ in branches]
+                for b in range(len(branches)):
+                    first = ChatCompletionStreamResponse(
+                        id=rid, model=req.model,
+                        choices=[ChatStreamChoice(
+                            index=b, delta=DeltaMessage(role="assistant",
+                                                        content=""))],
+                    )
+                    yield f"data: {first.model_dump_json()}\n\n"
+                gens = [engine.generate(prompt, bp, f"{rid}-{b}",
+                                        lora=lora)
+                        for b, bp in enumerate(branches)]
                 try:
-                    async for out in engine.generate(prompt, params, rid,
-                                                     lora=lora):
+                    async for b, out in _merge_streams(gens):
+                        st = states[b]
                         comp = out.outputs[0]
-                        new = len(comp.token_ids) - seen_toks
-                        seen_toks = len(comp.token_ids)
+                        new = len(comp.token_ids) - st.seen_toks
+                        st.seen_toks = len(comp.token_ids)
                         state.num_generation_tokens += new
                         timer.on_tokens(new)
                         n_gen += new
                         n_prompt = len(out.prompt_token_ids)
-                        msgs, saw = deltas(comp.text, comp.finish_reason)
+                        msgs, saw = st.deltas(comp.text, comp.finish_reason)
                         finish = comp.finish_reason
                         if finish and (saw or named):
                             finish = "tool_calls" if finish == "stop" \
@@ -763,7 +771,7 @@ def build_app(state: ServerState) -> FastAPI:
                             chunk = ChatCompletionStreamResponse(
                                 id=rid, model=req.model,
                                 choices=[ChatStreamChoice(
-                                    index=0, delta=d,
+                                    index=b, delta=d,
                                     finish_reason=finish if last else None,
                                 )],
                             )
@@ -772,7 +780,7 @@ def build_app(state: ServerState) -> FastAPI:
                             chunk = ChatCompletionStreamResponse(
                                 id=rid, model=req.model,
                                 choices=[ChatStreamChoice(
-                                    index=0, delta=DeltaMessage(),
+                                    index=b, delta=DeltaMessage(),
                                     finish_reason=finish)],
                             )
                             yield f"data: {chunk.model_dump_json()}\n\n"
