@@ -604,3 +604,35 @@ def test_custom_logits_processors():
         logits_processors=[force_42]))
     llm.shutdown()
     assert out.outputs[0].token_ids == [42] * 5
+
+
+def test_include_stop_str_in_output():
+    """The matched stop string is trimmed by default and kept with
+    include_stop_str_in_output=True."""
+    from vllm_amd.entrypoints.llm import LLM
+    from vllm_amd.sampling_params import SamplingParams
+    llm = LLM(model="tiny-llama", dtype="fp32", device="cpu",
+              block_size=16, num_gpu_blocks=64, max_model_len=128,
+              max_num_batched_tokens=64, max_num_seqs=2)
+    prompt = [5, 9, 13, 17]
+
+    # Force the byte-level mock tokenizer to spell "abcdab..." so the
+    # stop string is a real, printable substring.
+    pattern = [ord(c) for c in "abcdab"]
+
+    def spell(output_ids, logits):
+        logits = logits.clone()
+        logits[:] = float("-inf")
+        logits[pattern[len(output_ids) % len(pattern)]] = 0.0
+        return logits
+
+    stop = "cd"
+    common = dict(temperature=0.0, max_tokens=6, ignore_eos=True,
+                  logits_processors=[spell])
+    [a] = llm.generate([prompt], SamplingParams(stop=[stop], **common))
+    [b] = llm.generate([prompt], SamplingParams(
+        stop=[stop], include_stop_str_in_output=True, **common))
+    llm.shutdown()
+    assert stop not in a.outputs[0].text
+    assert b.outputs[0].text.endswith(stop)
+    assert b.outputs[0].text == a.outputs[0].text + stop

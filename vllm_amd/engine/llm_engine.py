@@ -97,7 +97,10 @@ class OutputProcessor:
                 for s in state.params.stop:
                     idx = text.find(s)
                     if idx != -1:
-                        state.detokenizer.output_text = text[:idx]
+                        end = (idx + len(s)
+                               if state.params.include_stop_str_in_output
+                               else idx)
+                        state.detokenizer.output_text = text[:end]
                         finish_reason = "stop"
                         stop_reason = s
                         if not co.finish_reason:

@@ -43,6 +43,8 @@ class SamplingParams:
     min_p: float = 0.0
     seed: Optional[int] = None
     stop: Union[None, str, list[str]] = None
+    # Keep the matched stop string in the output text (reference param).
+    include_stop_str_in_output: bool = False
     stop_token_ids: Optional[list[int]] = None
     ignore_eos: bool = False
     max_tokens: Optional[int] = 16

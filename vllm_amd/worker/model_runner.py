@@ -695,6 +695,10 @@ class ModelRunner:
             and not meta.generators and meta.logit_bias is None
             and meta.allowed_token_ids is None
             and meta.min_tokens_mask is None
+            # bad_words masks and custom processors read the CURRENT
+            # output tail — never reuse them across steps.
+            and meta.bad_token_ids is None
+            and meta.logits_processors is None
             and all(st.grammar is None for st in states)
         )
         if cacheable:
