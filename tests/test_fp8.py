@@ -152,3 +152,28 @@ def test_engine_fp8_gpu_e2e():
     assert all(len(o.outputs[0].token_ids) == 8 for o in outs)
     for a, b in zip(outs, outs2):
         assert a.outputs[0].token_ids == b.outputs[0].token_ids
+
+
+def test_engine_fp8_moe_and_mla_cpu():
+    """fp8 W8A8 on the MoE (mixtral) and MLA (deepseek) trunks: dense
+    linears quantized, expert weights stay in model dtype, decode is
+    deterministic."""
+    from vllm_amd.entrypoints.llm import LLM
+    from vllm_amd.sampling_params import SamplingParams
+
+    for model in ("tiny-mixtral", "tiny-deepseek"):
+        llm = LLM(model=model, dtype="fp32", device="cpu",
+                  quantization="fp8", block_size=16, num_gpu_blocks=128,
+                  max_model_len=128, max_num_batched_tokens=64,
+                  max_num_seqs=2)
+        p = SamplingParams(temperature=0.0, max_tokens=4, ignore_eos=True)
+        prompt = [[(3 * j) % 900 + 5 for j in range(12)]]
+        a = llm.generate(prompt, p)
+        b = llm.generate(prompt, p)
+        # at least one dense linear got quantized
+        runner_model = llm.engine.engine_core.worker.runner.model
+        n_q = sum(1 for m in runner_model.modules()
+                  if getattr(m, "weight_fp8", None) is not None)
+        llm.shutdown()
+        assert n_q > 0, model
+        assert a[0].outputs[0].token_ids == b[0].outputs[0].token_ids, model
