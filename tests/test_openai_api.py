@@ -618,3 +618,30 @@ def test_chat_stream_n2(client):
                         finishes[ch[0]["index"]] = ch[0]["finish_reason"]
     assert idxs == {0, 1}
     assert finishes == {0: "length", 1: "length"}
+
+
+def test_multi_turn_prefix_cache_hit(client):
+    """Second chat turn extends the first conversation: its prompt must
+    hit the prefix cache (prefix_cache_hits grows)."""
+    conv = [{"role": "user", "content": "tell me about caching " * 8}]
+    r1 = client.post("/v1/chat/completions", json={
+        "model": "tiny-llama", "messages": conv, "max_tokens": 4,
+        "temperature": 0.0, "ignore_eos": True})
+    assert r1.status_code == 200
+    before = {ln.split()[0]: float(ln.split()[-1])
+              for ln in client.get("/metrics").text.splitlines()
+              if ln.startswith("vllm_amd:prefix_cache")}
+    conv = conv + [
+        {"role": "assistant",
+         "content": r1.json()["choices"][0]["message"]["content"] or "ok"},
+        {"role": "user", "content": "more"},
+    ]
+    r2 = client.post("/v1/chat/completions", json={
+        "model": "tiny-llama", "messages": conv, "max_tokens": 4,
+        "temperature": 0.0, "ignore_eos": True})
+    assert r2.status_code == 200
+    after = {ln.split()[0]: float(ln.split()[-1])
+             for ln in client.get("/metrics").text.splitlines()
+             if ln.startswith("vllm_amd:prefix_cache")}
+    assert after["vllm_amd:prefix_cache_hits_total"] > \
+        before["vllm_amd:prefix_cache_hits_total"]
