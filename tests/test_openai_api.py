@@ -645,3 +645,20 @@ def test_multi_turn_prefix_cache_hit(client):
              if ln.startswith("vllm_amd:prefix_cache")}
     assert after["vllm_amd:prefix_cache_hits_total"] > \
         before["vllm_amd:prefix_cache_hits_total"]
+
+
+def test_priority_param_reaches_scheduler():
+    from vllm_amd.engine.arg_utils import EngineArgs
+    from vllm_amd.entrypoints.openai.api_server import make_server
+
+    args = EngineArgs(model="tiny-llama", dtype="fp32", device="cpu",
+                      block_size=16, num_gpu_blocks=64, max_model_len=128,
+                      max_num_batched_tokens=64, max_num_seqs=4,
+                      scheduling_policy="priority")
+    app, state = make_server(args)
+    with TestClient(app) as c:
+        r = c.post("/v1/completions", json={
+            "model": "tiny-llama", "prompt": "p", "max_tokens": 2,
+            "temperature": 0.0, "ignore_eos": True, "priority": -5})
+        assert r.status_code == 200, r.text
+    state.engine.shutdown()

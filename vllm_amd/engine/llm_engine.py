@@ -215,6 +215,7 @@ class LLMEngine:
             sampling_params=params,
             eos_token_id=eos,
             prompt=prompt_text,
+            priority=params.priority,
             lora_id=self.config.model_config.lora_id_of(lora),
         )
         if params.bad_words and params._bad_words_token_ids is None:

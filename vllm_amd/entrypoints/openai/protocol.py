@@ -48,6 +48,7 @@ class CompletionRequest(BaseModel):
     guided_json: Optional[Union[dict, str]] = None
     guided_grammar: Optional[str] = None
     bad_words: Optional[list[str]] = None
+    priority: int = 0  # lower = sooner (priority policy)
     # Extension (same name as the reference): per-prompt-token logprobs.
     prompt_logprobs: Optional[int] = None
 
@@ -80,6 +81,7 @@ class CompletionRequest(BaseModel):
             guided_json=self.guided_json,
             guided_grammar=self.guided_grammar,
             bad_words=self.bad_words,
+            priority=self.priority,
             output_kind=(RequestOutputKind.DELTA if stream
                          else RequestOutputKind.FINAL_ONLY),
         )
@@ -134,6 +136,7 @@ class ChatCompletionRequest(BaseModel):
     guided_json: Optional[Union[dict, str]] = None
     guided_grammar: Optional[str] = None
     bad_words: Optional[list[str]] = None
+    priority: int = 0  # lower = sooner (priority policy)
     response_format: Optional[dict[str, Any]] = None
     add_generation_prompt: bool = True
     # Tool calling (OpenAI function-calling surface).
@@ -199,6 +202,7 @@ class ChatCompletionRequest(BaseModel):
             guided_json=guided_json,
             guided_grammar=self.guided_grammar,
             bad_words=self.bad_words,
+            priority=self.priority,
             guided_json_object=json_object,
             output_kind=(RequestOutputKind.DELTA if stream
                          else RequestOutputKind.FINAL_ONLY),

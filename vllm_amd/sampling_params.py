@@ -79,6 +79,9 @@ class SamplingParams:
     # logits, applied in request order before temperature (role of the
     # reference's per-request logits_processors).
     logits_processors: Optional[list] = None
+    # Scheduling priority (lower = sooner under --scheduling-policy
+    # priority; ties broken FCFS).
+    priority: int = 0
     # response_format={"type": "json_object"}: any JSON object.
     guided_json_object: bool = False
     output_kind: RequestOutputKind = RequestOutputKind.CUMULATIVE
