@@ -662,3 +662,13 @@ def test_priority_param_reaches_scheduler():
             "temperature": 0.0, "ignore_eos": True, "priority": -5})
         assert r.status_code == 200, r.text
     state.engine.shutdown()
+
+
+def test_usage_reports_cached_tokens(client):
+    body = {"model": "tiny-llama",
+            "prompt": "repeatable cached prompt text " * 4,
+            "max_tokens": 2, "temperature": 0.0, "ignore_eos": True}
+    client.post("/v1/completions", json=body)
+    r = client.post("/v1/completions", json=body)
+    details = r.json()["usage"].get("prompt_tokens_details")
+    assert details and details["cached_tokens"] > 0

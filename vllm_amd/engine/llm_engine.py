@@ -37,6 +37,7 @@ class RequestState:
     finish_time: Optional[float] = None
     pooled: Optional[list[float]] = None
     prompt_logprobs: Optional[list] = None
+    num_cached_tokens: int = 0
 
 
 class OutputProcessor:
@@ -109,6 +110,8 @@ class OutputProcessor:
 
             if co.pooled is not None:
                 state.pooled = co.pooled
+            if co.num_cached_tokens:
+                state.num_cached_tokens = co.num_cached_tokens
             if co.new_prompt_logprobs:
                 state.prompt_logprobs = (
                     (state.prompt_logprobs or []) + co.new_prompt_logprobs)
@@ -153,6 +156,7 @@ class OutputProcessor:
             metrics=metrics,
             pooled=getattr(state, "pooled", None),
             prompt_logprobs=getattr(state, "prompt_logprobs", None),
+            num_cached_tokens=getattr(state, "num_cached_tokens", 0),
         )
 
     def release(self, request_id: str) -> None:

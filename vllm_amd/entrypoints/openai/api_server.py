@@ -398,10 +398,13 @@ def build_app(state: ServerState) -> FastAPI:
         timer.on_finish(n_prompt, n_gen)
         state.num_prompt_tokens += n_prompt
         state.num_generation_tokens += n_gen
+        cached = max(f.num_cached_tokens for f in finals)
         usage = UsageInfo(
             prompt_tokens=n_prompt,
             completion_tokens=n_gen,
             total_tokens=n_prompt + n_gen,
+            prompt_tokens_details=({"cached_tokens": cached}
+                                   if cached else None),
         )
         choices = []
         for b, final in enumerate(finals):
@@ -822,10 +825,13 @@ def build_app(state: ServerState) -> FastAPI:
         timer.on_finish(n_prompt, n_gen)
         state.num_prompt_tokens += n_prompt
         state.num_generation_tokens += n_gen
+        cached = max(f.num_cached_tokens for f in finals)
         usage = UsageInfo(
             prompt_tokens=n_prompt,
             completion_tokens=n_gen,
             total_tokens=n_prompt + n_gen,
+            prompt_tokens_details=({"cached_tokens": cached}
+                                   if cached else None),
         )
         choices = []
         for b, final in enumerate(finals):

@@ -213,6 +213,8 @@ class UsageInfo(BaseModel):
     prompt_tokens: int = 0
     completion_tokens: int = 0
     total_tokens: int = 0
+    # OpenAI prompt_tokens_details parity: prefix-cache hits.
+    prompt_tokens_details: Optional[dict[str, int]] = None
 
 
 class CompletionChoice(BaseModel):

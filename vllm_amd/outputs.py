@@ -34,6 +34,8 @@ class RequestOutput:
     # SamplingParams.prompt_logprobs: one {token: logprob} dict per
     # prompt token starting at index 1 (token 0 is unconditioned).
     prompt_logprobs: Optional[list[dict[int, float]]] = None
+    # Prompt tokens served from the prefix cache (usage reporting).
+    num_cached_tokens: int = 0
 
     def __repr__(self) -> str:
         return (
