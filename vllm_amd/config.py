@@ -32,6 +32,9 @@ class ModelSpec:
     head_dim: int = 128
     rope_theta: float = 500000.0
     rope_scaling: Optional[dict] = None
+    # Rotate only the first fraction of each head (GLM/Phi-2 style);
+    # 1.0 = full-head rope. The HIP rope kernel takes rotary_dim.
+    partial_rotary_factor: float = 1.0
     rms_norm_eps: float = 1e-5
     max_position_embeddings: int = 8192
     tie_word_embeddings: bool = False
@@ -546,6 +549,7 @@ def _spec_from_hf_config(path: str) -> ModelSpec:
         rms_norm_eps=hf.get("rms_norm_eps", 1e-5),
         max_position_embeddings=hf.get("max_position_embeddings", 8192),
         tie_word_embeddings=hf.get("tie_word_embeddings", False),
+        partial_rotary_factor=hf.get("partial_rotary_factor", 1.0) or 1.0,
         num_experts=(hf.get("num_local_experts")
                      or hf.get("n_routed_experts")
                      or hf.get("num_experts") or 0),

@@ -69,9 +69,11 @@ class LlamaAttention(nn.Module):
             spec.num_heads * spec.head_dim, spec.hidden_size,
             bias=spec.use_bias, dtype=dtype,
         )
+        rotary_dim = int(spec.head_dim * spec.partial_rotary_factor)
+        rotary_dim -= rotary_dim % 2
         self.rotary_emb = RotaryEmbedding(
             spec.head_dim,
-            spec.head_dim,
+            rotary_dim,
             max_position,
             theta=spec.rope_theta,
             rope_scaling=spec.rope_scaling,
