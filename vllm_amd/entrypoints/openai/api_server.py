@@ -311,6 +311,8 @@ def build_app(state: ServerState) -> FastAPI:
     async def completions(req: CompletionRequest, raw: Request):
         if (req.best_of or 1) != 1:
             return _error("best_of is not supported")
+        if not 1 <= req.n <= 64:
+            return _error("n must be in [1, 64]")
         prompts = req.prompt
         if isinstance(prompts, str):
             prompts = [prompts]
@@ -665,6 +667,8 @@ def build_app(state: ServerState) -> FastAPI:
     # ------------------------------------------------------------------
     @app.post("/v1/chat/completions")
     async def chat_completions(req: ChatCompletionRequest, raw: Request):
+        if not 1 <= req.n <= 64:
+            return _error("n must be in [1, 64]")
         from vllm_amd.entrypoints import tool_parser as tp
 
         tools_on = bool(req.tools) and req.tool_choice != "none"
