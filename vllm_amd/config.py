@@ -691,6 +691,10 @@ class ParallelConfig:
     # API process only tokenizes/detokenizes (reference EngineCoreProc).
     multiprocess_engine: bool = False
     data_parallel_size: int = 1
+    # Serve-level DP replicas: replica i pins its GPUs starting at this
+    # device index, and multiproc workers rendezvous on worker_port.
+    device_offset: int = 0
+    worker_port: Optional[int] = None
     # Filled from env (RANK/LOCAL_RANK/WORLD_SIZE) when launched by torchrun.
     rank: int = 0
     local_rank: int = 0

@@ -52,6 +52,7 @@ class EngineArgs:
 
     tensor_parallel_size: int = 1
     pipeline_parallel_size: int = 1
+    data_parallel_size: int = 1
     enable_expert_parallel: bool = False
     multiprocess_engine: bool = False
     device: str = "auto"
@@ -107,6 +108,10 @@ class EngineArgs:
                             default=1)
         parser.add_argument("--pipeline-parallel-size", "-pp", type=int,
                             default=1)
+        parser.add_argument("--data-parallel-size", "-dp", type=int,
+                            default=1,
+                            help="serve-level engine replicas "
+                                 "(least-loaded request routing)")
         parser.add_argument("--enable-expert-parallel", action="store_true")
         parser.add_argument("--multiprocess-engine", action="store_true",
                             help="run the engine core in its own process")
@@ -134,6 +139,7 @@ class EngineArgs:
         pc = ParallelConfig(
             tensor_parallel_size=self.tensor_parallel_size,
             pipeline_parallel_size=self.pipeline_parallel_size,
+            data_parallel_size=self.data_parallel_size,
             enable_expert_parallel=self.enable_expert_parallel,
             multiprocess_engine=self.multiprocess_engine,
             rank=int(os.environ.get("RANK", "0")),

@@ -11,6 +11,7 @@ from __future__ import annotations
 
 import asyncio
 import logging
+import os
 import queue
 import threading
 import time
@@ -208,3 +209,99 @@ class AsyncLLM:
         self._shutdown = True
         self._thread.join(timeout=5)
         self.engine.shutdown()
+
+
+class DPAsyncLLM:
+    """Serve-level data-parallel router (role of the reference's
+    vllm/v1/engine/core_client.py DPLBAsyncMPClient: --data-parallel-size
+    engine replicas behind one API server, least-loaded routing).
+
+    Replica i owns GPUs [i*tp, (i+1)*tp): its config gets
+    device_offset=i*tp (uniproc workers take the device directly from
+    local_rank; multiproc workers add the offset per rank) and a unique
+    worker_port so the replicas' TP process groups don't collide. Each
+    replica runs its own engine loop; new requests go to the replica
+    with the fewest in-flight requests. xGMI note: one replica's TP
+    traffic stays on its own GPUs' links, so replicas don't contend."""
+
+    def __init__(self, config: EngineConfig):
+        import copy
+
+        dp = config.parallel_config.data_parallel_size
+        tp = config.parallel_config.tensor_parallel_size
+        base_port = int(os.environ.get("VLLM_AMD_WORKER_PORT", "29533"))
+        self.replicas: list[AsyncLLM] = []
+        for i in range(dp):
+            c = copy.deepcopy(config)
+            c.parallel_config.data_parallel_size = 1
+            c.parallel_config.device_offset = i * tp
+            c.parallel_config.local_rank = i * tp
+            c.parallel_config.worker_port = base_port + i
+            self.replicas.append(AsyncLLM(c))
+        self.config = config
+        self._in_flight = [0] * dp
+        self._counter = 0
+        self._counter_lock = threading.Lock()
+
+    @property
+    def tokenizer(self):
+        return self.replicas[0].tokenizer
+
+    def _next_id(self) -> str:
+        with self._counter_lock:
+            self._counter += 1
+            return f"req-{self._counter}"
+
+    def _pick(self) -> int:
+        return min(range(len(self.replicas)),
+                   key=lambda i: self._in_flight[i])
+
+    async def generate(self, prompt, sampling_params=None,
+                       request_id=None, lora=None):
+        rid = request_id or self._next_id()
+        i = self._pick()
+        self._in_flight[i] += 1
+        try:
+            async for out in self.replicas[i].generate(
+                    prompt, sampling_params, request_id=rid, lora=lora):
+                yield out
+        finally:
+            self._in_flight[i] -= 1
+
+    async def abort(self, request_id: str) -> None:
+        for r in self.replicas:
+            await r.abort(request_id)
+
+    def start_profile(self) -> None:
+        for r in self.replicas:
+            r.start_profile()
+
+    def stop_profile(self):
+        return [r.stop_profile() for r in self.replicas]
+
+    def check_health(self) -> None:
+        for r in self.replicas:
+            r.check_health()
+
+    def sleep(self, level: int = 1) -> None:
+        for r in self.replicas:
+            r.sleep(level)
+
+    def wake_up(self) -> None:
+        for r in self.replicas:
+            r.wake_up()
+
+    def is_sleeping(self) -> bool:
+        return any(r.is_sleeping() for r in self.replicas)
+
+    def stats(self) -> dict:
+        agg: dict = {}
+        for r in self.replicas:
+            for k, v in r.stats().items():
+                agg[k] = agg.get(k, 0) + v
+        agg["dp_in_flight"] = list(self._in_flight)
+        return agg
+
+    def shutdown(self) -> None:
+        for r in self.replicas:
+            r.shutdown()

@@ -880,7 +880,13 @@ def make_server(engine_args: EngineArgs,
                 reasoning_parser: Optional[str] = None,
                 api_key: Optional[str] = None,
                 tool_call_parser: str = "hermes"):
-    engine = AsyncLLM(engine_args.create_engine_config())
+    config = engine_args.create_engine_config()
+    if config.parallel_config.data_parallel_size > 1:
+        from vllm_amd.engine.async_llm import DPAsyncLLM
+
+        engine = DPAsyncLLM(config)
+    else:
+        engine = AsyncLLM(config)
     state = ServerState(engine, served_model_name or engine_args.model,
                         reasoning_parser=reasoning_parser,
                         api_key=api_key,

@@ -31,7 +31,8 @@ def _worker_main(rank: int, config: EngineConfig, conn, master_port: int):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(master_port)
     config.parallel_config.rank = rank
-    config.parallel_config.local_rank = rank
+    config.parallel_config.local_rank = (
+        config.parallel_config.device_offset + rank)
     config.parallel_config.world_size = \
         config.parallel_config.tensor_parallel_size
 
@@ -138,7 +139,8 @@ class MultiprocExecutor:
         self.config = config
         tp = config.parallel_config.tensor_parallel_size
         ctx = mp.get_context("spawn")
-        port = int(os.environ.get("VLLM_AMD_WORKER_PORT", "29533"))
+        port = (config.parallel_config.worker_port
+                or int(os.environ.get("VLLM_AMD_WORKER_PORT", "29533")))
         self.conns = []
         self.procs = []
         for rank in range(tp):

@@ -7,6 +7,7 @@ every forward signature.
 
 from __future__ import annotations
 
+import threading
 from contextlib import contextmanager
 from dataclasses import dataclass
 from typing import Optional
@@ -40,23 +41,25 @@ class ForwardContext:
     lora_manager: object = None
 
 
-_forward_context: Optional[ForwardContext] = None
+# Thread-local: serve-level DP replicas run one engine loop per thread
+# in the same process, each with its own per-step context.
+_tls = threading.local()
 
 
 def get_forward_context() -> ForwardContext:
-    assert _forward_context is not None, (
+    ctx = getattr(_tls, "ctx", None)
+    assert ctx is not None, (
         "forward context not set — model forward must run under "
         "set_forward_context()"
     )
-    return _forward_context
+    return ctx
 
 
 @contextmanager
 def set_forward_context(ctx: ForwardContext):
-    global _forward_context
-    prev = _forward_context
-    _forward_context = ctx
+    prev = getattr(_tls, "ctx", None)
+    _tls.ctx = ctx
     try:
         yield
     finally:
-        _forward_context = prev
+        _tls.ctx = prev
