@@ -164,3 +164,58 @@ def test_dynamic_speculation_length():
     llm.shutdown()
     for a, b in zip(outs, base):
         assert a.outputs[0].token_ids == b.outputs[0].token_ids
+
+
+# ---------------------------------------------------------------- EAGLE
+
+def test_eagle_draft_chunk_shapes_and_determinism():
+    import torch
+
+    from vllm_amd.spec_decode.eagle import EagleDraft, _ReqKV
+
+    d = EagleDraft(hidden_size=64, num_heads=4, num_kv_heads=2,
+                   intermediate_size=128, dtype=torch.float32)
+    d.init_dummy(seed=0)
+    kv = _ReqKV(64, 2, 16, torch.float32, torch.device("cpu"))
+    inp = torch.randn(5, 128)
+    g = d.forward_chunk(inp, 0, kv)
+    assert g.shape == (5, 64)
+    assert kv.len == 5
+    # Incremental single-slot continuation matches a fresh full pass.
+    inp2 = torch.randn(1, 128)
+    g2 = d.forward_chunk(inp2, 5, kv)
+    kv_b = _ReqKV(64, 2, 16, torch.float32, torch.device("cpu"))
+    g_full = d.forward_chunk(torch.cat([inp, inp2]), 0, kv_b)
+    assert torch.allclose(g2[-1], g_full[-1], atol=1e-5)
+    # Rollback: rewriting slot 5 after truncation reuses the slot.
+    kv.len = 5
+    g3 = d.forward_chunk(inp2, 5, kv)
+    assert torch.allclose(g2, g3)
+
+
+def _generate_eagle(spec_tokens: int):
+    llm = LLM(model="tiny-llama", dtype="fp32", device="cpu",
+              block_size=16, num_gpu_blocks=256, max_model_len=512,
+              max_num_batched_tokens=512, max_num_seqs=4,
+              num_speculative_tokens=spec_tokens,
+              spec_decode_method="eagle")
+    prompt = [7, 8, 9, 10] * 12
+    outs = llm.generate(
+        [prompt, list(range(30, 60))],
+        SamplingParams(temperature=0.0, max_tokens=24, ignore_eos=True),
+    )
+    sched = llm.engine.engine_core.scheduler
+    stats = (sched.spec_stats_drafted, sched.spec_stats_accepted)
+    llm.shutdown()
+    return [o.outputs[0].token_ids for o in outs], stats
+
+
+def test_eagle_matches_baseline():
+    """Random (untrained) EAGLE draft: drafts are ~all rejected, but
+    greedy in-place verification keeps outputs EXACTLY equal to the
+    non-speculative run — correctness never depends on draft quality."""
+    base, _ = _generate(0)
+    egl, stats = _generate_eagle(3)
+    assert base == egl
+    assert all(len(t) == 24 for t in egl)
+    assert stats[0] > 0  # drafts were actually produced and verified

@@ -664,14 +664,16 @@ class SchedulerConfig:
     # role of the reference's AsyncScheduler + async model-runner output).
     async_scheduling: bool = True
     # Speculative decoding (0 = off). Spec decode forces synchronous
-    # scheduling. Methods: "ngram" (prompt-lookup, CPU) or "medusa"
-    # (model-based multi-head drafts from the runner).
+    # scheduling. Methods: "ngram" (prompt-lookup, CPU), "medusa"
+    # (model-based multi-head drafts from the runner) or "eagle"
+    # (autoregressive one-layer feature draft, runner-side).
     num_speculative_tokens: int = 0
     spec_decode_method: str = "ngram"
     ngram_prompt_lookup_min: int = 2
     ngram_prompt_lookup_max: int = 4
-    # Optional dir with medusa-head safetensors; None -> random heads.
+    # Optional dirs with draft safetensors; None -> random-init drafts.
     medusa_path: Optional[str] = None
+    eagle_path: Optional[str] = None
 
     def __post_init__(self) -> None:
         if self.num_speculative_tokens > 0:

@@ -56,6 +56,10 @@ class Scheduler:
                 from vllm_amd.spec_decode.medusa import MedusaProposer
 
                 self.spec_proposer = MedusaProposer()
+            elif sc.spec_decode_method == "eagle":
+                from vllm_amd.spec_decode.eagle import EagleProposer
+
+                self.spec_proposer = EagleProposer()
             else:
                 from vllm_amd.spec_decode.ngram import NgramProposer
 

@@ -46,6 +46,7 @@ class EngineArgs:
     num_speculative_tokens: int = 0
     spec_decode_method: str = "ngram"
     medusa_path: Optional[str] = None
+    eagle_path: Optional[str] = None
     ngram_prompt_lookup_min: int = 2
     ngram_prompt_lookup_max: int = 4
     async_scheduling: bool = True
@@ -98,8 +99,9 @@ class EngineArgs:
             help="LoRA adapters to serve (requests select by model name)")
         parser.add_argument("--num-speculative-tokens", type=int, default=0)
         parser.add_argument("--spec-decode-method", type=str,
-                            default="ngram", choices=["ngram", "medusa"])
+                            default="ngram", choices=["ngram", "medusa", "eagle"])
         parser.add_argument("--medusa-path", type=str, default=None)
+        parser.add_argument("--eagle-path", type=str, default=None)
         parser.add_argument("--ngram-prompt-lookup-min", type=int, default=2)
         parser.add_argument("--ngram-prompt-lookup-max", type=int, default=4)
         parser.add_argument("--no-async-scheduling",
@@ -179,6 +181,7 @@ class EngineArgs:
                 num_speculative_tokens=self.num_speculative_tokens,
                 spec_decode_method=self.spec_decode_method,
                 medusa_path=self.medusa_path,
+                eagle_path=self.eagle_path,
                 ngram_prompt_lookup_min=self.ngram_prompt_lookup_min,
                 ngram_prompt_lookup_max=self.ngram_prompt_lookup_max,
             ),

@@ -215,6 +215,7 @@ class ModelRunner:
         self.num_host_blocks = 0
         self.model: Optional[torch.nn.Module] = None
         self.medusa = None  # MedusaHeads when spec_decode_method=medusa
+        self.eagle = None  # EagleRunnerSide when spec_decode_method=eagle
         self.graph_runner: Optional[DecodeGraphRunner] = None
         self.sampler = Sampler()
         self.max_model_len = config.model_config.max_model_len
@@ -264,6 +265,27 @@ class ModelRunner:
             else:
                 self.medusa.init_dummy(self.config.model_config.seed)
             self.medusa = self.medusa.to(self.device).eval()
+        if (sc.num_speculative_tokens > 0
+                and sc.spec_decode_method == "eagle"):
+            from vllm_amd.spec_decode.eagle import (EagleDraft,
+                                                    EagleRunnerSide)
+
+            draft = EagleDraft(
+                self.spec.hidden_size, self.spec.num_heads,
+                self.spec.num_kv_heads, self.spec.intermediate_size,
+                self.dtype, rope_theta=self.spec.rope_theta)
+            if sc.eagle_path:
+                draft.load_safetensors(sc.eagle_path, self.dtype)
+            else:
+                draft.init_dummy(self.config.model_config.seed)
+            draft = draft.to(self.device).eval()
+            self.eagle = EagleRunnerSide(
+                draft,
+                embed=self.model.model.embed_tokens,
+                compute_logits=self.model.compute_logits,
+                k=sc.num_speculative_tokens,
+                max_len=self.max_model_len,
+                dtype=self.dtype, device=self.device)
         mc = self.config.model_config
         self.lora_manager = None
         if mc.lora_modules:
@@ -467,6 +489,8 @@ class ModelRunner:
             if self.requests.pop(req_id, None) is not None:
                 row = self._row_of.pop(req_id)
                 self._free_rows.append(row)
+                if self.eagle is not None:
+                    self.eagle.free(req_id)
         for nr in so.scheduled_new_reqs:
             self.requests[nr.req_id] = CachedReqState(
                 req_id=nr.req_id,
@@ -618,6 +642,16 @@ class ModelRunner:
             # the critical path anyway.
             draft_map = dict(zip(
                 req_ids, self.medusa.propose(hidden[:n]).cpu().tolist()))
+        if self.eagle is not None:
+            # EAGLE forces sync scheduling too: the draft loop needs the
+            # sampled token and this step's hidden on the host path.
+            toks = sampled_t.cpu().tolist()
+            draft_map = {}
+            for j, rid in enumerate(req_ids):
+                tok = int(toks[j])
+                self.eagle.observe(rid, int(positions[j]),
+                                   hidden[j:j + 1], [tok])
+                draft_map[rid] = self.eagle.propose(rid, tok)
 
         if self.device.type == "cuda" and s_out.logprobs is None:
             pin = self._pin_sampled[self._pin_idx]
@@ -1000,6 +1034,35 @@ class ModelRunner:
             h = hidden[torch.tensor(med_pos, device=dev)]
             drafts = self.medusa.propose(h).cpu().tolist()
             draft_map = {req_ids[r]: d for r, d in zip(med_rows, drafts)}
+        if self.eagle is not None:
+            # Feed every ACCEPTED position's (target hidden, next token)
+            # pair into the draft, then propose for sampling rows. Kept
+            # positions per row: chunk length minus rejected drafts.
+            draft_map = {}
+            sampling_set = {r: npos for r, npos
+                            in zip(sampling_rows, sampling_npos)}
+            for i, rid in enumerate(req_ids):
+                state = self.requests.get(rid)
+                if state is None or state.sampling_params.pooling:
+                    continue
+                q0 = int(query_start_loc[i])
+                q1 = int(query_start_loc[i + 1])
+                nq = q1 - q0
+                kept = nq
+                if i in sampling_set and sampling_set[i] > 1:
+                    kept = nq - (sampling_set[i]
+                                 - len(sampled_per_req[i]))
+                if kept <= 0:
+                    continue
+                p0 = int(positions[q0])
+                next_toks = state.token_ids[p0 + 1: p0 + kept + 1]
+                if len(next_toks) < kept:
+                    continue
+                self.eagle.observe(rid, p0, hidden[q0:q0 + kept],
+                                   next_toks)
+                if i in sampling_set and sampled_per_req[i]:
+                    draft_map[rid] = self.eagle.propose(
+                        rid, sampled_per_req[i][-1])
         # Mixed steps resolve on the CPU; invalidate the device-side
         # sampled-token carry so the next decode reads np_last_tok.
         self._last_sampled = None
