@@ -732,6 +732,9 @@ class ObservabilityConfig:
     first-token / finish timestamps, token counts, finish reason."""
 
     trace_file: Optional[str] = None
+    # OTLP/HTTP collector endpoint (reference --otlp-traces-endpoint):
+    # one OTEL span per finished request, gen_ai.* attributes.
+    otlp_traces_endpoint: Optional[str] = None
     # torch.profiler output dir for /start_profile `/stop_profile`
     # (kineto -> chrome trace; roctracer GPU events on ROCm).
     profile_dir: str = "profile_out"

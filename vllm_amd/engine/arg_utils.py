@@ -58,6 +58,7 @@ class EngineArgs:
     multiprocess_engine: bool = False
     device: str = "auto"
     trace_file: Optional[str] = None
+    otlp_traces_endpoint: Optional[str] = None
 
     @staticmethod
     def add_cli_args(parser: argparse.ArgumentParser) -> argparse.ArgumentParser:
@@ -120,6 +121,10 @@ class EngineArgs:
         parser.add_argument("--device", type=str, default="auto")
         parser.add_argument("--trace-file", type=str, default=None,
                             help="JSONL request-trace output path")
+        parser.add_argument("--otlp-traces-endpoint", type=str,
+                            default=None,
+                            help="OTLP/HTTP collector URL; one OTEL "
+                                 "span per finished request")
         return parser
 
     @classmethod
@@ -188,5 +193,6 @@ class EngineArgs:
             parallel_config=pc,
             device_config=DeviceConfig(device=self.device),
             observability_config=ObservabilityConfig(
-                trace_file=self.trace_file),
+                trace_file=self.trace_file,
+                otlp_traces_endpoint=self.otlp_traces_endpoint),
         )

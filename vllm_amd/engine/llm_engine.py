@@ -180,6 +180,12 @@ class LLMEngine:
         trace_file = config.observability_config.trace_file
         if trace_file:
             self._trace_fh = open(trace_file, "a", buffering=1)
+        self._otel = None
+        otlp = config.observability_config.otlp_traces_endpoint
+        if otlp:
+            from vllm_amd.tracing import OtelSpanExporter
+
+            self._otel = OtelSpanExporter(otlp)
 
     @property
     def is_driver(self) -> bool:
@@ -284,6 +290,15 @@ class LLMEngine:
             if out.finished:
                 if self._trace_fh is not None:
                     self._write_trace(out)
+                if self._otel is not None:
+                    comp = out.outputs[0]
+                    self._otel.export_request_span(
+                        out.request_id,
+                        self.config.model_config.model,
+                        out.metrics or {},
+                        len(out.prompt_token_ids),
+                        len(comp.token_ids),
+                        comp.finish_reason)
                 self.output_processor.release(out.request_id)
                 if self.engine_core.scheduler is not None:
                     self.engine_core.scheduler.release_request(
@@ -333,4 +348,7 @@ class LLMEngine:
         if self._trace_fh is not None:
             self._trace_fh.close()
             self._trace_fh = None
+        if self._otel is not None:
+            self._otel.shutdown()
+            self._otel = None
         self.engine_core.shutdown()
