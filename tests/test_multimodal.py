@@ -1,0 +1,85 @@
+"""Llava-style multimodal path on CPU (reference vllm/multimodal/ +
+models/llava.py): image placeholders expanded to per-patch tokens,
+vision features scattered over placeholder embeddings, prefix-cache
+salted by image content, chunked prefill invariance."""
+
+import numpy as np
+import torch
+
+from vllm_amd.entrypoints.llm import LLM
+from vllm_amd.multimodal import expand_image_placeholders
+from vllm_amd.sampling_params import SamplingParams
+
+IMG = 1000  # tiny-llava image_token_id; 16 patches (32px / 8px patch)
+
+
+def _llm(**kw):
+    return LLM(model="tiny-llava", dtype="fp32", device="cpu",
+               block_size=16, num_gpu_blocks=128, max_model_len=512,
+               max_num_batched_tokens=kw.pop("mnbt", 512),
+               max_num_seqs=4, **kw)
+
+
+def _gen(llm, image, extra=0):
+    prompt = {"prompt_token_ids": [5, 6, IMG, 7, 8, 9 + extra],
+              "multi_modal_data": {"image": image}}
+    outs = llm.generate([prompt], SamplingParams(
+        max_tokens=8, temperature=0.0, ignore_eos=True))
+    return outs[0].outputs[0].token_ids
+
+
+def test_expand_placeholders():
+    out = expand_image_placeholders([1, IMG, 2], IMG, 4, 1)
+    assert out == [1] + [IMG] * 4 + [2]
+
+
+def test_image_changes_output_and_is_deterministic():
+    rng = np.random.default_rng(0)
+    img_a = rng.normal(size=(3, 32, 32)).astype(np.float32)
+    img_b = rng.normal(size=(3, 32, 32)).astype(np.float32)
+    llm = _llm()
+    a1 = _gen(llm, img_a)
+    a2 = _gen(llm, img_a)
+    b = _gen(llm, img_b)
+    # Text-only same prompt (no image): also different.
+    t = llm.generate([[5, 6, 7, 8, 9]], SamplingParams(
+        max_tokens=8, temperature=0.0,
+        ignore_eos=True))[0].outputs[0].token_ids
+    llm.shutdown()
+    assert len(a1) == 8
+    assert a1 == a2              # deterministic + safe feature cache
+    assert a1 != b               # image content reaches the logits
+    assert a1 != t
+    # Different images, same tokens: prefix cache must NOT cross-hit
+    # (a1==a2 while a1!=b already proves the salt worked: the second
+    # a-run hit the cache, the b-run could not reuse those blocks).
+
+
+def test_chunked_prefill_invariance():
+    rng = np.random.default_rng(1)
+    img = rng.normal(size=(3, 32, 32)).astype(np.float32)
+    big = _llm()
+    whole = _gen(big, img)
+    big.shutdown()
+    small = _llm(mnbt=16)  # image span forced across prefill chunks
+    chunked = _gen(small, img)
+    small.shutdown()
+    assert whole == chunked
+
+
+def test_torch_tensor_image_and_rejects_text_only_model():
+    img = torch.randn(3, 32, 32)
+    llm = _llm()
+    assert len(_gen(llm, img)) == 8
+    llm.shutdown()
+    llm2 = LLM(model="tiny-llama", dtype="fp32", device="cpu",
+               block_size=16, num_gpu_blocks=64, max_model_len=256,
+               max_num_batched_tokens=256, max_num_seqs=2)
+    try:
+        import pytest
+        with pytest.raises(Exception, match="no vision encoder"):
+            llm2.generate([{"prompt_token_ids": [5, 6],
+                            "multi_modal_data": {"image": img}}],
+                          SamplingParams(max_tokens=2))
+    finally:
+        llm2.shutdown()

@@ -74,6 +74,14 @@ class ModelSpec:
     v_head_dim: int = 0
     eos_token_id: int = 2
     bos_token_id: int = 1
+    # Vision (llava-style): 0 layers = text-only. Prompts carry
+    # image_token_id placeholders, one per patch after expansion.
+    vision_layers: int = 0
+    vision_hidden_size: int = 0
+    vision_heads: int = 0
+    vision_patch: int = 14
+    image_size: int = 336
+    image_token_id: int = 0
 
     @property
     def is_moe(self) -> bool:
@@ -94,6 +102,27 @@ class ModelSpec:
 
 # Built-in presets matching BASELINE.json's named configs.
 MODEL_PRESETS: dict[str, ModelSpec] = {
+    # Tiny llava-style vision-language model (CPU tests).
+    "tiny-llava": ModelSpec(
+        name="tiny-llava",
+        architecture="llama",
+        vocab_size=1024,
+        hidden_size=64,
+        intermediate_size=128,
+        num_layers=2,
+        num_heads=4,
+        num_kv_heads=2,
+        head_dim=16,
+        rope_theta=10000.0,
+        max_position_embeddings=2048,
+        vision_layers=2,
+        vision_hidden_size=32,
+        vision_heads=2,
+        vision_patch=8,
+        image_size=32,
+        image_token_id=1000,
+        eos_token_id=2,
+    ),
     "opt-125m": ModelSpec(
         name="opt-125m",
         architecture="opt",

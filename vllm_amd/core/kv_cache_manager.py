@@ -79,7 +79,8 @@ class KVCacheManager:
             self.block_size,
             request.all_token_ids,
             prior_hashes=request.block_hashes,
-            salt=getattr(request, "lora_id", 0) or None,
+            salt=((getattr(request, "lora_id", 0)
+                   ^ getattr(request, "mm_hash", 0)) or None),
         )
         computed: list[KVCacheBlock] = []
         for h in request.block_hashes:
@@ -163,7 +164,8 @@ class KVCacheManager:
                 self.block_size,
                 request.all_token_ids,
                 prior_hashes=request.block_hashes,
-                salt=getattr(request, "lora_id", 0) or None,
+                salt=((getattr(request, "lora_id", 0)
+                   ^ getattr(request, "mm_hash", 0)) or None),
             )
             # Only blocks whose tokens are all known can be hashed; with
             # chunked prefill total_tokens <= num_tokens so this holds.

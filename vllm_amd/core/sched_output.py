@@ -22,6 +22,7 @@ class NewRequestData:
     sampling_params: SamplingParams
     grammar: object = None  # CompiledGrammar for structured output
     lora_id: int = 0
+    mm_data: object = None  # {"image": pixels} for vision models
 
 
 @dataclass

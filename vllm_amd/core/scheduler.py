@@ -306,6 +306,7 @@ class Scheduler:
                         sampling_params=request.sampling_params,
                         grammar=request.grammar,
                         lora_id=request.lora_id,
+                        mm_data=request.mm_data,
                     )
                 )
             num_scheduled_tokens[request.request_id] = num_new_tokens

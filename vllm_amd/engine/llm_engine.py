@@ -202,6 +202,36 @@ class LLMEngine:
             request_id = f"req-{self._request_counter}"
             self._request_counter += 1
         params = params or SamplingParams()
+        mm_data = None
+        mm_hash = 0
+        if isinstance(prompt, dict):
+            # {"prompt" | "prompt_token_ids", "multi_modal_data":
+            #  {"image": pixels [3,S,S]}} (reference TextPrompt /
+            # TokensPrompt with multi_modal_data).
+            mm_data = prompt.get("multi_modal_data")
+            if "prompt_token_ids" in prompt:
+                prompt = list(prompt["prompt_token_ids"])
+            else:
+                prompt = prompt["prompt"]
+            if mm_data and mm_data.get("image") is not None:
+                from vllm_amd.multimodal import (expand_image_placeholders,
+                                                 mm_content_hash)
+
+                spec = self.config.model_config.spec
+                if spec.vision_layers == 0:
+                    raise ValueError(
+                        f"model {spec.name} has no vision encoder")
+                if isinstance(prompt, str):
+                    prompt = self.tokenizer.encode(prompt)
+                npatch = (spec.image_size // spec.vision_patch) ** 2
+                import torch as _torch
+
+                nimg = (1 if _torch.as_tensor(
+                    mm_data["image"]).dim() == 3 else
+                    _torch.as_tensor(mm_data["image"]).shape[0])
+                prompt = expand_image_placeholders(
+                    prompt, spec.image_token_id, npatch, nimg)
+                mm_hash = mm_content_hash(mm_data)
         if isinstance(prompt, str):
             prompt_text = prompt
             prompt_token_ids = self.tokenizer.encode(prompt)
@@ -227,6 +257,8 @@ class LLMEngine:
             prompt=prompt_text,
             priority=params.priority,
             lora_id=self.config.model_config.lora_id_of(lora),
+            mm_data=mm_data,
+            mm_hash=mm_hash,
         )
         if params.bad_words and params._bad_words_token_ids is None:
             # Tokenize both bare and space-prefixed spellings (the

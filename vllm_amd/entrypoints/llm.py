@@ -69,7 +69,7 @@ class LLM:
         sorted best-first (token_ids exclude the prompt)."""
         from vllm_amd.sampling_params import SamplingParams
 
-        if isinstance(prompts, str) or (
+        if isinstance(prompts, (str, dict)) or (
             prompts and isinstance(prompts, list)
             and isinstance(prompts[0], int)
         ):
@@ -121,7 +121,7 @@ class LLM:
         ] = None,
         lora: Optional[str] = None,
     ) -> list[RequestOutput]:
-        if isinstance(prompts, str) or (
+        if isinstance(prompts, (str, dict)) or (
             prompts and isinstance(prompts, list)
             and isinstance(prompts[0], int)
         ):

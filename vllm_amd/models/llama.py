@@ -14,6 +14,7 @@ import torch.nn as nn
 from vllm_amd.config import ModelConfig, ModelSpec
 from vllm_amd.layers.activation import SiluAndMul
 from vllm_amd.layers.attention import Attention
+from vllm_amd.worker.forward_context import get_forward_context
 from vllm_amd.layers.embedding import ParallelLMHead, VocabParallelEmbedding
 from vllm_amd.layers.layernorm import RMSNorm
 from vllm_amd.layers.linear import (
@@ -174,6 +175,11 @@ class LlamaModel(nn.Module):
     def forward(self, input_ids, positions, hidden_in=None):
         if self.embed_tokens is not None:
             hidden = self.embed_tokens(input_ids)
+            mm = get_forward_context().mm_embeds
+            if mm is not None:
+                idx, feats = mm
+                hidden = hidden.index_copy(
+                    0, idx, feats.to(hidden.dtype))
             residual = None
         else:
             hidden = hidden_in

@@ -50,6 +50,8 @@ class Request:
         priority: int = 0,
         prompt: Optional[str] = None,
         lora_id: int = 0,
+        mm_data: Optional[dict] = None,
+        mm_hash: int = 0,
     ) -> None:
         self.request_id = request_id
         self.prompt_token_ids = prompt_token_ids
@@ -59,6 +61,10 @@ class Request:
         self.arrival_time = arrival_time if arrival_time is not None else time.time()
         self.priority = priority
         self.lora_id = lora_id
+        # Multimodal payload ({"image": pixels}) + content hash used to
+        # salt prefix-cache block hashes (vllm mm_hash extra_keys role).
+        self.mm_data = mm_data
+        self.mm_hash = mm_hash
 
         self.status = RequestStatus.WAITING
         self.stop_reason: Optional[object] = None

@@ -39,6 +39,9 @@ class ForwardContext:
     # Multi-LoRA: per-token adapter ids (0 = base) + the manager.
     lora_ids: Optional[torch.Tensor] = None
     lora_manager: object = None
+    # Vision: (flat token indices in this batch, feature rows) to
+    # scatter over placeholder embeddings right after embed_tokens.
+    mm_embeds: Optional[tuple] = None
 
 
 # Thread-local: serve-level DP replicas run one engine loop per thread

@@ -41,6 +41,9 @@ class CachedReqState:
     grammar_state: object = None
     lora_id: int = 0
     pool_acc: object = None  # running hidden sum for mean pooling
+    mm_data: object = None        # {"image": pixels}
+    mm_feats: object = None       # encoded-once projected patch rows
+    mm_img_pos: object = None     # np positions of image tokens in prompt
 
     @property
     def output_token_ids(self) -> list[int]:
@@ -215,6 +218,7 @@ class ModelRunner:
         self.num_host_blocks = 0
         self.model: Optional[torch.nn.Module] = None
         self.medusa = None  # MedusaHeads when spec_decode_method=medusa
+        self.vision = None  # VisionTower when spec.vision_layers > 0
         self.eagle = None  # EagleRunnerSide when spec_decode_method=eagle
         self.graph_runner: Optional[DecodeGraphRunner] = None
         self.sampler = Sampler()
@@ -252,6 +256,16 @@ class ModelRunner:
 
     def load_model(self) -> None:
         self.model = load_model(self.config.model_config, self.device)
+        if self.spec.vision_layers > 0:
+            from vllm_amd.multimodal import VisionTower
+
+            self.vision = VisionTower(
+                self.spec.image_size, self.spec.vision_patch,
+                self.spec.vision_hidden_size, self.spec.vision_layers,
+                self.spec.vision_heads, self.spec.hidden_size,
+                self.dtype)
+            self.vision.init_dummy(self.config.model_config.seed)
+            self.vision = self.vision.to(self.device).eval()
         sc = self.config.scheduler_config
         if (sc.num_speculative_tokens > 0
                 and sc.spec_decode_method == "medusa"):
@@ -503,6 +517,7 @@ class ModelRunner:
                 grammar_state=(nr.grammar.initial_state()
                                if nr.grammar is not None else None),
                 lora_id=nr.lora_id,
+                mm_data=nr.mm_data,
             )
             row = self._free_rows.pop()
             self._row_of[nr.req_id] = row
@@ -840,10 +855,42 @@ class ModelRunner:
                 np.full(n_, self.np_lora[self._row_of[rid]])
                 for rid, n_ in items
             ])
+        mm_embeds = None
+        if self.vision is not None:
+            img_id = self.spec.image_token_id
+            mm_idx, mm_rows = [], []
+            for i, (rid, _nq) in enumerate(items):
+                state = self.requests[rid]
+                if state.mm_data is None:
+                    continue
+                q0 = int(query_start_loc[i])
+                q1 = int(query_start_loc[i + 1])
+                sel = input_ids[q0:q1] == img_id
+                if not sel.any():
+                    continue
+                if state.mm_feats is None:
+                    pix = torch.as_tensor(
+                        state.mm_data["image"]).to(self.device)
+                    state.mm_feats = self.vision(pix)
+                    state.mm_img_pos = np.nonzero(np.asarray(
+                        state.token_ids[:state.prompt_len]) == img_id)[0]
+                # Feature row of an image token = its ordinal among the
+                # prompt's image tokens (chunking-invariant).
+                rows_np = np.searchsorted(state.mm_img_pos,
+                                          positions[q0:q1][sel])
+                mm_idx.append(np.nonzero(sel)[0] + q0)
+                mm_rows.append(state.mm_feats[
+                    torch.from_numpy(rows_np).to(self.device)])
+            if mm_idx:
+                mm_embeds = (
+                    torch.from_numpy(
+                        np.concatenate(mm_idx)).to(dev),
+                    torch.cat(mm_rows))
         ctx = ForwardContext(
             attn_metadata=meta, kv_caches=self.kv_caches,
             lora_ids=self._lora_ids_tensor(lora_np),
             lora_manager=self.lora_manager,
+            mm_embeds=mm_embeds,
         )
         if self.pp_size > 1:
             # Stage boundary: recv [T, hidden] from the previous
