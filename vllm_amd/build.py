@@ -22,6 +22,7 @@ OUT_SO = PKG_DIR / "_C.so"
 
 SOURCES = [
     CSRC / "elementwise.hip",
+    CSRC / "comms.hip",
     CSRC / "quant_fp8.hip",
     CSRC / "attention_decode.hip",
     CSRC / "attention_prefill.hip",

@@ -89,13 +89,23 @@ class KVCacheManager:
                 block = self._materialize_host_hit(h)
             if block is None:
                 break
+            # Hold a ref while the chain walk continues: a later
+            # _materialize_host_hit allocates a GPU block, and without
+            # the ref it could evict — or reuse as its H2D destination —
+            # a block already collected here (ref 0 in the free queue),
+            # silently corrupting the reported prefix.
+            self.block_pool.touch([block])
             computed.append(block)
         num_computed = len(computed) * self.block_size
         # Never report the full prompt as computed: at least the last token
         # must be recomputed to produce logits.
         if num_computed >= request.num_tokens:
-            computed.pop()
+            self.block_pool.free_blocks([computed.pop()])
             num_computed -= self.block_size
+        # Release the walk refs (blocks re-enter the free queue MRU-end);
+        # allocate_slots re-touches the hits it commits immediately after.
+        for b in computed:
+            self.block_pool.free_blocks([b])
         return computed, num_computed
 
     def allocate_slots(

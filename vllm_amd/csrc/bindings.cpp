@@ -38,6 +38,14 @@ void dynamic_quant_fp8(torch::Tensor out, torch::Tensor scales,
 void scale_rows_cols(torch::Tensor y, torch::Tensor row_scales,
                      torch::Tensor col_scales,
                      c10::optional<torch::Tensor> bias);
+torch::Tensor car_init(int64_t rank, int64_t world, int64_t max_bytes);
+void car_connect(torch::Tensor handles);
+bool car_is_ready();
+int64_t car_max_bytes();
+int64_t car_error();
+void car_destroy();
+void car_all_reduce(torch::Tensor t);
+void car_all_gather(torch::Tensor out, torch::Tensor t);
 
 }  // namespace vllm_amd
 
@@ -64,6 +72,22 @@ TORCH_LIBRARY(vllm_amd, m) {
         " -> ()");
   m.def("scale_rows_cols(Tensor(a!) y, Tensor row_scales, Tensor col_scales,"
         " Tensor? bias) -> ()");
+  // Custom xGMI collectives (comms.hip). init/connect/destroy are
+  // host-side control ops and live on the catch-all dispatch below.
+  m.def("car_init(int rank, int world, int max_bytes) -> Tensor");
+  m.def("car_connect(Tensor handles) -> ()");
+  m.def("car_is_ready() -> bool");
+  m.def("car_max_bytes() -> int");
+  m.def("car_error() -> int");
+  m.def("car_destroy() -> ()");
+  m.def("car_all_reduce(Tensor(a!) t) -> ()");
+  m.def("car_all_gather(Tensor(a!) out, Tensor t) -> ()");
+  m.impl("car_init", &vllm_amd::car_init);
+  m.impl("car_connect", &vllm_amd::car_connect);
+  m.impl("car_is_ready", &vllm_amd::car_is_ready);
+  m.impl("car_max_bytes", &vllm_amd::car_max_bytes);
+  m.impl("car_error", &vllm_amd::car_error);
+  m.impl("car_destroy", &vllm_amd::car_destroy);
 }
 
 TORCH_LIBRARY_IMPL(vllm_amd, CUDA, m) {
@@ -79,4 +103,6 @@ TORCH_LIBRARY_IMPL(vllm_amd, CUDA, m) {
   m.impl("lt_linear_fp8", &vllm_amd::lt_linear_fp8);
   m.impl("dynamic_quant_fp8", &vllm_amd::dynamic_quant_fp8);
   m.impl("scale_rows_cols", &vllm_amd::scale_rows_cols);
+  m.impl("car_all_reduce", &vllm_amd::car_all_reduce);
+  m.impl("car_all_gather", &vllm_amd::car_all_gather);
 }

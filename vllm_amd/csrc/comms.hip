@@ -1,0 +1,429 @@
+// Custom xGMI collectives for TP decode (gfx950).
+//
+// Role of the reference's csrc/custom_all_reduce.cuh (one/two-shot IPC
+// all-reduce) + the flag-color scheme of csrc/quickreduce/quick_reduce.h
+// (device-side monotonic counters => hipGraph-capturable), redesigned
+// for the MI355X topology: every GPU pair on one node has a dedicated
+// xGMI link (7 links/GPU, ~153 GB/s each), so the one-shot pattern —
+// each rank reads all peers' buffers concurrently — uses all links at
+// once, and a ring (RCCL) serializes on one link. Two-shot
+// (reduce-scatter + all-gather over owned chunks) halves per-link bytes
+// for large messages.
+//
+// Protocol (per block b, per call with per-block round counter c):
+//   1. wait all peers' ack[b] >= c-2   (buffer parity c&1 is free)
+//   2. copy my input slice -> my shm data[parity]
+//   3. fence.sys; ready1[b] = c
+//   4. wait all peers' ready1[b] >= c
+//   5. reduce my slice across all ranks' data[parity] -> output
+//      (two-shot: reduce owned chunk -> my shm result[parity];
+//       fence; ready2[b] = c; wait peers ready2[b] >= c; gather chunks)
+//   6. fence.sys; ack[b] = c
+//
+// All counters live in device memory and increment inside the kernel,
+// so graph replays are self-sequencing (no host-written flag colors).
+// Blocks cover index sets {i : i % gridDim == b} so per-block flags are
+// sufficient: block b's data on every rank is exactly what block b
+// everywhere else reads (grids are identical across ranks for a given
+// message size).
+
+#include <torch/all.h>
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+
+#include <algorithm>
+#include <cstring>
+
+#include "common.h"
+
+namespace vllm_amd {
+
+namespace car {
+
+constexpr int MAX_W = 8;
+constexpr int MAX_BLOCKS = 80;
+constexpr int FLAG_STRIDE = 16;  // 16 * 8B = one 128B line per block
+constexpr int THREADS = 256;
+
+// Offsets (bytes) inside each rank's shared buffer.
+constexpr size_t READY1_OFF = 0;
+constexpr size_t READY2_OFF = READY1_OFF + MAX_BLOCKS * FLAG_STRIDE * 8;
+constexpr size_t ACK_OFF = READY2_OFF + MAX_BLOCKS * FLAG_STRIDE * 8;
+constexpr size_t DATA_OFF = 64 * 1024;
+static_assert(ACK_OFF + MAX_BLOCKS * FLAG_STRIDE * 8 <= DATA_OFF);
+
+// Per-rank shared region layout:
+//   [flags 64K][parity0: input max_bytes | result max_bytes/2]
+//   [parity1: input max_bytes | result max_bytes/2]
+struct CarDev {
+  char* base[MAX_W];            // peer-mapped shared buffers
+  unsigned long long* round;    // [MAX_BLOCKS] local (not shared)
+  unsigned long long* err;      // local error flag (spin timeout)
+  long long max_bytes;
+  int rank;
+  int world;
+};
+
+DEVINL unsigned long long* flag_ptr(char* base, size_t off, int b) {
+  return reinterpret_cast<unsigned long long*>(base + off) +
+         (size_t)b * FLAG_STRIDE;
+}
+
+DEVINL unsigned long long flag_ld(const unsigned long long* p) {
+  return __hip_atomic_load(p, __ATOMIC_ACQUIRE, __HIP_MEMORY_SCOPE_SYSTEM);
+}
+
+DEVINL void flag_st(unsigned long long* p, unsigned long long v) {
+  __hip_atomic_store(p, v, __ATOMIC_RELEASE, __HIP_MEMORY_SCOPE_SYSTEM);
+}
+
+// Spin until *p >= target; returns false on timeout (~minutes). A
+// timeout leaves wrong data but not a hung GPU; the err flag records it.
+DEVINL bool spin_ge(const unsigned long long* p, unsigned long long target) {
+  for (long long i = 0; i < (1LL << 31); ++i) {
+    if (flag_ld(p) >= target) return true;
+    __builtin_amdgcn_s_sleep(2);
+  }
+  return false;
+}
+
+// Block-wide: publish `ready` flag after all threads' writes, then wait
+// for every rank's flag to reach c. Lanes 0..w-1 of wave 0 each watch
+// one peer.
+template <size_t OFF>
+DEVINL void barrier_at(const CarDev& d, int b, unsigned long long c) {
+  __threadfence_system();
+  __syncthreads();
+  if (threadIdx.x == 0) flag_st(flag_ptr(d.base[d.rank], OFF, b), c);
+  if (threadIdx.x < (unsigned)d.world && (int)threadIdx.x != d.rank) {
+    if (!spin_ge(flag_ptr(d.base[threadIdx.x], OFF, b), c))
+      __hip_atomic_store(d.err, 1ull, __ATOMIC_RELAXED,
+                         __HIP_MEMORY_SCOPE_AGENT);
+  }
+  __syncthreads();
+}
+
+// Entry/exit bookkeeping shared by all collectives: returns the round
+// counter; waits for peers to have finished reading the parity buffer
+// this round will overwrite.
+DEVINL unsigned long long enter(const CarDev& d, int b) {
+  __shared__ unsigned long long c_sh;
+  if (threadIdx.x == 0) {
+    unsigned long long c = d.round[b] + 1;
+    d.round[b] = c;
+    c_sh = c;
+  }
+  __syncthreads();
+  unsigned long long c = c_sh;
+  if (c > 2 && threadIdx.x < (unsigned)d.world &&
+      (int)threadIdx.x != d.rank) {
+    if (!spin_ge(flag_ptr(d.base[threadIdx.x], ACK_OFF, b), c - 2))
+      __hip_atomic_store(d.err, 1ull, __ATOMIC_RELAXED,
+                         __HIP_MEMORY_SCOPE_AGENT);
+  }
+  __syncthreads();
+  return c;
+}
+
+DEVINL void exit_ack(const CarDev& d, int b, unsigned long long c) {
+  __threadfence_system();
+  __syncthreads();
+  if (threadIdx.x == 0) flag_st(flag_ptr(d.base[d.rank], ACK_OFF, b), c);
+}
+
+// 16-byte vector unit; reduction converts through fp32.
+union alignas(16) Vec16 {
+  i32x4 i;
+  short h[8];
+  float f[4];
+};
+
+template <typename Tag>
+DEVINL Vec16 vec_add(Vec16 a, Vec16 b) {
+  Vec16 r;
+#pragma unroll
+  for (int j = 0; j < 8; ++j)
+    r.h[j] = from_f32<Tag>(to_f32<Tag>(a.h[j]) + to_f32<Tag>(b.h[j]));
+  return r;
+}
+
+struct F32Tag {};
+template <>
+DEVINL Vec16 vec_add<F32Tag>(Vec16 a, Vec16 b) {
+  Vec16 r;
+#pragma unroll
+  for (int j = 0; j < 4; ++j) r.f[j] = a.f[j] + b.f[j];
+  return r;
+}
+
+DEVINL Vec16* data_ptr(char* base, long long max_bytes, int parity) {
+  return reinterpret_cast<Vec16*>(base + DATA_OFF +
+                                  (size_t)parity * (max_bytes + max_bytes / 2));
+}
+
+DEVINL Vec16* result_ptr(char* base, long long max_bytes, int parity) {
+  return reinterpret_cast<Vec16*>(base + DATA_OFF +
+                                  (size_t)parity * (max_bytes + max_bytes / 2) +
+                                  max_bytes);
+}
+
+// ---------------------------------------------------------------------------
+// One-shot: copy-in, barrier, every rank reduces its full index set by
+// reading all peers (each peer read streams over a distinct xGMI link).
+template <typename Tag>
+__global__ void one_shot_kernel(Vec16* __restrict__ io, long long nvec,
+                                CarDev d) {
+  const int b = blockIdx.x;
+  const unsigned long long c = enter(d, b);
+  const int parity = (int)(c & 1);
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  Vec16* mine = data_ptr(d.base[d.rank], d.max_bytes, parity);
+  for (long long i = b * blockDim.x + threadIdx.x; i < nvec; i += stride)
+    mine[i] = io[i];
+  barrier_at<READY1_OFF>(d, b, c);
+  // Reduce: accumulate rank-major so every lane's 7 peer streams stay
+  // in flight; start at (rank+1) so ranks offset their link usage.
+  for (long long i = b * blockDim.x + threadIdx.x; i < nvec; i += stride) {
+    Vec16 acc = mine[i];
+    for (int k = 1; k < d.world; ++k) {
+      int p = (d.rank + k) % d.world;
+      acc = vec_add<Tag>(acc,
+                         data_ptr(d.base[p], d.max_bytes, parity)[i]);
+    }
+    io[i] = acc;
+  }
+  exit_ack(d, b, c);
+}
+
+// ---------------------------------------------------------------------------
+// Two-shot: reduce-scatter (each rank reduces its owned 1/w chunk into
+// its shm result area) + all-gather of result chunks. Per-link bytes
+// drop from nvec to 2*nvec/w. Requires nvec % w == 0.
+template <typename Tag>
+__global__ void two_shot_kernel(Vec16* __restrict__ io, long long nvec,
+                                CarDev d) {
+  const int b = blockIdx.x;
+  const unsigned long long c = enter(d, b);
+  const int parity = (int)(c & 1);
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  const long long chunk = nvec / d.world;
+  Vec16* mine = data_ptr(d.base[d.rank], d.max_bytes, parity);
+  for (long long i = b * blockDim.x + threadIdx.x; i < nvec; i += stride)
+    mine[i] = io[i];
+  barrier_at<READY1_OFF>(d, b, c);
+  // Phase 1: reduce my chunk, write result to shm + my own output rows.
+  Vec16* res = result_ptr(d.base[d.rank], d.max_bytes, parity);
+  const long long lo = (long long)d.rank * chunk;
+  for (long long i = b * blockDim.x + threadIdx.x; i < chunk; i += stride) {
+    Vec16 acc = mine[lo + i];
+    for (int k = 1; k < d.world; ++k) {
+      int p = (d.rank + k) % d.world;
+      acc = vec_add<Tag>(acc,
+                         data_ptr(d.base[p], d.max_bytes, parity)[lo + i]);
+    }
+    res[i] = acc;
+    io[lo + i] = acc;
+  }
+  barrier_at<READY2_OFF>(d, b, c);
+  // Phase 2: gather peers' result chunks.
+  for (int k = 1; k < d.world; ++k) {
+    int p = (d.rank + k) % d.world;
+    Vec16* pres = result_ptr(d.base[p], d.max_bytes, parity);
+    const long long plo = (long long)p * chunk;
+    for (long long i = b * blockDim.x + threadIdx.x; i < chunk; i += stride)
+      io[plo + i] = pres[i];
+  }
+  exit_ack(d, b, c);
+}
+
+// ---------------------------------------------------------------------------
+// All-gather: rank r's nvec input becomes out rows [r*nvec, (r+1)*nvec).
+__global__ void all_gather_kernel(Vec16* __restrict__ out,
+                                  const Vec16* __restrict__ in,
+                                  long long nvec, CarDev d) {
+  const int b = blockIdx.x;
+  const unsigned long long c = enter(d, b);
+  const int parity = (int)(c & 1);
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  Vec16* mine = data_ptr(d.base[d.rank], d.max_bytes, parity);
+  for (long long i = b * blockDim.x + threadIdx.x; i < nvec; i += stride)
+    mine[i] = in[i];
+  barrier_at<READY1_OFF>(d, b, c);
+  for (int k = 0; k < d.world; ++k) {
+    int p = (d.rank + k) % d.world;
+    const Vec16* src =
+        (p == d.rank) ? mine : data_ptr(d.base[p], d.max_bytes, parity);
+    Vec16* dst = out + (long long)p * nvec;
+    for (long long i = b * blockDim.x + threadIdx.x; i < nvec; i += stride)
+      dst[i] = src[i];
+  }
+  exit_ack(d, b, c);
+}
+
+// ---------------------------------------------------------------------------
+// Host-side state (one TP group per process).
+struct CarState {
+  bool ready = false;
+  int rank = -1;
+  int world = 0;
+  long long max_bytes = 0;
+  char* self_buf = nullptr;
+  char* peer[MAX_W] = {nullptr};
+  bool opened[MAX_W] = {false};
+  unsigned long long* local = nullptr;  // [MAX_BLOCKS round][1 err]
+  CarDev dev{};
+};
+
+static CarState g_car;
+
+static int pick_blocks(long long nvec) {
+  long long per_block = 4096;  // vec16 units; ~64KB per block minimum
+  int nb = (int)std::min<long long>((nvec + per_block - 1) / per_block,
+                                    MAX_BLOCKS);
+  return std::max(nb, 1);
+}
+
+}  // namespace car
+
+torch::Tensor car_init(int64_t rank, int64_t world, int64_t max_bytes) {
+  using namespace car;
+  TORCH_CHECK(!g_car.ready && g_car.self_buf == nullptr,
+              "custom all-reduce already initialized");
+  TORCH_CHECK(world >= 2 && world <= MAX_W, "world must be in [2,8]");
+  TORCH_CHECK(rank >= 0 && rank < world);
+  TORCH_CHECK(max_bytes % 4096 == 0);
+  size_t total = DATA_OFF + 2 * ((size_t)max_bytes + (size_t)max_bytes / 2);
+  hipError_t e = hipMalloc((void**)&g_car.self_buf, total);
+  TORCH_CHECK(e == hipSuccess, "car_init hipMalloc: ", hipGetErrorString(e));
+  e = hipMemset(g_car.self_buf, 0, DATA_OFF);
+  TORCH_CHECK(e == hipSuccess);
+  e = hipMalloc((void**)&g_car.local, (MAX_BLOCKS + 1) * 8);
+  TORCH_CHECK(e == hipSuccess);
+  e = hipMemset(g_car.local, 0, (MAX_BLOCKS + 1) * 8);
+  TORCH_CHECK(e == hipSuccess);
+  g_car.rank = (int)rank;
+  g_car.world = (int)world;
+  g_car.max_bytes = max_bytes;
+
+  hipIpcMemHandle_t handle;
+  e = hipIpcGetMemHandle(&handle, g_car.self_buf);
+  TORCH_CHECK(e == hipSuccess, "hipIpcGetMemHandle: ", hipGetErrorString(e));
+  auto out = torch::empty({(int64_t)sizeof(handle)},
+                          torch::dtype(torch::kUInt8));
+  memcpy(out.data_ptr(), &handle, sizeof(handle));
+  return out;
+}
+
+void car_connect(torch::Tensor handles) {
+  using namespace car;
+  TORCH_CHECK(g_car.self_buf != nullptr, "car_init first");
+  TORCH_CHECK(handles.dim() == 2 && handles.size(0) == g_car.world &&
+              handles.size(1) == (int64_t)sizeof(hipIpcMemHandle_t));
+  auto h = handles.contiguous();
+  const uint8_t* p = h.data_ptr<uint8_t>();
+  for (int r = 0; r < g_car.world; ++r) {
+    if (r == g_car.rank) {
+      g_car.peer[r] = g_car.self_buf;
+      continue;
+    }
+    hipIpcMemHandle_t handle;
+    memcpy(&handle, p + (size_t)r * sizeof(handle), sizeof(handle));
+    void* ptr = nullptr;
+    hipError_t e = hipIpcOpenMemHandle(&ptr, handle,
+                                       hipIpcMemLazyEnablePeerAccess);
+    TORCH_CHECK(e == hipSuccess, "hipIpcOpenMemHandle rank ", r, ": ",
+                hipGetErrorString(e));
+    g_car.peer[r] = (char*)ptr;
+    g_car.opened[r] = true;
+  }
+  for (int r = 0; r < g_car.world; ++r) g_car.dev.base[r] = g_car.peer[r];
+  g_car.dev.round = g_car.local;
+  g_car.dev.err = g_car.local + MAX_BLOCKS;
+  g_car.dev.max_bytes = g_car.max_bytes;
+  g_car.dev.rank = g_car.rank;
+  g_car.dev.world = g_car.world;
+  g_car.ready = true;
+}
+
+bool car_is_ready() { return car::g_car.ready; }
+
+int64_t car_max_bytes() { return car::g_car.max_bytes; }
+
+int64_t car_error() {
+  using namespace car;
+  if (!g_car.ready) return 0;
+  unsigned long long v = 0;
+  hipMemcpy(&v, g_car.local + MAX_BLOCKS, 8, hipMemcpyDeviceToHost);
+  return (int64_t)v;
+}
+
+void car_destroy() {
+  using namespace car;
+  for (int r = 0; r < MAX_W; ++r) {
+    if (g_car.opened[r] && g_car.peer[r]) hipIpcCloseMemHandle(g_car.peer[r]);
+    g_car.peer[r] = nullptr;
+    g_car.opened[r] = false;
+  }
+  if (g_car.self_buf) hipFree(g_car.self_buf);
+  if (g_car.local) hipFree(g_car.local);
+  g_car = CarState{};
+}
+
+void car_all_reduce(torch::Tensor t) {
+  using namespace car;
+  TORCH_CHECK(g_car.ready, "custom all-reduce not initialized");
+  TORCH_CHECK(t.is_cuda() && t.is_contiguous());
+  const long long nbytes = (long long)t.numel() * t.element_size();
+  TORCH_CHECK(nbytes % 16 == 0 && nbytes <= g_car.max_bytes,
+              "unsupported all-reduce size ", nbytes);
+  const long long nvec = nbytes / 16;
+  auto stream = c10::hip::getCurrentHIPStream();
+  const int nb = pick_blocks(nvec);
+  const bool two_shot = nbytes > (512 << 10) && nvec % g_car.world == 0;
+  Vec16* io = reinterpret_cast<Vec16*>(t.data_ptr());
+  auto launch = [&](auto tag) {
+    using Tag = decltype(tag);
+    if (two_shot)
+      hipLaunchKernelGGL((two_shot_kernel<Tag>), dim3(nb), dim3(THREADS), 0,
+                         stream.stream(), io, nvec, g_car.dev);
+    else
+      hipLaunchKernelGGL((one_shot_kernel<Tag>), dim3(nb), dim3(THREADS), 0,
+                         stream.stream(), io, nvec, g_car.dev);
+  };
+  switch (t.scalar_type()) {
+    case torch::kBFloat16:
+      launch(BF16Tag{});
+      break;
+    case torch::kHalf:
+      launch(FP16Tag{});
+      break;
+    case torch::kFloat:
+      launch(F32Tag{});
+      break;
+    default:
+      TORCH_CHECK(false, "car_all_reduce: unsupported dtype");
+  }
+  HIP_CHECK_KERNEL();
+}
+
+void car_all_gather(torch::Tensor out, torch::Tensor t) {
+  using namespace car;
+  TORCH_CHECK(g_car.ready, "custom all-reduce not initialized");
+  TORCH_CHECK(t.is_cuda() && t.is_contiguous() && out.is_contiguous());
+  const long long nbytes = (long long)t.numel() * t.element_size();
+  TORCH_CHECK(nbytes % 16 == 0 && nbytes <= g_car.max_bytes,
+              "unsupported all-gather size ", nbytes);
+  TORCH_CHECK(out.numel() == t.numel() * g_car.world &&
+              out.scalar_type() == t.scalar_type());
+  const long long nvec = nbytes / 16;
+  auto stream = c10::hip::getCurrentHIPStream();
+  const int nb = pick_blocks(nvec);
+  hipLaunchKernelGGL(all_gather_kernel, dim3(nb), dim3(THREADS), 0,
+                     stream.stream(),
+                     reinterpret_cast<Vec16*>(out.data_ptr()),
+                     reinterpret_cast<const Vec16*>(t.data_ptr()), nvec,
+                     g_car.dev);
+  HIP_CHECK_KERNEL();
+}
+
+}  // namespace vllm_amd

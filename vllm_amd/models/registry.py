@@ -45,16 +45,30 @@ def initialize_dummy_weights(model: torch.nn.Module, seed: int = 0) -> None:
     a pipeline-parallel stage holding layers [lo, hi) materializes
     exactly the weights a single-process model would — pp1 and pp2 runs
     are bit-comparable."""
+    import math
     import zlib
 
     for name, param in model.named_parameters():
         gen = torch.Generator()
         gen.manual_seed(seed ^ zlib.crc32(name.encode()))
-        # Uniform in [-1e-3, 1e-3]: keeps bf16 activations finite through
-        # 80-layer stacks while exercising real arithmetic.
         cpu_val = torch.rand(param.shape, generator=gen,
-                             dtype=torch.float32)
-        param.data.copy_(((cpu_val * 2.0 - 1.0) * 1e-3).to(param.dtype))
+                             dtype=torch.float32) * 2.0 - 1.0
+        # Scales chosen so the network is a FUNCTIONING transformer, not
+        # a constant map: norm gains at ~1 and matrices at 1/sqrt(fan_in)
+        # keep activations O(1) through deep stacks (RMSNorm re-centers
+        # every layer), so outputs genuinely depend on inputs — an
+        # all-tiny init drives fp32 logits to exactly-uniform and lets
+        # equality-style e2e tests pass vacuously.
+        if param.dim() == 1 and "norm" in name.lower():
+            val = 1.0 + 0.05 * cpu_val
+        elif param.dim() == 1:  # biases
+            val = 1e-3 * cpu_val
+        else:
+            fan_in = param.shape[-1] if param.dim() >= 2 else 1
+            if param.dim() > 2 and "patch" in name.lower():
+                fan_in = int(torch.tensor(param.shape[1:]).prod())
+            val = cpu_val / math.sqrt(fan_in)
+        param.data.copy_(val.to(param.dtype))
 
 
 def load_model(config: ModelConfig, device: torch.device) -> torch.nn.Module:

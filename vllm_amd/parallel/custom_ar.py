@@ -1,0 +1,131 @@
+"""Custom xGMI all-reduce / all-gather wrapper (csrc/comms.hip).
+
+Role of the reference's vllm/distributed/device_communicators/
+custom_all_reduce.py:74 (size-gated IPC all-reduce in front of NCCL),
+MI355X-shaped: one process per GPU on one node, buffers exchanged as
+hipIpc handles over the TP group's gloo CPU group, device-side flag
+counters so hipGraph capture/replay of TP decode steps needs no host
+involvement.
+
+Dispatch policy (GroupCoordinator.all_reduce):
+  <= max_bytes and 16B-aligned  -> custom kernel (one-shot <=512KB,
+                                    two-shot above — chosen in C++)
+  otherwise                     -> torch.distributed (RCCL)
+"""
+
+from __future__ import annotations
+
+import logging
+import os
+
+import torch
+import torch.distributed as dist
+
+logger = logging.getLogger(__name__)
+
+# Covers decode all-reduces ([tokens, hidden] bf16; 1024x8192x2 = 16 MiB)
+# and the logits all-gather chunk ([tokens, vocab/tp]; up to ~126 MiB at
+# llama-3 vocab TP=2, batch 1024) so the whole TP decode step is
+# graph-capturable. Buffer cost = 3x this per GPU — noise on 288 GB.
+DEFAULT_MAX_BYTES = int(
+    os.environ.get("VLLM_AMD_CAR_MAX_BYTES", 160 * 1024 * 1024))
+
+
+class CustomCollectives:
+    """Per-process handle to the comms.hip state (one TP group)."""
+
+    def __init__(self, rank_in_group: int, world_size: int,
+                 max_bytes: int = DEFAULT_MAX_BYTES):
+        from vllm_amd import ops  # loads _C.so
+
+        assert ops is not None
+        _C = torch.ops.vllm_amd
+        self._C = _C
+        self.world_size = world_size
+        self.max_bytes = max_bytes
+        self.handle = _C.car_init(rank_in_group, world_size, max_bytes)
+        self.disabled = False
+
+    def connect(self, cpu_group) -> None:
+        all_handles = [torch.empty_like(self.handle)
+                       for _ in range(self.world_size)]
+        dist.all_gather(all_handles, self.handle, group=cpu_group)
+        self._C.car_connect(torch.stack(all_handles))
+
+    def should_use(self, t: torch.Tensor) -> bool:
+        if self.disabled or not t.is_cuda or not t.is_contiguous():
+            return False
+        if t.dtype not in (torch.bfloat16, torch.float16, torch.float32):
+            return False
+        nbytes = t.numel() * t.element_size()
+        return nbytes % 16 == 0 and nbytes <= self.max_bytes
+
+    def all_reduce(self, t: torch.Tensor) -> torch.Tensor:
+        self._C.car_all_reduce(t)
+        return t
+
+    def can_gather(self, t: torch.Tensor) -> bool:
+        if self.disabled or not t.is_cuda:
+            return False
+        if t.dtype not in (torch.bfloat16, torch.float16, torch.float32):
+            return False
+        nbytes = t.numel() * t.element_size()
+        return nbytes % 16 == 0 and nbytes <= self.max_bytes
+
+    def all_gather_flat(self, t: torch.Tensor) -> torch.Tensor:
+        """Returns [world, *t.shape]."""
+        t = t.contiguous()
+        out = torch.empty((self.world_size,) + tuple(t.shape),
+                          dtype=t.dtype, device=t.device)
+        self._C.car_all_gather(out.view(-1), t.view(-1))
+        return out
+
+    def error(self) -> int:
+        return int(self._C.car_error())
+
+    def destroy(self) -> None:
+        if not self.disabled:
+            self._C.car_destroy()
+            self.disabled = True
+
+
+def try_init_custom_collectives(rank_in_group: int, world_size: int,
+                                cpu_group) -> "CustomCollectives | None":
+    """Best-effort init; falls back to None (RCCL-only) on any failure.
+    All ranks must agree — the boolean is all-reduced over the CPU group
+    so no rank is left spinning on a peer that failed."""
+    ok = True
+    inst = None
+    if os.environ.get("VLLM_AMD_DISABLE_CUSTOM_AR", "0") == "1":
+        ok = False
+    if not torch.cuda.is_available() or not (2 <= world_size <= 8):
+        ok = False
+    if ok:
+        try:
+            inst = CustomCollectives(rank_in_group, world_size)
+        except Exception as e:  # noqa: BLE001
+            logger.warning("custom all-reduce init failed (%s); "
+                           "falling back to RCCL", e)
+            ok = False
+    # Agree before the handle exchange so no rank blocks in all_gather
+    # against a peer whose car_init failed.
+    flag = torch.tensor([1 if ok else 0], dtype=torch.int32)
+    dist.all_reduce(flag, op=dist.ReduceOp.MIN, group=cpu_group)
+    if int(flag.item()) == 0:
+        if inst is not None:
+            inst.destroy()
+        return None
+    try:
+        inst.connect(cpu_group)
+    except Exception as e:  # noqa: BLE001
+        # Handle-open failure can be rank-local (IPC limits); every rank
+        # must agree again or peers would spin forever in the kernel.
+        logger.warning("custom all-reduce connect failed (%s)", e)
+        flag = torch.tensor([0], dtype=torch.int32)
+    else:
+        flag = torch.tensor([1], dtype=torch.int32)
+    dist.all_reduce(flag, op=dist.ReduceOp.MIN, group=cpu_group)
+    if int(flag.item()) == 0:
+        inst.destroy()
+        return None
+    return inst

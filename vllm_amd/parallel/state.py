@@ -30,6 +30,9 @@ class GroupCoordinator:
     rank_in_group: int
     device_group: Optional[dist.ProcessGroup] = None
     cpu_group: Optional[dist.ProcessGroup] = None
+    # Custom xGMI collectives (parallel/custom_ar.py); size-gated in
+    # front of RCCL, hipGraph-capturable. TP group only.
+    comms: Optional[object] = None
 
     @property
     def is_first_rank(self) -> bool:
@@ -38,6 +41,8 @@ class GroupCoordinator:
     def all_reduce(self, t: torch.Tensor) -> torch.Tensor:
         if self.world_size == 1:
             return t
+        if self.comms is not None and self.comms.should_use(t):
+            return self.comms.all_reduce(t)
         dist.all_reduce(t, group=self.device_group)
         return t
 
@@ -46,14 +51,18 @@ class GroupCoordinator:
             return t
         if dim < 0:
             dim += t.dim()
-        # Gather flat on dim 0 (collective layout), then view as
-        # [world, *shape] to concat on `dim`.
-        out = torch.empty(
-            (self.world_size * t.shape[0],) + tuple(t.shape[1:]),
-            dtype=t.dtype, device=t.device,
-        )
-        dist.all_gather_into_tensor(out, t.contiguous(), group=self.device_group)
-        out = out.view((self.world_size,) + tuple(t.shape))
+        if self.comms is not None and self.comms.can_gather(t):
+            out = self.comms.all_gather_flat(t)
+        else:
+            # Gather flat on dim 0 (collective layout), then view as
+            # [world, *shape] to concat on `dim`.
+            out = torch.empty(
+                (self.world_size * t.shape[0],) + tuple(t.shape[1:]),
+                dtype=t.dtype, device=t.device,
+            )
+            dist.all_gather_into_tensor(out, t.contiguous(),
+                                        group=self.device_group)
+            out = out.view((self.world_size,) + tuple(t.shape))
         if dim == 0:
             return out.reshape(-1, *t.shape[1:])
         pieces = out.unbind(0)
@@ -248,13 +257,28 @@ def init_distributed(
         if rank in ranks:
             tp_group = g
             my_tp_ranks = ranks
+    # Per-TP-group gloo group for the IPC handle exchange and object
+    # collectives (the global gloo group would collide across replicas).
+    tp_cpu_group = None
+    if backend != "gloo" and tensor_parallel_size > 1:
+        for start in range(0, world_size, tensor_parallel_size):
+            ranks = list(range(start, start + tensor_parallel_size))
+            g = dist.new_group(ranks, backend="gloo")
+            if rank in ranks:
+                tp_cpu_group = g
     _TP = GroupCoordinator(
         rank=rank,
         world_size=tensor_parallel_size,
         rank_in_group=my_tp_ranks.index(rank),
         device_group=tp_group,
-        cpu_group=cpu_group,
+        cpu_group=tp_cpu_group or cpu_group,
     )
+    if (tensor_parallel_size > 1 and backend == "nccl"
+            and torch.cuda.is_available()):
+        from vllm_amd.parallel.custom_ar import try_init_custom_collectives
+        _TP.comms = try_init_custom_collectives(
+            _TP.rank_in_group, tensor_parallel_size,
+            tp_cpu_group or cpu_group)
     if pipeline_parallel_size > 1:
         pp_group = None
         my_pp_ranks = None
@@ -309,6 +333,8 @@ def init_distributed(
 
 def destroy_distributed() -> None:
     global _TP, _EP, _WORLD, _PP, _REPLICA
+    if _TP is not None and _TP.comms is not None:
+        _TP.comms.destroy()
     _TP = _EP = _WORLD = _PP = _REPLICA = None
     if dist.is_initialized():
         dist.destroy_process_group()

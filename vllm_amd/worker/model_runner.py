@@ -362,13 +362,17 @@ class ModelRunner:
             )
             for _ in range(self._num_local_layers())
         ]
-        # MoE models: the segmented-GEMM fused_moe syncs for expert
-        # counts, which is not graph-capturable (until the grouped-GEMM
-        # kernel lands).
+        # TP decode graphs require the custom xGMI collectives: RCCL
+        # through torch.distributed is not hipGraph-capturable, and the
+        # per-layer all-reduce sits inside the captured region.
+        from vllm_amd.parallel.state import get_tp_group
+        tp_graph_ok = (get_tp_world_size() == 1
+                       or get_tp_group().comms is not None)
         if (self.device.type == "cuda"
                 and not self.config.model_config.enforce_eager
                 and not self.config.model_config.lora_modules
                 and self.pp_size == 1
+                and tp_graph_ok
                 and not self.spec.is_moe and not self.spec.is_mla):
             self.graph_runner = DecodeGraphRunner(self)
 
