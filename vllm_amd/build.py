@@ -23,6 +23,7 @@ OUT_SO = PKG_DIR / "_C.so"
 SOURCES = [
     CSRC / "elementwise.hip",
     CSRC / "comms.hip",
+    CSRC / "moe.hip",
     CSRC / "quant_fp8.hip",
     CSRC / "attention_decode.hip",
     CSRC / "attention_prefill.hip",

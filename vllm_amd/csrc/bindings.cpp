@@ -38,6 +38,15 @@ void dynamic_quant_fp8(torch::Tensor out, torch::Tensor scales,
 void scale_rows_cols(torch::Tensor y, torch::Tensor row_scales,
                      torch::Tensor col_scales,
                      c10::optional<torch::Tensor> bias);
+void moe_align(torch::Tensor topk_ids, int64_t num_experts,
+               torch::Tensor sorted, torch::Tensor expert_tiles,
+               torch::Tensor inv_perm, torch::Tensor counts,
+               torch::Tensor fill, torch::Tensor off);
+void moe_gemm(torch::Tensor a, torch::Tensor b, torch::Tensor c,
+              torch::Tensor sorted, torch::Tensor expert_tiles,
+              int64_t topk_div, int64_t total_flat);
+void moe_combine(torch::Tensor out, torch::Tensor y,
+                 torch::Tensor topk_weights, torch::Tensor inv_perm);
 torch::Tensor car_init(int64_t rank, int64_t world, int64_t max_bytes);
 void car_connect(torch::Tensor handles);
 bool car_is_ready();
@@ -72,6 +81,13 @@ TORCH_LIBRARY(vllm_amd, m) {
         " -> ()");
   m.def("scale_rows_cols(Tensor(a!) y, Tensor row_scales, Tensor col_scales,"
         " Tensor? bias) -> ()");
+  m.def("moe_align(Tensor topk_ids, int num_experts, Tensor(a!) sorted, "
+        "Tensor(b!) expert_tiles, Tensor(c!) inv_perm, Tensor(d!) counts, "
+        "Tensor(e!) fill, Tensor(f!) off) -> ()");
+  m.def("moe_gemm(Tensor a, Tensor b, Tensor(a!) c, Tensor sorted, "
+        "Tensor expert_tiles, int topk_div, int total_flat) -> ()");
+  m.def("moe_combine(Tensor(a!) out, Tensor y, Tensor topk_weights, "
+        "Tensor inv_perm) -> ()");
   // Custom xGMI collectives (comms.hip). init/connect/destroy are
   // host-side control ops and live on the catch-all dispatch below.
   m.def("car_init(int rank, int world, int max_bytes) -> Tensor");
@@ -105,4 +121,7 @@ TORCH_LIBRARY_IMPL(vllm_amd, CUDA, m) {
   m.impl("scale_rows_cols", &vllm_amd::scale_rows_cols);
   m.impl("car_all_reduce", &vllm_amd::car_all_reduce);
   m.impl("car_all_gather", &vllm_amd::car_all_gather);
+  m.impl("moe_align", &vllm_amd::moe_align);
+  m.impl("moe_gemm", &vllm_amd::moe_gemm);
+  m.impl("moe_combine", &vllm_amd::moe_combine);
 }

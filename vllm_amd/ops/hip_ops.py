@@ -180,14 +180,63 @@ from vllm_amd.ops._torch_ref import (  # noqa: E402,F401
 )
 
 
-def fused_moe(hidden, w13, w2, topk_weights, topk_ids, activation="silu"):
-    """MoE expert MLP on GPU: tokens sorted by expert, one tuned hipBLASLt
-    GEMM pair per non-empty expert, weighted scatter-add combine.
+_MOE_BM = 32  # must match moe::BM in csrc/moe.hip
 
-    hidden: [T, H]; w13: [E, 2I, H]; w2: [E, H, I]. One D2H sync per call
-    for the segment sizes (a dedicated grouped-GEMM MFMA kernel replaces
-    this; role of the reference's fused_moe.py:299 Triton grouped GEMM).
+
+def _moe_hip_ok(hidden, w13, w2, activation):
+    if activation != "silu":
+        return False
+    if hidden.dtype not in (torch.bfloat16, torch.float16):
+        return False
+    H = hidden.shape[1]
+    twoI = w13.shape[1]
+    # GEMM1: K=H%64, N=2I%128; GEMM2: K=I%64, N=H%128.
+    return H % 128 == 0 and twoI % 128 == 0 and (twoI // 2) % 64 == 0
+
+
+def fused_moe_hip(hidden, w13, w2, topk_weights, topk_ids):
+    """Grouped-GEMM MFMA MoE (csrc/moe.hip): on-device token->expert
+    sort into BM tiles, two MFMA GEMMs streaming each expert's weights,
+    deterministic weighted combine. No host sync — hipGraph-capturable
+    (role of the reference's fused_moe.py:299 + moe_align_sum_kernels.cu).
     """
+    T, H = hidden.shape
+    E, twoI, _ = w13.shape
+    I = twoI // 2
+    k = topk_ids.shape[1]
+    total = T * k
+    dev = hidden.device
+    em_max = -(-(total + E * (_MOE_BM - 1)) // _MOE_BM) * _MOE_BM
+    ids32 = topk_ids.to(torch.int32).contiguous()
+    sorted_ids = torch.empty(em_max, dtype=torch.int32, device=dev)
+    expert_tiles = torch.empty(em_max // _MOE_BM, dtype=torch.int32,
+                               device=dev)
+    inv_perm = torch.empty(total, dtype=torch.int32, device=dev)
+    scratch = torch.empty(3 * E, dtype=torch.int32, device=dev)
+    counts, fill, off = scratch[:E], scratch[E:2 * E], scratch[2 * E:]
+    _C.moe_align(ids32, E, sorted_ids, expert_tiles, inv_perm, counts,
+                 fill, off)
+    y = torch.empty(em_max, twoI, dtype=hidden.dtype, device=dev)
+    _C.moe_gemm(hidden, w13, y, sorted_ids, expert_tiles, k, total)
+    act = silu_and_mul(y)
+    y2 = torch.empty(em_max, H, dtype=hidden.dtype, device=dev)
+    _C.moe_gemm(act, w2, y2, sorted_ids, expert_tiles, 0, total)
+    out = torch.empty(T, H, dtype=hidden.dtype, device=dev)
+    _C.moe_combine(out, y2, topk_weights.float().contiguous(), inv_perm)
+    return out
+
+
+def fused_moe(hidden, w13, w2, topk_weights, topk_ids, activation="silu"):
+    """MoE expert MLP on GPU. Primary path: the grouped-GEMM MFMA kernel
+    (fused_moe_hip, no host sync). Fallback for unsupported shapes or
+    activations: tokens sorted by expert, one tuned hipBLASLt GEMM pair
+    per non-empty expert (one D2H sync per call), weighted scatter-add
+    combine.
+
+    hidden: [T, H]; w13: [E, 2I, H]; w2: [E, H, I].
+    """
+    if _moe_hip_ok(hidden, w13, w2, activation):
+        return fused_moe_hip(hidden, w13, w2, topk_weights, topk_ids)
     T, H = hidden.shape
     E = w13.shape[0]
     k = topk_ids.shape[1]

@@ -368,12 +368,27 @@ class ModelRunner:
         from vllm_amd.parallel.state import get_tp_group
         tp_graph_ok = (get_tp_world_size() == 1
                        or get_tp_group().comms is not None)
+        # MoE models are graph-capturable when the grouped-GEMM MFMA
+        # path applies (csrc/moe.hip — no host sync); the segmented
+        # hipBLASLt fallback syncs for expert counts and is not.
+        moe_graph_ok = not self.spec.is_moe
+        if self.spec.is_moe and self.device.type == "cuda":
+            from vllm_amd.ops import get_backend
+            i_shard = self.spec.moe_intermediate_size
+            if not self.config.parallel_config.enable_expert_parallel:
+                i_shard //= max(1, get_tp_world_size())
+            probe = torch.zeros(1, self.spec.hidden_size, dtype=self.dtype)
+            w13 = torch.zeros(1, 2 * i_shard, 1)
+            backend = get_backend(self.device)
+            moe_graph_ok = bool(
+                getattr(backend, "_moe_hip_ok", lambda *a: False)(
+                    probe, w13, None, "silu"))
         if (self.device.type == "cuda"
                 and not self.config.model_config.enforce_eager
                 and not self.config.model_config.lora_modules
                 and self.pp_size == 1
-                and tp_graph_ok
-                and not self.spec.is_moe and not self.spec.is_mla):
+                and tp_graph_ok and moe_graph_ok
+                and not self.spec.is_mla):
             self.graph_runner = DecodeGraphRunner(self)
 
     def allocate_host_kv_pool(self, num_host_blocks: int) -> None:
