@@ -438,3 +438,64 @@ def test_moe_hip_graph_capturable():
         expect = ref.fused_moe((hidden * scale).float(), w13.float(),
                                w2.float(), tw, ti)
         assert_close(out, expect, atol=8e-2, rtol=8e-2)
+
+
+@pytest.mark.parametrize("T,Hq,ctx", [
+    (1, 128, 33),       # single seq, TP=1 head count, partial subtile
+    (4, 128, 300),      # multi-seq single-partition
+    (3, 16, 2500),      # TP=8 head count, multi-partition (>1024)
+    (2, 32, 1024),      # exact partition boundary
+])
+def test_mla_decode_kernel(T, Hq, ctx):
+    """MFMA MLA decode (csrc/mla.hip) vs the fp32 torch composition over
+    the compressed latent cache (DeepSeek geometry: lora 512 + rope 64)."""
+    hip = _hip()
+    lora, rope, bs = 512, 64, 16
+    nb_per = (ctx + bs - 1) // bs
+    total_blocks = T * nb_per + 1
+    kv_cache = (torch.randn(total_blocks, bs, lora + rope,
+                            dtype=torch.bfloat16, device="cuda") * 0.5)
+    block_table = torch.zeros(T, nb_per, dtype=torch.int32, device="cuda")
+    perm = torch.randperm(T * nb_per, device="cuda").int() + 1
+    block_table[:] = perm.view(T, nb_per)
+    seq_lens = torch.full((T,), ctx, dtype=torch.int32, device="cuda")
+    seq_lens[0] = max(1, ctx - 7)  # ragged lengths
+    q_nope = torch.randn(T, Hq, lora, dtype=torch.bfloat16,
+                         device="cuda") * 0.3
+    q_pe = torch.randn(T, Hq, rope, dtype=torch.bfloat16, device="cuda") * 0.3
+    scale = (128 + 64) ** -0.5
+    out = hip.mla_decode(q_nope, q_pe, kv_cache, block_table, seq_lens,
+                         scale, ctx)
+    qsl = torch.arange(T + 1, dtype=torch.int32, device="cuda")
+    expect = ref.mla_attention(q_nope.float(), q_pe.float(),
+                               kv_cache.float(), block_table, qsl,
+                               seq_lens, scale)
+    assert_close(out, expect, atol=4e-2, rtol=4e-2)
+
+
+def test_mla_unified_dispatch_mixed_batch():
+    """ops.mla_attention routes decode rows to the kernel and prefill
+    rows to the torch path; both must agree with the all-torch result."""
+    hip = _hip()
+    lora, rope, bs, Hq = 512, 64, 16, 32
+    T_dec, ql_pre, ctx_pre = 3, 5, 40
+    ctx_dec = 77
+    nb = 16
+    kv_cache = torch.randn(64, bs, lora + rope, dtype=torch.bfloat16,
+                           device="cuda") * 0.5
+    bt = torch.arange(4 * nb, dtype=torch.int32, device="cuda").view(4, nb)
+    seq_lens = torch.tensor([ctx_dec, ctx_dec, ctx_dec, ctx_pre],
+                            dtype=torch.int32, device="cuda")
+    qsl = torch.tensor([0, 1, 2, 3, 3 + ql_pre], dtype=torch.int32,
+                       device="cuda")
+    Ttot = 3 + ql_pre
+    q_nope = torch.randn(Ttot, Hq, lora, dtype=torch.bfloat16,
+                         device="cuda") * 0.3
+    q_pe = torch.randn(Ttot, Hq, rope, dtype=torch.bfloat16,
+                       device="cuda") * 0.3
+    scale = 0.08
+    out = hip.mla_attention(q_nope, q_pe, kv_cache, bt, qsl, seq_lens,
+                            scale, num_decodes=T_dec, max_seq_len=ctx_dec)
+    expect = ref.mla_attention(q_nope.float(), q_pe.float(),
+                               kv_cache.float(), bt, qsl, seq_lens, scale)
+    assert_close(out, expect, atol=4e-2, rtol=4e-2)

@@ -24,6 +24,7 @@ SOURCES = [
     CSRC / "elementwise.hip",
     CSRC / "comms.hip",
     CSRC / "moe.hip",
+    CSRC / "mla.hip",
     CSRC / "quant_fp8.hip",
     CSRC / "attention_decode.hip",
     CSRC / "attention_prefill.hip",

@@ -47,6 +47,10 @@ void moe_gemm(torch::Tensor a, torch::Tensor b, torch::Tensor c,
               int64_t topk_div, int64_t total_flat);
 void moe_combine(torch::Tensor out, torch::Tensor y,
                  torch::Tensor topk_weights, torch::Tensor inv_perm);
+void mla_decode(torch::Tensor out, torch::Tensor q_nope, torch::Tensor q_pe,
+                torch::Tensor kv_cache, torch::Tensor block_table,
+                torch::Tensor seq_lens, double scale, int64_t max_seq_len,
+                torch::Tensor tmp_out, torch::Tensor tmp_lse);
 torch::Tensor car_init(int64_t rank, int64_t world, int64_t max_bytes);
 void car_connect(torch::Tensor handles);
 bool car_is_ready();
@@ -88,6 +92,10 @@ TORCH_LIBRARY(vllm_amd, m) {
         "Tensor expert_tiles, int topk_div, int total_flat) -> ()");
   m.def("moe_combine(Tensor(a!) out, Tensor y, Tensor topk_weights, "
         "Tensor inv_perm) -> ()");
+  m.def("mla_decode(Tensor(a!) out, Tensor q_nope, Tensor q_pe, "
+        "Tensor kv_cache, Tensor block_table, Tensor seq_lens, "
+        "float scale, int max_seq_len, Tensor(b!) tmp_out, "
+        "Tensor(c!) tmp_lse) -> ()");
   // Custom xGMI collectives (comms.hip). init/connect/destroy are
   // host-side control ops and live on the catch-all dispatch below.
   m.def("car_init(int rank, int world, int max_bytes) -> Tensor");
@@ -124,4 +132,5 @@ TORCH_LIBRARY_IMPL(vllm_amd, CUDA, m) {
   m.impl("moe_align", &vllm_amd::moe_align);
   m.impl("moe_gemm", &vllm_amd::moe_gemm);
   m.impl("moe_combine", &vllm_amd::moe_combine);
+  m.impl("mla_decode", &vllm_amd::mla_decode);
 }

@@ -136,6 +136,7 @@ class DeepseekMLAAttention(nn.Module):
             o_c = ops.mla_attention(
                 q_absorbed, q_pe, kv_cache, meta.block_table,
                 meta.query_start_loc, meta.seq_lens, self.scale,
+                num_decodes=meta.num_decodes, max_seq_len=meta.max_seq_len,
             )
         # Un-absorb W_UV: [T, h, v].
         out = torch.einsum("thl,hlv->thv", o_c.float(),

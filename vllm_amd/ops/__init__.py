@@ -133,8 +133,14 @@ def concat_and_cache_mla(c_kv, k_pe, kv_cache, slot_mapping):
 
 
 def mla_attention(q_nope, q_pe, kv_cache, block_table, query_start_loc,
-                  seq_lens, scale):
-    return get_backend(q_nope.device).mla_attention(
+                  seq_lens, scale, num_decodes=0, max_seq_len=0):
+    backend = get_backend(q_nope.device)
+    if backend is not _torch_ref:
+        return backend.mla_attention(
+            q_nope, q_pe, kv_cache, block_table, query_start_loc, seq_lens,
+            scale, num_decodes=num_decodes, max_seq_len=max_seq_len,
+        )
+    return backend.mla_attention(
         q_nope, q_pe, kv_cache, block_table, query_start_loc, seq_lens, scale
     )
 

@@ -169,15 +169,64 @@ def _warn_prefill_fallback(head_dim: int) -> None:
             "HIP kernel)", head_dim)
 
 
-# MoE routing (small [T, E] tensors — torch ops are fine here) and the
-# MLA compressed-attention path (torch composition today; dedicated HIP
-# MLA decode kernel is the known next step — see SURVEY §9).
+# MoE routing (small [T, E] tensors — torch ops are fine here).
+from vllm_amd.ops import _torch_ref  # noqa: E402
 from vllm_amd.ops._torch_ref import (  # noqa: E402,F401
     concat_and_cache_mla,
     grouped_topk,
-    mla_attention,
     topk_softmax,
 )
+
+_MLA_PART = 1024  # must match mla::PART in csrc/mla.hip
+
+
+def mla_decode(q_nope, q_pe, kv_cache, block_table, seq_lens, scale,
+               max_seq_len):
+    """Absorbed-MQA MLA decode on the MFMA kernel (csrc/mla.hip);
+    kv_lora_rank 512 + rope 64 latent cache."""
+    T, Hq, lora = q_nope.shape
+    out = torch.empty(T, Hq, lora, dtype=q_nope.dtype, device=q_nope.device)
+    if max_seq_len <= 0:
+        max_seq_len = int(seq_lens.max().item())
+    parts = max(1, -(-max_seq_len // _MLA_PART))
+    if parts > 1:
+        tmp_out = torch.empty(T, Hq, parts, lora, dtype=torch.float32,
+                              device=q_nope.device)
+        tmp_lse = torch.empty(T, Hq, parts, 2, dtype=torch.float32,
+                              device=q_nope.device)
+    else:
+        tmp_out = q_nope.new_empty(0, dtype=torch.float32)
+        tmp_lse = tmp_out
+    _C.mla_decode(out, q_nope, q_pe, kv_cache, block_table, seq_lens,
+                  scale, max_seq_len, tmp_out, tmp_lse)
+    return out
+
+
+def mla_attention(q_nope, q_pe, kv_cache, block_table, query_start_loc,
+                  seq_lens, scale, num_decodes=0, max_seq_len=0):
+    """Decode rows (query_len==1, first num_decodes) on the HIP MFMA
+    kernel; prefill rows on the torch composition (MLA prefill runs per
+    scheduler chunk — a bounded, non-steady-state path)."""
+    T = q_nope.shape[0]
+    num_reqs = seq_lens.shape[0]
+    use_hip = (q_nope.shape[2] == 512 and q_pe.shape[2] == 64
+               and q_nope.dtype in (torch.bfloat16, torch.float16)
+               and kv_cache.dtype == q_nope.dtype)
+    if num_decodes > 0 and use_hip:
+        out_dec = mla_decode(
+            q_nope[:num_decodes].contiguous(),
+            q_pe[:num_decodes].contiguous(), kv_cache,
+            block_table, seq_lens[:num_decodes], scale, max_seq_len)
+        if num_decodes == T:
+            return out_dec
+        qsl = query_start_loc[num_decodes:] - num_decodes
+        out_pre = _torch_ref.mla_attention(
+            q_nope[num_decodes:], q_pe[num_decodes:], kv_cache,
+            block_table[num_decodes:], qsl, seq_lens[num_decodes:], scale)
+        return torch.cat([out_dec, out_pre], dim=0)
+    return _torch_ref.mla_attention(
+        q_nope, q_pe, kv_cache, block_table, query_start_loc, seq_lens,
+        scale)
 
 
 _MOE_BM = 32  # must match moe::BM in csrc/moe.hip

@@ -36,9 +36,9 @@ class CustomCollectives:
 
     def __init__(self, rank_in_group: int, world_size: int,
                  max_bytes: int = DEFAULT_MAX_BYTES):
-        from vllm_amd import ops  # loads _C.so
+        from vllm_amd.ops import hip_ops  # loads _C.so
 
-        assert ops is not None
+        assert hip_ops is not None
         _C = torch.ops.vllm_amd
         self._C = _C
         self.world_size = world_size

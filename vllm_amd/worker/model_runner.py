@@ -383,12 +383,17 @@ class ModelRunner:
             moe_graph_ok = bool(
                 getattr(backend, "_moe_hip_ok", lambda *a: False)(
                     probe, w13, None, "silu"))
+        # MLA decode is graph-capturable on the MFMA kernel (csrc/mla.hip,
+        # kv_lora 512 + rope 64 — the DeepSeek geometry); other MLA dims
+        # fall back to the torch composition, which host-syncs.
+        mla_graph_ok = (not self.spec.is_mla
+                        or (self.spec.kv_lora_rank == 512
+                            and self.spec.qk_rope_head_dim == 64))
         if (self.device.type == "cuda"
                 and not self.config.model_config.enforce_eager
                 and not self.config.model_config.lora_modules
                 and self.pp_size == 1
-                and tp_graph_ok and moe_graph_ok
-                and not self.spec.is_mla):
+                and tp_graph_ok and moe_graph_ok and mla_graph_ok):
             self.graph_runner = DecodeGraphRunner(self)
 
     def allocate_host_kv_pool(self, num_host_blocks: int) -> None:
