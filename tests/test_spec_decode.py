@@ -49,7 +49,13 @@ def test_scheduler_spec_rollback():
     from vllm_amd.core.sched_output import ModelRunnerOutput
 
     sched = create_scheduler()
-    sched.spec_proposer = object()  # enable the propose path marker
+    class _NoneProposer:
+        model_based = False
+
+        def propose(self, toks):
+            return None
+
+    sched.spec_proposer = _NoneProposer()  # enable the propose path
     req = make_request("r1", num_tokens=16, max_tokens=32)
     req.sampling_params.ignore_eos = True
     sched.add_request(req)
@@ -219,3 +225,63 @@ def test_eagle_matches_baseline():
     assert base == egl
     assert all(len(t) == 24 for t in egl)
     assert stats[0] > 0  # drafts were actually produced and verified
+
+
+class _JunkProposer:
+    """Deterministic arbitrary proposer: rejection sampling must keep
+    the output distribution EXACTLY regardless of draft quality — bad
+    drafts only cost acceptance rate."""
+
+    model_based = False
+
+    def propose(self, toks):
+        t = toks[-1]
+        return [t, (t * 7 + 13) % 900 + 3, (t * 3 + 5) % 900 + 3]
+
+
+def _generate_sampled(spec_tokens: int, seed: int = 1234):
+    llm = LLM(model="tiny-llama", dtype="fp32", device="cpu",
+              block_size=16, num_gpu_blocks=256, max_model_len=512,
+              max_num_batched_tokens=512, max_num_seqs=4,
+              num_speculative_tokens=spec_tokens)
+    if spec_tokens > 0:
+        llm.engine.engine_core.scheduler.spec_proposer = _JunkProposer()
+    prompt = [7, 8, 9, 10] * 12
+    outs = llm.generate(
+        [prompt, list(range(30, 60))],
+        SamplingParams(temperature=1.0, seed=seed, max_tokens=24,
+                       ignore_eos=True),
+    )
+    sched = llm.engine.engine_core.scheduler
+    stats = (sched.spec_stats_drafted, sched.spec_stats_accepted)
+    llm.shutdown()
+    return [o.outputs[0].token_ids for o in outs], stats
+
+
+def test_sampled_spec_decode_preserves_distribution():
+    """Rejection sampling for temperature>0: with a fixed seed the
+    per-position target samples are identical with and without spec
+    decode, so outputs must match token-for-token — the strongest form
+    of the distribution-preservation property (reference
+    rejection_sampler.py:38 semantics for one-hot drafts)."""
+    base, _ = _generate_sampled(0)
+    spec, (drafted, accepted) = _generate_sampled(4)
+    assert base == spec
+    assert all(len(t) == 24 for t in spec)
+    assert drafted > 0  # drafts were actually scheduled at temp>0
+
+
+def test_sampled_spec_decode_accepts_drafts():
+    """Low-temperature sampled run on repetitive text: acceptance must
+    actually happen (not just scheduling) for temp>0 traffic."""
+    llm = LLM(model="tiny-llama", dtype="fp32", device="cpu",
+              block_size=16, num_gpu_blocks=256, max_model_len=512,
+              max_num_batched_tokens=512, max_num_seqs=4,
+              num_speculative_tokens=4)
+    prompt = [3, 4, 5, 6] * 14
+    llm.generate([prompt], SamplingParams(
+        temperature=0.05, seed=7, max_tokens=24, ignore_eos=True))
+    sched = llm.engine.engine_core.scheduler
+    assert sched.spec_stats_drafted > 0
+    assert sched.spec_stats_accepted > 0
+    llm.shutdown()

@@ -432,9 +432,13 @@ class Scheduler:
             if stopped:
                 self.running.remove(request)
                 self._free_request(request)
+            # Spec decode covers sampled requests too: the runner's
+            # verify loop is an exact rejection sampler for one-hot
+            # drafts (model_runner.py). Grammar requests stay gated —
+            # draft positions would need per-position grammar masks.
             elif (self.spec_proposer is not None
                     and request.grammar is None
-                    and request.sampling_params.temperature == 0.0):
+                    and not request.sampling_params.pooling):
                 if getattr(self.spec_proposer, "model_based", False):
                     drafts = (runner_output.draft_token_ids or {}).get(
                         req_id)

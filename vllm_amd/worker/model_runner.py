@@ -1046,12 +1046,22 @@ class ModelRunner:
         s_params = rep(lambda st: st.sampling_params)
         s_prompts = rep(lambda st: st.token_ids[: st.prompt_len])
         s_outputs = rep(lambda st: st.output_token_ids)
+        # Seed offset advances per POSITION within a spec-verify chunk:
+        # position j of a request with n output tokens draws the same
+        # u as the non-spec run would for token n+j, so seeded sampled
+        # outputs are bit-identical with and without spec decode (the
+        # acceptance rule below samples from the target distribution at
+        # every position).
+        s_seeds = []
+        for r, npos in zip(sampling_rows, sampling_npos):
+            base = len(self.requests[req_ids[r]].output_token_ids)
+            s_seeds.extend(base + j for j in range(npos))
         s_meta = SamplingMetadata.build(
             s_params,
             s_prompts,
             s_outputs,
             dev,
-            seeds_offset=[len(o) for o in s_outputs],
+            seeds_offset=s_seeds,
         )
         g_states = rep(lambda st: st)
         s_meta = self._with_grammar_masks(s_meta, g_states)
@@ -1071,8 +1081,15 @@ class ModelRunner:
             if npos == 1:
                 accepted = [int(row_sampled[0])]
             else:
-                # Greedy draft verification: accept draft j while the
-                # model's token at the previous position equals it.
+                # Rejection sampling (role of the reference's
+                # rejection_sampler.py:38), exact for one-hot draft
+                # distributions (our proposers emit tokens, not probs):
+                # row_sampled[j] ~ target dist p_j; accepting the draft
+                # iff the target's own sample equals it accepts with
+                # prob p_j(d_j), and on mismatch the sample itself is
+                # distributed as p_j conditioned on != d_j — which IS
+                # the adjusted distribution max(p - q, 0)/Z for one-hot
+                # q. Greedy (temp 0) degenerates to argmax matching.
                 drafts = spec_map[rid]
                 accepted = [int(row_sampled[0])]
                 for d_j in range(len(drafts)):
