@@ -370,9 +370,9 @@ def test_decode_attention_head_dims_window(hd):
 
 
 @pytest.mark.parametrize("hd", [64, 256])
-def test_prefill_torch_path_other_head_dims(hd):
-    """Mixed decode+prefill at head_dim != 128: decode rows on the HIP
-    kernel, prefill rows on the explicit torch path, one output."""
+def test_prefill_other_head_dims(hd):
+    """Mixed decode+prefill at head_dim 64/256: both row kinds now run
+    the templated MFMA/HIP kernels (Gemma3 256, OPT 64)."""
     hip = _hip()
     Hkv, group = 2, 2
     Hq = Hkv * group
@@ -499,3 +499,29 @@ def test_mla_unified_dispatch_mixed_batch():
     expect = ref.mla_attention(q_nope.float(), q_pe.float(),
                                kv_cache.float(), bt, qsl, seq_lens, scale)
     assert_close(out, expect, atol=4e-2, rtol=4e-2)
+
+
+@pytest.mark.parametrize("hd", [64, 128, 256])
+@pytest.mark.parametrize("q_lens,ctx_lens", [
+    ([16, 33], [16, 33]),                 # pure prefill from scratch
+    ([17, 70], [100, 300]),               # chunked-prefill continuation
+    ([1, 1, 5, 200], [40, 513, 60, 200]),  # mixed decode+prefill
+])
+def test_prefill_attention_head_dim_sweep(hd, q_lens, ctx_lens):
+    """head_dim-templated MFMA prefill (64/128/256) vs fp32 reference."""
+    hip = _hip()
+    Hkv = 2
+    Hq = Hkv * 4
+    n = len(q_lens)
+    num_decodes = sum(1 for x in q_lens if x == 1)
+    order = sorted(range(n), key=lambda i: q_lens[i] != 1)
+    q_lens = [q_lens[i] for i in order]
+    ctx_lens = [ctx_lens[i] for i in order]
+    q, cache, bt, qsl, sl = _make_paged(n, q_lens, ctx_lens, Hq, Hkv, D=hd)
+    scale = 1.0 / hd**0.5
+    out = hip.attention_unified(
+        q, cache, bt, qsl, sl, scale, num_decodes=num_decodes,
+        max_seq_len=max(ctx_lens), max_query_len=max(q_lens))
+    expect = ref.attention_unified(q, cache, bt, qsl, sl, scale,
+                                   num_decodes=num_decodes)
+    assert_close(out, expect, msg=f"hd={hd}")

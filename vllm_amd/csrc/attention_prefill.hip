@@ -29,7 +29,6 @@ namespace vllm_amd {
 constexpr int PF_BLOCK = 256;  // 4 waves
 constexpr int QTILE = 64;      // q rows per workgroup (16 per wave)
 constexpr int KVTILE = 64;     // keys per inner tile == cache block_size
-constexpr int HEAD_DIM = 128;
 constexpr float LOG2E = 1.4426950408889634f;
 
 // LDS byte-offset XOR swizzle: spread the 16 same-column rows of a
@@ -38,7 +37,7 @@ DEVINL int swz(int row, int col_byte) {
   return (col_byte ^ ((row & 7) << 4));
 }
 
-template <typename Tag, typename CTag>
+template <typename Tag, typename CTag, int HEAD_DIM>
 __global__ __launch_bounds__(PF_BLOCK) void prefill_attention_kernel(
     short* __restrict__ out,             // [T, Hq, D]
     const short* __restrict__ q,         // [T, Hq, D]
@@ -75,12 +74,12 @@ __global__ __launch_bounds__(PF_BLOCK) void prefill_attention_kernel(
   // A-frag layout (16x16x32): lane&15 = row, elems k = (lane>>4)*8 + j.
   const int q_row_frag = qtile0 + wave * 16 + l16;           // for A-frags
   const int q_row_safe = min(q_row_frag, ql - 1);
-  s16x8 qfrag[4];
+  s16x8 qfrag[HEAD_DIM / 32];
   {
     const short* qbase =
         q + (int64_t)(q_start + q_row_safe) * q_stride + (int64_t)h * HEAD_DIM;
 #pragma unroll
-    for (int kk = 0; kk < 4; ++kk)
+    for (int kk = 0; kk < HEAD_DIM / 32; ++kk)
       qfrag[kk] =
           *reinterpret_cast<const s16x8*>(qbase + kk * 32 + l4 * 8);
   }
@@ -93,9 +92,10 @@ __global__ __launch_bounds__(PF_BLOCK) void prefill_attention_kernel(
   for (int r = 0; r < 4; ++r) { m_st[r] = -3.0e38f; l_st[r] = 0.f; }
   // O^T accumulator: 8 d-subtiles; C layout col(lane&15)=q row,
   // row=(lane>>4)*4+reg = d within subtile.
-  f32x4 o_acc[8];
+  f32x4 o_acc[HEAD_DIM / 16];
 #pragma unroll
-  for (int i = 0; i < 8; ++i) o_acc[i] = f32x4{0.f, 0.f, 0.f, 0.f};
+  for (int i = 0; i < HEAD_DIM / 16; ++i)
+    o_acc[i] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   const float scale_log2e = scale * LOG2E;
   const int pos0 = ctx - ql;  // global position of local q row 0
@@ -116,24 +116,26 @@ __global__ __launch_bounds__(PF_BLOCK) void prefill_attention_kernel(
     const typename CT::elem* ksrc =
         kv_cache + ((int64_t)phys * num_kv_heads + kvh) * head_tile;
     const typename CT::elem* vsrc = ksrc + kv_plane_stride;
-    // K: 256 threads x 4 iters x 8 elements, swizzled ds_write_b128.
-    // (fp8 cache: converted to the compute dtype while staging to LDS —
-    // the MFMA tiles always run bf16/fp16.)
+    // K: 256 threads x (KVTILE*D/8/256) iters x 8 elements, swizzled
+    // ds_write_b128. (fp8 cache: converted to the compute dtype while
+    // staging to LDS — the MFMA tiles always run bf16/fp16.)
+    constexpr int CPR = HEAD_DIM / 8;  // 8-elem chunks per row
 #pragma unroll
-    for (int it = 0; it < 4; ++it) {
+    for (int it = 0; it < KVTILE * CPR / PF_BLOCK; ++it) {
       const int vec = it * PF_BLOCK + threadIdx.x;  // 8-elem chunk index
-      const int tok = vec >> 4;                     // 16 chunks per row
-      const int cb = (vec & 15) * 16;               // col byte
+      const int tok = vec / CPR;
+      const int cb = (vec % CPR) * 16;              // col byte
       cvec kraw = *reinterpret_cast<const cvec*>(ksrc + vec * 8);
       s16x8 kv8;
 #pragma unroll
       for (int j = 0; j < 8; ++j)
         kv8[j] = from_f32<Tag>(CT::get(kraw, j));
       *reinterpret_cast<s16x8*>(
-          reinterpret_cast<char*>(k_lds) + tok * 256 + swz(tok, cb)) = kv8;
+          reinterpret_cast<char*>(k_lds) + tok * (HEAD_DIM * 2) +
+          swz(tok, cb)) = kv8;
       // V: read the same shape, scatter-transpose into vt_lds.
       cvec vraw = *reinterpret_cast<const cvec*>(vsrc + vec * 8);
-      const int d0 = (vec & 15) * 8;
+      const int d0 = (vec % CPR) * 8;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         const int d = d0 + j;
@@ -152,10 +154,11 @@ __global__ __launch_bounds__(PF_BLOCK) void prefill_attention_kernel(
     for (int n = 0; n < 4; ++n) {       // key subtile
       const int tok = n * 16 + l16;     // B-frag col = token
 #pragma unroll
-      for (int kk = 0; kk < 4; ++kk) {  // d chunk of 32
+      for (int kk = 0; kk < HEAD_DIM / 32; ++kk) {  // d chunk of 32
         const int cb = (kk * 32 + l4 * 8) * 2;
         s16x8 bfrag = *reinterpret_cast<const s16x8*>(
-            reinterpret_cast<const char*>(k_lds) + tok * 256 + swz(tok, cb));
+            reinterpret_cast<const char*>(k_lds) + tok * (HEAD_DIM * 2) +
+            swz(tok, cb));
         s_frag[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             qfrag[kk], bfrag, s_frag[n], 0, 0, 0);
       }
@@ -211,7 +214,7 @@ __global__ __launch_bounds__(PF_BLOCK) void prefill_attention_kernel(
     // --- rescale O^T and accumulate PV -----------------------------------
     const float fac = f_lds[wave][l16];  // this lane's q column factor
 #pragma unroll
-    for (int i = 0; i < 8; ++i)
+    for (int i = 0; i < HEAD_DIM / 16; ++i)
 #pragma unroll
       for (int r = 0; r < 4; ++r) o_acc[i][r] *= fac;
 
@@ -224,7 +227,7 @@ __global__ __launch_bounds__(PF_BLOCK) void prefill_attention_kernel(
       s16x8 pfrag = *reinterpret_cast<const s16x8*>(
           &p_lds[wave][l16 * 80] + (kc * 32 + l4 * 8));
 #pragma unroll
-      for (int msub = 0; msub < 8; ++msub) {
+      for (int msub = 0; msub < HEAD_DIM / 16; ++msub) {
         const int d = msub * 16 + l16;
         s16x8 afrag = *reinterpret_cast<const s16x8*>(
             reinterpret_cast<const char*>(vt_lds) + d * 128 + swz(d, tb));
@@ -248,7 +251,7 @@ __global__ __launch_bounds__(PF_BLOCK) void prefill_attention_kernel(
     short* obase =
         out + ((int64_t)(q_start + qrow_o) * num_q_heads + h) * HEAD_DIM;
 #pragma unroll
-    for (int msub = 0; msub < 8; ++msub)
+    for (int msub = 0; msub < HEAD_DIM / 16; ++msub)
 #pragma unroll
       for (int r = 0; r < 4; ++r)
         obase[msub * 16 + l4 * 4 + r] = from_f32<Tag>(o_acc[msub][r] * inv_l);
@@ -269,7 +272,8 @@ void prefill_attention(torch::Tensor out, torch::Tensor q,
   const int num_q_heads = q.size(1);
   const int head_dim = q.size(2);
   const int num_kv_heads = kv_cache.size(2);
-  TORCH_CHECK(head_dim == HEAD_DIM, "prefill kernel supports head_dim=128");
+  TORCH_CHECK(head_dim == 64 || head_dim == 128 || head_dim == 256,
+              "prefill kernel supports head_dim 64/128/256");
   TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == head_dim &&
               q.stride(0) % 8 == 0, "q must be head-contiguous [T,H,D]");
   TORCH_CHECK(out.is_contiguous(), "out must be contiguous");
@@ -280,7 +284,7 @@ void prefill_attention(torch::Tensor out, torch::Tensor q,
   dim3 grid(max_tiles, num_q_heads, num_prefills);
 
 #define LAUNCH_PF(TAG, CTAG)                                                 \
-  hipLaunchKernelGGL((prefill_attention_kernel<TAG, CTAG>), grid,            \
+  hipLaunchKernelGGL((prefill_attention_kernel<TAG, CTAG, D>), grid,         \
                      dim3(PF_BLOCK),                                         \
                      0, stream, (short*)out.data_ptr(),                      \
                      (const short*)q.data_ptr(),                             \
@@ -293,13 +297,19 @@ void prefill_attention(torch::Tensor out, torch::Tensor q,
                      q.stride(0), (int)sliding_window)
 
   const bool fp8c = kv_cache.scalar_type() == torch::kFloat8_e4m3fn;
-  if (q.scalar_type() == torch::kBFloat16) {
-    if (fp8c) { LAUNCH_PF(BF16Tag, FP8CacheTag); }
-    else      { LAUNCH_PF(BF16Tag, BF16Tag); }
-  } else {
-    if (fp8c) { LAUNCH_PF(FP16Tag, FP8CacheTag); }
-    else      { LAUNCH_PF(FP16Tag, FP16Tag); }
-  }
+  auto dispatch = [&](auto dtag) {
+    constexpr int D = decltype(dtag)::value;
+    if (q.scalar_type() == torch::kBFloat16) {
+      if (fp8c) { LAUNCH_PF(BF16Tag, FP8CacheTag); }
+      else      { LAUNCH_PF(BF16Tag, BF16Tag); }
+    } else {
+      if (fp8c) { LAUNCH_PF(FP16Tag, FP8CacheTag); }
+      else      { LAUNCH_PF(FP16Tag, FP16Tag); }
+    }
+  };
+  if (head_dim == 64) dispatch(std::integral_constant<int, 64>{});
+  else if (head_dim == 128) dispatch(std::integral_constant<int, 128>{});
+  else dispatch(std::integral_constant<int, 256>{});
 #undef LAUNCH_PF
   HIP_CHECK_KERNEL();
 }

@@ -128,12 +128,11 @@ def attention_unified(
         )
 
     if num_decodes < num_reqs:
-        if head_dim != 128:
-            # The MFMA prefill kernel is 128-specific (LDS tiling); other
-            # head dims (Gemma3's 256, OPT's 64) take an explicit torch
-            # path for the one-shot prefill rows — decode, the
-            # steady-state loop, runs the HIP kernel above at any
-            # supported head_dim.
+        if head_dim not in (64, 128, 256):
+            # Off-template head dims take an explicit torch path for the
+            # prefill rows (decode runs the HIP kernel at any supported
+            # head_dim). Gemma3 (256) and OPT (64) are on the MFMA
+            # kernel since the head-dim templating.
             _warn_prefill_fallback(head_dim)
             from vllm_amd.ops import _torch_ref
 
