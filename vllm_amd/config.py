@@ -91,6 +91,13 @@ class ModelSpec:
     def is_mla(self) -> bool:
         return self.kv_lora_rank > 0
 
+    @property
+    def is_mixed_attn(self) -> bool:
+        """Mixed sliding-window + global layers (Gemma3 pattern): window
+        layers get their own KV block-table group so their out-of-window
+        blocks are reclaimed while global layers keep full-length KV."""
+        return self.sliding_window > 0 and self.global_attn_every_n_layers > 0
+
     def num_kv_heads_per_rank(self, tp_size: int) -> int:
         if self.num_kv_heads >= tp_size:
             assert self.num_kv_heads % tp_size == 0

@@ -24,8 +24,19 @@ class KVCacheManager:
         enable_caching: bool = True,
         num_host_blocks: int = 0,
         sliding_window: int = 0,
+        mixed_window: int = 0,
     ) -> None:
         self.block_size = block_size
+        # Mixed sliding+global models run two KV groups; prefix caching
+        # across groups needs hybrid hash alignment (reference
+        # kv_cache_coordinator.py HybridKVCacheCoordinator) — out of
+        # scope, so caching is disabled for them.
+        self.mixed_window = mixed_window
+        if mixed_window > 0 and enable_caching:
+            import logging
+            logging.getLogger(__name__).info(
+                "prefix caching disabled: mixed sliding+global KV groups")
+            enable_caching = False
         self.enable_caching = enable_caching
         # Host (CPU RAM) tier for evicted prefix blocks: GPU evictions
         # are saved D2H, host hits are restored H2D (core/host_kv_pool.py).
@@ -51,6 +62,14 @@ class KVCacheManager:
         self.sliding_window = sliding_window
         # req_id -> number of leading blocks already returned to the pool.
         self.num_reclaimed: dict[str, int] = defaultdict(int)
+        # Window-group structures (mixed models): a second block list per
+        # request, positionally aligned with the full-group list (stale
+        # leading ids stay in the table after reclaim; only the memory is
+        # returned). Role of the reference's per-group block tables
+        # (kv_cache_coordinator.py:60, SlidingWindowManager:878).
+        self.req_to_blocks_w: dict[str, list[KVCacheBlock]] = \
+            defaultdict(list)
+        self.num_reclaimed_w: dict[str, int] = defaultdict(int)
 
     def _on_evict(self, block_hash, block_id: int) -> None:
         slot = self.host_pool.put(block_hash)
@@ -135,7 +154,10 @@ class KVCacheManager:
             num_required_blocks - len(req_blocks) - len(new_computed_blocks)
         )
 
-        if num_new_blocks > self.block_pool.get_num_free_blocks():
+        need = max(num_new_blocks, 0)
+        if self.mixed_window > 0:
+            need *= 2  # window group allocates in lockstep
+        if need > self.block_pool.get_num_free_blocks():
             return None
 
         # Commit cache hits (bump refs) only after we know allocation fits.
@@ -153,6 +175,9 @@ class KVCacheManager:
         else:
             new_blocks = self.block_pool.get_new_blocks(num_new_blocks)
             req_blocks.extend(new_blocks)
+        if self.mixed_window > 0 and num_new_blocks > 0:
+            self.req_to_blocks_w[request.request_id].extend(
+                self.block_pool.get_new_blocks(num_new_blocks))
 
         # Sliding window: return leading blocks whose every token lies
         # outside the window of ALL future queries. One block of margin
@@ -166,6 +191,18 @@ class KVCacheManager:
                 stale = req_blocks[done:reclaim_below]
                 self.block_pool.free_blocks(list(reversed(stale)))
                 self.num_reclaimed[request.request_id] = reclaim_below
+        if self.mixed_window > 0:
+            # Same reclaim rule, applied to the WINDOW group only; global
+            # layers keep their full-length blocks.
+            reclaim_below = (
+                total_tokens - self.mixed_window - self.block_size
+            ) // self.block_size
+            done = self.num_reclaimed_w[request.request_id]
+            if reclaim_below > done:
+                blocks_w = self.req_to_blocks_w[request.request_id]
+                stale = blocks_w[done:reclaim_below]
+                self.block_pool.free_blocks(list(reversed(stale)))
+                self.num_reclaimed_w[request.request_id] = reclaim_below
 
         # Content-cache the blocks that become full after this step.
         if self.enable_caching:
@@ -221,9 +258,29 @@ class KVCacheManager:
         # are evicted first (LRU queue order). Skip blocks the sliding
         # window already returned.
         self.block_pool.free_blocks(list(reversed(blocks[reclaimed:])))
+        blocks_w = self.req_to_blocks_w.pop(request.request_id, [])
+        if blocks_w:
+            reclaimed_w = self.num_reclaimed_w.pop(request.request_id, 0)
+            self.block_pool.free_blocks(
+                list(reversed(blocks_w[reclaimed_w:])))
 
     def get_block_ids(self, request_id: str) -> list[int]:
         return [b.block_id for b in self.req_to_blocks[request_id]]
+
+    def get_block_ids_w(self, request_id: str) -> Optional[list[int]]:
+        if self.mixed_window <= 0:
+            return None
+        return [b.block_id for b in self.req_to_blocks_w[request_id]]
+
+    def last_w_block_ids(self, request_id: str,
+                         n: int) -> Optional[list[int]]:
+        """Ids of the n window-group blocks the latest allocate_slots
+        appended (positionally aligned with its returned full-group
+        blocks)."""
+        if self.mixed_window <= 0:
+            return None
+        blocks = self.req_to_blocks_w[request_id]
+        return [b.block_id for b in blocks[len(blocks) - n:]] if n else []
 
     def reset_prefix_cache(self) -> bool:
         if self.host_pool is not None:

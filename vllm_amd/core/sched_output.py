@@ -23,6 +23,9 @@ class NewRequestData:
     grammar: object = None  # CompiledGrammar for structured output
     lora_id: int = 0
     mm_data: object = None  # {"image": pixels} for vision models
+    # Window-group block ids (mixed sliding+global models); None for
+    # single-group models.
+    block_ids_w: object = None
 
 
 @dataclass
@@ -36,6 +39,9 @@ class CachedRequestData:
     new_block_ids: list[list[int]] = field(default_factory=list)
     num_computed_tokens: list[int] = field(default_factory=list)
     new_token_ids: list[list[int]] = field(default_factory=list)
+    # Parallel to new_block_ids for the window group (mixed models);
+    # entries are None for single-group models.
+    new_block_ids_w: list = field(default_factory=list)
 
     @property
     def num_reqs(self) -> int:

@@ -90,6 +90,8 @@ class Scheduler:
             enable_caching=config.cache_config.enable_prefix_caching,
             num_host_blocks=num_host_blocks,
             sliding_window=uniform_window,
+            mixed_window=(spec.sliding_window
+                          if spec.is_mixed_attn else 0),
         )
 
         self.requests: dict[str, Request] = {}
@@ -207,6 +209,9 @@ class Scheduler:
             cached.req_ids.append(request.request_id)
             cached.resumed.append(False)
             cached.new_block_ids.append([b.block_id for b in new_blocks])
+            cached.new_block_ids_w.append(
+                self.kv_cache_manager.last_w_block_ids(
+                    request.request_id, len(new_blocks)))
             cached.num_computed_tokens.append(request.num_computed_tokens)
             # Tokens the runner hasn't seen yet: any output tokens generated
             # since the prompt (runner keeps its own copy; for non-resumed
@@ -290,10 +295,14 @@ class Scheduler:
             all_block_ids = self.kv_cache_manager.get_block_ids(
                 request.request_id
             )
+            all_block_ids_w = self.kv_cache_manager.get_block_ids_w(
+                request.request_id
+            )
             if resumed:
                 cached.req_ids.append(request.request_id)
                 cached.resumed.append(True)
                 cached.new_block_ids.append(all_block_ids)
+                cached.new_block_ids_w.append(all_block_ids_w)
                 cached.num_computed_tokens.append(num_computed)
                 cached.new_token_ids.append(list(request.all_token_ids))
             else:
@@ -307,6 +316,7 @@ class Scheduler:
                         grammar=request.grammar,
                         lora_id=request.lora_id,
                         mm_data=request.mm_data,
+                        block_ids_w=all_block_ids_w,
                     )
                 )
             num_scheduled_tokens[request.request_id] = num_new_tokens

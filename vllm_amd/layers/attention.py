@@ -25,6 +25,7 @@ class Attention(nn.Module):
         num_kv_heads: int,
         layer_idx: int,
         sliding_window: int = 0,
+        kv_group: str = "full",
     ):
         super().__init__()
         self.num_heads = num_heads
@@ -33,6 +34,9 @@ class Attention(nn.Module):
         self.num_kv_heads = num_kv_heads
         self.layer_idx = layer_idx
         self.sliding_window = sliding_window
+        # "window": this layer reads/writes through the window group's
+        # block table when the model has mixed sliding+global layers.
+        self.kv_group = kv_group
 
     def forward(
         self, q: torch.Tensor, k: torch.Tensor, v: torch.Tensor
@@ -59,11 +63,15 @@ class Attention(nn.Module):
             return self._profile_attention(q, k, v, meta).reshape(T, -1)
 
         kv_cache = ctx.kv_caches[self.layer_idx]
-        ops.reshape_and_cache(k, v, kv_cache, meta.slot_mapping)
+        use_w = (self.kv_group == "window"
+                 and meta.block_table_w is not None)
+        slot_mapping = meta.slot_mapping_w if use_w else meta.slot_mapping
+        block_table = meta.block_table_w if use_w else meta.block_table
+        ops.reshape_and_cache(k, v, kv_cache, slot_mapping)
         out = ops.attention_unified(
             q,
             kv_cache,
-            meta.block_table,
+            block_table,
             meta.query_start_loc,
             meta.seq_lens,
             self.scale,

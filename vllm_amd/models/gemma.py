@@ -74,6 +74,7 @@ class GemmaAttention(nn.Module):
             self.num_heads, spec.head_dim, scale=scalar**-0.5,
             num_kv_heads=self.num_kv_heads, layer_idx=layer_idx,
             sliding_window=0 if self.is_global else spec.sliding_window,
+            kv_group="full" if self.is_global else "window",
         )
 
     def forward(self, positions, hidden):

@@ -97,6 +97,8 @@ class LlamaAttention(nn.Module):
             num_kv_heads=self.num_kv_heads,
             layer_idx=cache_idx if cache_idx >= 0 else layer_idx,
             sliding_window=window,
+            kv_group=("window" if window and spec.is_mixed_attn
+                      else "full"),
         )
 
     def forward(self, positions, hidden):

@@ -30,6 +30,13 @@ class AttentionMetadata:
     max_seq_len: int
     # Requests are ordered decodes-first; first num_decodes have query_len 1.
     num_decodes: int = 0
+    # Hybrid KV (mixed sliding+global models, e.g. Gemma3): window layers
+    # use their own block table / slot mapping so their out-of-window
+    # blocks can be reclaimed while global layers keep full-length KV
+    # (role of the reference's per-group block tables,
+    # kv_cache_coordinator.py:60). None = single-group model.
+    block_table_w: Optional[torch.Tensor] = None
+    slot_mapping_w: Optional[torch.Tensor] = None
 
 
 @dataclass

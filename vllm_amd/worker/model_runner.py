@@ -42,6 +42,7 @@ class CachedReqState:
     lora_id: int = 0
     pool_acc: object = None  # running hidden sum for mean pooling
     mm_data: object = None        # {"image": pixels}
+    block_ids_w: object = None    # window-group blocks (mixed models)
     mm_feats: object = None       # encoded-once projected patch rows
     mm_img_pos: object = None     # np positions of image tokens in prompt
 
@@ -108,8 +109,16 @@ class DecodeGraphRunner:
                                             device=dev)
         self.block_table = torch.zeros(n, self.max_blocks, dtype=torch.int32,
                                        device=dev)
+        self.mixed_attn = runner.spec.is_mixed_attn
+        if self.mixed_attn:
+            self.slot_mapping_w = torch.full((n,), -1, dtype=torch.int64,
+                                             device=dev)
+            self.block_table_w = torch.zeros(n, self.max_blocks,
+                                             dtype=torch.int32, device=dev)
+            self.pin_bt_w = torch.empty(n, self.max_blocks,
+                                        dtype=torch.int32, pin_memory=True)
         # Pinned staging buffers for async H2D.
-        self.pin_i64 = torch.empty(3, n, dtype=torch.int64, pin_memory=True)
+        self.pin_i64 = torch.empty(4, n, dtype=torch.int64, pin_memory=True)
         self.pin_seq = torch.empty(n, dtype=torch.int32, pin_memory=True)
         self.pin_bt = torch.empty(n, self.max_blocks, dtype=torch.int32,
                                   pin_memory=True)
@@ -140,6 +149,10 @@ class DecodeGraphRunner:
             max_query_len=1,
             max_seq_len=parts * self.DEC_PART,
             num_decodes=nb,
+            block_table_w=(self.block_table_w[:nb]
+                           if self.mixed_attn else None),
+            slot_mapping_w=(self.slot_mapping_w[:nb]
+                            if self.mixed_attn else None),
         )
 
     def _capture(self, nb: int, parts: int):
@@ -165,7 +178,8 @@ class DecodeGraphRunner:
 
     def run(self, n: int, nb: int, parts: int, input_ids, positions,
             slot_mapping, seq_lens, block_table,
-            ids_dev: Optional[torch.Tensor] = None) -> torch.Tensor:
+            ids_dev: Optional[torch.Tensor] = None,
+            slot_mapping_w=None, block_table_w=None) -> torch.Tensor:
         """Stage inputs (n real rows, nb bucket) and replay. Returns
         logits for the n real rows. ids_dev, when given, is a device
         tensor of decode input ids (previous step's sampled tokens — no
@@ -180,6 +194,13 @@ class DecodeGraphRunner:
         self.pin_seq[n:nb] = 0
         w = block_table.shape[1]
         self.pin_bt[:n, :w] = torch.from_numpy(block_table)
+        if self.mixed_attn:
+            pi[3, :n] = torch.from_numpy(slot_mapping_w)
+            pi[3, n:nb] = -1
+            self.pin_bt_w[:n, :w] = torch.from_numpy(block_table_w)
+            self.slot_mapping_w[:nb].copy_(pi[3, :nb], non_blocking=True)
+            self.block_table_w[:n, :w].copy_(self.pin_bt_w[:n, :w],
+                                             non_blocking=True)
         if ids_dev is None:
             self.input_ids[:nb].copy_(pi[0, :nb], non_blocking=True)
         else:
@@ -234,6 +255,11 @@ class ModelRunner:
         self.np_last_tok = np.zeros(n, dtype=np.int64)
         self.np_computed = np.zeros(n, dtype=np.int64)
         self.np_block_table = np.zeros((n, mb), dtype=np.int32)
+        # Window-group table for mixed sliding+global models (hybrid KV):
+        # same positional indexing, separate physical blocks.
+        self.mixed_attn = self.spec.is_mixed_attn
+        self.np_block_table_w = (np.zeros((n, mb), dtype=np.int32)
+                                 if self.mixed_attn else None)
         self.np_nblocks = np.zeros(n, dtype=np.int32)
         self.np_lora = np.zeros(n, dtype=np.int64)
         self.lora_manager = None
@@ -542,11 +568,16 @@ class ModelRunner:
                                if nr.grammar is not None else None),
                 lora_id=nr.lora_id,
                 mm_data=nr.mm_data,
+                block_ids_w=(list(nr.block_ids_w)
+                             if nr.block_ids_w is not None else None),
             )
             row = self._free_rows.pop()
             self._row_of[nr.req_id] = row
             nb = len(nr.block_ids)
             self.np_block_table[row, :nb] = nr.block_ids
+            if self.mixed_attn and nr.block_ids_w:
+                self.np_block_table_w[row, :len(nr.block_ids_w)] = \
+                    nr.block_ids_w
             self.np_nblocks[row] = nb
             self.np_computed[row] = nr.num_computed_tokens
             self.np_lora[row] = nr.lora_id
@@ -554,6 +585,8 @@ class ModelRunner:
         for i, req_id in enumerate(cr.req_ids):
             state = self.requests[req_id]
             row = self._row_of[req_id]
+            newb_w = (cr.new_block_ids_w[i]
+                      if i < len(cr.new_block_ids_w) else None)
             if cr.resumed[i]:
                 state.block_ids = list(cr.new_block_ids[i])
                 state.token_ids = list(cr.new_token_ids[i])
@@ -561,6 +594,9 @@ class ModelRunner:
                 nb = len(state.block_ids)
                 self.np_block_table[row, :nb] = state.block_ids
                 self.np_nblocks[row] = nb
+                if self.mixed_attn and newb_w is not None:
+                    state.block_ids_w = list(newb_w)
+                    self.np_block_table_w[row, :len(newb_w)] = newb_w
             else:
                 state.block_ids.extend(cr.new_block_ids[i])
                 state.num_computed_tokens = cr.num_computed_tokens[i]
@@ -569,6 +605,12 @@ class ModelRunner:
                     nb0 = self.np_nblocks[row]
                     self.np_block_table[row, nb0:nb0 + len(newb)] = newb
                     self.np_nblocks[row] = nb0 + len(newb)
+                if self.mixed_attn and newb_w:
+                    if state.block_ids_w is None:
+                        state.block_ids_w = []
+                    w0 = len(state.block_ids_w)
+                    state.block_ids_w.extend(newb_w)
+                    self.np_block_table_w[row, w0:w0 + len(newb_w)] = newb_w
             self.np_computed[row] = cr.num_computed_tokens[i]
 
     # ------------------------------------------------------------------
@@ -626,6 +668,12 @@ class ModelRunner:
         max_seq_len = int(seq_lens.max())
         w = _cdiv(max_seq_len, self.block_size)
         block_table = self.np_block_table[rows][:, :w]
+        slot_mapping_w = block_table_w = None
+        if self.mixed_attn:
+            slot_mapping_w = (
+                self.np_block_table_w[rows, blk].astype(np.int64)
+                * self.block_size + positions % self.block_size)
+            block_table_w = self.np_block_table_w[rows][:, :w]
 
         dev = self.device
         nb = (self.graph_runner.bucket_for(n)
@@ -636,6 +684,7 @@ class ModelRunner:
             logits = self.graph_runner.run(
                 n, nb, parts, input_ids, positions, slot_mapping, seq_lens,
                 block_table, ids_dev=ids_dev,
+                slot_mapping_w=slot_mapping_w, block_table_w=block_table_w,
             )
             hidden = self.graph_runner.last_hidden
         else:
@@ -651,6 +700,11 @@ class ModelRunner:
                 max_query_len=1,
                 max_seq_len=max_seq_len,
                 num_decodes=n,
+                block_table_w=(torch.from_numpy(np.ascontiguousarray(
+                    block_table_w)).to(dev) if block_table_w is not None
+                    else None),
+                slot_mapping_w=(torch.from_numpy(slot_mapping_w).to(dev)
+                                if slot_mapping_w is not None else None),
             )
             ids_t = (ids_dev if ids_dev is not None
                      else torch.from_numpy(input_ids).to(dev))
@@ -815,6 +869,11 @@ class ModelRunner:
             (len(self.requests[rid].block_ids) for rid in req_ids), default=1
         )
         block_table = np.zeros((len(items), max_blocks), dtype=np.int32)
+        slot_mapping_w = block_table_w = None
+        if self.mixed_attn:
+            slot_mapping_w = np.empty(total, dtype=np.int64)
+            block_table_w = np.zeros((len(items), max_blocks),
+                                     dtype=np.int32)
 
         spec_map = so.scheduled_spec_decode_tokens or {}
         sampling_rows: list[int] = []  # row in `items` order
@@ -841,6 +900,14 @@ class ModelRunner:
             query_start_loc[i + 1] = query_start_loc[i] + n
             seq_lens[i] = start + n
             block_table[i, : len(state.block_ids)] = state.block_ids
+            if self.mixed_attn:
+                bids_w = np.asarray(state.block_ids_w or state.block_ids,
+                                    dtype=np.int64)
+                slot_mapping_w[t : t + n] = (
+                    bids_w[pos // self.block_size] * self.block_size
+                    + pos % self.block_size
+                )
+                block_table_w[i, : len(bids_w)] = bids_w
             if rid in spec_map:
                 sampling_rows.append(i)
                 sampling_npos.append(len(spec_map[rid]) + 1)
@@ -863,6 +930,10 @@ class ModelRunner:
             query_start_loc=torch.from_numpy(query_start_loc).to(dev),
             seq_lens=torch.from_numpy(seq_lens).to(dev),
             block_table=torch.from_numpy(block_table).to(dev),
+            block_table_w=(torch.from_numpy(block_table_w).to(dev)
+                           if block_table_w is not None else None),
+            slot_mapping_w=(torch.from_numpy(slot_mapping_w).to(dev)
+                            if slot_mapping_w is not None else None),
             slot_mapping=torch.from_numpy(slot_mapping).to(dev),
             num_reqs=len(items),
             num_actual_tokens=total,
