@@ -4,11 +4,11 @@
 // (vllm/model_executor/layers/fused_moe/fused_moe.py:299) and
 // moe_align_block_size (csrc/libtorch_stable/moe/moe_align_sum_kernels.cu:326),
 // redesigned as hand-written MFMA kernels: tokens are sorted by expert
-// into BLOCK_M=32 tiles on-device (no host sync — shapes depend only on
+// into BLOCK_M=64 tiles on-device (no host sync — shapes depend only on
 // (T, topk, E), so the whole MoE layer is hipGraph-capturable), then a
 // tiled 16x16x32-bf16 MFMA GEMM streams each expert's weights from HBM
 // once per m-tile. Decode-shape MoE is weight-bound (~32 flops per
-// weight byte at BM=32), so the kernel targets the HBM roofline, not
+// weight byte at BM=64), so the kernel targets the HBM roofline, not
 // peak MFMA.
 //
 // Pipeline (ops/hip_ops.py fused_moe):
@@ -29,7 +29,7 @@ namespace vllm_amd {
 
 namespace moe {
 
-constexpr int BM = 32;    // token rows per tile (must divide align padding)
+constexpr int BM = 64;    // token rows per tile (must divide align padding)
 constexpr int BN = 128;   // output cols per block
 constexpr int BK = 64;    // K step staged in LDS
 constexpr int THREADS = 256;
@@ -141,28 +141,33 @@ void moe_gemm_kernel(const short* __restrict__ A, long lda,
   const int lane = tid % WAVE_SIZE;
   const int l16 = lane & 15;
   const int l4 = lane >> 4;       // 0..3: k-quarter for A/B frags
-  const int wm = wave & 1;        // m 16-half
-  const int wn = wave >> 1;       // n 64-half
+  const int wm = wave & 1;        // m 32-half (2 m-frags of 16)
+  const int wn = wave >> 1;       // n 64-half (4 n-frags of 16)
 
-  f32x4 acc[4];
+  f32x4 acc[2][4];
 #pragma unroll
-  for (int i = 0; i < 4; ++i) acc[i] = f32x4{0.f, 0.f, 0.f, 0.f};
+  for (int m = 0; m < 2; ++m)
+#pragma unroll
+    for (int i = 0; i < 4; ++i) acc[m][i] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   const short* b_exp = B + (long)expert * expert_stride;
 
   for (int k0 = 0; k0 < K; k0 += BK) {
-    // Stage A: 32 rows x 64 cols bf16 = 32 x 128B; 256 threads take 16B
-    // each (8 threads per row).
+    // Stage A: BM rows x 64 cols bf16 = BM x 128B; 16B per thread-chunk.
     {
-      const int r = tid >> 3;          // 0..31
-      const int cb = (tid & 7) * 16;   // byte col 0..112
-      const int arow = rows_sh[r];
-      s16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
-      if (arow >= 0)
-        v = *reinterpret_cast<const s16x8*>(A + (long)arow * lda + k0 +
-                                            (cb >> 1));
-      *reinterpret_cast<s16x8*>(
-          reinterpret_cast<char*>(a_lds) + r * 128 + swz(r, cb)) = v;
+#pragma unroll
+      for (int it = 0; it < BM * 8 / THREADS; ++it) {
+        const int idx = it * THREADS + tid;
+        const int r = idx >> 3;
+        const int cb = (idx & 7) * 16;   // byte col 0..112
+        const int arow = rows_sh[r];
+        s16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+        if (arow >= 0)
+          v = *reinterpret_cast<const s16x8*>(A + (long)arow * lda + k0 +
+                                              (cb >> 1));
+        *reinterpret_cast<s16x8*>(
+            reinterpret_cast<char*>(a_lds) + r * 128 + swz(r, cb)) = v;
+      }
     }
     // Stage B: 128 rows (n) x 64 cols = 128 x 128B; 4 x 16B per thread.
     {
@@ -181,18 +186,25 @@ void moe_gemm_kernel(const short* __restrict__ A, long lda,
     // 2 K-steps of 32 within the tile.
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
-      const int arow = wm * 16 + l16;
       const int kb = kk * 64 + l4 * 16;  // byte offset of this lane's 8 elems
-      s16x8 afrag = *reinterpret_cast<const s16x8*>(
-          reinterpret_cast<const char*>(a_lds) + arow * 128 + swz(arow, kb));
+      s16x8 afrag[2];
+#pragma unroll
+      for (int m = 0; m < 2; ++m) {
+        const int arow = wm * 32 + m * 16 + l16;
+        afrag[m] = *reinterpret_cast<const s16x8*>(
+            reinterpret_cast<const char*>(a_lds) + arow * 128 +
+            swz(arow, kb));
+      }
 #pragma unroll
       for (int nf = 0; nf < 4; ++nf) {
         const int bcol = wn * 64 + nf * 16 + l16;
         s16x8 bfrag = *reinterpret_cast<const s16x8*>(
             reinterpret_cast<const char*>(b_lds) + bcol * 128 +
             swz(bcol, kb));
-        acc[nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag,
-                                                          acc[nf], 0, 0, 0);
+#pragma unroll
+        for (int m = 0; m < 2; ++m)
+          acc[m][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[m], bfrag, acc[m][nf], 0, 0, 0);
       }
     }
     __syncthreads();
@@ -200,13 +212,17 @@ void moe_gemm_kernel(const short* __restrict__ A, long lda,
 
   // Epilogue: C rows are sorted positions; skip padding rows.
 #pragma unroll
-  for (int nf = 0; nf < 4; ++nf) {
-    const int n = n0 + wn * 64 + nf * 16 + l16;
+  for (int m = 0; m < 2; ++m) {
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int mrow = wm * 16 + l4 * 4 + r;
-      if (cval_sh[mrow])
-        C[(long)(tile_m * BM + mrow) * ldc + n] = from_f32<Tag>(acc[nf][r]);
+    for (int nf = 0; nf < 4; ++nf) {
+      const int n = n0 + wn * 64 + nf * 16 + l16;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int mrow = wm * 32 + m * 16 + l4 * 4 + r;
+        if (cval_sh[mrow])
+          C[(long)(tile_m * BM + mrow) * ldc + n] =
+              from_f32<Tag>(acc[m][nf][r]);
+      }
     }
   }
 }

@@ -228,7 +228,7 @@ def mla_attention(q_nope, q_pe, kv_cache, block_table, query_start_loc,
         scale)
 
 
-_MOE_BM = 32  # must match moe::BM in csrc/moe.hip
+_MOE_BM = 64  # must match moe::BM in csrc/moe.hip
 
 
 def _moe_hip_ok(hidden, w13, w2, activation):
