@@ -525,3 +525,26 @@ def test_prefill_attention_head_dim_sweep(hd, q_lens, ctx_lens):
     expect = ref.attention_unified(q, cache, bt, qsl, sl, scale,
                                    num_decodes=num_decodes)
     assert_close(out, expect, msg=f"hd={hd}")
+
+
+@pytest.mark.parametrize("T,H,I,E,K", [
+    (64, 256, 512, 8, 2),
+    (37, 256, 128, 16, 4),
+])
+def test_moe_shuffled_weight_gemm(T, H, I, E, K):
+    """Fragment-major pre-shuffled weight stream (moe_gemm_shuf) must
+    match the staged-LDS grouped GEMM and the fp32 reference."""
+    hip = _hip()
+    hidden = torch.randn(T, H, dtype=torch.bfloat16, device="cuda") * 0.3
+    w13 = torch.randn(E, 2 * I, H, dtype=torch.bfloat16, device="cuda") * 0.2
+    w2 = torch.randn(E, H, I, dtype=torch.bfloat16, device="cuda") * 0.2
+    logits = torch.randn(T, E, dtype=torch.float32, device="cuda")
+    tw, ti = hip.topk_softmax(logits, K)
+    w13s = hip.moe_shuffle_weights(w13)
+    w2s = hip.moe_shuffle_weights(w2)
+    out = hip.fused_moe_hip(hidden, w13, w2, tw, ti,
+                            w13_shuf=w13s, w2_shuf=w2s)
+    base = hip.fused_moe_hip(hidden, w13, w2, tw, ti)
+    expect = ref.fused_moe(hidden.float(), w13.float(), w2.float(), tw, ti)
+    assert_close(out, base, atol=1e-3, rtol=1e-3)  # same math, same order
+    assert_close(out, expect, atol=8e-2, rtol=8e-2)

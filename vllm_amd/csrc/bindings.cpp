@@ -47,6 +47,10 @@ void moe_gemm(torch::Tensor a, torch::Tensor b, torch::Tensor c,
               int64_t topk_div, int64_t total_flat);
 void moe_combine(torch::Tensor out, torch::Tensor y,
                  torch::Tensor topk_weights, torch::Tensor inv_perm);
+void moe_gemm_shuf(torch::Tensor a, torch::Tensor b_shuf, torch::Tensor c,
+                   torch::Tensor sorted, torch::Tensor expert_tiles,
+                   int64_t n, int64_t k, int64_t topk_div,
+                   int64_t total_flat);
 void mla_decode(torch::Tensor out, torch::Tensor q_nope, torch::Tensor q_pe,
                 torch::Tensor kv_cache, torch::Tensor block_table,
                 torch::Tensor seq_lens, double scale, int64_t max_seq_len,
@@ -92,6 +96,9 @@ TORCH_LIBRARY(vllm_amd, m) {
         "Tensor expert_tiles, int topk_div, int total_flat) -> ()");
   m.def("moe_combine(Tensor(a!) out, Tensor y, Tensor topk_weights, "
         "Tensor inv_perm) -> ()");
+  m.def("moe_gemm_shuf(Tensor a, Tensor b_shuf, Tensor(a!) c, "
+        "Tensor sorted, Tensor expert_tiles, int n, int k, int topk_div, "
+        "int total_flat) -> ()");
   m.def("mla_decode(Tensor(a!) out, Tensor q_nope, Tensor q_pe, "
         "Tensor kv_cache, Tensor block_table, Tensor seq_lens, "
         "float scale, int max_seq_len, Tensor(b!) tmp_out, "
@@ -131,6 +138,7 @@ TORCH_LIBRARY_IMPL(vllm_amd, CUDA, m) {
   m.impl("car_all_gather", &vllm_amd::car_all_gather);
   m.impl("moe_align", &vllm_amd::moe_align);
   m.impl("moe_gemm", &vllm_amd::moe_gemm);
+  m.impl("moe_gemm_shuf", &vllm_amd::moe_gemm_shuf);
   m.impl("moe_combine", &vllm_amd::moe_combine);
   m.impl("mla_decode", &vllm_amd::mla_decode);
 }

@@ -228,6 +228,122 @@ void moe_gemm_kernel(const short* __restrict__ A, long lda,
 }
 
 // --------------------------------------------------------------------------
+// Fragment-major grouped GEMM: B is pre-shuffled at weight-load time
+// into MFMA-fragment order [E][N/16][K/32][64 lanes][8 elems] so each
+// wave reads its B-fragment as ONE coalesced 16B/lane chunk straight
+// from HBM — no LDS staging, no per-kstep barrier for the (static)
+// weights. Only the small A tile stages through LDS, in ABK=256 chunks
+// (one barrier per 8 k-steps). This is the MI355X-first inference
+// layout: weights are written once and streamed billions of times, so
+// they belong in consumer order (role of the reference's Marlin-style
+// weight repacking, csrc/quantization/marlin — re-thought for bf16
+// MFMA fragments instead of int4 tiles).
+constexpr int ABK = 256;  // A-tile K chunk per barrier (8 k-steps of 32)
+
+template <typename Tag>
+__launch_bounds__(THREADS) __global__
+void moe_gemm_shuf_kernel(const short* __restrict__ A, long lda,
+                          const short* __restrict__ B, long expert_stride,
+                          short* __restrict__ C, long ldc,
+                          const int* __restrict__ sorted,
+                          const int* __restrict__ expert_tiles, int K,
+                          int N, int topk_div, int total_flat) {
+  const int tile_m = blockIdx.y;
+  const int n0 = blockIdx.x * BN;
+  const int expert = expert_tiles[tile_m];
+  if (expert < 0) return;  // padding tile
+
+  __shared__ short a_lds[BM * ABK];  // [row][k], swizzled 16B slots
+  __shared__ int rows_sh[BM];
+  __shared__ int cval_sh[BM];
+
+  const int tid = threadIdx.x;
+  if (tid < BM) {
+    int sid = sorted[tile_m * BM + tid];
+    cval_sh[tid] = (sid < total_flat) ? 1 : 0;
+    rows_sh[tid] = (sid < total_flat)
+                       ? (topk_div > 0 ? sid / topk_div : tile_m * BM + tid)
+                       : -1;
+  }
+  __syncthreads();
+
+  const int wave = tid / WAVE_SIZE;
+  const int lane = tid % WAVE_SIZE;
+  const int l16 = lane & 15;
+  const int l4 = lane >> 4;
+  const int wm = wave & 1;   // m 32-half (2 m-frags)
+  const int wn = wave >> 1;  // n 64-half (4 n-frags)
+
+  f32x4 acc[2][4];
+#pragma unroll
+  for (int m = 0; m < 2; ++m)
+#pragma unroll
+    for (int i = 0; i < 4; ++i) acc[m][i] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  // B fragment stream base for this wave: 4 n-frags at cols
+  // n0 + wn*64 + nf*16; fragment granule = 64 lanes x 8 elems = 512.
+  const int ksteps_total = K / 32;
+  const short* bbase = B + (long)expert * expert_stride;
+
+  for (int k0 = 0; k0 < K; k0 += ABK) {
+    const int kchunk = min(ABK, K - k0);
+    // Stage A chunk: BM x kchunk; 16B per thread-chunk, swizzled.
+    for (int idx = tid; idx < BM * (kchunk / 8); idx += THREADS) {
+      const int cpr = kchunk / 8;
+      const int r = idx / cpr;
+      const int cb = (idx % cpr) * 16;
+      const int arow = rows_sh[r];
+      s16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+      if (arow >= 0)
+        v = *reinterpret_cast<const s16x8*>(A + (long)arow * lda + k0 +
+                                            (cb >> 1));
+      *reinterpret_cast<s16x8*>(
+          reinterpret_cast<char*>(a_lds) + r * (ABK * 2) + swz(r, cb)) = v;
+    }
+    __syncthreads();
+#pragma unroll 2
+    for (int kk = 0; kk < kchunk / 32; ++kk) {
+      const int ks = (k0 >> 5) + kk;  // global kstep
+      const int kb = (kk * 32 + l4 * 8) * 2;
+      s16x8 afrag[2];
+#pragma unroll
+      for (int m = 0; m < 2; ++m) {
+        const int arow = wm * 32 + m * 16 + l16;
+        afrag[m] = *reinterpret_cast<const s16x8*>(
+            reinterpret_cast<const char*>(a_lds) + arow * (ABK * 2) +
+            swz(arow, kb));
+      }
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf) {
+        const int ntile = (n0 >> 4) + wn * 4 + nf;
+        s16x8 bfrag = *reinterpret_cast<const s16x8*>(
+            bbase + ((long)ntile * ksteps_total + ks) * 512 + lane * 8);
+#pragma unroll
+        for (int m = 0; m < 2; ++m)
+          acc[m][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[m], bfrag, acc[m][nf], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int m = 0; m < 2; ++m) {
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) {
+      const int n = n0 + wn * 64 + nf * 16 + l16;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int mrow = wm * 32 + m * 16 + l4 * 4 + r;
+        if (cval_sh[mrow])
+          C[(long)(tile_m * BM + mrow) * ldc + n] =
+              from_f32<Tag>(acc[m][nf][r]);
+      }
+    }
+  }
+}
+
+// --------------------------------------------------------------------------
 // Weighted combine: out[t, :] = sum_k w[t*K+k] * Y[inv_perm[t*K+k], :].
 template <typename Tag>
 __global__ void combine_kernel(short* __restrict__ out,
@@ -348,6 +464,47 @@ void moe_combine(torch::Tensor out, torch::Tensor y,
     launch(FP16Tag{});
   else
     TORCH_CHECK(false, "moe_combine: bf16/fp16 only");
+  HIP_CHECK_KERNEL();
+}
+
+
+void moe_gemm_shuf(torch::Tensor a, torch::Tensor b_shuf, torch::Tensor c,
+                   torch::Tensor sorted, torch::Tensor expert_tiles,
+                   int64_t n, int64_t k, int64_t topk_div,
+                   int64_t total_flat) {
+  using namespace moe;
+  TORCH_CHECK(a.dim() == 2 && c.dim() == 2);
+  TORCH_CHECK(a.is_contiguous() && b_shuf.is_contiguous() &&
+              c.is_contiguous());
+  TORCH_CHECK(a.scalar_type() == b_shuf.scalar_type() &&
+              a.scalar_type() == c.scalar_type());
+  const int K = (int)k;
+  const int N = (int)n;
+  TORCH_CHECK(a.size(1) == K);
+  TORCH_CHECK(K % 64 == 0 && N % BN == 0);
+  const int em = (int)c.size(0);
+  TORCH_CHECK(em % BM == 0 && em / BM <= (int)expert_tiles.numel());
+  TORCH_CHECK(c.size(1) == N);
+  const long expert_stride = (long)(N / 16) * (K / 32) * 512;
+  TORCH_CHECK(b_shuf.numel() % expert_stride == 0);
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  dim3 grid(N / BN, em / BM);
+  auto launch = [&](auto tag) {
+    using Tag = decltype(tag);
+    hipLaunchKernelGGL(
+        (moe_gemm_shuf_kernel<Tag>), grid, dim3(THREADS), 0, stream,
+        (const short*)a.data_ptr(), a.size(1),
+        (const short*)b_shuf.data_ptr(), expert_stride,
+        (short*)c.data_ptr(), c.size(1), sorted.data_ptr<int>(),
+        expert_tiles.data_ptr<int>(), K, N, (int)topk_div,
+        (int)total_flat);
+  };
+  if (a.scalar_type() == torch::kBFloat16)
+    launch(BF16Tag{});
+  else if (a.scalar_type() == torch::kHalf)
+    launch(FP16Tag{});
+  else
+    TORCH_CHECK(false, "moe_gemm_shuf: bf16/fp16 only");
   HIP_CHECK_KERNEL();
 }
 

@@ -80,6 +80,41 @@ class FusedMoE(nn.Module):
                         dtype=dtype),
             requires_grad=False,
         )
+        # Fragment-major weight copies for the HIP grouped GEMM (built
+        # lazily on the first GPU forward; freed by invalidate_shuffled
+        # when weights change). Costs a second copy of the expert
+        # weights — the MI355X trade: 288 GB HBM buys coalesced,
+        # barrier-free weight streams. VLLM_AMD_MOE_SHUF=0 disables.
+        self._w13_shuf = None
+        self._w2_shuf = None
+
+    def refresh_shuffled(self) -> None:
+        """Call after in-place weight updates (RL update_weights):
+        re-shuffles INTO THE EXISTING storage so captured hipGraphs that
+        reference the fragment-major tensors stay valid."""
+        if self._w13_shuf is None:
+            return
+        from vllm_amd.ops import get_backend
+        backend = get_backend(self.w13.device)
+        self._w13_shuf.copy_(backend.moe_shuffle_weights(self.w13.data))
+        self._w2_shuf.copy_(backend.moe_shuffle_weights(self.w2.data))
+
+    def _maybe_shuffled(self, hidden):
+        import os
+
+        if (not hidden.is_cuda
+                or os.environ.get("VLLM_AMD_MOE_SHUF", "1") == "0"):
+            return None, None
+        if self._w13_shuf is None:
+            from vllm_amd.ops import get_backend
+            backend = get_backend(hidden.device)
+            shuffle = getattr(backend, "moe_shuffle_weights", None)
+            if shuffle is None or not backend._moe_hip_ok(
+                    hidden, self.w13, self.w2, self.activation):
+                return None, None
+            self._w13_shuf = shuffle(self.w13.data)
+            self._w2_shuf = shuffle(self.w2.data)
+        return self._w13_shuf, self._w2_shuf
 
     def forward(self, hidden: torch.Tensor) -> torch.Tensor:
         router_logits = self.gate(hidden)
@@ -108,17 +143,19 @@ class FusedMoE(nn.Module):
                 return torch.zeros_like(hidden)
             sel_ids = (topk_ids[slot_tok, slot_k] - lo).unsqueeze(1)
             sel_w = topk_weights[slot_tok, slot_k].unsqueeze(1)
+            w13s, w2s = self._maybe_shuffled(hidden)
             y = ops.fused_moe(
                 hidden[slot_tok], self.w13, self.w2, sel_w, sel_ids,
-                activation=self.activation,
+                activation=self.activation, w13_shuf=w13s, w2_shuf=w2s,
             )
             out = torch.zeros(hidden.shape, dtype=torch.float32,
                               device=hidden.device)
             out.index_add_(0, slot_tok, y.float())
             return out.to(hidden.dtype)
+        w13s, w2s = self._maybe_shuffled(hidden)
         return ops.fused_moe(
             hidden, self.w13, self.w2, topk_weights, topk_ids,
-            activation=self.activation,
+            activation=self.activation, w13_shuf=w13s, w2_shuf=w2s,
         )
 
     def load_full_weights(self, w1_full, w3_full, w2_full) -> None:

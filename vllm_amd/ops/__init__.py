@@ -160,7 +160,12 @@ def topk_softmax(gating_logits, topk, renormalize=True):
     )
 
 
-def fused_moe(hidden, w13, w2, topk_weights, topk_ids, activation="silu"):
-    return get_backend(hidden.device).fused_moe(
-        hidden, w13, w2, topk_weights, topk_ids, activation
-    )
+def fused_moe(hidden, w13, w2, topk_weights, topk_ids, activation="silu",
+              w13_shuf=None, w2_shuf=None):
+    backend = get_backend(hidden.device)
+    if backend is _torch_ref:
+        return backend.fused_moe(hidden, w13, w2, topk_weights, topk_ids,
+                                 activation)
+    return backend.fused_moe(hidden, w13, w2, topk_weights, topk_ids,
+                             activation, w13_shuf=w13_shuf,
+                             w2_shuf=w2_shuf)

@@ -242,7 +242,19 @@ def _moe_hip_ok(hidden, w13, w2, activation):
     return H % 128 == 0 and twoI % 128 == 0 and (twoI // 2) % 64 == 0
 
 
-def fused_moe_hip(hidden, w13, w2, topk_weights, topk_ids):
+def moe_shuffle_weights(w):
+    """Pre-shuffle expert weights [E, N, K] into MFMA fragment-major
+    order [E][N/16][K/32][64 lanes][8 elems]: each wave's B-fragment
+    becomes one coalesced 16B/lane HBM read (csrc/moe.hip
+    moe_gemm_shuf_kernel). One-time cost at weight load."""
+    E, N, K = w.shape
+    v = w.view(E, N // 16, 16, K // 32, 4, 8)
+    v = v.permute(0, 1, 3, 4, 2, 5)
+    return v.reshape(E, (N // 16) * (K // 32) * 512).contiguous()
+
+
+def fused_moe_hip(hidden, w13, w2, topk_weights, topk_ids,
+                  w13_shuf=None, w2_shuf=None):
     """Grouped-GEMM MFMA MoE (csrc/moe.hip): on-device token->expert
     sort into BM tiles, two MFMA GEMMs streaming each expert's weights,
     deterministic weighted combine. No host sync — hipGraph-capturable
@@ -265,26 +277,37 @@ def fused_moe_hip(hidden, w13, w2, topk_weights, topk_ids):
     _C.moe_align(ids32, E, sorted_ids, expert_tiles, inv_perm, counts,
                  fill, off)
     y = torch.empty(em_max, twoI, dtype=hidden.dtype, device=dev)
-    _C.moe_gemm(hidden, w13, y, sorted_ids, expert_tiles, k, total)
+    if w13_shuf is not None:
+        _C.moe_gemm_shuf(hidden, w13_shuf, y, sorted_ids, expert_tiles,
+                         twoI, H, k, total)
+    else:
+        _C.moe_gemm(hidden, w13, y, sorted_ids, expert_tiles, k, total)
     act = silu_and_mul(y)
     y2 = torch.empty(em_max, H, dtype=hidden.dtype, device=dev)
-    _C.moe_gemm(act, w2, y2, sorted_ids, expert_tiles, 0, total)
+    if w2_shuf is not None:
+        _C.moe_gemm_shuf(act, w2_shuf, y2, sorted_ids, expert_tiles,
+                         H, I, 0, total)
+    else:
+        _C.moe_gemm(act, w2, y2, sorted_ids, expert_tiles, 0, total)
     out = torch.empty(T, H, dtype=hidden.dtype, device=dev)
     _C.moe_combine(out, y2, topk_weights.float().contiguous(), inv_perm)
     return out
 
 
-def fused_moe(hidden, w13, w2, topk_weights, topk_ids, activation="silu"):
+def fused_moe(hidden, w13, w2, topk_weights, topk_ids, activation="silu",
+              w13_shuf=None, w2_shuf=None):
     """MoE expert MLP on GPU. Primary path: the grouped-GEMM MFMA kernel
-    (fused_moe_hip, no host sync). Fallback for unsupported shapes or
-    activations: tokens sorted by expert, one tuned hipBLASLt GEMM pair
-    per non-empty expert (one D2H sync per call), weighted scatter-add
-    combine.
+    (fused_moe_hip, no host sync; fragment-major weight stream when the
+    layer provides pre-shuffled copies). Fallback for unsupported shapes
+    or activations: tokens sorted by expert, one tuned hipBLASLt GEMM
+    pair per non-empty expert (one D2H sync per call), weighted
+    scatter-add combine.
 
     hidden: [T, H]; w13: [E, 2I, H]; w2: [E, H, I].
     """
     if _moe_hip_ok(hidden, w13, w2, activation):
-        return fused_moe_hip(hidden, w13, w2, topk_weights, topk_ids)
+        return fused_moe_hip(hidden, w13, w2, topk_weights, topk_ids,
+                             w13_shuf=w13_shuf, w2_shuf=w2_shuf)
     T, H = hidden.shape
     E = w13.shape[0]
     k = topk_ids.shape[1]

@@ -518,6 +518,10 @@ class ModelRunner:
         cfg = dataclasses.replace(mc, model_path=model_path,
                                   load_format="safetensors")
         load_safetensors_weights(self.model, cfg)
+        from vllm_amd.layers.fused_moe import FusedMoE
+        for m in self.model.modules():
+            if isinstance(m, FusedMoE):
+                m.refresh_shuffled()  # in-place: hipGraphs stay valid
 
     def save_sharded_state(self, out_dir: str) -> str:
         """Write THIS rank's (TP-sharded, PP-sliced) parameters to
