@@ -32,6 +32,46 @@ from vllm_amd.core.sched_output import (
 from vllm_amd.request import Request, RequestStatus
 
 
+
+class PriorityWaiting:
+    """Heap-backed waiting queue with the deque surface schedule() uses
+    ([0], popleft, append, appendleft, remove, len, iter). Ordering is
+    Request.__lt__ = (priority, arrival). appendleft (preemption
+    requeue) is just a push: the heap restores priority order, which is
+    exactly what preempted requests need."""
+
+    def __init__(self):
+        import heapq
+        self._heapq = heapq
+        self._h: list = []
+
+    def append(self, request) -> None:
+        self._heapq.heappush(self._h, request)
+
+    appendleft = append
+
+    def popleft(self):
+        return self._heapq.heappop(self._h)
+
+    def remove(self, request) -> None:
+        self._h.remove(request)  # rare (abort); O(n) + heapify
+        self._heapq.heapify(self._h)
+
+    def __getitem__(self, idx):
+        if idx != 0:
+            raise IndexError("PriorityWaiting only exposes the head")
+        return self._h[0]
+
+    def __len__(self) -> int:
+        return len(self._h)
+
+    def __bool__(self) -> bool:
+        return bool(self._h)
+
+    def __iter__(self):
+        return iter(sorted(self._h))
+
+
 class Scheduler:
 
     def __init__(
@@ -95,7 +135,11 @@ class Scheduler:
         )
 
         self.requests: dict[str, Request] = {}
-        self.waiting: deque[Request] = deque()
+        # FCFS: plain deque. Priority: a heap with the same deque-like
+        # surface — O(log n) insert instead of re-sorting per add
+        # (reference request_queue.py priority queue role).
+        self.waiting = (PriorityWaiting() if self.policy == "priority"
+                        else deque())
         self.running: list[Request] = []
         # req_ids finished since the last schedule() call; the runner uses
         # this to clear its persistent-batch rows.
@@ -111,14 +155,7 @@ class Scheduler:
     def add_request(self, request: Request) -> None:
         self.requests[request.request_id] = request
         request.status = RequestStatus.WAITING
-        if self.policy == "priority":
-            # Maintain the waiting deque sorted by (priority, arrival).
-            items = list(self.waiting)
-            items.append(request)
-            items.sort()
-            self.waiting = deque(items)
-        else:
-            self.waiting.append(request)
+        self.waiting.append(request)
 
     def finish_requests(
         self, request_ids: list[str], status: RequestStatus
@@ -153,7 +190,7 @@ class Scheduler:
     # ------------------------------------------------------------------
     # Scheduling
 
-    def schedule(self) -> SchedulerOutput:
+    def schedule(self) -> SchedulerOutput:  # noqa: C901
         token_budget = self.max_num_batched_tokens
         scheduled_spec_tokens: dict[str, list[int]] = {}
         scheduled_new_reqs: list[NewRequestData] = []

@@ -231,15 +231,14 @@ void moe_gemm_kernel(const short* __restrict__ A, long lda,
 // Fragment-major grouped GEMM: B is pre-shuffled at weight-load time
 // into MFMA-fragment order [E][N/16][K/32][64 lanes][8 elems] so each
 // wave reads its B-fragment as ONE coalesced 16B/lane chunk straight
-// from HBM — no LDS staging, no per-kstep barrier for the (static)
-// weights. Only the small A tile stages through LDS, in ABK=256 chunks
-// (one barrier per 8 k-steps). This is the MI355X-first inference
-// layout: weights are written once and streamed billions of times, so
-// they belong in consumer order (role of the reference's Marlin-style
-// weight repacking, csrc/quantization/marlin — re-thought for bf16
-// MFMA fragments instead of int4 tiles).
-constexpr int ABK = 256;  // A-tile K chunk per barrier (8 k-steps of 32)
-
+// from HBM. A-fragments load per-lane from global (scattered but tiny —
+// <1% of traffic, L3-resident across the N/128 blocks sharing the
+// m-tile). ZERO LDS and ZERO barriers in the k-loop: nothing forces a
+// vmcnt(0) drain, so the compiler pipelines the weight stream at full
+// depth (the staged-LDS variant stalls on the barrier drain — guide
+// §Composed models). This is the MI355X-first inference layout:
+// weights are written once and streamed billions of times, so they
+// belong in consumer order.
 template <typename Tag>
 __launch_bounds__(THREADS) __global__
 void moe_gemm_shuf_kernel(const short* __restrict__ A, long lda,
@@ -253,20 +252,7 @@ void moe_gemm_shuf_kernel(const short* __restrict__ A, long lda,
   const int expert = expert_tiles[tile_m];
   if (expert < 0) return;  // padding tile
 
-  __shared__ short a_lds[BM * ABK];  // [row][k], swizzled 16B slots
-  __shared__ int rows_sh[BM];
-  __shared__ int cval_sh[BM];
-
   const int tid = threadIdx.x;
-  if (tid < BM) {
-    int sid = sorted[tile_m * BM + tid];
-    cval_sh[tid] = (sid < total_flat) ? 1 : 0;
-    rows_sh[tid] = (sid < total_flat)
-                       ? (topk_div > 0 ? sid / topk_div : tile_m * BM + tid)
-                       : -1;
-  }
-  __syncthreads();
-
   const int wave = tid / WAVE_SIZE;
   const int lane = tid % WAVE_SIZE;
   const int l16 = lane & 15;
@@ -274,57 +260,60 @@ void moe_gemm_shuf_kernel(const short* __restrict__ A, long lda,
   const int wm = wave & 1;   // m 32-half (2 m-frags)
   const int wn = wave >> 1;  // n 64-half (4 n-frags)
 
+  // Per-lane A row pointers (A-frag: lane&15 = row, k = (lane>>4)*8+j).
+  // Padding rows read row 0 (real data); their C rows are masked at the
+  // epilogue and never mix into valid rows (MFMA rows are independent).
+  const short* a_ptr[2];
+  bool cval[2][4];
+#pragma unroll
+  for (int m = 0; m < 2; ++m) {
+    const int r = tile_m * BM + wm * 32 + m * 16 + l16;
+    const int sid = sorted[r];
+    const long arow =
+        (sid < total_flat)
+            ? (topk_div > 0 ? (long)(sid / topk_div) : (long)r)
+            : 0;
+    a_ptr[m] = A + arow * lda + l4 * 8;
+#pragma unroll
+    for (int rr = 0; rr < 4; ++rr) {
+      const int crow = tile_m * BM + wm * 32 + m * 16 + l4 * 4 + rr;
+      cval[m][rr] = sorted[crow] < total_flat;
+    }
+  }
+
   f32x4 acc[2][4];
 #pragma unroll
   for (int m = 0; m < 2; ++m)
 #pragma unroll
     for (int i = 0; i < 4; ++i) acc[m][i] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  // B fragment stream base for this wave: 4 n-frags at cols
-  // n0 + wn*64 + nf*16; fragment granule = 64 lanes x 8 elems = 512.
-  const int ksteps_total = K / 32;
-  const short* bbase = B + (long)expert * expert_stride;
+  // B fragment stream for this wave: frags at n-tiles n0/16 + wn*4 + nf,
+  // granule 64 lanes x 8 elems; consecutive k-steps are contiguous 1 KB
+  // lines, so each wave walks 4 linear HBM streams. A re-reads hit the
+  // L3 (the m-tile's rows are shared by all N/128 blocks).
+  const int ksteps = K / 32;
+  const short* bptr[4];
+#pragma unroll
+  for (int nf = 0; nf < 4; ++nf)
+    bptr[nf] = B + (long)expert * expert_stride +
+               ((long)((n0 >> 4) + wn * 4 + nf) * ksteps) * 512 + lane * 8;
 
-  for (int k0 = 0; k0 < K; k0 += ABK) {
-    const int kchunk = min(ABK, K - k0);
-    // Stage A chunk: BM x kchunk; 16B per thread-chunk, swizzled.
-    for (int idx = tid; idx < BM * (kchunk / 8); idx += THREADS) {
-      const int cpr = kchunk / 8;
-      const int r = idx / cpr;
-      const int cb = (idx % cpr) * 16;
-      const int arow = rows_sh[r];
-      s16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
-      if (arow >= 0)
-        v = *reinterpret_cast<const s16x8*>(A + (long)arow * lda + k0 +
-                                            (cb >> 1));
-      *reinterpret_cast<s16x8*>(
-          reinterpret_cast<char*>(a_lds) + r * (ABK * 2) + swz(r, cb)) = v;
+  // No LDS, no barriers: the compiler software-pipelines the 6 loads +
+  // 8 MFMAs per k-step with fine-grained s_waitcnt.
+  for (int ks = 0; ks < ksteps; ++ks) {
+    s16x8 afrag[2];
+#pragma unroll
+    for (int m = 0; m < 2; ++m)
+      afrag[m] = *reinterpret_cast<const s16x8*>(a_ptr[m] + ks * 32);
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) {
+      const s16x8 bfrag =
+          *reinterpret_cast<const s16x8*>(bptr[nf] + (long)ks * 512);
+#pragma unroll
+      for (int m = 0; m < 2; ++m)
+        acc[m][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afrag[m], bfrag, acc[m][nf], 0, 0, 0);
     }
-    __syncthreads();
-#pragma unroll 2
-    for (int kk = 0; kk < kchunk / 32; ++kk) {
-      const int ks = (k0 >> 5) + kk;  // global kstep
-      const int kb = (kk * 32 + l4 * 8) * 2;
-      s16x8 afrag[2];
-#pragma unroll
-      for (int m = 0; m < 2; ++m) {
-        const int arow = wm * 32 + m * 16 + l16;
-        afrag[m] = *reinterpret_cast<const s16x8*>(
-            reinterpret_cast<const char*>(a_lds) + arow * (ABK * 2) +
-            swz(arow, kb));
-      }
-#pragma unroll
-      for (int nf = 0; nf < 4; ++nf) {
-        const int ntile = (n0 >> 4) + wn * 4 + nf;
-        s16x8 bfrag = *reinterpret_cast<const s16x8*>(
-            bbase + ((long)ntile * ksteps_total + ks) * 512 + lane * 8);
-#pragma unroll
-        for (int m = 0; m < 2; ++m)
-          acc[m][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              afrag[m], bfrag, acc[m][nf], 0, 0, 0);
-      }
-    }
-    __syncthreads();
   }
 
 #pragma unroll
@@ -334,9 +323,8 @@ void moe_gemm_shuf_kernel(const short* __restrict__ A, long lda,
       const int n = n0 + wn * 64 + nf * 16 + l16;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        const int mrow = wm * 32 + m * 16 + l4 * 4 + r;
-        if (cval_sh[mrow])
-          C[(long)(tile_m * BM + mrow) * ldc + n] =
+        if (cval[m][r])
+          C[(long)(tile_m * BM + wm * 32 + m * 16 + l4 * 4 + r) * ldc + n] =
               from_f32<Tag>(acc[m][nf][r]);
       }
     }

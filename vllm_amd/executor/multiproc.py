@@ -1,7 +1,11 @@
 """Multi-process executor: one worker process per GPU, spawned by the
 engine (role of the reference's MultiprocExecutor,
-vllm/v1/executor/multiproc_executor.py:108 — pipes instead of the shm
-MessageQueue; the data plane between workers is still RCCL over xGMI).
+vllm/v1/executor/multiproc_executor.py:108). Engine -> workers: a
+single-writer shared-memory broadcast ring (executor/shm_queue.py, role
+of shm_broadcast.py:464 MessageQueue) — each SchedulerOutput is pickled
+ONCE, not once per worker. Workers -> engine: pipes (low volume; rank 0
+ships the step result). The data plane between workers is RCCL over
+xGMI.
 
 The engine process owns the scheduler; worker rank 0 returns sampled
 tokens. Workers execute steps asynchronously (a sender thread ships each
@@ -24,7 +28,8 @@ from vllm_amd.config import EngineConfig
 logger = logging.getLogger(__name__)
 
 
-def _worker_main(rank: int, config: EngineConfig, conn, master_port: int):
+def _worker_main(rank: int, config: EngineConfig, conn, master_port: int,
+                 ring_name: str, num_readers: int, ring_size: int):
     os.environ["RANK"] = str(rank)
     os.environ["LOCAL_RANK"] = str(rank)
     os.environ["WORLD_SIZE"] = str(config.parallel_config.tensor_parallel_size)
@@ -38,9 +43,19 @@ def _worker_main(rank: int, config: EngineConfig, conn, master_port: int):
 
     from vllm_amd.worker.worker import Worker
 
+    from vllm_amd.executor.shm_queue import ShmRing
+
+    ring = ShmRing(num_readers, size=ring_size, name=ring_name,
+                   create=False)
+
     worker = Worker(config)
     worker.init_device()
     worker.load_model()
+
+    def orphan_check():
+        # Engine death leaves readers spinning: exit when reparented.
+        if os.getppid() == 1:
+            raise EOFError("engine process died")
 
     send_lock = threading.Lock()
 
@@ -71,7 +86,7 @@ def _worker_main(rank: int, config: EngineConfig, conn, master_port: int):
 
     try:
         while True:
-            msg = conn.recv()
+            msg = ring.read(rank, poll=orphan_check)
             kind = msg[0]
             if kind == "rpc":
                 _, method, args, kwargs = msg
@@ -104,6 +119,8 @@ def _worker_main(rank: int, config: EngineConfig, conn, master_port: int):
                 return
     except (EOFError, KeyboardInterrupt):
         pass
+    finally:
+        ring.close()
 
 
 class _Future:
@@ -141,12 +158,18 @@ class MultiprocExecutor:
         ctx = mp.get_context("spawn")
         port = (config.parallel_config.worker_port
                 or int(os.environ.get("VLLM_AMD_WORKER_PORT", "29533")))
+        from vllm_amd.executor.shm_queue import ShmRing
+
+        ring_size = 8 << 20
+        self.ring = ShmRing(tp, size=ring_size)
         self.conns = []
         self.procs = []
         for rank in range(tp):
             parent, child = ctx.Pipe()
             p = ctx.Process(
-                target=_worker_main, args=(rank, config, child, port),
+                target=_worker_main,
+                args=(rank, config, child, port, self.ring.name, tp,
+                      ring_size),
                 daemon=True,
             )
             p.start()
@@ -169,12 +192,13 @@ class MultiprocExecutor:
         return conn.recv()
 
     def _send_all(self, msg) -> None:
-        for conn in self.conns:
-            try:
-                conn.send(msg)
-            except (BrokenPipeError, OSError) as e:
-                self.check_health()
-                raise EngineDeadError(f"worker pipe closed: {e}") from e
+        try:
+            self.ring.write(msg, health_check=self.check_health)
+        except EngineDeadError:
+            raise
+        except (ValueError, OSError) as e:
+            self.check_health()
+            raise EngineDeadError(f"shm ring write failed: {e}") from e
 
     # ---- control-plane RPC -------------------------------------------
     def collective_rpc(self, method: str, *args: Any, **kwargs: Any) -> list:
@@ -238,12 +262,12 @@ class MultiprocExecutor:
         return _Future(self.conns[0], self)
 
     def shutdown(self) -> None:
-        for conn in self.conns:
-            try:
-                conn.send(("shutdown",))
-            except (BrokenPipeError, OSError):
-                pass
+        try:
+            self.ring.write(("shutdown",))
+        except Exception:  # noqa: BLE001
+            pass
         for p in self.procs:
             p.join(timeout=10)
             if p.is_alive():
                 p.terminate()
+        self.ring.close()

@@ -102,8 +102,12 @@ class FusedMoE(nn.Module):
     def _maybe_shuffled(self, hidden):
         import os
 
+        # Fragment-major streaming is opt-in (VLLM_AMD_MOE_SHUF=1): on
+        # MI355X the staged-LDS grouped GEMM measured faster at decode
+        # batch >= 256 (the barrier-free stream wins only at tiny
+        # batches; see profiles/r02_summary.md).
         if (not hidden.is_cuda
-                or os.environ.get("VLLM_AMD_MOE_SHUF", "1") == "0"):
+                or os.environ.get("VLLM_AMD_MOE_SHUF", "0") != "1"):
             return None, None
         if self._w13_shuf is None:
             from vllm_amd.ops import get_backend
