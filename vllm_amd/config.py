@@ -774,6 +774,10 @@ class ObservabilityConfig:
     # torch.profiler output dir for /start_profile `/stop_profile`
     # (kineto -> chrome trace; roctracer GPU events on ROCm).
     profile_dir: str = "profile_out"
+    # KV cache event stream endpoint "host:port" (reference
+    # kv_events.py role): block_stored/block_removed/all_blocks_cleared
+    # as JSONL over TCP for cache-aware routers.
+    kv_events_endpoint: Optional[str] = None
 
 
 @dataclass

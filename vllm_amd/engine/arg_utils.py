@@ -59,6 +59,7 @@ class EngineArgs:
     device: str = "auto"
     trace_file: Optional[str] = None
     otlp_traces_endpoint: Optional[str] = None
+    kv_events_endpoint: Optional[str] = None
 
     @staticmethod
     def add_cli_args(parser: argparse.ArgumentParser) -> argparse.ArgumentParser:
@@ -121,6 +122,10 @@ class EngineArgs:
         parser.add_argument("--device", type=str, default="auto")
         parser.add_argument("--trace-file", type=str, default=None,
                             help="JSONL request-trace output path")
+        parser.add_argument("--kv-events-endpoint", type=str,
+                            default=None,
+                            help="host:port to stream KV-cache block "
+                                 "events (JSONL over TCP)")
         parser.add_argument("--otlp-traces-endpoint", type=str,
                             default=None,
                             help="OTLP/HTTP collector URL; one OTEL "
@@ -194,5 +199,6 @@ class EngineArgs:
             device_config=DeviceConfig(device=self.device),
             observability_config=ObservabilityConfig(
                 trace_file=self.trace_file,
-                otlp_traces_endpoint=self.otlp_traces_endpoint),
+                otlp_traces_endpoint=self.otlp_traces_endpoint,
+                kv_events_endpoint=self.kv_events_endpoint),
         )

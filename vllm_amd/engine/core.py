@@ -98,6 +98,14 @@ class EngineCore:
         self.async_scheduling = config.scheduler_config.async_scheduling
         # One in-flight (SchedulerOutput, AsyncModelOutput) when pipelining.
         self._pending = None
+        # KV-cache event stream for cache-aware routers (kv_events.py).
+        self.kv_events = None
+        ep = getattr(config.observability_config, "kv_events_endpoint", None)
+        if ep and self.scheduler is not None:
+            from vllm_amd.kv_events import KVEventPublisher
+            self.kv_events = KVEventPublisher(ep)
+            self.kv_events.attach(
+                self.scheduler.kv_cache_manager.block_pool)
 
     # ------------------------------------------------------------------
     def add_request(self, request: Request) -> None:
@@ -218,6 +226,8 @@ class EngineCore:
             pass
 
     def shutdown(self) -> None:
+        if self.kv_events is not None:
+            self.kv_events.close()
         if self.is_driver and self._pending is not None:
             try:
                 self._drain()

@@ -59,8 +59,15 @@ class TokenizerWrapper:
 
 
 class IncrementalDetokenizer:
-    """Streams text from a growing token list without re-decoding from
-    scratch; holds back bytes that may still change (utf-8 continuation)."""
+    """Streams text from a growing token list.
+
+    Fast path (role of the reference's FastIncrementalDetokenizer,
+    vllm/v1/engine/detokenizer.py:168): HF fast tokenizers expose the
+    Rust `DecodeStream`, which emits each token's stabilized text in
+    O(1) — the fallback below re-decodes the whole list per step, which
+    is O(n^2) over a long generation and shows up in the API-process CPU
+    budget. The fallback remains for the byte-level mock tokenizer and
+    slow (non-Rust) tokenizers."""
 
     def __init__(self, tokenizer: TokenizerWrapper, prompt_len: int,
                  skip_special_tokens: bool = True):
@@ -70,10 +77,29 @@ class IncrementalDetokenizer:
         self.output_text = ""
         # Index of the first token not yet surely decoded.
         self._stable_len = 0
+        self._stream = None
+        self._raw = getattr(tokenizer.tokenizer, "_tokenizer", None)
+        if self._raw is not None:
+            try:
+                from tokenizers.decoders import DecodeStream
+
+                self._stream = DecodeStream(
+                    skip_special_tokens=skip_special_tokens)
+            except Exception:  # noqa: BLE001
+                self._stream = None
 
     def update(self, new_token_ids: list[int]) -> str:
         """Append tokens, return newly stabilized text delta."""
         self.token_ids.extend(new_token_ids)
+        if self._stream is not None:
+            parts = []
+            for t in new_token_ids:
+                piece = self._stream.step(self._raw, int(t))
+                if piece:
+                    parts.append(piece)
+            delta = "".join(parts)
+            self.output_text += delta
+            return delta
         full = self.tokenizer.decode(
             self.token_ids, skip_special_tokens=self.skip_special_tokens
         )

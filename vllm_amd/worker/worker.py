@@ -10,6 +10,8 @@ from __future__ import annotations
 import gc
 import logging
 
+import os
+
 import torch
 
 from vllm_amd.config import EngineConfig
@@ -33,6 +35,13 @@ class Worker:
         if self.config.device_config.device == "cuda":
             torch.cuda.set_device(pc.local_rank)
             self.device = torch.device("cuda", pc.local_rank)
+            # Debug sync-check (role of the reference's
+            # with_gpu_sync_check, gpu_worker.py:1023): surface
+            # accidental host synchronizations in the async decode path.
+            # VLLM_AMD_SYNC_CHECK=warn|error.
+            mode = os.environ.get("VLLM_AMD_SYNC_CHECK")
+            if mode in ("warn", "error"):
+                torch.cuda.set_sync_debug_mode(mode)
         else:
             self.device = torch.device("cpu")
         if pc.needs_distributed:
