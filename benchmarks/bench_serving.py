@@ -36,6 +36,12 @@ def parse_args():
     p.add_argument("--max-num-seqs", type=int, default=256)
     p.add_argument("--num-gpu-blocks", type=int, default=None)
     p.add_argument("--block-size", type=int, default=64)
+    p.add_argument("--server-mode", choices=("subprocess", "inproc"),
+                   default="subprocess",
+                   help="subprocess: real uvicorn server + TCP sockets "
+                        "(the reference methodology — client cost does "
+                        "not steal the server's event loop); inproc: "
+                        "ASGI transport in one process")
     return p.parse_args()
 
 
@@ -63,13 +69,56 @@ async def run(args) -> dict:
         max_num_batched_tokens=max(32768, args.input_len * 8),
         max_num_seqs=args.max_num_seqs,
     )
-    app, state = make_server(engine_args, served_model_name=args.model)
     rng = np.random.default_rng(0)
-    vocab = state.engine.config.model_config.spec.vocab_size
+    proc = None
+    state = None
+    if args.server_mode == "subprocess":
+        import socket as _socket
+        import subprocess
+        import urllib.request
 
-    transport = httpx.ASGITransport(app=app)
-    client = httpx.AsyncClient(transport=transport,
-                               base_url="http://bench", timeout=600.0)
+        from vllm_amd.config import MODEL_PRESETS
+
+        vocab = MODEL_PRESETS[args.model].vocab_size
+        srv = _socket.socket()
+        srv.bind(("127.0.0.1", 0))
+        port = srv.getsockname()[1]
+        srv.close()
+        cmd = [sys.executable, "-m", "vllm_amd", "serve", args.model,
+               "--dtype", args.dtype, "--device",
+               "cuda" if on_gpu else "cpu",
+               "--block-size", str(args.block_size),
+               "--max-model-len", str(args.input_len + args.output_len + 32),
+               "--max-num-batched-tokens",
+               str(max(32768, args.input_len * 8)),
+               "--max-num-seqs", str(args.max_num_seqs),
+               "--host", "127.0.0.1", "--port", str(port)]
+        if args.num_gpu_blocks:
+            cmd += ["--num-gpu-blocks", str(args.num_gpu_blocks)]
+        proc = subprocess.Popen(cmd, stdout=subprocess.DEVNULL,
+                                stderr=subprocess.DEVNULL)
+        deadline = time.time() + 420
+        while time.time() < deadline:
+            try:
+                with urllib.request.urlopen(
+                        f"http://127.0.0.1:{port}/health", timeout=2) as r:
+                    if r.status == 200:
+                        break
+            except Exception:
+                if proc.poll() is not None:
+                    raise RuntimeError("server process died during boot")
+                await asyncio.sleep(0.5)
+        else:
+            proc.kill()
+            raise RuntimeError("server did not come up in 420s")
+        client = httpx.AsyncClient(base_url=f"http://127.0.0.1:{port}",
+                                   timeout=600.0)
+    else:
+        app, state = make_server(engine_args, served_model_name=args.model)
+        vocab = state.engine.config.model_config.spec.vocab_size
+        transport = httpx.ASGITransport(app=app)
+        client = httpx.AsyncClient(transport=transport,
+                                   base_url="http://bench", timeout=600.0)
 
     results = []
 
@@ -113,7 +162,14 @@ async def run(args) -> dict:
     await asyncio.gather(*tasks)
     elapsed = time.perf_counter() - t_start
     await client.aclose()
-    state.engine.shutdown()
+    if state is not None:
+        state.engine.shutdown()
+    if proc is not None:
+        proc.terminate()
+        try:
+            proc.wait(timeout=15)
+        except Exception:  # noqa: BLE001
+            proc.kill()
 
     ttfts = np.array([r["ttft"] for r in results])
     e2es = np.array([r["e2e"] for r in results])

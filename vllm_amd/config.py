@@ -641,6 +641,9 @@ class ModelConfig:
     # name -> local PEFT checkpoint dir; ids are 1-based in listed order.
     lora_modules: Optional[dict] = None
     enable_expert_parallel: bool = False
+    # EPLB: rebalance expert->rank placement every N MoE forwards
+    # (0 = off). Mirrored from ParallelConfig like the EP flag.
+    eplb_window: int = 0
     # "fp8": W8A8 e4m3 — per-channel weight scales (quantized after
     # load), per-token dynamic activation scales, fp8 MFMA GEMMs. Dense
     # linears only; lm_head and MoE expert weights stay in model dtype.
@@ -725,6 +728,10 @@ class ParallelConfig:
     # Shard MoE experts across the world (EP) instead of TP-sharding each
     # expert's intermediate dim; tokens stay replicated (AgRs combine).
     enable_expert_parallel: bool = False
+    # EPLB window (expert load balancing): every N MoE-layer forwards,
+    # re-pack experts onto ranks by EWMA token load (0 = off). Role of
+    # the reference's vllm/distributed/eplb (eplb_state.py:220).
+    eplb_window: int = 0
     # Run the engine core (scheduler + executor) in its own process; the
     # API process only tokenizes/detokenizes (reference EngineCoreProc).
     multiprocess_engine: bool = False
@@ -800,3 +807,4 @@ class EngineConfig:
         # Models read the EP flag from ModelConfig at construction.
         self.model_config.enable_expert_parallel = \
             self.parallel_config.enable_expert_parallel
+        self.model_config.eplb_window = self.parallel_config.eplb_window

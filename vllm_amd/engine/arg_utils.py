@@ -55,6 +55,7 @@ class EngineArgs:
     pipeline_parallel_size: int = 1
     data_parallel_size: int = 1
     enable_expert_parallel: bool = False
+    eplb_window: int = 0
     multiprocess_engine: bool = False
     device: str = "auto"
     trace_file: Optional[str] = None
@@ -117,6 +118,9 @@ class EngineArgs:
                             help="serve-level engine replicas "
                                  "(least-loaded request routing)")
         parser.add_argument("--enable-expert-parallel", action="store_true")
+        parser.add_argument("--eplb-window", type=int, default=0,
+                            help="rebalance expert placement every N MoE "
+                                 "forwards (0=off)")
         parser.add_argument("--multiprocess-engine", action="store_true",
                             help="run the engine core in its own process")
         parser.add_argument("--device", type=str, default="auto")
@@ -153,6 +157,7 @@ class EngineArgs:
             pipeline_parallel_size=self.pipeline_parallel_size,
             data_parallel_size=self.data_parallel_size,
             enable_expert_parallel=self.enable_expert_parallel,
+            eplb_window=self.eplb_window,
             multiprocess_engine=self.multiprocess_engine,
             rank=int(os.environ.get("RANK", "0")),
             local_rank=int(os.environ.get("LOCAL_RANK", "0")),

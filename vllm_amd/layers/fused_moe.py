@@ -38,6 +38,7 @@ class FusedMoE(nn.Module):
         activation: str = "silu",
         dtype: torch.dtype = None,
         enable_expert_parallel: bool = False,
+        eplb_window: int = 0,
     ):
         super().__init__()
         from vllm_amd.config import EngineConfig  # noqa: F401 (doc only)
@@ -80,6 +81,32 @@ class FusedMoE(nn.Module):
                         dtype=dtype),
             requires_grad=False,
         )
+        # EPLB (role of the reference's vllm/distributed/eplb,
+        # eplb_state.py:220): dynamic expert->rank placement. owner_of /
+        # slot_of are device lookup tensors consulted by the EP routing
+        # mask; rebalance() re-packs experts by EWMA token load and
+        # moves weights IN PLACE (hipGraphs stay valid). All ranks see
+        # identical router outputs (replicated activations), so every
+        # rank computes the same plan with no control-plane sync — only
+        # the weight broadcasts communicate.
+        self.eplb_window = eplb_window if self.ep_size > 1 else 0
+        if self.ep_size > 1:
+            n = self.num_local_experts
+            self.assignment = [e // n for e in range(num_experts)]
+            self.local_slots = list(range(self.expert_lo,
+                                          self.expert_lo + n))
+            self.register_buffer(
+                "owner_of",
+                torch.tensor(self.assignment, dtype=torch.int64),
+                persistent=False)
+            self.register_buffer(
+                "slot_of",
+                torch.tensor([e % n for e in range(num_experts)],
+                             dtype=torch.int64),
+                persistent=False)
+            self._eplb_load = torch.zeros(num_experts, dtype=torch.float64)
+            self._eplb_ewma = None
+            self._eplb_calls = 0
         # Fragment-major weight copies for the HIP grouped GEMM (built
         # lazily on the first GPU forward; freed by invalidate_shuffled
         # when weights change). Costs a second copy of the expert
@@ -87,6 +114,96 @@ class FusedMoE(nn.Module):
         # barrier-free weight streams. VLLM_AMD_MOE_SHUF=0 disables.
         self._w13_shuf = None
         self._w2_shuf = None
+
+    # -- EPLB ----------------------------------------------------------
+    def _eplb_observe(self, topk_ids) -> None:
+        counts = torch.bincount(topk_ids.reshape(-1),
+                                minlength=self.num_experts)
+        self._eplb_load += counts.double().cpu()
+        self._eplb_calls += 1
+        if self._eplb_calls % self.eplb_window == 0:
+            self.rebalance()
+
+    def _plan_assignment(self, load) -> list[int]:
+        """Greedy balanced packing: experts by load (desc), each to the
+        least-loaded rank with free slots. Deterministic (ties broken by
+        expert id / rank id) so every rank computes the same plan."""
+        w = self.ep_size
+        cap = self.num_local_experts
+        order = sorted(range(self.num_experts),
+                       key=lambda e: (-load[e], e))
+        rank_load = [0.0] * w
+        rank_cnt = [0] * w
+        plan = [0] * self.num_experts
+        for e in order:
+            r = min((r for r in range(w) if rank_cnt[r] < cap),
+                    key=lambda r: (rank_load[r], r))
+            plan[e] = r
+            rank_load[r] += load[e]
+            rank_cnt[r] += 1
+        return plan
+
+    @torch.no_grad()
+    def rebalance(self) -> None:
+        """Re-pack experts onto ranks by EWMA load and move weights.
+        Collective: every rank must call this at the same step (SPMD
+        lockstep guarantees it when triggered by the shared counter)."""
+        import torch.distributed as dist
+
+        from vllm_amd.parallel.state import get_ep_group
+
+        ep = get_ep_group()
+        cur = self._eplb_load.numpy()
+        if self._eplb_ewma is None:
+            self._eplb_ewma = cur.copy()
+        else:
+            self._eplb_ewma = 0.5 * self._eplb_ewma + 0.5 * cur
+        self._eplb_load.zero_()
+        plan = self._plan_assignment(self._eplb_ewma.tolist())
+        if plan == self.assignment:
+            return
+        # Per-rank slot order: owned experts ascending by global id.
+        new_slots = [e for e in range(self.num_experts)
+                     if plan[e] == self.ep_rank]
+        slot_index = {}
+        for r in range(self.ep_size):
+            owned = [e for e in range(self.num_experts) if plan[e] == r]
+            for s, e in enumerate(owned):
+                slot_index[e] = s
+        # Stash current local weights (sources for broadcasts AND for
+        # intra-rank slot moves).
+        stash13 = self.w13.data.clone()
+        stash2 = self.w2.data.clone()
+        old_slot = {e: s for s, e in enumerate(self.local_slots)}
+        buf13 = torch.empty_like(self.w13.data[0])
+        buf2 = torch.empty_like(self.w2.data[0])
+        for e in range(self.num_experts):
+            src = self.assignment[e]
+            dst = plan[e]
+            if src == dst and e in old_slot and slot_index[e] == old_slot[e]:
+                continue  # unchanged owner and slot
+            if src == dst:
+                # intra-rank slot move only
+                if dst == self.ep_rank:
+                    self.w13.data[slot_index[e]].copy_(stash13[old_slot[e]])
+                    self.w2.data[slot_index[e]].copy_(stash2[old_slot[e]])
+                continue
+            if self.ep_rank == src:
+                buf13.copy_(stash13[old_slot[e]])
+                buf2.copy_(stash2[old_slot[e]])
+            src_global = (ep.group_ranks[src]
+                          if hasattr(ep, "group_ranks") else src)
+            dist.broadcast(buf13, src=src_global, group=ep.device_group)
+            dist.broadcast(buf2, src=src_global, group=ep.device_group)
+            if self.ep_rank == dst:
+                self.w13.data[slot_index[e]].copy_(buf13)
+                self.w2.data[slot_index[e]].copy_(buf2)
+        self.assignment = plan
+        self.local_slots = new_slots
+        self.owner_of.copy_(torch.tensor(plan, dtype=torch.int64))
+        slot_list = [slot_index[e] for e in range(self.num_experts)]
+        self.slot_of.copy_(torch.tensor(slot_list, dtype=torch.int64))
+        self.refresh_shuffled()
 
     def refresh_shuffled(self) -> None:
         """Call after in-place weight updates (RL update_weights):
@@ -139,13 +256,15 @@ class FusedMoE(nn.Module):
         once activations are DP/sequence-sharded — a round-2 item
         together with DP attention for MoE.)"""
         if self.ep_size > 1:
-            lo = self.expert_lo
-            hi = lo + self.num_local_experts
-            mask = (topk_ids >= lo) & (topk_ids < hi)
+            if self.eplb_window > 0:
+                self._eplb_observe(topk_ids)
+            owner = self.owner_of.to(topk_ids.device)
+            slot_map = self.slot_of.to(topk_ids.device)
+            mask = owner[topk_ids] == self.ep_rank
             slot_tok, slot_k = mask.nonzero(as_tuple=True)
             if slot_tok.numel() == 0:
                 return torch.zeros_like(hidden)
-            sel_ids = (topk_ids[slot_tok, slot_k] - lo).unsqueeze(1)
+            sel_ids = slot_map[topk_ids[slot_tok, slot_k]].unsqueeze(1)
             sel_w = topk_weights[slot_tok, slot_k].unsqueeze(1)
             w13s, w2s = self._maybe_shuffled(hidden)
             y = ops.fused_moe(

@@ -194,6 +194,7 @@ class DeepseekMoE(nn.Module):
             renormalize=spec.norm_topk_prob,
             dtype=dtype,
             enable_expert_parallel=ep,
+            eplb_window=eplb_window,
         )
         # DeepSeek-V3 aux-loss-free balancing bias (inference: applied to
         # selection only).
@@ -235,7 +236,7 @@ class DeepseekMoE(nn.Module):
 
 class DeepseekDecoderLayer(nn.Module):
     def __init__(self, spec: ModelSpec, layer_idx: int, dtype, max_position,
-                 ep: bool = False):
+                 ep: bool = False, eplb_window: int = 0):
         super().__init__()
         self.self_attn = DeepseekMLAAttention(spec, layer_idx, dtype,
                                               max_position)
@@ -272,7 +273,8 @@ class DeepseekModel(nn.Module):
         )
         self.layers = nn.ModuleList([
             DeepseekDecoderLayer(spec, i, dtype, config.max_model_len,
-                                 ep=config.enable_expert_parallel)
+                                 ep=config.enable_expert_parallel,
+                                eplb_window=config.eplb_window)
             for i in range(spec.num_layers)
         ])
         self.norm = RMSNorm(spec.hidden_size, spec.rms_norm_eps, dtype=dtype)

@@ -66,7 +66,7 @@ class MixtralAttention(nn.Module):
 
 class MixtralDecoderLayer(nn.Module):
     def __init__(self, spec: ModelSpec, layer_idx: int, dtype, max_position,
-                 ep: bool = False):
+                 ep: bool = False, eplb_window: int = 0):
         super().__init__()
         self.self_attn = MixtralAttention(spec, layer_idx, dtype,
                                           max_position)
@@ -78,6 +78,7 @@ class MixtralDecoderLayer(nn.Module):
             renormalize=spec.norm_topk_prob,
             dtype=dtype,
             enable_expert_parallel=ep,
+            eplb_window=eplb_window,
         )
         self.input_layernorm = RMSNorm(spec.hidden_size, spec.rms_norm_eps,
                                        dtype=dtype)
@@ -107,7 +108,8 @@ class MixtralModel(nn.Module):
         )
         self.layers = nn.ModuleList([
             MixtralDecoderLayer(spec, i, dtype, config.max_model_len,
-                                ep=config.enable_expert_parallel)
+                                ep=config.enable_expert_parallel,
+                                eplb_window=config.eplb_window)
             for i in range(spec.num_layers)
         ])
         self.norm = RMSNorm(spec.hidden_size, spec.rms_norm_eps, dtype=dtype)
