@@ -682,3 +682,26 @@ def test_partial_rotary_factor():
         assert not torch.equal(q[..., :rd], q0[..., :rd])
     finally:
         MODEL_PRESETS.pop("tiny-partial-rope", None)
+
+
+def test_new_model_family_presets_generate():
+    """Round-2 family presets (GLM-style partial-rotary+bias via the
+    tiny spec; registry entries for llama-3.1/glm-4/internlm2/yi/nemo)
+    instantiate and decode deterministically."""
+    from vllm_amd.config import MODEL_PRESETS
+    from vllm_amd.entrypoints.llm import LLM
+    from vllm_amd.sampling_params import SamplingParams
+
+    for name in ("llama-3.1-8b", "glm-4-9b", "internlm2-7b", "yi-6b",
+                 "mistral-nemo-12b"):
+        spec = MODEL_PRESETS[name]
+        assert spec.num_layers > 0 and spec.head_dim in (64, 128)
+
+    llm = LLM(model="tiny-glm", dtype="fp32", device="cpu",
+              block_size=16, num_gpu_blocks=64, max_model_len=128,
+              max_num_batched_tokens=128, max_num_seqs=2)
+    p = SamplingParams(temperature=0.0, max_tokens=8, ignore_eos=True)
+    a = llm.generate([[5, 6, 7, 8]], p)[0].outputs[0].token_ids
+    b = llm.generate([[5, 6, 7, 8]], p)[0].outputs[0].token_ids
+    llm.shutdown()
+    assert len(a) == 8 and a == b

@@ -300,3 +300,55 @@ def test_streaming_llama3_json_parser():
     c, _ = p2.feed("hello world")
     c2, k = p2.flush()
     assert c + c2 == "hello world" and not k
+
+
+def test_pythonic_parser_complete():
+    from vllm_amd.entrypoints.tool_parser import parse_tool_calls
+
+    content, calls = parse_tool_calls(
+        "pythonic", '[get_weather(city="SF", unit="c"), t2(x=3, ok=True)]')
+    assert content == ""
+    assert [c.name for c in calls] == ["get_weather", "t2"]
+    import json
+    assert json.loads(calls[0].arguments) == {"city": "SF", "unit": "c"}
+    assert json.loads(calls[1].arguments) == {"x": 3, "ok": True}
+    # Non-call text passes through.
+    content, calls = parse_tool_calls("pythonic", "just text [1, 2]")
+    assert calls == [] and "just text" in content
+
+
+def test_granite_parser_complete_and_streaming():
+    import json
+
+    from vllm_amd.entrypoints.tool_parser import (
+        make_streaming_tool_parser, parse_tool_calls)
+
+    msg = ('I will call a tool. <|tool_call|>'
+           '[{"name": "lookup", "arguments": {"q": "x"}}]')
+    content, calls = parse_tool_calls("granite", msg)
+    assert content == "I will call a tool."
+    assert calls[0].name == "lookup"
+    assert json.loads(calls[0].arguments) == {"q": "x"}
+
+    sp = make_streaming_tool_parser("granite")
+    streamed = ""
+    for i in range(0, len(msg), 7):
+        c, _ = sp.feed(msg[i:i + 7])
+        streamed += c
+    c, out = sp.flush()
+    streamed += c
+    assert streamed.strip() == "I will call a tool."
+    assert out[0]["function"]["name"] == "lookup"
+
+
+def test_internlm2_parser():
+    import json
+
+    from vllm_amd.entrypoints.tool_parser import parse_tool_calls
+
+    msg = ('Let me check.<|action_start|><|plugin|>'
+           '{"name": "calc", "parameters": {"a": 1}}<|action_end|>')
+    content, calls = parse_tool_calls("internlm2", msg)
+    assert content == "Let me check."
+    assert calls[0].name == "calc"
+    assert json.loads(calls[0].arguments) == {"a": 1}

@@ -224,6 +224,111 @@ MODEL_PRESETS: dict[str, ModelSpec] = {
         v_head_dim=128,
         eos_token_id=1,
     ),
+    # Llama-3.1: llama3 rope scaling over the 3.0 geometry (the
+    # long-context member of the family; reference rope_scaling
+    # rope_type="llama3").
+    "llama-3.1-8b": ModelSpec(
+        name="llama-3.1-8b",
+        architecture="llama",
+        vocab_size=128256,
+        hidden_size=4096,
+        intermediate_size=14336,
+        num_layers=32,
+        num_heads=32,
+        num_kv_heads=8,
+        head_dim=128,
+        rope_theta=500000.0,
+        rope_scaling={"rope_type": "llama3", "factor": 8.0,
+                      "low_freq_factor": 1.0, "high_freq_factor": 4.0,
+                      "original_max_position_embeddings": 8192},
+        max_position_embeddings=131072,
+        eos_token_id=128001,
+        bos_token_id=128000,
+    ),
+    # GLM-4: llama-structured with half-head rotary + qkv bias
+    # (reference models/glm4.py geometry).
+    "glm-4-9b": ModelSpec(
+        name="glm-4-9b",
+        architecture="llama",
+        vocab_size=151552,
+        hidden_size=4096,
+        intermediate_size=13696,
+        num_layers=40,
+        num_heads=32,
+        num_kv_heads=2,
+        head_dim=128,
+        partial_rotary_factor=0.5,
+        qkv_bias=True,
+        rope_theta=10000.0,
+        max_position_embeddings=8192,
+        eos_token_id=151329,
+    ),
+    # InternLM2: llama geometry, 1e6 theta, 92k vocab
+    # (reference models/internlm2.py).
+    "internlm2-7b": ModelSpec(
+        name="internlm2-7b",
+        architecture="llama",
+        vocab_size=92544,
+        hidden_size=4096,
+        intermediate_size=14336,
+        num_layers=32,
+        num_heads=32,
+        num_kv_heads=8,
+        head_dim=128,
+        rope_theta=1000000.0,
+        max_position_embeddings=32768,
+        use_bias=False,
+        eos_token_id=2,
+    ),
+    # Yi: llama geometry with 4 kv heads / 64k vocab
+    # (reference models/llama.py covers Yi checkpoints).
+    "yi-6b": ModelSpec(
+        name="yi-6b",
+        architecture="llama",
+        vocab_size=64000,
+        hidden_size=4096,
+        intermediate_size=11008,
+        num_layers=32,
+        num_heads=32,
+        num_kv_heads=4,
+        head_dim=128,
+        rope_theta=5000000.0,
+        max_position_embeddings=4096,
+        eos_token_id=2,
+    ),
+    # Mistral-NeMo 12B: 5120 hidden with 128 head_dim (head_dim !=
+    # hidden/heads — exercises the explicit head_dim path).
+    "mistral-nemo-12b": ModelSpec(
+        name="mistral-nemo-12b",
+        architecture="llama",
+        vocab_size=131072,
+        hidden_size=5120,
+        intermediate_size=14336,
+        num_layers=40,
+        num_heads=32,
+        num_kv_heads=8,
+        head_dim=128,
+        rope_theta=1000000.0,
+        max_position_embeddings=32768,
+        eos_token_id=2,
+    ),
+    # Tiny GLM-style spec for CPU tests: partial rotary + qkv bias.
+    "tiny-glm": ModelSpec(
+        name="tiny-glm",
+        architecture="llama",
+        vocab_size=1024,
+        hidden_size=64,
+        intermediate_size=128,
+        num_layers=2,
+        num_heads=4,
+        num_kv_heads=2,
+        head_dim=16,
+        partial_rotary_factor=0.5,
+        qkv_bias=True,
+        rope_theta=10000.0,
+        max_position_embeddings=2048,
+        eos_token_id=2,
+    ),
     "mistral-7b": ModelSpec(
         name="mistral-7b",
         architecture="llama",

@@ -353,6 +353,89 @@ class StreamingBufferedToolParser:
         return content, out
 
 
+
+
+# Pythonic format (Llama-4 / reference pythonic_tool_parser.py): the
+# message is a Python-literal list of calls: [get_weather(city="SF"),
+# f(x=3)]. Parsed with ast — no eval.
+def parse_pythonic_tool_calls(text: str) -> tuple[str, list[ParsedToolCall]]:
+    import ast
+
+    stripped = text.strip()
+    if not (stripped.startswith("[") and stripped.endswith("]")):
+        return stripped, []
+    try:
+        tree = ast.parse(stripped, mode="eval")
+    except SyntaxError:
+        return stripped, []
+    if not isinstance(tree.body, ast.List):
+        return stripped, []
+    calls = []
+    for node in tree.body.elts:
+        if not (isinstance(node, ast.Call)
+                and isinstance(node.func, ast.Name)):
+            return stripped, []
+        args = {}
+        for kw in node.keywords:
+            try:
+                args[kw.arg] = ast.literal_eval(kw.value)
+            except (ValueError, SyntaxError):
+                return stripped, []
+        calls.append(ParsedToolCall(id=_call_id(), name=node.func.id,
+                                    arguments=json.dumps(args)))
+    return "", calls
+
+
+# Granite-3 format (reference granite_tool_parser.py): the message
+# starts with "<|tool_call|>" followed by a JSON array of
+# {"name": ..., "arguments": {...}}.
+GRANITE_MARKER = "<|tool_call|>"
+
+
+def parse_granite_tool_calls(text: str) -> tuple[str, list[ParsedToolCall]]:
+    idx = text.find(GRANITE_MARKER)
+    if idx < 0:
+        return text.strip(), []
+    content = text[:idx].strip()
+    body = text[idx + len(GRANITE_MARKER):].strip()
+    try:
+        arr = json.loads(body)
+    except (json.JSONDecodeError, ValueError):
+        return text.strip(), []
+    calls = []
+    for obj in arr if isinstance(arr, list) else [arr]:
+        if isinstance(obj, dict) and "name" in obj:
+            args = obj.get("arguments", obj.get("parameters", {}))
+            if not isinstance(args, str):
+                args = json.dumps(args)
+            calls.append(ParsedToolCall(id=_call_id(),
+                                        name=str(obj["name"]),
+                                        arguments=args))
+    return content, calls
+
+
+# InternLM2 format (reference internlm2_tool_parser.py):
+# content<|action_start|><|plugin|>{json}<|action_end|>
+IL2_START = "<|action_start|><|plugin|>"
+IL2_END = "<|action_end|>"
+
+
+def parse_internlm2_tool_calls(
+        text: str) -> tuple[str, list[ParsedToolCall]]:
+    idx = text.find(IL2_START)
+    if idx < 0:
+        return text.strip(), []
+    content = text[:idx].strip()
+    body = text[idx + len(IL2_START):]
+    end = body.find(IL2_END)
+    if end >= 0:
+        body = body[:end]
+    call = _parse_call_body(body.strip())
+    if call is None:
+        return text.strip(), []
+    return content, [call]
+
+
 def make_streaming_tool_parser(fmt: str):
     if fmt == "mistral":
         return StreamingBufferedToolParser(
@@ -367,6 +450,27 @@ def make_streaming_tool_parser(fmt: str):
             parse=parse_llama3_json_tool_calls, trigger=trig)
         p._hold_len = lambda text: 0
         return p
+    if fmt == "pythonic":
+        def trig(s):
+            st = s.lstrip()
+            return (len(s) - len(st)) if st.startswith("[") else -1
+
+        p = StreamingBufferedToolParser(
+            parse=parse_pythonic_tool_calls, trigger=trig)
+        p._hold_len = lambda text: 0
+        return p
+    if fmt == "granite":
+        p = StreamingBufferedToolParser(
+            parse=parse_granite_tool_calls,
+            trigger=lambda s: s.find(GRANITE_MARKER))
+        p._hold_len = lambda text: _longest_tag_prefix(text, GRANITE_MARKER)
+        return p
+    if fmt == "internlm2":
+        p = StreamingBufferedToolParser(
+            parse=parse_internlm2_tool_calls,
+            trigger=lambda s: s.find(IL2_START))
+        p._hold_len = lambda text: _longest_tag_prefix(text, IL2_START)
+        return p
     return StreamingToolParser()
 
 
@@ -375,4 +479,12 @@ def parse_tool_calls(fmt: str, text: str):
         return parse_mistral_tool_calls(text)
     if fmt == "llama3_json":
         return parse_llama3_json_tool_calls(text)
+    if fmt == "pythonic":
+        return parse_pythonic_tool_calls(text)
+    if fmt == "granite":
+        return parse_granite_tool_calls(text)
+    if fmt == "internlm2":
+        return parse_internlm2_tool_calls(text)
+    # "hermes" and aliases that use the same <tool_call> JSON format
+    # (qwen, qwen3, glm4 — reference hermes_tool_parser.py family).
     return parse_hermes_tool_calls(text)
