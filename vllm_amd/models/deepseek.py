@@ -183,7 +183,8 @@ class DeepseekDenseMLP(nn.Module):
 class DeepseekMoE(nn.Module):
     """Routed experts (sigmoid + group-limited top-k) + shared experts."""
 
-    def __init__(self, spec: ModelSpec, dtype, ep: bool = False):
+    def __init__(self, spec: ModelSpec, dtype, ep: bool = False,
+                 eplb_window: int = 0):
         super().__init__()
         self.spec = spec
         self.moe = FusedMoE(
@@ -244,7 +245,8 @@ class DeepseekDecoderLayer(nn.Module):
             self.mlp = DeepseekDenseMLP(spec.hidden_size,
                                         spec.intermediate_size, dtype)
         else:
-            self.mlp = DeepseekMoE(spec, dtype, ep=ep)
+            self.mlp = DeepseekMoE(spec, dtype, ep=ep,
+                                   eplb_window=eplb_window)
         self.input_layernorm = RMSNorm(spec.hidden_size, spec.rms_norm_eps,
                                        dtype=dtype)
         self.post_attention_layernorm = RMSNorm(
