@@ -49,13 +49,15 @@ def parse_args():
     p.add_argument("--quantization", type=str, default=None,
                    choices=["fp8"],
                    help="fp8 W8A8 dense linears (~2x MFMA rate on gfx950)")
-    p.add_argument("--parallelism", type=str, default="dp",
+    p.add_argument("--parallelism", type=str, default="tp",
                    choices=["dp", "tp"],
-                   help="dp: one independent engine replica per rank, each "
-                        "with its own --batch (weak scaling; xGMI-free, the "
-                        "throughput-optimal layout for an 8B model). tp: one "
-                        "engine, weights sharded over all ranks (strong "
-                        "scaling; the 70B layout)")
+                   help="tp (default): one engine, weights sharded over all "
+                        "ranks — strong scaling over xGMI (RCCL + custom "
+                        "IPC all-reduce; the 70B layout, and the honest "
+                        "multi-GPU measurement). dp: one independent engine "
+                        "replica per rank, each with its own --batch (weak "
+                        "scaling, xGMI-free — the throughput-optimal "
+                        "serving layout for an 8B model)")
     return p.parse_args()
 
 

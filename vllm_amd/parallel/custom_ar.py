@@ -42,6 +42,7 @@ class CustomCollectives:
         _C = torch.ops.vllm_amd
         self._C = _C
         self.world_size = world_size
+        self.rank_in_group = rank_in_group
         self.max_bytes = max_bytes
         self.handle = _C.car_init(rank_in_group, world_size, max_bytes)
         self.disabled = False
@@ -117,15 +118,44 @@ def try_init_custom_collectives(rank_in_group: int, world_size: int,
         return None
     try:
         inst.connect(cpu_group)
+        ok = _self_test(inst, world_size)
     except Exception as e:  # noqa: BLE001
         # Handle-open failure can be rank-local (IPC limits); every rank
         # must agree again or peers would spin forever in the kernel.
         logger.warning("custom all-reduce connect failed (%s)", e)
-        flag = torch.tensor([0], dtype=torch.int32)
-    else:
-        flag = torch.tensor([1], dtype=torch.int32)
+        ok = False
+    flag = torch.tensor([1 if ok else 0], dtype=torch.int32)
     dist.all_reduce(flag, op=dist.ReduceOp.MIN, group=cpu_group)
     if int(flag.item()) == 0:
+        logger.warning("custom all-reduce self-test failed; using RCCL")
         inst.destroy()
         return None
     return inst
+
+
+def _self_test(inst: CustomCollectives, world_size: int) -> bool:
+    """Verify the IPC kernels against known sums on THIS topology before
+    trusting them with traffic: one-shot, two-shot and all-gather sizes,
+    three rounds each (parity/ack protocol), plus the device error flag.
+    Rank- and position-dependent payloads catch peer-mapping swaps and
+    chunk-offset bugs, not just visibility failures. A failure
+    downgrades the whole group to RCCL instead of corrupting traffic."""
+    r = inst.rank_in_group
+    sum_r = world_size * (world_size - 1) / 2.0
+    for nbytes in (4096, 1 << 20):  # one-shot and two-shot regimes
+        n = nbytes // 4
+        pos = torch.arange(n, dtype=torch.float32, device="cuda") % 17
+        for round_i in range(3):
+            t = pos + 1000.0 * r + round_i
+            inst.all_reduce(t)
+            torch.cuda.synchronize()
+            expect = (pos + round_i) * world_size + 1000.0 * sum_r
+            if not torch.allclose(t, expect):
+                return False
+    g = inst.all_gather_flat(
+        torch.full((1024,), 1.0 + r, dtype=torch.float32, device="cuda"))
+    torch.cuda.synchronize()
+    for p in range(world_size):
+        if not torch.allclose(g[p], torch.full_like(g[p], 1.0 + p)):
+            return False
+    return inst.error() == 0 and rank == 0

@@ -10,6 +10,7 @@ computed after this step.
 from __future__ import annotations
 
 import dataclasses
+import logging
 import gc
 from dataclasses import dataclass, field
 from typing import Optional
@@ -684,14 +685,24 @@ class ModelRunner:
               if self.graph_runner is not None else None)
         hidden = None
         if nb is not None:
-            parts = self.graph_runner.parts_bucket(max_seq_len)
-            logits = self.graph_runner.run(
-                n, nb, parts, input_ids, positions, slot_mapping, seq_lens,
-                block_table, ids_dev=ids_dev,
-                slot_mapping_w=slot_mapping_w, block_table_w=block_table_w,
-            )
-            hidden = self.graph_runner.last_hidden
-        else:
+            try:
+                parts = self.graph_runner.parts_bucket(max_seq_len)
+                logits = self.graph_runner.run(
+                    n, nb, parts, input_ids, positions, slot_mapping,
+                    seq_lens, block_table, ids_dev=ids_dev,
+                    slot_mapping_w=slot_mapping_w,
+                    block_table_w=block_table_w,
+                )
+                hidden = self.graph_runner.last_hidden
+            except Exception:  # noqa: BLE001
+                # A capture/replay failure (e.g. an uncapturable op on an
+                # untested multi-GPU topology) downgrades to eager decode
+                # for the process lifetime instead of killing the engine.
+                logging.getLogger(__name__).exception(
+                    "hipGraph decode failed; falling back to eager decode")
+                self.graph_runner = None
+                nb = None
+        if nb is None:
             meta = AttentionMetadata(
                 query_start_loc=torch.arange(n + 1, dtype=torch.int32,
                                              device=dev),
