@@ -24,14 +24,20 @@ def test_manager_reclaims_out_of_window_blocks():
     blocks = m.allocate_slots(r, 200)
     assert blocks is not None
     r.num_computed_tokens = 200
-    # 200 tokens, window 16, margin 16 -> blocks below (200-32)//16 = 10
-    # returned to the pool already.
+    # Reclaim is bounded by the CHUNK START (the first new token's
+    # window), not the chunk end: the 200-token prefill chunk itself
+    # still reads its early blocks, so nothing is reclaimed during it.
+    assert m.num_reclaimed["a"] == 0
+    # The next allocation (chunk start 200) reclaims blocks below
+    # (200 - window 16 - margin 16) // 16 = 10.
+    m.allocate_slots(r, 8)
+    r.num_computed_tokens = 208
     assert m.num_reclaimed["a"] == 10
     used = free0 - m.block_pool.get_num_free_blocks()
     assert used == 13 - 10  # 13 allocated, 10 reclaimed
-    # decode steps keep the window: allocate a few more tokens
+    # decode steps keep the window moving
     m.allocate_slots(r, 8)
-    r.num_computed_tokens = 208
+    r.num_computed_tokens = 216
     assert m.num_reclaimed["a"] == 11
     # request end: remaining blocks freed exactly once (no double free)
     m.free(r)

@@ -93,6 +93,19 @@ class BlockPool:
             block.block_hash = h
             self.cached_block_hash_to_block[h.value] = block
 
+    def uncache(self, block: KVCacheBlock) -> None:
+        """Drop a block's prefix-cache registration WITHOUT eviction
+        side effects (no host-tier save). Used by sliding-window
+        reclaim: a window-reclaimed block's content is semantically dead
+        — a later prefix hit on it would resume from KV the window
+        model can no longer extend (and the memory may be reused)."""
+        if block.block_hash is None:
+            return
+        cached = self.cached_block_hash_to_block.get(block.block_hash.value)
+        if cached is block:
+            del self.cached_block_hash_to_block[block.block_hash.value]
+        block.reset_hash()
+
     def free_blocks(self, ordered_blocks: list[KVCacheBlock]) -> None:
         """Deref blocks; zero-ref blocks go to the free queue in the given
         order (callers pass eviction-preference order: tail blocks first)."""

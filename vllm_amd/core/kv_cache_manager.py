@@ -154,6 +154,45 @@ class KVCacheManager:
             num_required_blocks - len(req_blocks) - len(new_computed_blocks)
         )
 
+        # Sliding window: return leading blocks whose every token lies
+        # outside the window of ALL queries from this chunk onward. The
+        # bound uses the FIRST new token of this allocation, not the
+        # last: earlier query rows of a prefill chunk still read (and
+        # write) positions inside their own windows, and a block freed
+        # here can be handed to another request scheduled in the SAME
+        # round — basing reclaim on total_tokens corrupted KV under pool
+        # pressure (caught by test_mixed_model_preemption_resume). Runs
+        # BEFORE the capacity check so this request's own stale blocks
+        # fund its new allocation; reclaimed blocks are UNCACHED (a
+        # prefix hit on window-reclaimed KV cannot be extended). One
+        # block of margin absorbs spec-decode rollbacks.
+        chunk_start = total_tokens - num_new_tokens
+        if self.sliding_window > 0:
+            reclaim_below = (
+                chunk_start - self.sliding_window - self.block_size
+            ) // self.block_size
+            done = self.num_reclaimed[request.request_id]
+            if reclaim_below > done:
+                stale = req_blocks[done:reclaim_below]
+                for b in stale:
+                    self.block_pool.uncache(b)
+                self.block_pool.free_blocks(list(reversed(stale)))
+                self.num_reclaimed[request.request_id] = reclaim_below
+        if self.mixed_window > 0:
+            # Same reclaim rule, applied to the WINDOW group only; global
+            # layers keep their full-length blocks.
+            reclaim_below = (
+                chunk_start - self.mixed_window - self.block_size
+            ) // self.block_size
+            done = self.num_reclaimed_w[request.request_id]
+            if reclaim_below > done:
+                blocks_w = self.req_to_blocks_w[request.request_id]
+                stale = blocks_w[done:reclaim_below]
+                for b in stale:
+                    self.block_pool.uncache(b)
+                self.block_pool.free_blocks(list(reversed(stale)))
+                self.num_reclaimed_w[request.request_id] = reclaim_below
+
         need = max(num_new_blocks, 0)
         if self.mixed_window > 0:
             need *= 2  # window group allocates in lockstep
@@ -178,31 +217,6 @@ class KVCacheManager:
         if self.mixed_window > 0 and num_new_blocks > 0:
             self.req_to_blocks_w[request.request_id].extend(
                 self.block_pool.get_new_blocks(num_new_blocks))
-
-        # Sliding window: return leading blocks whose every token lies
-        # outside the window of ALL future queries. One block of margin
-        # absorbs spec-decode rollbacks (<= a few tokens).
-        if self.sliding_window > 0:
-            reclaim_below = (
-                total_tokens - self.sliding_window - self.block_size
-            ) // self.block_size
-            done = self.num_reclaimed[request.request_id]
-            if reclaim_below > done:
-                stale = req_blocks[done:reclaim_below]
-                self.block_pool.free_blocks(list(reversed(stale)))
-                self.num_reclaimed[request.request_id] = reclaim_below
-        if self.mixed_window > 0:
-            # Same reclaim rule, applied to the WINDOW group only; global
-            # layers keep their full-length blocks.
-            reclaim_below = (
-                total_tokens - self.mixed_window - self.block_size
-            ) // self.block_size
-            done = self.num_reclaimed_w[request.request_id]
-            if reclaim_below > done:
-                blocks_w = self.req_to_blocks_w[request.request_id]
-                stale = blocks_w[done:reclaim_below]
-                self.block_pool.free_blocks(list(reversed(stale)))
-                self.num_reclaimed_w[request.request_id] = reclaim_below
 
         # Content-cache the blocks that become full after this step.
         if self.enable_caching:
