@@ -47,9 +47,8 @@ def _eplb_worker(rank: int, world: int, port: int, q):
 
         # Heavily skewed synthetic load: experts 0..3 hot. The greedy
         # plan must split hot experts across ranks.
-        moe._eplb_load = torch.tensor(
-            [100.0, 90.0, 80.0, 70.0, 1.0, 1.0, 1.0, 1.0],
-            dtype=torch.float64)
+        moe.eplb_load.copy_(torch.tensor(
+            [100, 90, 80, 70, 1, 1, 1, 1], dtype=torch.int64))
         old_assignment = list(moe.assignment)
         moe.rebalance()
         assert moe.assignment != old_assignment, "plan should change"
@@ -66,9 +65,9 @@ def _eplb_worker(rank: int, world: int, port: int, q):
 
         # Second rebalance with a different skew: weights keep moving
         # correctly (slots shuffle within and across ranks).
-        moe._eplb_load = torch.tensor(
-            [1.0, 1.0, 1.0, 1.0, 50.0, 60.0, 70.0, 80.0],
-            dtype=torch.float64)
+        moe._eplb_ewma = None  # fresh EWMA for the new skew
+        moe.eplb_load.copy_(torch.tensor(
+            [1, 1, 1, 1, 50, 60, 70, 80], dtype=torch.int64))
         moe.rebalance()
         out2 = moe(x)
         assert torch.allclose(out0, out2, atol=1e-5, rtol=1e-5)

@@ -119,8 +119,8 @@ class EngineArgs:
                                  "(least-loaded request routing)")
         parser.add_argument("--enable-expert-parallel", action="store_true")
         parser.add_argument("--eplb-window", type=int, default=0,
-                            help="rebalance expert placement every N MoE "
-                                 "forwards (0=off)")
+                            help="rebalance expert placement every N engine "
+                                 "steps (0=off)")
         parser.add_argument("--multiprocess-engine", action="store_true",
                             help="run the engine core in its own process")
         parser.add_argument("--device", type=str, default="auto")

@@ -104,9 +104,15 @@ class FusedMoE(nn.Module):
                 torch.tensor([e % n for e in range(num_experts)],
                              dtype=torch.int64),
                 persistent=False)
-            self._eplb_load = torch.zeros(num_experts, dtype=torch.float64)
+            # Device-side token-load accumulator: the += in forward is
+            # hipGraph-capturable and keeps accumulating across graph
+            # REPLAYS; the runner triggers rebalance() between steps
+            # (collectives and the D2H read cannot run inside a capture).
+            self.register_buffer("eplb_load",
+                                 torch.zeros(num_experts,
+                                             dtype=torch.int64),
+                                 persistent=False)
             self._eplb_ewma = None
-            self._eplb_calls = 0
         # Fragment-major weight copies for the HIP grouped GEMM (built
         # lazily on the first GPU forward; freed by invalidate_shuffled
         # when weights change). Costs a second copy of the expert
@@ -117,12 +123,8 @@ class FusedMoE(nn.Module):
 
     # -- EPLB ----------------------------------------------------------
     def _eplb_observe(self, topk_ids) -> None:
-        counts = torch.bincount(topk_ids.reshape(-1),
-                                minlength=self.num_experts)
-        self._eplb_load += counts.double().cpu()
-        self._eplb_calls += 1
-        if self._eplb_calls % self.eplb_window == 0:
-            self.rebalance()
+        self.eplb_load += torch.bincount(topk_ids.reshape(-1),
+                                         minlength=self.num_experts)
 
     def _plan_assignment(self, load) -> list[int]:
         """Greedy balanced packing: experts by load (desc), each to the
@@ -153,12 +155,12 @@ class FusedMoE(nn.Module):
         from vllm_amd.parallel.state import get_ep_group
 
         ep = get_ep_group()
-        cur = self._eplb_load.numpy()
+        cur = self.eplb_load.detach().cpu().double().numpy()
         if self._eplb_ewma is None:
             self._eplb_ewma = cur.copy()
         else:
             self._eplb_ewma = 0.5 * self._eplb_ewma + 0.5 * cur
-        self._eplb_load.zero_()
+        self.eplb_load.zero_()
         plan = self._plan_assignment(self._eplb_ewma.tolist())
         if plan == self.assignment:
             return

@@ -264,6 +264,12 @@ class ModelRunner:
         self.np_nblocks = np.zeros(n, dtype=np.int32)
         self.np_lora = np.zeros(n, dtype=np.int64)
         self.lora_manager = None
+        # EPLB: rebalance expert placement every N executed steps
+        # (runner-side — outside graph capture; all ranks hit the same
+        # counter in lockstep).
+        self._eplb_every = (config.parallel_config.eplb_window
+                            if self.spec.is_moe else 0)
+        self._eplb_step = 0
         # Sampling-metadata cache for steady-state decode batches.
         self._samp_cache_key = None
         self._samp_cache_val = None
@@ -552,6 +558,14 @@ class ModelRunner:
 
     # ------------------------------------------------------------------
     def _update_states(self, so: SchedulerOutput) -> None:
+        if (self._eplb_every and so.total_num_scheduled_tokens > 0
+                and self.model is not None):
+            self._eplb_step += 1
+            if self._eplb_step % self._eplb_every == 0:
+                from vllm_amd.layers.fused_moe import FusedMoE
+                for m in self.model.modules():
+                    if isinstance(m, FusedMoE) and m.ep_size > 1:
+                        m.rebalance()
         if so.kv_swap_ops:
             self._run_kv_swaps(so.kv_swap_ops)
         for req_id in so.finished_req_ids:
