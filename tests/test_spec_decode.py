@@ -285,3 +285,57 @@ def test_sampled_spec_decode_accepts_drafts():
     assert sched.spec_stats_drafted > 0
     assert sched.spec_stats_accepted > 0
     llm.shutdown()
+
+
+def test_draft_model_spec_decode_matches_baseline():
+    """Draft-model proposer with the TARGET model as its own draft:
+    greedy outputs must equal the non-spec engine, and acceptance must
+    be near-perfect (the draft argmaxes the same distribution)."""
+    llm = LLM(model="tiny-llama", dtype="fp32", device="cpu",
+              block_size=16, num_gpu_blocks=256, max_model_len=512,
+              max_num_batched_tokens=512, max_num_seqs=4,
+              num_speculative_tokens=3, spec_decode_method="draft",
+              speculative_model="tiny-llama")
+    prompts = [[7, 8, 9, 10] * 12, list(range(30, 60))]
+    outs = llm.generate(prompts, SamplingParams(
+        temperature=0.0, max_tokens=24, ignore_eos=True))
+    spec = [o.outputs[0].token_ids for o in outs]
+    sched = llm.engine.engine_core.scheduler
+    drafted, accepted = sched.spec_stats_drafted, sched.spec_stats_accepted
+    llm.shutdown()
+
+    base_llm = LLM(model="tiny-llama", dtype="fp32", device="cpu",
+                   block_size=16, num_gpu_blocks=256, max_model_len=512,
+                   max_num_batched_tokens=512, max_num_seqs=4)
+    base = [o.outputs[0].token_ids for o in base_llm.generate(
+        prompts, SamplingParams(temperature=0.0, max_tokens=24,
+                                ignore_eos=True))]
+    base_llm.shutdown()
+    assert spec == base
+    assert drafted > 0
+    # Identical draft and target: nearly every draft position accepted.
+    assert accepted / drafted > 0.9, (accepted, drafted)
+
+
+def test_draft_model_different_draft_still_exact():
+    """A DIFFERENT (random-init) draft model cannot change outputs —
+    only the acceptance rate."""
+    llm = LLM(model="tiny-llama", dtype="fp32", device="cpu",
+              block_size=16, num_gpu_blocks=256, max_model_len=512,
+              max_num_batched_tokens=512, max_num_seqs=4,
+              num_speculative_tokens=3, spec_decode_method="draft",
+              speculative_model="tiny-qwen3")
+    prompts = [[5, 6, 7] * 10]
+    outs = llm.generate(prompts, SamplingParams(
+        temperature=0.0, max_tokens=16, ignore_eos=True))
+    spec = outs[0].outputs[0].token_ids
+    llm.shutdown()
+
+    base_llm = LLM(model="tiny-llama", dtype="fp32", device="cpu",
+                   block_size=16, num_gpu_blocks=256, max_model_len=512,
+                   max_num_batched_tokens=512, max_num_seqs=4)
+    base = base_llm.generate(prompts, SamplingParams(
+        temperature=0.0, max_tokens=16,
+        ignore_eos=True))[0].outputs[0].token_ids
+    base_llm.shutdown()
+    assert spec == base

@@ -818,6 +818,9 @@ class SchedulerConfig:
     # Optional dirs with draft safetensors; None -> random-init drafts.
     medusa_path: Optional[str] = None
     eagle_path: Optional[str] = None
+    # Draft-model spec decode (spec_decode_method="draft"): preset name
+    # or local HF dir of the independent small proposer model.
+    speculative_model: Optional[str] = None
 
     def __post_init__(self) -> None:
         if self.num_speculative_tokens > 0:

@@ -100,6 +100,11 @@ class Scheduler:
                 from vllm_amd.spec_decode.eagle import EagleProposer
 
                 self.spec_proposer = EagleProposer()
+            elif sc.spec_decode_method == "draft":
+                from vllm_amd.spec_decode.draft_model import (
+                    DraftModelProposer)
+
+                self.spec_proposer = DraftModelProposer()
             else:
                 from vllm_amd.spec_decode.ngram import NgramProposer
 
@@ -492,7 +497,16 @@ class Scheduler:
                 else:
                     drafts = self.spec_proposer.propose(
                         request.all_token_ids)
-                request.spec_token_ids = (drafts or [])[: self.spec_k]
+                # Guard: a draft model with a larger vocab could propose
+                # ids the target cannot embed — truncate at the first
+                # out-of-range id (later drafts condition on it).
+                drafts = (drafts or [])[: self.spec_k]
+                vocab = self.config.model_config.spec.vocab_size
+                for di, t in enumerate(drafts):
+                    if not 0 <= t < vocab:
+                        drafts = drafts[:di]
+                        break
+                request.spec_token_ids = drafts
         return outputs
 
     def _check_stop(self, request: Request, last_token: int) -> bool:

@@ -47,6 +47,7 @@ class EngineArgs:
     spec_decode_method: str = "ngram"
     medusa_path: Optional[str] = None
     eagle_path: Optional[str] = None
+    speculative_model: Optional[str] = None
     ngram_prompt_lookup_min: int = 2
     ngram_prompt_lookup_max: int = 4
     async_scheduling: bool = True
@@ -102,8 +103,12 @@ class EngineArgs:
             help="LoRA adapters to serve (requests select by model name)")
         parser.add_argument("--num-speculative-tokens", type=int, default=0)
         parser.add_argument("--spec-decode-method", type=str,
-                            default="ngram", choices=["ngram", "medusa", "eagle"])
+                            default="ngram", choices=["ngram", "medusa", "eagle", "draft"])
         parser.add_argument("--medusa-path", type=str, default=None)
+        parser.add_argument("--speculative-model", type=str,
+                            default=None,
+                            help="draft model (preset name or HF dir) for "
+                                 "--spec-decode-method draft")
         parser.add_argument("--eagle-path", type=str, default=None)
         parser.add_argument("--ngram-prompt-lookup-min", type=int, default=2)
         parser.add_argument("--ngram-prompt-lookup-max", type=int, default=4)
@@ -197,6 +202,7 @@ class EngineArgs:
                 spec_decode_method=self.spec_decode_method,
                 medusa_path=self.medusa_path,
                 eagle_path=self.eagle_path,
+                speculative_model=self.speculative_model,
                 ngram_prompt_lookup_min=self.ngram_prompt_lookup_min,
                 ngram_prompt_lookup_max=self.ngram_prompt_lookup_max,
             ),

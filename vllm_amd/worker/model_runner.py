@@ -242,6 +242,7 @@ class ModelRunner:
         self.medusa = None  # MedusaHeads when spec_decode_method=medusa
         self.vision = None  # VisionTower when spec.vision_layers > 0
         self.eagle = None  # EagleRunnerSide when spec_decode_method=eagle
+        self.draft_lm = None  # DraftModelRunnerSide for method=draft
         self.graph_runner: Optional[DecodeGraphRunner] = None
         self.sampler = Sampler()
         self.max_model_len = config.model_config.max_model_len
@@ -332,6 +333,22 @@ class ModelRunner:
                 compute_logits=self.model.compute_logits,
                 k=sc.num_speculative_tokens,
                 max_len=self.max_model_len,
+                dtype=self.dtype, device=self.device)
+        if (sc.num_speculative_tokens > 0
+                and sc.spec_decode_method == "draft"):
+            import dataclasses as _dc
+
+            from vllm_amd.spec_decode.draft_model import DraftModelRunnerSide
+
+            assert sc.speculative_model, (
+                "--spec-decode-method draft needs --speculative-model")
+            dmc = _dc.replace(self.config.model_config,
+                              model=sc.speculative_model, spec=None)
+            draft_model = load_model(dmc, self.device)
+            self.draft_lm = DraftModelRunnerSide(
+                draft_model, dmc.spec, k=sc.num_speculative_tokens,
+                max_len=self.max_model_len,
+                max_seqs=self.config.scheduler_config.max_num_seqs,
                 dtype=self.dtype, device=self.device)
         mc = self.config.model_config
         self.lora_manager = None
@@ -574,6 +591,8 @@ class ModelRunner:
                 self._free_rows.append(row)
                 if self.eagle is not None:
                     self.eagle.free(req_id)
+                if self.draft_lm is not None:
+                    self.draft_lm.free(req_id)
         for nr in so.scheduled_new_reqs:
             self.requests[nr.req_id] = CachedReqState(
                 req_id=nr.req_id,
@@ -774,6 +793,16 @@ class ModelRunner:
                 self.eagle.observe(rid, int(positions[j]),
                                    hidden[j:j + 1], [tok])
                 draft_map[rid] = self.eagle.propose(rid, tok)
+        if self.draft_lm is not None:
+            # Commit this step's input token to the draft KV, then
+            # speculate from the freshly sampled one (spec decode runs
+            # sync-scheduled, so np-side inputs are current).
+            toks = sampled_t.cpu().tolist()
+            draft_map = {}
+            for j, rid in enumerate(req_ids):
+                self.draft_lm.observe(rid, int(positions[j]),
+                                      [int(input_ids[j])])
+                draft_map[rid] = self.draft_lm.propose(rid, int(toks[j]))
 
         if self.device.type == "cuda" and s_out.logprobs is None:
             pin = self._pin_sampled[self._pin_idx]
@@ -1250,6 +1279,34 @@ class ModelRunner:
                                    next_toks)
                 if i in sampling_set and sampled_per_req[i]:
                     draft_map[rid] = self.eagle.propose(
+                        rid, sampled_per_req[i][-1])
+        if self.draft_lm is not None:
+            # Token-conditioned: commit the chunk's FINAL tokens (kept
+            # positions after verify) to the draft KV, then speculate
+            # from the newest accepted token.
+            draft_map = {}
+            sampling_set = {r: npos for r, npos
+                            in zip(sampling_rows, sampling_npos)}
+            for i, rid in enumerate(req_ids):
+                state = self.requests.get(rid)
+                if state is None or state.sampling_params.pooling:
+                    continue
+                q0 = int(query_start_loc[i])
+                q1 = int(query_start_loc[i + 1])
+                nq = q1 - q0
+                kept = nq
+                if i in sampling_set and sampling_set[i] > 1:
+                    kept = nq - (sampling_set[i]
+                                 - len(sampled_per_req[i]))
+                if kept <= 0:
+                    continue
+                p0 = int(positions[q0])
+                chunk_toks = state.token_ids[p0: p0 + kept]
+                if len(chunk_toks) < kept:
+                    continue
+                self.draft_lm.observe(rid, p0, chunk_toks)
+                if i in sampling_set and sampled_per_req[i]:
+                    draft_map[rid] = self.draft_lm.propose(
                         rid, sampled_per_req[i][-1])
         # Mixed steps resolve on the CPU; invalidate the device-side
         # sampled-token carry so the next decode reads np_last_tok.
