@@ -342,8 +342,14 @@ class ModelRunner:
 
             assert sc.speculative_model, (
                 "--spec-decode-method draft needs --speculative-model")
-            dmc = _dc.replace(self.config.model_config,
-                              model=sc.speculative_model, spec=None)
+            import os as _os
+            _draft_dir = _os.path.isdir(sc.speculative_model)
+            dmc = _dc.replace(
+                self.config.model_config,
+                model=sc.speculative_model, spec=None,
+                model_path=sc.speculative_model if _draft_dir else None,
+                load_format="safetensors" if _draft_dir else "dummy",
+                quantization=None, lora_modules=None)
             draft_model = load_model(dmc, self.device)
             self.draft_lm = DraftModelRunnerSide(
                 draft_model, dmc.spec, k=sc.num_speculative_tokens,
