@@ -234,3 +234,30 @@ def test_gptq_dequant_roundtrip():
         torch.float32)
     assert torch.equal(out["m.weight"], deq)
     assert "other" in out and "m.qzeros" not in out
+
+
+def test_fp8_block_checkpoint_dequant(tmp_path):
+    """DeepSeek-V3-style block-wise fp8 serialization: weight fp8 +
+    weight_scale_inv [N/128, K/128] dequantizes at load to the exact
+    block product (incl. ragged tail blocks)."""
+    import torch
+    from safetensors.torch import save_file
+
+    from vllm_amd.models.weight_loader import _iter_safetensors
+
+    torch.manual_seed(0)
+    n, k = 200, 300  # ragged vs 128 blocks
+    w = torch.randn(n, k) * 0.05
+    scale = torch.rand((n + 127) // 128, (k + 127) // 128) + 0.5
+    w_fp8 = (w / scale.repeat_interleave(128, 0)[:n]
+             .repeat_interleave(128, 1)[:, :k]).to(torch.float8_e4m3fn)
+    save_file({"model.layers.0.mlp.w.weight": w_fp8.contiguous(),
+               "model.layers.0.mlp.w.weight_scale_inv": scale,
+               "model.norm.weight": torch.ones(8)},
+              str(tmp_path / "model.safetensors"))
+    got = dict(_iter_safetensors(str(tmp_path), dtype=torch.float32))
+    expect = (w_fp8.float() * scale.repeat_interleave(128, 0)[:n]
+              .repeat_interleave(128, 1)[:, :k])
+    assert torch.allclose(got["model.layers.0.mlp.w.weight"], expect)
+    assert torch.equal(got["model.norm.weight"], torch.ones(8))
+    assert "model.layers.0.mlp.w.weight_scale_inv" not in got

@@ -36,6 +36,14 @@ def _iter_safetensors(path: str, dtype=None):
                 for name, t in dequantize_awq_stream(
                         tensors, dtype or torch.float32).items():
                     yield name, t
+            elif any(n.endswith(".weight_scale_inv") for n in names):
+                from vllm_amd.quant_loaders import (
+                    dequantize_fp8_block_stream)
+
+                tensors = {n: sf.get_tensor(n) for n in names}
+                for name, t in dequantize_fp8_block_stream(
+                        tensors, dtype or torch.float32).items():
+                    yield name, t
             else:
                 for name in names:
                     yield name, sf.get_tensor(name)

@@ -163,9 +163,9 @@ def _warn_prefill_fallback(head_dim: int) -> None:
         import logging
 
         logging.getLogger(__name__).warning(
-            "prefill head_dim=%d: MFMA prefill kernel is head_dim-128 "
-            "only; prefill rows use the torch path (decode stays on the "
-            "HIP kernel)", head_dim)
+            "prefill head_dim=%d: off-template (64/128/256); prefill "
+            "rows use the torch path (decode stays on the HIP kernel)",
+            head_dim)
 
 
 # MoE routing (small [T, E] tensors — torch ops are fine here).

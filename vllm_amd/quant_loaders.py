@@ -151,3 +151,31 @@ def dequantize_awq_stream(tensors: "dict[str, torch.Tensor]",
             continue
         out[name] = t
     return out
+
+
+def dequantize_fp8_block_stream(tensors: "dict[str, torch.Tensor]",
+                                dtype: torch.dtype,
+                                block: int = 128):
+    """DeepSeek-V3-style block-wise fp8 checkpoints (reference
+    layers/quantization/fp8.py weight_block_size=[128,128] path):
+    {prefix}.weight is float8_e4m3fn [N, K] with
+    {prefix}.weight_scale_inv float32 [ceil(N/128), ceil(K/128)];
+    w[n, k] = fp8[n, k] * scale_inv[n//128, k//128]. Dequantized to the
+    model dtype at load (the fp8 MFMA runtime path re-quantizes W8A8
+    per-channel under --quantization fp8)."""
+    out = {}
+    for name, t in tensors.items():
+        if name.endswith(".weight_scale_inv"):
+            continue
+        scale_name = f"{name}_scale_inv"
+        if (name.endswith(".weight")
+                and t.dtype == torch.float8_e4m3fn
+                and scale_name in tensors):
+            s = tensors[scale_name].float()
+            n, k = t.shape
+            s_full = s.repeat_interleave(block, 0)[:n] \
+                      .repeat_interleave(block, 1)[:, :k]
+            out[name] = (t.float() * s_full).to(dtype)
+        else:
+            out[name] = t
+    return out
