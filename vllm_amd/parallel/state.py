@@ -43,6 +43,11 @@ class GroupCoordinator:
             return t
         if self.comms is not None and self.comms.should_use(t):
             return self.comms.all_reduce(t)
+        if t.is_cuda:
+            r = self._maybe_rccl()
+            if r is not None:
+                r.all_reduce(t)
+                return t
         dist.all_reduce(t, group=self.device_group)
         return t
 
