@@ -94,3 +94,29 @@ def test_torch_tensor_image_and_rejects_text_only_model():
                           SamplingParams(max_tokens=2))
     finally:
         llm2.shutdown()
+
+
+def test_multi_image_prompt():
+    """Two placeholders + a stacked [2, 3, H, W] image batch: each
+    placeholder span gets its own image's features (order-sensitive),
+    deterministic across runs."""
+    rng = np.random.default_rng(3)
+    a = rng.normal(size=(3, 32, 32)).astype(np.float32)
+    b = rng.normal(size=(3, 32, 32)).astype(np.float32)
+    llm = _llm()
+
+    def gen(images):
+        prompt = {"prompt_token_ids": [5, IMG, 6, IMG, 7],
+                  "multi_modal_data": {"image": np.stack(images)}}
+        outs = llm.generate([prompt], SamplingParams(
+            max_tokens=6, temperature=0.0, ignore_eos=True, logprobs=1))
+        o = outs[0].outputs[0]
+        lp = o.logprobs[0][o.token_ids[0]]
+        return o.token_ids, float(getattr(lp, "logprob", lp))
+
+    t_ab, lp_ab = gen([a, b])
+    t_ab2, lp_ab2 = gen([a, b])
+    t_ba, lp_ba = gen([b, a])
+    llm.shutdown()
+    assert t_ab == t_ab2 and lp_ab == lp_ab2   # deterministic + cache-safe
+    assert lp_ab != lp_ba                       # image ORDER reaches logits

@@ -14,7 +14,7 @@ import pytest
 
 
 def _tp_worker(rank: int, world: int, port: int, model: str, q,
-               ep: bool = False):
+               ep: bool = False, sp: bool = False):
     os.environ["RANK"] = str(rank)
     os.environ["LOCAL_RANK"] = str(rank)
     os.environ["WORLD_SIZE"] = str(world)
@@ -41,6 +41,7 @@ def _tp_worker(rank: int, world: int, port: int, model: str, q,
                                              max_num_seqs=4),
             parallel_config=ParallelConfig(tensor_parallel_size=world,
                                            enable_expert_parallel=ep,
+                                           enable_sequence_parallel=sp,
                                            distributed_backend="gloo"),
             device_config=DeviceConfig(device="cpu"),
         )
@@ -194,3 +195,34 @@ def test_fp8_quantization_under_tp2():
     toks = next(p for s, p in outs if p is not None)
     assert len(toks["run0"]) == 6
     assert toks["run0"] == toks["run1"]
+
+
+def test_sequence_parallel_tp2_matches_tp1():
+    """SP decode (residual stream sharded, AG/RS instead of per-layer
+    all-reduce) must be numerically identical to plain TP, which is
+    itself tested equal to TP1. Runs through the SPMD engine with
+    world=2 on gloo; odd batch sizes exercise the SP row padding."""
+    ctx = mp.get_context("spawn")
+    results = {}
+    for sp, port in ((False, 29731), (True, 29741)):
+        q = ctx.Queue()
+        procs = [
+            ctx.Process(target=_tp_worker,
+                        args=(r, 2, port, "tiny-llama", q),
+                        kwargs={"sp": sp})
+            for r in range(2)
+        ]
+        for p in procs:
+            p.start()
+        res = None
+        for _ in range(2):
+            status, payload = q.get(timeout=180)
+            assert status == "ok", payload
+            if payload is not None:
+                res = payload
+        for p in procs:
+            p.join(timeout=30)
+            if p.is_alive():
+                p.kill()
+        results[sp] = res
+    assert results[True] == results[False]

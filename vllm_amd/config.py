@@ -840,6 +840,11 @@ class ParallelConfig:
     # re-pack experts onto ranks by EWMA token load (0 = off). Role of
     # the reference's vllm/distributed/eplb (eplb_state.py:220).
     eplb_window: int = 0
+    # Sequence parallelism on decode steps (llama-family): residual
+    # stream sharded across TP ranks between blocks; per-layer
+    # all-reduces become all-gather + reduce-scatter (role of the
+    # reference's SP compile pass, parallel_state.py:164-250).
+    enable_sequence_parallel: bool = False
     # Run the engine core (scheduler + executor) in its own process; the
     # API process only tokenizes/detokenizes (reference EngineCoreProc).
     multiprocess_engine: bool = False

@@ -63,6 +63,7 @@ int64_t car_error();
 void car_destroy();
 void car_all_reduce(torch::Tensor t);
 void car_all_gather(torch::Tensor out, torch::Tensor t);
+void car_reduce_scatter(torch::Tensor out, torch::Tensor t);
 
 }  // namespace vllm_amd
 
@@ -113,6 +114,7 @@ TORCH_LIBRARY(vllm_amd, m) {
   m.def("car_destroy() -> ()");
   m.def("car_all_reduce(Tensor(a!) t) -> ()");
   m.def("car_all_gather(Tensor(a!) out, Tensor t) -> ()");
+  m.def("car_reduce_scatter(Tensor(a!) out, Tensor t) -> ()");
   m.impl("car_init", &vllm_amd::car_init);
   m.impl("car_connect", &vllm_amd::car_connect);
   m.impl("car_is_ready", &vllm_amd::car_is_ready);
@@ -136,6 +138,7 @@ TORCH_LIBRARY_IMPL(vllm_amd, CUDA, m) {
   m.impl("scale_rows_cols", &vllm_amd::scale_rows_cols);
   m.impl("car_all_reduce", &vllm_amd::car_all_reduce);
   m.impl("car_all_gather", &vllm_amd::car_all_gather);
+  m.impl("car_reduce_scatter", &vllm_amd::car_reduce_scatter);
   m.impl("moe_align", &vllm_amd::moe_align);
   m.impl("moe_gemm", &vllm_amd::moe_gemm);
   m.impl("moe_gemm_shuf", &vllm_amd::moe_gemm_shuf);

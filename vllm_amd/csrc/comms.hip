@@ -239,6 +239,36 @@ __global__ void two_shot_kernel(Vec16* __restrict__ io, long long nvec,
   exit_ack(d, b, c);
 }
 
+
+// ---------------------------------------------------------------------------
+// Reduce-scatter: rank r's output = sum over ranks of input chunk r.
+// Same protocol as two_shot phase 1, writing straight to the local out.
+template <typename Tag>
+__global__ void reduce_scatter_kernel(Vec16* __restrict__ out,
+                                      const Vec16* __restrict__ in,
+                                      long long nvec, CarDev d) {
+  const int b = blockIdx.x;
+  const unsigned long long c = enter(d, b);
+  const int parity = (int)(c & 1);
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  const long long chunk = nvec / d.world;
+  Vec16* mine = data_ptr(d.base[d.rank], d.max_bytes, parity);
+  for (long long i = b * blockDim.x + threadIdx.x; i < nvec; i += stride)
+    mine[i] = in[i];
+  barrier_at<READY1_OFF>(d, b, c);
+  const long long lo = (long long)d.rank * chunk;
+  for (long long i = b * blockDim.x + threadIdx.x; i < chunk; i += stride) {
+    Vec16 acc = mine[lo + i];
+    for (int k = 1; k < d.world; ++k) {
+      int p = (d.rank + k) % d.world;
+      acc = vec_add<Tag>(acc,
+                         data_ptr(d.base[p], d.max_bytes, parity)[lo + i]);
+    }
+    out[i] = acc;
+  }
+  exit_ack(d, b, c);
+}
+
 // ---------------------------------------------------------------------------
 // All-gather: rank r's nvec input becomes out rows [r*nvec, (r+1)*nvec).
 __global__ void all_gather_kernel(Vec16* __restrict__ out,
@@ -426,6 +456,45 @@ void car_all_gather(torch::Tensor out, torch::Tensor t) {
                      reinterpret_cast<Vec16*>(out.data_ptr()),
                      reinterpret_cast<const Vec16*>(t.data_ptr()), nvec,
                      g_car.dev);
+  HIP_CHECK_KERNEL();
+}
+
+
+void car_reduce_scatter(torch::Tensor out, torch::Tensor t) {
+  using namespace car;
+  TORCH_CHECK(g_car.ready, "custom all-reduce not initialized");
+  TORCH_CHECK(t.is_cuda() && t.is_contiguous() && out.is_contiguous());
+  const long long nbytes = (long long)t.numel() * t.element_size();
+  TORCH_CHECK(nbytes % 16 == 0 && nbytes <= g_car.max_bytes,
+              "unsupported reduce-scatter size ", nbytes);
+  const long long nvec = nbytes / 16;
+  TORCH_CHECK(nvec % g_car.world == 0,
+              "reduce-scatter needs numel divisible by world*8");
+  TORCH_CHECK(out.numel() * g_car.world == t.numel() &&
+              out.scalar_type() == t.scalar_type());
+  auto stream = c10::hip::getCurrentHIPStream();
+  const int nb = pick_blocks(nvec);
+  auto launch = [&](auto tag) {
+    using Tag = decltype(tag);
+    hipLaunchKernelGGL((reduce_scatter_kernel<Tag>), dim3(nb), dim3(THREADS),
+                       0, stream.stream(),
+                       reinterpret_cast<Vec16*>(out.data_ptr()),
+                       reinterpret_cast<const Vec16*>(t.data_ptr()), nvec,
+                       g_car.dev);
+  };
+  switch (t.scalar_type()) {
+    case torch::kBFloat16:
+      launch(BF16Tag{});
+      break;
+    case torch::kHalf:
+      launch(FP16Tag{});
+      break;
+    case torch::kFloat:
+      launch(F32Tag{});
+      break;
+    default:
+      TORCH_CHECK(false, "car_reduce_scatter: unsupported dtype");
+  }
   HIP_CHECK_KERNEL();
 }
 

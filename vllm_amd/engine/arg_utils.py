@@ -57,6 +57,7 @@ class EngineArgs:
     data_parallel_size: int = 1
     enable_expert_parallel: bool = False
     eplb_window: int = 0
+    enable_sequence_parallel: bool = False
     multiprocess_engine: bool = False
     device: str = "auto"
     trace_file: Optional[str] = None
@@ -123,6 +124,10 @@ class EngineArgs:
                             help="serve-level engine replicas "
                                  "(least-loaded request routing)")
         parser.add_argument("--enable-expert-parallel", action="store_true")
+        parser.add_argument("--enable-sequence-parallel",
+                            action="store_true",
+                            help="shard the residual stream across TP "
+                                 "ranks on decode steps (llama-family)")
         parser.add_argument("--eplb-window", type=int, default=0,
                             help="rebalance expert placement every N engine "
                                  "steps (0=off)")
@@ -163,6 +168,7 @@ class EngineArgs:
             data_parallel_size=self.data_parallel_size,
             enable_expert_parallel=self.enable_expert_parallel,
             eplb_window=self.eplb_window,
+            enable_sequence_parallel=self.enable_sequence_parallel,
             multiprocess_engine=self.multiprocess_engine,
             rank=int(os.environ.get("RANK", "0")),
             local_rank=int(os.environ.get("LOCAL_RANK", "0")),

@@ -240,12 +240,17 @@ class RowParallelLinear(nn.Module):
             else None
         )
 
-    def forward(self, x):
+    def forward(self, x, reduce_results=None):
         y = _run_gemm(self, x, None)
         y = _maybe_apply_lora(self, x, y)
-        if self.reduce_results:
+        reduce = (self.reduce_results if reduce_results is None
+                  else reduce_results)
+        if reduce:
             y = tensor_model_parallel_all_reduce(y)
-        if self.bias is not None:
+        if self.bias is not None and (reduce or get_tp_rank() == 0):
+            # Partial (sequence-parallel) outputs add the bias once:
+            # rank 0's contribution carries it through the later
+            # reduce-scatter.
             y = y + self.bias
         return y
 

@@ -45,10 +45,10 @@ class LlamaMLP(nn.Module):
         )
         self.act_fn = SiluAndMul()
 
-    def forward(self, x):
+    def forward(self, x, reduce_results: bool = True):
         x = self.gate_up_proj(x)
         x = self.act_fn(x)
-        return self.down_proj(x)
+        return self.down_proj(x, reduce_results=reduce_results)
 
 
 class LlamaAttention(nn.Module):
@@ -101,7 +101,7 @@ class LlamaAttention(nn.Module):
                       else "full"),
         )
 
-    def forward(self, positions, hidden):
+    def forward(self, positions, hidden, reduce_results: bool = True):
         T = hidden.shape[0]
         qkv = self.qkv_proj(hidden)
         q, k, v = self.qkv_proj.split_qkv(qkv)
@@ -116,7 +116,7 @@ class LlamaAttention(nn.Module):
                 k.reshape(-1, self.head_dim).contiguous()).view(k.shape)
         self.rotary_emb(positions, q, k)
         out = self.attn(q, k, v)
-        return self.o_proj(out)
+        return self.o_proj(out, reduce_results=reduce_results)
 
 
 class LlamaDecoderLayer(nn.Module):
@@ -133,6 +133,10 @@ class LlamaDecoderLayer(nn.Module):
         )
 
     def forward(self, positions, hidden, residual):
+        from vllm_amd.worker.forward_context import get_forward_context
+        sp = get_forward_context().sp_size
+        if sp > 1:
+            return self._forward_sp(positions, hidden, residual)
         if residual is None:
             residual = hidden
             hidden = self.input_layernorm(hidden)
@@ -141,6 +145,31 @@ class LlamaDecoderLayer(nn.Module):
         hidden = self.self_attn(positions, hidden)
         hidden, residual = self.post_attention_layernorm(hidden, residual)
         hidden = self.mlp(hidden)
+        return hidden, residual
+
+    def _forward_sp(self, positions, hidden, residual):
+        """Sequence-parallel block (role of the reference's SP compile
+        pass, parallel_state.py:164-250): the residual stream is sharded
+        across TP ranks between blocks — norms, residual adds and
+        activations run on T/tp rows; the per-layer all-reduces become
+        all-gather (before the sharded-weight matmuls) + reduce-scatter
+        (after the row-parallel ones). Same bytes on the wire as TP, but
+        the elementwise work is 1/tp per rank and the RS output lands
+        pre-sharded for the next block's norm."""
+        from vllm_amd.parallel.state import (sp_all_gather_rows,
+                                             sp_reduce_scatter_rows)
+        if residual is None:
+            residual = hidden
+            hidden = self.input_layernorm(hidden)
+        else:
+            hidden, residual = self.input_layernorm(hidden, residual)
+        full = sp_all_gather_rows(hidden)
+        part = self.self_attn(positions, full, reduce_results=False)
+        hidden = sp_reduce_scatter_rows(part)
+        hidden, residual = self.post_attention_layernorm(hidden, residual)
+        full = sp_all_gather_rows(hidden)
+        part = self.mlp(full, reduce_results=False)
+        hidden = sp_reduce_scatter_rows(part)
         return hidden, residual
 
 
@@ -175,9 +204,10 @@ class LlamaModel(nn.Module):
                      if is_last_pp_rank() else None)
 
     def forward(self, input_ids, positions, hidden_in=None):
+        ctx = get_forward_context()
         if self.embed_tokens is not None:
             hidden = self.embed_tokens(input_ids)
-            mm = get_forward_context().mm_embeds
+            mm = ctx.mm_embeds
             if mm is not None:
                 idx, feats = mm
                 hidden = hidden.index_copy(
@@ -186,11 +216,24 @@ class LlamaModel(nn.Module):
         else:
             hidden = hidden_in
             residual = None
+        sp = ctx.sp_size
+        if sp > 1:
+            # Sequence parallelism: shard the residual stream rows
+            # (runner pads the batch to a multiple of tp).
+            from vllm_amd.parallel.state import get_tp_rank
+            T = hidden.shape[0]
+            assert T % sp == 0, "SP needs the batch padded to tp"
+            shard = T // sp
+            r = get_tp_rank()
+            hidden = hidden[r * shard:(r + 1) * shard]
         for layer in self.layers[self.lo:self.hi]:
             hidden, residual = layer(positions, hidden, residual)
         if self.norm is None:
             return hidden + residual  # stage-boundary activation
         hidden, _ = self.norm(hidden, residual)
+        if sp > 1:
+            from vllm_amd.parallel.state import sp_all_gather_rows
+            hidden = sp_all_gather_rows(hidden)
         return hidden
 
 

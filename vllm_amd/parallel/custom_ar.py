@@ -73,6 +73,23 @@ class CustomCollectives:
         nbytes = t.numel() * t.element_size()
         return nbytes % 16 == 0 and nbytes <= self.max_bytes
 
+    def can_reduce_scatter(self, t: torch.Tensor) -> bool:
+        if self.disabled or not t.is_cuda:
+            return False
+        if t.dtype not in (torch.bfloat16, torch.float16, torch.float32):
+            return False
+        nbytes = t.numel() * t.element_size()
+        return (nbytes % 16 == 0 and nbytes <= self.max_bytes
+                and (nbytes // 16) % self.world_size == 0
+                and t.shape[0] % self.world_size == 0)
+
+    def reduce_scatter_rows(self, t: torch.Tensor) -> torch.Tensor:
+        t = t.contiguous()
+        out_shape = (t.shape[0] // self.world_size,) + tuple(t.shape[1:])
+        out = torch.empty(out_shape, dtype=t.dtype, device=t.device)
+        self._C.car_reduce_scatter(out.view(-1), t.view(-1))
+        return out
+
     def all_gather_flat(self, t: torch.Tensor) -> torch.Tensor:
         """Returns [world, *t.shape]."""
         t = t.contiguous()
@@ -158,4 +175,4 @@ def _self_test(inst: CustomCollectives, world_size: int) -> bool:
     for p in range(world_size):
         if not torch.allclose(g[p], torch.full_like(g[p], 1.0 + p)):
             return False
-    return inst.error() == 0 and rank == 0
+    return inst.error() == 0

@@ -77,6 +77,8 @@ class GroupCoordinator:
         if self.world_size == 1:
             return t
         assert dim == 0
+        if self.comms is not None and self.comms.can_reduce_scatter(t):
+            return self.comms.reduce_scatter_rows(t)
         out_shape = (t.shape[0] // self.world_size,) + tuple(t.shape[1:])
         out = torch.empty(out_shape, dtype=t.dtype, device=t.device)
         dist.reduce_scatter_tensor(out, t.contiguous(), group=self.device_group)
@@ -348,6 +350,17 @@ def destroy_distributed() -> None:
 # Convenience wrappers used by layers.
 def tensor_model_parallel_all_reduce(t: torch.Tensor) -> torch.Tensor:
     return get_tp_group().all_reduce(t)
+
+
+def sp_all_gather_rows(t: torch.Tensor) -> torch.Tensor:
+    """Sequence parallelism: sharded rows [T/tp, H] -> full [T, H]."""
+    return get_tp_group().all_gather(t, dim=0)
+
+
+def sp_reduce_scatter_rows(t: torch.Tensor) -> torch.Tensor:
+    """Sequence parallelism: partial full rows [T, H] -> reduced shard
+    [T/tp, H] (this rank's contiguous row chunk)."""
+    return get_tp_group().reduce_scatter(t, dim=0)
 
 
 def tensor_model_parallel_all_gather(t: torch.Tensor, dim: int = -1) -> torch.Tensor:

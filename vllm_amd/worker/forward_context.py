@@ -49,6 +49,10 @@ class ForwardContext:
     # Vision: (flat token indices in this batch, feature rows) to
     # scatter over placeholder embeddings right after embed_tokens.
     mm_embeds: Optional[tuple] = None
+    # Sequence parallelism (decode steps): residual stream sharded over
+    # this many TP ranks between blocks; 1 = off. The runner pads the
+    # token count to a multiple.
+    sp_size: int = 1
 
 
 # Thread-local: serve-level DP replicas run one engine loop per thread

@@ -97,10 +97,16 @@ class DecodeGraphRunner:
         self.buckets = [b for b in self.BATCH_BUCKETS if b <= self.max_seqs]
         if self.buckets[-1] != self.max_seqs:
             self.buckets.append(self.max_seqs)
+        self.sp_size = runner.sp_size
+        if self.sp_size > 1:
+            # SP shards rows across ranks: buckets round to tp multiples.
+            sp = self.sp_size
+            self.buckets = sorted({-(-b // sp) * sp for b in self.buckets})
         self.max_parts = max(1, _cdiv(runner.max_model_len, self.DEC_PART))
         self.max_blocks = _cdiv(runner.max_model_len, runner.block_size)
         dev = runner.device
-        n = self.max_seqs
+        n = max(self.max_seqs, self.buckets[-1])
+        self.max_seqs = n
         self.input_ids = torch.zeros(n, dtype=torch.int64, device=dev)
         self.positions = torch.zeros(n, dtype=torch.int64, device=dev)
         self.slot_mapping = torch.full((n,), -1, dtype=torch.int64,
@@ -159,7 +165,8 @@ class DecodeGraphRunner:
     def _capture(self, nb: int, parts: int):
         runner = self.runner
         ctx = ForwardContext(attn_metadata=self._meta(nb, parts),
-                             kv_caches=runner.kv_caches)
+                             kv_caches=runner.kv_caches,
+                             sp_size=self.sp_size)
         # Warmup on a side stream (allocator state, RCCL lazy init).
         s = torch.cuda.Stream()
         s.wait_stream(torch.cuda.current_stream())
@@ -271,6 +278,17 @@ class ModelRunner:
         self._eplb_every = (config.parallel_config.eplb_window
                             if self.spec.is_moe else 0)
         self._eplb_step = 0
+        # Sequence parallelism on pure-decode steps (llama-family
+        # blocks consult ForwardContext.sp_size; other archs ignore it
+        # and run plain TP).
+        from vllm_amd.parallel.state import get_pp_world_size as _gpw
+        self.sp_size = (config.parallel_config.tensor_parallel_size
+                        if (config.parallel_config.enable_sequence_parallel
+                            and config.parallel_config
+                            .tensor_parallel_size > 1
+                            and config.parallel_config
+                            .pipeline_parallel_size == 1)
+                        else 1)
         # Sampling-metadata cache for steady-state decode batches.
         self._samp_cache_key = None
         self._samp_cache_val = None
@@ -742,18 +760,49 @@ class ModelRunner:
                 self.graph_runner = None
                 nb = None
         if nb is None:
+            np_pad = 0
+            if self.sp_size > 1 and n % self.sp_size:
+                # Sequence parallelism shards rows across ranks: pad the
+                # batch to tp with inert rows (ctx 1 token of block 0,
+                # slot -1 so nothing is written; outputs sliced off).
+                np_pad = self.sp_size - n % self.sp_size
+                input_ids = np.concatenate(
+                    [input_ids, np.zeros(np_pad, dtype=np.int64)])
+                positions = np.concatenate(
+                    [positions, np.zeros(np_pad, dtype=np.int64)])
+                slot_mapping = np.concatenate(
+                    [slot_mapping, np.full(np_pad, -1, dtype=np.int64)])
+                seq_lens = np.concatenate(
+                    [seq_lens, np.ones(np_pad, dtype=np.int32)])
+                block_table = np.concatenate(
+                    [block_table,
+                     np.zeros((np_pad, block_table.shape[1]),
+                              dtype=np.int32)])
+                if self.mixed_attn:
+                    slot_mapping_w = np.concatenate(
+                        [slot_mapping_w, np.full(np_pad, -1,
+                                                 dtype=np.int64)])
+                    block_table_w = np.concatenate(
+                        [block_table_w,
+                         np.zeros((np_pad, block_table_w.shape[1]),
+                                  dtype=np.int32)])
+                if ids_dev is not None:
+                    ids_dev = torch.cat(
+                        [ids_dev, torch.zeros(np_pad, dtype=ids_dev.dtype,
+                                              device=ids_dev.device)])
+            ntot = n + np_pad
             meta = AttentionMetadata(
-                query_start_loc=torch.arange(n + 1, dtype=torch.int32,
+                query_start_loc=torch.arange(ntot + 1, dtype=torch.int32,
                                              device=dev),
                 seq_lens=torch.from_numpy(seq_lens).to(dev),
                 block_table=torch.from_numpy(
                     np.ascontiguousarray(block_table)).to(dev),
                 slot_mapping=torch.from_numpy(slot_mapping).to(dev),
-                num_reqs=n,
-                num_actual_tokens=n,
+                num_reqs=ntot,
+                num_actual_tokens=ntot,
                 max_query_len=1,
                 max_seq_len=max_seq_len,
-                num_decodes=n,
+                num_decodes=ntot,
                 block_table_w=(torch.from_numpy(np.ascontiguousarray(
                     block_table_w)).to(dev) if block_table_w is not None
                     else None),
@@ -762,14 +811,20 @@ class ModelRunner:
             )
             ids_t = (ids_dev if ids_dev is not None
                      else torch.from_numpy(input_ids).to(dev))
+            lora_np = self.np_lora[rows]
+            if np_pad:
+                lora_np = np.concatenate(
+                    [lora_np, np.zeros(np_pad, dtype=lora_np.dtype)])
             ctx = ForwardContext(
                 attn_metadata=meta, kv_caches=self.kv_caches,
-                lora_ids=self._lora_ids_tensor(self.np_lora[rows]),
+                lora_ids=self._lora_ids_tensor(lora_np),
                 lora_manager=self.lora_manager,
+                sp_size=self.sp_size,
             )
             with set_forward_context(ctx):
                 hidden = self.model(ids_t, torch.from_numpy(positions).to(dev))
-            logits = self.model.compute_logits(hidden)
+            logits = self.model.compute_logits(hidden)[:n]
+            hidden = hidden[:n]
 
         self.np_computed[rows] += 1
         states = [self.requests[r] for r in req_ids]
