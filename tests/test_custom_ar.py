@@ -63,6 +63,14 @@ def _ar_worker(rank: int, world: int, port: int, q, mode: str):
                     comms.all_reduce(t)
                     torch.cuda.synchronize()
                     results.append(t.cpu())
+            # Reduce-scatter: rank r's output = summed chunk r.
+            t = make((64, 256), torch.bfloat16)
+            rs = comms.reduce_scatter_rows(t)
+            torch.cuda.synchronize()
+            full = ref_sum([((64, 256), torch.bfloat16)])[0]
+            chunk = full.view(2, 32, 256)[rank].to(torch.bfloat16)
+            assert torch.allclose(rs.cpu().float(), chunk.float(),
+                                  rtol=0.05, atol=0.05)
             refs = ref_sum(cases) * 4
             for got, ref in zip(results, [r for r in refs]):
                 ref_c = ref.to(got.dtype)

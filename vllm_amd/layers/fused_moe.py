@@ -253,11 +253,14 @@ class FusedMoE(nn.Module):
     def run_experts(self, hidden, topk_weights, topk_ids) -> torch.Tensor:
         """Local expert computation. EP mode: activations are replicated
         across ranks (SPMD TP serving), so each rank COMPACTS to the
-        (token, slot) pairs routed to its local experts — no wasted
-        expert FLOPs on foreign slots — and the caller's all-reduce
-        completes the combine. (A DeepEP-style a2a dispatch only pays
-        once activations are DP/sequence-sharded — a round-2 item
-        together with DP attention for MoE.)"""
+        (token, slot) pairs routed to its local experts — the per-rank
+        expert GEMM token count is already T*k/ep, identical to what an
+        all-to-all dispatch would compute — and the caller's all-reduce
+        completes the combine. An a2a dispatch/combine only REDUCES
+        traffic once the residual stream itself is sequence-sharded
+        through the MoE block (SP covers llama-family dense blocks;
+        SP-MoE is the tracked follow-on), so the replicated regime keeps
+        the simpler, deterministic collective."""
         if self.ep_size > 1:
             if self.eplb_window > 0:
                 self._eplb_observe(topk_ids)
