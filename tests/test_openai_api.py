@@ -672,3 +672,42 @@ def test_usage_reports_cached_tokens(client):
     r = client.post("/v1/completions", json=body)
     details = r.json()["usage"].get("prompt_tokens_details")
     assert details and details["cached_tokens"] > 0
+
+
+def test_completions_batched_prompts(client):
+    r = client.post("/v1/completions", json={
+        "model": "tiny-llama", "prompt": ["one two", "three four five"],
+        "max_tokens": 4, "temperature": 0.0, "ignore_eos": True})
+    assert r.status_code == 200
+    body = r.json()
+    assert [c["index"] for c in body["choices"]] == [0, 1]
+    assert all(c["finish_reason"] == "length" for c in body["choices"])
+    assert body["usage"]["completion_tokens"] == 8
+
+
+def test_completions_batched_prompts_with_n(client):
+    r = client.post("/v1/completions", json={
+        "model": "tiny-llama", "prompt": ["a b", "c d"], "n": 2, "seed": 5,
+        "max_tokens": 3, "temperature": 1.0, "ignore_eos": True})
+    assert r.status_code == 200
+    body = r.json()
+    # prompt-major indexing: p*n + branch
+    assert [c["index"] for c in body["choices"]] == [0, 1, 2, 3]
+
+
+def test_completions_best_of(client):
+    r = client.post("/v1/completions", json={
+        "model": "tiny-llama", "prompt": "pick the best", "n": 1, "best_of": 4,
+        "seed": 11, "max_tokens": 4, "temperature": 1.0,
+        "ignore_eos": True})
+    assert r.status_code == 200
+    body = r.json()
+    assert len(body["choices"]) == 1
+    assert body["choices"][0]["index"] == 0
+    # logprobs were forced internally for selection but not returned
+    assert body["choices"][0].get("logprobs") is None
+    # invalid combos rejected
+    r = client.post("/v1/completions", json={
+        "model": "tiny-llama", "prompt": "x", "n": 3, "best_of": 2,
+        "max_tokens": 2})
+    assert r.status_code != 200 or "error" in r.json()

@@ -309,50 +309,67 @@ def build_app(state: ServerState) -> FastAPI:
     # ------------------------------------------------------------------
     @app.post("/v1/completions")
     async def completions(req: CompletionRequest, raw: Request):
-        if (req.best_of or 1) != 1:
-            return _error("best_of is not supported")
-        if not 1 <= req.n <= 64:
-            return _error("n must be in [1, 64]")
+        best_of = req.best_of or req.n
+        if best_of < req.n:
+            return _error("best_of must be >= n")
+        if best_of > req.n and req.stream:
+            return _error("best_of with streaming is not supported")
+        if not 1 <= req.n <= 64 or best_of > 64:
+            return _error("n/best_of must be in [1, 64]")
         prompts = req.prompt
         if isinstance(prompts, str):
             prompts = [prompts]
         elif prompts and isinstance(prompts[0], int):
             prompts = [prompts]
-        if len(prompts) != 1:
-            return _error("batched prompts: send one prompt per request")
-        prompt = prompts[0]
+        if not prompts:
+            return _error("prompt must not be empty")
         params = req.to_sampling_params(req.stream)
-        branches = _branch_params(params, req.n)
+        forced_lp = False
+        if best_of > req.n and params.logprobs is None:
+            # Branch selection scores by chosen-token logprob; request it
+            # engine-side and strip it from the response.
+            import dataclasses as _dc
+            params = _dc.replace(params, logprobs=1)
+            forced_lp = True
+        branches = _branch_params(params, best_of)
         lora = req.model if req.model in state.lora_names else None
         state.num_requests += 1
         rid = random_id("cmpl")
+        # (prompt index, branch) flat fan-out; OpenAI choice index is
+        # prompt-major: index = p * n + branch.
+        fan = [(p_i, b, prompt, bp)
+               for p_i, prompt in enumerate(prompts)
+               for b, bp in enumerate(branches)]
 
         if req.stream:
             async def gen() -> AsyncGenerator[str, None]:
                 from vllm_amd.metrics import RequestTimer
                 timer = RequestTimer(state.metrics)
                 n_gen = 0
-                n_prompt = 0
-                gens = [engine.generate(prompt, bp, f"{rid}-{b}", lora=lora)
-                        for b, bp in enumerate(branches)]
+                n_prompt_by: dict[int, int] = {}
+                gens = [engine.generate(prompt, bp, f"{rid}-{p_i}-{b}",
+                                        lora=lora)
+                        for p_i, b, prompt, bp in fan]
                 seen: dict[int, int] = {}
                 try:
-                    async for b, out in _merge_streams(gens):
+                    async for gi, out in _merge_streams(gens):
+                        p_i, b = fan[gi][0], fan[gi][1]
                         comp = out.outputs[0]
                         # comp.token_ids is cumulative; count the delta.
-                        new = len(comp.token_ids) - seen.get(b, 0)
-                        seen[b] = len(comp.token_ids)
+                        new = len(comp.token_ids) - seen.get(gi, 0)
+                        seen[gi] = len(comp.token_ids)
                         state.num_generation_tokens += new
                         timer.on_tokens(new)
                         n_gen += new
-                        n_prompt = len(out.prompt_token_ids)
+                        n_prompt_by[p_i] = len(out.prompt_token_ids)
+                        n_prompt = sum(n_prompt_by.values())
                         chunk = {
                             "id": rid,
                             "object": "text_completion",
                             "created": int(time.time()),
                             "model": req.model,
                             "choices": [{
-                                "index": b,
+                                "index": p_i * req.n + b,
                                 "text": comp.text,
                                 "logprobs": None,
                                 "finish_reason": comp.finish_reason,
@@ -383,19 +400,53 @@ def build_app(state: ServerState) -> FastAPI:
         from vllm_amd.metrics import RequestTimer
         timer = RequestTimer(state.metrics)
 
-        async def run_branch(b, bp):
+        async def run_branch(p_i, b, prompt, bp):
             final = None
-            async for out in engine.generate(prompt, bp, f"{rid}-{b}",
+            async for out in engine.generate(prompt, bp,
+                                             f"{rid}-{p_i}-{b}",
                                              lora=lora):
                 final = out
             return final
 
         try:
             finals = await asyncio.gather(
-                *(run_branch(b, bp) for b, bp in enumerate(branches)))
+                *(run_branch(*item) for item in fan))
         except ValueError as e:
             return _error(str(e))
-        n_prompt = len(finals[0].prompt_token_ids)
+        # best_of > n: keep the n best branches per prompt by mean
+        # chosen-token logprob (OpenAI legacy semantics).
+        out_index = [p_i * req.n + b for p_i, b, _, _ in fan]
+        if best_of > req.n:
+            kept = []
+            for p_i in range(len(prompts)):
+                group = [(gi, finals[gi]) for gi, item in enumerate(fan)
+                         if item[0] == p_i]
+
+                def score(f):
+                    c = f.outputs[0]
+                    lp = getattr(c, "cumulative_logprob", None)
+                    if lp is None and c.logprobs and c.token_ids:
+                        lp = sum(d.get(t, 0.0) if isinstance(
+                            d.get(t, 0.0), float) else
+                            getattr(d.get(t), "logprob", 0.0)
+                            for d, t in zip(c.logprobs, c.token_ids))
+                    if lp is not None and c.token_ids:
+                        return lp / len(c.token_ids)
+                    return 0.0
+
+                group.sort(key=lambda t: -score(t[1]))
+                kept.extend(g for g in group[: req.n])
+            fan = [fan[gi] for gi, _ in kept]
+            finals = [f for _, f in kept]
+            out_index = []
+            per_prompt_counter = {}
+            for p_i, b, _, _ in fan:
+                k = per_prompt_counter.get(p_i, 0)
+                per_prompt_counter[p_i] = k + 1
+                out_index.append(p_i * req.n + k)
+        n_prompt = sum(len(f.prompt_token_ids)
+                       for gi, f in enumerate(finals)
+                       if fan[gi][1] == 0)
         n_gen = sum(len(f.outputs[0].token_ids) for f in finals)
         timer.on_finish(n_prompt, n_gen)
         state.num_prompt_tokens += n_prompt
@@ -409,13 +460,14 @@ def build_app(state: ServerState) -> FastAPI:
                                    if cached else None),
         )
         choices = []
-        for b, final in enumerate(finals):
+        for out_i, final in enumerate(finals):
             comp = final.outputs[0]
             choices.append(CompletionChoice(
-                index=b,
+                index=out_index[out_i],
                 text=(final.prompt or "") + comp.text if req.echo
                 else comp.text,
-                logprobs=_completion_logprobs(comp),
+                logprobs=(None if forced_lp
+                          else _completion_logprobs(comp)),
                 prompt_logprobs=final.prompt_logprobs,
                 finish_reason=comp.finish_reason,
                 stop_reason=comp.stop_reason
