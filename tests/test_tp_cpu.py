@@ -197,18 +197,24 @@ def test_fp8_quantization_under_tp2():
     assert toks["run0"] == toks["run1"]
 
 
-def test_sequence_parallel_tp2_matches_tp1():
+@pytest.mark.parametrize("model,base_port", [
+    ("tiny-llama", 29731),
+    # llama-arch with mixed sliding+global windows: SP row padding must
+    # also pad the hybrid W-group tables.
+    ("tiny-mistral", 29751),
+])
+def test_sequence_parallel_tp2_matches_tp1(model, base_port):
     """SP decode (residual stream sharded, AG/RS instead of per-layer
     all-reduce) must be numerically identical to plain TP, which is
     itself tested equal to TP1. Runs through the SPMD engine with
     world=2 on gloo; odd batch sizes exercise the SP row padding."""
     ctx = mp.get_context("spawn")
     results = {}
-    for sp, port in ((False, 29731), (True, 29741)):
+    for sp, port in ((False, base_port), (True, base_port + 5)):
         q = ctx.Queue()
         procs = [
             ctx.Process(target=_tp_worker,
-                        args=(r, 2, port, "tiny-llama", q),
+                        args=(r, 2, port, model, q),
                         kwargs={"sp": sp})
             for r in range(2)
         ]
