@@ -705,3 +705,27 @@ def test_new_model_family_presets_generate():
     b = llm.generate([[5, 6, 7, 8]], p)[0].outputs[0].token_ids
     llm.shutdown()
     assert len(a) == 8 and a == b
+
+
+def test_neox_parallel_residual_family():
+    """GPT-NeoX/Falcon-style blocks (LayerNorm, parallel residual,
+    partial rotary, biases): deterministic decode + chunked-prefill
+    invariance through the single-all-reduce parallel form."""
+    from vllm_amd.entrypoints.llm import LLM
+    from vllm_amd.sampling_params import SamplingParams
+
+    p = SamplingParams(temperature=0.0, max_tokens=8, ignore_eos=True)
+    prompt = list(range(3, 80))
+    big = LLM(model="tiny-neox", dtype="fp32", device="cpu",
+              block_size=16, num_gpu_blocks=64, max_model_len=256,
+              max_num_batched_tokens=256, max_num_seqs=2)
+    whole = big.generate([prompt], p)[0].outputs[0].token_ids
+    again = big.generate([prompt], p)[0].outputs[0].token_ids
+    big.shutdown()
+    assert whole == again and len(whole) == 8
+    small = LLM(model="tiny-neox", dtype="fp32", device="cpu",
+                block_size=16, num_gpu_blocks=64, max_model_len=256,
+                max_num_batched_tokens=32, max_num_seqs=2)
+    chunked = small.generate([prompt], p)[0].outputs[0].token_ids
+    small.shutdown()
+    assert chunked == whole

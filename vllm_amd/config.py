@@ -42,6 +42,15 @@ class ModelSpec:
     activation: str = "silu"  # silu | relu | gelu
     use_layernorm: bool = False  # OPT uses LayerNorm, llama uses RMSNorm
     use_bias: bool = False
+    # GPT-NeoX / Falcon / Command-R family: attn and MLP read (their
+    # own) norms of the SAME block input and their outputs add into one
+    # residual update — x + attn(ln1(x)) + mlp(ln2(x)). With TP this
+    # also halves the per-layer all-reduces (both row-parallel partials
+    # sum before one reduce).
+    parallel_residual: bool = False
+    # False: plain 2-layer MLP (up -> activation -> down, the
+    # NeoX/Falcon/OPT form) instead of the gated SwiGLU pair.
+    gated_mlp: bool = True
     # Qwen-family specifics
     qkv_bias: bool = False  # Qwen2: bias on q/k/v projections only
     qk_norm: bool = False  # Qwen3: per-head RMSNorm on q and k pre-RoPE
@@ -313,6 +322,68 @@ MODEL_PRESETS: dict[str, ModelSpec] = {
         eos_token_id=2,
     ),
     # Tiny GLM-style spec for CPU tests: partial rotary + qkv bias.
+    # GPT-NeoX family (Pythia): LayerNorm, parallel residual, partial
+    # rotary, biases everywhere (reference models/gpt_neox.py).
+    "pythia-6.9b": ModelSpec(
+        name="pythia-6.9b",
+        architecture="llama",
+        vocab_size=50432,
+        hidden_size=4096,
+        intermediate_size=16384,
+        num_layers=32,
+        num_heads=32,
+        num_kv_heads=32,
+        head_dim=128,
+        rope_theta=10000.0,
+        partial_rotary_factor=0.25,
+        max_position_embeddings=2048,
+        use_layernorm=True,
+        use_bias=True,
+        parallel_residual=True,
+        gated_mlp=False,
+        activation="gelu",
+        eos_token_id=0,
+    ),
+    # Falcon-7B: multi-query attention (1 kv head), LayerNorm, parallel
+    # residual (reference models/falcon.py).
+    "falcon-7b": ModelSpec(
+        name="falcon-7b",
+        architecture="llama",
+        vocab_size=65024,
+        hidden_size=4544,
+        intermediate_size=18176,
+        num_layers=32,
+        num_heads=71,
+        num_kv_heads=1,
+        head_dim=64,
+        rope_theta=10000.0,
+        max_position_embeddings=2048,
+        use_layernorm=True,
+        parallel_residual=True,
+        gated_mlp=False,
+        activation="gelu",
+        eos_token_id=11,
+    ),
+    "tiny-neox": ModelSpec(
+        name="tiny-neox",
+        architecture="llama",
+        vocab_size=1024,
+        hidden_size=64,
+        intermediate_size=128,
+        num_layers=2,
+        num_heads=4,
+        num_kv_heads=4,
+        head_dim=16,
+        rope_theta=10000.0,
+        partial_rotary_factor=0.25,
+        max_position_embeddings=2048,
+        use_layernorm=True,
+        use_bias=True,
+        parallel_residual=True,
+        gated_mlp=False,
+        activation="gelu",
+        eos_token_id=2,
+    ),
     "tiny-glm": ModelSpec(
         name="tiny-glm",
         architecture="llama",

@@ -16,7 +16,7 @@ from vllm_amd.layers.activation import SiluAndMul
 from vllm_amd.layers.attention import Attention
 from vllm_amd.worker.forward_context import get_forward_context
 from vllm_amd.layers.embedding import ParallelLMHead, VocabParallelEmbedding
-from vllm_amd.layers.layernorm import RMSNorm
+from vllm_amd.layers.layernorm import LayerNorm, RMSNorm
 from vllm_amd.layers.linear import (
     MergedColumnParallelLinear,
     QKVParallelLinear,
@@ -25,6 +25,7 @@ from vllm_amd.layers.linear import (
 from vllm_amd.layers.rotary import RotaryEmbedding
 from vllm_amd.parallel.state import (
     is_first_pp_rank,
+    tensor_model_parallel_all_reduce,
     is_last_pp_rank,
     pp_layer_range,
 )
@@ -33,21 +34,37 @@ from vllm_amd.parallel.state import (
 class LlamaMLP(nn.Module):
     def __init__(self, spec: ModelSpec, dtype: torch.dtype):
         super().__init__()
-        self.gate_up_proj = MergedColumnParallelLinear(
-            spec.hidden_size,
-            [spec.intermediate_size, spec.intermediate_size],
-            bias=spec.use_bias,
-            dtype=dtype,
-        )
+        self.gated = spec.gated_mlp
+        if self.gated:
+            self.gate_up_proj = MergedColumnParallelLinear(
+                spec.hidden_size,
+                [spec.intermediate_size, spec.intermediate_size],
+                bias=spec.use_bias,
+                dtype=dtype,
+            )
+            self.act_fn = SiluAndMul()
+        else:
+            # NeoX/Falcon plain MLP: up -> act -> down.
+            from vllm_amd.layers.linear import ColumnParallelLinear
+            self.up_proj = ColumnParallelLinear(
+                spec.hidden_size, spec.intermediate_size,
+                bias=spec.use_bias, dtype=dtype)
+            self.plain_act = spec.activation
         self.down_proj = RowParallelLinear(
             spec.intermediate_size, spec.hidden_size, bias=spec.use_bias,
             dtype=dtype,
         )
-        self.act_fn = SiluAndMul()
 
     def forward(self, x, reduce_results: bool = True):
-        x = self.gate_up_proj(x)
-        x = self.act_fn(x)
+        if self.gated:
+            x = self.gate_up_proj(x)
+            x = self.act_fn(x)
+        else:
+            x = self.up_proj(x)
+            if self.plain_act == "gelu":
+                x = torch.nn.functional.gelu(x, approximate="tanh")
+            else:
+                x = torch.nn.functional.relu(x)
         return self.down_proj(x, reduce_results=reduce_results)
 
 
@@ -126,13 +143,29 @@ class LlamaDecoderLayer(nn.Module):
         self.self_attn = LlamaAttention(spec, layer_idx, dtype, max_position,
                                         cache_idx)
         self.mlp = LlamaMLP(spec, dtype)
-        self.input_layernorm = RMSNorm(spec.hidden_size, spec.rms_norm_eps,
-                                       dtype=dtype)
-        self.post_attention_layernorm = RMSNorm(
+        norm_cls = LayerNorm if spec.use_layernorm else RMSNorm
+        self.parallel_residual = spec.parallel_residual
+        self.input_layernorm = norm_cls(spec.hidden_size, spec.rms_norm_eps,
+                                        dtype=dtype)
+        self.post_attention_layernorm = norm_cls(
             spec.hidden_size, spec.rms_norm_eps, dtype=dtype
         )
 
     def forward(self, positions, hidden, residual):
+        if self.parallel_residual:
+            # GPT-NeoX / Falcon form: x + attn(ln1(x)) + mlp(ln2(x)).
+            # Both branches produce row-parallel PARTIALS summed before
+            # ONE all-reduce — half the per-layer comm of sequential
+            # blocks under TP.
+            if residual is not None:
+                hidden = hidden + residual
+            attn_p = self.self_attn(positions,
+                                    self.input_layernorm(hidden),
+                                    reduce_results=False)
+            mlp_p = self.mlp(self.post_attention_layernorm(hidden),
+                             reduce_results=False)
+            out = tensor_model_parallel_all_reduce(attn_p + mlp_p)
+            return out, hidden
         from vllm_amd.worker.forward_context import get_forward_context
         sp = get_forward_context().sp_size
         if sp > 1:
@@ -199,8 +232,9 @@ class LlamaModel(nn.Module):
                 for i in range(spec.num_layers)
             ]
         )
-        self.norm = (RMSNorm(spec.hidden_size, spec.rms_norm_eps,
-                             dtype=dtype)
+        norm_cls = LayerNorm if spec.use_layernorm else RMSNorm
+        self.norm = (norm_cls(spec.hidden_size, spec.rms_norm_eps,
+                              dtype=dtype)
                      if is_last_pp_rank() else None)
 
     def forward(self, input_ids, positions, hidden_in=None):
