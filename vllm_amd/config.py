@@ -364,6 +364,50 @@ MODEL_PRESETS: dict[str, ModelSpec] = {
         activation="gelu",
         eos_token_id=11,
     ),
+    # StarCoder2: LayerNorm + plain GELU MLP + biases, sequential
+    # residual, sliding window on all layers (reference
+    # models/starcoder2.py).
+    "starcoder2-7b": ModelSpec(
+        name="starcoder2-7b",
+        architecture="llama",
+        vocab_size=49152,
+        hidden_size=4608,
+        intermediate_size=18432,
+        num_layers=32,
+        num_heads=36,
+        num_kv_heads=4,
+        head_dim=128,
+        rope_theta=1000000.0,
+        max_position_embeddings=16384,
+        sliding_window=4096,
+        use_layernorm=True,
+        use_bias=True,
+        gated_mlp=False,
+        activation="gelu",
+        eos_token_id=0,
+    ),
+    # Phi-2: LayerNorm, parallel residual, partial rotary 0.4, plain
+    # GELU MLP (reference models/phi.py).
+    "phi-2": ModelSpec(
+        name="phi-2",
+        architecture="llama",
+        vocab_size=51200,
+        hidden_size=2560,
+        intermediate_size=10240,
+        num_layers=32,
+        num_heads=32,
+        num_kv_heads=32,
+        head_dim=80,
+        rope_theta=10000.0,
+        partial_rotary_factor=0.4,
+        max_position_embeddings=2048,
+        use_layernorm=True,
+        use_bias=True,
+        parallel_residual=True,
+        gated_mlp=False,
+        activation="gelu",
+        eos_token_id=50256,
+    ),
     "tiny-neox": ModelSpec(
         name="tiny-neox",
         architecture="llama",
@@ -740,6 +784,9 @@ def _spec_from_hf_config(path: str) -> ModelSpec:
         architecture = "qwen2"
     else:
         architecture = "llama"
+    neoxish = any(k in arch for k in ("gptneox", "falcon", "cohere",
+                                      "phiforcausallm"))
+    plain_mlp = neoxish or "starcoder2" in arch or "opt" in arch
     # Gemma3 nests the text config under text_config in the multimodal
     # checkpoint layout.
     if architecture == "gemma3" and "text_config" in hf:
@@ -778,7 +825,18 @@ def _spec_from_hf_config(path: str) -> ModelSpec:
         query_pre_attn_scalar=hf.get("query_pre_attn_scalar", 0) or 0,
         scale_embeddings=architecture == "gemma3",
         rmsnorm_unit_offset=architecture == "gemma3",
-        activation="gelu" if architecture == "gemma3" else "silu",
+        use_layernorm=(architecture == "opt" or neoxish
+                       or "starcoder2" in arch),
+        use_bias=bool(hf.get("use_bias") or hf.get("attention_bias")
+                      or neoxish or "starcoder2" in arch
+                      or architecture == "opt"),
+        parallel_residual=bool(
+            hf.get("use_parallel_residual",
+                   hf.get("parallel_attn", neoxish and "cohere" not in arch))
+            if neoxish else False),
+        gated_mlp=not plain_mlp,
+        activation=("gelu" if architecture == "gemma3" or plain_mlp
+                    else "silu"),
     )
 
 
