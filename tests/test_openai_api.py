@@ -711,3 +711,12 @@ def test_completions_best_of(client):
         "model": "tiny-llama", "prompt": "x", "n": 3, "best_of": 2,
         "max_tokens": 2})
     assert r.status_code != 200 or "error" in r.json()
+
+
+def test_anthropic_count_tokens(client):
+    r = client.post("/v1/messages/count_tokens", json={
+        "model": "tiny-llama", "max_tokens": 8,
+        "messages": [{"role": "user", "content": "count these tokens"}]})
+    assert r.status_code == 200
+    n = r.json()["input_tokens"]
+    assert isinstance(n, int) and n > 0

@@ -74,6 +74,14 @@ def build_anthropic_router(state) -> APIRouter:
         parts.append("<|assistant|>\n")
         return "\n".join(parts)
 
+    @router.post("/v1/messages/count_tokens")
+    async def count_tokens(req: MessagesRequest):
+        """Anthropic token-counting endpoint: tokenize the rendered
+        prompt without generating."""
+        prompt = to_prompt(req)
+        ids = state.engine.tokenizer.encode(prompt)
+        return {"input_tokens": len(ids)}
+
     @router.post("/v1/messages")
     async def messages(req: MessagesRequest):
         params = SamplingParams(
