@@ -196,6 +196,12 @@ class KVCacheManager:
         need = max(num_new_blocks, 0)
         if self.mixed_window > 0:
             need *= 2  # window group allocates in lockstep
+        # Committing ref-0 prefix hits (touch) PULLS them off the free
+        # queue, so they must be counted against the free pool alongside
+        # the new blocks — counting them as "free" let the check pass
+        # and get_new_blocks raise under pressure (caught by the soak
+        # test).
+        need += sum(1 for b in new_computed_blocks if b.ref_cnt == 0)
         if need > self.block_pool.get_num_free_blocks():
             return None
 
