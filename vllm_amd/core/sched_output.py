@@ -28,7 +28,7 @@ class NewRequestData:
     block_ids_w: object = None
 
 
-@dataclass
+@dataclass(slots=True)
 class CachedRequestData:
     """Columnar diffs for requests the runner has already seen."""
 
@@ -96,7 +96,7 @@ class ModelRunnerOutput:
 EMPTY_MODEL_RUNNER_OUTPUT = ModelRunnerOutput(req_ids=[], sampled_token_ids=[])
 
 
-@dataclass
+@dataclass(slots=True)
 class EngineCoreOutput:
     req_id: str
     new_token_ids: list[int]

@@ -196,6 +196,92 @@ class Scheduler:
     # Scheduling
 
     def schedule(self) -> SchedulerOutput:  # noqa: C901
+        fast = self._try_schedule_decode_fast()
+        if fast is not None:
+            return fast
+        return self._schedule_slow()
+
+    def _try_schedule_decode_fast(self) -> Optional[SchedulerOutput]:
+        """Steady-state decode fast path: every running request advances
+        exactly one token, nothing is waiting, no spec/grammar state.
+        The general loop costs ~20 us of Python per request per step —
+        ~20 ms at batch 1024, which out-budgets a ~5 ms TP=8 decode
+        step. Here only block-boundary crossings (1 in block_size
+        steps) take the full allocate_slots; everything else is O(1)
+        bookkeeping. Bails (with rollback) to the general path whenever
+        any precondition fails, so semantics are identical by
+        construction."""
+        running = self.running
+        if (self.waiting or not running
+                or len(running) > self.max_num_batched_tokens):
+            return None
+        mgr = self.kv_cache_manager
+        if mgr.sliding_window > 0 or mgr.mixed_window > 0:
+            return None  # window reclaim stays on the audited path
+        bs = self.block_size
+        max_len = self.max_model_len
+        # Precondition scan (no mutation).
+        for r in running:
+            if (r.spec_token_ids or r.grammar is not None
+                    or r.num_tokens + r.num_output_placeholders
+                    - r.num_computed_tokens != 1
+                    or r.num_computed_tokens + 1 > max_len):
+                return None
+        req_to_blocks = mgr.req_to_blocks
+        cached = CachedRequestData()
+        req_ids = cached.req_ids
+        new_block_ids = cached.new_block_ids
+        resumed = cached.resumed
+        num_computed_list = cached.num_computed_tokens
+        new_token_ids = cached.new_token_ids
+        committed: list = []
+        ok = True
+        for r in running:
+            total = r.num_computed_tokens + 1
+            blocks_needed = (total + bs - 1) // bs
+            if total % bs == 0 or blocks_needed > len(
+                    req_to_blocks[r.request_id]):
+                nb = mgr.allocate_slots(r, 1)
+                if nb is None:
+                    ok = False
+                    break
+                ids = [b.block_id for b in nb]
+            else:
+                ids = []
+            req_ids.append(r.request_id)
+            resumed.append(False)
+            new_block_ids.append(ids)
+            num_computed_list.append(r.num_computed_tokens)
+            new_token_ids.append(
+                r.all_token_ids[r.num_computed_tokens:total])
+            r.num_computed_tokens = total
+            if total == r.num_tokens + r.num_output_placeholders:
+                r.num_output_placeholders += 1
+            committed.append(r)
+        if not ok:
+            # Roll back and let the general path preempt.
+            for r in committed:
+                r.num_computed_tokens -= 1
+                # The placeholder was bumped iff (restored computed + 2)
+                # == tokens + bumped placeholders.
+                if (r.num_computed_tokens + 2
+                        == r.num_tokens + r.num_output_placeholders):
+                    r.num_output_placeholders -= 1
+            return None
+        n = len(running)
+        out = SchedulerOutput(
+            scheduled_new_reqs=[],
+            scheduled_cached_reqs=cached,
+            num_scheduled_tokens={r.request_id: 1 for r in running},
+            total_num_scheduled_tokens=n,
+            finished_req_ids=self.finished_req_ids,
+            scheduled_spec_decode_tokens={},
+            kv_swap_ops=mgr.take_swap_ops(),
+        )
+        self.finished_req_ids = set()
+        return out
+
+    def _schedule_slow(self) -> SchedulerOutput:  # noqa: C901
         token_budget = self.max_num_batched_tokens
         scheduled_spec_tokens: dict[str, list[int]] = {}
         scheduled_new_reqs: list[NewRequestData] = []
@@ -405,13 +491,17 @@ class Scheduler:
             zip(runner_output.req_ids, runner_output.sampled_token_ids)
         )
         logprobs_by_req = runner_output.logprobs or {}
+        pooled_by_req = runner_output.pooled or {}
+        plp_by_req = runner_output.prompt_logprobs or {}
+        spec_sched = scheduler_output.scheduled_spec_decode_tokens
+        requests = self.requests
 
         for req_id, num_sched in scheduler_output.num_scheduled_tokens.items():
-            request = self.requests.get(req_id)
+            request = requests.get(req_id)
             if request is None or request.is_finished():
                 continue  # aborted mid-step
 
-            pooled_vec = (runner_output.pooled or {}).get(req_id)
+            pooled_vec = pooled_by_req.get(req_id) if pooled_by_req else None
             if pooled_vec is not None:
                 # Pooling request: prefill done, no tokens — finish now.
                 request.status = RequestStatus.FINISHED_STOPPED
@@ -426,15 +516,16 @@ class Scheduler:
                 self._free_request(request)
                 continue
 
-            new_plp = (runner_output.prompt_logprobs or {}).get(req_id)
-            if new_plp and not request.is_finished():
-                request.prompt_logprob_chunks = (
-                    getattr(request, "prompt_logprob_chunks", []) + new_plp)
+            if plp_by_req:
+                new_plp = plp_by_req.get(req_id)
+                if new_plp and not request.is_finished():
+                    request.prompt_logprob_chunks = (
+                        getattr(request, "prompt_logprob_chunks", [])
+                        + new_plp)
 
             new_token_ids = sampled_by_req.get(req_id) or []
-            num_spec_sched = len(
-                scheduler_output.scheduled_spec_decode_tokens.get(req_id, ())
-            )
+            num_spec_sched = (len(spec_sched.get(req_id, ()))
+                              if spec_sched else 0)
             if num_spec_sched > 0:
                 # Roll back KV positions of rejected draft tokens: they
                 # were computed from draft values that turned out wrong
@@ -462,9 +553,9 @@ class Scheduler:
             kept_tokens: list[int] = []
             for tok in new_token_ids:
                 kept_tokens.append(tok)
-                request.append_output_token_ids([tok])
-                request.num_output_placeholders = max(
-                    0, request.num_output_placeholders - 1)
+                request.append_output_token(tok)
+                if request.num_output_placeholders:
+                    request.num_output_placeholders -= 1
                 stopped = self._check_stop(request, tok)
                 if stopped:
                     break

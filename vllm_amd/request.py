@@ -81,6 +81,8 @@ class Request:
         # All token ids: prompt + generated. Kept as one list so attention
         # metadata / block hashing index into a single sequence.
         self._all_token_ids: list[int] = list(prompt_token_ids)
+        self.num_tokens: int = len(self._all_token_ids)
+        self.num_output_tokens: int = 0
 
         # Number of tokens whose KV is already computed (incl. prefix-cache
         # hits). Catches up to num_tokens under the scheduler token budget;
@@ -101,13 +103,10 @@ class Request:
     def num_prompt_tokens(self) -> int:
         return len(self.prompt_token_ids)
 
-    @property
-    def num_output_tokens(self) -> int:
-        return len(self.output_token_ids)
-
-    @property
-    def num_tokens(self) -> int:
-        return len(self._all_token_ids)
+    # num_tokens / num_output_tokens are PLAIN INTS maintained by
+    # append_output_token_ids: they sit on the scheduler's per-request
+    # per-step hot path, and Python property+len() cost is measurable at
+    # batch 1024 (profiles/r02_summary.md control-plane section).
 
     @property
     def all_token_ids(self) -> list[int]:
@@ -121,6 +120,15 @@ class Request:
     def append_output_token_ids(self, token_ids: list[int]) -> None:
         self.output_token_ids.extend(token_ids)
         self._all_token_ids.extend(token_ids)
+        n = len(token_ids)
+        self.num_tokens += n
+        self.num_output_tokens += n
+
+    def append_output_token(self, token_id: int) -> None:
+        self.output_token_ids.append(token_id)
+        self._all_token_ids.append(token_id)
+        self.num_tokens += 1
+        self.num_output_tokens += 1
 
     def is_finished(self) -> bool:
         return RequestStatus.is_finished(self.status)
