@@ -254,3 +254,49 @@ class TestAbort:
         # Next schedule() reports it finished so the runner can clear state.
         out = sched.schedule()
         assert "r1" in out.finished_req_ids
+
+
+def test_decode_fast_path_abort_and_length_cap():
+    """The steady-decode fast path must propagate finished_req_ids from
+    aborts and bail to the general path at the context-length cap."""
+    from vllm_amd.core.sched_output import ModelRunnerOutput
+    from vllm_amd.request import RequestStatus
+
+    sched = create_scheduler(max_num_seqs=4, max_num_batched_tokens=512,
+                             num_gpu_blocks=64, max_model_len=48)
+    for i in range(3):
+        r = make_request(f"r{i}", num_tokens=32, max_tokens=64,
+                         prompt=[3 + i * 7 + j for j in range(32)])
+        r.sampling_params.ignore_eos = True
+        sched.add_request(r)
+    # prefill
+    so = sched.schedule()
+    sched.update_from_output(so, ModelRunnerOutput(
+        req_ids=list(so.num_scheduled_tokens),
+        sampled_token_ids=[[5]] * len(so.num_scheduled_tokens)))
+    # steady decode (fast path)
+    so = sched.schedule()
+    assert all(v == 1 for v in so.num_scheduled_tokens.values())
+    sched.update_from_output(so, ModelRunnerOutput(
+        req_ids=list(so.num_scheduled_tokens),
+        sampled_token_ids=[[6]] * len(so.num_scheduled_tokens)))
+    # abort one request; its id must reach the next SchedulerOutput.
+    sched.finish_requests(["r1"], RequestStatus.FINISHED_ABORTED)
+    so = sched.schedule()
+    assert "r1" in so.finished_req_ids
+    assert "r1" not in so.num_scheduled_tokens
+    sched.update_from_output(so, ModelRunnerOutput(
+        req_ids=list(so.num_scheduled_tokens),
+        sampled_token_ids=[[7]] * len(so.num_scheduled_tokens)))
+    # run to the 48-token context cap: every step must stay exactly one
+    # token per request and stop scheduling at the cap (slow-path clamp).
+    for _ in range(40):
+        so = sched.schedule()
+        if not so.num_scheduled_tokens:
+            break
+        assert all(v == 1 for v in so.num_scheduled_tokens.values())
+        sched.update_from_output(so, ModelRunnerOutput(
+            req_ids=list(so.num_scheduled_tokens),
+            sampled_token_ids=[[8]] * len(so.num_scheduled_tokens)))
+    for r in sched.running:
+        assert r.num_computed_tokens <= 48
