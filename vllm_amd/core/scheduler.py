@@ -41,7 +41,6 @@ class PriorityWaiting:
     exactly what preempted requests need."""
 
     def __init__(self):
-        import heapq
         self._heapq = heapq
         self._h: list = []
 

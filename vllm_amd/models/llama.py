@@ -166,7 +166,6 @@ class LlamaDecoderLayer(nn.Module):
                              reduce_results=False)
             out = tensor_model_parallel_all_reduce(attn_p + mlp_p)
             return out, hidden
-        from vllm_amd.worker.forward_context import get_forward_context
         sp = get_forward_context().sp_size
         if sp > 1:
             return self._forward_sp(positions, hidden, residual)
