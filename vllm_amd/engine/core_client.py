@@ -13,7 +13,6 @@ from __future__ import annotations
 import logging
 import queue
 import threading
-from typing import Optional
 
 import torch.multiprocessing as mp
 

@@ -10,7 +10,7 @@ from typing import Any, AsyncGenerator, Optional, Union
 
 from fastapi import APIRouter
 from fastapi.responses import JSONResponse, StreamingResponse
-from pydantic import BaseModel, Field
+from pydantic import BaseModel
 
 from vllm_amd.sampling_params import RequestOutputKind, SamplingParams
 

@@ -19,7 +19,7 @@ from __future__ import annotations
 import logging
 import os
 import threading
-from typing import Any, Optional
+from typing import Any
 
 import torch.multiprocessing as mp
 

@@ -7,7 +7,6 @@ with a single MI355X backend family instead of a backend registry.
 
 from __future__ import annotations
 
-from typing import Optional
 
 import torch
 import torch.nn as nn

@@ -7,7 +7,6 @@ shards after the LM-head GEMM).
 
 from __future__ import annotations
 
-from typing import Optional
 
 import torch
 import torch.nn as nn

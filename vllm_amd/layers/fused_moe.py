@@ -41,7 +41,6 @@ class FusedMoE(nn.Module):
         eplb_window: int = 0,
     ):
         super().__init__()
-        from vllm_amd.config import EngineConfig  # noqa: F401 (doc only)
         from vllm_amd.parallel.state import get_ep_group
 
         tp = get_tp_world_size()

@@ -8,7 +8,7 @@ seed, stop conditions, token limits, logprobs.
 from __future__ import annotations
 
 import dataclasses
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from enum import Enum
 from typing import Optional, Union
 

@@ -21,7 +21,7 @@ KV rollback and preemption re-runs stay consistent automatically.
 from __future__ import annotations
 
 import math
-from typing import Callable, Optional
+from typing import Callable
 
 import torch
 import torch.nn as nn

@@ -12,7 +12,7 @@ from __future__ import annotations
 import dataclasses
 import logging
 import gc
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Optional
 
 import numpy as np
@@ -281,7 +281,6 @@ class ModelRunner:
         # Sequence parallelism on pure-decode steps (llama-family
         # blocks consult ForwardContext.sp_size; other archs ignore it
         # and run plain TP).
-        from vllm_amd.parallel.state import get_pp_world_size as _gpw
         self.sp_size = (config.parallel_config.tensor_parallel_size
                         if (config.parallel_config.enable_sequence_parallel
                             and config.parallel_config
