@@ -42,16 +42,17 @@ def _error(msg: str, code: int = 400) -> JSONResponse:
 
 
 def apply_chat_template(tokenizer, messages, add_generation_prompt=True,
-                        tools=None):
+                        tools=None, image_sentinel=None):
     """HF chat template when a real tokenizer is loaded; otherwise a
     simple role-tagged fallback (mock tokenizer / no template). `tools`
     go through the template's `tools=` kwarg when it takes one, else a
-    hermes-style system preamble."""
+    hermes-style system preamble. `image_sentinel` marks image_url
+    content parts for post-template token splicing."""
     hf = getattr(tokenizer, "tokenizer", None)
     if hf is not None and getattr(hf, "chat_template", None):
         dicts = []
         for m in messages:
-            d = {"role": m.role, "content": m.text()}
+            d = {"role": m.role, "content": m.text(image_sentinel)}
             if getattr(m, "tool_calls", None):
                 d["tool_calls"] = m.tool_calls
             if getattr(m, "tool_call_id", None):
@@ -75,10 +76,79 @@ def apply_chat_template(tokenizer, messages, add_generation_prompt=True,
     if tools:
         from vllm_amd.entrypoints.tool_parser import render_tools_block
         parts.append(f"<|system|>\n{render_tools_block(tools)}")
-    parts += [f"<|{m.role}|>\n{m.text()}" for m in messages]
+    parts += [f"<|{m.role}|>\n{m.text(image_sentinel)}" for m in messages]
     if add_generation_prompt:
         parts.append("<|assistant|>\n")
     return "\n".join(parts)
+
+
+# Marks image positions in templated text; NUL bytes cannot appear in
+# JSON chat content, so the sentinel never collides with user text.
+_IMG_SENTINEL = "\x00<image>\x00"
+
+
+def _decode_image_data_url(url: str, size: int):
+    """`data:image/...;base64,...` -> [3, size, size] float tensor scaled
+    to [-1, 1] (the vision tower's expected range). Only data: URLs are
+    accepted: this server has no network egress, so remote image_url
+    fetching (which the reference does) fails fast with a clear error."""
+    if not url.startswith("data:"):
+        raise ValueError(
+            "only data: image URLs are supported (no network egress)")
+    import base64
+    import io
+
+    payload = url.split(",", 1)
+    if len(payload) != 2:
+        raise ValueError("malformed data: URL (missing comma)")
+    raw = base64.b64decode(payload[1])
+    from PIL import Image
+
+    img = Image.open(io.BytesIO(raw)).convert("RGB")
+    img = img.resize((size, size), Image.BILINEAR)
+    import numpy as np
+    import torch
+
+    arr = torch.from_numpy(np.asarray(img, dtype=np.float32).copy())
+    return arr.permute(2, 0, 1) / 127.5 - 1.0
+
+
+def build_mm_chat_prompt(engine, messages, add_generation_prompt,
+                         image_urls, tools=None):
+    """Multimodal chat prompt: template with sentinels, splice one
+    image placeholder token per image into the token ids, decode the
+    data-URL images to a stacked pixel tensor. Returns the engine dict
+    prompt ({"prompt_token_ids", "multi_modal_data"}); the engine
+    expands each placeholder to per-patch tokens (llm_engine.py
+    add_request) and salts prefix-cache hashes with the pixel content."""
+    spec = engine.config.model_config.spec
+    if spec.vision_layers == 0:
+        raise ValueError(
+            f"model {spec.name} does not support image input")
+    text = apply_chat_template(
+        engine.tokenizer, messages, add_generation_prompt,
+        tools=tools, image_sentinel=_IMG_SENTINEL)
+    pieces = text.split(_IMG_SENTINEL)
+    if len(pieces) - 1 != len(image_urls):
+        raise ValueError("chat template dropped image placeholders")
+    ids: list[int] = []
+    for i, piece in enumerate(pieces):
+        if piece:
+            ids.extend(engine.tokenizer.encode(
+                piece, add_special_tokens=(i == 0)))
+        if i < len(pieces) - 1:
+            ids.append(spec.image_token_id)
+    import torch
+
+    try:
+        imgs = [_decode_image_data_url(u, spec.image_size)
+                for u in image_urls]
+    except ValueError:
+        raise
+    except Exception as e:  # binascii / PIL decode errors -> 400
+        raise ValueError(f"could not decode image: {e}") from e
+    image = torch.stack(imgs) if len(imgs) > 1 else imgs[0]
+    return {"prompt_token_ids": ids, "multi_modal_data": {"image": image}}
 
 
 class ServerState:
@@ -725,10 +795,21 @@ def build_app(state: ServerState) -> FastAPI:
 
         tools_on = bool(req.tools) and req.tool_choice != "none"
         named = req.named_tool() if tools_on else None
-        prompt = apply_chat_template(
-            engine.tokenizer, req.messages, req.add_generation_prompt,
-            tools=req.tools if tools_on and not named else None,
-        )
+        image_urls = [u for m in req.messages for u in m.image_urls()]
+        if image_urls:
+            try:
+                prompt = build_mm_chat_prompt(
+                    engine, req.messages, req.add_generation_prompt,
+                    image_urls,
+                    tools=req.tools if tools_on and not named else None,
+                )
+            except ValueError as e:
+                return _error(str(e))
+        else:
+            prompt = apply_chat_template(
+                engine.tokenizer, req.messages, req.add_generation_prompt,
+                tools=req.tools if tools_on and not named else None,
+            )
         default_max = state.max_model_len
         params = req.to_sampling_params(req.stream, default_max)
         lora = req.model if req.model in state.lora_names else None

@@ -94,7 +94,7 @@ class ChatMessage(BaseModel):
     tool_calls: Optional[list[dict[str, Any]]] = None
     tool_call_id: Optional[str] = None
 
-    def text(self) -> str:
+    def text(self, image_sentinel: Optional[str] = None) -> str:
         if isinstance(self.content, str):
             return self.content
         if self.content is None:
@@ -103,7 +103,24 @@ class ChatMessage(BaseModel):
         for seg in self.content:
             if seg.get("type") == "text":
                 parts.append(seg.get("text", ""))
+            elif seg.get("type") == "image_url" and image_sentinel:
+                # Placeholder spliced back into token ids by the server
+                # (reference: multimodal chat content parts become
+                # per-model placeholder tokens via the mm processor).
+                parts.append(image_sentinel)
         return "".join(parts)
+
+    def image_urls(self) -> list[str]:
+        """URLs of image_url content parts, in order of appearance."""
+        if not isinstance(self.content, list):
+            return []
+        out = []
+        for seg in self.content:
+            if seg.get("type") == "image_url":
+                u = seg.get("image_url")
+                out.append(u.get("url", "") if isinstance(u, dict) else
+                           str(u or ""))
+        return out
 
 
 class ChatCompletionRequest(BaseModel):

@@ -49,8 +49,15 @@ class TokenizerWrapper:
     def eos_token_id(self) -> Optional[int]:
         return getattr(self.tokenizer, "eos_token_id", None)
 
-    def encode(self, text: str) -> list[int]:
-        return self.tokenizer.encode(text)
+    def encode(self, text: str, add_special_tokens: bool = True) -> list[int]:
+        if add_special_tokens:
+            return self.tokenizer.encode(text)
+        # Piece-wise encoding (multimodal prompt splicing) must not repeat
+        # BOS/prefix specials at every image boundary.
+        try:
+            return self.tokenizer.encode(text, add_special_tokens=False)
+        except TypeError:
+            return self.tokenizer.encode(text)
 
     def decode(self, ids: list[int], skip_special_tokens=True) -> str:
         return self.tokenizer.decode(
