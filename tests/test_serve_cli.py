@@ -69,3 +69,34 @@ def test_serve_cli_boot():
         except subprocess.TimeoutExpired:
             proc.kill()
             proc.wait(timeout=10)
+
+
+def _run_cli(args):
+    out = subprocess.run(
+        [sys.executable, "-m", "vllm_amd"] + args,
+        capture_output=True, text=True, timeout=240,
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    assert out.returncode == 0, out.stdout + out.stderr
+    # Last stdout line is the machine-readable JSON summary.
+    return json.loads(out.stdout.strip().splitlines()[-1])
+
+
+def test_bench_latency_cli():
+    res = _run_cli(["bench", "latency", "--model", "tiny-llama",
+                    "--device", "cpu", "--dtype", "fp32",
+                    "--input-len", "8", "--output-len", "4",
+                    "--batch-size", "2", "--num-iters-warmup", "1",
+                    "--num-iters", "2"])
+    assert res["avg_latency"] > 0
+    assert len(res["latencies"]) == 2
+    assert "50" in res["percentiles"]
+
+
+def test_bench_throughput_cli():
+    res = _run_cli(["bench", "throughput", "--model", "tiny-llama",
+                    "--device", "cpu", "--dtype", "fp32",
+                    "--num-prompts", "6", "--input-len", "16",
+                    "--output-len", "8"])
+    assert res["num_requests"] == 6
+    assert res["total_output_tokens"] == 6 * 8
+    assert res["requests_per_second"] > 0

@@ -9,19 +9,27 @@ import sys
 def main() -> None:
     if len(sys.argv) < 2 or sys.argv[1] in ("-h", "--help"):
         print("usage: python -m vllm_amd {serve,bench,run-batch} [args]\n"
-              "  serve          — start the OpenAI-compatible API server\n"
-              "  bench serving  — TTFT/ITL + throughput at fixed QPS\n"
-              "  run-batch      — offline OpenAI batch-format JSONL runner\n"
-              "  collect-env    — print environment diagnostics")
+              "  serve            — start the OpenAI-compatible API server\n"
+              "  bench serving    — TTFT/ITL + throughput at fixed QPS\n"
+              "  bench latency    — offline batch end-to-end latency\n"
+              "  bench throughput — offline requests/s and tokens/s\n"
+              "  run-batch        — offline OpenAI batch-format JSONL "
+              "runner\n"
+              "  collect-env      — print environment diagnostics")
         return
     cmd = sys.argv.pop(1)
     if cmd == "bench":
-        # `python -m vllm_amd bench serving [args]`
+        # `python -m vllm_amd bench {serving,latency,throughput} [args]`
         sub = sys.argv.pop(1) if len(sys.argv) > 1 else "serving"
-        if sub != "serving":
-            raise SystemExit("bench subcommands: serving")
-        from benchmarks.bench_serving import main as bench_main
-
+        if sub == "serving":
+            from benchmarks.bench_serving import main as bench_main
+        elif sub == "latency":
+            from benchmarks.bench_latency import main as bench_main
+        elif sub == "throughput":
+            from benchmarks.bench_throughput import main as bench_main
+        else:
+            raise SystemExit(
+                "bench subcommands: serving, latency, throughput")
         bench_main()
         return
     if cmd == "collect-env":
