@@ -91,10 +91,21 @@ class ModelSpec:
     vision_patch: int = 14
     image_size: int = 336
     image_token_id: int = 0
+    # Mamba / SSM family (architecture "mamba"): constant-size recurrent
+    # state per request instead of growing KV (role of the reference's
+    # MambaSpec, kv_cache_interface.py:710).
+    mamba_d_state: int = 16
+    mamba_d_conv: int = 4
+    mamba_expand: int = 2
+    mamba_dt_rank: int = 0  # 0 -> ceil(hidden_size / 16)
 
     @property
     def is_moe(self) -> bool:
         return self.num_experts > 0
+
+    @property
+    def is_mamba(self) -> bool:
+        return self.architecture == "mamba"
 
     @property
     def is_mla(self) -> bool:
@@ -676,6 +687,37 @@ MODEL_PRESETS: dict[str, ModelSpec] = {
         max_position_embeddings=2048,
         eos_token_id=2,
     ),
+    # state-spaces/mamba-2.8b geometry (HF MambaForCausalLM):
+    # d_model 2560, 64 layers, d_state 16, d_conv 4, expand 2.
+    "mamba-2.8b": ModelSpec(
+        name="mamba-2.8b",
+        architecture="mamba",
+        vocab_size=50280,
+        hidden_size=2560,
+        intermediate_size=0,
+        num_layers=64,
+        num_heads=1,
+        num_kv_heads=1,
+        head_dim=1,
+        max_position_embeddings=1_000_000,  # no positional cap in SSMs
+        tie_word_embeddings=True,
+        eos_token_id=0,
+    ),
+    "tiny-mamba": ModelSpec(
+        name="tiny-mamba",
+        architecture="mamba",
+        vocab_size=1024,
+        hidden_size=64,
+        intermediate_size=0,
+        num_layers=2,
+        num_heads=1,
+        num_kv_heads=1,
+        head_dim=1,
+        max_position_embeddings=2048,
+        mamba_d_state=8,
+        tie_word_embeddings=True,
+        eos_token_id=2,
+    ),
     # Tiny model with the HIP kernels' native geometry (head_dim=128) —
     # used by GPU e2e tests and smoke runs.
     "tiny-llama-128": ModelSpec(
@@ -763,6 +805,30 @@ def _spec_from_hf_config(path: str) -> ModelSpec:
     with open(cfg_path) as f:
         hf = json.load(f)
     arch = (hf.get("architectures") or ["LlamaForCausalLM"])[0].lower()
+    if "mamba" in arch:
+        # HF MambaForCausalLM (state-spaces/mamba-*): SSM geometry only,
+        # attention fields unused.
+        tsr = hf.get("time_step_rank", "auto")
+        return ModelSpec(
+            name=os.path.basename(path.rstrip("/")),
+            architecture="mamba",
+            vocab_size=hf["vocab_size"],
+            hidden_size=hf["hidden_size"],
+            intermediate_size=0,
+            num_layers=hf.get("num_hidden_layers", hf.get("n_layer", 64)),
+            num_heads=1,
+            num_kv_heads=1,
+            head_dim=1,
+            max_position_embeddings=1_000_000,
+            mamba_d_state=hf.get("state_size", 16),
+            mamba_d_conv=hf.get("conv_kernel", 4),
+            mamba_expand=hf.get("expand", 2),
+            mamba_dt_rank=0 if tsr == "auto" else int(tsr),
+            tie_word_embeddings=hf.get("tie_word_embeddings", True),
+            rms_norm_eps=hf.get("layer_norm_epsilon", 1e-5),
+            eos_token_id=hf.get("eos_token_id", 0) or 0,
+            bos_token_id=hf.get("bos_token_id", 0) or 0,
+        )
     if "opt" in arch:
         architecture = "opt"
     elif "mixtral" in arch:

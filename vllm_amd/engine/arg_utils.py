@@ -161,6 +161,20 @@ class EngineArgs:
     def create_engine_config(self) -> EngineConfig:
         import os
 
+        from vllm_amd.config import get_model_spec
+        spec = get_model_spec(self.model)
+        enable_prefix_caching = self.enable_prefix_caching
+        if spec.is_mamba:
+            # SSM state is not content-addressable: a prefix hit would
+            # skip tokens the recurrent state never saw. Every
+            # (re)admission scans from position 0.
+            enable_prefix_caching = False
+            if self.num_speculative_tokens > 0 or self.speculative_model:
+                raise ValueError(
+                    "speculative decoding is not supported for SSM "
+                    "(mamba) models: rejected draft tokens cannot be "
+                    "rolled back out of the recurrent state")
+
         world_size = int(os.environ.get("WORLD_SIZE", "1"))
         pc = ParallelConfig(
             tensor_parallel_size=self.tensor_parallel_size,
@@ -192,7 +206,7 @@ class EngineArgs:
                 block_size=self.block_size,
                 gpu_memory_utilization=self.gpu_memory_utilization,
                 num_gpu_blocks=self.num_gpu_blocks,
-                enable_prefix_caching=self.enable_prefix_caching,
+                enable_prefix_caching=enable_prefix_caching,
                 kv_cache_dtype=self.kv_cache_dtype,
                 cpu_offload_gb=self.cpu_offload_gb,
             ),

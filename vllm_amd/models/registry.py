@@ -29,6 +29,10 @@ def get_model_class(architecture: str):
         from vllm_amd.models.gemma import GemmaForCausalLM
 
         return GemmaForCausalLM
+    if architecture == "mamba":
+        from vllm_amd.models.mamba import MambaForCausalLM
+
+        return MambaForCausalLM
     if architecture == "deepseek":
         from vllm_amd.models.deepseek import DeepseekForCausalLM
 

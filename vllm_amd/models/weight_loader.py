@@ -84,6 +84,12 @@ def load_safetensors_weights(model: torch.nn.Module, config) -> None:
         # RMSNorm kernel stays a plain x̂*w.
         return w + 1 if spec.rmsnorm_unit_offset else w
 
+    if hasattr(model, "load_hf_weights"):
+        # Families with non-llama checkpoint layouts (e.g. mamba's
+        # backbone.*) map their own names.
+        model.load_hf_weights(_iter_safetensors(path, dtype), config)
+        return
+
     # Collect q/k/v and gate/up pieces so fused layers load atomically.
     pending: dict[str, dict[str, torch.Tensor]] = {}
     moe_pending: dict[str, dict[str, torch.Tensor]] = {}

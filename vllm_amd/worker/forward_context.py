@@ -37,6 +37,10 @@ class AttentionMetadata:
     # kv_cache_coordinator.py:60). None = single-group model.
     block_table_w: Optional[torch.Tensor] = None
     slot_mapping_w: Optional[torch.Tensor] = None
+    # SSM (Mamba) models: per-request persistent state row indices, in
+    # this batch's request order (decodes first) — indexes the runner's
+    # conv/ssm state tensors. None for attention models.
+    state_rows: Optional[torch.Tensor] = None
 
 
 @dataclass
@@ -53,6 +57,10 @@ class ForwardContext:
     # this many TP ranks between blocks; 1 = off. The runner pads the
     # token count to a multiple.
     sp_size: int = 1
+    # SSM models: (conv_states [L, rows, d_inner, d_conv-1],
+    # ssm_states [L, rows, d_inner, d_state]) owned by the runner;
+    # None for attention models and during memory profiling.
+    mamba_states: Optional[tuple] = None
 
 
 # Thread-local: serve-level DP replicas run one engine loop per thread

@@ -267,6 +267,11 @@ class ModelRunner:
         # Window-group table for mixed sliding+global models (hybrid KV):
         # same positional indexing, separate physical blocks.
         self.mixed_attn = self.spec.is_mixed_attn
+        # SSM (Mamba) models: per-row recurrent state, allocated with
+        # the KV pool in allocate_kv_cache.
+        self.is_mamba = self.spec.is_mamba
+        self.mamba_conv: Optional[torch.Tensor] = None
+        self.mamba_ssm: Optional[torch.Tensor] = None
         self.np_block_table_w = (np.zeros((n, mb), dtype=np.int32)
                                  if self.mixed_attn else None)
         self.np_nblocks = np.zeros(n, dtype=np.int32)
@@ -398,6 +403,12 @@ class ModelRunner:
 
         spec = self.spec
         elt = torch.tensor([], dtype=self.cache_dtype).element_size()
+        if spec.is_mamba:
+            # No paged KV exists; blocks are pure admission-control
+            # accounting. Price a block like a single-layer KV slab so
+            # the profiled pool lands at a sane size (state tensors are
+            # tiny and allocated separately in allocate_kv_cache).
+            return self.block_size * 2 * spec.hidden_size * elt
         if spec.is_mla:
             # Compressed MLA cache: kv_lora + rope values per token,
             # replicated across TP ranks (vs per-head K+V for GQA).
@@ -415,6 +426,27 @@ class ModelRunner:
 
         self.num_gpu_blocks = num_blocks
         spec = self.spec
+        if spec.is_mamba:
+            # SSM models: constant-size recurrent state per request row
+            # instead of paged KV (reference MambaSpec role). Block
+            # accounting still runs in the scheduler (admission control)
+            # but no block storage exists. One extra scratch row absorbs
+            # padded batch entries. SSM state is fp32 (recurrence
+            # stability); the conv lookback window stays in model dtype.
+            nl = self._num_local_layers()
+            rows = self.config.scheduler_config.max_num_seqs + 1
+            d_inner = spec.mamba_expand * spec.hidden_size
+            self.mamba_conv = torch.zeros(
+                nl, rows, d_inner, spec.mamba_d_conv - 1,
+                dtype=self.dtype, device=self.device)
+            self.mamba_ssm = torch.zeros(
+                nl, rows, d_inner, spec.mamba_d_state,
+                dtype=torch.float32, device=self.device)
+            self.kv_caches = []
+            # Decode stays eager: the state gather/scatter is cheap and
+            # graph capture of dynamic row indices is untested on HW.
+            self.graph_runner = None
+            return
         if spec.is_mla:
             # MLA cache stays at model precision (fp8 MLA cache later).
             per_tok = spec.kv_lora_rank + spec.qk_rope_head_dim
@@ -634,6 +666,11 @@ class ModelRunner:
             )
             row = self._free_rows.pop()
             self._row_of[nr.req_id] = row
+            if self.is_mamba and self.mamba_conv is not None:
+                # Recycled row: fresh requests always scan from position
+                # 0 (prefix caching is off for SSM models).
+                self.mamba_conv[:, row] = 0
+                self.mamba_ssm[:, row] = 0
             nb = len(nr.block_ids)
             self.np_block_table[row, :nb] = nr.block_ids
             if self.mixed_attn and nr.block_ids_w:
@@ -649,6 +686,9 @@ class ModelRunner:
             newb_w = (cr.new_block_ids_w[i]
                       if i < len(cr.new_block_ids_w) else None)
             if cr.resumed[i]:
+                if self.is_mamba and self.mamba_conv is not None:
+                    self.mamba_conv[:, row] = 0
+                    self.mamba_ssm[:, row] = 0
                 state.block_ids = list(cr.new_block_ids[i])
                 state.token_ids = list(cr.new_token_ids[i])
                 state.num_computed_tokens = cr.num_computed_tokens[i]
@@ -689,6 +729,17 @@ class ModelRunner:
             return self._execute_decode(so, list(items.keys()))
         out = self._execute_inner(so)
         return AsyncModelOutput(lambda: out)
+
+    def _state_rows(self, rows: "np.ndarray", np_pad: int,
+                    dev) -> torch.Tensor:
+        """Per-request SSM state row indices for this decode batch;
+        padded entries land on the scratch row (last)."""
+        sr = rows
+        if np_pad:
+            scratch = self.mamba_conv.shape[1] - 1
+            sr = np.concatenate(
+                [sr, np.full(np_pad, scratch, dtype=np.int64)])
+        return torch.from_numpy(np.ascontiguousarray(sr)).to(dev)
 
     # ------------------------------------------------------------------
     @torch.inference_mode()
@@ -807,6 +858,8 @@ class ModelRunner:
                     else None),
                 slot_mapping_w=(torch.from_numpy(slot_mapping_w).to(dev)
                                 if slot_mapping_w is not None else None),
+                state_rows=(self._state_rows(rows, np_pad, dev)
+                            if self.is_mamba else None),
             )
             ids_t = (ids_dev if ids_dev is not None
                      else torch.from_numpy(input_ids).to(dev))
@@ -819,6 +872,8 @@ class ModelRunner:
                 lora_ids=self._lora_ids_tensor(lora_np),
                 lora_manager=self.lora_manager,
                 sp_size=self.sp_size,
+                mamba_states=((self.mamba_conv, self.mamba_ssm)
+                              if self.is_mamba else None),
             )
             with set_forward_context(ctx):
                 hidden = self.model(ids_t, torch.from_numpy(positions).to(dev))
@@ -1058,6 +1113,10 @@ class ModelRunner:
             max_query_len=int(max(n for _, n in items)),
             max_seq_len=max_seq_len,
             num_decodes=num_decodes,
+            state_rows=(torch.tensor(
+                [self._row_of[rid] for rid, _ in items],
+                dtype=torch.int64, device=dev)
+                if self.is_mamba else None),
         )
         input_ids_t = torch.from_numpy(input_ids).to(dev)
         positions_t = torch.from_numpy(positions).to(dev)
@@ -1104,6 +1163,8 @@ class ModelRunner:
             lora_ids=self._lora_ids_tensor(lora_np),
             lora_manager=self.lora_manager,
             mm_embeds=mm_embeds,
+            mamba_states=((self.mamba_conv, self.mamba_ssm)
+                          if self.is_mamba else None),
         )
         if self.pp_size > 1:
             # Stage boundary: recv [T, hidden] from the previous
