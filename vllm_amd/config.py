@@ -98,6 +98,10 @@ class ModelSpec:
     mamba_d_conv: int = 4
     mamba_expand: int = 2
     mamba_dt_rank: int = 0  # 0 -> ceil(hidden_size / 16)
+    # Jamba-style hybrid (architecture "jamba"): layer i is ATTENTION
+    # when i % attn_layer_period == attn_layer_offset, mamba otherwise.
+    attn_layer_period: int = 0
+    attn_layer_offset: int = 0
 
     @property
     def is_moe(self) -> bool:
@@ -106,6 +110,19 @@ class ModelSpec:
     @property
     def is_mamba(self) -> bool:
         return self.architecture == "mamba"
+
+    @property
+    def has_mamba(self) -> bool:
+        """Any SSM state present (pure mamba or jamba hybrid)."""
+        return self.architecture in ("mamba", "jamba")
+
+    def is_attn_layer(self, i: int) -> bool:
+        if self.architecture == "mamba":
+            return False
+        if self.architecture == "jamba":
+            return (self.attn_layer_period > 0
+                    and i % self.attn_layer_period == self.attn_layer_offset)
+        return True
 
     @property
     def is_mla(self) -> bool:
@@ -703,6 +720,25 @@ MODEL_PRESETS: dict[str, ModelSpec] = {
         tie_word_embeddings=True,
         eos_token_id=0,
     ),
+    # Jamba-mini geometry scaled down is impractical offline; this tiny
+    # hybrid (attention at layers 1 and 3, mamba at 0 and 2) covers the
+    # paged-KV + SSM-state coexistence path in CPU tests.
+    "tiny-jamba": ModelSpec(
+        name="tiny-jamba",
+        architecture="jamba",
+        vocab_size=1024,
+        hidden_size=64,
+        intermediate_size=128,
+        num_layers=4,
+        num_heads=4,
+        num_kv_heads=2,
+        head_dim=16,
+        max_position_embeddings=2048,
+        mamba_d_state=8,
+        attn_layer_period=2,
+        attn_layer_offset=1,
+        eos_token_id=2,
+    ),
     "tiny-mamba": ModelSpec(
         name="tiny-mamba",
         architecture="mamba",
@@ -805,6 +841,34 @@ def _spec_from_hf_config(path: str) -> ModelSpec:
     with open(cfg_path) as f:
         hf = json.load(f)
     arch = (hf.get("architectures") or ["LlamaForCausalLM"])[0].lower()
+    if "jamba" in arch:
+        # HF JambaForCausalLM: hybrid attn/mamba layer pattern + SSM
+        # geometry (MoE variants load as dense — tracked follow-up).
+        heads = hf["num_attention_heads"]
+        return ModelSpec(
+            name=os.path.basename(path.rstrip("/")),
+            architecture="jamba",
+            vocab_size=hf["vocab_size"],
+            hidden_size=hf["hidden_size"],
+            intermediate_size=hf["intermediate_size"],
+            num_layers=hf["num_hidden_layers"],
+            num_heads=heads,
+            num_kv_heads=hf.get("num_key_value_heads", heads),
+            head_dim=hf["hidden_size"] // heads,
+            max_position_embeddings=hf.get("max_position_embeddings",
+                                           262144),
+            attn_layer_period=hf.get("attn_layer_period", 8),
+            attn_layer_offset=hf.get("attn_layer_offset", 4),
+            mamba_d_state=hf.get("mamba_d_state", 16),
+            mamba_d_conv=hf.get("mamba_d_conv", 4),
+            mamba_expand=hf.get("mamba_expand", 2),
+            mamba_dt_rank=(0 if hf.get("mamba_dt_rank", "auto") == "auto"
+                           else int(hf["mamba_dt_rank"])),
+            rms_norm_eps=hf.get("rms_norm_eps", 1e-6),
+            tie_word_embeddings=hf.get("tie_word_embeddings", False),
+            eos_token_id=hf.get("eos_token_id", 2) or 2,
+            bos_token_id=hf.get("bos_token_id", 1) or 1,
+        )
     if "mamba" in arch:
         # HF MambaForCausalLM (state-spaces/mamba-*): SSM geometry only,
         # attention fields unused.

@@ -164,7 +164,7 @@ class EngineArgs:
         from vllm_amd.config import get_model_spec
         spec = get_model_spec(self.model)
         enable_prefix_caching = self.enable_prefix_caching
-        if spec.is_mamba:
+        if spec.has_mamba:
             # SSM state is not content-addressable: a prefix hit would
             # skip tokens the recurrent state never saw. Every
             # (re)admission scans from position 0.

@@ -29,6 +29,10 @@ def get_model_class(architecture: str):
         from vllm_amd.models.gemma import GemmaForCausalLM
 
         return GemmaForCausalLM
+    if architecture == "jamba":
+        from vllm_amd.models.jamba import JambaForCausalLM
+
+        return JambaForCausalLM
     if architecture == "mamba":
         from vllm_amd.models.mamba import MambaForCausalLM
 

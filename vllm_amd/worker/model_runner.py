@@ -269,7 +269,7 @@ class ModelRunner:
         self.mixed_attn = self.spec.is_mixed_attn
         # SSM (Mamba) models: per-row recurrent state, allocated with
         # the KV pool in allocate_kv_cache.
-        self.is_mamba = self.spec.is_mamba
+        self.is_mamba = self.spec.has_mamba
         self.mamba_conv: Optional[torch.Tensor] = None
         self.mamba_ssm: Optional[torch.Tensor] = None
         self.np_block_table_w = (np.zeros((n, mb), dtype=np.int32)
@@ -397,6 +397,15 @@ class ModelRunner:
         lo, hi = pp_layer_range(self.spec.num_layers)
         return hi - lo
 
+    def _num_local_kv_layers(self) -> int:
+        """Attention layers in this PP stage's slice (== all layers for
+        non-hybrid models; 0 for pure mamba)."""
+        from vllm_amd.parallel.state import pp_layer_range
+
+        lo, hi = pp_layer_range(self.spec.num_layers)
+        return sum(1 for i in range(lo, hi)
+                   if self.spec.is_attn_layer(i))
+
     def kv_cache_page_bytes(self) -> int:
         """Bytes per KV block across all layers on THIS rank."""
         from vllm_amd.parallel.state import get_tp_world_size
@@ -409,6 +418,10 @@ class ModelRunner:
             # the profiled pool lands at a sane size (state tensors are
             # tiny and allocated separately in allocate_kv_cache).
             return self.block_size * 2 * spec.hidden_size * elt
+        if spec.architecture == "jamba":
+            kv_heads = spec.num_kv_heads_per_rank(get_tp_world_size())
+            return (2 * self.block_size * kv_heads * spec.head_dim * elt
+                    * max(1, self._num_local_kv_layers()))
         if spec.is_mla:
             # Compressed MLA cache: kv_lora + rope values per token,
             # replicated across TP ranks (vs per-head K+V for GQA).
@@ -426,14 +439,15 @@ class ModelRunner:
 
         self.num_gpu_blocks = num_blocks
         spec = self.spec
-        if spec.is_mamba:
-            # SSM models: constant-size recurrent state per request row
-            # instead of paged KV (reference MambaSpec role). Block
-            # accounting still runs in the scheduler (admission control)
-            # but no block storage exists. One extra scratch row absorbs
+        if spec.has_mamba:
+            # SSM state: constant-size recurrent state per request row
+            # (reference MambaSpec role) — the whole cache for pure
+            # mamba, the SSM layers' share for jamba hybrids (whose
+            # attention layers get paged KV below). Block accounting
+            # still runs in the scheduler. One extra scratch row absorbs
             # padded batch entries. SSM state is fp32 (recurrence
             # stability); the conv lookback window stays in model dtype.
-            nl = self._num_local_layers()
+            nl = self._num_local_layers() - self._num_local_kv_layers()
             rows = self.config.scheduler_config.max_num_seqs + 1
             d_inner = spec.mamba_expand * spec.hidden_size
             self.mamba_conv = torch.zeros(
@@ -442,11 +456,12 @@ class ModelRunner:
             self.mamba_ssm = torch.zeros(
                 nl, rows, d_inner, spec.mamba_d_state,
                 dtype=torch.float32, device=self.device)
-            self.kv_caches = []
             # Decode stays eager: the state gather/scatter is cheap and
             # graph capture of dynamic row indices is untested on HW.
             self.graph_runner = None
-            return
+            if spec.is_mamba:
+                self.kv_caches = []
+                return
         if spec.is_mla:
             # MLA cache stays at model precision (fp8 MLA cache later).
             per_tok = spec.kv_lora_rank + spec.qk_rope_head_dim
@@ -457,6 +472,7 @@ class ModelRunner:
             ]
             return
         kv_heads = spec.num_kv_heads_per_rank(get_tp_world_size())
+        n_kv_layers = self._num_local_kv_layers()
         # Head-major block layout: each (block, head) KV tile is one
         # contiguous block_size*head_dim chunk (16 KB at 64x128 bf16) —
         # the unit the HIP attention kernels read/stage.
@@ -465,8 +481,10 @@ class ModelRunner:
                 2, num_blocks, kv_heads, self.block_size, spec.head_dim,
                 dtype=self.cache_dtype, device=self.device,
             )
-            for _ in range(self._num_local_layers())
+            for _ in range(n_kv_layers)
         ]
+        if spec.has_mamba:
+            return  # jamba: eager only (graph_runner cleared above)
         # TP decode graphs require the custom xGMI collectives: RCCL
         # through torch.distributed is not hipGraph-capturable, and the
         # per-layer all-reduce sits inside the captured region.
