@@ -93,3 +93,36 @@ def test_soak_mixed_window_model():
     for t, p in zip(t1, params):
         assert len(t) == p.max_tokens
     assert pre1 > 0, "no pool pressure exercised"
+
+
+def test_soak_jamba_hybrid_state():
+    """Soak shape on the attention+SSM hybrid: paged KV and recurrent
+    state rows under preemption and chunked prefill. Bit-determinism
+    across runs catches state-row leaks (a finished request's recycled
+    row bleeding into a new request) that targeted tests can miss."""
+    def run(seed):
+        rng = np.random.default_rng(seed)
+        llm = LLM(model="tiny-jamba", dtype="fp32", device="cpu",
+                  block_size=16, num_gpu_blocks=24, max_model_len=256,
+                  max_num_batched_tokens=96, max_num_seqs=4,
+                  scheduling_policy="priority")
+        prompts = [[int(x) for x in
+                    rng.integers(3, 900, size=int(rng.integers(4, 100)))]
+                   for _ in range(16)]
+        params = [SamplingParams(
+            temperature=float(rng.choice([0.0, 1.0])),
+            seed=int(rng.integers(0, 2**31)),
+            max_tokens=int(rng.integers(1, 50)), ignore_eos=True,
+            priority=int(rng.integers(0, 3))) for _ in range(16)]
+        outs = llm.generate(prompts, params)
+        sched = llm.engine.engine_core.scheduler
+        pre = sched.num_preemptions_total
+        llm.shutdown()
+        return [o.outputs[0].token_ids for o in outs], pre, params
+
+    t1, pre1, params = run(99)
+    t2, _, _ = run(99)
+    assert t1 == t2
+    for t, p in zip(t1, params):
+        assert len(t) == p.max_tokens
+    assert pre1 > 0, "no pool pressure exercised"
