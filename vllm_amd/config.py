@@ -116,6 +116,11 @@ class ModelSpec:
         """Any SSM state present (pure mamba or jamba hybrid)."""
         return self.architecture in ("mamba", "jamba")
 
+    @property
+    def pooling_only(self) -> bool:
+        """Bidirectional encoder: serves /v1/embeddings only."""
+        return self.architecture == "bert"
+
     def is_attn_layer(self, i: int) -> bool:
         if self.architecture == "mamba":
             return False
@@ -737,6 +742,36 @@ MODEL_PRESETS: dict[str, ModelSpec] = {
         mamba_d_state=8,
         attn_layer_period=2,
         attn_layer_offset=1,
+        eos_token_id=2,
+    ),
+    # bge-base-en-v1.5 geometry (BERT-base): 12 layers, hidden 768.
+    "bge-base": ModelSpec(
+        name="bge-base",
+        architecture="bert",
+        vocab_size=30522,
+        hidden_size=768,
+        intermediate_size=3072,
+        num_layers=12,
+        num_heads=12,
+        num_kv_heads=12,
+        head_dim=64,
+        max_position_embeddings=512,
+        rms_norm_eps=1e-12,
+        eos_token_id=102,
+        bos_token_id=101,
+    ),
+    "tiny-bert": ModelSpec(
+        name="tiny-bert",
+        architecture="bert",
+        vocab_size=1024,
+        hidden_size=64,
+        intermediate_size=128,
+        num_layers=2,
+        num_heads=4,
+        num_kv_heads=4,
+        head_dim=16,
+        max_position_embeddings=512,
+        rms_norm_eps=1e-12,
         eos_token_id=2,
     ),
     "tiny-mamba": ModelSpec(

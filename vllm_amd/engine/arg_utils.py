@@ -164,6 +164,13 @@ class EngineArgs:
         from vllm_amd.config import get_model_spec
         spec = get_model_spec(self.model)
         enable_prefix_caching = self.enable_prefix_caching
+        enable_chunked_prefill = self.enable_chunked_prefill
+        if spec.pooling_only:
+            # Bidirectional attention cannot span prefill chunks, and a
+            # prefix hit would skip positions whose hidden states the
+            # pooler needs: encoders run whole-prompt, uncached.
+            enable_prefix_caching = False
+            enable_chunked_prefill = False
         if spec.has_mamba:
             # SSM state is not content-addressable: a prefix hit would
             # skip tokens the recurrent state never saw. Every
@@ -213,7 +220,7 @@ class EngineArgs:
             scheduler_config=SchedulerConfig(
                 max_num_batched_tokens=self.max_num_batched_tokens,
                 max_num_seqs=self.max_num_seqs,
-                enable_chunked_prefill=self.enable_chunked_prefill,
+                enable_chunked_prefill=enable_chunked_prefill,
                 long_prefill_token_threshold=(
                     self.long_prefill_token_threshold),
                 policy=self.scheduling_policy,

@@ -246,6 +246,18 @@ class LLMEngine:
                 f"prompt length {len(prompt_token_ids)} >= max_model_len "
                 f"{max_len}"
             )
+        spec_cfg = self.config.model_config.spec
+        if spec_cfg.pooling_only:
+            if not params.pooling:
+                raise ValueError(
+                    f"model {spec_cfg.name} is an embedding encoder: "
+                    "only pooling requests are accepted")
+            budget = self.config.scheduler_config.max_num_batched_tokens
+            if len(prompt_token_ids) > budget:
+                raise ValueError(
+                    f"prompt length {len(prompt_token_ids)} exceeds the "
+                    f"encoder's single-pass budget {budget} "
+                    "(bidirectional attention cannot be chunked)")
         eos = self.tokenizer.eos_token_id
         if eos is None:
             eos = self.config.model_config.spec.eos_token_id

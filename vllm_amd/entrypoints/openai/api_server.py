@@ -1035,10 +1035,16 @@ def main() -> None:
     parser.add_argument("--host", type=str, default="0.0.0.0")
     parser.add_argument("--port", type=int, default=8000)
     parser.add_argument("--served-model-name", type=str, default=None)
+    # All listed reasoning models emit the same <think>...</think> span;
+    # the aliases map to one splitter (reference vllm/reasoning/ keeps a
+    # class per name over the same format).
     parser.add_argument("--reasoning-parser", type=str, default=None,
-                        choices=["deepseek_r1"])
+                        choices=["deepseek_r1", "qwen3", "glm45",
+                                 "nemotron"])
     parser.add_argument("--tool-call-parser", type=str, default="hermes",
-                        choices=["hermes", "mistral", "llama3_json"])
+                        choices=["hermes", "qwen", "glm4", "mistral",
+                                 "llama3_json", "pythonic", "granite",
+                                 "internlm2"])
     parser.add_argument("--api-key", type=str, default=None,
                         help="require this bearer token on /v1 routes")
     parser.add_argument("--grpc-port", type=int, default=None,

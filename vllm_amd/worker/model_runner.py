@@ -412,7 +412,7 @@ class ModelRunner:
 
         spec = self.spec
         elt = torch.tensor([], dtype=self.cache_dtype).element_size()
-        if spec.is_mamba:
+        if spec.is_mamba or spec.pooling_only:
             # No paged KV exists; blocks are pure admission-control
             # accounting. Price a block like a single-layer KV slab so
             # the profiled pool lands at a sane size (state tensors are
@@ -439,6 +439,13 @@ class ModelRunner:
 
         self.num_gpu_blocks = num_blocks
         spec = self.spec
+        if spec.pooling_only:
+            # Bidirectional encoders keep no KV at all: every prompt is
+            # encoded whole in one forward (chunking disabled). Blocks
+            # remain admission accounting.
+            self.kv_caches = []
+            self.graph_runner = None
+            return
         if spec.has_mamba:
             # SSM state: constant-size recurrent state per request row
             # (reference MambaSpec role) — the whole cache for pure
