@@ -112,9 +112,11 @@ def test_gates():
 
 @pytest.mark.gpu
 def test_jamba_gpu_smoke():
-    llm = LLM(model="tiny-jamba", dtype="bf16", device="cuda",
-              block_size=16, num_gpu_blocks=64, max_model_len=256,
-              max_num_batched_tokens=256, max_num_seqs=4)
+    # head_dim-128 / block-64 geometry: the attention layers run the
+    # HIP paged kernels, the mamba layers the torch SSM path.
+    llm = LLM(model="tiny-jamba-128", dtype="bf16", device="cuda",
+              block_size=64, num_gpu_blocks=64, max_model_len=2048,
+              max_num_batched_tokens=2048, max_num_seqs=4)
     p = _prompt(42)
     a = _gen(llm, p)
     b = _gen(llm, p)

@@ -805,6 +805,23 @@ MODEL_PRESETS: dict[str, ModelSpec] = {
         max_position_embeddings=2048,
         eos_token_id=2,
     ),
+    # GPU-geometry tiny jamba (head_dim 128 for the HIP decode kernel).
+    "tiny-jamba-128": ModelSpec(
+        name="tiny-jamba-128",
+        architecture="jamba",
+        vocab_size=1024,
+        hidden_size=512,
+        intermediate_size=1024,
+        num_layers=4,
+        num_heads=4,
+        num_kv_heads=2,
+        head_dim=128,
+        max_position_embeddings=2048,
+        mamba_d_state=16,
+        attn_layer_period=2,
+        attn_layer_offset=1,
+        eos_token_id=2,
+    ),
     "tiny-mixtral-128": ModelSpec(
         name="tiny-mixtral-128",
         architecture="mixtral",
