@@ -82,6 +82,9 @@ def _tp_worker(rank: int, world: int, port: int, model: str, q,
     ("tiny-llama", 29611, False),
     ("tiny-mixtral", 29613, False),
     ("tiny-mixtral", 29615, True),   # expert parallelism
+    # Hybrid SSM: attention layers shard across TP, mamba mixers run
+    # replicated (zero extra collectives) — TP2 must equal TP1 exactly.
+    ("tiny-jamba", 29617, False),
 ])
 def test_tp2_spmd_cpu(model, port, ep):
     ctx = mp.get_context("spawn")

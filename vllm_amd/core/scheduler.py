@@ -128,10 +128,17 @@ class Scheduler:
         uniform_window = (spec.sliding_window
                           if spec.sliding_window > 0
                           and spec.global_attn_every_n_layers == 0 else 0)
+        enable_caching = config.cache_config.enable_prefix_caching
+        if spec.has_mamba or spec.pooling_only:
+            # Defense in depth (EngineArgs also clears the flag): SSM
+            # state is not content-addressable and encoders need every
+            # position's hidden state — a prefix hit would skip tokens
+            # those models must actually process.
+            enable_caching = False
         self.kv_cache_manager = KVCacheManager(
             num_gpu_blocks=num_gpu_blocks,
             block_size=config.cache_config.block_size,
-            enable_caching=config.cache_config.enable_prefix_caching,
+            enable_caching=enable_caching,
             num_host_blocks=num_host_blocks,
             sliding_window=uniform_window,
             mixed_window=(spec.sliding_window
