@@ -238,6 +238,11 @@ class LLMEngine:
         else:
             prompt_text = None
             prompt_token_ids = list(prompt)
+        if params.truncate_prompt_tokens is not None:
+            if params.truncate_prompt_tokens < 1:
+                raise ValueError("truncate_prompt_tokens must be >= 1")
+            prompt_token_ids = prompt_token_ids[
+                -params.truncate_prompt_tokens:]
         if not prompt_token_ids:
             raise ValueError("empty prompt")
         max_len = self.config.model_config.max_model_len

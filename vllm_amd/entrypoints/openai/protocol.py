@@ -48,6 +48,8 @@ class CompletionRequest(BaseModel):
     guided_json: Optional[Union[dict, str]] = None
     guided_grammar: Optional[str] = None
     bad_words: Optional[list[str]] = None
+    allowed_token_ids: Optional[list[int]] = None
+    truncate_prompt_tokens: Optional[int] = None
     priority: int = 0  # lower = sooner (priority policy)
     # Extension (same name as the reference): per-prompt-token logprobs.
     prompt_logprobs: Optional[int] = None
@@ -81,6 +83,8 @@ class CompletionRequest(BaseModel):
             guided_json=self.guided_json,
             guided_grammar=self.guided_grammar,
             bad_words=self.bad_words,
+            allowed_token_ids=self.allowed_token_ids,
+            truncate_prompt_tokens=self.truncate_prompt_tokens,
             priority=self.priority,
             output_kind=(RequestOutputKind.DELTA if stream
                          else RequestOutputKind.FINAL_ONLY),
@@ -153,6 +157,8 @@ class ChatCompletionRequest(BaseModel):
     guided_json: Optional[Union[dict, str]] = None
     guided_grammar: Optional[str] = None
     bad_words: Optional[list[str]] = None
+    allowed_token_ids: Optional[list[int]] = None
+    truncate_prompt_tokens: Optional[int] = None
     priority: int = 0  # lower = sooner (priority policy)
     response_format: Optional[dict[str, Any]] = None
     add_generation_prompt: bool = True
@@ -219,6 +225,8 @@ class ChatCompletionRequest(BaseModel):
             guided_json=guided_json,
             guided_grammar=self.guided_grammar,
             bad_words=self.bad_words,
+            allowed_token_ids=self.allowed_token_ids,
+            truncate_prompt_tokens=self.truncate_prompt_tokens,
             priority=self.priority,
             guided_json_object=json_object,
             output_kind=(RequestOutputKind.DELTA if stream

@@ -60,6 +60,9 @@ class SamplingParams:
     spaces_between_special_tokens: bool = True
     logit_bias: Optional[dict[int, float]] = None
     allowed_token_ids: Optional[list[int]] = None
+    # Keep only the LAST N prompt tokens (reference extension of the
+    # same name; applied at admission in llm_engine.add_request).
+    truncate_prompt_tokens: Optional[int] = None
     bad_words: Optional[list[str]] = None
     # Structured output: generation constrained to one of these strings
     # (compiled to a token trie by the engine; see structured_output.py).
