@@ -67,3 +67,111 @@ def test_guided_grammar_api():
         assert r.status_code == 200, r.text
         assert r.json()["choices"][0]["text"] in ("aa", "bb")
     state.engine.shutdown()
+
+
+# ---------------------------------------------------------------------------
+# Recursive grammars (pushdown matcher, guided_ebnf.py)
+
+PAREN_GRAMMAR = 'root ::= expr\nexpr ::= "(" expr ")" | "x"'
+
+LIST_GRAMMAR = '''
+root  ::= list
+list  ::= "[" items "]"
+items ::= "" | value ("," value)*
+value ::= [0-9]+ | list
+'''
+
+
+def _accepts(fsm, s):
+    sid = fsm.start
+    for ch in s:
+        sid = fsm.step(sid, ord(ch))
+        if sid is None:
+            return False
+    return fsm.is_accepting(sid)
+
+
+def test_ebnf_pda_matching():
+    from vllm_amd.guided_ebnf import EbnfFSM
+
+    f = EbnfFSM(PAREN_GRAMMAR)
+    for ok in ["x", "(x)", "(((x)))"]:
+        assert _accepts(f, ok), ok
+    for bad in ["", "()", "((x)", "(x))", "y"]:
+        assert not _accepts(f, bad), bad
+    f2 = EbnfFSM(LIST_GRAMMAR)
+    for ok in ["[]", "[7]", "[12,3]", "[[],[1,[2,33]]]"]:
+        assert _accepts(f2, ok), ok
+    for bad in ["[1,]", "[", "1", "[,1]", "[[]"]:
+        assert not _accepts(f2, bad), bad
+
+
+def test_ebnf_left_recursion_rejected():
+    from vllm_amd.guided_ebnf import EbnfFSM
+
+    for g in ['root ::= root "x" | "y"',
+              'root ::= a\na ::= b "z"\nb ::= a | "w"',
+              'root ::= opt root "x" | "y"\nopt ::= "q"?']:
+        with pytest.raises(GrammarError, match="left-recursive"):
+            EbnfFSM(g)
+
+
+def _balanced(text):
+    depth = 0
+    for ch in text:
+        if ch == "(":
+            depth += 1
+        elif ch == ")":
+            depth -= 1
+            if depth < 0:
+                return False
+    return depth == 0
+
+
+def test_recursive_grammar_e2e():
+    """The engine routes cyclic grammars to the PDA automatically; a
+    sampled (temperature 1) run must still emit a valid derivation —
+    checked with an independent balance/JSON verifier, not the PDA."""
+    import json
+
+    from vllm_amd.entrypoints.llm import LLM
+    from vllm_amd.sampling_params import SamplingParams
+
+    llm = LLM(model="tiny-llama", dtype="fp32", device="cpu",
+              block_size=16, num_gpu_blocks=128, max_model_len=128,
+              max_num_batched_tokens=128, max_num_seqs=2)
+    # A sampled derivation may hit max_tokens mid-grammar (the request
+    # finishes with reason "length" — same as the reference); completed
+    # outputs must be exact derivations, truncated ones valid prefixes.
+    done = 0
+    for seed in range(1, 9):
+        [out] = llm.generate(
+            [[5, 9, 13, 17]],
+            SamplingParams(temperature=1.0, seed=seed, max_tokens=40,
+                           guided_grammar=PAREN_GRAMMAR))
+        o = out.outputs[0]
+        if o.finish_reason == "stop":
+            done += 1
+            assert _balanced(o.text) and o.text.count("x") == 1, o.text
+        else:
+            depth = 0
+            for ch in o.text:  # prefix validity: depth never negative
+                depth += 1 if ch == "(" else -1 if ch == ")" else 0
+                assert depth >= 0, o.text
+    assert done > 0, "no sampled derivation completed in 8 seeds"
+    done = 0
+    for seed in range(1, 9):
+        [out2] = llm.generate(
+            [[6, 10, 14]],
+            SamplingParams(temperature=1.0, seed=seed, max_tokens=60,
+                           # bias toward ']' (byte 93) so the random
+                           # model actually closes lists within budget
+                           logit_bias={93: 6.0},
+                           guided_grammar=LIST_GRAMMAR))
+        o2 = out2.outputs[0]
+        if o2.finish_reason == "stop":
+            done += 1
+            parsed = json.loads(o2.text)  # nested int lists ARE JSON
+            assert isinstance(parsed, list), o2.text
+    assert done > 0, "no sampled list derivation completed in 8 seeds"
+    llm.shutdown()

@@ -308,15 +308,32 @@ class LLMEngine:
             if params.guided_regex:
                 pattern = params.guided_regex
             elif params.guided_grammar:
-                from vllm_amd.guided_grammar import grammar_to_regex
+                from vllm_amd.guided_grammar import (
+                    GrammarError,
+                    grammar_to_regex,
+                )
 
-                pattern = grammar_to_regex(params.guided_grammar)
+                try:
+                    pattern = grammar_to_regex(params.guided_grammar)
+                except GrammarError as e:
+                    if "recursive rule" not in str(e):
+                        raise
+                    # Rule cycles (balanced parens, nested JSON, ...)
+                    # need the pushdown matcher; acyclic grammars stay
+                    # on the cheaper regex DFA.
+                    from vllm_amd.guided_ebnf import EbnfGrammar
+
+                    pattern = None
+                    request.grammar = EbnfGrammar(
+                        params.guided_grammar, self.tokenizer, eos)
             elif params.guided_json is not None \
                     and params.guided_json is not True:
                 pattern = schema_to_regex(params.guided_json)
             else:
                 pattern = any_json_regex()
-            request.grammar = RegexGrammar(pattern, self.tokenizer, eos)
+            if request.grammar is None:
+                request.grammar = RegexGrammar(
+                    pattern, self.tokenizer, eos)
         self.engine_core.add_request(request)
         self.output_processor.add_request(
             request_id, prompt_text, prompt_token_ids, params
