@@ -1,0 +1,163 @@
+"""Whisper-style STT on CPU (reference vllm whisper + /v1/audio
+endpoints): encoder-decoder with per-request cached cross-attention
+states over the paged-KV decoder."""
+
+import base64
+import io
+import struct
+import wave
+
+import numpy as np
+import pytest
+
+from vllm_amd.entrypoints.llm import LLM
+from vllm_amd.sampling_params import SamplingParams
+
+
+def _llm(**kw):
+    return LLM(model="tiny-whisper", dtype="fp32", device="cpu",
+               block_size=16, num_gpu_blocks=64, max_model_len=256,
+               max_num_batched_tokens=kw.pop("mnbt", 256),
+               max_num_seqs=4, **kw)
+
+
+def _wav(seed, secs=0.5, sr=16000):
+    rng = np.random.default_rng(seed)
+    return rng.normal(0, 0.1, size=int(secs * sr)).astype(np.float32)
+
+
+GREEDY = SamplingParams(max_tokens=8, temperature=0.0, ignore_eos=True,
+                       logprobs=1)
+
+
+def _gen(llm, wav, prompt=(3, 4, 5), params=GREEDY):
+    o = llm.generate([{"prompt_token_ids": list(prompt),
+                       "multi_modal_data": {"audio": wav}}], params)[0]
+    out = o.outputs[0]
+    lp = out.logprobs[0][out.token_ids[0]]
+    return out.token_ids, float(getattr(lp, "logprob", lp))
+
+
+def test_audio_reaches_logits_and_deterministic():
+    llm = _llm()
+    t1, lp1 = _gen(llm, _wav(0))
+    t2, lp2 = _gen(llm, _wav(0))
+    t3, lp3 = _gen(llm, _wav(1))
+    llm.shutdown()
+    assert len(t1) == 8
+    assert (t1, lp1) == (t2, lp2)  # deterministic; mm-hash salt safe
+    assert lp1 != lp3              # audio content reaches the decoder
+
+
+def test_chunked_prefill_invariance():
+    wav = _wav(2)
+    prompt = tuple(range(10, 30))
+    big = _llm()
+    whole = _gen(big, wav, prompt)[0]
+    big.shutdown()
+    small = _llm(mnbt=8)
+    chunked = _gen(small, wav, prompt)[0]
+    small.shutdown()
+    assert whole == chunked
+
+
+def test_rejects_audio_on_text_model_and_text_on_whisper_ok():
+    llm = LLM(model="tiny-llama", dtype="fp32", device="cpu",
+              block_size=16, num_gpu_blocks=64, max_model_len=128,
+              max_num_batched_tokens=128, max_num_seqs=2)
+    with pytest.raises(Exception, match="no audio encoder"):
+        llm.generate([{"prompt_token_ids": [5, 6],
+                       "multi_modal_data": {"audio": _wav(3)}}],
+                     SamplingParams(max_tokens=2))
+    llm.shutdown()
+    # Text-only requests on the whisper model run (cross-attn is zero).
+    w = _llm()
+    outs = w.generate([{"prompt_token_ids": [3, 4, 5]}],
+                      SamplingParams(max_tokens=4, temperature=0.0,
+                                     ignore_eos=True))
+    assert len(outs[0].outputs[0].token_ids) == 4
+    w.shutdown()
+
+
+def _wav_bytes(seed, secs=0.4, sr=16000):
+    data = (_wav(seed, secs, sr) * 32767).astype(np.int16)
+    buf = io.BytesIO()
+    with wave.open(buf, "wb") as w:
+        w.setnchannels(1)
+        w.setsampwidth(2)
+        w.setframerate(sr)
+        w.writeframes(data.tobytes())
+    return buf.getvalue()
+
+
+@pytest.fixture(scope="module")
+def client():
+    from fastapi.testclient import TestClient
+
+    from vllm_amd.engine.arg_utils import EngineArgs
+    from vllm_amd.entrypoints.openai.api_server import make_server
+
+    args = EngineArgs(model="tiny-whisper", dtype="fp32", device="cpu",
+                      block_size=16, num_gpu_blocks=64, max_model_len=256,
+                      max_num_batched_tokens=256, max_num_seqs=4)
+    app, state = make_server(args, served_model_name="tiny-whisper")
+    with TestClient(app) as c:
+        yield c
+    state.engine.shutdown()
+
+
+def test_transcriptions_multipart(client):
+    wav = _wav_bytes(10)
+    boundary = "testboundary123"
+    body = b""
+    for name, payload in [("model", b"tiny-whisper"), ("file", wav),
+                          ("response_format", b"json")]:
+        body += (f"--{boundary}\r\nContent-Disposition: form-data; "
+                 f'name="{name}"; filename="a.wav"\r\n\r\n').encode()
+        body += payload + b"\r\n"
+    body += f"--{boundary}--\r\n".encode()
+    r = client.post("/v1/audio/transcriptions", content=body, headers={
+        "content-type": f"multipart/form-data; boundary={boundary}"})
+    assert r.status_code == 200, r.text
+    assert isinstance(r.json()["text"], str)
+
+
+def test_transcriptions_json_base64_and_translations(client):
+    b64 = base64.b64encode(_wav_bytes(11)).decode()
+    r = client.post("/v1/audio/transcriptions", json={
+        "model": "tiny-whisper", "file": b64})
+    assert r.status_code == 200, r.text
+    text_a = r.json()["text"]
+    # Same audio -> same transcription (temperature 0 default).
+    r2 = client.post("/v1/audio/translations", json={
+        "model": "tiny-whisper", "file": b64})
+    assert r2.json()["text"] == text_a
+    # Different audio -> (random-init model) may differ; just succeed.
+    r3 = client.post("/v1/audio/transcriptions", json={
+        "model": "tiny-whisper", "file":
+        base64.b64encode(_wav_bytes(12)).decode(),
+        "response_format": "text"})
+    assert r3.status_code == 200
+    assert r3.headers["content-type"].startswith("text/plain")
+
+
+def test_transcriptions_errors(client):
+    r = client.post("/v1/audio/transcriptions", json={})
+    assert r.status_code == 400
+    r = client.post("/v1/audio/transcriptions", json={
+        "file": base64.b64encode(b"notawav").decode()})
+    assert r.status_code == 400
+    assert "WAV" in r.json()["message"]
+
+
+@pytest.mark.gpu
+def test_whisper_gpu_smoke():
+    # head_dim-64 / block-64 geometry: decoder self-attention runs the
+    # HIP paged kernels; encoder + cross-attention run torch on ROCm.
+    llm = LLM(model="tiny-whisper-64", dtype="bf16", device="cuda",
+              block_size=64, num_gpu_blocks=64, max_model_len=256,
+              max_num_batched_tokens=256, max_num_seqs=4)
+    t1, _ = _gen(llm, _wav(42))
+    t2, _ = _gen(llm, _wav(42))
+    llm.shutdown()
+    assert len(t1) == 8 and t1 == t2

@@ -102,6 +102,12 @@ class ModelSpec:
     # when i % attn_layer_period == attn_layer_offset, mamba otherwise.
     attn_layer_period: int = 0
     attn_layer_offset: int = 0
+    # Whisper-style encoder-decoder (architecture "whisper"): audio
+    # encoder geometry; decoder fields above. 0 layers = no audio.
+    audio_encoder_layers: int = 0
+    audio_mel_bins: int = 80
+    audio_heads: int = 8
+    audio_max_frames: int = 1500
 
     @property
     def is_moe(self) -> bool:
@@ -115,6 +121,10 @@ class ModelSpec:
     def has_mamba(self) -> bool:
         """Any SSM state present (pure mamba or jamba hybrid)."""
         return self.architecture in ("mamba", "jamba")
+
+    @property
+    def is_encoder_decoder(self) -> bool:
+        return self.architecture == "whisper"
 
     @property
     def pooling_only(self) -> bool:
@@ -772,6 +782,66 @@ MODEL_PRESETS: dict[str, ModelSpec] = {
         head_dim=16,
         max_position_embeddings=512,
         rms_norm_eps=1e-12,
+        eos_token_id=2,
+    ),
+    # whisper-base geometry (openai/whisper-base): 512 hidden, 6+6
+    # layers, 8 heads, 80 mel bins.
+    "whisper-base": ModelSpec(
+        name="whisper-base",
+        architecture="whisper",
+        vocab_size=51865,
+        hidden_size=512,
+        intermediate_size=2048,
+        num_layers=6,
+        num_heads=8,
+        num_kv_heads=8,
+        head_dim=64,
+        max_position_embeddings=448,
+        audio_encoder_layers=6,
+        audio_mel_bins=80,
+        audio_heads=8,
+        tie_word_embeddings=True,
+        rms_norm_eps=1e-5,
+        eos_token_id=50257,
+        bos_token_id=50258,
+    ),
+    "tiny-whisper": ModelSpec(
+        name="tiny-whisper",
+        architecture="whisper",
+        vocab_size=1024,
+        hidden_size=64,
+        intermediate_size=128,
+        num_layers=2,
+        num_heads=4,
+        num_kv_heads=4,
+        head_dim=16,
+        max_position_embeddings=448,
+        audio_encoder_layers=2,
+        audio_mel_bins=16,
+        audio_heads=2,
+        audio_max_frames=128,
+        tie_word_embeddings=True,
+        rms_norm_eps=1e-5,
+        eos_token_id=2,
+    ),
+    # GPU-geometry tiny whisper (head_dim 64 for the HIP decode kernel).
+    "tiny-whisper-64": ModelSpec(
+        name="tiny-whisper-64",
+        architecture="whisper",
+        vocab_size=1024,
+        hidden_size=256,
+        intermediate_size=512,
+        num_layers=2,
+        num_heads=4,
+        num_kv_heads=4,
+        head_dim=64,
+        max_position_embeddings=448,
+        audio_encoder_layers=2,
+        audio_mel_bins=16,
+        audio_heads=2,
+        audio_max_frames=128,
+        tie_word_embeddings=True,
+        rms_norm_eps=1e-5,
         eos_token_id=2,
     ),
     "tiny-mamba": ModelSpec(

@@ -213,6 +213,14 @@ class LLMEngine:
                 prompt = list(prompt["prompt_token_ids"])
             else:
                 prompt = prompt["prompt"]
+            if mm_data and mm_data.get("audio") is not None:
+                from vllm_amd.audio import audio_content_hash
+
+                spec = self.config.model_config.spec
+                if spec.audio_encoder_layers == 0:
+                    raise ValueError(
+                        f"model {spec.name} has no audio encoder")
+                mm_hash = audio_content_hash(mm_data)
             if mm_data and mm_data.get("image") is not None:
                 from vllm_amd.multimodal import (expand_image_placeholders,
                                                  mm_content_hash)

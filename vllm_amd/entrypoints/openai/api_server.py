@@ -151,6 +151,62 @@ def build_mm_chat_prompt(engine, messages, add_generation_prompt,
     return {"prompt_token_ids": ids, "multi_modal_data": {"image": image}}
 
 
+def _parse_multipart(body: bytes, content_type: str) -> dict:
+    """Minimal multipart/form-data parser (the environment has no
+    python-multipart): field name -> raw bytes."""
+    import re as _re
+
+    m = _re.search(r'boundary="?([^";,\s]+)"?', content_type)
+    if not m:
+        raise ValueError("multipart body without boundary")
+    delim = b"--" + m.group(1).encode()
+    out: dict[str, bytes] = {}
+    for part in body.split(delim):
+        part = part.strip(b"\r\n")
+        if not part or part == b"--":
+            continue
+        head, sep, data = part.partition(b"\r\n\r\n")
+        if not sep:
+            continue
+        hm = _re.search(rb'name="([^"]+)"', head)
+        if hm:
+            out[hm.group(1).decode()] = data
+    return out
+
+
+def _decode_wav(data: bytes):
+    """WAV (PCM 8/16-bit) -> mono float32 waveform at 16 kHz. Only WAV:
+    compressed formats would need ffmpeg, which this image lacks."""
+    import io
+    import wave as _wave
+
+    import numpy as _np
+
+    try:
+        with _wave.open(io.BytesIO(data)) as w:
+            sr = w.getframerate()
+            nch = w.getnchannels()
+            sw = w.getsampwidth()
+            raw = w.readframes(w.getnframes())
+    except Exception as e:
+        raise ValueError(f"could not decode WAV audio: {e}") from e
+    if sw == 2:
+        arr = _np.frombuffer(raw, dtype=_np.int16).astype(
+            _np.float32) / 32768.0
+    elif sw == 1:
+        arr = (_np.frombuffer(raw, dtype=_np.uint8).astype(_np.float32)
+               - 128.0) / 128.0
+    else:
+        raise ValueError(f"unsupported WAV sample width {sw}")
+    if nch > 1:
+        arr = arr.reshape(-1, nch).mean(axis=1)
+    if sr != 16_000 and len(arr) > 1:
+        n_out = int(len(arr) * 16_000 / sr)
+        arr = _np.interp(_np.linspace(0, len(arr) - 1, n_out),
+                         _np.arange(len(arr)), arr).astype(_np.float32)
+    return arr
+
+
 class ServerState:
     def __init__(self, engine: AsyncLLM, model_name: str,
                  reasoning_parser: Optional[str] = None,
@@ -668,6 +724,72 @@ def build_app(state: ServerState) -> FastAPI:
         return body
 
     # ------------------------------------------------------------------
+    @app.post("/v1/audio/transcriptions")
+    @app.post("/v1/audio/translations")
+    async def audio_transcriptions(raw: Request):
+        """Whisper-style STT (reference /v1/audio endpoints): accepts
+        multipart/form-data with a WAV `file` (parsed in-process — no
+        python-multipart in this image) or JSON {"file": base64-wav}.
+        The decoded waveform rides the engine's multimodal dict-prompt
+        path; the decoder generates over cross-attention."""
+        spec = engine.config.model_config.spec
+        if spec.audio_encoder_layers == 0:
+            return _error(
+                f"model {state.model_name} has no audio encoder")
+        ctype = raw.headers.get("content-type", "")
+        try:
+            if ctype.startswith("multipart/form-data"):
+                fields = _parse_multipart(await raw.body(), ctype)
+                wav_bytes = fields.get("file")
+                if wav_bytes is None:
+                    return _error("missing file field")
+                opts = {k: v.decode(errors="replace")
+                        for k, v in fields.items() if k != "file"}
+            else:
+                import base64
+
+                data = await raw.json()
+                b64 = data.get("file")
+                if not b64:
+                    return _error("missing file field")
+                wav_bytes = base64.b64decode(b64)
+                opts = {k: str(v) for k, v in data.items()
+                        if k != "file"}
+            waveform = _decode_wav(wav_bytes)
+        except ValueError as e:
+            return _error(str(e))
+        max_secs = ((spec.audio_max_frames * 2 - 1) * 160 + 400) / 16000
+        if len(waveform) / 16000 > max_secs:
+            return _error(
+                f"audio longer than {max_secs:.0f}s is not supported")
+        from vllm_amd.sampling_params import SamplingParams
+
+        ids = [spec.bos_token_id]
+        if opts.get("prompt"):
+            ids += engine.tokenizer.encode(opts["prompt"],
+                                           add_special_tokens=False)
+        params = SamplingParams(
+            temperature=float(opts.get("temperature", 0.0)),
+            max_tokens=min(state.max_model_len - len(ids) - 1, 200),
+        )
+        rid = random_id("transcription")
+        state.num_requests += 1
+        final = None
+        try:
+            async for out in engine.generate(
+                    {"prompt_token_ids": ids,
+                     "multi_modal_data": {"audio": waveform}},
+                    params, rid):
+                final = out
+        except ValueError as e:
+            return _error(str(e))
+        text = final.outputs[0].text
+        if opts.get("response_format") == "text":
+            from fastapi.responses import PlainTextResponse
+
+            return PlainTextResponse(text)
+        return {"text": text}
+
     @app.post("/v1/embeddings")
     async def embeddings(req: EmbeddingRequest):
         import asyncio

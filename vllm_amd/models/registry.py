@@ -29,6 +29,10 @@ def get_model_class(architecture: str):
         from vllm_amd.models.gemma import GemmaForCausalLM
 
         return GemmaForCausalLM
+    if architecture == "whisper":
+        from vllm_amd.models.whisper import WhisperForConditionalGeneration
+
+        return WhisperForConditionalGeneration
     if architecture == "bert":
         from vllm_amd.models.bert_embed import BertEmbeddingModel
 

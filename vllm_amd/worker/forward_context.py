@@ -61,6 +61,10 @@ class ForwardContext:
     # ssm_states [L, rows, d_inner, d_state]) owned by the runner;
     # None for attention models and during memory profiling.
     mamba_states: Optional[tuple] = None
+    # Encoder-decoder (whisper): per-request cached audio-encoder
+    # states, in this batch's request order ([T_audio, H] tensors or
+    # None for requests without audio / padded rows).
+    cross_feats: Optional[list] = None
 
 
 # Thread-local: serve-level DP replicas run one engine loop per thread

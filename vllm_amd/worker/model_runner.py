@@ -248,6 +248,7 @@ class ModelRunner:
         self.model: Optional[torch.nn.Module] = None
         self.medusa = None  # MedusaHeads when spec_decode_method=medusa
         self.vision = None  # VisionTower when spec.vision_layers > 0
+        self.audio = None  # AudioEncoder when spec.audio_encoder_layers > 0
         self.eagle = None  # EagleRunnerSide when spec_decode_method=eagle
         self.draft_lm = None  # DraftModelRunnerSide for method=draft
         self.graph_runner: Optional[DecodeGraphRunner] = None
@@ -322,6 +323,15 @@ class ModelRunner:
                 self.dtype)
             self.vision.init_dummy(self.config.model_config.seed)
             self.vision = self.vision.to(self.device).eval()
+        if self.spec.audio_encoder_layers > 0:
+            from vllm_amd.audio import AudioEncoder
+
+            self.audio = AudioEncoder(
+                self.spec.audio_mel_bins, self.spec.hidden_size,
+                self.spec.audio_encoder_layers, self.spec.audio_heads,
+                self.spec.audio_max_frames, self.dtype)
+            self.audio.init_dummy(self.config.model_config.seed)
+            self.audio = self.audio.to(self.device).eval()
         sc = self.config.scheduler_config
         if (sc.num_speculative_tokens > 0
                 and sc.spec_decode_method == "medusa"):
@@ -490,8 +500,11 @@ class ModelRunner:
             )
             for _ in range(n_kv_layers)
         ]
-        if spec.has_mamba:
-            return  # jamba: eager only (graph_runner cleared above)
+        if spec.has_mamba or spec.is_encoder_decoder:
+            # jamba: eager only (graph_runner cleared above).
+            # whisper: the cross-attention loop over per-request encoder
+            # states is not capture-shaped; graph_runner stays None.
+            return
         # TP decode graphs require the custom xGMI collectives: RCCL
         # through torch.distributed is not hipGraph-capturable, and the
         # per-layer all-reduce sits inside the captured region.
@@ -892,6 +905,10 @@ class ModelRunner:
             if np_pad:
                 lora_np = np.concatenate(
                     [lora_np, np.zeros(np_pad, dtype=lora_np.dtype)])
+            cross = None
+            if self.audio is not None:
+                cross = [self.requests[r].mm_feats for r in req_ids]
+                cross += [None] * np_pad
             ctx = ForwardContext(
                 attn_metadata=meta, kv_caches=self.kv_caches,
                 lora_ids=self._lora_ids_tensor(lora_np),
@@ -899,6 +916,7 @@ class ModelRunner:
                 sp_size=self.sp_size,
                 mamba_states=((self.mamba_conv, self.mamba_ssm)
                               if self.is_mamba else None),
+                cross_feats=cross,
             )
             with set_forward_context(ctx):
                 hidden = self.model(ids_t, torch.from_numpy(positions).to(dev))
@@ -1153,6 +1171,24 @@ class ModelRunner:
                 for rid, n_ in items
             ])
         mm_embeds = None
+        cross_feats = None
+        if self.audio is not None:
+            from vllm_amd.audio import log_mel_spectrogram
+
+            cross_feats = []
+            for rid, _nq in items:
+                state = self.requests[rid]
+                if state.mm_data is None or \
+                        state.mm_data.get("audio") is None:
+                    cross_feats.append(None)
+                    continue
+                if state.mm_feats is None:
+                    wav = torch.as_tensor(
+                        state.mm_data["audio"]).to(self.device)
+                    mel = log_mel_spectrogram(
+                        wav, self.spec.audio_mel_bins)
+                    state.mm_feats = self.audio(mel)
+                cross_feats.append(state.mm_feats)
         if self.vision is not None:
             img_id = self.spec.image_token_id
             mm_idx, mm_rows = [], []
@@ -1190,6 +1226,7 @@ class ModelRunner:
             mm_embeds=mm_embeds,
             mamba_states=((self.mamba_conv, self.mamba_ssm)
                           if self.is_mamba else None),
+            cross_feats=cross_feats,
         )
         if self.pp_size > 1:
             # Stage boundary: recv [T, hidden] from the previous
