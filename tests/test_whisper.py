@@ -161,3 +161,39 @@ def test_whisper_gpu_smoke():
     t2, _ = _gen(llm, _wav(42))
     llm.shutdown()
     assert len(t1) == 8 and t1 == t2
+
+
+def test_preemption_and_prefix_cache_with_audio():
+    """Pool pressure forces preemption; resumed requests reuse their
+    cached encoder states and recover prefix blocks (hashes salted by
+    waveform content, so two requests with identical token prompts but
+    different audio never share KV)."""
+    wavs = [_wav(20 + i) for i in range(4)]
+    prompt = list(range(10, 40))  # SAME tokens for all -> salt matters
+    params = SamplingParams(max_tokens=12, temperature=0.0,
+                            ignore_eos=True, logprobs=1)
+
+    def run(blocks):
+        llm = LLM(model="tiny-whisper", dtype="fp32", device="cpu",
+                  block_size=16, num_gpu_blocks=blocks, max_model_len=256,
+                  max_num_batched_tokens=96, max_num_seqs=4)
+        outs = llm.generate(
+            [{"prompt_token_ids": list(prompt),
+              "multi_modal_data": {"audio": w}} for w in wavs], params)
+        pre = llm.engine.engine_core.scheduler.num_preemptions_total
+        llm.shutdown()
+        res = []
+        for o in outs:
+            out = o.outputs[0]
+            lp = out.logprobs[0][out.token_ids[0]]
+            res.append((out.token_ids, float(getattr(lp, "logprob", lp))))
+        return res, pre
+
+    calm, _ = run(64)
+    tight, pre = run(9)
+    assert pre > 0, "no preemption exercised"
+    assert tight == calm
+    # Different audio with identical prompt tokens -> logits differ
+    # (the mm-hash salt kept their KV apart).
+    lps = {lp for _, lp in calm}
+    assert len(lps) > 1, calm
