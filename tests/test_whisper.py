@@ -4,7 +4,6 @@ states over the paged-KV decoder."""
 
 import base64
 import io
-import struct
 import wave
 
 import numpy as np
