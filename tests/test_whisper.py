@@ -196,3 +196,22 @@ def test_preemption_and_prefix_cache_with_audio():
     # (the mm-hash salt kept their KV apart).
     lps = {lp for _, lp in calm}
     assert len(lps) > 1, calm
+
+
+def test_spec_decode_exact_over_cross_attention():
+    """ngram spec decode verifies drafts through the cross-attending
+    decoder; greedy output must be identical with and without spec."""
+    wav = _wav(5)
+    prompt = {"prompt_token_ids": [3, 4, 5, 6, 7, 8, 9, 10],
+              "multi_modal_data": {"audio": wav}}
+    params = SamplingParams(max_tokens=16, temperature=0.0,
+                            ignore_eos=True)
+
+    def run(**kw):
+        llm = _llm(**kw)
+        out = llm.generate([dict(prompt)],
+                           params)[0].outputs[0].token_ids
+        llm.shutdown()
+        return out
+
+    assert run() == run(num_speculative_tokens=3)
