@@ -108,6 +108,9 @@ class ModelSpec:
     audio_mel_bins: int = 80
     audio_heads: int = 8
     audio_max_frames: int = 1500
+    # BART-style TEXT encoder-decoder (architecture "bart"): layers of
+    # the bidirectional text encoder; decoder fields above.
+    encoder_layers: int = 0
 
     @property
     def is_moe(self) -> bool:
@@ -124,7 +127,7 @@ class ModelSpec:
 
     @property
     def is_encoder_decoder(self) -> bool:
-        return self.architecture == "whisper"
+        return self.architecture in ("whisper", "bart")
 
     @property
     def pooling_only(self) -> bool:
@@ -840,6 +843,40 @@ MODEL_PRESETS: dict[str, ModelSpec] = {
         audio_mel_bins=16,
         audio_heads=2,
         audio_max_frames=128,
+        tie_word_embeddings=True,
+        rms_norm_eps=1e-5,
+        eos_token_id=2,
+    ),
+    # bart-large geometry: 1024 hidden, 12+12 layers, 16 heads.
+    "bart-large": ModelSpec(
+        name="bart-large",
+        architecture="bart",
+        vocab_size=50265,
+        hidden_size=1024,
+        intermediate_size=4096,
+        num_layers=12,
+        num_heads=16,
+        num_kv_heads=16,
+        head_dim=64,
+        max_position_embeddings=1024,
+        encoder_layers=12,
+        tie_word_embeddings=True,
+        rms_norm_eps=1e-5,
+        eos_token_id=2,
+        bos_token_id=0,
+    ),
+    "tiny-bart": ModelSpec(
+        name="tiny-bart",
+        architecture="bart",
+        vocab_size=1024,
+        hidden_size=64,
+        intermediate_size=128,
+        num_layers=2,
+        num_heads=4,
+        num_kv_heads=4,
+        head_dim=16,
+        max_position_embeddings=512,
+        encoder_layers=2,
         tie_word_embeddings=True,
         rms_norm_eps=1e-5,
         eos_token_id=2,

@@ -209,10 +209,31 @@ class LLMEngine:
             #  {"image": pixels [3,S,S]}} (reference TextPrompt /
             # TokensPrompt with multi_modal_data).
             mm_data = prompt.get("multi_modal_data")
+            # Encoder-decoder text prompts (reference ExplicitEncoder
+            # DecoderPrompt shape): the encoder prompt rides the same
+            # per-request cached-states path as audio.
+            enc = prompt.get("encoder_prompt_token_ids")
+            if enc is None and prompt.get("encoder_prompt") is not None:
+                enc = self.tokenizer.encode(prompt["encoder_prompt"])
+            if enc is not None:
+                mm_data = dict(mm_data or {})
+                mm_data["encoder_tokens"] = list(enc)
             if "prompt_token_ids" in prompt:
                 prompt = list(prompt["prompt_token_ids"])
             else:
                 prompt = prompt["prompt"]
+            if mm_data and mm_data.get("encoder_tokens") is not None:
+                spec = self.config.model_config.spec
+                if spec.encoder_layers == 0:
+                    raise ValueError(
+                        f"model {spec.name} has no text encoder "
+                        "(encoder_prompt requires an encoder-decoder "
+                        "model)")
+                import hashlib as _hashlib
+
+                mm_hash = int.from_bytes(_hashlib.sha256(
+                    bytes(str(mm_data["encoder_tokens"]),
+                          "utf-8")).digest()[:8], "little")
             if mm_data and mm_data.get("audio") is not None:
                 from vllm_amd.audio import audio_content_hash
 

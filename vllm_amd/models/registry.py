@@ -29,6 +29,10 @@ def get_model_class(architecture: str):
         from vllm_amd.models.gemma import GemmaForCausalLM
 
         return GemmaForCausalLM
+    if architecture == "bart":
+        from vllm_amd.models.bart import BartForConditionalGeneration
+
+        return BartForConditionalGeneration
     if architecture == "whisper":
         from vllm_amd.models.whisper import WhisperForConditionalGeneration
 

@@ -906,7 +906,7 @@ class ModelRunner:
                 lora_np = np.concatenate(
                     [lora_np, np.zeros(np_pad, dtype=lora_np.dtype)])
             cross = None
-            if self.audio is not None:
+            if self.spec.is_encoder_decoder:
                 cross = [self.requests[r].mm_feats for r in req_ids]
                 cross += [None] * np_pad
             ctx = ForwardContext(
@@ -1172,22 +1172,27 @@ class ModelRunner:
             ])
         mm_embeds = None
         cross_feats = None
-        if self.audio is not None:
-            from vllm_amd.audio import log_mel_spectrogram
-
+        if self.spec.is_encoder_decoder:
             cross_feats = []
             for rid, _nq in items:
                 state = self.requests[rid]
-                if state.mm_data is None or \
-                        state.mm_data.get("audio") is None:
+                if state.mm_data is None:
                     cross_feats.append(None)
                     continue
                 if state.mm_feats is None:
-                    wav = torch.as_tensor(
-                        state.mm_data["audio"]).to(self.device)
-                    mel = log_mel_spectrogram(
-                        wav, self.spec.audio_mel_bins)
-                    state.mm_feats = self.audio(mel)
+                    if state.mm_data.get("audio") is not None:
+                        from vllm_amd.audio import log_mel_spectrogram
+
+                        wav = torch.as_tensor(
+                            state.mm_data["audio"]).to(self.device)
+                        mel = log_mel_spectrogram(
+                            wav, self.spec.audio_mel_bins)
+                        state.mm_feats = self.audio(mel)
+                    elif state.mm_data.get("encoder_tokens"):
+                        ids = torch.as_tensor(
+                            state.mm_data["encoder_tokens"],
+                            dtype=torch.int64, device=self.device)
+                        state.mm_feats = self.model.encoder(ids)
                 cross_feats.append(state.mm_feats)
         if self.vision is not None:
             img_id = self.spec.image_token_id
