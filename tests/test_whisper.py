@@ -215,3 +215,46 @@ def test_spec_decode_exact_over_cross_attention():
         return out
 
     assert run() == run(num_speculative_tokens=3)
+
+
+def test_chat_input_audio_content_part():
+    """OpenAI chat with an input_audio content part (base64 WAV) routes
+    through the same engine audio path as /v1/audio/transcriptions."""
+    from fastapi.testclient import TestClient
+
+    from vllm_amd.engine.arg_utils import EngineArgs
+    from vllm_amd.entrypoints.openai.api_server import make_server
+
+    args = EngineArgs(model="tiny-whisper", dtype="fp32", device="cpu",
+                      block_size=16, num_gpu_blocks=64, max_model_len=256,
+                      max_num_batched_tokens=256, max_num_seqs=4)
+    app, state = make_server(args, served_model_name="tiny-whisper")
+    try:
+        with TestClient(app) as c:
+            b64 = base64.b64encode(_wav_bytes(30)).decode()
+
+            def chat(data_b64):
+                return c.post("/v1/chat/completions", json={
+                    "model": "tiny-whisper",
+                    "messages": [{"role": "user", "content": [
+                        {"type": "text", "text": "transcribe"},
+                        {"type": "input_audio",
+                         "input_audio": {"data": data_b64,
+                                         "format": "wav"}}]}],
+                    "max_tokens": 6, "temperature": 0.0,
+                    "ignore_eos": True, "logprobs": True,
+                    "top_logprobs": 1})
+
+            r = chat(b64)
+            assert r.status_code == 200, r.text
+            lp = r.json()["choices"][0]["logprobs"]["content"][0][
+                "logprob"]
+            r2 = chat(base64.b64encode(_wav_bytes(31)).decode())
+            lp2 = r2.json()["choices"][0]["logprobs"]["content"][0][
+                "logprob"]
+            assert lp != lp2  # audio content reaches the logits
+            # Garbage audio payload -> clean 400.
+            r3 = chat("!!!")
+            assert r3.status_code == 400
+    finally:
+        state.engine.shutdown()

@@ -918,7 +918,31 @@ def build_app(state: ServerState) -> FastAPI:
         tools_on = bool(req.tools) and req.tool_choice != "none"
         named = req.named_tool() if tools_on else None
         image_urls = [u for m in req.messages for u in m.image_urls()]
-        if image_urls:
+        audio_ins = [a for m in req.messages for a in m.audio_inputs()]
+        if audio_ins:
+            # OpenAI input_audio content parts (base64 WAV): the decoded
+            # waveform rides the engine's multimodal dict-prompt path;
+            # the templated text becomes the decoder prompt (no
+            # placeholder tokens — the decoder cross-attends).
+            spec = engine.config.model_config.spec
+            if spec.audio_encoder_layers == 0:
+                return _error(f"model {state.model_name} does not "
+                              "support audio input")
+            if len(audio_ins) > 1:
+                return _error("at most one input_audio part per request")
+            import base64
+
+            try:
+                waveform = _decode_wav(
+                    base64.b64decode(audio_ins[0].get("data", "")))
+            except Exception as e:  # noqa: BLE001
+                return _error(f"could not decode input_audio: {e}")
+            text_prompt = apply_chat_template(
+                engine.tokenizer, req.messages, req.add_generation_prompt)
+            prompt = {"prompt_token_ids":
+                      engine.tokenizer.encode(text_prompt),
+                      "multi_modal_data": {"audio": waveform}}
+        elif image_urls:
             try:
                 prompt = build_mm_chat_prompt(
                     engine, req.messages, req.add_generation_prompt,

@@ -114,6 +114,13 @@ class ChatMessage(BaseModel):
                 parts.append(image_sentinel)
         return "".join(parts)
 
+    def audio_inputs(self) -> list[dict]:
+        """input_audio content parts ({"data": b64, "format": ...})."""
+        if not isinstance(self.content, list):
+            return []
+        return [seg.get("input_audio") or {} for seg in self.content
+                if seg.get("type") == "input_audio"]
+
     def image_urls(self) -> list[str]:
         """URLs of image_url content parts, in order of appearance."""
         if not isinstance(self.content, list):
