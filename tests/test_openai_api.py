@@ -760,3 +760,11 @@ def test_truncate_prompt_tokens_and_allowed_ids(client):
     # Mock tokenizer is byte-level: id 42 -> '*', 43 -> '+'. Every
     # sampled token must come from the allowed set.
     assert set(toks) <= {"*", "+"}, toks
+
+
+def test_suffix_rejected(client):
+    r = client.post("/v1/completions", json={
+        "model": "tiny-llama", "prompt": "abc", "suffix": "def",
+        "max_tokens": 2})
+    assert r.status_code == 400
+    assert "suffix" in r.json()["message"]

@@ -442,6 +442,10 @@ def build_app(state: ServerState) -> FastAPI:
             return _error("best_of with streaming is not supported")
         if not 1 <= req.n <= 64 or best_of > 64:
             return _error("n/best_of must be in [1, 64]")
+        if req.suffix:
+            # Same behavior as the reference: insertion mode is not
+            # implemented, reject rather than silently ignore.
+            return _error("suffix is not supported")
         prompts = req.prompt
         if isinstance(prompts, str):
             prompts = [prompts]
