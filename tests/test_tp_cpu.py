@@ -235,3 +235,173 @@ def test_sequence_parallel_tp2_matches_tp1(model, base_port):
                 p.kill()
         results[sp] = res
     assert results[True] == results[False]
+
+
+def _whisper_baseline(q):
+    try:
+        import numpy as np
+
+        from vllm_amd.entrypoints.llm import LLM
+        from vllm_amd.sampling_params import SamplingParams
+
+        llm = LLM(model="tiny-whisper", dtype="fp32", device="cpu",
+                  block_size=16, num_gpu_blocks=64, max_model_len=256,
+                  max_num_batched_tokens=256, max_num_seqs=4)
+        wav = np.random.default_rng(7).normal(
+            0, 0.1, size=8000).astype(np.float32)
+        outs = llm.generate(
+            [{"prompt_token_ids": [3, 4, 5, 6],
+              "multi_modal_data": {"audio": wav}}],
+            SamplingParams(temperature=0.0, max_tokens=8,
+                           ignore_eos=True))
+        llm.shutdown()
+        q.put(("ok", outs[0].outputs[0].token_ids))
+    except Exception as e:  # noqa: BLE001
+        import traceback
+
+        q.put(("err", f"{e}\n{traceback.format_exc()}"))
+
+
+def _whisper_tp_worker(rank: int, world: int, port: int, q):
+    os.environ["RANK"] = str(rank)
+    os.environ["LOCAL_RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    try:
+        import numpy as np
+        import torch
+
+        torch.set_num_threads(1)
+        from vllm_amd.audio import audio_content_hash
+        from vllm_amd.config import (
+            CacheConfig, DeviceConfig, EngineConfig, ModelConfig,
+            ParallelConfig, SchedulerConfig,
+        )
+        from vllm_amd.engine.core import EngineCore
+        from vllm_amd.request import Request
+        from vllm_amd.sampling_params import SamplingParams
+
+        config = EngineConfig(
+            model_config=ModelConfig(model="tiny-whisper", dtype="fp32",
+                                     max_model_len=256),
+            cache_config=CacheConfig(block_size=16, num_gpu_blocks=64),
+            scheduler_config=SchedulerConfig(max_num_batched_tokens=256,
+                                             max_num_seqs=4),
+            parallel_config=ParallelConfig(tensor_parallel_size=world,
+                                           distributed_backend="gloo"),
+            device_config=DeviceConfig(device="cpu"),
+        )
+        engine = EngineCore(config)
+        if rank == 0:
+            wav = np.random.default_rng(7).normal(
+                0, 0.1, size=8000).astype(np.float32)
+            mm = {"audio": wav}
+            engine.add_request(Request(
+                request_id="r0", prompt_token_ids=[3, 4, 5, 6],
+                sampling_params=SamplingParams(
+                    temperature=0.0, max_tokens=8, ignore_eos=True),
+                mm_data=mm, mm_hash=audio_content_hash(mm)))
+            toks = []
+            while engine.has_unfinished_requests():
+                for out in engine.step():
+                    toks.extend(out.new_token_ids)
+            engine.shutdown()
+            q.put(("ok", toks))
+        else:
+            engine.run_spmd_worker_loop()
+            q.put(("ok", None))
+        import torch.distributed as dist
+
+        if dist.is_initialized():
+            dist.destroy_process_group()
+    except Exception as e:  # noqa: BLE001
+        import traceback
+
+        q.put(("err", f"{e}\n{traceback.format_exc()}"))
+
+
+def test_tp2_whisper_cross_attention_matches_tp1():
+    """TP2 over the encoder-decoder: the cross-attention q/kv shards
+    must each hold their own heads' K and V (a naive 2N column split
+    breaks exactly here), so tp2 greedy tokens == single-process."""
+    ctx = mp.get_context("spawn")
+    q0 = ctx.Queue()
+    pb = ctx.Process(target=_whisper_baseline, args=(q0,))
+    pb.start()
+    status, baseline = q0.get(timeout=180)
+    pb.join(timeout=30)
+    assert status == "ok", baseline
+
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_whisper_tp_worker, args=(r, 2, 29641, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    outs = []
+    try:
+        for _ in range(2):
+            outs.append(q.get(timeout=180))
+    finally:
+        for p in procs:
+            p.join(timeout=30)
+            if p.is_alive():
+                p.terminate()
+    for status, payload in outs:
+        assert status == "ok", payload
+    tp_tokens = next(p for s, p in outs if p is not None)
+    assert tp_tokens == baseline
+
+
+def _llama_tp_baseline(q):
+    try:
+        from vllm_amd.entrypoints.llm import LLM
+        from vllm_amd.sampling_params import SamplingParams
+
+        llm = LLM(model="tiny-llama", dtype="fp32", device="cpu",
+                  block_size=16, num_gpu_blocks=64, max_model_len=256,
+                  max_num_batched_tokens=256, max_num_seqs=4)
+        prompts = [[i * 7 + j + 3 for j in range(10)] for i in range(3)]
+        outs = llm.generate(prompts, SamplingParams(
+            temperature=0.0, max_tokens=8, ignore_eos=True))
+        llm.shutdown()
+        q.put(("ok", [o.outputs[0].token_ids for o in outs]))
+    except Exception as e:  # noqa: BLE001
+        import traceback
+
+        q.put(("err", f"{e}\n{traceback.format_exc()}"))
+
+
+def test_tp2_llama_matches_tp1():
+    """Dummy init is TP-partition-invariant (full-shape name-seeded
+    tensors sliced per shard), so tp2 greedy tokens must equal the
+    single-process run exactly — catches any wrong shard mapping in the
+    column/row/QKV splits, not just nondeterminism."""
+    ctx = mp.get_context("spawn")
+    q0 = ctx.Queue()
+    pb = ctx.Process(target=_llama_tp_baseline, args=(q0,))
+    pb.start()
+    status, baseline = q0.get(timeout=180)
+    pb.join(timeout=30)
+    assert status == "ok", baseline
+
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_tp_worker,
+                         args=(r, 2, 29643, "tiny-llama", q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    outs = []
+    try:
+        for _ in range(2):
+            outs.append(q.get(timeout=180))
+    finally:
+        for p in procs:
+            p.join(timeout=30)
+            if p.is_alive():
+                p.terminate()
+    for status, payload in outs:
+        assert status == "ok", payload
+    results = next(p for s, p in outs if p is not None)
+    run0 = {k.split("-", 1)[1]: v for k, v in results[0].items()}
+    assert [run0[f"r{i}"] for i in range(3)] == baseline

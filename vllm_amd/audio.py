@@ -57,16 +57,20 @@ def log_mel_spectrogram(waveform: torch.Tensor,
                         n_mels: int) -> torch.Tensor:
     """waveform [S] float in [-1, 1] -> [n_mels, frames] normalized
     log-mel (Whisper's normalization: clamp to max-8 dB, /4 shift)."""
-    waveform = torch.as_tensor(waveform, dtype=torch.float32).flatten()
-    window = torch.hann_window(N_FFT, device=waveform.device)
+    device = getattr(waveform, "device", torch.device("cpu"))
+    # The STFT runs on CPU regardless of input device: audio is a few
+    # seconds of 1-D samples (microseconds of work) and this keeps the
+    # hot path independent of the FFT backend on the GPU.
+    waveform = torch.as_tensor(waveform, dtype=torch.float32)
+    waveform = waveform.flatten().cpu()
+    window = torch.hann_window(N_FFT)
     stft = torch.stft(waveform, N_FFT, HOP, window=window,
                       center=True, return_complex=True)
     power = stft.abs() ** 2  # [n_freqs, frames]
-    fb = mel_filterbank(n_mels).to(waveform.device)
-    mel = fb @ power
+    mel = mel_filterbank(n_mels) @ power
     log = torch.clamp(mel, min=1e-10).log10()
     log = torch.maximum(log, log.max() - 8.0)
-    return (log + 4.0) / 4.0
+    return ((log + 4.0) / 4.0).to(device)
 
 
 def _sinusoids(length: int, channels: int) -> torch.Tensor:

@@ -61,6 +61,19 @@ class VocabParallelEmbedding(nn.Module):
             )
         self.weight.data.copy_(full_weight.chunk(tp, dim=0)[get_tp_rank()])
 
+    def dummy_shard_shapes(self) -> dict:
+        return {"weight": (self.num_embeddings, self.weight.shape[1])}
+
+    def dummy_shard(self, pname: str, full: torch.Tensor) -> torch.Tensor:
+        tp = get_tp_world_size()
+        if tp == 1:
+            return full
+        pad = self.num_embeddings_padded - full.shape[0]
+        if pad:
+            full = torch.cat(
+                [full, full.new_zeros(pad, full.shape[1])], dim=0)
+        return full.chunk(tp, dim=0)[get_tp_rank()]
+
 
 class ParallelLMHead(VocabParallelEmbedding):
     """LM head sharing the vocab-parallel sharding; logits are gathered."""

@@ -137,6 +137,20 @@ class ColumnParallelLinear(nn.Module):
         tp, rank = get_tp_world_size(), get_tp_rank()
         self.bias.data.copy_(full_bias.chunk(tp, dim=0)[rank])
 
+    # Dummy-init partition invariance: the initializer generates the
+    # FULL-shape tensor (name-seeded) and each rank copies its shard
+    # through the same mapping the checkpoint loaders use, so tp=N
+    # dummy weights are exact slices of the tp=1 weights.
+    def dummy_shard_shapes(self) -> dict:
+        d = {"weight": (self.output_size, self.input_size)}
+        if self.bias is not None:
+            d["bias"] = (self.output_size,)
+        return d
+
+    def dummy_shard(self, pname: str, full: torch.Tensor) -> torch.Tensor:
+        tp, rank = get_tp_world_size(), get_tp_rank()
+        return full.chunk(tp, dim=0)[rank]
+
 
 class MergedColumnParallelLinear(ColumnParallelLinear):
     """Several column-parallel projections fused into one GEMM
@@ -153,6 +167,12 @@ class MergedColumnParallelLinear(ColumnParallelLinear):
         size = self.output_sizes[idx] // tp
         shard = full_weight.chunk(tp, dim=0)[rank]
         self.weight.data[offset : offset + size].copy_(shard)
+
+    def dummy_shard(self, pname: str, full: torch.Tensor) -> torch.Tensor:
+        tp, rank = get_tp_world_size(), get_tp_rank()
+        parts = full.split(self.output_sizes, dim=0)
+        return torch.cat([p.chunk(tp, dim=0)[rank] for p in parts],
+                         dim=0)
 
 
 class QKVParallelLinear(ColumnParallelLinear):
@@ -206,6 +226,26 @@ class QKVParallelLinear(ColumnParallelLinear):
             kb = k_b.chunk(kv_chunks, dim=0)[kv_rank % kv_chunks]
             vb = v_b.chunk(kv_chunks, dim=0)[kv_rank % kv_chunks]
             self.bias.data.copy_(torch.cat([qb, kb, vb], dim=0))
+
+    def dummy_shard_shapes(self) -> dict:
+        rows = (self.total_num_heads
+                + 2 * self.total_num_kv_heads) * self.head_dim
+        d = {"weight": (rows, self.hidden_size)}
+        if self.bias is not None:
+            d["bias"] = (rows,)
+        return d
+
+    def dummy_shard(self, pname: str, full: torch.Tensor) -> torch.Tensor:
+        q_rows = self.total_num_heads * self.head_dim
+        kv_rows = self.total_num_kv_heads * self.head_dim
+        q, k, v = full.split([q_rows, kv_rows, kv_rows], dim=0)
+        tp, rank = get_tp_world_size(), get_tp_rank()
+        kv_chunks = max(
+            self.total_num_kv_heads // max(self.num_kv_heads, 1), 1)
+        kv_rank = (rank // self.num_kv_head_replicas) % kv_chunks
+        return torch.cat([q.chunk(tp, dim=0)[rank],
+                          k.chunk(kv_chunks, dim=0)[kv_rank],
+                          v.chunk(kv_chunks, dim=0)[kv_rank]], dim=0)
 
 
 class RowParallelLinear(nn.Module):
@@ -261,5 +301,17 @@ class RowParallelLinear(nn.Module):
 
     def load_bias(self, full_bias: torch.Tensor) -> None:
         self.bias.data.copy_(full_bias)
+
+    def dummy_shard_shapes(self) -> dict:
+        d = {"weight": (self.output_size, self.input_size)}
+        if self.bias is not None:
+            d["bias"] = (self.output_size,)
+        return d
+
+    def dummy_shard(self, pname: str, full: torch.Tensor) -> torch.Tensor:
+        if pname == "bias":  # bias applied once post-reduce, unsharded
+            return full
+        tp, rank = get_tp_world_size(), get_tp_rank()
+        return full.chunk(tp, dim=1)[rank]
 
     quantize_fp8 = _quantize_fp8

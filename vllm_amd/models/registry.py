@@ -60,17 +60,23 @@ def initialize_dummy_weights(model: torch.nn.Module, seed: int = 0) -> None:
     """Random-init all weights (reference dummy_loader.py:22 pattern) —
     the offline benchmark path; no checkpoints are available.
 
-    Seeded PER PARAMETER NAME (stable crc32), not by iteration order, so
-    a pipeline-parallel stage holding layers [lo, hi) materializes
-    exactly the weights a single-process model would — pp1 and pp2 runs
-    are bit-comparable."""
+    PARTITION-INVARIANT across both parallel axes:
+    - Seeded PER PARAMETER NAME (stable crc32), not by iteration order,
+      so a pipeline-parallel stage holding layers [lo, hi) materializes
+      exactly the weights a single-process model would.
+    - TP-sharded layers expose dummy_shard_shapes()/dummy_shard(): the
+      FULL-shape tensor is generated and each rank copies its slice
+      through the same mapping the checkpoint loaders use — so tp=N
+      dummy weights are exact slices of the tp=1 weights and tp2==tp1
+      greedy outputs are bit-comparable (fp reduction order aside).
+    """
     import math
     import zlib
 
-    for name, param in model.named_parameters():
+    def gen_full(name: str, shape: tuple) -> torch.Tensor:
         gen = torch.Generator()
         gen.manual_seed(seed ^ zlib.crc32(name.encode()))
-        cpu_val = torch.rand(param.shape, generator=gen,
+        cpu_val = torch.rand(shape, generator=gen,
                              dtype=torch.float32) * 2.0 - 1.0
         # Scales chosen so the network is a FUNCTIONING transformer, not
         # a constant map: norm gains at ~1 and matrices at 1/sqrt(fan_in)
@@ -78,15 +84,33 @@ def initialize_dummy_weights(model: torch.nn.Module, seed: int = 0) -> None:
         # every layer), so outputs genuinely depend on inputs — an
         # all-tiny init drives fp32 logits to exactly-uniform and lets
         # equality-style e2e tests pass vacuously.
-        if param.dim() == 1 and "norm" in name.lower():
-            val = 1.0 + 0.05 * cpu_val
-        elif param.dim() == 1:  # biases
-            val = 1e-3 * cpu_val
-        else:
-            fan_in = param.shape[-1] if param.dim() >= 2 else 1
-            if param.dim() > 2 and "patch" in name.lower():
-                fan_in = int(torch.tensor(param.shape[1:]).prod())
-            val = cpu_val / math.sqrt(fan_in)
+        if len(shape) == 1 and "norm" in name.lower():
+            return 1.0 + 0.05 * cpu_val
+        if len(shape) == 1:  # biases
+            return 1e-3 * cpu_val
+        fan_in = shape[-1] if len(shape) >= 2 else 1
+        if len(shape) > 2 and "patch" in name.lower():
+            fan_in = int(torch.tensor(shape[1:]).prod())
+        return cpu_val / math.sqrt(fan_in)
+
+    handled: set[int] = set()
+    for mod_name, mod in model.named_modules():
+        shapes_fn = getattr(mod, "dummy_shard_shapes", None)
+        if shapes_fn is None:
+            continue
+        for pname, full_shape in shapes_fn().items():
+            param = getattr(mod, pname, None)
+            if param is None or id(param) in handled:
+                continue
+            gname = f"{mod_name}.{pname}" if mod_name else pname
+            full = gen_full(gname, tuple(full_shape))
+            param.data.copy_(
+                mod.dummy_shard(pname, full).to(param.dtype))
+            handled.add(id(param))
+    for name, param in model.named_parameters():
+        if id(param) in handled:
+            continue
+        val = gen_full(name, tuple(param.shape))
         param.data.copy_(val.to(param.dtype))
 
 

@@ -30,6 +30,7 @@ from vllm_amd.layers.embedding import ParallelLMHead, VocabParallelEmbedding
 from vllm_amd.layers.layernorm import LayerNorm
 from vllm_amd.layers.linear import (
     ColumnParallelLinear,
+    MergedColumnParallelLinear,
     QKVParallelLinear,
     RowParallelLinear,
 )
@@ -75,9 +76,12 @@ class WhisperCrossAttention(nn.Module):
         self.q_proj = ColumnParallelLinear(
             spec.hidden_size, spec.num_heads * spec.head_dim, bias=True,
             dtype=dtype)
-        self.kv_proj = ColumnParallelLinear(
-            spec.hidden_size, 2 * spec.num_heads * spec.head_dim,
-            bias=False, dtype=dtype)
+        # Merged column-parallel so each rank's shard holds ITS OWN
+        # K and V slices (a single 2N column split would give rank 0
+        # only K columns and rank 1 only V columns under tp=2).
+        n = spec.num_heads * spec.head_dim
+        self.kv_proj = MergedColumnParallelLinear(
+            spec.hidden_size, [n, n], bias=False, dtype=dtype)
         from vllm_amd.parallel.state import get_tp_world_size
 
         self.num_heads = spec.num_heads // max(1, get_tp_world_size())
