@@ -353,12 +353,12 @@ def test_tp2_whisper_cross_attention_matches_tp1():
     assert tp_tokens == baseline
 
 
-def _llama_tp_baseline(q):
+def _exact_tp_baseline(model, q):
     try:
         from vllm_amd.entrypoints.llm import LLM
         from vllm_amd.sampling_params import SamplingParams
 
-        llm = LLM(model="tiny-llama", dtype="fp32", device="cpu",
+        llm = LLM(model=model, dtype="fp32", device="cpu",
                   block_size=16, num_gpu_blocks=64, max_model_len=256,
                   max_num_batched_tokens=256, max_num_seqs=4)
         prompts = [[i * 7 + j + 3 for j in range(10)] for i in range(3)]
@@ -372,14 +372,19 @@ def _llama_tp_baseline(q):
         q.put(("err", f"{e}\n{traceback.format_exc()}"))
 
 
-def test_tp2_llama_matches_tp1():
+@pytest.mark.parametrize("model,port", [
+    ("tiny-llama", 29643),   # plain GQA column/row/QKV splits
+    ("tiny-qwen3", 29645),   # + qkv bias sharding and per-head qk-norm
+    ("tiny-jamba", 29647),   # sharded attention + REPLICATED ssm mixers
+])
+def test_tp2_matches_tp1_exact(model, port):
     """Dummy init is TP-partition-invariant (full-shape name-seeded
     tensors sliced per shard), so tp2 greedy tokens must equal the
     single-process run exactly — catches any wrong shard mapping in the
     column/row/QKV splits, not just nondeterminism."""
     ctx = mp.get_context("spawn")
     q0 = ctx.Queue()
-    pb = ctx.Process(target=_llama_tp_baseline, args=(q0,))
+    pb = ctx.Process(target=_exact_tp_baseline, args=(model, q0))
     pb.start()
     status, baseline = q0.get(timeout=180)
     pb.join(timeout=30)
@@ -387,7 +392,7 @@ def test_tp2_llama_matches_tp1():
 
     q = ctx.Queue()
     procs = [ctx.Process(target=_tp_worker,
-                         args=(r, 2, 29643, "tiny-llama", q))
+                         args=(r, 2, port, model, q))
              for r in range(2)]
     for p in procs:
         p.start()
@@ -402,6 +407,6 @@ def test_tp2_llama_matches_tp1():
                 p.terminate()
     for status, payload in outs:
         assert status == "ok", payload
-    results = next(p for s, p in outs if p is not None)
+    results = next(p for s_, p in outs if p is not None)
     run0 = {k.split("-", 1)[1]: v for k, v in results[0].items()}
     assert [run0[f"r{i}"] for i in range(3)] == baseline
