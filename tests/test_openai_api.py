@@ -768,3 +768,43 @@ def test_suffix_rejected(client):
         "max_tokens": 2})
     assert r.status_code == 400
     assert "suffix" in r.json()["message"]
+
+
+def test_responses_store_background_and_chaining(client):
+    import time as _time
+
+    # store=true (default): result retrievable by id.
+    r = client.post("/v1/responses", json={
+        "model": "tiny-llama", "input": "hello", "max_output_tokens": 6,
+        "temperature": 0.0})
+    assert r.status_code == 200, r.text
+    rid = r.json()["id"]
+    got = client.get(f"/v1/responses/{rid}")
+    assert got.status_code == 200
+    assert got.json()["output_text"] == r.json()["output_text"]
+    # previous_response_id chains the stored output into the context.
+    r2 = client.post("/v1/responses", json={
+        "model": "tiny-llama", "input": "continue",
+        "previous_response_id": rid, "max_output_tokens": 4,
+        "temperature": 0.0})
+    assert r2.status_code == 200, r2.text
+    # background: immediate queued envelope, poll to completion.
+    rb = client.post("/v1/responses", json={
+        "model": "tiny-llama", "input": "bg", "background": True,
+        "max_output_tokens": 4, "temperature": 0.0})
+    assert rb.status_code == 200, rb.text
+    bid = rb.json()["id"]
+    assert rb.json()["status"] in ("queued", "in_progress")
+    for _ in range(100):
+        body = client.get(f"/v1/responses/{bid}").json()
+        if body["status"] == "completed":
+            break
+        _time.sleep(0.05)
+    assert body["status"] == "completed", body
+    assert isinstance(body["output_text"], str)
+    # unknown id -> 404; background without store -> 400.
+    assert client.get("/v1/responses/resp-nope").status_code == 404
+    bad = client.post("/v1/responses", json={
+        "model": "tiny-llama", "input": "x", "background": True,
+        "store": False})
+    assert bad.status_code == 400

@@ -218,6 +218,18 @@ class ServerState:
         self.api_key = api_key
         # hermes | mistral | llama3_json (tool_parser.py formats).
         self.tool_call_parser = tool_call_parser
+        # Stored /v1/responses bodies (bounded LRU) + background tasks.
+        from collections import OrderedDict
+
+        class _LRU(OrderedDict):
+            def __setitem__(self, k, v):
+                super().__setitem__(k, v)
+                self.move_to_end(k)
+                while len(self) > 256:
+                    self.popitem(last=False)
+
+        self.responses_store: dict = _LRU()
+        self.responses_tasks: dict = {}
         # "deepseek_r1" enables <think> splitting into reasoning_content.
         self.reasoning_parser = reasoning_parser
         self.lora_names = list(
@@ -620,6 +632,19 @@ def build_app(state: ServerState) -> FastAPI:
         if req.instructions:
             msgs.append(ChatMessage(role="system",
                                     content=req.instructions))
+        if req.previous_response_id:
+            prev = state.responses_store.get(req.previous_response_id)
+            if prev is None:
+                return _error(
+                    f"response {req.previous_response_id!r} not found",
+                    404)
+            for item in prev.get("output", []):
+                if item.get("type") == "message":
+                    msgs.append(ChatMessage(
+                        role=item.get("role", "assistant"),
+                        content="".join(
+                            c.get("text", "")
+                            for c in item.get("content", []))))
         if isinstance(req.input, str):
             msgs.append(ChatMessage(role="user", content=req.input))
         else:
@@ -658,6 +683,45 @@ def build_app(state: ServerState) -> FastAPI:
                     "status": status,
                     "content": [{"type": "output_text", "text": text,
                                  "annotations": []}]}
+
+        if req.background:
+            if req.stream:
+                return _error(
+                    "background and stream are mutually exclusive")
+            if not req.store:
+                return _error("background requires store=true")
+            state.responses_store[rid] = envelope("queued", [])
+
+            async def run_background():
+                state.responses_store[rid] = envelope("in_progress", [])
+                try:
+                    final = None
+                    async for out in engine.generate(prompt, params, rid):
+                        final = out
+                    comp = final.outputs[0]
+                    usage = {
+                        "input_tokens": len(final.prompt_token_ids),
+                        "output_tokens": len(comp.token_ids),
+                        "total_tokens": len(final.prompt_token_ids)
+                        + len(comp.token_ids)}
+                    body = envelope(
+                        "completed", [msg_item("completed", comp.text)],
+                        usage)
+                    body["output_text"] = comp.text
+                    state.responses_store[rid] = body
+                except Exception as e:  # noqa: BLE001
+                    err = envelope("failed", [])
+                    err["error"] = {"message": str(e)}
+                    state.responses_store[rid] = err
+
+            import asyncio as _asyncio
+
+            task = _asyncio.get_running_loop().create_task(
+                run_background())
+            state.responses_tasks[rid] = task
+            task.add_done_callback(
+                lambda _t: state.responses_tasks.pop(rid, None))
+            return envelope("queued", [])
 
         if req.stream:
             async def gen() -> AsyncGenerator[str, None]:
@@ -725,6 +789,30 @@ def build_app(state: ServerState) -> FastAPI:
         body = envelope("completed", [msg_item("completed", comp.text)],
                         usage)
         body["output_text"] = comp.text
+        if req.store:
+            state.responses_store[rid] = body
+        return body
+
+    @app.get("/v1/responses/{response_id}")
+    async def get_response(response_id: str):
+        body = state.responses_store.get(response_id)
+        if body is None:
+            return _error(f"response {response_id!r} not found", 404)
+        return body
+
+    @app.post("/v1/responses/{response_id}/cancel")
+    async def cancel_response(response_id: str):
+        body = state.responses_store.get(response_id)
+        if body is None:
+            return _error(f"response {response_id!r} not found", 404)
+        if body.get("status") in ("queued", "in_progress"):
+            await engine.abort(response_id)
+            task = state.responses_tasks.get(response_id)
+            if task is not None:
+                task.cancel()
+            body = dict(body)
+            body["status"] = "cancelled"
+            state.responses_store[response_id] = body
         return body
 
     # ------------------------------------------------------------------

@@ -345,6 +345,13 @@ class ResponsesRequest(BaseModel):
     top_p: float = 1.0
     stream: bool = False
     metadata: Optional[dict[str, Any]] = None
+    # Stored-response surface (reference responses API): store lets
+    # GET /v1/responses/{id} retrieve the result; background returns
+    # immediately and generates server-side; previous_response_id
+    # chains onto a stored response's output.
+    store: bool = True
+    background: bool = False
+    previous_response_id: Optional[str] = None
 
 
 class ScoreRequest(BaseModel):
