@@ -142,3 +142,46 @@ def test_text_only_model_rejects_images():
             assert "image" in r.json()["message"]
     finally:
         state.engine.shutdown()
+
+
+def test_anthropic_messages_image_blocks():
+    """Anthropic /v1/messages with base64 image content blocks routes
+    through the same vision path as the OpenAI chat mm prompts."""
+    import base64 as _b64
+    import io as _io
+
+    from PIL import Image
+
+    def png_b64(seed):
+        rng = np.random.default_rng(seed)
+        arr = rng.integers(0, 256, size=(32, 32, 3), dtype=np.uint8)
+        buf = _io.BytesIO()
+        Image.fromarray(arr).save(buf, format="PNG")
+        return _b64.b64encode(buf.getvalue()).decode()
+
+    args = EngineArgs(model="tiny-llava", dtype="fp32", device="cpu",
+                      block_size=16, num_gpu_blocks=128, max_model_len=512,
+                      max_num_batched_tokens=512, max_num_seqs=4)
+    app, state = make_server(args, served_model_name="tiny-llava")
+    try:
+        with TestClient(app) as c:
+            def msg(b64):
+                return c.post("/v1/messages", json={
+                    "model": "tiny-llava", "max_tokens": 6,
+                    "temperature": 0.0,
+                    "messages": [{"role": "user", "content": [
+                        {"type": "text", "text": "look"},
+                        {"type": "image",
+                         "source": {"type": "base64",
+                                    "media_type": "image/png",
+                                    "data": b64}}]}]})
+
+            r = msg(png_b64(0))
+            assert r.status_code == 200, r.text
+            assert r.json()["stop_reason"] in ("max_tokens", "end_turn")
+            # Undecodable image -> clean Anthropic-shaped 400.
+            bad = msg("!!!")
+            assert bad.status_code == 400
+            assert bad.json()["error"]["type"] == "invalid_request_error"
+    finally:
+        state.engine.shutdown()

@@ -19,11 +19,25 @@ class AnthropicMessage(BaseModel):
     role: str
     content: Union[str, list[dict[str, Any]]]
 
-    def text(self) -> str:
+    def text(self, image_sentinel: str = "") -> str:
         if isinstance(self.content, str):
             return self.content
-        return "".join(seg.get("text", "") for seg in self.content
-                       if seg.get("type") == "text")
+        parts = []
+        for seg in self.content:
+            if seg.get("type") == "text":
+                parts.append(seg.get("text", ""))
+            elif seg.get("type") == "image" and image_sentinel:
+                parts.append(image_sentinel)
+        return "".join(parts)
+
+    def image_payloads(self) -> list[str]:
+        """base64 payloads of image content blocks, in order."""
+        if not isinstance(self.content, list):
+            return []
+        return [(seg.get("source") or {}).get("data", "")
+                for seg in self.content
+                if seg.get("type") == "image"
+                and (seg.get("source") or {}).get("type") == "base64"]
 
 
 class MessagesRequest(BaseModel):
@@ -46,7 +60,7 @@ def build_anthropic_router(state) -> APIRouter:
     router = APIRouter()
     engine = state.engine
 
-    def to_prompt(req: MessagesRequest) -> str:
+    def to_prompt(req: MessagesRequest, image_sentinel: str = "") -> str:
         parts = []
         if req.system:
             sys_text = req.system if isinstance(req.system, str) else \
@@ -70,9 +84,47 @@ def build_anthropic_router(state) -> APIRouter:
                         parts.append(
                             "<|tool|>\n"
                             + json.dumps(seg.get("content", "")))
-            parts.append(f"<|{m.role}|>\n{m.text()}")
+            parts.append(f"<|{m.role}|>\n{m.text(image_sentinel)}")
         parts.append("<|assistant|>\n")
         return "\n".join(parts)
+
+    def build_prompt(req: MessagesRequest):
+        """Engine prompt: plain text, or the multimodal dict prompt
+        when messages carry Anthropic base64 image blocks (decoded and
+        spliced as placeholder tokens, same route as the OpenAI chat mm
+        path)."""
+        payloads = [p for m in req.messages for p in m.image_payloads()]
+        if not payloads:
+            return to_prompt(req)
+        import base64
+
+        from vllm_amd.entrypoints.openai.api_server import (
+            _IMG_SENTINEL, decode_image_bytes)
+
+        spec = engine.config.model_config.spec
+        if spec.vision_layers == 0:
+            raise ValueError(
+                f"model {spec.name} does not support image input")
+        text = to_prompt(req, image_sentinel=_IMG_SENTINEL)
+        pieces = text.split(_IMG_SENTINEL)
+        ids = []
+        for i, piece in enumerate(pieces):
+            if piece:
+                ids.extend(engine.tokenizer.encode(
+                    piece, add_special_tokens=(i == 0)))
+            if i < len(pieces) - 1:
+                ids.append(spec.image_token_id)
+        import torch
+
+        try:
+            imgs = [decode_image_bytes(base64.b64decode(p),
+                                       spec.image_size)
+                    for p in payloads]
+        except Exception as e:  # noqa: BLE001
+            raise ValueError(f"could not decode image: {e}") from e
+        image = torch.stack(imgs) if len(imgs) > 1 else imgs[0]
+        return {"prompt_token_ids": ids,
+                "multi_modal_data": {"image": image}}
 
     @router.post("/v1/messages/count_tokens")
     async def count_tokens(req: MessagesRequest):
@@ -93,7 +145,15 @@ def build_anthropic_router(state) -> APIRouter:
             output_kind=(RequestOutputKind.DELTA if req.stream
                          else RequestOutputKind.FINAL_ONLY),
         )
-        prompt = to_prompt(req)
+        try:
+            prompt = build_prompt(req)
+        except ValueError as e:
+            from fastapi.responses import JSONResponse
+
+            return JSONResponse(
+                {"type": "error",
+                 "error": {"type": "invalid_request_error",
+                           "message": str(e)}}, status_code=400)
         lora = req.model if req.model in state.lora_names else None
         rid = f"msg_{int(time.time() * 1e6):x}"
         state.num_requests += 1

@@ -101,7 +101,13 @@ def _decode_image_data_url(url: str, size: int):
     payload = url.split(",", 1)
     if len(payload) != 2:
         raise ValueError("malformed data: URL (missing comma)")
-    raw = base64.b64decode(payload[1])
+    return decode_image_bytes(base64.b64decode(payload[1]), size)
+
+
+def decode_image_bytes(raw: bytes, size: int):
+    """Encoded image bytes -> [3, size, size] float tensor in [-1, 1]."""
+    import io
+
     from PIL import Image
 
     img = Image.open(io.BytesIO(raw)).convert("RGB")
