@@ -120,3 +120,34 @@ def test_multi_image_prompt():
     llm.shutdown()
     assert t_ab == t_ab2 and lp_ab == lp_ab2   # deterministic + cache-safe
     assert lp_ab != lp_ba                       # image ORDER reaches logits
+
+
+def test_encoder_admission_budget():
+    """--max-encoder-tokens-per-step meters how many encoder tokens'
+    worth of encode runs start per step (reference encoder-budget
+    role): with a budget of one image (16 patches), three simultaneous
+    image requests admit across separate steps — outputs unchanged."""
+    rng = np.random.default_rng(9)
+    imgs = [rng.normal(size=(3, 32, 32)).astype(np.float32)
+            for _ in range(3)]
+    prompts = [{"prompt_token_ids": [5, 6, IMG, 7 + i],
+                "multi_modal_data": {"image": img}}
+               for i, img in enumerate(imgs)]
+    params = SamplingParams(max_tokens=4, temperature=0.0,
+                            ignore_eos=True)
+
+    def run(**kw):
+        llm = _llm(**kw)
+        outs = llm.generate([dict(p) for p in prompts], params)
+        defer = llm.engine.engine_core.scheduler.num_encoder_deferrals
+        llm.shutdown()
+        return [o.outputs[0].token_ids for o in outs], defer
+
+    free, d0 = run()
+    metered, d1 = run(max_encoder_tokens_per_step=16)
+    assert d0 == 0
+    assert d1 > 0, "budget never deferred an encode"
+    assert metered == free
+    # A single request larger than the whole budget still admits.
+    big, _ = run(max_encoder_tokens_per_step=4)
+    assert big == free

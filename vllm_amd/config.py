@@ -1222,6 +1222,12 @@ class SchedulerConfig:
     max_num_batched_tokens: int = 8192
     max_num_seqs: int = 256
     enable_chunked_prefill: bool = True
+    # Multimodal encoder admission budget (role of the reference's
+    # encoder budget, scheduler.py:1478 _try_schedule_encoder_inputs):
+    # cap the ENCODER tokens (vision patches / audio frames / encoder
+    # prompt tokens) whose encode runs are started per step. 0 = the
+    # token budget (max_num_batched_tokens).
+    max_encoder_tokens_per_step: int = 0
     # Pipeline CPU scheduling with GPU execution (one step in flight;
     # role of the reference's AsyncScheduler + async model-runner output).
     async_scheduling: bool = True

@@ -159,6 +159,12 @@ class Scheduler:
         self.prefix_cache_queries = 0
         self.prefix_cache_hits = 0
         self.num_preemptions_total = 0
+        # Multimodal encoder admission budget (encoder tokens whose
+        # encode runs start per step; reference encoder-budget role).
+        self.max_encoder_tokens = (
+            sched_cfg.max_encoder_tokens_per_step
+            or sched_cfg.max_num_batched_tokens)
+        self.num_encoder_deferrals = 0
 
     # ------------------------------------------------------------------
     # Request admission / removal
@@ -371,6 +377,7 @@ class Scheduler:
             req_index += 1
 
         # ---- WAITING loop (reference scheduler.py:692) ----
+        encoder_budget = self.max_encoder_tokens
         while (
             self.waiting
             and token_budget > 0
@@ -379,6 +386,16 @@ class Scheduler:
         ):
             request = self.waiting[0]
             resumed = request.status == RequestStatus.PREEMPTED
+            if (request.encoder_tokens > 0
+                    and request.num_computed_tokens == 0
+                    and request.encoder_tokens > encoder_budget
+                    and encoder_budget < self.max_encoder_tokens):
+                # Budget partially spent and this request would start
+                # another encode run: defer to the next step (FCFS order
+                # holds). A request bigger than the WHOLE budget still
+                # admits alone — otherwise it could never run.
+                self.num_encoder_deferrals += 1
+                break
 
             # Prefix-cache lookup for fresh requests.
             new_computed_blocks = []
@@ -424,6 +441,8 @@ class Scheduler:
             self.waiting.popleft()
             self.running.append(request)
             request.status = RequestStatus.RUNNING
+            if request.encoder_tokens > 0 and num_computed == 0:
+                encoder_budget -= request.encoder_tokens
             request.num_computed_tokens = num_computed
 
             all_block_ids = self.kv_cache_manager.get_block_ids(

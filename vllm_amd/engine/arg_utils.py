@@ -39,6 +39,7 @@ class EngineArgs:
     max_num_batched_tokens: int = 8192
     max_num_seqs: int = 256
     enable_chunked_prefill: bool = True
+    max_encoder_tokens_per_step: int = 0
     long_prefill_token_threshold: int = 0
     scheduling_policy: str = "fcfs"
 
@@ -89,6 +90,10 @@ class EngineArgs:
         parser.add_argument("--max-num-batched-tokens", type=int,
                             default=8192)
         parser.add_argument("--max-num-seqs", type=int, default=256)
+        parser.add_argument("--max-encoder-tokens-per-step", type=int,
+                            default=0,
+                            help="cap multimodal encoder tokens started "
+                                 "per step (0 = token budget)")
         parser.add_argument("--no-enable-chunked-prefill",
                             dest="enable_chunked_prefill",
                             action="store_false")
@@ -221,6 +226,8 @@ class EngineArgs:
                 max_num_batched_tokens=self.max_num_batched_tokens,
                 max_num_seqs=self.max_num_seqs,
                 enable_chunked_prefill=enable_chunked_prefill,
+                max_encoder_tokens_per_step=(
+                    self.max_encoder_tokens_per_step),
                 long_prefill_token_threshold=(
                     self.long_prefill_token_threshold),
                 policy=self.scheduling_policy,

@@ -204,6 +204,7 @@ class LLMEngine:
         params = params or SamplingParams()
         mm_data = None
         mm_hash = 0
+        encoder_tokens = 0
         if isinstance(prompt, dict):
             # {"prompt" | "prompt_token_ids", "multi_modal_data":
             #  {"image": pixels [3,S,S]}} (reference TextPrompt /
@@ -234,6 +235,7 @@ class LLMEngine:
                 mm_hash = int.from_bytes(_hashlib.sha256(
                     bytes(str(mm_data["encoder_tokens"]),
                           "utf-8")).digest()[:8], "little")
+                encoder_tokens = len(mm_data["encoder_tokens"])
             if mm_data and mm_data.get("audio") is not None:
                 from vllm_amd.audio import audio_content_hash
 
@@ -242,6 +244,12 @@ class LLMEngine:
                     raise ValueError(
                         f"model {spec.name} has no audio encoder")
                 mm_hash = audio_content_hash(mm_data)
+                import torch as _torch
+
+                n_samples = _torch.as_tensor(
+                    mm_data["audio"]).numel()
+                # conv2 downsamples 2x: encoder rows ~= frames / 2.
+                encoder_tokens = max(1, (1 + n_samples // 160) // 2)
             if mm_data and mm_data.get("image") is not None:
                 from vllm_amd.multimodal import (expand_image_placeholders,
                                                  mm_content_hash)
@@ -261,6 +269,7 @@ class LLMEngine:
                 prompt = expand_image_placeholders(
                     prompt, spec.image_token_id, npatch, nimg)
                 mm_hash = mm_content_hash(mm_data)
+                encoder_tokens = npatch * nimg
         if isinstance(prompt, str):
             prompt_text = prompt
             prompt_token_ids = self.tokenizer.encode(prompt)
@@ -305,6 +314,7 @@ class LLMEngine:
             lora_id=self.config.model_config.lora_id_of(lora),
             mm_data=mm_data,
             mm_hash=mm_hash,
+            encoder_tokens=encoder_tokens,
         )
         if params.bad_words and params._bad_words_token_ids is None:
             # Tokenize both bare and space-prefixed spellings (the

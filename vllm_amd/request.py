@@ -52,6 +52,7 @@ class Request:
         lora_id: int = 0,
         mm_data: Optional[dict] = None,
         mm_hash: int = 0,
+        encoder_tokens: int = 0,
     ) -> None:
         self.request_id = request_id
         self.prompt_token_ids = prompt_token_ids
@@ -65,6 +66,10 @@ class Request:
         # salt prefix-cache block hashes (vllm mm_hash extra_keys role).
         self.mm_data = mm_data
         self.mm_hash = mm_hash
+        # Encoder tokens this request's first scheduled chunk triggers
+        # (vision patches / audio frames / encoder prompt length);
+        # metered by the scheduler's encoder admission budget.
+        self.encoder_tokens = encoder_tokens
 
         self.status = RequestStatus.WAITING
         self.stop_reason: Optional[object] = None
