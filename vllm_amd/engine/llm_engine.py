@@ -166,6 +166,9 @@ class OutputProcessor:
 class LLMEngine:
 
     def __init__(self, config: EngineConfig):
+        from vllm_amd.plugins import load_plugins
+
+        load_plugins()
         self.config = config
         if config.parallel_config.multiprocess_engine:
             from vllm_amd.engine.core_client import EngineCoreClient
