@@ -151,3 +151,18 @@ def test_encoder_admission_budget():
     # A single request larger than the whole budget still admits.
     big, _ = run(max_encoder_tokens_per_step=4)
     assert big == free
+
+
+def test_truncate_rejected_with_images():
+    img = np.random.default_rng(4).normal(size=(3, 32, 32)).astype(
+        np.float32)
+    llm = _llm()
+    try:
+        import pytest
+        with pytest.raises(Exception, match="truncate_prompt_tokens"):
+            llm.generate([{"prompt_token_ids": [5, IMG, 6],
+                           "multi_modal_data": {"image": img}}],
+                         SamplingParams(max_tokens=2,
+                                        truncate_prompt_tokens=4))
+    finally:
+        llm.shutdown()

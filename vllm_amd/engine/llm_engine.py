@@ -282,6 +282,12 @@ class LLMEngine:
         if params.truncate_prompt_tokens is not None:
             if params.truncate_prompt_tokens < 1:
                 raise ValueError("truncate_prompt_tokens must be >= 1")
+            if mm_data and mm_data.get("image") is not None:
+                # Truncation could cut through the expanded per-patch
+                # placeholder spans and desync them from the features.
+                raise ValueError(
+                    "truncate_prompt_tokens is not supported with "
+                    "image prompts")
             prompt_token_ids = prompt_token_ids[
                 -params.truncate_prompt_tokens:]
         if not prompt_token_ids:
