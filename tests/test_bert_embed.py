@@ -117,3 +117,37 @@ def test_bert_gpu_smoke():
     llm.shutdown()
     assert len(a) == 64
     assert a == b
+
+
+def test_score_and_rerank_on_encoder():
+    """/v1/score and /v1/rerank (embedding-similarity path) work over
+    the bidirectional encoder — the cross-encoder-ish serving shape."""
+    from fastapi.testclient import TestClient
+
+    from vllm_amd.engine.arg_utils import EngineArgs
+    from vllm_amd.entrypoints.openai.api_server import make_server
+
+    args = EngineArgs(model="tiny-bert", dtype="fp32", device="cpu",
+                      block_size=16, num_gpu_blocks=64, max_model_len=256,
+                      max_num_batched_tokens=128, max_num_seqs=4)
+    app, state = make_server(args, served_model_name="tiny-bert")
+    try:
+        with TestClient(app) as c:
+            r = c.post("/v1/score", json={
+                "model": "tiny-bert", "text_1": "query text",
+                "text_2": ["doc one", "doc two"]})
+            assert r.status_code == 200, r.text
+            scores = [d["score"] for d in r.json()["data"]]
+            assert len(scores) == 2
+            assert all(-1.0 <= s <= 1.0 for s in scores)
+            r = c.post("/v1/rerank", json={
+                "model": "tiny-bert", "query": "query text",
+                "documents": ["aaa", "bbb", "query text"]})
+            assert r.status_code == 200, r.text
+            results = r.json()["results"]
+            assert len(results) == 3
+            # The identical document must rank first (cosine 1.0).
+            assert results[0]["document"]["text"] == "query text"
+            assert results[0]["relevance_score"] > 0.999
+    finally:
+        state.engine.shutdown()
