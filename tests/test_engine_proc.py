@@ -66,3 +66,42 @@ def test_multiprocess_engine_control_plane(tmp_path):
     again = llm.generate([[3, 4, 5, 6]], p)[0].outputs[0].token_ids
     assert again == before
     llm.shutdown()
+
+
+def test_multiprocess_engine_multimodal_payloads():
+    """Multimodal payloads (pixels / waveforms / encoder prompts) must
+    survive the pickle hop into the engine-core process and produce the
+    same outputs as the in-process engine."""
+    import numpy as np
+
+    rng = np.random.default_rng(6)
+    img = rng.normal(size=(3, 32, 32)).astype(np.float32)
+    p = SamplingParams(temperature=0.0, max_tokens=6, ignore_eos=True)
+    prompt = {"prompt_token_ids": [5, 6, 1000, 7],
+              "multi_modal_data": {"image": img}}
+
+    def run(mp_engine):
+        llm = LLM(model="tiny-llava", dtype="fp32", device="cpu",
+                  block_size=16, num_gpu_blocks=128, max_model_len=512,
+                  max_num_batched_tokens=512, max_num_seqs=4,
+                  multiprocess_engine=mp_engine)
+        out = llm.generate([dict(prompt)], p)[0].outputs[0].token_ids
+        llm.shutdown()
+        return out
+
+    assert run(True) == run(False)
+
+    wav = rng.normal(0, 0.1, size=6000).astype(np.float32)
+    aprompt = {"prompt_token_ids": [3, 4, 5],
+               "multi_modal_data": {"audio": wav}}
+
+    def run_a(mp_engine):
+        llm = LLM(model="tiny-whisper", dtype="fp32", device="cpu",
+                  block_size=16, num_gpu_blocks=64, max_model_len=256,
+                  max_num_batched_tokens=256, max_num_seqs=4,
+                  multiprocess_engine=mp_engine)
+        out = llm.generate([dict(aprompt)], p)[0].outputs[0].token_ids
+        llm.shutdown()
+        return out
+
+    assert run_a(True) == run_a(False)
