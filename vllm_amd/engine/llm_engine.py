@@ -310,6 +310,14 @@ class LLMEngine:
                     f"prompt length {len(prompt_token_ids)} exceeds the "
                     f"encoder's single-pass budget {budget} "
                     "(bidirectional attention cannot be chunked)")
+        if params.logit_bias:
+            vocab = self.config.model_config.spec.vocab_size
+            bad = [t for t in params.logit_bias if not 0 <= t < vocab]
+            if bad:
+                # An out-of-range index would crash the sampler's
+                # index_add_ mid-step and poison the engine loop.
+                raise ValueError(
+                    f"logit_bias token ids out of vocab range: {bad[:5]}")
         eos = self.tokenizer.eos_token_id
         if eos is None:
             eos = self.config.model_config.spec.eos_token_id
@@ -376,7 +384,13 @@ class LLMEngine:
                         params.guided_grammar, self.tokenizer, eos)
             elif params.guided_json is not None \
                     and params.guided_json is not True:
-                pattern = schema_to_regex(params.guided_json)
+                try:
+                    pattern = schema_to_regex(params.guided_json)
+                except ValueError:
+                    raise
+                except Exception as e:  # malformed schema ($ref, types)
+                    raise ValueError(
+                        f"invalid guided_json schema: {e!r}") from e
             else:
                 pattern = any_json_regex()
             if request.grammar is None:

@@ -471,7 +471,10 @@ def build_app(state: ServerState) -> FastAPI:
             prompts = [prompts]
         if not prompts:
             return _error("prompt must not be empty")
-        params = req.to_sampling_params(req.stream)
+        try:
+            params = req.to_sampling_params(req.stream)
+        except ValueError as e:
+            return _error(str(e))
         forced_lp = False
         if best_of > req.n and params.logprobs is None:
             # Branch selection scores by chosen-token logprob; request it
@@ -1055,7 +1058,10 @@ def build_app(state: ServerState) -> FastAPI:
                 tools=req.tools if tools_on and not named else None,
             )
         default_max = state.max_model_len
-        params = req.to_sampling_params(req.stream, default_max)
+        try:
+            params = req.to_sampling_params(req.stream, default_max)
+        except ValueError as e:
+            return _error(str(e))
         lora = req.model if req.model in state.lora_names else None
         state.num_requests += 1
         rid = random_id("chatcmpl")
