@@ -84,6 +84,27 @@ def test_misc_endpoints_never_500(client, body):
         _check(client, url, body)
 
 
+def test_extreme_logprob_counts_and_empty_allowed(client):
+    """Regression: logprobs/prompt_logprobs beyond the vocab crashed
+    the topk mid-step (poisoning the loop); an empty allowed_token_ids
+    would mask every token. Huge counts now mean full-vocab, empty
+    allowed sets 400."""
+    for body, want in (
+        ({"prompt": "x", "max_tokens": 2, "logprobs": 10**6}, 200),
+        ({"prompt": "x", "max_tokens": 2,
+          "prompt_logprobs": 10**6}, 200),
+        ({"prompt": "x", "max_tokens": 2,
+          "allowed_token_ids": []}, 400),
+    ):
+        r = client.post("/v1/completions",
+                        json={"model": "tiny-llama", **body})
+        assert r.status_code == want, (body, r.status_code, r.text[:200])
+        ok = client.post("/v1/completions", json={
+            "model": "tiny-llama", "prompt": "ok", "max_tokens": 2,
+            "ignore_eos": True})
+        assert ok.status_code == 200, (body, ok.text[:200])
+
+
 def test_poison_requests_do_not_kill_the_engine(client):
     """Regression: an out-of-vocab logit_bias crashed the sampler's
     index_add_ mid-step and poisoned the engine loop for every later

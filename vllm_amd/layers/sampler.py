@@ -213,12 +213,14 @@ class Sampler(torch.nn.Module):
         logprobs_out = None
         if meta.max_num_logprobs > 0:
             lp = torch.log_softmax(logits, dim=-1)
-            k = meta.max_num_logprobs
+            # Clamp to vocab: a larger request means "full vocab", and
+            # an unclamped topk would crash the step.
+            k = min(meta.max_num_logprobs, logits.shape[-1])
             topv, topi = lp.topk(k, dim=-1)
             sampled_lp = lp.gather(-1, sampled.unsqueeze(-1))
             logprobs_out = []
             for i in range(B):
-                n = meta.num_logprobs[i]
+                n = min(meta.num_logprobs[i], k)
                 if n == 0:
                     logprobs_out.append(None)
                     continue

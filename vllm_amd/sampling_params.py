@@ -121,6 +121,11 @@ class SamplingParams:
             raise ValueError("max_tokens must be >= 1")
         if self.pooling not in (None, "last", "mean"):
             raise ValueError("pooling must be 'last' or 'mean'")
+        if self.allowed_token_ids is not None \
+                and not self.allowed_token_ids:
+            raise ValueError(
+                "allowed_token_ids must be non-empty when given "
+                "(an empty set would mask every token)")
         if self.min_tokens < 0:
             raise ValueError("min_tokens must be >= 0")
         if self.temperature < _SAMPLING_EPS:

@@ -1266,7 +1266,8 @@ class ModelRunner:
             # {token: logprob} dict per covered prompt position (top-k
             # plus the actual next prompt token, reference convention).
             state = self.requests[rid]
-            k = state.sampling_params.prompt_logprobs
+            k = min(state.sampling_params.prompt_logprobs,
+                    self.spec.vocab_size)  # clamp: topk(k>V) crashes
             s0 = int(query_start_loc[i])
             # positions start..start+n-1 predict tokens start+1..start+n,
             # clipped to prompt tokens only (token 0 has no logprob —
