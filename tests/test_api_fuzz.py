@@ -151,3 +151,10 @@ def test_sampler_and_grammar_payloads_never_500(client, val, field):
     body = {"model": "tiny-llama", "prompt": "x", "max_tokens": 2,
             field: val}
     _check(client, "/v1/completions", body)
+
+
+def test_detokenize_out_of_range_ids(client):
+    r = client.post("/detokenize", json={"tokens": [-5, 99999999]})
+    assert r.status_code == 400
+    r = client.post("/detokenize", json={"tokens": [72, 105]})
+    assert r.status_code == 200

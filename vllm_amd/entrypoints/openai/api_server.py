@@ -301,8 +301,12 @@ def build_app(state: ServerState) -> FastAPI:
                                 max_model_len=state.max_model_len)
 
     @app.post("/detokenize")
-    async def detokenize(req: DetokenizeRequest) -> DetokenizeResponse:
-        return DetokenizeResponse(prompt=engine.tokenizer.decode(req.tokens))
+    async def detokenize(req: DetokenizeRequest):
+        try:
+            text = engine.tokenizer.decode(req.tokens)
+        except Exception as e:  # out-of-range / negative ids
+            return _error(f"could not detokenize: {e}")
+        return DetokenizeResponse(prompt=text)
 
     @app.post("/sleep")
     async def sleep(raw: Request):
