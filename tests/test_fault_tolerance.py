@@ -85,3 +85,49 @@ def test_health_endpoint_503_on_dead_engine():
         proc.join(timeout=10)
         assert c.get("/health").status_code == 503
     state.engine.shutdown()
+
+
+def test_engine_loop_recovers_after_step_crash():
+    """A crashing step must fail the in-flight request AND drain it from
+    the scheduler — the loop recovers for the next request instead of
+    re-crashing forever behind a green /health."""
+    import asyncio
+
+    from vllm_amd.engine.async_llm import AsyncLLM
+    from vllm_amd.engine.arg_utils import EngineArgs
+    from vllm_amd.sampling_params import SamplingParams
+
+    llm = AsyncLLM(EngineArgs(
+        model="tiny-llama", dtype="fp32", device="cpu", block_size=16,
+        num_gpu_blocks=64, max_model_len=128,
+        max_num_batched_tokens=128, max_num_seqs=2
+    ).create_engine_config())
+    real_step = llm.engine.step
+    crashes = {"n": 0}
+
+    def bad_step():
+        crashes["n"] += 1
+        raise RuntimeError("injected step crash")
+
+    async def drive():
+        llm.engine.step = bad_step
+        p = SamplingParams(max_tokens=4, temperature=0.0,
+                           ignore_eos=True)
+        try:
+            async for _ in llm.generate([5, 6, 7], p, "poisoned"):
+                pass
+            raise AssertionError("poisoned request did not error")
+        except RuntimeError:
+            pass
+        llm.engine.step = real_step
+        out = None
+        async for o in llm.generate([8, 9, 10], p, "healthy"):
+            out = o
+        assert len(out.outputs[0].token_ids) == 4
+        llm.check_health()  # loop recovered, engine not marked dead
+
+    try:
+        asyncio.run(drive())
+        assert crashes["n"] == 1, crashes
+    finally:
+        llm.shutdown()

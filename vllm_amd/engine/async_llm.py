@@ -90,10 +90,26 @@ class AsyncLLM:
                     # Unrecoverable: mark errored (health reports it) and
                     # fail all in-flight streams.
                     self._engine_error = e
+                failed = list(self._streams.keys())
                 for rid, (loop, out_q) in self._streams.items():
                     loop.call_soon_threadsafe(out_q.put_nowait, e)
                 self._streams.clear()
                 if self._engine_error is not None:
+                    return
+                # Drain the requests that were in flight when the step
+                # crashed — otherwise a poisoned request re-crashes
+                # every subsequent step and the loop wedges while
+                # /health still reports green.
+                try:
+                    if failed:
+                        self.engine.abort_request(failed)
+                except Exception:  # noqa: BLE001
+                    logger.exception("abort after step failure failed")
+                if self.engine.has_unfinished_requests():
+                    # Could not drain: stop pretending to be healthy.
+                    self._engine_error = EngineDeadError(
+                        "engine step failed and the scheduler could "
+                        "not be drained; engine marked dead")
                     return
                 continue
             for out in outputs:
