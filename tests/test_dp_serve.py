@@ -97,3 +97,26 @@ def test_dp_api_server_end_to_end():
             assert r.json()["usage"]["completion_tokens"] == 4
         assert c.get("/health").status_code == 200
     state.engine.shutdown()
+
+
+def test_dp_router_skips_dead_replica():
+    """A replica whose engine died is routed around — requests keep
+    succeeding on the survivors."""
+    import asyncio
+
+    dp = DPAsyncLLM(_args(2).create_engine_config())
+    try:
+        # Kill replica 0's engine loop.
+        dp.replicas[0]._engine_error = RuntimeError("injected death")
+        p = SamplingParams(max_tokens=4, temperature=0.0,
+                           ignore_eos=True)
+
+        async def drive():
+            for i in range(4):
+                out = None
+                async for o in dp.generate([5 + i, 6, 7], p, f"r{i}"):
+                    out = o
+                assert len(out.outputs[0].token_ids) == 4
+        asyncio.run(drive())
+    finally:
+        dp.shutdown()

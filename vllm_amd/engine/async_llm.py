@@ -269,8 +269,20 @@ class DPAsyncLLM:
             return f"req-{self._counter}"
 
     def _pick(self) -> int:
-        return min(range(len(self.replicas)),
-                   key=lambda i: self._in_flight[i])
+        """Least-loaded HEALTHY replica; a replica whose engine died
+        (check_health raises) is routed around instead of failing every
+        request that lands on it. All-dead falls through to replica 0
+        so the caller gets the real error."""
+        alive = []
+        for i, r in enumerate(self.replicas):
+            try:
+                r.check_health()
+                alive.append(i)
+            except Exception:  # noqa: BLE001
+                continue
+        if not alive:
+            alive = [0]
+        return min(alive, key=lambda i: self._in_flight[i])
 
     async def generate(self, prompt, sampling_params=None,
                        request_id=None, lora=None):
