@@ -334,7 +334,7 @@ def test_tp2_whisper_cross_attention_matches_tp1():
     assert status == "ok", baseline
 
     q = ctx.Queue()
-    procs = [ctx.Process(target=_whisper_tp_worker, args=(r, 2, 29641, q))
+    procs = [ctx.Process(target=_whisper_tp_worker, args=(r, 2, 29657, q))
              for r in range(2)]
     for p in procs:
         p.start()
