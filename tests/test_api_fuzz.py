@@ -158,3 +158,25 @@ def test_detokenize_out_of_range_ids(client):
     assert r.status_code == 400
     r = client.post("/detokenize", json={"tokens": [72, 105]})
     assert r.status_code == 200
+
+
+_anthropic_content = st.one_of(
+    st.text(max_size=6),
+    st.lists(st.dictionaries(
+        st.sampled_from(["type", "text", "source", "content", "id"]),
+        st.one_of(st.text(max_size=6), st.none(),
+                  st.dictionaries(st.sampled_from(
+                      ["type", "data", "media_type"]),
+                      st.text(max_size=8), max_size=3)),
+        max_size=3), max_size=2),
+)
+
+
+@settings(max_examples=40, deadline=None, derandomize=True,
+          suppress_health_check=[HealthCheck.function_scoped_fixture])
+@given(content=_anthropic_content,
+       role=st.sampled_from(["user", "assistant", "tool", "zzz"]))
+def test_anthropic_messages_never_500(client, content, role):
+    body = {"model": "tiny-llama", "max_tokens": 2,
+            "messages": [{"role": role, "content": content}]}
+    _check(client, "/v1/messages", body)
