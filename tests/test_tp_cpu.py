@@ -376,6 +376,7 @@ def _exact_tp_baseline(model, q):
     ("tiny-llama", 29643),   # plain GQA column/row/QKV splits
     ("tiny-qwen3", 29645),   # + qkv bias sharding and per-head qk-norm
     ("tiny-jamba", 29647),   # sharded attention + REPLICATED ssm mixers
+    ("tiny-llama-mqa", 29659),  # tp > num_kv_heads: KV-head replication
 ])
 def test_tp2_matches_tp1_exact(model, port):
     """Dummy init is TP-partition-invariant (full-shape name-seeded

@@ -722,6 +722,23 @@ MODEL_PRESETS: dict[str, ModelSpec] = {
         max_position_embeddings=2048,
         eos_token_id=2,
     ),
+    # MQA tiny model (1 KV head): under tp=2 the KV heads REPLICATE
+    # (tp > num_kv_heads) — covers that branch of QKV sharding and the
+    # partition-invariant dummy init.
+    "tiny-llama-mqa": ModelSpec(
+        name="tiny-llama-mqa",
+        architecture="llama",
+        vocab_size=1024,
+        hidden_size=128,
+        intermediate_size=256,
+        num_layers=2,
+        num_heads=4,
+        num_kv_heads=1,
+        head_dim=32,
+        rope_theta=10000.0,
+        max_position_embeddings=2048,
+        eos_token_id=2,
+    ),
     # state-spaces/mamba-2.8b geometry (HF MambaForCausalLM):
     # d_model 2560, 64 layers, d_state 16, d_conv 4, expand 2.
     "mamba-2.8b": ModelSpec(
