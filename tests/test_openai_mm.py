@@ -185,3 +185,27 @@ def test_anthropic_messages_image_blocks():
             assert bad.json()["error"]["type"] == "invalid_request_error"
     finally:
         state.engine.shutdown()
+
+
+def test_offline_llm_chat_with_images():
+    """LLM.chat() accepts image_url content parts like the server."""
+    from vllm_amd.entrypoints.llm import LLM
+    from vllm_amd.sampling_params import SamplingParams
+
+    llm = LLM(model="tiny-llava", dtype="fp32", device="cpu",
+              block_size=16, num_gpu_blocks=128, max_model_len=512,
+              max_num_batched_tokens=512, max_num_seqs=4)
+    try:
+        msgs = [{"role": "user", "content": [
+            {"type": "text", "text": "describe"},
+            {"type": "image_url", "image_url": {"url": _data_url(7)}}]}]
+        outs = llm.chat(msgs, SamplingParams(
+            max_tokens=4, temperature=0.0, ignore_eos=True))
+        assert len(outs[0].outputs[0].token_ids) == 4
+        # Text-only conversations keep the plain path.
+        outs2 = llm.chat([{"role": "user", "content": "hi"}],
+                         SamplingParams(max_tokens=3, temperature=0.0,
+                                        ignore_eos=True))
+        assert len(outs2[0].outputs[0].token_ids) == 3
+    finally:
+        llm.shutdown()

@@ -126,3 +126,40 @@ def test_soak_jamba_hybrid_state():
     for t, p in zip(t1, params):
         assert len(t) == p.max_tokens
     assert pre1 > 0, "no pool pressure exercised"
+
+
+def test_soak_moe_model():
+    """Soak shape on the MoE model (expert routing + spec decode +
+    preemption + prefix caching together)."""
+    def run(seed):
+        rng = np.random.default_rng(seed)
+        llm = LLM(model="tiny-mixtral", dtype="fp32", device="cpu",
+                  block_size=16, num_gpu_blocks=16, max_model_len=256,
+                  max_num_batched_tokens=96, max_num_seqs=4,
+                  scheduling_policy="priority",
+                  num_speculative_tokens=2)
+        shared = [int(x) for x in rng.integers(3, 900, size=20)]
+        prompts = []
+        for i in range(14):
+            plen = int(rng.integers(4, 90))
+            p = (shared + [int(x) for x in
+                           rng.integers(3, 900, size=max(1, plen - 20))]
+                 if i % 3 == 0 else
+                 [int(x) for x in rng.integers(3, 900, size=plen)])
+            prompts.append(p[:180])
+        params = [SamplingParams(
+            temperature=float(rng.choice([0.0, 1.0])),
+            seed=int(rng.integers(0, 2**31)),
+            max_tokens=int(rng.integers(20, 60)), ignore_eos=True,
+            priority=int(rng.integers(0, 3))) for _ in range(14)]
+        outs = llm.generate(prompts, params)
+        pre = llm.engine.engine_core.scheduler.num_preemptions_total
+        llm.shutdown()
+        return [o.outputs[0].token_ids for o in outs], pre, params
+
+    t1, pre1, params = run(55)
+    t2, _, _ = run(55)
+    assert t1 == t2
+    for t, p in zip(t1, params):
+        assert len(t) == p.max_tokens
+    assert pre1 > 0, "no pool pressure exercised"

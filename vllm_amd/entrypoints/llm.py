@@ -41,14 +41,23 @@ class LLM:
 
         if messages and isinstance(messages[0], dict):
             messages = [messages]
-        prompts = [
-            apply_chat_template(
-                self.engine.tokenizer,
-                [ChatMessage(**m) for m in conv],
-                add_generation_prompt,
-            )
-            for conv in messages
-        ]
+        prompts = []
+        for conv in messages:
+            msgs = [ChatMessage(**m) for m in conv]
+            image_urls = [u for m in msgs for u in m.image_urls()]
+            if image_urls:
+                # Same path as the server's multimodal chat: data: URLs
+                # decoded, placeholder tokens spliced, pixels on the
+                # engine dict prompt.
+                from vllm_amd.entrypoints.openai.api_server import (
+                    build_mm_chat_prompt)
+
+                prompts.append(build_mm_chat_prompt(
+                    self.engine, msgs, add_generation_prompt,
+                    image_urls))
+            else:
+                prompts.append(apply_chat_template(
+                    self.engine.tokenizer, msgs, add_generation_prompt))
         return self.generate(prompts, sampling_params)
 
     def beam_search(
