@@ -300,3 +300,54 @@ def test_decode_fast_path_abort_and_length_cap():
             sampled_token_ids=[[8]] * len(so.num_scheduled_tokens)))
     for r in sched.running:
         assert r.num_computed_tokens <= 48
+
+
+def test_fast_path_bail_releases_partial_allocations():
+    """Regression (found by the MoE soak): when the fast path allocated
+    a boundary block for an early request and then bailed because a
+    later request could not allocate, the early request's block stayed
+    in req_to_blocks without ever reaching the runner — the manager's
+    boundary check then skipped the allocation the runner actually
+    needed, and decode wrote through a stale block-table entry
+    (silent cross-request KV corruption)."""
+    sched = create_scheduler(max_num_seqs=4, max_num_batched_tokens=64,
+                             num_gpu_blocks=5, block_size=16,
+                             enable_prefix_caching=False,
+                             max_model_len=256)
+    # Two requests, each owning 2 full blocks (32 tokens); pool has 1
+    # free block left. Both sit exactly at a block boundary, so the
+    # fast path must allocate for BOTH: the first succeeds (consuming
+    # the last free block), the second fails -> bail.
+    reqs = []
+    for i in range(2):
+        r = make_request(f"b{i}", num_tokens=32, max_tokens=32,
+                         prompt=[3 + i * 5 + j for j in range(32)])
+        sched.add_request(r)
+        reqs.append(r)
+    so = sched.schedule()  # prefill both (2 blocks each)
+    sched.update_from_output(
+        so, fake_runner_output(so, {r.request_id: r for r in reqs}))
+    mgr = sched.kv_cache_manager
+    assert mgr.block_pool.get_num_free_blocks() == 1
+    blocks_before = [len(mgr.req_to_blocks[r.request_id]) for r in reqs]
+    computed_before = [r.num_computed_tokens for r in reqs]
+
+    so2 = sched.schedule()  # decode step: boundary for both -> bail
+    # Whatever path served the step, manager state and the emitted
+    # diffs must agree: every request's runner-visible block count
+    # (prefill blocks + diffs) equals the manager's.
+    cr = so2.scheduled_cached_reqs
+    seen = {rid: nb for rid, nb in zip(
+        cr.req_ids, (len(b) for b in cr.new_block_ids))}
+    for r, nb0 in zip(reqs, blocks_before):
+        known = nb0 + seen.get(r.request_id, 0)
+        mgr_blocks = len(mgr.req_to_blocks.get(r.request_id, []))
+        if r.status == RequestStatus.RUNNING \
+                and r.request_id in seen:
+            assert known == mgr_blocks, (
+                r.request_id, known, mgr_blocks)
+    # No block may be both leaked and unreported: total pool accounting
+    # stays consistent.
+    total_owned = sum(len(b) for b in mgr.req_to_blocks.values())
+    assert total_owned + mgr.block_pool.get_num_free_blocks() == 5
+    del computed_before

@@ -284,6 +284,19 @@ class KVCacheManager:
             self.block_pool.free_blocks(
                 list(reversed(blocks_w[reclaimed_w:])))
 
+    def release_trailing(self, request: Request, n: int) -> None:
+        """Undo the last `n` blocks appended by allocate_slots (fast-path
+        bail rollback): without this the manager keeps blocks the runner
+        was never told about, and later boundary checks think the
+        request already owns them — the runner then writes through a
+        stale zero entry in its block table."""
+        if n <= 0:
+            return
+        blocks = self.req_to_blocks[request.request_id]
+        tail = blocks[len(blocks) - n:]
+        del blocks[len(blocks) - n:]
+        self.block_pool.free_blocks(list(reversed(tail)))
+
     def get_block_ids(self, request_id: str) -> list[int]:
         return [b.block_id for b in self.req_to_blocks[request_id]]
 

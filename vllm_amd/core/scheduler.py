@@ -269,16 +269,21 @@ class Scheduler:
             r.num_computed_tokens = total
             if total == r.num_tokens + r.num_output_placeholders:
                 r.num_output_placeholders += 1
-            committed.append(r)
+            committed.append((r, len(ids)))
         if not ok:
-            # Roll back and let the general path preempt.
-            for r in committed:
+            # Roll back and let the general path preempt. Blocks
+            # allocated in this failed pass MUST be released too:
+            # leaving them in req_to_blocks makes the manager think the
+            # request owns them while the runner was never told — its
+            # stale block table then silently corrupts block 0.
+            for r, n_alloc in committed:
                 r.num_computed_tokens -= 1
                 # The placeholder was bumped iff (restored computed + 2)
                 # == tokens + bumped placeholders.
                 if (r.num_computed_tokens + 2
                         == r.num_tokens + r.num_output_placeholders):
                     r.num_output_placeholders -= 1
+                mgr.release_trailing(r, n_alloc)
             return None
         n = len(running)
         out = SchedulerOutput(
