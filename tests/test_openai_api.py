@@ -808,3 +808,11 @@ def test_responses_store_background_and_chaining(client):
         "model": "tiny-llama", "input": "x", "background": True,
         "store": False})
     assert bad.status_code == 400
+
+
+def test_model_retrieval_and_ping(client):
+    r = client.get("/v1/models/tiny-llama")
+    assert r.status_code == 200 and r.json()["id"] == "tiny-llama"
+    assert client.get("/v1/models/nope").status_code == 404
+    assert client.get("/ping").status_code == 200
+    assert client.post("/ping").status_code == 200

@@ -294,6 +294,23 @@ def build_app(state: ServerState) -> FastAPI:
                   for n in state.lora_names]
         return ModelList(data=cards)
 
+    @app.get("/v1/models/{model_id}")
+    async def get_model(model_id: str):
+        if model_id == state.model_name or model_id in state.lora_names:
+            return ModelCard(id=model_id,
+                             max_model_len=state.max_model_len)
+        return _error(f"model {model_id!r} not found", 404)
+
+    @app.get("/ping")
+    @app.post("/ping")
+    async def ping():
+        # SageMaker-style liveness alias for /health (reference parity).
+        try:
+            engine.check_health()
+        except Exception as e:  # noqa: BLE001
+            return _error(str(e), 503)
+        return Response(status_code=200)
+
     @app.post("/tokenize")
     async def tokenize(req: TokenizeRequest) -> TokenizeResponse:
         ids = engine.tokenizer.encode(req.prompt)
