@@ -131,3 +131,58 @@ def test_engine_loop_recovers_after_step_crash():
         assert crashes["n"] == 1, crashes
     finally:
         llm.shutdown()
+
+
+def test_concurrent_streams_with_disconnects():
+    """Many concurrent streaming requests; half the clients disconnect
+    mid-stream. Survivors must complete with the full token count and
+    the engine must drain to zero unfinished requests (aborts release
+    scheduler state and KV)."""
+    import asyncio
+
+    from vllm_amd.engine.arg_utils import EngineArgs
+    from vllm_amd.engine.async_llm import AsyncLLM
+    from vllm_amd.sampling_params import SamplingParams
+
+    llm = AsyncLLM(EngineArgs(
+        model="tiny-llama", dtype="fp32", device="cpu", block_size=16,
+        num_gpu_blocks=96, max_model_len=256,
+        max_num_batched_tokens=128, max_num_seqs=6
+    ).create_engine_config())
+
+    async def client(i, cancel_after):
+        p = SamplingParams(max_tokens=24, temperature=0.0,
+                           ignore_eos=True)
+        got = 0
+        gen = llm.generate([3 + i] * 12, p, f"c{i}")
+        try:
+            async for out in gen:
+                got = len(out.outputs[0].token_ids)
+                if cancel_after and got >= cancel_after:
+                    return ("cancelled", got)
+            return ("done", got)
+        finally:
+            await gen.aclose()
+
+    async def drive():
+        tasks = [client(i, cancel_after=(4 if i % 2 else 0))
+                 for i in range(10)]
+        results = await asyncio.gather(*tasks)
+        for i, (status, got) in enumerate(results):
+            if i % 2:
+                assert status == "cancelled" and got >= 4, (i, results)
+            else:
+                assert status == "done" and got == 24, (i, results)
+        # Aborts must drain: no zombie requests holding KV.
+        for _ in range(100):
+            if not llm.engine.has_unfinished_requests():
+                break
+            await asyncio.sleep(0.05)
+        assert not llm.engine.has_unfinished_requests()
+        mgr = llm.engine.engine_core.scheduler.kv_cache_manager
+        assert not mgr.req_to_blocks, list(mgr.req_to_blocks)
+
+    try:
+        asyncio.run(drive())
+    finally:
+        llm.shutdown()
