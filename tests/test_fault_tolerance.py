@@ -186,3 +186,61 @@ def test_concurrent_streams_with_disconnects():
         asyncio.run(drive())
     finally:
         llm.shutdown()
+
+
+def test_aborts_interleaved_with_preemption():
+    """Clients cancel while pool pressure preempts others (and spec
+    decode + priority scheduling run): survivors complete exactly, and
+    after the dust settles the KV manager holds zero request state."""
+    import asyncio
+
+    import numpy as np
+
+    from vllm_amd.engine.arg_utils import EngineArgs
+    from vllm_amd.engine.async_llm import AsyncLLM
+    from vllm_amd.sampling_params import SamplingParams
+
+    llm = AsyncLLM(EngineArgs(
+        model="tiny-llama", dtype="fp32", device="cpu", block_size=16,
+        num_gpu_blocks=20, max_model_len=256,
+        max_num_batched_tokens=80, max_num_seqs=5,
+        scheduling_policy="priority",
+        num_speculative_tokens=2).create_engine_config())
+    rng = np.random.default_rng(17)
+
+    async def client(i):
+        plen = int(rng.integers(4, 120))
+        p = SamplingParams(
+            temperature=float(rng.choice([0.0, 1.0])),
+            seed=int(rng.integers(0, 2**31)),
+            max_tokens=int(rng.integers(5, 50)), ignore_eos=True,
+            priority=int(rng.integers(0, 3)))
+        cancel_at = int(rng.integers(0, 8)) if i % 3 == 0 else 0
+        prompt = [int(x) for x in rng.integers(3, 900, size=plen)]
+        gen = llm.generate(prompt, p, f"c{i}")
+        got = 0
+        try:
+            async for out in gen:
+                got = len(out.outputs[0].token_ids)
+                if cancel_at and got >= cancel_at:
+                    return
+            assert got == p.max_tokens, (i, got, p.max_tokens)
+        finally:
+            await gen.aclose()
+
+    async def drive():
+        await asyncio.gather(*[client(i) for i in range(16)])
+        for _ in range(200):
+            if not llm.engine.has_unfinished_requests():
+                break
+            await asyncio.sleep(0.05)
+        assert not llm.engine.has_unfinished_requests()
+        mgr = llm.engine.engine_core.scheduler.kv_cache_manager
+        assert not mgr.req_to_blocks, list(mgr.req_to_blocks)
+        assert llm.engine.engine_core.scheduler \
+            .num_preemptions_total > 0
+
+    try:
+        asyncio.run(drive())
+    finally:
+        llm.shutdown()
