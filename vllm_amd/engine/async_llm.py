@@ -217,6 +217,7 @@ class AsyncLLM:
             "prefix_cache_queries": sched.prefix_cache_queries,
             "prefix_cache_hits": sched.prefix_cache_hits,
             "num_preemptions": sched.num_preemptions_total,
+            "num_encoder_deferrals": sched.num_encoder_deferrals,
             "spec_tokens_drafted": sched.spec_stats_drafted,
             "spec_tokens_accepted": sched.spec_stats_accepted,
         }

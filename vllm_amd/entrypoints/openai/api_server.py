@@ -382,6 +382,9 @@ def build_app(state: ServerState) -> FastAPI:
             f"vllm_amd:prefix_cache_hits_total {s.get('prefix_cache_hits', 0)}",
             "# TYPE vllm_amd:num_preemptions_total counter",
             f"vllm_amd:num_preemptions_total {s.get('num_preemptions', 0)}",
+            "# TYPE vllm_amd:encoder_deferrals_total counter",
+            "vllm_amd:encoder_deferrals_total "
+            f"{s.get('num_encoder_deferrals', 0)}",
             "# TYPE vllm_amd:spec_decode_num_draft_tokens_total counter",
             "vllm_amd:spec_decode_num_draft_tokens_total "
             f"{s.get('spec_tokens_drafted', 0)}",
