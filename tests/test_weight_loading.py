@@ -261,3 +261,49 @@ def test_fp8_block_checkpoint_dequant(tmp_path):
     assert torch.allclose(got["model.layers.0.mlp.w.weight"], expect)
     assert torch.equal(got["model.norm.weight"], torch.ones(8))
     assert "model.layers.0.mlp.w.weight_scale_inv" not in got
+
+
+def test_dummy_shard_matches_loaders_at_high_tp(monkeypatch):
+    """dummy_shard must produce EXACTLY what the checkpoint loaders
+    produce from the same full tensors, for every rank at tp=4 and
+    tp=8 (incl. KV-head replication) — the gloo tests only cover tp=2,
+    but the driver's scale run shards at tp=8."""
+    import torch
+
+    import vllm_amd.layers.linear as L
+
+    hidden, head_dim, heads, kv_heads = 64, 8, 8, 2
+    torch.manual_seed(0)
+    q_w = torch.randn(heads * head_dim, hidden)
+    k_w = torch.randn(kv_heads * head_dim, hidden)
+    v_w = torch.randn(kv_heads * head_dim, hidden)
+    full = torch.cat([q_w, k_w, v_w], dim=0)
+    row_w = torch.randn(16, hidden)
+    merged_a = torch.randn(32, hidden)
+    merged_b = torch.randn(32, hidden)
+    merged_full = torch.cat([merged_a, merged_b], dim=0)
+
+    for tp in (4, 8):
+        for rank in range(tp):
+            monkeypatch.setattr(L, "get_tp_world_size", lambda: tp)
+            monkeypatch.setattr(L, "get_tp_rank", lambda: rank)
+            qkv = L.QKVParallelLinear(hidden, head_dim, heads, kv_heads,
+                                      dtype=torch.float32)
+            qkv.load_qkv(q_w, k_w, v_w)
+            via_loader = qkv.weight.data.clone()
+            via_dummy = qkv.dummy_shard("weight", full)
+            assert torch.equal(via_loader, via_dummy), (tp, rank, "qkv")
+
+            row = L.RowParallelLinear(hidden, 16, dtype=torch.float32)
+            row.load_weight(row_w)
+            assert torch.equal(row.weight.data,
+                               row.dummy_shard("weight", row_w)), \
+                (tp, rank, "row")
+
+            mg = L.MergedColumnParallelLinear(hidden, [32, 32],
+                                              dtype=torch.float32)
+            mg.load_sub_weight(0, merged_a)
+            mg.load_sub_weight(1, merged_b)
+            assert torch.equal(mg.weight.data,
+                               mg.dummy_shard("weight", merged_full)), \
+                (tp, rank, "merged")
