@@ -816,3 +816,24 @@ def test_model_retrieval_and_ping(client):
     assert client.get("/v1/models/nope").status_code == 404
     assert client.get("/ping").status_code == 200
     assert client.post("/ping").status_code == 200
+
+
+def test_chat_stream_logprobs(client):
+    toks = []
+    with client.stream("POST", "/v1/chat/completions", json={
+        "model": "tiny-llama",
+        "messages": [{"role": "user", "content": "hi"}],
+        "max_tokens": 4, "temperature": 0.0, "ignore_eos": True,
+        "stream": True, "logprobs": True, "top_logprobs": 1,
+    }) as r:
+        assert r.status_code == 200
+        for ln in r.iter_lines():
+            if ln.startswith("data: ") and ln != "data: [DONE]":
+                ch = json.loads(ln[6:])["choices"][0]
+                lp = ch.get("logprobs")
+                if lp:
+                    toks += lp["content"]
+    assert len(toks) == 4
+    for t in toks:
+        assert isinstance(t["logprob"], float)
+        assert t["top_logprobs"]

@@ -410,13 +410,14 @@ def build_app(state: ServerState) -> FastAPI:
         return {"tokens": tokens, "token_logprobs": token_logprobs,
                 "top_logprobs": top, "text_offset": []}
 
-    def _chat_logprobs(comp):
-        """OpenAI chat logprobs object ({"content": [...]})."""
+    def _chat_logprobs(comp, lo: int = 0, hi: Optional[int] = None):
+        """OpenAI chat logprobs object ({"content": [...]}); lo/hi
+        slice the (cumulative) token list for streaming chunks."""
         if not comp.logprobs:
             return None
         tok = engine.tokenizer
         content = []
-        for tid, d in zip(comp.token_ids, comp.logprobs):
+        for tid, d in zip(comp.token_ids[lo:hi], comp.logprobs[lo:hi]):
             tstr = tok.decode([tid])
             content.append({
                 "token": tstr,
@@ -1169,6 +1170,7 @@ def build_app(state: ServerState) -> FastAPI:
                     async for b, out in _merge_streams(gens):
                         st = states[b]
                         comp = out.outputs[0]
+                        prev_seen = st.seen_toks
                         new = len(comp.token_ids) - st.seen_toks
                         st.seen_toks = len(comp.token_ids)
                         state.num_generation_tokens += new
@@ -1180,6 +1182,19 @@ def build_app(state: ServerState) -> FastAPI:
                         if finish and (saw or named):
                             finish = "tool_calls" if finish == "stop" \
                                 else finish
+                        # Streamed logprobs for plain content (no tool/
+                        # reasoning parsers buffering text — there the
+                        # token<->delta alignment is undefined, same
+                        # limitation as the reference).
+                        lp_chunk = None
+                        if (req.logprobs and new and st.rparse is None
+                                and st.tparse is None and not named):
+                            lp_chunk = _chat_logprobs(
+                                comp, prev_seen, prev_seen + new)
+                        if lp_chunk is not None and not msgs:
+                            # Tokens that detokenize to nothing (byte
+                            # fragments) still stream their logprobs.
+                            msgs = [DeltaMessage()]
                         for i, d in enumerate(msgs):
                             last = comp.finish_reason and i == len(msgs) - 1
                             chunk = ChatCompletionStreamResponse(
@@ -1187,8 +1202,10 @@ def build_app(state: ServerState) -> FastAPI:
                                 choices=[ChatStreamChoice(
                                     index=b, delta=d,
                                     finish_reason=finish if last else None,
+                                    logprobs=lp_chunk,
                                 )],
                             )
+                            lp_chunk = None  # attach once
                             yield f"data: {chunk.model_dump_json()}\n\n"
                         if comp.finish_reason and not msgs:
                             chunk = ChatCompletionStreamResponse(

@@ -301,6 +301,7 @@ class ChatStreamChoice(BaseModel):
     index: int
     delta: DeltaMessage
     finish_reason: Optional[str] = None
+    logprobs: Optional[dict[str, Any]] = None
 
 
 class ChatCompletionStreamResponse(BaseModel):
