@@ -837,3 +837,19 @@ def test_chat_stream_logprobs(client):
     for t in toks:
         assert isinstance(t["logprob"], float)
         assert t["top_logprobs"]
+
+
+def test_completions_stream_logprobs(client):
+    got = 0
+    with client.stream("POST", "/v1/completions", json={
+        "model": "tiny-llama", "prompt": "hi", "max_tokens": 4,
+        "temperature": 0.0, "ignore_eos": True, "stream": True,
+        "logprobs": 1,
+    }) as r:
+        assert r.status_code == 200
+        for ln in r.iter_lines():
+            if ln.startswith("data: ") and ln != "data: [DONE]":
+                ch = json.loads(ln[6:])["choices"]
+                if ch and ch[0].get("logprobs"):
+                    got += len(ch[0]["logprobs"]["token_logprobs"])
+    assert got == 4

@@ -396,14 +396,15 @@ def build_app(state: ServerState) -> FastAPI:
         return Response("\n".join(lines) + "\n",
                         media_type="text/plain; version=0.0.4")
 
-    def _completion_logprobs(comp):
+    def _completion_logprobs(comp, lo: int = 0,
+                             hi: Optional[int] = None):
         """OpenAI completions logprobs object from the engine's raw
         per-token {token_id: logprob} dicts."""
         if not comp.logprobs:
             return None
         tok = engine.tokenizer
         tokens, token_logprobs, top = [], [], []
-        for tid, d in zip(comp.token_ids, comp.logprobs):
+        for tid, d in zip(comp.token_ids[lo:hi], comp.logprobs[lo:hi]):
             tokens.append(tok.decode([tid]))
             token_logprobs.append(d.get(tid))
             top.append({tok.decode([t]): lp for t, lp in d.items()})
@@ -539,6 +540,11 @@ def build_app(state: ServerState) -> FastAPI:
                         n_gen += new
                         n_prompt_by[p_i] = len(out.prompt_token_ids)
                         n_prompt = sum(n_prompt_by.values())
+                        lp_chunk = None
+                        if req.logprobs is not None and new:
+                            prev = seen[gi] - new
+                            lp_chunk = _completion_logprobs(
+                                comp, prev, prev + new)
                         chunk = {
                             "id": rid,
                             "object": "text_completion",
@@ -547,7 +553,7 @@ def build_app(state: ServerState) -> FastAPI:
                             "choices": [{
                                 "index": p_i * req.n + b,
                                 "text": comp.text,
-                                "logprobs": None,
+                                "logprobs": lp_chunk,
                                 "finish_reason": comp.finish_reason,
                             }],
                         }
