@@ -853,3 +853,16 @@ def test_completions_stream_logprobs(client):
                 if ch and ch[0].get("logprobs"):
                     got += len(ch[0]["logprobs"]["token_logprobs"])
     assert got == 4
+
+
+def test_echo_with_logprobs_covers_prompt(client):
+    r = client.post("/v1/completions", json={
+        "model": "tiny-llama", "prompt": [5, 6, 7], "max_tokens": 3,
+        "temperature": 0.0, "ignore_eos": True,
+        "echo": True, "logprobs": 1})
+    assert r.status_code == 200, r.text
+    lp = r.json()["choices"][0]["logprobs"]
+    # 3 prompt tokens (first with null logprob) + 3 generated.
+    assert len(lp["tokens"]) == 6
+    assert lp["token_logprobs"][0] is None
+    assert all(isinstance(v, float) for v in lp["token_logprobs"][1:])

@@ -501,6 +501,13 @@ def build_app(state: ServerState) -> FastAPI:
             params = req.to_sampling_params(req.stream)
         except ValueError as e:
             return _error(str(e))
+        echo_lp = bool(req.echo) and req.logprobs is not None \
+            and not req.stream
+        if echo_lp and params.prompt_logprobs is None:
+            # OpenAI echo+logprobs includes the prompt tokens' logprobs
+            # (first token null); request them engine-side.
+            import dataclasses as _dc
+            params = _dc.replace(params, prompt_logprobs=req.logprobs)
         forced_lp = False
         if best_of > req.n and params.logprobs is None:
             # Branch selection scores by chosen-token logprob; request it
@@ -641,6 +648,25 @@ def build_app(state: ServerState) -> FastAPI:
             prompt_tokens_details=({"cached_tokens": cached}
                                    if cached else None),
         )
+        def _echo_logprobs(final, comp):
+            tok = engine.tokenizer
+            ids = final.prompt_token_ids
+            tokens = [tok.decode([t]) for t in ids]
+            token_logprobs = [None]
+            top = [None]
+            plps = final.prompt_logprobs or []
+            for i, t in enumerate(ids[1:]):  # plps[i] -> prompt tok i+1
+                d = plps[i] if i < len(plps) else {}
+                token_logprobs.append(d.get(t))
+                top.append({tok.decode([k]): lp for k, lp in d.items()})
+            base = _completion_logprobs(comp) or {
+                "tokens": [], "token_logprobs": [], "top_logprobs": []}
+            return {"tokens": tokens + base["tokens"],
+                    "token_logprobs":
+                        token_logprobs + base["token_logprobs"],
+                    "top_logprobs": top + base["top_logprobs"],
+                    "text_offset": []}
+
         choices = []
         for out_i, final in enumerate(finals):
             comp = final.outputs[0]
@@ -649,8 +675,11 @@ def build_app(state: ServerState) -> FastAPI:
                 text=(final.prompt or "") + comp.text if req.echo
                 else comp.text,
                 logprobs=(None if forced_lp
+                          else _echo_logprobs(final, comp) if echo_lp
                           else _completion_logprobs(comp)),
-                prompt_logprobs=final.prompt_logprobs,
+                prompt_logprobs=(final.prompt_logprobs
+                                 if req.prompt_logprobs is not None
+                                 else None),
                 finish_reason=comp.finish_reason,
                 stop_reason=comp.stop_reason
                 if isinstance(comp.stop_reason, (int, str)) else None,
