@@ -866,3 +866,26 @@ def test_echo_with_logprobs_covers_prompt(client):
     assert len(lp["tokens"]) == 6
     assert lp["token_logprobs"][0] is None
     assert all(isinstance(v, float) for v in lp["token_logprobs"][1:])
+
+
+def test_include_stop_str_in_output(client):
+    # Find a character the greedy continuation actually produces.
+    base = client.post("/v1/completions", json={
+        "model": "tiny-llama", "prompt": "hello there friend",
+        "max_tokens": 12, "temperature": 0.0, "ignore_eos": True,
+    }).json()["choices"][0]["text"]
+    printable = [ch for ch in base if ch.isprintable() and ch != " "]
+    if not printable:
+        import pytest
+        pytest.skip("tiny model produced no printable text")
+    stop = printable[len(printable) // 2]
+    common = {"model": "tiny-llama", "prompt": "hello there friend",
+              "max_tokens": 12, "temperature": 0.0, "ignore_eos": True,
+              "stop": [stop]}
+    r1 = client.post("/v1/completions", json=common).json()["choices"][0]
+    r2 = client.post("/v1/completions", json={
+        **common, "include_stop_str_in_output": True,
+    }).json()["choices"][0]
+    assert r1["finish_reason"] == "stop"
+    assert stop not in r1["text"]
+    assert r2["text"].endswith(stop)

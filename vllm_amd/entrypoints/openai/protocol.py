@@ -50,6 +50,7 @@ class CompletionRequest(BaseModel):
     bad_words: Optional[list[str]] = None
     allowed_token_ids: Optional[list[int]] = None
     truncate_prompt_tokens: Optional[int] = None
+    include_stop_str_in_output: bool = False
     priority: int = 0  # lower = sooner (priority policy)
     # Extension (same name as the reference): per-prompt-token logprobs.
     prompt_logprobs: Optional[int] = None
@@ -85,6 +86,7 @@ class CompletionRequest(BaseModel):
             bad_words=self.bad_words,
             allowed_token_ids=self.allowed_token_ids,
             truncate_prompt_tokens=self.truncate_prompt_tokens,
+            include_stop_str_in_output=self.include_stop_str_in_output,
             priority=self.priority,
             output_kind=(RequestOutputKind.DELTA if stream
                          else RequestOutputKind.FINAL_ONLY),
@@ -166,6 +168,7 @@ class ChatCompletionRequest(BaseModel):
     bad_words: Optional[list[str]] = None
     allowed_token_ids: Optional[list[int]] = None
     truncate_prompt_tokens: Optional[int] = None
+    include_stop_str_in_output: bool = False
     priority: int = 0  # lower = sooner (priority policy)
     response_format: Optional[dict[str, Any]] = None
     add_generation_prompt: bool = True
@@ -234,6 +237,7 @@ class ChatCompletionRequest(BaseModel):
             bad_words=self.bad_words,
             allowed_token_ids=self.allowed_token_ids,
             truncate_prompt_tokens=self.truncate_prompt_tokens,
+            include_stop_str_in_output=self.include_stop_str_in_output,
             priority=self.priority,
             guided_json_object=json_object,
             output_kind=(RequestOutputKind.DELTA if stream
