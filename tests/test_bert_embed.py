@@ -151,3 +151,30 @@ def test_score_and_rerank_on_encoder():
             assert results[0]["relevance_score"] > 0.999
     finally:
         state.engine.shutdown()
+
+
+def test_embeddings_dimensions_truncation():
+    from fastapi.testclient import TestClient
+
+    from vllm_amd.engine.arg_utils import EngineArgs
+    from vllm_amd.entrypoints.openai.api_server import make_server
+
+    args = EngineArgs(model="tiny-bert", dtype="fp32", device="cpu",
+                      block_size=16, num_gpu_blocks=64, max_model_len=256,
+                      max_num_batched_tokens=128, max_num_seqs=4)
+    app, state = make_server(args, served_model_name="tiny-bert")
+    try:
+        with TestClient(app) as c:
+            r = c.post("/v1/embeddings", json={
+                "model": "tiny-bert", "input": "hello",
+                "dimensions": 16})
+            assert r.status_code == 200, r.text
+            vec = r.json()["data"][0]["embedding"]
+            assert len(vec) == 16
+            norm = sum(v * v for v in vec) ** 0.5
+            assert abs(norm - 1.0) < 1e-4  # re-normalized
+            bad = c.post("/v1/embeddings", json={
+                "model": "tiny-bert", "input": "x", "dimensions": 0})
+            assert bad.status_code == 400
+    finally:
+        state.engine.shutdown()

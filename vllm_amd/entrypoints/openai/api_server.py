@@ -978,11 +978,18 @@ def build_app(state: ServerState) -> FastAPI:
             finals = await asyncio.gather(*(one(p) for p in inputs))
         except ValueError as e:
             return _error(str(e))
+        if req.dimensions is not None and req.dimensions < 1:
+            return _error("dimensions must be >= 1")
         data = []
         n_prompt = 0
         for i, final in enumerate(finals):
             n_prompt += len(final.prompt_token_ids)
             vec = final.pooled or []
+            if req.dimensions is not None and vec:
+                vec = vec[:req.dimensions]
+                norm = sum(v * v for v in vec) ** 0.5
+                if norm > 0:
+                    vec = [v / norm for v in vec]
             if req.encoding_format == "base64":
                 emb = base64.b64encode(
                     struct.pack(f"<{len(vec)}f", *vec)).decode()

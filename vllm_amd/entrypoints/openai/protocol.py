@@ -321,6 +321,9 @@ class EmbeddingRequest(BaseModel):
     model: str
     input: Union[str, list[str], list[int], list[list[int]]]
     encoding_format: Literal["float", "base64"] = "float"
+    # Matryoshka-style truncation: keep the first N dimensions and
+    # re-normalize (OpenAI `dimensions` semantics).
+    dimensions: Optional[int] = None
     user: Optional[str] = None
     # Extension: pooling strategy (reference pools per model config).
     pooling: Literal["last", "mean"] = "last"
