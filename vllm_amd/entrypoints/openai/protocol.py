@@ -51,6 +51,7 @@ class CompletionRequest(BaseModel):
     allowed_token_ids: Optional[list[int]] = None
     truncate_prompt_tokens: Optional[int] = None
     include_stop_str_in_output: bool = False
+    spaces_between_special_tokens: bool = True
     priority: int = 0  # lower = sooner (priority policy)
     # Extension (same name as the reference): per-prompt-token logprobs.
     prompt_logprobs: Optional[int] = None
@@ -87,6 +88,7 @@ class CompletionRequest(BaseModel):
             allowed_token_ids=self.allowed_token_ids,
             truncate_prompt_tokens=self.truncate_prompt_tokens,
             include_stop_str_in_output=self.include_stop_str_in_output,
+            spaces_between_special_tokens=self.spaces_between_special_tokens,
             priority=self.priority,
             output_kind=(RequestOutputKind.DELTA if stream
                          else RequestOutputKind.FINAL_ONLY),
@@ -169,6 +171,7 @@ class ChatCompletionRequest(BaseModel):
     allowed_token_ids: Optional[list[int]] = None
     truncate_prompt_tokens: Optional[int] = None
     include_stop_str_in_output: bool = False
+    spaces_between_special_tokens: bool = True
     priority: int = 0  # lower = sooner (priority policy)
     response_format: Optional[dict[str, Any]] = None
     add_generation_prompt: bool = True
@@ -238,6 +241,7 @@ class ChatCompletionRequest(BaseModel):
             allowed_token_ids=self.allowed_token_ids,
             truncate_prompt_tokens=self.truncate_prompt_tokens,
             include_stop_str_in_output=self.include_stop_str_in_output,
+            spaces_between_special_tokens=self.spaces_between_special_tokens,
             priority=self.priority,
             guided_json_object=json_object,
             output_kind=(RequestOutputKind.DELTA if stream
